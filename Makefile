@@ -1,0 +1,62 @@
+# ucc_amd build: everything compiled by hipcc for gfx950, objects shared
+# between the standalone C library and the python extension. In-tree .so
+# outputs (they travel to the GPU box with the repo snapshot).
+
+HIPCC    ?= hipcc
+ARCH     ?= gfx950
+PYTHON   ?= python3
+PY_EXT   := $(shell $(PYTHON) -c "import sysconfig;print(sysconfig.get_config_var('EXT_SUFFIX'))")
+PY_INC   := $(shell $(PYTHON) -m pybind11 --includes)
+
+CXXFLAGS := --offload-arch=$(ARCH) -O3 -std=c++17 -fPIC -Wall -Wextra \
+            -Wno-unused-parameter -DUCC_AMD_HAS_HIP
+LDFLAGS  := -shared -fPIC
+
+BUILD := build
+
+# library sources (no python dependency)
+LIB_SRCS := $(wildcard src/utils/*.cc) \
+            $(wildcard src/core/*.cc) \
+            $(wildcard src/schedule/*.cc) \
+            $(wildcard src/coll_score/*.cc) \
+            $(wildcard src/coll_patterns/*.cc) \
+            $(wildcard src/mc/*.cc) \
+            $(wildcard src/ec/*.cc) \
+            $(wildcard src/topo/*.cc) \
+            $(wildcard src/tl/self/*.cc) \
+            $(wildcard src/tl/shm/*.cc) \
+            $(wildcard src/tl/cdna4/*.cc) \
+            $(wildcard src/tl/rccl/*.cc)
+KERNEL_SRCS := $(wildcard src/ec/kernels/*.hip) $(wildcard src/tl/cdna4/kernels/*.hip)
+
+LIB_OBJS := $(patsubst %.cc,$(BUILD)/%.o,$(LIB_SRCS)) \
+            $(patsubst %.hip,$(BUILD)/%.o,$(KERNEL_SRCS))
+BIND_OBJ := $(BUILD)/src/bind/module.o
+
+MODULE := ucc_amd/_core$(PY_EXT)
+SHLIB  := $(BUILD)/libucc_amd.so
+
+all: $(MODULE) $(SHLIB)
+
+$(BUILD)/%.o: %.cc
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+$(BUILD)/%.o: %.hip
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+$(BIND_OBJ): src/bind/module.cc
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) $(PY_INC) -c $< -o $@
+
+$(MODULE): $(LIB_OBJS) $(BIND_OBJ)
+	$(HIPCC) $(LDFLAGS) $^ -o $@ -lrt
+
+$(SHLIB): $(LIB_OBJS)
+	$(HIPCC) $(LDFLAGS) $^ -o $@ -lrt
+
+clean:
+	rm -rf $(BUILD) ucc_amd/_core*.so
+
+.PHONY: all clean

@@ -1,0 +1,618 @@
+/* Public API implementation: lib/context/team/collective lifecycle.
+ * Parity targets: reference core/ucc_lib.c, core/ucc_context.c,
+ * core/ucc_team.c (nonblocking create state machine), core/ucc_coll.c
+ * (score-map dispatch, zero-size fast path, persistent re-post). */
+#include "core.h"
+
+#include <cstdlib>
+#include <random>
+#include <unistd.h>
+
+namespace ucc {
+
+/* ------------------------------------------------------------- ProcInfo */
+static uint64_t str_hash(const char *s)
+{
+    uint64_t h = 1469598103934665603ull;
+    while (*s) {
+        h = (h ^ (uint8_t)*s++) * 1099511628211ull;
+    }
+    return h;
+}
+
+ProcInfo local_proc_info()
+{
+    ProcInfo pi;
+    char     host[256] = {0};
+    gethostname(host, sizeof(host) - 1);
+    pi.host_hash = str_hash(host);
+    pi.pid       = (int32_t)getpid();
+    pi.device    = -1;
+    return pi;
+}
+
+bool Team::all_same_node() const
+{
+    for (const auto &p : procs) {
+        if (p.host_hash != procs[0].host_hash) {
+            return false;
+        }
+    }
+    return true;
+}
+
+bool Team::all_have_device() const
+{
+    for (const auto &p : procs) {
+        if (p.device < 0) {
+            return false;
+        }
+    }
+    return true;
+}
+
+/* ------------------------------------------------------------- OobRound */
+ucc_status_t OobRound::start(const void *src, size_t size)
+{
+    send_.assign((const uint8_t *)src, (const uint8_t *)src + size);
+    recv_.resize(size * oob_.n_oob_eps);
+    ucc_status_t st = oob_.allgather(send_.data(), recv_.data(), size,
+                                     oob_.coll_info, &req_);
+    if (st != UCC_OK) {
+        return st;
+    }
+    active_ = true;
+    return UCC_OK;
+}
+
+ucc_status_t OobRound::test()
+{
+    if (!active_) {
+        return UCC_ERR_INVALID_PARAM;
+    }
+    ucc_status_t st = oob_.req_test(req_);
+    if (st == UCC_INPROGRESS) {
+        return st;
+    }
+    oob_.req_free(req_);
+    req_    = nullptr;
+    active_ = false;
+    return st;
+}
+
+/* ------------------------------------------------------------ registry  */
+std::vector<Tl *> &tl_registry()
+{
+    static std::vector<Tl *> reg;
+    return reg;
+}
+
+void register_tl(Tl *tl) { tl_registry().push_back(tl); }
+
+} // namespace ucc
+
+using namespace ucc;
+
+/* ===================================================== lib object (C API) */
+struct ucc_lib_config {
+    std::string prefix;
+};
+struct ucc_context_config {
+    Lib *lib;
+};
+
+extern "C" {
+
+ucc_status_t ucc_lib_config_read(const char *env_prefix, const char *filename,
+                                 ucc_lib_config_h *config)
+{
+    (void)filename;
+    auto *c   = new ucc_lib_config;
+    c->prefix = env_prefix ? env_prefix : "";
+    *config   = c;
+    return UCC_OK;
+}
+
+void ucc_lib_config_release(ucc_lib_config_h config) { delete config; }
+
+ucc_status_t ucc_lib_config_modify(ucc_lib_config_h config, const char *name,
+                                   const char *value)
+{
+    (void)config;
+    Config::instance().set("", name, value);
+    return UCC_OK;
+}
+
+void ucc_lib_config_print(const ucc_lib_config_h config, void *stream,
+                          const char *title, int print_flags)
+{
+    (void)config;
+    (void)print_flags;
+    FILE *f = stream ? (FILE *)stream : stdout;
+    fprintf(f, "# %s\n", title ? title : "ucc_amd config");
+    for (const auto &e : Config::instance().entries()) {
+        fprintf(f, "%s=%s  # %s\n",
+                Config::key(e.component, e.name).c_str(), e.dflt.c_str(),
+                e.doc.c_str());
+    }
+}
+
+ucc_status_t ucc_init_version(unsigned api_major, unsigned api_minor,
+                              const ucc_lib_params_t *params,
+                              const ucc_lib_config_h  config,
+                              ucc_lib_h *lib_p)
+{
+    (void)config;
+    if (api_major != UCC_API_MAJOR || api_minor > UCC_API_MINOR) {
+        ucc_warn("requested api %u.%u, library is %u.%u", api_major, api_minor,
+                 UCC_API_MAJOR, UCC_API_MINOR);
+    }
+    ensure_builtin_tls();
+    auto *lib = new Lib;
+    if (params) {
+        lib->params = *params;
+        if (params->mask & UCC_LIB_PARAM_FIELD_THREAD_MODE) {
+            lib->thread_mode = params->thread_mode;
+        }
+    }
+    *lib_p = reinterpret_cast<ucc_lib_h>(lib);
+    return UCC_OK;
+}
+
+ucc_status_t ucc_finalize(ucc_lib_h lib)
+{
+    delete reinterpret_cast<Lib *>(lib);
+    return UCC_OK;
+}
+
+ucc_status_t ucc_lib_get_attr(ucc_lib_h lib, ucc_lib_attr_t *attr)
+{
+    auto *l = reinterpret_cast<Lib *>(lib);
+    if (attr->mask & UCC_LIB_ATTR_FIELD_THREAD_MODE) {
+        attr->thread_mode = l->thread_mode;
+    }
+    if (attr->mask & UCC_LIB_ATTR_FIELD_COLL_TYPES) {
+        attr->coll_types = UCC_COLL_TYPE_ALL;
+    }
+    return UCC_OK;
+}
+
+/* ---------------------------------------------------------------- context */
+ucc_status_t ucc_context_config_read(ucc_lib_h lib, const char *filename,
+                                     ucc_context_config_h *config)
+{
+    (void)filename;
+    auto *c = new ucc_context_config;
+    c->lib  = reinterpret_cast<Lib *>(lib);
+    *config = c;
+    return UCC_OK;
+}
+
+void ucc_context_config_release(ucc_context_config_h config) { delete config; }
+
+ucc_status_t ucc_context_config_modify(ucc_context_config_h config,
+                                       const char *component,
+                                       const char *name, const char *value)
+{
+    (void)config;
+    Config::instance().set(component ? component : "", name, value);
+    return UCC_OK;
+}
+
+void ucc_context_config_print(const ucc_context_config_h config, void *stream,
+                              const char *title, int print_flags)
+{
+    ucc_lib_config_print(nullptr, stream, title, print_flags);
+    (void)config;
+}
+
+ucc_status_t ucc_context_create(ucc_lib_h lib_h,
+                                const ucc_context_params_t *params,
+                                const ucc_context_config_h  config,
+                                ucc_context_h *context)
+{
+    (void)config;
+    auto *lib = reinterpret_cast<Lib *>(lib_h);
+    auto *ctx = new Context;
+    ctx->lib  = lib;
+    if (params) {
+        ctx->params = *params;
+        ctx->has_oob = params->mask & UCC_CONTEXT_PARAM_FIELD_OOB;
+    }
+    ctx->mt   = lib->thread_mode == UCC_THREAD_MULTIPLE;
+    ctx->seq  = lib->next_ctx_seq++;
+    ctx->proc = local_proc_info();
+    ctx->proc.ctx_seq = ((uint64_t)ctx->proc.pid << 20) | ctx->seq;
+    for (Tl *tl : tl_registry()) {
+        TlContext *tlc = tl->context_create(ctx);
+        if (tlc) {
+            ctx->tl_ctxs.emplace_back(tlc);
+            ucc_debug("context %p: tl %s available", (void *)ctx, tl->name());
+        }
+    }
+    *context = reinterpret_cast<ucc_context_h>(ctx);
+    return UCC_OK;
+}
+
+ucc_status_t ucc_context_destroy(ucc_context_h context)
+{
+    delete reinterpret_cast<Context *>(context);
+    return UCC_OK;
+}
+
+ucc_status_t ucc_context_get_attr(ucc_context_h context,
+                                  ucc_context_attr_t *attr)
+{
+    (void)context;
+    if (attr->mask & UCC_CONTEXT_ATTR_FIELD_CTX_ADDR_LEN) {
+        attr->ctx_addr_len = 0;
+    }
+    return UCC_OK;
+}
+
+ucc_status_t ucc_context_progress(ucc_context_h context)
+{
+    return reinterpret_cast<Context *>(context)->progress();
+}
+
+/* ------------------------------------------------------------------ team */
+ucc_status_t ucc_team_create_post(ucc_context_h *contexts,
+                                  uint32_t num_contexts,
+                                  const ucc_team_params_t *team_params,
+                                  ucc_team_h *new_team)
+{
+    if (num_contexts != 1 || !contexts || !team_params) {
+        return UCC_ERR_NOT_SUPPORTED; /* one context per process */
+    }
+    auto *ctx  = reinterpret_cast<Context *>(contexts[0]);
+    auto *team = new Team;
+    team->ctx    = ctx;
+    team->params = *team_params;
+    team->has_oob = team_params->mask & UCC_TEAM_PARAM_FIELD_OOB;
+    if (team->has_oob) {
+        team->oob  = team_params->oob;
+        team->rank = team->oob.oob_ep;
+        team->size = team->oob.n_oob_eps;
+    } else if ((team_params->mask & UCC_TEAM_PARAM_FIELD_TEAM_SIZE) &&
+               team_params->team_size == 1) {
+        team->rank = 0;
+        team->size = 1;
+    } else {
+        delete team;
+        return UCC_ERR_INVALID_PARAM; /* OOB required for size > 1 */
+    }
+    team->id = (team_params->mask & UCC_TEAM_PARAM_FIELD_ID)
+                   ? team_params->id
+                   : (uint16_t)(ctx->next_team_id++);
+
+    if (team->size == 1) {
+        team->procs.assign(1, ctx->proc);
+        team->team_uid = ctx->proc.ctx_seq;
+        team->state    = Team::TL_CREATE;
+        /* create TL teams immediately (self at least) */
+        for (auto &tlc : ctx->tl_ctxs) {
+            TlTeam *tt = tlc->iface()->team_create(tlc.get(), team);
+            if (tt) {
+                team->tl_teams.emplace_back(tt);
+            }
+        }
+    } else {
+        /* round 1: proc info + rank0's uid proposal */
+        struct R1 {
+            ProcInfo pi;
+            uint64_t uid;
+        } r1;
+        r1.pi = ctx->proc;
+        std::random_device rd;
+        r1.uid = ((uint64_t)rd() << 32) ^ rd() ^
+                 ((uint64_t)ctx->proc.pid << 16) ^ ctx->seq;
+        team->oobr.init(team->oob);
+        ucc_status_t st = team->oobr.start(&r1, sizeof(r1));
+        if (st != UCC_OK) {
+            delete team;
+            return st;
+        }
+        team->state = Team::ADDR_EXCHANGE;
+    }
+    *new_team = reinterpret_cast<ucc_team_h>(team);
+    return UCC_OK;
+}
+
+ucc_status_t ucc_team_create_test(ucc_team_h team_h)
+{
+    auto *team = reinterpret_cast<Team *>(team_h);
+    auto *ctx  = team->ctx;
+
+    switch (team->state) {
+    case Team::ADDR_EXCHANGE: {
+        ucc_status_t st = team->oobr.test();
+        if (st == UCC_INPROGRESS) {
+            return UCC_INPROGRESS;
+        }
+        if (st != UCC_OK) {
+            team->state = Team::FAILED;
+            return (team->err = st);
+        }
+        struct R1 {
+            ProcInfo pi;
+            uint64_t uid;
+        };
+        const R1 *all = (const R1 *)team->oobr.data();
+        team->procs.resize(team->size);
+        for (uint32_t i = 0; i < team->size; i++) {
+            team->procs[i] = all[i].pi;
+        }
+        team->team_uid = all[0].uid;
+        /* create TL team objects and the combined exchange blob */
+        for (auto &tlc : ctx->tl_ctxs) {
+            TlTeam *tt = tlc->iface()->team_create(tlc.get(), team);
+            if (tt) {
+                team->tl_teams.emplace_back(tt);
+            }
+        }
+        size_t off = 0;
+        team->exchg_off.clear();
+        for (auto &tt : team->tl_teams) {
+            team->exchg_off.push_back(off);
+            off += tt->exchg_size();
+        }
+        team->exchg_stride = off;
+        if (off == 0) {
+            team->state = Team::TL_CREATE;
+            return ucc_team_create_test(team_h);
+        }
+        team->exchg_buf.assign(off, 0);
+        for (size_t i = 0; i < team->tl_teams.size(); i++) {
+            team->tl_teams[i]->exchg_pack(team->exchg_buf.data() +
+                                          team->exchg_off[i]);
+        }
+        ucc_status_t st2 =
+            team->oobr.start(team->exchg_buf.data(), team->exchg_stride);
+        if (st2 != UCC_OK) {
+            team->state = Team::FAILED;
+            return (team->err = st2);
+        }
+        team->state = Team::TL_EXCHANGE;
+        return UCC_INPROGRESS;
+    }
+    case Team::TL_EXCHANGE: {
+        ucc_status_t st = team->oobr.test();
+        if (st == UCC_INPROGRESS) {
+            return UCC_INPROGRESS;
+        }
+        if (st != UCC_OK) {
+            team->state = Team::FAILED;
+            return (team->err = st);
+        }
+        const uint8_t *all = (const uint8_t *)team->oobr.data();
+        for (size_t i = 0; i < team->tl_teams.size();) {
+            ucc_status_t us = team->tl_teams[i]->exchg_unpack(
+                all + team->exchg_off[i], team->exchg_stride);
+            if (us != UCC_OK) {
+                ucc_warn("tl %s team exchange failed (%d), dropping",
+                         team->tl_teams[i]->tlc_->iface()->name(), us);
+                team->tl_teams.erase(team->tl_teams.begin() + i);
+                team->exchg_off.erase(team->exchg_off.begin() + i);
+            } else {
+                i++;
+            }
+        }
+        team->state = Team::TL_CREATE;
+        return ucc_team_create_test(team_h);
+    }
+    case Team::TL_CREATE: {
+        bool all_done = true;
+        for (size_t i = 0; i < team->tl_teams.size();) {
+            ucc_status_t st = team->tl_teams[i]->create_test();
+            if (st == UCC_INPROGRESS) {
+                all_done = false;
+                i++;
+            } else if (st != UCC_OK) {
+                ucc_warn("tl %s team create failed (%d), dropping",
+                         team->tl_teams[i]->tlc_->iface()->name(), st);
+                team->tl_teams.erase(team->tl_teams.begin() + i);
+            } else {
+                i++;
+            }
+        }
+        if (!all_done) {
+            return UCC_INPROGRESS;
+        }
+        if (team->tl_teams.empty()) {
+            team->state = Team::FAILED;
+            return (team->err = UCC_ERR_NO_RESOURCE);
+        }
+        for (auto &tt : team->tl_teams) {
+            tt->get_scores(team, team->score_map);
+        }
+        std::string tune = Config::instance().get("", "TUNE", "");
+        if (!tune.empty()) {
+            team->score_map.apply_str(tune);
+        }
+        if (log_level() >= LogLevel::INFO) {
+            ucc_info("team %u size %u score map:\n%s", team->id, team->size,
+                     team->score_map.to_string().c_str());
+        }
+        team->state = Team::ACTIVE;
+        return UCC_OK;
+    }
+    case Team::ACTIVE: return UCC_OK;
+    case Team::FAILED: return team->err;
+    }
+    return UCC_ERR_INVALID_PARAM;
+}
+
+ucc_status_t ucc_team_destroy(ucc_team_h team_h)
+{
+    delete reinterpret_cast<Team *>(team_h);
+    return UCC_OK;
+}
+
+ucc_status_t ucc_team_get_attr(ucc_team_h team_h, ucc_team_attr_t *attr)
+{
+    auto *team = reinterpret_cast<Team *>(team_h);
+    if (attr->mask & UCC_TEAM_ATTR_FIELD_EP) {
+        attr->ep = team->rank;
+    }
+    if (attr->mask & UCC_TEAM_ATTR_FIELD_SIZE) {
+        attr->size = team->size;
+    }
+    if (attr->mask & UCC_TEAM_ATTR_FIELD_EP_RANGE) {
+        attr->ep_range = UCC_COLLECTIVE_EP_RANGE_CONTIG;
+    }
+    return UCC_OK;
+}
+
+ucc_status_t ucc_team_get_size(ucc_team_h team_h, uint32_t *size)
+{
+    *size = reinterpret_cast<Team *>(team_h)->size;
+    return UCC_OK;
+}
+
+ucc_status_t ucc_team_get_my_ep(ucc_team_h team_h, uint64_t *ep)
+{
+    *ep = reinterpret_cast<Team *>(team_h)->rank;
+    return UCC_OK;
+}
+
+ucc_status_t ucc_team_get_all_eps(ucc_team_h team_h, uint64_t **ep,
+                                  uint64_t *num_eps)
+{
+    (void)team_h; (void)ep; (void)num_eps;
+    return UCC_ERR_NOT_IMPLEMENTED;
+}
+
+/* ------------------------------------------------------------ collective */
+namespace {
+class StubTask final : public Task {
+  public:
+    using Task::Task;
+    ucc_status_t post() override { return UCC_OK; }
+};
+} // namespace
+
+ucc_status_t ucc_collective_init(ucc_coll_args_t *coll_args,
+                                 ucc_coll_req_h *request, ucc_team_h team_h)
+{
+    auto *team = reinterpret_cast<Team *>(team_h);
+    if (team->state != Team::ACTIVE) {
+        return UCC_ERR_INVALID_PARAM;
+    }
+    auto  *req = new CollRequest;
+    req->args  = *coll_args;
+    req->team  = team;
+    req->persistent = coll_args->mask & UCC_COLL_ARGS_FIELD_FLAGS &&
+                      (coll_args->flags & UCC_COLL_ARGS_FLAG_PERSISTENT);
+    if (!(coll_args->mask & UCC_COLL_ARGS_FIELD_FLAGS)) {
+        req->args.flags = 0;
+    }
+    size_t msgsize = coll_args_msgsize(req->args, team->rank, team->size);
+    /* Zero-size fast path only where a local zero implies a global zero
+     * (fixed-count colls; v-variants may be locally empty but must still
+     * take part in the exchange). Reference: ucc_coll.c:191-208. */
+    const bool v_coll = req->args.coll_type == UCC_COLL_TYPE_ALLGATHERV ||
+                        req->args.coll_type == UCC_COLL_TYPE_ALLTOALLV ||
+                        req->args.coll_type == UCC_COLL_TYPE_GATHERV ||
+                        req->args.coll_type == UCC_COLL_TYPE_SCATTERV ||
+                        req->args.coll_type == UCC_COLL_TYPE_REDUCE_SCATTERV;
+    if (msgsize == 0 && !v_coll &&
+        req->args.coll_type != UCC_COLL_TYPE_BARRIER &&
+        req->args.coll_type != UCC_COLL_TYPE_FANIN &&
+        req->args.coll_type != UCC_COLL_TYPE_FANOUT) {
+        req->task = new StubTask(team->ctx);
+    } else {
+        ucc_status_t st =
+            team->score_map.init_coll(req->args, team, msgsize, &req->task);
+        if (st != UCC_OK) {
+            delete req;
+            return st;
+        }
+    }
+    req->task->req_ = req;
+    if ((req->args.mask & UCC_COLL_ARGS_FIELD_FLAGS) &&
+        (req->args.flags & UCC_COLL_ARGS_FLAG_TIMEOUT)) {
+        req->task->timeout = req->args.timeout;
+    }
+    req->super.status = UCC_OPERATION_INITIALIZED;
+    *request          = &req->super;
+    return UCC_OK;
+}
+
+ucc_status_t ucc_collective_post(ucc_coll_req_h request)
+{
+    auto *req = reinterpret_cast<CollRequest *>(request);
+    if (req->posted && req->task->status == UCC_INPROGRESS) {
+        return UCC_ERR_INVALID_PARAM; /* re-post of in-flight coll */
+    }
+    req->posted       = true;
+    req->seq          = req->team->coll_seq++;
+    req->super.status = UCC_INPROGRESS;
+    task_start(req->task);
+    if (req->task->status != UCC_INPROGRESS) {
+        req->super.status = req->task->status;
+    }
+    return req->task->status < 0 ? req->task->status : UCC_OK;
+}
+
+ucc_status_t ucc_collective_test(ucc_coll_req_h request)
+{
+    auto *req = reinterpret_cast<CollRequest *>(request);
+    return req->super.status;
+}
+
+ucc_status_t ucc_collective_finalize(ucc_coll_req_h request)
+{
+    auto *req = reinterpret_cast<CollRequest *>(request);
+    delete req->task;
+    delete req;
+    return UCC_OK;
+}
+
+/* ------------------------------------------------------------ EE (stubs) */
+ucc_status_t ucc_ee_create(ucc_team_h team, const ucc_ee_params_t *params,
+                           ucc_ee_h *ee)
+{
+    (void)team; (void)params; (void)ee;
+    return UCC_ERR_NOT_IMPLEMENTED;
+}
+ucc_status_t ucc_ee_destroy(ucc_ee_h ee) { (void)ee; return UCC_ERR_NOT_IMPLEMENTED; }
+ucc_status_t ucc_ee_get_event(ucc_ee_h ee, ucc_ev_t **ev)
+{
+    (void)ee; (void)ev;
+    return UCC_ERR_NOT_IMPLEMENTED;
+}
+ucc_status_t ucc_ee_ack_event(ucc_ee_h ee, ucc_ev_t *ev)
+{
+    (void)ee; (void)ev;
+    return UCC_ERR_NOT_IMPLEMENTED;
+}
+ucc_status_t ucc_ee_set_event(ucc_ee_h ee, ucc_ev_t *ev)
+{
+    (void)ee; (void)ev;
+    return UCC_ERR_NOT_IMPLEMENTED;
+}
+ucc_status_t ucc_ee_wait(ucc_ee_h ee, ucc_ev_t *ev)
+{
+    (void)ee; (void)ev;
+    return UCC_ERR_NOT_IMPLEMENTED;
+}
+ucc_status_t ucc_collective_triggered_post(ucc_ee_h ee, ucc_ev_t *ev)
+{
+    (void)ee; (void)ev;
+    return UCC_ERR_NOT_IMPLEMENTED;
+}
+
+ucc_status_t ucc_mem_map(ucc_context_h context, ucc_mem_map_flags_t flags,
+                         ucc_mem_map_params_t *params, size_t *memh_size,
+                         ucc_mem_map_mem_h *memh)
+{
+    (void)context; (void)flags; (void)params; (void)memh_size; (void)memh;
+    return UCC_ERR_NOT_IMPLEMENTED;
+}
+ucc_status_t ucc_mem_unmap(ucc_mem_map_mem_h *memh)
+{
+    (void)memh;
+    return UCC_ERR_NOT_IMPLEMENTED;
+}
+
+} /* extern "C" */

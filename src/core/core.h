@@ -1,0 +1,292 @@
+/* ucc_amd — internal object model.
+ *
+ * Reference parity map (structure re-derived, not ported):
+ *  - Task/Schedule      <- src/schedule/ucc_schedule.h (task fn ptrs, deps,
+ *                          event manager) — here: virtual post/progress +
+ *                          dependency counters on a C++ class.
+ *  - Context progress   <- core/ucc_context.c:1063 progress queue loop.
+ *  - Team state machine <- core/ucc_team.c (ADDR_EXCHANGE->TL_CREATE->ACTIVE).
+ *  - Score map dispatch <- src/coll_score/ (ranges per coll x memtype,
+ *                          fallback chain).
+ *  - TL static registry <- components/{cl,tl}/ dlopen zoo collapsed to a
+ *                          compiled-in registry (design deviation per
+ *                          SURVEY.md section 7).
+ */
+#ifndef UCC_AMD_CORE_H_
+#define UCC_AMD_CORE_H_
+
+#include <cstdint>
+#include <cstring>
+#include <deque>
+#include <functional>
+#include <memory>
+#include <mutex>
+#include <string>
+#include <vector>
+
+#include "../api/ucc.h"
+#include "../utils/config.h"
+#include "../utils/log.h"
+
+namespace ucc {
+
+struct Lib;
+struct Context;
+struct Team;
+class Task;
+class Schedule;
+class TlContext;
+class TlTeam;
+class Tl;
+struct CollRequest;
+
+double time_sec(); /* monotonic seconds */
+
+/* ------------------------------------------------------------------ Task */
+/* Unit of the progress engine. A collective algorithm is a Task (or a
+ * Schedule of Tasks). status lifecycle:
+ *   OPERATION_INITIALIZED -> (post) -> INPROGRESS|OK|err -> (progress)* */
+class Task {
+  public:
+    explicit Task(Context *ctx) : ctx_(ctx) {}
+    virtual ~Task() = default;
+
+    virtual ucc_status_t post()     = 0;
+    virtual ucc_status_t progress() { return status; }
+    /* Called exactly once when the task reaches a terminal status. */
+    virtual void on_complete() {}
+
+    ucc_status_t status = UCC_OPERATION_INITIALIZED;
+    Context     *ctx_;
+    CollRequest *req_        = nullptr; /* set on the root task           */
+    Schedule    *sched       = nullptr; /* parent schedule, if any        */
+    int          n_deps      = 0;
+    int          n_satisfied = 0;
+    std::vector<Task *> dependents; /* started when I complete            */
+    double       start_time = 0;
+    double       timeout    = 0; /* seconds; 0 = none                     */
+    bool         in_pq      = false;
+
+    void depends_on(Task *producer)
+    {
+        producer->dependents.push_back(this);
+        n_deps++;
+    }
+};
+
+/* Run a task: post it and hand it to the progress queue if needed. */
+void task_start(Task *t);
+/* Called by the engine when t->status became terminal. */
+void task_completed(Task *t);
+
+/* -------------------------------------------------------------- Schedule */
+class Schedule : public Task {
+  public:
+    explicit Schedule(Context *ctx) : Task(ctx) {}
+    ~Schedule() override
+    {
+        for (auto *t : tasks_) {
+            delete t;
+        }
+    }
+    void add(Task *t)
+    {
+        t->sched = this;
+        tasks_.push_back(t);
+    }
+    ucc_status_t post() override;
+    ucc_status_t progress() override { return status; }
+    void         subtask_completed(Task *t);
+
+    std::vector<Task *> tasks_;
+    size_t              n_completed_ = 0;
+    bool                posting_     = false;
+};
+
+/* ------------------------------------------------------------- ScoreMap  */
+using CollInitFn =
+    std::function<ucc_status_t(const ucc_coll_args_t &, Team *, Task **)>;
+
+struct ScoreRange {
+    size_t     start = 0;
+    size_t     end   = SIZE_MAX; /* inclusive range [start, end]           */
+    int        score = 0;
+    CollInitFn init;
+    std::string tl_name;
+    std::string alg_name;
+};
+
+class ScoreMap {
+  public:
+    /* indexed [coll_type_idx][mem_type] -> ranges sorted by score desc */
+    std::vector<ScoreRange> ranges[UCC_COLL_TYPE_NUM][UCC_MEMORY_TYPE_LAST];
+
+    void add(ucc_coll_type_t ct, ucc_memory_type_t mt, ScoreRange r);
+    /* Apply a user tuning string: "coll:msgrange:mem:@alg:score,..." */
+    ucc_status_t apply_str(const std::string &s);
+    /* Dispatch with fallback: highest score first. */
+    ucc_status_t init_coll(const ucc_coll_args_t &args, Team *team,
+                           size_t msgsize, Task **task) const;
+    std::string  to_string() const;
+};
+
+int coll_type_index(ucc_coll_type_t ct);
+const char *coll_type_name(ucc_coll_type_t ct);
+const char *mem_type_name(ucc_memory_type_t mt);
+ucc_status_t coll_type_from_name(const std::string &s, ucc_coll_type_t *ct);
+
+/* -------------------------------------------------------------- ProcInfo */
+struct ProcInfo {
+    uint64_t host_hash = 0;
+    int32_t  pid       = 0;
+    int32_t  device    = -1; /* hip device id, -1 = none                   */
+    uint64_t ctx_seq   = 0;  /* unique per (pid, context)                  */
+};
+ProcInfo local_proc_info();
+
+/* --------------------------------------------------------------- OobPoll */
+/* Drives repeated nonblocking OOB allgather rounds. */
+class OobRound {
+  public:
+    void init(const ucc_oob_coll_t &oob) { oob_ = oob; }
+    /* Begin an allgather of `size` bytes per rank. */
+    ucc_status_t start(const void *src, size_t size);
+    /* UCC_INPROGRESS / UCC_OK / error. On OK, data() is valid. */
+    ucc_status_t test();
+    void        *data() { return recv_.data(); }
+    uint32_t     size() const { return oob_.n_oob_eps; }
+    uint32_t     rank() const { return oob_.oob_ep; }
+
+  private:
+    ucc_oob_coll_t        oob_{};
+    std::vector<uint8_t>  send_, recv_;
+    void                 *req_ = nullptr;
+    bool                  active_ = false;
+};
+
+/* ------------------------------------------------------------------- TL  */
+class TlContext {
+  public:
+    explicit TlContext(Context *ctx) : ctx_(ctx) {}
+    virtual ~TlContext() = default;
+    virtual Tl *iface()  = 0;
+    Context    *ctx_;
+};
+
+class TlTeam {
+  public:
+    TlTeam(TlContext *tlc, Team *team) : tlc_(tlc), team_(team) {}
+    virtual ~TlTeam() = default;
+
+    /* Two-phase create with one combined OOB exchange round:
+     *  exchg_size/pack: contribute a fixed-size blob before the round;
+     *  exchg_unpack(all): receives n_ranks blobs at my TL's offset;
+     *  create_test: poll until resources are ready. */
+    virtual size_t       exchg_size() { return 0; }
+    virtual void         exchg_pack(void *buf) { (void)buf; }
+    virtual ucc_status_t exchg_unpack(const void *all, size_t stride)
+    {
+        (void)all; (void)stride;
+        return UCC_OK;
+    }
+    virtual ucc_status_t create_test() { return UCC_OK; }
+    /* Populate the team score map with this TL's algorithms. */
+    virtual void get_scores(Team *team, ScoreMap &map) = 0;
+
+    TlContext *tlc_;
+    Team      *team_;
+};
+
+class Tl {
+  public:
+    virtual ~Tl() = default;
+    virtual const char *name() const   = 0;
+    virtual int         default_score() const = 0;
+    /* nullptr if TL cannot run in this process context. */
+    virtual TlContext  *context_create(Context *ctx) = 0;
+    /* nullptr if TL cannot serve this team (e.g. multi-node for shm). */
+    virtual TlTeam     *team_create(TlContext *tlc, Team *team) = 0;
+};
+
+std::vector<Tl *> &tl_registry();
+void               register_tl(Tl *tl);
+void               ensure_builtin_tls(); /* registers self/shm/cdna4/rccl */
+
+/* --------------------------------------------------------------- Context */
+struct Context {
+    Lib                 *lib = nullptr;
+    ucc_context_params_t params{};
+    bool                 has_oob = false;
+    uint64_t             seq     = 0; /* unique id of this ctx in process  */
+    ProcInfo             proc;
+    std::deque<Task *>   pq;
+    std::recursive_mutex pq_mtx; /* used only in THREAD_MULTIPLE           */
+    bool                 mt = false;
+    std::vector<std::unique_ptr<TlContext>> tl_ctxs;
+    uint64_t             next_team_id = 1;
+    uint32_t             n_progress_calls = 0;
+
+    void pq_push(Task *t);
+    ucc_status_t progress();
+};
+
+/* ------------------------------------------------------------------ Team */
+struct Team {
+    enum State {
+        ADDR_EXCHANGE,
+        TL_EXCHANGE,
+        TL_CREATE,
+        ACTIVE,
+        FAILED,
+    };
+
+    Context            *ctx = nullptr;
+    ucc_team_params_t   params{};
+    ucc_oob_coll_t      oob{};
+    bool                has_oob = false;
+    uint32_t            rank = 0, size = 1;
+    uint16_t            id = 0;
+    State               state = ADDR_EXCHANGE;
+    ucc_status_t        err   = UCC_OK;
+    std::vector<ProcInfo>                  procs;
+    std::vector<std::unique_ptr<TlTeam>>   tl_teams;
+    std::vector<size_t>                    exchg_off; /* per-TL blob offset */
+    size_t                                 exchg_stride = 0;
+    std::vector<uint8_t>                   exchg_buf;
+    OobRound            oobr;
+    ScoreMap            score_map;
+    uint64_t            team_uid = 0; /* rank0 random, shared: shm naming  */
+    uint64_t            coll_seq = 0;
+
+    bool all_same_node() const;
+    bool all_have_device() const;
+};
+
+/* ------------------------------------------------------------------ Lib  */
+struct Lib {
+    ucc_lib_params_t  params{};
+    ucc_thread_mode_t thread_mode = UCC_THREAD_SINGLE;
+    uint64_t          next_ctx_seq = 1;
+};
+
+/* ----------------------------------------------------------- CollRequest */
+struct CollRequest {
+    ucc_coll_req_t  super; /* must be first: public handle casts here      */
+    ucc_coll_args_t args{};
+    Team           *team = nullptr;
+    Task           *task = nullptr;
+    bool            persistent = false;
+    bool            posted     = false;
+    uint64_t        seq        = 0;
+};
+
+/* msgsize used for score-map range selection (bytes, per reference
+ * ucc_coll_utils msgsize conventions). */
+size_t coll_args_msgsize(const ucc_coll_args_t &args, uint32_t rank,
+                         uint32_t size);
+ucc_memory_type_t coll_args_mem_type(const ucc_coll_args_t &args,
+                                     uint32_t rank);
+
+} // namespace ucc
+
+#endif

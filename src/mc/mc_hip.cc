@@ -1,0 +1,88 @@
+#include "mc.h"
+#include "../utils/log.h"
+
+#include <cstring>
+#include <hip/hip_runtime.h>
+
+namespace ucc {
+namespace mc {
+
+int hip_device_count()
+{
+    static int count = []() {
+        int n = 0;
+        if (hipGetDeviceCount(&n) != hipSuccess) {
+            return 0;
+        }
+        return n;
+    }();
+    return count;
+}
+
+bool hip_available() { return hip_device_count() > 0; }
+
+ucc_status_t mem_query(const void *ptr, ucc_memory_type_t *mt)
+{
+    if (!hip_available()) {
+        *mt = UCC_MEMORY_TYPE_HOST;
+        return UCC_OK;
+    }
+    hipPointerAttribute_t attr;
+    hipError_t            err = hipPointerGetAttributes(&attr, ptr);
+    if (err != hipSuccess) {
+        (void)hipGetLastError();
+        *mt = UCC_MEMORY_TYPE_HOST;
+        return UCC_OK;
+    }
+    switch (attr.type) {
+    case hipMemoryTypeDevice: *mt = UCC_MEMORY_TYPE_CUDA; break;
+    case hipMemoryTypeManaged: *mt = UCC_MEMORY_TYPE_CUDA_MANAGED; break;
+    default: *mt = UCC_MEMORY_TYPE_HOST; break;
+    }
+    return UCC_OK;
+}
+
+ucc_status_t alloc(void **ptr, size_t size, ucc_memory_type_t mt)
+{
+    if (is_device_mt(mt)) {
+        if (hipMalloc(ptr, size) != hipSuccess) {
+            return UCC_ERR_NO_MEMORY;
+        }
+        return UCC_OK;
+    }
+    *ptr = malloc(size);
+    return *ptr ? UCC_OK : UCC_ERR_NO_MEMORY;
+}
+
+ucc_status_t mem_free(void *ptr, ucc_memory_type_t mt)
+{
+    if (is_device_mt(mt)) {
+        return hipFree(ptr) == hipSuccess ? UCC_OK : UCC_ERR_INVALID_PARAM;
+    }
+    free(ptr);
+    return UCC_OK;
+}
+
+ucc_status_t copy(void *dst, ucc_memory_type_t dst_mt, const void *src,
+                  ucc_memory_type_t src_mt, size_t bytes)
+{
+    if (!is_device_mt(dst_mt) && !is_device_mt(src_mt)) {
+        memcpy(dst, src, bytes);
+        return UCC_OK;
+    }
+    hipError_t err = hipMemcpy(dst, src, bytes, hipMemcpyDefault);
+    return err == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
+}
+
+ucc_status_t memset_(void *ptr, ucc_memory_type_t mt, int value, size_t bytes)
+{
+    if (!is_device_mt(mt)) {
+        memset(ptr, value, bytes);
+        return UCC_OK;
+    }
+    return hipMemset(ptr, value, bytes) == hipSuccess ? UCC_OK
+                                                      : UCC_ERR_NO_RESOURCE;
+}
+
+} // namespace mc
+} // namespace ucc

@@ -1,0 +1,145 @@
+/* Task engine: dependency-counted tasks + schedules driven by the context
+ * progress queue. Parity: reference src/schedule/ (ucc_coll_task_t event
+ * manager + ucc_schedule_t) and core/ucc_progress_queue*. */
+#include "../core/core.h"
+
+#include <ctime>
+
+namespace ucc {
+
+double time_sec()
+{
+    struct timespec ts;
+    clock_gettime(CLOCK_MONOTONIC, &ts);
+    return (double)ts.tv_sec + 1e-9 * (double)ts.tv_nsec;
+}
+
+void Context::pq_push(Task *t)
+{
+    if (t->in_pq) {
+        return;
+    }
+    t->in_pq = true;
+    if (mt) {
+        std::lock_guard<std::recursive_mutex> lk(pq_mtx);
+        pq.push_back(t);
+    } else {
+        pq.push_back(t);
+    }
+}
+
+void task_start(Task *t)
+{
+    t->start_time = time_sec();
+    ucc_status_t st = t->post();
+    t->status       = st;
+    if (st == UCC_INPROGRESS) {
+        t->ctx_->pq_push(t);
+    } else {
+        task_completed(t);
+    }
+}
+
+void task_completed(Task *t)
+{
+    t->on_complete();
+    if (t->req_) {
+        t->req_->super.status = t->status;
+        if ((t->req_->args.mask & UCC_COLL_ARGS_FIELD_CB) &&
+            t->req_->args.cb.cb) {
+            t->req_->args.cb.cb(t->req_->args.cb.data, t->status);
+        }
+    }
+    if (t->sched) {
+        t->sched->subtask_completed(t);
+    }
+    for (Task *d : t->dependents) {
+        d->n_satisfied++;
+        if (d->n_satisfied == d->n_deps) {
+            task_start(d);
+        }
+    }
+}
+
+ucc_status_t Schedule::post()
+{
+    status       = UCC_INPROGRESS;
+    n_completed_ = 0;
+    posting_     = true;
+    /* reset dep counters for persistent re-post */
+    for (auto *t : tasks_) {
+        t->n_satisfied = 0;
+        t->status      = UCC_OPERATION_INITIALIZED;
+    }
+    for (auto *t : tasks_) {
+        if (t->n_deps == 0) {
+            task_start(t);
+        }
+    }
+    posting_ = false;
+    if (tasks_.empty() || n_completed_ == tasks_.size()) {
+        if (status == UCC_INPROGRESS) {
+            status = UCC_OK;
+        }
+    }
+    /* task_start(this) handles the terminal-status notification */
+    return status;
+}
+
+void Schedule::subtask_completed(Task *t)
+{
+    if (t->status < 0 && status >= 0) {
+        status = t->status; /* propagate first error */
+        if (!posting_) {
+            task_completed(this);
+        }
+        return;
+    }
+    n_completed_++;
+    if (n_completed_ == tasks_.size() && status == UCC_INPROGRESS) {
+        if (!posting_) {
+            status = UCC_OK;
+            task_completed(this);
+        }
+        /* else: Schedule::post finalizes the status itself */
+    }
+}
+
+ucc_status_t Context::progress()
+{
+    n_progress_calls++;
+    Task *t = nullptr;
+    {
+        if (mt) {
+            pq_mtx.lock();
+        }
+        if (!pq.empty()) {
+            t = pq.front();
+            pq.pop_front();
+        }
+        if (mt) {
+            pq_mtx.unlock();
+        }
+    }
+    if (!t) {
+        return UCC_OK;
+    }
+    t->in_pq        = false;
+    ucc_status_t st = t->progress();
+    t->status       = st;
+    if (st == UCC_INPROGRESS) {
+        if (t->timeout > 0 && time_sec() - t->start_time > t->timeout) {
+            ucc_warn("collective task %p timed out after %.1fs", (void *)t,
+                     t->timeout);
+            t->status = UCC_ERR_TIMED_OUT;
+            task_completed(t);
+            return UCC_ERR_TIMED_OUT;
+        }
+        pq_push(t);
+        return UCC_OK;
+    }
+    task_completed(t);
+    return st < 0 ? st : UCC_OK;
+}
+
+} // namespace ucc
