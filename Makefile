@@ -9,7 +9,7 @@ PY_EXT   := $(shell $(PYTHON) -c "import sysconfig;print(sysconfig.get_config_va
 PY_INC   := $(shell $(PYTHON) -m pybind11 --includes)
 
 CXXFLAGS := --offload-arch=$(ARCH) -O3 -std=c++17 -fPIC -Wall -Wextra \
-            -Wno-unused-parameter -DUCC_AMD_HAS_HIP
+            -Wno-unused-parameter -DUCC_AMD_HAS_HIP -DUCC_AMD_HAS_TL_CDNA4
 LDFLAGS  := -shared -fPIC
 
 BUILD := build

@@ -1,0 +1,633 @@
+/* gfx950 execution-component kernels.
+ *
+ * Design (MI355X-first, see /opt/skills/guides/cdna_hip_programming.md):
+ *  - 256-thread workgroups (4 waves of 64), grid-stride loops capped so the
+ *    scheduler keeps the chip busy without oversubscription (G11).
+ *  - 16 B per lane vector packs (G13): bf16/fp16 loaded as ushort8-class
+ *    packs, converted to f32, accumulated in f32; fp64 in f64; ints in T.
+ *  - fused allreduce: cross-GPU arrival flags are u64 system-scope atomics
+ *    on fine-grained memory; payload ordering = __threadfence_system on the
+ *    producer before the flag store, one system acquire on the consumer
+ *    after the poll matches (the G16 protocol lifted to system scope for
+ *    xGMI peers). Every spin is bounded and reports via error_word.
+ */
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <type_traits>
+
+#include "../ec_hip.h"
+
+namespace ucc {
+namespace ec_hip {
+
+/* ------------------------------------------------------------ convert  */
+__device__ __forceinline__ float d_bf16_to_f(uint16_t v)
+{
+    union {
+        uint32_t u;
+        float    f;
+    } c;
+    c.u = (uint32_t)v << 16;
+    return c.f;
+}
+__device__ __forceinline__ uint16_t d_f_to_bf16(float f)
+{
+    union {
+        uint32_t u;
+        float    f;
+    } c;
+    c.f = f;
+    uint32_t r = 0x7fff + ((c.u >> 16) & 1);
+    return (uint16_t)((c.u + r) >> 16);
+}
+
+/* fp8 (OCP): decode/encode via f32, same tables as the host reference. */
+__device__ __forceinline__ float d_e4m3_to_f(uint8_t v)
+{
+    uint32_t sign = v >> 7, exp = (v >> 3) & 0xf, man = v & 7;
+    float    val;
+    if (exp == 0xf && man == 7) {
+        return __builtin_nanf("");
+    }
+    if (exp == 0) {
+        val = (float)man * (1.0f / 512.0f);
+    } else {
+        val = (1.0f + (float)man * 0.125f) * exp2f((float)exp - 7.0f);
+    }
+    return sign ? -val : val;
+}
+__device__ __forceinline__ uint8_t d_f_to_e4m3(float f)
+{
+    if (isnan(f)) {
+        return 0x7f;
+    }
+    uint8_t sign = f < 0.0f ? 0x80 : 0;
+    float   a    = fabsf(f);
+    if (a >= 448.0f) {
+        return sign | 0x7e;
+    }
+    if (a < 0.001953125f) { /* 2^-9 */
+        return sign;
+    }
+    int   e;
+    float m   = frexpf(a, &e);
+    int   exp = e - 1 + 7;
+    if (exp <= 0) {
+        int mi = (int)rintf(a * 512.0f);
+        if (mi >= 8) {
+            return sign | 0x08;
+        }
+        return sign | (uint8_t)mi;
+    }
+    int mi = (int)rintf((m * 2.0f - 1.0f) * 8.0f);
+    if (mi >= 8) {
+        mi = 0;
+        exp++;
+    }
+    if (exp > 0xf || (exp == 0xf && mi == 7)) {
+        return sign | 0x7e;
+    }
+    return sign | (uint8_t)(exp << 3) | (uint8_t)mi;
+}
+__device__ __forceinline__ float d_e5m2_to_f(uint8_t v)
+{
+    uint32_t sign = v >> 7, exp = (v >> 2) & 0x1f, man = v & 3;
+    if (exp == 0x1f) {
+        return man ? __builtin_nanf("")
+                   : (sign ? -__builtin_inff() : __builtin_inff());
+    }
+    float val;
+    if (exp == 0) {
+        val = (float)man * 0.25f * exp2f(-14.0f);
+    } else {
+        val = (1.0f + (float)man * 0.25f) * exp2f((float)exp - 15.0f);
+    }
+    return sign ? -val : val;
+}
+__device__ __forceinline__ uint8_t d_f_to_e5m2(float f)
+{
+    if (isnan(f)) {
+        return 0x7f;
+    }
+    uint8_t sign = signbit(f) ? 0x80 : 0;
+    float   a    = fabsf(f);
+    if (isinf(f) || a > 57344.0f) {
+        return sign | 0x7c;
+    }
+    if (a < exp2f(-17.0f)) {
+        return sign;
+    }
+    int   e;
+    float m   = frexpf(a, &e);
+    int   exp = e - 1 + 15;
+    if (exp <= 0) {
+        int mi = (int)rintf(a * exp2f(14.0f) * 4.0f);
+        if (mi >= 4) {
+            return sign | 0x04;
+        }
+        return sign | (uint8_t)mi;
+    }
+    int mi = (int)rintf((m * 2.0f - 1.0f) * 4.0f);
+    if (mi >= 4) {
+        mi = 0;
+        exp++;
+    }
+    if (exp >= 0x1f) {
+        return sign | 0x7c;
+    }
+    return sign | (uint8_t)(exp << 2) | (uint8_t)mi;
+}
+
+/* type traits: storage type T <-> accumulator A */
+template <typename T> struct Cvt {
+    using A = T;
+    static __device__ __forceinline__ A load(T v) { return v; }
+    static __device__ __forceinline__ T store(A v) { return (T)v; }
+};
+struct bf16_t {
+    uint16_t v;
+};
+template <> struct Cvt<bf16_t> {
+    using A = float;
+    static __device__ __forceinline__ float load(bf16_t x)
+    {
+        return d_bf16_to_f(x.v);
+    }
+    static __device__ __forceinline__ bf16_t store(float f)
+    {
+        return {d_f_to_bf16(f)};
+    }
+};
+struct fp16_t {
+    _Float16 v;
+};
+template <> struct Cvt<fp16_t> {
+    using A = float;
+    static __device__ __forceinline__ float load(fp16_t x)
+    {
+        return (float)x.v;
+    }
+    static __device__ __forceinline__ fp16_t store(float f)
+    {
+        return {(_Float16)f};
+    }
+};
+struct e4m3_t {
+    uint8_t v;
+};
+template <> struct Cvt<e4m3_t> {
+    using A = float;
+    static __device__ __forceinline__ float load(e4m3_t x)
+    {
+        return d_e4m3_to_f(x.v);
+    }
+    static __device__ __forceinline__ e4m3_t store(float f)
+    {
+        return {d_f_to_e4m3(f)};
+    }
+};
+struct e5m2_t {
+    uint8_t v;
+};
+template <> struct Cvt<e5m2_t> {
+    using A = float;
+    static __device__ __forceinline__ float load(e5m2_t x)
+    {
+        return d_e5m2_to_f(x.v);
+    }
+    static __device__ __forceinline__ e5m2_t store(float f)
+    {
+        return {d_f_to_e5m2(f)};
+    }
+};
+
+/* op ids match ucc_reduction_op_t */
+template <typename A, int OP>
+__device__ __forceinline__ A red(A a, A b)
+{
+    if constexpr (OP == 0 || OP == 12) { /* SUM / AVG */
+        return a + b;
+    } else if constexpr (OP == 1) {
+        return a * b;
+    } else if constexpr (OP == 2) {
+        return a > b ? a : b;
+    } else if constexpr (OP == 3) {
+        return a < b ? a : b;
+    } else if constexpr (OP == 4) {
+        return (A)((a != (A)0) && (b != (A)0));
+    } else if constexpr (OP == 5) {
+        return (A)((a != (A)0) || (b != (A)0));
+    } else if constexpr (OP == 6) {
+        return (A)((a != (A)0) != (b != (A)0));
+    } else if constexpr (OP == 7) {
+        if constexpr (!std::is_floating_point<A>::value) {
+            return a & b;
+        }
+    } else if constexpr (OP == 8) {
+        if constexpr (!std::is_floating_point<A>::value) {
+            return a | b;
+        }
+    } else if constexpr (OP == 9) {
+        if constexpr (!std::is_floating_point<A>::value) {
+            return a ^ b;
+        }
+    }
+    return a;
+}
+
+template <typename A>
+__device__ __forceinline__ A apply_alpha(A v, float alpha)
+{
+    if constexpr (std::is_floating_point<A>::value) {
+        return (A)(v * (A)alpha);
+    } else {
+        (void)alpha;
+        return v;
+    }
+}
+
+/* ------------------------------------------------------------- reduce  */
+/* VEC chosen so VEC*sizeof(T) == 16 bytes (one dwordx4 per lane). */
+template <typename T> struct VecOf {
+    static constexpr int value = 16 / (int)sizeof(T);
+};
+
+template <typename T, int OP, int VEC>
+struct __align__(16) Pack {
+    T v[VEC];
+};
+
+template <typename T, int OP, int VEC>
+__global__ void k_reduce(ReduceArgs a)
+{
+    using A            = typename Cvt<T>::A;
+    const uint64_t nv  = a.count / VEC;
+    const int      n   = a.n_srcs;
+    uint64_t       i   = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const uint64_t str = (uint64_t)gridDim.x * blockDim.x;
+    using P            = Pack<T, OP, VEC>;
+    for (; i < nv; i += str) {
+        P acc = ((const P *)a.srcs[0])[i];
+        A  r[VEC];
+#pragma unroll
+        for (int k = 0; k < VEC; k++) {
+            r[k] = Cvt<T>::load(acc.v[k]);
+        }
+        for (int s = 1; s < n; s++) {
+            P x = ((const P *)a.srcs[s])[i];
+#pragma unroll
+            for (int k = 0; k < VEC; k++) {
+                r[k] = red<A, OP>(r[k], Cvt<T>::load(x.v[k]));
+            }
+        }
+        P out;
+#pragma unroll
+        for (int k = 0; k < VEC; k++) {
+            out.v[k] = Cvt<T>::store(apply_alpha<A>(r[k], a.alpha));
+        }
+        ((P *)a.dst)[i] = out;
+    }
+    /* tail (scalar) */
+    uint64_t t = nv * VEC + ((uint64_t)blockIdx.x * blockDim.x + threadIdx.x);
+    for (; t < a.count; t += str) {
+        A r = Cvt<T>::load(((const T *)a.srcs[0])[t]);
+        for (int s = 1; s < n; s++) {
+            r = red<A, OP>(r, Cvt<T>::load(((const T *)a.srcs[s])[t]));
+        }
+        ((T *)a.dst)[t] = Cvt<T>::store(apply_alpha<A>(r, a.alpha));
+    }
+}
+
+template <typename T, int OP>
+__global__ void k_reduce_scalar(ReduceArgs a)
+{
+    using A            = typename Cvt<T>::A;
+    uint64_t       i   = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const uint64_t str = (uint64_t)gridDim.x * blockDim.x;
+    for (; i < a.count; i += str) {
+        A r = Cvt<T>::load(((const T *)a.srcs[0])[i]);
+        for (int s = 1; s < a.n_srcs; s++) {
+            r = red<A, OP>(r, Cvt<T>::load(((const T *)a.srcs[s])[i]));
+        }
+        ((T *)a.dst)[i] = Cvt<T>::store(apply_alpha<A>(r, a.alpha));
+    }
+}
+
+/* -------------------------------------------------------- gather copy  */
+__global__ void k_gather_copy(GatherArgs a)
+{
+    for (int s = 0; s < a.n; s++) {
+        uint8_t       *dst = (uint8_t *)a.dst_base + a.offs[s];
+        const uint8_t *src = (const uint8_t *)a.srcs[s];
+        uint64_t       len = a.lens[s];
+        const uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+        const uint64_t str = (uint64_t)gridDim.x * blockDim.x;
+        if ((((uintptr_t)dst | (uintptr_t)src) & 15) == 0) {
+            uint64_t nv = len / 16;
+            for (uint64_t i = tid; i < nv; i += str) {
+                ((uint4 *)dst)[i] = ((const uint4 *)src)[i];
+            }
+            for (uint64_t i = nv * 16 + tid; i < len; i += str) {
+                dst[i] = src[i];
+            }
+        } else if ((((uintptr_t)dst | (uintptr_t)src) & 3) == 0) {
+            uint64_t nv = len / 4;
+            for (uint64_t i = tid; i < nv; i += str) {
+                ((uint32_t *)dst)[i] = ((const uint32_t *)src)[i];
+            }
+            for (uint64_t i = nv * 4 + tid; i < len; i += str) {
+                dst[i] = src[i];
+            }
+        } else {
+            for (uint64_t i = tid; i < len; i += str) {
+                dst[i] = src[i];
+            }
+        }
+    }
+}
+
+/* ---------------------------------------------------- fused allreduce  */
+__device__ __forceinline__ uint64_t
+sys_load(const uint64_t *p)
+{
+    return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+}
+__device__ __forceinline__ void sys_store(uint64_t *p, uint64_t v)
+{
+    __hip_atomic_store(p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+}
+
+constexpr uint64_t kSpinLimit = 200u * 1000u * 1000u;
+
+template <typename T, int OP, int VEC>
+__global__ void k_fused_allreduce(FusedArgs a)
+{
+    using A = typename Cvt<T>::A;
+    /* 1. stage src -> my scratch (all blocks, grid-stride, 16B packs) */
+    {
+        const uint64_t tid =
+            (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+        const uint64_t str = (uint64_t)gridDim.x * blockDim.x;
+        const uint64_t nv  = a.count / VEC;
+        using P            = Pack<T, OP, VEC>;
+        for (uint64_t i = tid; i < nv; i += str) {
+            ((P *)a.my_scratch)[i] = ((const P *)a.src)[i];
+        }
+        for (uint64_t i = nv * VEC + tid; i < a.count; i += str) {
+            ((T *)a.my_scratch)[i] = ((const T *)a.src)[i];
+        }
+    }
+    __threadfence_system();
+    __syncthreads();
+    /* 2. grid arrival: each block bumps the staging counter; block 0
+     * waits for all blocks, then signals every peer (incl. self).
+     * staging counter is u64 index kMaxRanks*kMaxSlots + slot (monotone,
+     * grows by nblocks per use -> target seq*nblocks). */
+    uint64_t *stage_cnt = a.local_flags + 8 * kMaxRanks + a.slot;
+    if (threadIdx.x == 0) {
+        __hip_atomic_fetch_add(stage_cnt, 1, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+    }
+    if (blockIdx.x == 0) {
+        if (threadIdx.x == 0) {
+            uint64_t spins = 0;
+            while (sys_load(stage_cnt) < a.seq * (uint64_t)a.nblocks) {
+                if (++spins > kSpinLimit) {
+                    sys_store(a.error_word, 1);
+                    return;
+                }
+                __builtin_amdgcn_s_sleep(2);
+            }
+        }
+        __syncthreads();
+        if ((int)threadIdx.x < a.nranks) {
+            sys_store(a.peer_flags[threadIdx.x] +
+                          (uint64_t)a.slot * kMaxRanks + a.rank,
+                      a.seq);
+        }
+    }
+    /* 3. all blocks wait for all ranks' arrivals */
+    if (threadIdx.x < 64) {
+        int      j     = (int)threadIdx.x;
+        uint64_t spins = 0;
+        if (j < a.nranks) {
+            const uint64_t *f =
+                a.local_flags + (uint64_t)a.slot * kMaxRanks + j;
+            while (sys_load(f) < a.seq) {
+                if (++spins > kSpinLimit) {
+                    sys_store(a.error_word, 1);
+                    return;
+                }
+                __builtin_amdgcn_s_sleep(2);
+            }
+        }
+    }
+    __syncthreads();
+    __threadfence_system(); /* acquire: drop stale lines before peer reads */
+    /* 4. reduce all peers' scratch into dst */
+    {
+        const uint64_t tid =
+            (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+        const uint64_t str = (uint64_t)gridDim.x * blockDim.x;
+        const uint64_t nv  = a.count / VEC;
+        using P            = Pack<T, OP, VEC>;
+        const int n        = a.nranks;
+        for (uint64_t i = tid; i < nv; i += str) {
+            P acc = ((const P *)a.peer_scratch[0])[i];
+            A r[VEC];
+#pragma unroll
+            for (int k = 0; k < VEC; k++) {
+                r[k] = Cvt<T>::load(acc.v[k]);
+            }
+            for (int s = 1; s < n; s++) {
+                P x = ((const P *)a.peer_scratch[s])[i];
+#pragma unroll
+                for (int k = 0; k < VEC; k++) {
+                    r[k] = red<A, OP>(r[k], Cvt<T>::load(x.v[k]));
+                }
+            }
+            P out;
+#pragma unroll
+            for (int k = 0; k < VEC; k++) {
+                out.v[k] = Cvt<T>::store(apply_alpha<A>(r[k], a.alpha));
+            }
+            ((P *)a.dst)[i] = out;
+        }
+        for (uint64_t t = nv * VEC + tid; t < a.count; t += str) {
+            A r = Cvt<T>::load(((const T *)a.peer_scratch[0])[t]);
+            for (int s = 1; s < n; s++) {
+                r = red<A, OP>(r, Cvt<T>::load(((const T *)a.peer_scratch[s])[t]));
+            }
+            ((T *)a.dst)[t] = Cvt<T>::store(apply_alpha<A>(r, a.alpha));
+        }
+    }
+}
+
+/* ----------------------------------------------------------- launchers */
+static bool aligned16(const void *p) { return (((uintptr_t)p) & 15) == 0; }
+
+template <typename T, int OP>
+static ucc_status_t launch_reduce(const ReduceArgs &a, hipStream_t s)
+{
+    constexpr int VEC = VecOf<T>::value;
+    bool          vec = aligned16(a.dst);
+    for (int i = 0; i < a.n_srcs; i++) {
+        vec = vec && aligned16(a.srcs[i]);
+    }
+    int  threads = 256;
+    long work    = (long)(a.count / (vec ? VEC : 1));
+    int  blocks  = (int)std::min<long>((work + threads - 1) / threads, 2048);
+    if (blocks < 1) {
+        blocks = 1;
+    }
+    if (vec) {
+        hipLaunchKernelGGL((k_reduce<T, OP, VEC>), dim3(blocks),
+                           dim3(threads), 0, s, a);
+    } else {
+        hipLaunchKernelGGL((k_reduce_scalar<T, OP>), dim3(blocks),
+                           dim3(threads), 0, s, a);
+    }
+    return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
+}
+
+template <typename T, int OP>
+static ucc_status_t launch_fused(const FusedArgs &a, hipStream_t s)
+{
+    constexpr int VEC = VecOf<T>::value;
+    hipLaunchKernelGGL((k_fused_allreduce<T, OP, VEC>), dim3(a.nblocks),
+                       dim3(256), 0, s, a);
+    return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
+}
+
+#define UCC_DT_CASE_FLOAT(DT, T, LFN)                                        \
+    case DT:                                                                 \
+        switch (op) {                                                        \
+        case UCC_OP_SUM: return LFN<T, 0>(a, s);                             \
+        case UCC_OP_AVG: return LFN<T, 12>(a, s);                            \
+        case UCC_OP_PROD: return LFN<T, 1>(a, s);                            \
+        case UCC_OP_MAX: return LFN<T, 2>(a, s);                             \
+        case UCC_OP_MIN: return LFN<T, 3>(a, s);                             \
+        default: return UCC_ERR_NOT_SUPPORTED;                               \
+        }
+
+#define UCC_DT_CASE_INT(DT, T, LFN)                                          \
+    case DT:                                                                 \
+        switch (op) {                                                        \
+        case UCC_OP_SUM: return LFN<T, 0>(a, s);                             \
+        case UCC_OP_PROD: return LFN<T, 1>(a, s);                            \
+        case UCC_OP_MAX: return LFN<T, 2>(a, s);                             \
+        case UCC_OP_MIN: return LFN<T, 3>(a, s);                             \
+        case UCC_OP_LAND: return LFN<T, 4>(a, s);                            \
+        case UCC_OP_LOR: return LFN<T, 5>(a, s);                             \
+        case UCC_OP_LXOR: return LFN<T, 6>(a, s);                            \
+        case UCC_OP_BAND: return LFN<T, 7>(a, s);                            \
+        case UCC_OP_BOR: return LFN<T, 8>(a, s);                             \
+        case UCC_OP_BXOR: return LFN<T, 9>(a, s);                            \
+        default: return UCC_ERR_NOT_SUPPORTED;                               \
+        }
+
+ucc_status_t reduce(const ReduceArgs &a, hipStream_t s)
+{
+    ucc_reduction_op_t op = a.op;
+    switch (a.dt) {
+        UCC_DT_CASE_FLOAT(UCC_DT_BFLOAT16, bf16_t, launch_reduce)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT16, fp16_t, launch_reduce)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT32, float, launch_reduce)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT64, double, launch_reduce)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT8_E4M3, e4m3_t, launch_reduce)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT8_E5M2, e5m2_t, launch_reduce)
+        UCC_DT_CASE_INT(UCC_DT_INT8, int8_t, launch_reduce)
+        UCC_DT_CASE_INT(UCC_DT_UINT8, uint8_t, launch_reduce)
+        UCC_DT_CASE_INT(UCC_DT_INT16, int16_t, launch_reduce)
+        UCC_DT_CASE_INT(UCC_DT_UINT16, uint16_t, launch_reduce)
+        UCC_DT_CASE_INT(UCC_DT_INT32, int32_t, launch_reduce)
+        UCC_DT_CASE_INT(UCC_DT_UINT32, uint32_t, launch_reduce)
+        UCC_DT_CASE_INT(UCC_DT_INT64, int64_t, launch_reduce)
+        UCC_DT_CASE_INT(UCC_DT_UINT64, uint64_t, launch_reduce)
+    default: return UCC_ERR_NOT_SUPPORTED;
+    }
+}
+
+ucc_status_t fused_allreduce(const FusedArgs &a, hipStream_t s)
+{
+    ucc_reduction_op_t op = a.op;
+    switch (a.dt) {
+        UCC_DT_CASE_FLOAT(UCC_DT_BFLOAT16, bf16_t, launch_fused)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT16, fp16_t, launch_fused)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT32, float, launch_fused)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT64, double, launch_fused)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT8_E4M3, e4m3_t, launch_fused)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT8_E5M2, e5m2_t, launch_fused)
+        UCC_DT_CASE_INT(UCC_DT_INT8, int8_t, launch_fused)
+        UCC_DT_CASE_INT(UCC_DT_UINT8, uint8_t, launch_fused)
+        UCC_DT_CASE_INT(UCC_DT_INT16, int16_t, launch_fused)
+        UCC_DT_CASE_INT(UCC_DT_UINT16, uint16_t, launch_fused)
+        UCC_DT_CASE_INT(UCC_DT_INT32, int32_t, launch_fused)
+        UCC_DT_CASE_INT(UCC_DT_UINT32, uint32_t, launch_fused)
+        UCC_DT_CASE_INT(UCC_DT_INT64, int64_t, launch_fused)
+        UCC_DT_CASE_INT(UCC_DT_UINT64, uint64_t, launch_fused)
+    default: return UCC_ERR_NOT_SUPPORTED;
+    }
+}
+
+ucc_status_t gather_copy(const GatherArgs &a, hipStream_t s)
+{
+    uint64_t total = 0;
+    for (int i = 0; i < a.n; i++) {
+        total += a.lens[i];
+    }
+    int threads = 256;
+    int blocks =
+        (int)std::min<uint64_t>((total / 16 + threads - 1) / threads, 2048);
+    if (blocks < 1) {
+        blocks = 1;
+    }
+    hipLaunchKernelGGL(k_gather_copy, dim3(blocks), dim3(threads), 0, s, a);
+    return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
+}
+
+bool dt_supported(ucc_datatype_t dt)
+{
+    switch (dt) {
+    case UCC_DT_BFLOAT16:
+    case UCC_DT_FLOAT16:
+    case UCC_DT_FLOAT32:
+    case UCC_DT_FLOAT64:
+    case UCC_DT_FLOAT8_E4M3:
+    case UCC_DT_FLOAT8_E5M2:
+    case UCC_DT_INT8:
+    case UCC_DT_UINT8:
+    case UCC_DT_INT16:
+    case UCC_DT_UINT16:
+    case UCC_DT_INT32:
+    case UCC_DT_UINT32:
+    case UCC_DT_INT64:
+    case UCC_DT_UINT64: return true;
+    default: return false;
+    }
+}
+
+bool op_supported(ucc_datatype_t dt, ucc_reduction_op_t op)
+{
+    bool is_float = dt == UCC_DT_BFLOAT16 || dt == UCC_DT_FLOAT16 ||
+                    dt == UCC_DT_FLOAT32 || dt == UCC_DT_FLOAT64 ||
+                    dt == UCC_DT_FLOAT8_E4M3 || dt == UCC_DT_FLOAT8_E5M2;
+    switch (op) {
+    case UCC_OP_SUM:
+    case UCC_OP_PROD:
+    case UCC_OP_MAX:
+    case UCC_OP_MIN: return dt_supported(dt);
+    case UCC_OP_AVG: return is_float;
+    case UCC_OP_LAND:
+    case UCC_OP_LOR:
+    case UCC_OP_LXOR:
+    case UCC_OP_BAND:
+    case UCC_OP_BOR:
+    case UCC_OP_BXOR: return dt_supported(dt) && !is_float;
+    default: return false;
+    }
+}
+
+} // namespace ec_hip
+} // namespace ucc

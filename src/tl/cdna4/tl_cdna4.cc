@@ -1,0 +1,856 @@
+/* TL "cdna4": the xGMI device transport for up to 8 fully-connected
+ * MI355X GPUs (one process per GPU, or the in-process multi-rank jig on a
+ * single GPU).
+ *
+ * Reference parity: replaces tl/cuda (tl_cuda_team.c scratch+IPC exchange,
+ * tl_cuda_coll.h slot protocol, *_linear.c algorithms) and the NVLS
+ * allreduce (tl/cuda/kernels/allreduce_kernel.cu), re-designed for CDNA4:
+ *
+ *  - control plane: host POSIX-shm slot segment (slot_seg.h) with monotonic
+ *    step counters — same protocol as tl/shm.
+ *  - data plane: per-rank device scratch (hipMalloc, HIP-IPC exported) laid
+ *    out as [slot][parity][in|out] areas; peers read each other's areas
+ *    directly over xGMI from gfx950 kernels (ec_hip reduce/gather), so a
+ *    staged linear collective touches all 7 links simultaneously:
+ *    reduce-scatter phase + allgather phase each move S/8 per link =>
+ *    aggregate bus bandwidth approaches 7 x 153 GB/s per GPU.
+ *  - small messages: ONE fused kernel per rank (ec_hip::fused_allreduce)
+ *    that stages, signals arrival on every peer's fine-grained flag word
+ *    (system-scope atomics), spins bounded, reduces — no host round trips
+ *    between ranks, single kernel-launch latency.
+ *
+ * Algorithms here (v1): allreduce (fused small / staged-linear large),
+ * allgather(v), reduce_scatter(v), bcast, reduce. Host-memory colls and
+ * everything else fall back per the score map.
+ */
+#include <hip/hip_runtime.h>
+
+#include "../../core/core.h"
+#include "../../ec/ec_hip.h"
+#include "../../mc/mc.h"
+#include "../shm/slot_seg.h"
+
+namespace ucc {
+namespace {
+
+using ec_hip::kMaxRanks;
+
+#define HIPCHK(expr)                                                         \
+    do {                                                                     \
+        hipError_t _e = (expr);                                              \
+        if (_e != hipSuccess) {                                              \
+            ucc_error("%s failed: %s", #expr, hipGetErrorString(_e));        \
+            return UCC_ERR_NO_RESOURCE;                                      \
+        }                                                                    \
+    } while (0)
+
+struct Cdna4Cfg {
+    uint32_t nslots;
+    size_t   chunk;     /* staging fragment bytes                     */
+    size_t   fused_max; /* msg sizes <= this take the fused kernel    */
+};
+
+class Cdna4Tl;
+
+class Cdna4TlContext final : public TlContext {
+  public:
+    Cdna4TlContext(Context *ctx, Tl *tl, int dev) : TlContext(ctx), tl_(tl),
+                                                    dev_(dev)
+    {
+        hipStreamCreateWithFlags(&comp_, hipStreamNonBlocking);
+        hipStreamCreateWithFlags(&copy_, hipStreamNonBlocking);
+    }
+    ~Cdna4TlContext() override
+    {
+        if (comp_) {
+            hipStreamDestroy(comp_);
+        }
+        if (copy_) {
+            hipStreamDestroy(copy_);
+        }
+    }
+    Tl *iface() override;
+
+    Tl         *tl_;
+    int         dev_;
+    hipStream_t comp_ = nullptr, copy_ = nullptr;
+};
+
+struct PeerRes {
+    uint8_t  *scratch = nullptr;
+    uint64_t *flags   = nullptr;
+    bool      ipc_s = false, ipc_f = false;
+};
+
+struct ExchgBlob {
+    int32_t           pid;
+    int32_t           device;
+    uint64_t          scratch_ptr;
+    uint64_t          flags_ptr;
+    hipIpcMemHandle_t sh;
+    hipIpcMemHandle_t fh;
+};
+
+class Cdna4TlTeam final : public TlTeam {
+  public:
+    Cdna4TlTeam(TlContext *tlc, Team *team, const Cdna4Cfg &cfg)
+        : TlTeam(tlc, team), cfg_(cfg)
+    {
+        char buf[96];
+        snprintf(buf, sizeof(buf), "/uccamd-c4-%016llx-%u",
+                 (unsigned long long)team->team_uid, team->id);
+        seg_name_ = buf;
+        init_st_  = local_init();
+        if (init_st_ == UCC_OK && team->rank == 0) {
+            init_st_ = seg_.create(seg_name_, team->size, cfg_.nslots, 0);
+        }
+    }
+
+    ~Cdna4TlTeam() override
+    {
+        for (uint32_t r = 0; r < (uint32_t)peers_.size(); r++) {
+            if (peers_[r].ipc_s) {
+                hipIpcCloseMemHandle(peers_[r].scratch);
+            }
+            if (peers_[r].ipc_f) {
+                hipIpcCloseMemHandle(peers_[r].flags);
+            }
+        }
+        if (scratch_) {
+            hipFree(scratch_);
+        }
+        if (flags_) {
+            hipFree(flags_);
+        }
+        if (err_host_) {
+            hipHostFree(err_host_);
+        }
+    }
+
+    ucc_status_t local_init()
+    {
+        /* scratch: [slot][parity][in|out] areas of chunk bytes each */
+        scratch_bytes_ = (size_t)cfg_.nslots * 2 * 2 * cfg_.chunk;
+        HIPCHK(hipMalloc((void **)&scratch_, scratch_bytes_));
+        /* flags: fine-grained for cross-GPU system-scope atomics */
+        hipError_t e = hipExtMallocWithFlags((void **)&flags_, 4096,
+                                             hipDeviceMallocFinegrained);
+        if (e != hipSuccess) {
+            ucc_warn("finegrained alloc failed (%s), using hipMalloc",
+                     hipGetErrorString(e));
+            HIPCHK(hipMalloc((void **)&flags_, 4096));
+        }
+        HIPCHK(hipMemset(flags_, 0, 4096));
+        HIPCHK(hipHostMalloc((void **)&err_host_, 64, hipHostMallocDefault));
+        *err_host_ = 0;
+        HIPCHK(hipDeviceSynchronize());
+        return UCC_OK;
+    }
+
+    size_t exchg_size() override { return sizeof(ExchgBlob); }
+
+    void exchg_pack(void *buf) override
+    {
+        ExchgBlob b{};
+        b.pid         = (int32_t)getpid();
+        b.device      = ((Cdna4TlContext *)tlc_)->dev_;
+        b.scratch_ptr = (uint64_t)(uintptr_t)scratch_;
+        b.flags_ptr   = (uint64_t)(uintptr_t)flags_;
+        if (init_st_ == UCC_OK) {
+            hipIpcGetMemHandle(&b.sh, scratch_);
+            hipIpcGetMemHandle(&b.fh, flags_);
+        }
+        memcpy(buf, &b, sizeof(b));
+    }
+
+    ucc_status_t exchg_unpack(const void *all, size_t stride) override
+    {
+        if (init_st_ != UCC_OK) {
+            return init_st_;
+        }
+        const uint32_t n  = team_->size;
+        const int32_t  me = (int32_t)getpid();
+        peers_.resize(n);
+        for (uint32_t r = 0; r < n; r++) {
+            ExchgBlob b;
+            memcpy(&b, (const uint8_t *)all + r * stride, sizeof(b));
+            if (r == team_->rank) {
+                peers_[r].scratch = scratch_;
+                peers_[r].flags   = flags_;
+            } else if (b.pid == me) {
+                /* in-process jig: same address space */
+                peers_[r].scratch = (uint8_t *)(uintptr_t)b.scratch_ptr;
+                peers_[r].flags   = (uint64_t *)(uintptr_t)b.flags_ptr;
+            } else {
+                void *p = nullptr;
+                hipError_t e = hipIpcOpenMemHandle(
+                    &p, b.sh, hipIpcMemLazyEnablePeerAccess);
+                if (e != hipSuccess) {
+                    ucc_warn("IPC open scratch of rank %u failed: %s", r,
+                             hipGetErrorString(e));
+                    return UCC_ERR_NO_RESOURCE;
+                }
+                peers_[r].scratch = (uint8_t *)p;
+                peers_[r].ipc_s   = true;
+                e = hipIpcOpenMemHandle(&p, b.fh,
+                                        hipIpcMemLazyEnablePeerAccess);
+                if (e != hipSuccess) {
+                    ucc_warn("IPC open flags of rank %u failed: %s", r,
+                             hipGetErrorString(e));
+                    return UCC_ERR_NO_RESOURCE;
+                }
+                peers_[r].flags = (uint64_t *)p;
+                peers_[r].ipc_f = true;
+            }
+        }
+        return UCC_OK;
+    }
+
+    ucc_status_t create_test() override
+    {
+        if (init_st_ != UCC_OK) {
+            return init_st_;
+        }
+        if (team_->rank == 0) {
+            return UCC_OK;
+        }
+        if (!attached_) {
+            ucc_status_t st =
+                seg_.attach(seg_name_, team_->size, cfg_.nslots, 0);
+            if (st == UCC_INPROGRESS) {
+                return UCC_INPROGRESS;
+            }
+            if (st != UCC_OK) {
+                return st;
+            }
+            attached_ = true;
+        }
+        return seg_.ready() ? UCC_OK : UCC_INPROGRESS;
+    }
+
+    void get_scores(Team *team, ScoreMap &map) override;
+
+    uint8_t *area(uint32_t rank, uint32_t slot, uint32_t parity, int which)
+    {
+        return peers_[rank].scratch +
+               (((size_t)slot * 2 + parity) * 2 + which) * cfg_.chunk;
+    }
+
+    Cdna4Cfg    cfg_;
+    ShmSeg      seg_;
+    std::string seg_name_;
+    uint64_t    seq_ = 0;
+    uint8_t    *scratch_ = nullptr;
+    size_t      scratch_bytes_ = 0;
+    uint64_t   *flags_ = nullptr;  /* device fine-grained               */
+    uint64_t   *err_host_ = nullptr; /* host-pinned, device-writable    */
+    std::vector<PeerRes> peers_;
+    ucc_status_t init_st_ = UCC_OK;
+    bool         attached_ = false;
+};
+
+/* -------------------------------------------------------------- tasks  */
+class Cdna4Task : public Task {
+  public:
+    Cdna4Task(Context *ctx, Cdna4TlTeam *tt, const ucc_coll_args_t &args)
+        : Task(ctx), tt_(tt), a_(args)
+    {
+    }
+    ~Cdna4Task() override
+    {
+        for (auto ev : evs_) {
+            if (ev) {
+                hipEventDestroy(ev);
+            }
+        }
+    }
+
+  protected:
+    hipEvent_t ev(int i)
+    {
+        while ((int)evs_.size() <= i) {
+            evs_.push_back(nullptr);
+        }
+        if (!evs_[i]) {
+            hipEventCreateWithFlags(&evs_[i], hipEventDisableTiming);
+        }
+        return evs_[i];
+    }
+
+    void publish(uint64_t k)
+    {
+        tt_->seg_.step(slot_, me_)->store(base_ + k,
+                                          std::memory_order_release);
+    }
+    void close_slot()
+    {
+        tt_->seg_.step(slot_, me_)->store(base_ + kCap,
+                                          std::memory_order_release);
+    }
+    bool all_ge(uint64_t k)
+    {
+        for (uint32_t r = 0; r < n_; r++) {
+            if (tt_->seg_.step(slot_, r)->load(std::memory_order_acquire) <
+                base_ + k) {
+                return false;
+            }
+        }
+        return true;
+    }
+
+    void begin_use()
+    {
+        me_          = tt_->team_->rank;
+        n_           = tt_->team_->size;
+        uint64_t use = tt_->seq_++;
+        slot_        = (uint32_t)(use % tt_->cfg_.nslots);
+        base_        = (use / tt_->cfg_.nslots) * kCap;
+        fseq_        = use / tt_->cfg_.nslots + 1;
+        phase_       = 0;
+        frag_        = 0;
+    }
+
+    hipStream_t comp() { return ((Cdna4TlContext *)tt_->tlc_)->comp_; }
+    hipStream_t copy_s() { return ((Cdna4TlContext *)tt_->tlc_)->copy_; }
+
+    Cdna4TlTeam    *tt_;
+    ucc_coll_args_t a_;
+    std::vector<hipEvent_t> evs_;
+    uint32_t me_ = 0, n_ = 1, slot_ = 0;
+    uint64_t base_ = 0, fseq_ = 0;
+    int      phase_ = 0;
+    size_t   frag_  = 0;
+};
+
+/* Fused single-kernel allreduce (small messages). */
+class FusedAllreduceTask final : public Cdna4Task {
+  public:
+    using Cdna4Task::Cdna4Task;
+
+    ucc_status_t post() override
+    {
+        begin_use();
+        bytes_ = a_.dst.info.count * ucc_dt_size(a_.dst.info.datatype);
+        status = UCC_INPROGRESS;
+        return progress();
+    }
+
+    ucc_status_t progress() override
+    {
+        if (phase_ == 0) {
+            if (!all_ge(0)) {
+                return UCC_INPROGRESS;
+            }
+            ec_hip::FusedArgs fa{};
+            const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+            fa.dst     = a_.dst.info.buffer;
+            fa.src     = inplace ? fa.dst : a_.src.info.buffer;
+            fa.count   = a_.dst.info.count;
+            fa.dt      = a_.dst.info.datatype;
+            fa.op      = a_.op == UCC_OP_AVG ? UCC_OP_SUM : a_.op;
+            fa.alpha   = a_.op == UCC_OP_AVG ? 1.0f / (float)n_ : 1.0f;
+            if (a_.op == UCC_OP_AVG) {
+                fa.op = (ucc_reduction_op_t)12; /* SUM+alpha path */
+            }
+            fa.my_scratch = tt_->area(me_, slot_, 0, 0);
+            for (uint32_t r = 0; r < n_; r++) {
+                fa.peer_scratch[r] = tt_->area(r, slot_, 0, 0);
+                fa.peer_flags[r]   = tt_->peers_[r].flags;
+            }
+            fa.local_flags = tt_->flags_;
+            fa.rank        = (int)me_;
+            fa.nranks      = (int)n_;
+            fa.slot        = (int)slot_;
+            fa.seq         = fseq_;
+            fa.error_word  = tt_->err_host_;
+            size_t blocks  = (bytes_ + 128 * 1024 - 1) / (128 * 1024);
+            fa.nblocks     = (int)(blocks < 1 ? 1 : blocks > 16 ? 16 : blocks);
+            ucc_status_t st = ec_hip::fused_allreduce(fa, comp());
+            if (st != UCC_OK) {
+                return st;
+            }
+            hipEventRecord(ev(0), comp());
+            phase_ = 1;
+        }
+        if (phase_ == 1) {
+            hipError_t e = hipEventQuery(ev(0));
+            if (e == hipErrorNotReady) {
+                return UCC_INPROGRESS;
+            }
+            if (e != hipSuccess) {
+                return UCC_ERR_NO_RESOURCE;
+            }
+            if (*tt_->err_host_ != 0) {
+                ucc_error("fused allreduce timed out waiting for peers");
+                return UCC_ERR_TIMED_OUT;
+            }
+            close_slot();
+            return UCC_OK;
+        }
+        return UCC_INPROGRESS;
+    }
+
+  private:
+    size_t bytes_ = 0;
+};
+
+/* Staged linear collectives: allreduce / allgather(v) / reduce_scatter(v) /
+ * bcast / reduce. Per fragment (parity p), phases:
+ *   A-issue (gated) -> A-wait (event) -> publish kA
+ *   B-gate all>=kA  -> issue copy(f+1), launch compute -> B-wait -> kB
+ *   [allreduce only] C-gate all>=kB -> gather -> C-wait -> kC           */
+class StagedTask final : public Cdna4Task {
+  public:
+    using Cdna4Task::Cdna4Task;
+
+    ucc_status_t post() override
+    {
+        begin_use();
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        ct_     = a_.coll_type;
+        steps_  = ct_ == UCC_COLL_TYPE_ALLREDUCE ? 3 : 2;
+        alpha_  = 1.0f;
+        op_     = a_.op;
+        if (op_ == UCC_OP_AVG) {
+            op_    = (ucc_reduction_op_t)12;
+            alpha_ = 1.0f / (float)n_;
+        }
+        cnt_.assign(n_, 0);
+        dsp_.assign(n_, 0);
+        switch (ct_) {
+        case UCC_COLL_TYPE_ALLREDUCE:
+            dt_    = a_.dst.info.datatype;
+            dtsz_  = ucc_dt_size(dt_);
+            total_ = a_.dst.info.count * dtsz_;
+            dbuf_  = (uint8_t *)a_.dst.info.buffer;
+            sbuf_  = inplace ? dbuf_ : (const uint8_t *)a_.src.info.buffer;
+            nfrags_ = (total_ + tt_->cfg_.chunk - 1) / tt_->cfg_.chunk;
+            break;
+        case UCC_COLL_TYPE_REDUCE:
+            dt_    = a_.src.info.datatype;
+            dtsz_  = ucc_dt_size(dt_);
+            total_ = a_.src.info.count * dtsz_;
+            dbuf_  = (uint8_t *)a_.dst.info.buffer;
+            sbuf_  = (inplace && me_ == a_.root)
+                         ? dbuf_
+                         : (const uint8_t *)a_.src.info.buffer;
+            nfrags_ = (total_ + tt_->cfg_.chunk - 1) / tt_->cfg_.chunk;
+            break;
+        case UCC_COLL_TYPE_BCAST:
+            dt_    = a_.src.info.datatype;
+            dtsz_  = ucc_dt_size(dt_);
+            total_ = a_.src.info.count * dtsz_;
+            dbuf_  = (uint8_t *)a_.src.info.buffer;
+            sbuf_  = dbuf_;
+            nfrags_ = (total_ + tt_->cfg_.chunk - 1) / tt_->cfg_.chunk;
+            break;
+        case UCC_COLL_TYPE_REDUCE_SCATTER: {
+            dt_   = a_.dst.info.datatype;
+            dtsz_ = ucc_dt_size(dt_);
+            size_t out_b;
+            if (inplace) {
+                total_ = a_.dst.info.count * dtsz_;
+                out_b  = total_ / n_;
+                sbuf_  = (uint8_t *)a_.dst.info.buffer;
+                dbuf_  = (uint8_t *)a_.dst.info.buffer + me_ * out_b;
+            } else {
+                out_b  = a_.dst.info.count * dtsz_;
+                total_ = out_b * n_;
+                sbuf_  = (const uint8_t *)a_.src.info.buffer;
+                dbuf_  = (uint8_t *)a_.dst.info.buffer;
+            }
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = out_b;
+                dsp_[r] = (size_t)r * out_b;
+            }
+            nfrags_ = (total_ + tt_->cfg_.chunk - 1) / tt_->cfg_.chunk;
+            break;
+        }
+        case UCC_COLL_TYPE_ALLGATHER: {
+            dt_          = a_.dst.info.datatype;
+            dtsz_        = ucc_dt_size(dt_);
+            size_t block = a_.dst.info.count * dtsz_ / n_;
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = block;
+                dsp_[r] = (size_t)r * block;
+            }
+            dbuf_ = (uint8_t *)a_.dst.info.buffer;
+            sbuf_ = inplace ? dbuf_ + dsp_[me_]
+                            : (const uint8_t *)a_.src.info.buffer;
+            total_  = block;
+            nfrags_ = (block + tt_->cfg_.chunk - 1) / tt_->cfg_.chunk;
+            break;
+        }
+        default: return UCC_ERR_NOT_SUPPORTED;
+        }
+        if (nfrags_ == 0) {
+            nfrags_ = 1;
+        }
+        status = UCC_INPROGRESS;
+        return progress();
+    }
+
+    ucc_status_t progress() override
+    {
+        const size_t chunk = tt_->cfg_.chunk;
+        while (frag_ < nfrags_) {
+            const size_t   f   = frag_;
+            const uint32_t p   = (uint32_t)(f & 1);
+            const uint64_t kA  = steps_ * f + 1;
+            const uint64_t kB  = steps_ * f + 2;
+            const uint64_t kC  = steps_ * f + 3;
+            const size_t   off = f * chunk;
+            const size_t   len =
+                total_ - off < chunk ? total_ - off : chunk;
+            switch (phase_) {
+            case 0: /* entry: first copy */
+                if (!all_ge(0)) {
+                    return UCC_INPROGRESS;
+                }
+                issue_copy(0);
+                phase_ = 1;
+                break;
+            case 1: { /* A-wait */
+                if (copy_pending_) {
+                    hipError_t e = hipEventQuery(ev(0 + (int)p));
+                    if (e == hipErrorNotReady) {
+                        return UCC_INPROGRESS;
+                    }
+                    if (e != hipSuccess) {
+                        return UCC_ERR_NO_RESOURCE;
+                    }
+                }
+                publish(kA);
+                phase_ = 2;
+                break;
+            }
+            case 2: { /* B-gate */
+                if (!all_ge(kA)) {
+                    return UCC_INPROGRESS;
+                }
+                if (f + 1 < nfrags_) {
+                    issue_copy(f + 1);
+                }
+                bool launched = launch_compute(f, p, off, len);
+                if (launched) {
+                    hipEventRecord(ev(2), comp());
+                }
+                b_launched_ = launched;
+                phase_      = 3;
+                break;
+            }
+            case 3: { /* B-wait */
+                if (b_launched_) {
+                    hipError_t e = hipEventQuery(ev(2));
+                    if (e == hipErrorNotReady) {
+                        return UCC_INPROGRESS;
+                    }
+                    if (e != hipSuccess) {
+                        return UCC_ERR_NO_RESOURCE;
+                    }
+                }
+                publish(kB);
+                if (steps_ == 2) {
+                    phase_ = 1;
+                    frag_++;
+                    break;
+                }
+                phase_ = 4;
+                break;
+            }
+            case 4: { /* C-gate (allreduce gather) */
+                if (!all_ge(kB)) {
+                    return UCC_INPROGRESS;
+                }
+                launch_gather(f, p, off, len);
+                hipEventRecord(ev(3), comp());
+                phase_ = 5;
+                break;
+            }
+            case 5: { /* C-wait */
+                hipError_t e = hipEventQuery(ev(3));
+                if (e == hipErrorNotReady) {
+                    return UCC_INPROGRESS;
+                }
+                if (e != hipSuccess) {
+                    return UCC_ERR_NO_RESOURCE;
+                }
+                publish(kC);
+                phase_ = 1;
+                frag_++;
+                break;
+            }
+            }
+        }
+        close_slot();
+        return UCC_OK;
+    }
+
+  private:
+    /* stage my contribution of fragment f into in[slot][f%2] */
+    void issue_copy(size_t f)
+    {
+        const uint32_t p   = (uint32_t)(f & 1);
+        const size_t   off = f * tt_->cfg_.chunk;
+        size_t         len = 0;
+        const uint8_t *src = nullptr;
+        if (ct_ == UCC_COLL_TYPE_ALLGATHER) {
+            if (off < cnt_[me_]) {
+                len = cnt_[me_] - off < tt_->cfg_.chunk ? cnt_[me_] - off
+                                                        : tt_->cfg_.chunk;
+                src = sbuf_ + off;
+            }
+        } else if (ct_ == UCC_COLL_TYPE_BCAST) {
+            if (me_ == a_.root && off < total_) {
+                len = total_ - off < tt_->cfg_.chunk ? total_ - off
+                                                     : tt_->cfg_.chunk;
+                src = sbuf_ + off;
+            }
+        } else {
+            if (off < total_) {
+                len = total_ - off < tt_->cfg_.chunk ? total_ - off
+                                                     : tt_->cfg_.chunk;
+                src = sbuf_ + off;
+            }
+        }
+        copy_pending_ = len > 0;
+        if (len > 0) {
+            hipMemcpyAsync(tt_->area(me_, slot_, p, 0), src, len,
+                           hipMemcpyDeviceToDevice, copy_s());
+        }
+        hipEventRecord(ev(0 + (int)p), copy_s());
+    }
+
+    /* B compute for fragment f; returns whether a kernel was launched */
+    bool launch_compute(size_t f, uint32_t p, size_t off, size_t len)
+    {
+        switch (ct_) {
+        case UCC_COLL_TYPE_ALLREDUCE: {
+            /* reduce my 1/n slice of the fragment into out area */
+            size_t nelem = len / dtsz_;
+            size_t gran  = 256 / dtsz_;
+            size_t per   = (nelem / n_) / gran * gran;
+            size_t b     = me_ * per;
+            size_t e     = me_ == n_ - 1 ? nelem : b + per;
+            if (e <= b) {
+                return false;
+            }
+            ec_hip::ReduceArgs ra{};
+            ra.dst    = tt_->area(me_, slot_, p, 1);
+            ra.n_srcs = (int)n_;
+            ra.count  = e - b;
+            ra.dt     = dt_;
+            ra.op     = op_;
+            ra.alpha  = alpha_;
+            for (uint32_t r = 0; r < n_; r++) {
+                ra.srcs[r] = tt_->area(r, slot_, p, 0) + b * dtsz_;
+            }
+            ec_hip::reduce(ra, comp());
+            return true;
+        }
+        case UCC_COLL_TYPE_REDUCE: {
+            if (me_ != a_.root) {
+                return false;
+            }
+            ec_hip::ReduceArgs ra{};
+            ra.dst    = dbuf_ + off;
+            ra.n_srcs = (int)n_;
+            ra.count  = len / dtsz_;
+            ra.dt     = dt_;
+            ra.op     = op_;
+            ra.alpha  = alpha_;
+            for (uint32_t r = 0; r < n_; r++) {
+                ra.srcs[r] = tt_->area(r, slot_, p, 0);
+            }
+            ec_hip::reduce(ra, comp());
+            return true;
+        }
+        case UCC_COLL_TYPE_REDUCE_SCATTER: {
+            /* my dst slice ∩ fragment */
+            size_t s0 = dsp_[me_], s1 = dsp_[me_] + cnt_[me_];
+            size_t f0 = off, f1 = off + len;
+            size_t b  = s0 > f0 ? s0 : f0;
+            size_t e  = s1 < f1 ? s1 : f1;
+            if (e <= b) {
+                return false;
+            }
+            ec_hip::ReduceArgs ra{};
+            ra.dst    = dbuf_ + (b - s0);
+            ra.n_srcs = (int)n_;
+            ra.count  = (e - b) / dtsz_;
+            ra.dt     = dt_;
+            ra.op     = op_;
+            ra.alpha  = alpha_;
+            for (uint32_t r = 0; r < n_; r++) {
+                ra.srcs[r] = tt_->area(r, slot_, p, 0) + (b - off);
+            }
+            ec_hip::reduce(ra, comp());
+            return true;
+        }
+        case UCC_COLL_TYPE_ALLGATHER: {
+            ec_hip::GatherArgs ga{};
+            int k = 0;
+            for (uint32_t r = 0; r < n_; r++) {
+                if (off >= cnt_[r]) {
+                    continue;
+                }
+                size_t l = cnt_[r] - off < tt_->cfg_.chunk
+                               ? cnt_[r] - off
+                               : tt_->cfg_.chunk;
+                if (r == me_ && (a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE)) {
+                    continue;
+                }
+                ga.srcs[k] = tt_->area(r, slot_, p, 0);
+                ga.offs[k] = dsp_[r] + off;
+                ga.lens[k] = l;
+                k++;
+            }
+            if (k == 0) {
+                return false;
+            }
+            ga.dst_base = dbuf_;
+            ga.n        = k;
+            ec_hip::gather_copy(ga, comp());
+            return true;
+        }
+        case UCC_COLL_TYPE_BCAST: {
+            if (me_ == a_.root) {
+                return false;
+            }
+            ec_hip::GatherArgs ga{};
+            ga.dst_base = dbuf_;
+            ga.srcs[0]  = tt_->area((uint32_t)a_.root, slot_, p, 0);
+            ga.offs[0]  = off;
+            ga.lens[0]  = len;
+            ga.n        = 1;
+            ec_hip::gather_copy(ga, comp());
+            return true;
+        }
+        default: return false;
+        }
+    }
+
+    /* allreduce stage C: gather every rank's reduced slice */
+    void launch_gather(size_t f, uint32_t p, size_t off, size_t len)
+    {
+        (void)f;
+        size_t nelem = len / dtsz_;
+        size_t gran  = 256 / dtsz_;
+        size_t per   = (nelem / n_) / gran * gran;
+        ec_hip::GatherArgs ga{};
+        ga.dst_base = dbuf_ + off;
+        int k       = 0;
+        for (uint32_t r = 0; r < n_; r++) {
+            size_t b = r * per;
+            size_t e = r == n_ - 1 ? nelem : b + per;
+            if (e <= b) {
+                continue;
+            }
+            ga.srcs[k] = tt_->area(r, slot_, p, 1);
+            ga.offs[k] = b * dtsz_;
+            ga.lens[k] = (e - b) * dtsz_;
+            k++;
+        }
+        ga.n = k;
+        ec_hip::gather_copy(ga, comp());
+    }
+
+    ucc_coll_type_t ct_ = UCC_COLL_TYPE_ALLREDUCE;
+    const uint8_t  *sbuf_ = nullptr;
+    uint8_t        *dbuf_ = nullptr;
+    ucc_datatype_t  dt_   = UCC_DT_INT8;
+    size_t          dtsz_ = 1;
+    ucc_reduction_op_t op_ = UCC_OP_SUM;
+    float           alpha_ = 1.0f;
+    size_t          total_ = 0, nfrags_ = 0;
+    std::vector<size_t> cnt_, dsp_;
+    int             steps_ = 3;
+    bool            copy_pending_ = false, b_launched_ = false;
+};
+
+/* ------------------------------------------------------------ scoring  */
+class Cdna4Tl final : public Tl {
+  public:
+    const char *name() const override { return "cdna4"; }
+    int         default_score() const override { return 80; }
+
+    TlContext *context_create(Context *ctx) override
+    {
+        if (!mc::hip_available()) {
+            return nullptr;
+        }
+        int dev = 0;
+        if (hipGetDevice(&dev) != hipSuccess) {
+            return nullptr;
+        }
+        ctx->proc.device = dev;
+        return new Cdna4TlContext(ctx, this, dev);
+    }
+
+    TlTeam *team_create(TlContext *tlc, Team *team) override
+    {
+        if (team->size < 2 || team->size > (uint32_t)kMaxRanks ||
+            !team->all_same_node() || !team->all_have_device()) {
+            return nullptr;
+        }
+        auto    &cfg = Config::instance();
+        Cdna4Cfg c;
+        c.nslots    = (uint32_t)cfg.get_int("TL_CDNA4", "MAX_CONCURRENT", 4);
+        c.chunk     = cfg.get_size("TL_CDNA4", "CHUNK_SIZE", 8 * 1024 * 1024);
+        c.fused_max = cfg.get_size("TL_CDNA4", "FUSED_MAX", 512 * 1024);
+        return new Cdna4TlTeam(tlc, team, c);
+    }
+};
+
+static Cdna4Tl g_cdna4_tl;
+
+Tl *Cdna4TlContext::iface() { return &g_cdna4_tl; }
+
+void Cdna4TlTeam::get_scores(Team *team, ScoreMap &map)
+{
+    (void)team;
+    Cdna4TlTeam *self = this;
+    auto add = [&](ucc_coll_type_t ct, size_t lo, size_t hi, int score,
+                   const char *alg, bool fused) {
+        ScoreRange r;
+        r.start    = lo;
+        r.end      = hi;
+        r.score    = score;
+        r.tl_name  = "cdna4";
+        r.alg_name = alg;
+        r.init     = [self, fused](const ucc_coll_args_t &args, Team *t,
+                               Task **task) -> ucc_status_t {
+            /* reductions need a supported dtype x op on device */
+            if (args.coll_type == UCC_COLL_TYPE_ALLREDUCE ||
+                args.coll_type == UCC_COLL_TYPE_REDUCE ||
+                args.coll_type == UCC_COLL_TYPE_REDUCE_SCATTER) {
+                ucc_datatype_t dt = args.dst.info.datatype;
+                if (!ec_hip::op_supported(dt, args.op)) {
+                    return UCC_ERR_NOT_SUPPORTED;
+                }
+            }
+            if (fused) {
+                *task = new FusedAllreduceTask(t->ctx, self, args);
+            } else {
+                *task = new StagedTask(t->ctx, self, args);
+            }
+            return UCC_OK;
+        };
+        for (auto mt : {UCC_MEMORY_TYPE_CUDA, UCC_MEMORY_TYPE_CUDA_MANAGED}) {
+            map.add(ct, mt, r);
+        }
+    };
+    add(UCC_COLL_TYPE_ALLREDUCE, 0, cfg_.fused_max, 100, "fused", true);
+    add(UCC_COLL_TYPE_ALLREDUCE, 0, SIZE_MAX, 80, "staged_linear", false);
+    add(UCC_COLL_TYPE_ALLGATHER, 0, SIZE_MAX, 80, "staged_linear", false);
+    add(UCC_COLL_TYPE_REDUCE_SCATTER, 0, SIZE_MAX, 80, "staged_linear",
+        false);
+    add(UCC_COLL_TYPE_BCAST, 0, SIZE_MAX, 80, "staged_linear", false);
+    add(UCC_COLL_TYPE_REDUCE, 0, SIZE_MAX, 80, "staged_linear", false);
+}
+
+} // namespace
+
+Tl *tl_cdna4_iface() { return &g_cdna4_tl; }
+
+} // namespace ucc
