@@ -22,6 +22,7 @@ namespace py = pybind11;
 
 extern "C" ucc_status_t ucc_amd_coll_from_name_c(const char *,
                                                  ucc_coll_type_t *);
+extern "C" int ucc_amd_hip_device_count_c();
 
 /* ------------------------------------------------------------- LocalOob */
 struct LocalOobShared {
@@ -389,4 +390,6 @@ PYBIND11_MODULE(_core, m)
     m.attr("FLAG_PERSISTENT") = (uint64_t)UCC_COLL_ARGS_FLAG_PERSISTENT;
     m.def("dt_size", [](int dt) { return ucc_dt_size((ucc_datatype_t)dt); });
     m.def("version", []() { return std::string(ucc_get_version_string()); });
+    m.def("hip_device_count",
+          []() { return ucc_amd_hip_device_count_c(); });
 }

@@ -58,6 +58,8 @@ struct FusedArgs {
     uint64_t   *peer_flags[kMaxRanks];
     int         rank, nranks, slot;
     uint64_t    seq;
+    uint64_t    stage_target; /* cumulative block arrivals for this slot
+                                 including this launch (host-tracked)   */
     ucc_datatype_t     dt;
     ucc_reduction_op_t op;
     float       alpha;

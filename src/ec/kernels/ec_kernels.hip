@@ -392,7 +392,7 @@ __global__ void k_fused_allreduce(FusedArgs a)
     if (blockIdx.x == 0) {
         if (threadIdx.x == 0) {
             uint64_t spins = 0;
-            while (sys_load(stage_cnt) < a.seq * (uint64_t)a.nblocks) {
+            while (sys_load(stage_cnt) < a.stage_target) {
                 if (++spins > kSpinLimit) {
                     sys_store(a.error_word, 1);
                     return;

@@ -10,14 +10,19 @@ namespace mc {
 int hip_device_count()
 {
     static int count = []() {
-        int n = 0;
-        if (hipGetDeviceCount(&n) != hipSuccess) {
+        int        n = 0;
+        hipError_t e = hipGetDeviceCount(&n);
+        if (e != hipSuccess) {
+            UCC_LOG(LogLevel::DEBUG, "mc", "hipGetDeviceCount: %s",
+                    hipGetErrorString(e));
             return 0;
         }
         return n;
     }();
     return count;
 }
+
+extern "C" int ucc_amd_hip_device_count_c() { return hip_device_count(); }
 
 bool hip_available() { return hip_device_count() > 0; }
 

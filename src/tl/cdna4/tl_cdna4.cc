@@ -240,6 +240,7 @@ class Cdna4TlTeam final : public TlTeam {
     ShmSeg      seg_;
     std::string seg_name_;
     uint64_t    seq_ = 0;
+    std::vector<uint64_t> stage_cum_; /* per-slot fused block arrivals */
     uint8_t    *scratch_ = nullptr;
     size_t      scratch_bytes_ = 0;
     uint64_t   *flags_ = nullptr;  /* device fine-grained               */
@@ -365,6 +366,11 @@ class FusedAllreduceTask final : public Cdna4Task {
             fa.error_word  = tt_->err_host_;
             size_t blocks  = (bytes_ + 128 * 1024 - 1) / (128 * 1024);
             fa.nblocks     = (int)(blocks < 1 ? 1 : blocks > 16 ? 16 : blocks);
+            if (tt_->stage_cum_.size() < tt_->cfg_.nslots) {
+                tt_->stage_cum_.assign(tt_->cfg_.nslots, 0);
+            }
+            tt_->stage_cum_[slot_] += (uint64_t)fa.nblocks;
+            fa.stage_target = tt_->stage_cum_[slot_];
             ucc_status_t st = ec_hip::fused_allreduce(fa, comp());
             if (st != UCC_OK) {
                 return st;
@@ -382,6 +388,7 @@ class FusedAllreduceTask final : public Cdna4Task {
             }
             if (*tt_->err_host_ != 0) {
                 ucc_error("fused allreduce timed out waiting for peers");
+                close_slot();
                 return UCC_ERR_TIMED_OUT;
             }
             close_slot();
@@ -830,6 +837,20 @@ void Cdna4TlTeam::get_scores(Team *team, ScoreMap &map)
                 }
             }
             if (fused) {
+                /* The fused kernel spin-waits for every peer's kernel.
+                 * With several ranks of one process on one GPU (the
+                 * in-process jig) their streams can share a HW queue and
+                 * event barrier packets serialize the dispatches ->
+                 * deadlock. One process per GPU (production) is safe;
+                 * same-process multi-rank falls back to the host-gated
+                 * staged path. */
+                int same_proc = 0;
+                for (auto &p : t->procs) {
+                    same_proc += (p.pid == t->ctx->proc.pid);
+                }
+                if (same_proc > 1) {
+                    return UCC_ERR_NOT_SUPPORTED;
+                }
                 *task = new FusedAllreduceTask(t->ctx, self, args);
             } else {
                 *task = new StagedTask(t->ctx, self, args);

@@ -1,0 +1,120 @@
+"""Multi-process collectives: 2 OS processes, gloo bootstrap, ucc_amd
+teams. The CPU variant runs everywhere (host shm TL); the GPU variant
+shares one MI355X between 2 processes and exercises the REAL production
+shape of the cdna4 TL: HIP-IPC handle exchange + the fused single-kernel
+allreduce with cross-process system-scope arrival flags.
+"""
+
+import multiprocessing as mp
+import os
+
+import numpy as np
+import pytest
+
+
+def _worker_host(rank, world, port, q):
+    try:
+        import torch.distributed as dist
+
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from ucc_amd.parallel import Communicator
+
+        import torch
+
+        comm = Communicator()
+        t = torch.arange(1000, dtype=torch.float32) * (rank + 1)
+        expected = t.clone() / (rank + 1) * sum(r + 1 for r in range(world))
+        comm.allreduce(t)
+        assert torch.allclose(t, expected), (t[:5], expected[:5])
+        comm.barrier()
+        q.put((rank, "ok"))
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"fail: {e!r}"))
+
+
+def _worker_gpu(rank, world, port, q):
+    try:
+        import torch
+        import torch.distributed as dist
+
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        torch.cuda.set_device(0)  # both procs share GPU 0 on the test box
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from ucc_amd.parallel import Communicator
+
+        comm = Communicator()
+        # small -> fused single-kernel path (cross-process arrival flags)
+        torch.manual_seed(100 + rank)
+        small = torch.randn(2048, device="cuda")
+        ref = small.clone()
+        comm.allreduce(small)
+        # verify against gloo host reduction of the same values
+        hostref = ref.cpu()
+        dist.all_reduce(hostref)
+        torch.cuda.synchronize()
+        assert torch.allclose(small.cpu(), hostref, rtol=1e-5, atol=1e-5)
+        # large -> staged linear path (peer reads over IPC mapping)
+        big = torch.randn(3_000_000, device="cuda")
+        bref = big.cpu()
+        comm.allreduce(big)
+        dist.all_reduce(bref)
+        torch.cuda.synchronize()
+        assert torch.allclose(big.cpu(), bref, rtol=1e-5, atol=1e-4)
+        # reduce_scatter + allgather (ZeRO pattern)
+        per = 250_000
+        src = torch.randn(per * world, device="cuda")
+        sref = src.cpu()
+        dst = torch.zeros(per, device="cuda")
+        comm.reduce_scatter(src, dst)
+        dist.all_reduce(sref)
+        assert torch.allclose(dst.cpu(),
+                              sref[rank * per:(rank + 1) * per],
+                              rtol=1e-5, atol=1e-4)
+        ag = torch.zeros(per * world, device="cuda")
+        comm.allgather(dst, ag)
+        torch.cuda.synchronize()
+        q.put((rank, "ok"))
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"fail: {e!r}"))
+
+
+def _run(worker, world=2, timeout=180):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29531 + os.getpid() % 1000
+    procs = [ctx.Process(target=worker, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    import queue as qmod
+
+    try:
+        for _ in range(world):
+            rank, msg = q.get(timeout=timeout)
+            results[rank] = msg
+    except qmod.Empty:
+        raise TimeoutError(f"workers did not finish: got {results}")
+    finally:
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
+    assert all(m == "ok" for m in results.values()), results
+
+
+def test_multiproc_host_allreduce():
+    _run(_worker_host)
+
+
+@pytest.mark.gpu
+def test_multiproc_gpu_colls():
+    torch = pytest.importorskip("torch")
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    _run(_worker_gpu)
