@@ -35,6 +35,23 @@ namespace {
 
 using ec_hip::kMaxRanks;
 
+/* v-coll count/displacement accessors honoring the 64-bit flags
+ * (reference: ucc_coll_utils count conventions). */
+static inline size_t coll_count_at(const ucc_coll_args_t &a, const void *c,
+                                   uint32_t i)
+{
+    return (a.flags & UCC_COLL_ARGS_FLAG_COUNT_64BIT)
+               ? (size_t)((const uint64_t *)c)[i]
+               : (size_t)((const uint32_t *)c)[i];
+}
+static inline size_t coll_disp_at(const ucc_coll_args_t &a, const void *d,
+                                  uint32_t i)
+{
+    return (a.flags & UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT)
+               ? (size_t)((const uint64_t *)d)[i]
+               : (size_t)((const uint32_t *)d)[i];
+}
+
 #define HIPCHK(expr)                                                         \
     do {                                                                     \
         hipError_t _e = (expr);                                              \
@@ -402,10 +419,16 @@ class FusedAllreduceTask final : public Cdna4Task {
 };
 
 /* Staged linear collectives: allreduce / allgather(v) / reduce_scatter(v) /
- * bcast / reduce. Per fragment (parity p), phases:
+ * bcast / reduce / alltoall(v) / gather(v) / scatter(v). Per fragment
+ * (parity p), phases:
  *   A-issue (gated) -> A-wait (event) -> publish kA
  *   B-gate all>=kA  -> issue copy(f+1), launch compute -> B-wait -> kB
- *   [allreduce only] C-gate all>=kB -> gather -> C-wait -> kC           */
+ *   [allreduce only] C-gate all>=kB -> gather -> C-wait -> kC
+ * Block colls fragment with granularity `chunk`; per-peer colls (alltoall,
+ * scatter) partition the staging area into n cells of `cell_` bytes and
+ * fragment per-peer blocks with granularity cell_. Ranks may have different
+ * nfrags (v-colls): close_slot() publishes base+kCap so early finishers
+ * never stall peers' all_ge() gates (slot_seg.h protocol).               */
 class StagedTask final : public Cdna4Task {
   public:
     using Cdna4Task::Cdna4Task;
@@ -424,6 +447,10 @@ class StagedTask final : public Cdna4Task {
         }
         cnt_.assign(n_, 0);
         dsp_.assign(n_, 0);
+        rcnt_.assign(n_, 0);
+        rdsp_.assign(n_, 0);
+        gran_ = tt_->cfg_.chunk;
+        cell_ = (tt_->cfg_.chunk / n_) & ~(size_t)255;
         switch (ct_) {
         case UCC_COLL_TYPE_ALLREDUCE:
             dt_    = a_.dst.info.datatype;
@@ -431,7 +458,6 @@ class StagedTask final : public Cdna4Task {
             total_ = a_.dst.info.count * dtsz_;
             dbuf_  = (uint8_t *)a_.dst.info.buffer;
             sbuf_  = inplace ? dbuf_ : (const uint8_t *)a_.src.info.buffer;
-            nfrags_ = (total_ + tt_->cfg_.chunk - 1) / tt_->cfg_.chunk;
             break;
         case UCC_COLL_TYPE_REDUCE:
             dt_    = a_.src.info.datatype;
@@ -441,7 +467,6 @@ class StagedTask final : public Cdna4Task {
             sbuf_  = (inplace && me_ == a_.root)
                          ? dbuf_
                          : (const uint8_t *)a_.src.info.buffer;
-            nfrags_ = (total_ + tt_->cfg_.chunk - 1) / tt_->cfg_.chunk;
             break;
         case UCC_COLL_TYPE_BCAST:
             dt_    = a_.src.info.datatype;
@@ -449,7 +474,6 @@ class StagedTask final : public Cdna4Task {
             total_ = a_.src.info.count * dtsz_;
             dbuf_  = (uint8_t *)a_.src.info.buffer;
             sbuf_  = dbuf_;
-            nfrags_ = (total_ + tt_->cfg_.chunk - 1) / tt_->cfg_.chunk;
             break;
         case UCC_COLL_TYPE_REDUCE_SCATTER: {
             dt_   = a_.dst.info.datatype;
@@ -470,7 +494,24 @@ class StagedTask final : public Cdna4Task {
                 cnt_[r] = out_b;
                 dsp_[r] = (size_t)r * out_b;
             }
-            nfrags_ = (total_ + tt_->cfg_.chunk - 1) / tt_->cfg_.chunk;
+            break;
+        }
+        case UCC_COLL_TYPE_REDUCE_SCATTERV: {
+            dt_   = a_.dst.info_v.datatype;
+            dtsz_ = ucc_dt_size(dt_);
+            size_t off = 0;
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = coll_count_at(a_, a_.dst.info_v.counts, r) * dtsz_;
+                dsp_[r] = off;
+                off += cnt_[r];
+            }
+            total_ = off;
+            sbuf_  = inplace ? (const uint8_t *)a_.dst.info_v.buffer
+                             : (const uint8_t *)a_.src.info.buffer;
+            dbuf_  = inplace
+                         ? (uint8_t *)a_.dst.info_v.buffer + dsp_[me_]
+                         : (uint8_t *)a_.dst.info_v.buffer;
+            ct_ = UCC_COLL_TYPE_REDUCE_SCATTER; /* same B compute */
             break;
         }
         case UCC_COLL_TYPE_ALLGATHER: {
@@ -485,11 +526,160 @@ class StagedTask final : public Cdna4Task {
             sbuf_ = inplace ? dbuf_ + dsp_[me_]
                             : (const uint8_t *)a_.src.info.buffer;
             total_  = block;
-            nfrags_ = (block + tt_->cfg_.chunk - 1) / tt_->cfg_.chunk;
+            break;
+        }
+        case UCC_COLL_TYPE_ALLGATHERV: {
+            dt_   = a_.dst.info_v.datatype;
+            dtsz_ = ucc_dt_size(dt_);
+            size_t maxb = 0;
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = coll_count_at(a_, a_.dst.info_v.counts, r) * dtsz_;
+                dsp_[r] =
+                    coll_disp_at(a_, a_.dst.info_v.displacements, r) * dtsz_;
+                maxb = cnt_[r] > maxb ? cnt_[r] : maxb;
+            }
+            dbuf_ = (uint8_t *)a_.dst.info_v.buffer;
+            sbuf_ = inplace ? dbuf_ + dsp_[me_]
+                            : (const uint8_t *)a_.src.info.buffer;
+            total_ = maxb;
+            ct_    = UCC_COLL_TYPE_ALLGATHER; /* same A/B machinery */
+            break;
+        }
+        case UCC_COLL_TYPE_ALLTOALL: {
+            if (inplace || cell_ == 0) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            dt_          = a_.dst.info.datatype;
+            dtsz_        = ucc_dt_size(dt_);
+            size_t block = a_.dst.info.count * dtsz_ / n_;
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = rcnt_[r] = block;
+                dsp_[r] = rdsp_[r] = (size_t)r * block;
+            }
+            sbuf_  = (const uint8_t *)a_.src.info.buffer;
+            dbuf_  = (uint8_t *)a_.dst.info.buffer;
+            total_ = block;
+            gran_  = cell_;
+            break;
+        }
+        case UCC_COLL_TYPE_ALLTOALLV: {
+            if (inplace || cell_ == 0) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            dt_   = a_.src.info_v.datatype;
+            dtsz_ = ucc_dt_size(dt_);
+            size_t rdtsz = ucc_dt_size(a_.dst.info_v.datatype);
+            size_t maxb  = 0;
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = coll_count_at(a_, a_.src.info_v.counts, r) * dtsz_;
+                dsp_[r] =
+                    coll_disp_at(a_, a_.src.info_v.displacements, r) * dtsz_;
+                rcnt_[r] =
+                    coll_count_at(a_, a_.dst.info_v.counts, r) * rdtsz;
+                rdsp_[r] =
+                    coll_disp_at(a_, a_.dst.info_v.displacements, r) * rdtsz;
+                maxb = cnt_[r] > maxb ? cnt_[r] : maxb;
+                maxb = rcnt_[r] > maxb ? rcnt_[r] : maxb;
+            }
+            sbuf_  = (const uint8_t *)a_.src.info_v.buffer;
+            dbuf_  = (uint8_t *)a_.dst.info_v.buffer;
+            total_ = maxb;
+            gran_  = cell_;
+            ct_    = UCC_COLL_TYPE_ALLTOALL;
+            break;
+        }
+        case UCC_COLL_TYPE_GATHER:
+        case UCC_COLL_TYPE_GATHERV: {
+            const bool is_v = ct_ == UCC_COLL_TYPE_GATHERV;
+            if (me_ == a_.root) {
+                dt_   = is_v ? a_.dst.info_v.datatype : a_.dst.info.datatype;
+                dtsz_ = ucc_dt_size(dt_);
+                size_t maxb = 0;
+                if (is_v) {
+                    for (uint32_t r = 0; r < n_; r++) {
+                        rcnt_[r] =
+                            coll_count_at(a_, a_.dst.info_v.counts, r) *
+                            dtsz_;
+                        rdsp_[r] =
+                            coll_disp_at(a_, a_.dst.info_v.displacements,
+                                         r) *
+                            dtsz_;
+                        maxb = rcnt_[r] > maxb ? rcnt_[r] : maxb;
+                    }
+                    dbuf_ = (uint8_t *)a_.dst.info_v.buffer;
+                } else {
+                    size_t block = a_.dst.info.count * dtsz_ / n_;
+                    for (uint32_t r = 0; r < n_; r++) {
+                        rcnt_[r] = block;
+                        rdsp_[r] = (size_t)r * block;
+                    }
+                    maxb  = block;
+                    dbuf_ = (uint8_t *)a_.dst.info.buffer;
+                }
+                sbuf_ = inplace ? dbuf_ + rdsp_[me_]
+                                : (const uint8_t *)a_.src.info.buffer;
+                total_ = maxb;
+            } else {
+                dt_    = a_.src.info.datatype;
+                dtsz_  = ucc_dt_size(dt_);
+                sbuf_  = (const uint8_t *)a_.src.info.buffer;
+                dbuf_  = nullptr;
+                total_ = a_.src.info.count * dtsz_;
+                cnt_[me_] = total_;
+            }
+            ct_ = UCC_COLL_TYPE_GATHER;
+            break;
+        }
+        case UCC_COLL_TYPE_SCATTER:
+        case UCC_COLL_TYPE_SCATTERV: {
+            const bool is_v = ct_ == UCC_COLL_TYPE_SCATTERV;
+            if (cell_ == 0) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            if (me_ == a_.root) {
+                dt_   = is_v ? a_.src.info_v.datatype : a_.src.info.datatype;
+                dtsz_ = ucc_dt_size(dt_);
+                size_t maxb = 0;
+                if (is_v) {
+                    for (uint32_t r = 0; r < n_; r++) {
+                        cnt_[r] =
+                            coll_count_at(a_, a_.src.info_v.counts, r) *
+                            dtsz_;
+                        dsp_[r] =
+                            coll_disp_at(a_, a_.src.info_v.displacements,
+                                         r) *
+                            dtsz_;
+                        maxb = cnt_[r] > maxb ? cnt_[r] : maxb;
+                    }
+                    sbuf_ = (const uint8_t *)a_.src.info_v.buffer;
+                } else {
+                    size_t block = a_.src.info.count * dtsz_ / n_;
+                    for (uint32_t r = 0; r < n_; r++) {
+                        cnt_[r] = block;
+                        dsp_[r] = (size_t)r * block;
+                    }
+                    maxb  = block;
+                    sbuf_ = (const uint8_t *)a_.src.info.buffer;
+                }
+                dbuf_ = inplace ? nullptr : (uint8_t *)a_.dst.info.buffer;
+                rcnt_[me_] = inplace ? 0 : cnt_[me_];
+                rdsp_[me_] = 0;
+                total_     = maxb;
+            } else {
+                dt_    = a_.dst.info.datatype;
+                dtsz_  = ucc_dt_size(dt_);
+                dbuf_  = (uint8_t *)a_.dst.info.buffer;
+                sbuf_  = nullptr;
+                total_ = a_.dst.info.count * dtsz_;
+                rcnt_[a_.root] = total_;
+            }
+            gran_ = cell_;
+            ct_   = UCC_COLL_TYPE_SCATTER;
             break;
         }
         default: return UCC_ERR_NOT_SUPPORTED;
         }
+        nfrags_ = (total_ + gran_ - 1) / gran_;
         if (nfrags_ == 0) {
             nfrags_ = 1;
         }
@@ -499,16 +689,15 @@ class StagedTask final : public Cdna4Task {
 
     ucc_status_t progress() override
     {
-        const size_t chunk = tt_->cfg_.chunk;
         while (frag_ < nfrags_) {
             const size_t   f   = frag_;
             const uint32_t p   = (uint32_t)(f & 1);
             const uint64_t kA  = steps_ * f + 1;
             const uint64_t kB  = steps_ * f + 2;
             const uint64_t kC  = steps_ * f + 3;
-            const size_t   off = f * chunk;
+            const size_t   off = f * gran_;
             const size_t   len =
-                total_ - off < chunk ? total_ - off : chunk;
+                total_ - off < gran_ ? total_ - off : gran_;
             switch (phase_) {
             case 0: /* entry: first copy */
                 if (!all_ge(0)) {
@@ -598,27 +787,78 @@ class StagedTask final : public Cdna4Task {
     void issue_copy(size_t f)
     {
         const uint32_t p   = (uint32_t)(f & 1);
-        const size_t   off = f * tt_->cfg_.chunk;
+        const size_t   off = f * gran_;
         size_t         len = 0;
         const uint8_t *src = nullptr;
-        if (ct_ == UCC_COLL_TYPE_ALLGATHER) {
+        copy_pending_ = false;
+        switch (ct_) {
+        case UCC_COLL_TYPE_ALLTOALL: {
+            /* per-dest cells: in[r*cell_] <- sbuf[dsp_[r]+off ..] */
+            ec_hip::GatherArgs ga{};
+            ga.dst_base = tt_->area(me_, slot_, p, 0);
+            int k       = 0;
+            for (uint32_t r = 0; r < n_; r++) {
+                if (off >= cnt_[r]) {
+                    continue;
+                }
+                size_t l = cnt_[r] - off < cell_ ? cnt_[r] - off : cell_;
+                ga.srcs[k] = sbuf_ + dsp_[r] + off;
+                ga.offs[k] = (uint64_t)r * cell_;
+                ga.lens[k] = l;
+                k++;
+            }
+            if (k > 0) {
+                ga.n = k;
+                ec_hip::gather_copy(ga, copy_s());
+                copy_pending_ = true;
+            }
+            hipEventRecord(ev(0 + (int)p), copy_s());
+            return;
+        }
+        case UCC_COLL_TYPE_SCATTER: {
+            if (me_ == a_.root) {
+                ec_hip::GatherArgs ga{};
+                ga.dst_base = tt_->area(me_, slot_, p, 0);
+                int k       = 0;
+                for (uint32_t r = 0; r < n_; r++) {
+                    if (r == me_ || off >= cnt_[r]) {
+                        continue;
+                    }
+                    size_t l = cnt_[r] - off < cell_ ? cnt_[r] - off
+                                                     : cell_;
+                    ga.srcs[k] = sbuf_ + dsp_[r] + off;
+                    ga.offs[k] = (uint64_t)r * cell_;
+                    ga.lens[k] = l;
+                    k++;
+                }
+                if (k > 0) {
+                    ga.n = k;
+                    ec_hip::gather_copy(ga, copy_s());
+                    copy_pending_ = true;
+                }
+            }
+            hipEventRecord(ev(0 + (int)p), copy_s());
+            return;
+        }
+        case UCC_COLL_TYPE_ALLGATHER:
+        case UCC_COLL_TYPE_GATHER:
             if (off < cnt_[me_]) {
-                len = cnt_[me_] - off < tt_->cfg_.chunk ? cnt_[me_] - off
-                                                        : tt_->cfg_.chunk;
+                len = cnt_[me_] - off < gran_ ? cnt_[me_] - off : gran_;
                 src = sbuf_ + off;
             }
-        } else if (ct_ == UCC_COLL_TYPE_BCAST) {
+            break;
+        case UCC_COLL_TYPE_BCAST:
             if (me_ == a_.root && off < total_) {
-                len = total_ - off < tt_->cfg_.chunk ? total_ - off
-                                                     : tt_->cfg_.chunk;
+                len = total_ - off < gran_ ? total_ - off : gran_;
                 src = sbuf_ + off;
             }
-        } else {
+            break;
+        default:
             if (off < total_) {
-                len = total_ - off < tt_->cfg_.chunk ? total_ - off
-                                                     : tt_->cfg_.chunk;
+                len = total_ - off < gran_ ? total_ - off : gran_;
                 src = sbuf_ + off;
             }
+            break;
         }
         copy_pending_ = len > 0;
         if (len > 0) {
@@ -733,6 +973,91 @@ class StagedTask final : public Cdna4Task {
             ec_hip::gather_copy(ga, comp());
             return true;
         }
+        case UCC_COLL_TYPE_ALLTOALL: {
+            /* read my cell from every peer's staged fragment */
+            ec_hip::GatherArgs ga{};
+            ga.dst_base = dbuf_;
+            int k       = 0;
+            for (uint32_t r = 0; r < n_; r++) {
+                if (off >= rcnt_[r]) {
+                    continue;
+                }
+                size_t l = rcnt_[r] - off < cell_ ? rcnt_[r] - off : cell_;
+                ga.srcs[k] = tt_->area(r, slot_, p, 0) + me_ * cell_;
+                ga.offs[k] = rdsp_[r] + off;
+                ga.lens[k] = l;
+                k++;
+            }
+            if (k == 0) {
+                return false;
+            }
+            ga.n = k;
+            ec_hip::gather_copy(ga, comp());
+            return true;
+        }
+        case UCC_COLL_TYPE_GATHER: {
+            if (me_ != a_.root) {
+                return false;
+            }
+            ec_hip::GatherArgs ga{};
+            ga.dst_base = dbuf_;
+            int k       = 0;
+            for (uint32_t r = 0; r < n_; r++) {
+                if (off >= rcnt_[r]) {
+                    continue;
+                }
+                size_t l = rcnt_[r] - off < gran_ ? rcnt_[r] - off : gran_;
+                if (r == me_) {
+                    if (a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE) {
+                        continue; /* already in place */
+                    }
+                    ga.srcs[k] = sbuf_ + off;
+                } else {
+                    ga.srcs[k] = tt_->area(r, slot_, p, 0);
+                }
+                ga.offs[k] = rdsp_[r] + off;
+                ga.lens[k] = l;
+                k++;
+            }
+            if (k == 0) {
+                return false;
+            }
+            ga.n = k;
+            ec_hip::gather_copy(ga, comp());
+            return true;
+        }
+        case UCC_COLL_TYPE_SCATTER: {
+            if (me_ == a_.root) {
+                /* my own block: direct local copy (skipped if inplace) */
+                if (!dbuf_ || off >= cnt_[me_]) {
+                    return false;
+                }
+                size_t l = cnt_[me_] - off < cell_ ? cnt_[me_] - off
+                                                   : cell_;
+                ec_hip::GatherArgs ga{};
+                ga.dst_base = dbuf_;
+                ga.srcs[0]  = sbuf_ + dsp_[me_] + off;
+                ga.offs[0]  = off;
+                ga.lens[0]  = l;
+                ga.n        = 1;
+                ec_hip::gather_copy(ga, comp());
+                return true;
+            }
+            if (off >= rcnt_[a_.root]) {
+                return false;
+            }
+            size_t l = rcnt_[a_.root] - off < cell_ ? rcnt_[a_.root] - off
+                                                    : cell_;
+            ec_hip::GatherArgs ga{};
+            ga.dst_base = dbuf_;
+            ga.srcs[0] =
+                tt_->area((uint32_t)a_.root, slot_, p, 0) + me_ * cell_;
+            ga.offs[0] = off;
+            ga.lens[0] = l;
+            ga.n       = 1;
+            ec_hip::gather_copy(ga, comp());
+            return true;
+        }
         default: return false;
         }
     }
@@ -770,7 +1095,8 @@ class StagedTask final : public Cdna4Task {
     ucc_reduction_op_t op_ = UCC_OP_SUM;
     float           alpha_ = 1.0f;
     size_t          total_ = 0, nfrags_ = 0;
-    std::vector<size_t> cnt_, dsp_;
+    size_t          gran_ = 0, cell_ = 0;
+    std::vector<size_t> cnt_, dsp_, rcnt_, rdsp_;
     int             steps_ = 3;
     bool            copy_pending_ = false, b_launched_ = false;
 };
@@ -830,8 +1156,15 @@ void Cdna4TlTeam::get_scores(Team *team, ScoreMap &map)
             /* reductions need a supported dtype x op on device */
             if (args.coll_type == UCC_COLL_TYPE_ALLREDUCE ||
                 args.coll_type == UCC_COLL_TYPE_REDUCE ||
-                args.coll_type == UCC_COLL_TYPE_REDUCE_SCATTER) {
-                ucc_datatype_t dt = args.dst.info.datatype;
+                args.coll_type == UCC_COLL_TYPE_REDUCE_SCATTER ||
+                args.coll_type == UCC_COLL_TYPE_REDUCE_SCATTERV) {
+                ucc_datatype_t dt =
+                    args.coll_type == UCC_COLL_TYPE_REDUCE_SCATTERV
+                        ? args.dst.info_v.datatype
+                        : args.dst.info.datatype;
+                if (args.coll_type == UCC_COLL_TYPE_REDUCE) {
+                    dt = args.src.info.datatype;
+                }
                 if (!ec_hip::op_supported(dt, args.op)) {
                     return UCC_ERR_NOT_SUPPORTED;
                 }
@@ -864,10 +1197,19 @@ void Cdna4TlTeam::get_scores(Team *team, ScoreMap &map)
     add(UCC_COLL_TYPE_ALLREDUCE, 0, cfg_.fused_max, 100, "fused", true);
     add(UCC_COLL_TYPE_ALLREDUCE, 0, SIZE_MAX, 80, "staged_linear", false);
     add(UCC_COLL_TYPE_ALLGATHER, 0, SIZE_MAX, 80, "staged_linear", false);
+    add(UCC_COLL_TYPE_ALLGATHERV, 0, SIZE_MAX, 80, "staged_linear", false);
     add(UCC_COLL_TYPE_REDUCE_SCATTER, 0, SIZE_MAX, 80, "staged_linear",
+        false);
+    add(UCC_COLL_TYPE_REDUCE_SCATTERV, 0, SIZE_MAX, 80, "staged_linear",
         false);
     add(UCC_COLL_TYPE_BCAST, 0, SIZE_MAX, 80, "staged_linear", false);
     add(UCC_COLL_TYPE_REDUCE, 0, SIZE_MAX, 80, "staged_linear", false);
+    add(UCC_COLL_TYPE_ALLTOALL, 0, SIZE_MAX, 80, "staged_linear", false);
+    add(UCC_COLL_TYPE_ALLTOALLV, 0, SIZE_MAX, 80, "staged_linear", false);
+    add(UCC_COLL_TYPE_GATHER, 0, SIZE_MAX, 80, "staged_linear", false);
+    add(UCC_COLL_TYPE_GATHERV, 0, SIZE_MAX, 80, "staged_linear", false);
+    add(UCC_COLL_TYPE_SCATTER, 0, SIZE_MAX, 80, "staged_linear", false);
+    add(UCC_COLL_TYPE_SCATTERV, 0, SIZE_MAX, 80, "staged_linear", false);
 }
 
 } // namespace

@@ -184,3 +184,116 @@ def test_persistent_allreduce_device(job):
         for d in dsts:
             torch.testing.assert_close(d.cpu(), expected, rtol=1e-5,
                                        atol=1e-4)
+
+
+@pytest.mark.parametrize("per", [512, 300_000])
+def test_alltoall_device(job, per):
+    torch.manual_seed(15)
+    n = job.n
+    srcs = [torch.randn(per * n, device="cuda") for _ in range(n)]
+    dsts = [torch.zeros(per * n, device="cuda") for _ in range(n)]
+    _run_device(job, "alltoall", [
+        dict(src=srcs[r].data_ptr(), dst=dsts[r].data_ptr(), count=per * n,
+             dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA)
+        for r in range(n)
+    ])
+    for r in range(n):
+        expected = torch.cat(
+            [srcs[s].cpu()[r * per:(r + 1) * per] for s in range(n)])
+        torch.testing.assert_close(dsts[r].cpu(), expected)
+
+
+def test_alltoallv_device_skewed(job):
+    """MoE expert-parallel pattern: skewed per-peer splits (config #4)."""
+    torch.manual_seed(16)
+    n = job.n
+    # rank r sends (r+1)*(d+1)*37 elements to rank d
+    scnt = [[(r + 1) * (d + 1) * 37 for d in range(n)] for r in range(n)]
+    rcnt = [[scnt[s][r] for s in range(n)] for r in range(n)]
+    sdsp = [np.concatenate([[0], np.cumsum(c)[:-1]]).tolist() for c in scnt]
+    rdsp = [np.concatenate([[0], np.cumsum(c)[:-1]]).tolist() for c in rcnt]
+    srcs = [torch.randn(sum(scnt[r]), device="cuda") for r in range(n)]
+    dsts = [torch.zeros(sum(rcnt[r]), device="cuda") for r in range(n)]
+    _run_device(job, "alltoallv", [
+        dict(src=srcs[r].data_ptr(), dst=dsts[r].data_ptr(), count=0,
+             dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA,
+             src_counts=scnt[r], src_displs=sdsp[r],
+             dst_counts=rcnt[r], dst_displs=rdsp[r])
+        for r in range(n)
+    ])
+    for r in range(n):
+        for s in range(n):
+            got = dsts[r].cpu()[rdsp[r][s]:rdsp[r][s] + rcnt[r][s]]
+            exp = srcs[s].cpu()[sdsp[s][r]:sdsp[s][r] + scnt[s][r]]
+            torch.testing.assert_close(got, exp)
+
+
+def test_allgatherv_device(job):
+    torch.manual_seed(17)
+    n = job.n
+    cnts = [(r + 1) * 1001 for r in range(n)]
+    dsps = np.concatenate([[0], np.cumsum(cnts)[:-1]]).tolist()
+    total = sum(cnts)
+    srcs = [torch.randn(cnts[r], device="cuda") for r in range(n)]
+    dsts = [torch.zeros(total, device="cuda") for _ in range(n)]
+    expected = torch.cat([s.cpu() for s in srcs])
+    _run_device(job, "allgatherv", [
+        dict(src=srcs[r].data_ptr(), dst=dsts[r].data_ptr(), count=cnts[r],
+             dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA,
+             dst_counts=cnts, dst_displs=dsps)
+        for r in range(n)
+    ])
+    for d in dsts:
+        torch.testing.assert_close(d.cpu(), expected)
+
+
+def test_reduce_scatterv_device(job):
+    torch.manual_seed(18)
+    n = job.n
+    cnts = [(r + 1) * 777 for r in range(n)]
+    total = sum(cnts)
+    dsps = np.concatenate([[0], np.cumsum(cnts)[:-1]]).tolist()
+    srcs = [torch.randn(total, device="cuda") for _ in range(n)]
+    dsts = [torch.zeros(cnts[r], device="cuda") for r in range(n)]
+    expected = sum(s.cpu() for s in srcs)
+    _run_device(job, "reduce_scatterv", [
+        dict(src=srcs[r].data_ptr(), dst=dsts[r].data_ptr(), count=0,
+             dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA,
+             dst_counts=cnts)
+        for r in range(n)
+    ])
+    for r in range(n):
+        torch.testing.assert_close(
+            dsts[r].cpu(), expected[dsps[r]:dsps[r] + cnts[r]],
+            rtol=1e-5, atol=1e-5)
+
+
+def test_gather_scatter_device(job):
+    torch.manual_seed(19)
+    n = job.n
+    per = 123_000
+    root = 0
+    srcs = [torch.randn(per, device="cuda") for _ in range(n)]
+    gdst = [torch.zeros(per * n, device="cuda") if r == root
+            else torch.zeros(1, device="cuda") for r in range(n)]
+    _run_device(job, "gather", [
+        dict(src=srcs[r].data_ptr(), dst=gdst[r].data_ptr(),
+             count=per * n if r == root else per,
+             dt=dtypes.FLOAT32, root=root, mem_type=dtypes.MEM_CUDA)
+        for r in range(n)
+    ])
+    expected = torch.cat([s.cpu() for s in srcs])
+    torch.testing.assert_close(gdst[root].cpu(), expected)
+
+    sdst = [torch.zeros(per, device="cuda") for _ in range(n)]
+    big = torch.randn(per * n, device="cuda")
+    _run_device(job, "scatter", [
+        dict(src=big.data_ptr() if r == root else 0,
+             dst=sdst[r].data_ptr(),
+             count=per * n if r == root else per,
+             dt=dtypes.FLOAT32, root=root, mem_type=dtypes.MEM_CUDA)
+        for r in range(n)
+    ])
+    for r in range(n):
+        torch.testing.assert_close(sdst[r].cpu(),
+                                   big.cpu()[r * per:(r + 1) * per])
