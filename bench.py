@@ -64,7 +64,7 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
     local = int(os.environ.get("LOCAL_RANK", "0"))
-    torch.cuda.set_device(local)
+    torch.cuda.set_device(local % torch.cuda.device_count())
 
     dist = None
     group = None

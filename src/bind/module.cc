@@ -372,6 +372,26 @@ PYBIND11_MODULE(_core, m)
         py::arg("dst_counts") = std::vector<uint64_t>(),
         py::arg("dst_displs") = std::vector<uint64_t>());
 
+    /* ------------------------------------------------- EE / triggered */
+    m.def("ee_create", [](std::shared_ptr<CoreTeam> t, uintptr_t stream) {
+        ucc_ee_params_t p{};
+        p.ee_type         = UCC_EE_ROCM_STREAM;
+        p.ee_context      = (void *)stream;
+        p.ee_context_size = sizeof(void *);
+        ucc_ee_h ee;
+        check(ucc_ee_create(t->team, &p, &ee), "ucc_ee_create");
+        return (uintptr_t)ee;
+    });
+    m.def("ee_destroy",
+          [](uintptr_t ee) { ucc_ee_destroy((ucc_ee_h)ee); });
+    m.def("triggered_post", [](uintptr_t ee, std::shared_ptr<CoreReq> r) {
+        ucc_ev_t ev{};
+        ev.ev_type = UCC_EVENT_COMPUTE_COMPLETE;
+        ev.req     = r->req;
+        check(ucc_collective_triggered_post((ucc_ee_h)ee, &ev),
+              "triggered_post");
+    });
+
     py::class_<CoreReq, std::shared_ptr<CoreReq>>(m, "Request")
         .def("post",
              [](CoreReq &r) { check(ucc_collective_post(r.req), "post"); })

@@ -568,38 +568,83 @@ ucc_status_t ucc_collective_finalize(ucc_coll_req_h request)
     return UCC_OK;
 }
 
-/* ------------------------------------------------------------ EE (stubs) */
+/* ------------------------------------------------------------------- EE */
+/* Execution-engine API (reference core/ucc_ee.c + triggered-post path in
+ * core/ucc_coll.c:423-659, re-derived): an EE wraps a user HIP stream;
+ * ucc_collective_triggered_post launches the collective's device work
+ * stream-ordered on it (tasks that support it: cdna4 fused allreduce,
+ * which is also hipGraph-capture legal). */
 ucc_status_t ucc_ee_create(ucc_team_h team, const ucc_ee_params_t *params,
                            ucc_ee_h *ee)
 {
-    (void)team; (void)params; (void)ee;
-    return UCC_ERR_NOT_IMPLEMENTED;
+    if (!team || !params || !ee) {
+        return UCC_ERR_INVALID_PARAM;
+    }
+    if (params->ee_type != UCC_EE_CUDA_STREAM &&
+        params->ee_type != UCC_EE_ROCM_STREAM) {
+        return UCC_ERR_NOT_SUPPORTED;
+    }
+    auto *e   = new Ee();
+    e->team   = reinterpret_cast<Team *>(team);
+    e->params = *params;
+    e->stream = params->ee_context;
+    *ee       = reinterpret_cast<ucc_ee_h>(e);
+    return UCC_OK;
 }
-ucc_status_t ucc_ee_destroy(ucc_ee_h ee) { (void)ee; return UCC_ERR_NOT_IMPLEMENTED; }
+ucc_status_t ucc_ee_destroy(ucc_ee_h ee)
+{
+    delete reinterpret_cast<Ee *>(ee);
+    return UCC_OK;
+}
 ucc_status_t ucc_ee_get_event(ucc_ee_h ee, ucc_ev_t **ev)
 {
-    (void)ee; (void)ev;
-    return UCC_ERR_NOT_IMPLEMENTED;
+    auto *e = reinterpret_cast<Ee *>(ee);
+    if (e->events.empty()) {
+        return UCC_ERR_NOT_FOUND;
+    }
+    *ev = &e->events.front();
+    return UCC_OK;
 }
 ucc_status_t ucc_ee_ack_event(ucc_ee_h ee, ucc_ev_t *ev)
 {
-    (void)ee; (void)ev;
-    return UCC_ERR_NOT_IMPLEMENTED;
+    auto *e = reinterpret_cast<Ee *>(ee);
+    if (!e->events.empty() && ev == &e->events.front()) {
+        e->events.pop_front();
+        return UCC_OK;
+    }
+    return UCC_ERR_INVALID_PARAM;
 }
 ucc_status_t ucc_ee_set_event(ucc_ee_h ee, ucc_ev_t *ev)
 {
-    (void)ee; (void)ev;
-    return UCC_ERR_NOT_IMPLEMENTED;
+    auto *e = reinterpret_cast<Ee *>(ee);
+    e->events.push_back(*ev);
+    return UCC_OK;
 }
 ucc_status_t ucc_ee_wait(ucc_ee_h ee, ucc_ev_t *ev)
 {
     (void)ee; (void)ev;
-    return UCC_ERR_NOT_IMPLEMENTED;
+    return UCC_ERR_NOT_SUPPORTED; /* host-side wait: use stream sync */
 }
 ucc_status_t ucc_collective_triggered_post(ucc_ee_h ee, ucc_ev_t *ev)
 {
-    (void)ee; (void)ev;
-    return UCC_ERR_NOT_IMPLEMENTED;
+    auto *e = reinterpret_cast<Ee *>(ee);
+    if (!e || !ev || !ev->req) {
+        return UCC_ERR_INVALID_PARAM;
+    }
+    auto *req = reinterpret_cast<CollRequest *>(ev->req);
+    if (req->posted && !req->persistent) {
+        return UCC_ERR_INVALID_PARAM;
+    }
+    ucc_status_t st = req->task->triggered_post(e->stream);
+    if (st != UCC_OK && st != UCC_INPROGRESS) {
+        return st;
+    }
+    req->posted        = true;
+    req->super.status  = req->task->status;
+    ucc_ev_t done      = *ev;
+    done.ev_type       = UCC_EVENT_COLLECTIVE_POST;
+    e->events.push_back(done);
+    return UCC_OK;
 }
 
 ucc_status_t ucc_mem_map(ucc_context_h context, ucc_mem_map_flags_t flags,

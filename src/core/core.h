@@ -53,6 +53,14 @@ class Task {
 
     virtual ucc_status_t post()     = 0;
     virtual ucc_status_t progress() { return status; }
+    /* Triggered (stream) post: launch the collective's device work onto
+     * the execution-engine stream; completion is stream-ordered. Tasks
+     * that cannot run stream-ordered return NOT_SUPPORTED. */
+    virtual ucc_status_t triggered_post(void *ee_stream)
+    {
+        (void)ee_stream;
+        return UCC_ERR_NOT_SUPPORTED;
+    }
     /* Called exactly once when the task reaches a terminal status. */
     virtual void on_complete() {}
 
@@ -267,6 +275,16 @@ struct Lib {
     ucc_lib_params_t  params{};
     ucc_thread_mode_t thread_mode = UCC_THREAD_SINGLE;
     uint64_t          next_ctx_seq = 1;
+};
+
+/* -------------------------------------------------------------------- EE */
+/* Execution engine: a user HIP stream colls can be triggered onto
+ * (reference ucc_ee_create ucc.h events section; core/ucc_ee.c). */
+struct Ee {
+    Team            *team = nullptr;
+    ucc_ee_params_t  params{};
+    void            *stream = nullptr; /* hipStream_t */
+    std::deque<ucc_ev_t> events;
 };
 
 /* ----------------------------------------------------------- CollRequest */

@@ -68,6 +68,40 @@ struct FusedArgs {
 };
 ucc_status_t fused_allreduce(const FusedArgs &a, hipStream_t stream);
 
+/* Graph-capturable fused allreduce: identical to fused_allreduce but the
+ * per-iteration sequence number is derived ON DEVICE from per-block launch
+ * counters (flags u64[kGraphCntBase + slot*kMaxGraphBlocks + block]), so
+ * the SAME launch can be captured once into a hipGraph and replayed: every
+ * replay is a new collective iteration. Requires a slot dedicated to one
+ * persistent request (all ranks replay in lockstep). */
+constexpr int kMaxGraphBlocks = 64;
+constexpr int kStageCntBase   = 8 * kMaxRanks;      /* u64 idx of stage cnts */
+constexpr int kGraphCntBase   = 128;                /* u64 idx of blk cnts   */
+constexpr int kFlagsBytes     = 8192;               /* flags alloc size      */
+
+struct GraphFusedArgs {
+    const void *src;
+    void       *dst;
+    uint64_t    count;
+    void       *my_scratch;
+    const void *peer_scratch[kMaxRanks];
+    uint64_t   *local_flags;
+    uint64_t   *peer_flags[kMaxRanks];
+    int         rank, nranks, slot;
+    ucc_datatype_t     dt;
+    ucc_reduction_op_t op;
+    float       alpha;
+    uint64_t   *error_word;
+    int         nblocks;       /* <= kMaxGraphBlocks */
+    uint64_t    parity_stride; /* bytes between parity-0 and parity-1
+                                  staging areas: iteration i stages into
+                                  base + (i&1)*parity_stride so a peer's
+                                  replay i+1 never overwrites data still
+                                  being read for replay i */
+};
+ucc_status_t fused_allreduce_graph(const GraphFusedArgs &a,
+                                   hipStream_t stream);
+
 bool dt_supported(ucc_datatype_t dt);
 bool op_supported(ucc_datatype_t dt, ucc_reduction_op_t op);
 

@@ -464,6 +464,132 @@ __global__ void k_fused_allreduce(FusedArgs a)
     }
 }
 
+/* ------------------------------------- graph-capturable fused allreduce */
+template <typename T, int OP, int VEC>
+__global__ void k_fused_allreduce_graph(GraphFusedArgs a)
+{
+    using A = typename Cvt<T>::A;
+    /* 0. device-side iteration number: this block's launch count. Every
+     * block of every launch/replay increments its own counter exactly
+     * once, so all blocks of one launch observe the same value. */
+    uint64_t *my_cnt = a.local_flags + kGraphCntBase +
+                       (uint64_t)a.slot * kMaxGraphBlocks + blockIdx.x;
+    __shared__ uint64_t s_seq;
+    if (threadIdx.x == 0) {
+        s_seq = __hip_atomic_fetch_add(my_cnt, 1, __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT) +
+                1;
+    }
+    __syncthreads();
+    const uint64_t seq = s_seq;
+    const uint64_t par = (seq & 1) * a.parity_stride;
+    /* 1. stage src -> my scratch (parity area for this iteration) */
+    {
+        const uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+        const uint64_t str = (uint64_t)gridDim.x * blockDim.x;
+        const uint64_t nv  = a.count / VEC;
+        using P            = Pack<T, OP, VEC>;
+        P *mysc = (P *)((uint8_t *)a.my_scratch + par);
+        for (uint64_t i = tid; i < nv; i += str) {
+            mysc[i] = ((const P *)a.src)[i];
+        }
+        for (uint64_t i = nv * VEC + tid; i < a.count; i += str) {
+            ((T *)mysc)[i] = ((const T *)a.src)[i];
+        }
+    }
+    __threadfence_system();
+    __syncthreads();
+    /* 2. grid arrival; block 0 signals all peers with seq */
+    uint64_t *stage_cnt = a.local_flags + kStageCntBase + a.slot;
+    if (threadIdx.x == 0) {
+        __hip_atomic_fetch_add(stage_cnt, 1, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+    }
+    if (blockIdx.x == 0) {
+        if (threadIdx.x == 0) {
+            uint64_t spins  = 0;
+            uint64_t target = seq * (uint64_t)a.nblocks;
+            while (sys_load(stage_cnt) < target) {
+                if (++spins > kSpinLimit) {
+                    sys_store(a.error_word, 1);
+                    return;
+                }
+                __builtin_amdgcn_s_sleep(2);
+            }
+        }
+        __syncthreads();
+        if ((int)threadIdx.x < a.nranks) {
+            sys_store(a.peer_flags[threadIdx.x] +
+                          (uint64_t)a.slot * kMaxRanks + a.rank,
+                      seq);
+        }
+    }
+    /* 3. wait all ranks arrived for this iteration */
+    if (threadIdx.x < 64) {
+        int      j     = (int)threadIdx.x;
+        uint64_t spins = 0;
+        if (j < a.nranks) {
+            const uint64_t *f =
+                a.local_flags + (uint64_t)a.slot * kMaxRanks + j;
+            while (sys_load(f) < seq) {
+                if (++spins > kSpinLimit) {
+                    sys_store(a.error_word, 1);
+                    return;
+                }
+                __builtin_amdgcn_s_sleep(2);
+            }
+        }
+    }
+    __syncthreads();
+    __threadfence_system();
+    /* 4. reduce all peers' scratch (this iteration's parity) into dst.
+     * Why parity is sufficient: a peer can start staging replay i+1 while
+     * I still read its replay-i data (staging precedes the handshake),
+     * but it writes the other parity area. It cannot reach replay i+2
+     * (same parity as i) until the i+1 handshake completed, which needs
+     * MY i+1 launch — stream-ordered after my reduce of i finished. */
+    {
+        const uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+        const uint64_t str = (uint64_t)gridDim.x * blockDim.x;
+        const uint64_t nv  = a.count / VEC;
+        using P            = Pack<T, OP, VEC>;
+        const int n        = a.nranks;
+        for (uint64_t i = tid; i < nv; i += str) {
+            P acc = ((const P *)((const uint8_t *)a.peer_scratch[0] + par))[i];
+            A r[VEC];
+#pragma unroll
+            for (int k = 0; k < VEC; k++) {
+                r[k] = Cvt<T>::load(acc.v[k]);
+            }
+            for (int s = 1; s < n; s++) {
+                P x = ((const P *)((const uint8_t *)a.peer_scratch[s] +
+                                   par))[i];
+#pragma unroll
+                for (int k = 0; k < VEC; k++) {
+                    r[k] = red<A, OP>(r[k], Cvt<T>::load(x.v[k]));
+                }
+            }
+            P out;
+#pragma unroll
+            for (int k = 0; k < VEC; k++) {
+                out.v[k] = Cvt<T>::store(apply_alpha<A>(r[k], a.alpha));
+            }
+            ((P *)a.dst)[i] = out;
+        }
+        for (uint64_t t = nv * VEC + tid; t < a.count; t += str) {
+            A r = Cvt<T>::load(
+                ((const T *)((const uint8_t *)a.peer_scratch[0] + par))[t]);
+            for (int s = 1; s < n; s++) {
+                r = red<A, OP>(
+                    r, Cvt<T>::load(((const T *)((const uint8_t *)
+                                                     a.peer_scratch[s] +
+                                                 par))[t]));
+            }
+            ((T *)a.dst)[t] = Cvt<T>::store(apply_alpha<A>(r, a.alpha));
+        }
+    }
+}
+
 /* ----------------------------------------------------------- launchers */
 static bool aligned16(const void *p) { return (((uintptr_t)p) & 15) == 0; }
 
@@ -497,6 +623,16 @@ static ucc_status_t launch_fused(const FusedArgs &a, hipStream_t s)
     constexpr int VEC = VecOf<T>::value;
     hipLaunchKernelGGL((k_fused_allreduce<T, OP, VEC>), dim3(a.nblocks),
                        dim3(256), 0, s, a);
+    return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
+}
+
+template <typename T, int OP>
+static ucc_status_t launch_fused_graph(const GraphFusedArgs &a,
+                                       hipStream_t s)
+{
+    constexpr int VEC = VecOf<T>::value;
+    hipLaunchKernelGGL((k_fused_allreduce_graph<T, OP, VEC>),
+                       dim3(a.nblocks), dim3(256), 0, s, a);
     return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
 }
 
@@ -567,6 +703,28 @@ ucc_status_t fused_allreduce(const FusedArgs &a, hipStream_t s)
         UCC_DT_CASE_INT(UCC_DT_UINT32, uint32_t, launch_fused)
         UCC_DT_CASE_INT(UCC_DT_INT64, int64_t, launch_fused)
         UCC_DT_CASE_INT(UCC_DT_UINT64, uint64_t, launch_fused)
+    default: return UCC_ERR_NOT_SUPPORTED;
+    }
+}
+
+ucc_status_t fused_allreduce_graph(const GraphFusedArgs &a, hipStream_t s)
+{
+    ucc_reduction_op_t op = a.op;
+    switch (a.dt) {
+        UCC_DT_CASE_FLOAT(UCC_DT_BFLOAT16, bf16_t, launch_fused_graph)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT16, fp16_t, launch_fused_graph)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT32, float, launch_fused_graph)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT64, double, launch_fused_graph)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT8_E4M3, e4m3_t, launch_fused_graph)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT8_E5M2, e5m2_t, launch_fused_graph)
+        UCC_DT_CASE_INT(UCC_DT_INT8, int8_t, launch_fused_graph)
+        UCC_DT_CASE_INT(UCC_DT_UINT8, uint8_t, launch_fused_graph)
+        UCC_DT_CASE_INT(UCC_DT_INT16, int16_t, launch_fused_graph)
+        UCC_DT_CASE_INT(UCC_DT_UINT16, uint16_t, launch_fused_graph)
+        UCC_DT_CASE_INT(UCC_DT_INT32, int32_t, launch_fused_graph)
+        UCC_DT_CASE_INT(UCC_DT_UINT32, uint32_t, launch_fused_graph)
+        UCC_DT_CASE_INT(UCC_DT_INT64, int64_t, launch_fused_graph)
+        UCC_DT_CASE_INT(UCC_DT_UINT64, uint64_t, launch_fused_graph)
     default: return UCC_ERR_NOT_SUPPORTED;
     }
 }

@@ -63,6 +63,8 @@ static inline size_t coll_disp_at(const ucc_coll_args_t &a, const void *d,
 
 struct Cdna4Cfg {
     uint32_t nslots;
+    uint32_t npers;     /* dedicated slots for persistent triggered
+                           (graph-captured) collectives              */
     size_t   chunk;     /* staging fragment bytes                     */
     size_t   fused_max; /* msg sizes <= this take the fused kernel    */
 };
@@ -146,18 +148,24 @@ class Cdna4TlTeam final : public TlTeam {
 
     ucc_status_t local_init()
     {
-        /* scratch: [slot][parity][in|out] areas of chunk bytes each */
-        scratch_bytes_ = (size_t)cfg_.nslots * 2 * 2 * cfg_.chunk;
+        /* scratch: [slot][parity][in|out] areas of chunk bytes each;
+         * slots [0, nslots) rotate for host-driven colls, slots
+         * [nslots, nslots+npers) are pinned to persistent triggered
+         * (graph-replayable) requests */
+        scratch_bytes_ =
+            (size_t)(cfg_.nslots + cfg_.npers) * 2 * 2 * cfg_.chunk;
         HIPCHK(hipMalloc((void **)&scratch_, scratch_bytes_));
+        pslot_used_.assign(cfg_.npers, false);
         /* flags: fine-grained for cross-GPU system-scope atomics */
-        hipError_t e = hipExtMallocWithFlags((void **)&flags_, 4096,
-                                             hipDeviceMallocFinegrained);
+        hipError_t e =
+            hipExtMallocWithFlags((void **)&flags_, ec_hip::kFlagsBytes,
+                                  hipDeviceMallocFinegrained);
         if (e != hipSuccess) {
             ucc_warn("finegrained alloc failed (%s), using hipMalloc",
                      hipGetErrorString(e));
-            HIPCHK(hipMalloc((void **)&flags_, 4096));
+            HIPCHK(hipMalloc((void **)&flags_, ec_hip::kFlagsBytes));
         }
-        HIPCHK(hipMemset(flags_, 0, 4096));
+        HIPCHK(hipMemset(flags_, 0, ec_hip::kFlagsBytes));
         HIPCHK(hipHostMalloc((void **)&err_host_, 64, hipHostMallocDefault));
         *err_host_ = 0;
         HIPCHK(hipDeviceSynchronize());
@@ -253,9 +261,30 @@ class Cdna4TlTeam final : public TlTeam {
                (((size_t)slot * 2 + parity) * 2 + which) * cfg_.chunk;
     }
 
+    /* Persistent-slot allocator. Deterministic across ranks as long as
+     * persistent triggered requests are created/finalized in the same
+     * order on every rank (collective semantics): lowest free index. */
+    int alloc_pslot()
+    {
+        for (uint32_t i = 0; i < cfg_.npers; i++) {
+            if (!pslot_used_[i]) {
+                pslot_used_[i] = true;
+                return (int)(cfg_.nslots + i);
+            }
+        }
+        return -1;
+    }
+    void free_pslot(int slot)
+    {
+        if (slot >= (int)cfg_.nslots) {
+            pslot_used_[slot - cfg_.nslots] = false;
+        }
+    }
+
     Cdna4Cfg    cfg_;
     ShmSeg      seg_;
     std::string seg_name_;
+    std::vector<bool> pslot_used_;
     uint64_t    seq_ = 0;
     std::vector<uint64_t> stage_cum_; /* per-slot fused block arrivals */
     uint8_t    *scratch_ = nullptr;
@@ -414,8 +443,73 @@ class FusedAllreduceTask final : public Cdna4Task {
         return UCC_INPROGRESS;
     }
 
+    /* Stream-triggered (and hipGraph-capturable) post: launch ONE fused
+     * kernel with a device-derived iteration number onto the EE stream.
+     * Completion is stream-ordered (request status = OK on return); the
+     * same capture replays as a fresh collective iteration each time.
+     * Requires a per-team dedicated slot (all ranks replay in lockstep).
+     * Reference parity: ucc_triggered_post (core/ucc_coll.c:510-659) +
+     * persistent executor-in-stream design, collapsed to one kernel. */
+    ucc_status_t triggered_post(void *ee_stream) override
+    {
+        size_t bytes =
+            a_.dst.info.count * ucc_dt_size(a_.dst.info.datatype);
+        if (bytes > tt_->cfg_.chunk) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        if (pslot_ < 0) {
+            pslot_ = tt_->alloc_pslot();
+            if (pslot_ < 0) {
+                ucc_error("no free persistent slot (TL_CDNA4 "
+                          "PERSISTENT_SLOTS exhausted)");
+                return UCC_ERR_NO_RESOURCE;
+            }
+            me_ = tt_->team_->rank;
+            n_  = tt_->team_->size;
+        }
+        ec_hip::GraphFusedArgs ga{};
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        ga.dst     = a_.dst.info.buffer;
+        ga.src     = inplace ? ga.dst : a_.src.info.buffer;
+        ga.count   = a_.dst.info.count;
+        ga.dt      = a_.dst.info.datatype;
+        ga.op      = a_.op == UCC_OP_AVG ? (ucc_reduction_op_t)12 : a_.op;
+        ga.alpha   = a_.op == UCC_OP_AVG ? 1.0f / (float)n_ : 1.0f;
+        ga.my_scratch = tt_->area(me_, (uint32_t)pslot_, 0, 0);
+        for (uint32_t r = 0; r < n_; r++) {
+            ga.peer_scratch[r] = tt_->area(r, (uint32_t)pslot_, 0, 0);
+            ga.peer_flags[r]   = tt_->peers_[r].flags;
+        }
+        ga.local_flags   = tt_->flags_;
+        ga.rank          = (int)me_;
+        ga.nranks        = (int)n_;
+        ga.slot          = pslot_;
+        ga.error_word    = tt_->err_host_;
+        ga.parity_stride = 2 * tt_->cfg_.chunk;
+        size_t blocks    = (bytes + 128 * 1024 - 1) / (128 * 1024);
+        ga.nblocks       = (int)(blocks < 1 ? 1
+                                 : blocks > (size_t)ec_hip::kMaxGraphBlocks
+                                     ? (size_t)ec_hip::kMaxGraphBlocks
+                                     : blocks);
+        ucc_status_t st =
+            ec_hip::fused_allreduce_graph(ga, (hipStream_t)ee_stream);
+        if (st != UCC_OK) {
+            return st;
+        }
+        status = UCC_OK;
+        return UCC_OK;
+    }
+
+    ~FusedAllreduceTask() override
+    {
+        if (pslot_ >= 0) {
+            tt_->free_pslot(pslot_);
+        }
+    }
+
   private:
     size_t bytes_ = 0;
+    int    pslot_ = -1;
 };
 
 /* Staged linear collectives: allreduce / allgather(v) / reduce_scatter(v) /
@@ -1129,6 +1223,10 @@ class Cdna4Tl final : public Tl {
         auto    &cfg = Config::instance();
         Cdna4Cfg c;
         c.nslots    = (uint32_t)cfg.get_int("TL_CDNA4", "MAX_CONCURRENT", 4);
+        c.npers  = (uint32_t)cfg.get_int("TL_CDNA4", "PERSISTENT_SLOTS", 2);
+        if (c.nslots + c.npers > 8) { /* flags layout: 8 slots max */
+            c.npers = c.nslots < 8 ? 8 - c.nslots : 0;
+        }
         c.chunk     = cfg.get_size("TL_CDNA4", "CHUNK_SIZE", 8 * 1024 * 1024);
         c.fused_max = cfg.get_size("TL_CDNA4", "FUSED_MAX", 512 * 1024);
         return new Cdna4TlTeam(tlc, team, c);

@@ -1,0 +1,169 @@
+"""Cross-process device-path worker: run under
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node=2 \
+      --master-addr 127.0.0.1 tests/xproc_worker.py
+
+Both ranks share cuda:0 on a 1-GPU box (or map to their own GPU on an
+8-GPU node): this exercises the production path the in-process jig cannot
+— hipIpcOpenMemHandle peer mappings across processes, the cross-process
+fused single-kernel allreduce, stream-triggered post (ucc_ee_create +
+ucc_collective_triggered_post) and hipGraph capture/replay of the fused
+kernel (BASELINE config #5).
+"""
+
+import os
+import sys
+
+import torch
+import torch.distributed as dist
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from ucc_amd import core, dtypes  # noqa: E402
+
+
+def oob(group, world):
+    def allgather(data: bytes):
+        n = len(data)
+        t = torch.frombuffer(bytearray(data), dtype=torch.uint8).clone()
+        outs = [torch.empty(n, dtype=torch.uint8) for _ in range(world)]
+        dist.all_gather(outs, t, group=group)
+        return [o.numpy().tobytes() for o in outs]
+
+    return allgather
+
+
+def wait(req, ctx):
+    req.post()
+    it = 0
+    while req.test() == core().INPROGRESS:
+        ctx.progress()
+        it += 1
+        if it > 200_000_000:
+            raise TimeoutError("collective stuck")
+
+
+def main():
+    rank = int(os.environ["RANK"])
+    world = int(os.environ["WORLD_SIZE"])
+    local = int(os.environ["LOCAL_RANK"])
+    torch.cuda.set_device(local % torch.cuda.device_count())
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    c = core()
+    lib = c.Lib()
+    ctx = c.Context(lib)
+    team = c.team_create_post(ctx, py_allgather=oob(dist.group.WORLD, world),
+                              rank=rank, n_ranks=world)
+    while True:
+        st = c.team_create_test(team)
+        if st == c.OK:
+            break
+        if st < 0:
+            raise RuntimeError(f"team create failed: {st}")
+
+    g0 = torch.Generator(device="cpu").manual_seed(1234)
+    results = []
+
+    # 1. small fused cross-process allreduce (bf16)
+    count = 8192
+    full = torch.randn(world, count, generator=g0)
+    src = full[rank].to(torch.bfloat16).cuda()
+    dst = torch.zeros(count, dtype=torch.bfloat16, device="cuda")
+    expected = sum(full[r].to(torch.bfloat16).float() for r in range(world))
+    r1 = c.coll_init(team, "allreduce", src=src.data_ptr(),
+                     dst=dst.data_ptr(), count=count, dt=dtypes.BFLOAT16,
+                     mem_type=dtypes.MEM_CUDA)
+    wait(r1, ctx)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(dst.cpu().float(), expected, rtol=2e-2,
+                               atol=2e-1)
+    results.append("fused_xproc")
+
+    # 2. large staged cross-process allreduce (fp32)
+    count = 20_000_000
+    full = torch.randn(world, count, generator=g0)
+    src = full[rank].cuda()
+    dst = torch.zeros(count, device="cuda")
+    expected = full.sum(0)
+    r2 = c.coll_init(team, "allreduce", src=src.data_ptr(),
+                     dst=dst.data_ptr(), count=count, dt=dtypes.FLOAT32,
+                     mem_type=dtypes.MEM_CUDA)
+    wait(r2, ctx)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(dst.cpu(), expected, rtol=1e-5, atol=1e-4)
+    results.append("staged_xproc")
+
+    # 3. cross-process alltoallv (skewed, fp16)
+    scnt = [[(r + 1) * (d + 1) * 1024 for d in range(world)]
+            for r in range(world)]
+    rcnt = [[scnt[s][r] for s in range(world)] for r in range(world)]
+    def cumsum0(v):
+        out, t = [], 0
+        for x in v:
+            out.append(t)
+            t += x
+        return out
+    sdsp = [cumsum0(x) for x in scnt]
+    rdsp = [cumsum0(x) for x in rcnt]
+    sfull = [torch.randn(sum(scnt[r]), generator=g0).to(torch.float16)
+             for r in range(world)]
+    src = sfull[rank].cuda()
+    dst = torch.zeros(sum(rcnt[rank]), dtype=torch.float16, device="cuda")
+    r3 = c.coll_init(team, "alltoallv", src=src.data_ptr(),
+                     dst=dst.data_ptr(), count=0, dt=dtypes.FLOAT16,
+                     mem_type=dtypes.MEM_CUDA,
+                     src_counts=scnt[rank], src_displs=sdsp[rank],
+                     dst_counts=rcnt[rank], dst_displs=rdsp[rank])
+    wait(r3, ctx)
+    torch.cuda.synchronize()
+    for s in range(world):
+        got = dst.cpu()[rdsp[rank][s]:rdsp[rank][s] + rcnt[rank][s]]
+        exp = sfull[s][sdsp[s][rank]:sdsp[s][rank] + scnt[s][rank]]
+        torch.testing.assert_close(got, exp)
+    results.append("alltoallv_xproc")
+
+    # 4. triggered post on a user stream (persistent, re-triggered)
+    count = 65536
+    full = torch.randn(world, count, generator=g0)
+    src = full[rank].cuda()
+    dst = torch.zeros(count, device="cuda")
+    expected = full.sum(0)
+    rp = c.coll_init(team, "allreduce", src=src.data_ptr(),
+                     dst=dst.data_ptr(), count=count, dt=dtypes.FLOAT32,
+                     mem_type=dtypes.MEM_CUDA, flags=c.FLAG_PERSISTENT)
+    s = torch.cuda.Stream()
+    ee = c.ee_create(team, s.cuda_stream)
+    for it in range(3):
+        dst.zero_()
+        torch.cuda.synchronize()
+        dist.barrier()
+        c.triggered_post(ee, rp)
+        s.synchronize()
+        torch.testing.assert_close(dst.cpu(), expected, rtol=1e-5,
+                                   atol=1e-4)
+    results.append("triggered")
+
+    # 5. hipGraph capture + replay of the triggered fused allreduce
+    g = torch.cuda.CUDAGraph()
+    dist.barrier()
+    with torch.cuda.graph(g, stream=s):
+        c.triggered_post(ee, rp)
+    dist.barrier()
+    for it in range(5):
+        src.copy_(full[rank] + it)
+        dst.zero_()
+        torch.cuda.synchronize()
+        dist.barrier()
+        g.replay()
+        torch.cuda.synchronize()
+        exp = (full + it).sum(0)
+        torch.testing.assert_close(dst.cpu(), exp, rtol=1e-5, atol=1e-4)
+    results.append("hipgraph_replay")
+
+    c.ee_destroy(ee)
+    dist.barrier()
+    print(f"XPROC_OK rank={rank} {'+'.join(results)}", flush=True)
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
