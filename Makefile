@@ -36,7 +36,20 @@ BIND_OBJ := $(BUILD)/src/bind/module.o
 MODULE := ucc_amd/_core$(PY_EXT)
 SHLIB  := $(BUILD)/libucc_amd.so
 
-all: $(MODULE) $(SHLIB)
+PERFTEST := build/ucc_perftest
+INFO     := build/ucc_info
+
+all: $(MODULE) $(SHLIB) $(PERFTEST) $(INFO)
+
+$(BUILD)/tools/%.o: tools/%.cc tools/shm_oob.h
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+$(PERFTEST): $(BUILD)/tools/perftest.o $(LIB_OBJS)
+	$(HIPCC) $^ -o $@ -lrt
+
+$(INFO): $(BUILD)/tools/info.o $(LIB_OBJS)
+	$(HIPCC) $^ -o $@ -lrt
 
 $(BUILD)/%.o: %.cc
 	@mkdir -p $(dir $@)
