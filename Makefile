@@ -9,8 +9,8 @@ PY_EXT   := $(shell $(PYTHON) -c "import sysconfig;print(sysconfig.get_config_va
 PY_INC   := $(shell $(PYTHON) -m pybind11 --includes)
 
 CXXFLAGS := --offload-arch=$(ARCH) -O3 -std=c++17 -fPIC -Wall -Wextra \
-            -Wno-unused-parameter -DUCC_AMD_HAS_HIP -DUCC_AMD_HAS_TL_CDNA4
-LDFLAGS  := -shared -fPIC
+            -Wno-unused-parameter -DUCC_AMD_HAS_HIP -DUCC_AMD_HAS_TL_CDNA4 -DUCC_AMD_HAS_TL_RCCL
+LDFLAGS  := -shared -fPIC -L/opt/rocm/lib -lrccl -Wl,-rpath,/opt/rocm/lib
 
 BUILD := build
 
@@ -46,10 +46,10 @@ $(BUILD)/tools/%.o: tools/%.cc tools/shm_oob.h
 	$(HIPCC) $(CXXFLAGS) -c $< -o $@
 
 $(PERFTEST): $(BUILD)/tools/perftest.o $(LIB_OBJS)
-	$(HIPCC) $^ -o $@ -lrt
+	$(HIPCC) $^ -o $@ -lrt -L/opt/rocm/lib -lrccl -Wl,-rpath,/opt/rocm/lib
 
 $(INFO): $(BUILD)/tools/info.o $(LIB_OBJS)
-	$(HIPCC) $^ -o $@ -lrt
+	$(HIPCC) $^ -o $@ -lrt -L/opt/rocm/lib -lrccl -Wl,-rpath,/opt/rocm/lib
 
 $(BUILD)/%.o: %.cc
 	@mkdir -p $(dir $@)
