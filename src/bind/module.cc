@@ -17,6 +17,7 @@
 #include <vector>
 
 #include "../api/ucc.h"
+#include "../topo/topo.h"
 
 namespace py = pybind11;
 
@@ -371,6 +372,23 @@ PYBIND11_MODULE(_core, m)
         py::arg("src_displs") = std::vector<uint64_t>(),
         py::arg("dst_counts") = std::vector<uint64_t>(),
         py::arg("dst_displs") = std::vector<uint64_t>());
+
+    m.def("topo_sbgps", [](std::shared_ptr<CoreTeam> t) {
+        auto *team = reinterpret_cast<ucc::Team *>(t->team);
+        auto  node = ucc::topo::build_sbgp(team, ucc::topo::SbgpType::NODE);
+        auto  ldr =
+            ucc::topo::build_sbgp(team, ucc::topo::SbgpType::NODE_LEADERS);
+        py::dict d;
+        d["node_size"]    = node.ranks.size();
+        d["node_idx"]     = node.my_idx;
+        d["leaders_size"] = ldr.ranks.size();
+        d["leaders_idx"]  = ldr.my_idx;
+        return d;
+    });
+    m.def("gpu_link_hops", []() {
+        auto &g = ucc::topo::gpu_links();
+        return g.hops;
+    });
 
     /* ------------------------------------------------- EE / triggered */
     m.def("ee_create", [](std::shared_ptr<CoreTeam> t, uintptr_t stream) {

@@ -1203,7 +1203,19 @@ class Cdna4Tl final : public Tl {
 
     TlContext *context_create(Context *ctx) override
     {
-        if (!mc::hip_available()) {
+        auto &cfg = Config::instance();
+        cfg.declare("TL_CDNA4", "ENABLE", "1",
+                    "enable the native xGMI device transport");
+        cfg.declare("TL_CDNA4", "MAX_CONCURRENT", "4",
+                    "rotating in-flight collective slots");
+        cfg.declare("TL_CDNA4", "PERSISTENT_SLOTS", "2",
+                    "dedicated slots for persistent triggered colls");
+        cfg.declare("TL_CDNA4", "CHUNK_SIZE", "8m",
+                    "staging fragment bytes per slot area");
+        cfg.declare("TL_CDNA4", "FUSED_MAX", "512k",
+                    "max msg bytes for the fused single-kernel allreduce");
+        if (!cfg.get_bool("TL_CDNA4", "ENABLE", true) ||
+            !mc::hip_available()) {
             return nullptr;
         }
         int dev = 0;

@@ -313,6 +313,10 @@ class RcclTl final : public Tl {
     TlContext *context_create(Context *ctx) override
     {
         auto &cfg = Config::instance();
+        cfg.declare("TL_RCCL", "ENABLE", "1",
+                    "enable the RCCL comparison/fallback transport");
+        cfg.declare("TL_RCCL", "SCORE", "20",
+                    "score for rccl ranges (cdna4 default is 80)");
         if (!cfg.get_bool("TL_RCCL", "ENABLE", true) ||
             !mc::hip_available()) {
             return nullptr;

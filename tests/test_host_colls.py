@@ -314,3 +314,15 @@ def test_concurrent_colls(job):
         for o in all_outs[k]:
             np.testing.assert_allclose(o.astype(np.float64), all_exp[k],
                                        rtol=1e-5, atol=1e-5)
+
+
+def test_topo_sbgps(job):
+    """Single-node placement: NODE == FULL, one leader (SURVEY topo/sbgp
+    parity, reference ucc_sbgp.c subgroup discovery)."""
+    from ucc_amd import core
+    c = core()
+    info = c.topo_sbgps(job.teams[0])
+    assert info["node_size"] == job.n
+    assert info["node_idx"] == 0
+    assert info["leaders_size"] == 1
+    assert info["leaders_idx"] == 0
