@@ -3,6 +3,7 @@
  * core/ucc_team.c (nonblocking create state machine), core/ucc_coll.c
  * (score-map dispatch, zero-size fast path, persistent re-post). */
 #include "core.h"
+#include "../utils/profile.h"
 
 #include <cstdlib>
 #include <random>
@@ -535,6 +536,8 @@ ucc_status_t ucc_collective_init(ucc_coll_args_t *coll_args,
     }
     req->super.status = UCC_OPERATION_INITIALIZED;
     *request          = &req->super;
+    UCC_PROFILE_REQUEST_NEW(coll_type_name(req->args.coll_type),
+                            (uintptr_t)req);
     return UCC_OK;
 }
 
@@ -547,6 +550,7 @@ ucc_status_t ucc_collective_post(ucc_coll_req_h request)
     req->posted       = true;
     req->seq          = req->team->coll_seq++;
     req->super.status = UCC_INPROGRESS;
+    UCC_PROFILE_REQUEST_EVENT("post", (uintptr_t)req);
     task_start(req->task);
     if (req->task->status != UCC_INPROGRESS) {
         req->super.status = req->task->status;
@@ -563,6 +567,7 @@ ucc_status_t ucc_collective_test(ucc_coll_req_h request)
 ucc_status_t ucc_collective_finalize(ucc_coll_req_h request)
 {
     auto *req = reinterpret_cast<CollRequest *>(request);
+    UCC_PROFILE_REQUEST_FREE("finalize", (uintptr_t)req);
     delete req->task;
     delete req;
     return UCC_OK;
