@@ -25,6 +25,7 @@ LIB_SRCS := $(wildcard src/utils/*.cc) \
             $(wildcard src/topo/*.cc) \
             $(wildcard src/tl/self/*.cc) \
             $(wildcard src/tl/shm/*.cc) \
+            $(wildcard src/tl/tcp/*.cc) \
             $(wildcard src/tl/cdna4/*.cc) \
             $(wildcard src/tl/rccl/*.cc)
 KERNEL_SRCS := $(wildcard src/ec/kernels/*.hip) $(wildcard src/tl/cdna4/kernels/*.hip)

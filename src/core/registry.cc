@@ -6,6 +6,7 @@ namespace ucc {
 
 Tl *tl_self_iface();
 Tl *tl_shm_iface();
+Tl *tl_tcp_iface();
 #ifdef UCC_AMD_HAS_TL_CDNA4
 Tl *tl_cdna4_iface();
 #endif
@@ -18,6 +19,7 @@ void ensure_builtin_tls()
     static bool done = [] {
         register_tl(tl_self_iface());
         register_tl(tl_shm_iface());
+        register_tl(tl_tcp_iface());
 #ifdef UCC_AMD_HAS_TL_CDNA4
         register_tl(tl_cdna4_iface());
 #endif

@@ -874,6 +874,9 @@ class ShmTl final : public Tl {
     int         default_score() const override { return 40; }
     TlContext  *context_create(Context *ctx) override
     {
+        if (!Config::instance().get_bool("TL_SHM", "ENABLE", true)) {
+            return nullptr;
+        }
         return new ShmTlContext(ctx, this);
     }
     TlTeam *team_create(TlContext *tlc, Team *team) override

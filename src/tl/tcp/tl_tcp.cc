@@ -1,0 +1,1231 @@
+/* TL "tcp": host-memory inter-node transport — the tl/ucp role
+ *  without UCX: tagged p2p over nonblocking full-mesh TCP
+ * (p2p.h) with re-derived pattern algorithms:
+ *   allreduce       recursive doubling w/ non-power-of-2 fold
+ *                   (tl/ucp allreduce knomial family role)
+ *   bcast/reduce    binomial tree
+ *   (all)gather(v)  ring / linear-to-root
+ *   alltoall(v)     pairwise exchange (throttle-free small-n)
+ *   reduce_scatter(v) reduce + scatter schedule
+ *   barrier/fanin/fanout  binomial fanin-fanout
+ * Scores below tl/shm (5 vs 50): same-node teams stay on shared memory;
+ * TCP serves teams spanning nodes (team create declines nothing — any
+ * host-memory team works, it is the universal fallback).
+ *
+ * Team bootstrap: listen socket per context; {ip, port} published through
+ * the team's combined OOB exchange; mesh built lazily in create_test
+ * (lower rank accepts, higher connects, 4-byte rank hello). */
+#include <arpa/inet.h>
+#include <errno.h>
+#include <fcntl.h>
+#include <ifaddrs.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <algorithm>
+
+#include "../../core/core.h"
+#include "../../ec/ec_cpu.h"
+#include "p2p.h"
+
+namespace ucc {
+namespace tcp {
+
+/* ------------------------------------------------------- Conn progress */
+static bool io_fatal(ssize_t r)
+{
+    return r == 0 || (r < 0 && errno != EAGAIN && errno != EWOULDBLOCK &&
+                      errno != EINTR);
+}
+
+void Conn::match_header()
+{
+    /* find first posted recv with this tag */
+    for (auto it = recvq.begin(); it != recvq.end(); ++it) {
+        if ((*it)->tag == hdr.tag) {
+            cur_recv = *it;
+            recvq.erase(it);
+            return;
+        }
+    }
+    unexp.push_back(UnexpMsg{hdr.tag, std::vector<uint8_t>(hdr.len)});
+    cur_un = &unexp.back().data;
+}
+
+void Conn::progress()
+{
+    if (fd < 0) {
+        return;
+    }
+    /* ---- sends */
+    while (!sendq.empty()) {
+        SendOp *op = sendq.front();
+        if (op->off < sizeof(WireHdr)) {
+            WireHdr h{op->tag, op->len};
+            ssize_t r = ::send(fd, (uint8_t *)&h + op->off,
+                               sizeof(h) - op->off, MSG_NOSIGNAL);
+            if (r <= 0) {
+                if (io_fatal(r)) {
+                    ucc_error("tcp send hdr failed: %s", strerror(errno));
+                }
+                break;
+            }
+            op->off += (size_t)r;
+            if (op->off < sizeof(WireHdr)) {
+                break;
+            }
+        }
+        size_t poff = op->off - sizeof(WireHdr);
+        while (poff < op->len) {
+            ssize_t r = ::send(fd, op->buf + poff, op->len - poff,
+                               MSG_NOSIGNAL);
+            if (r <= 0) {
+                break;
+            }
+            poff += (size_t)r;
+            op->off += (size_t)r;
+        }
+        if (poff < op->len) {
+            break;
+        }
+        op->done = true;
+        sendq.pop_front();
+    }
+    /* ---- recvs */
+    while (true) {
+        if (hdr_got < sizeof(WireHdr)) {
+            ssize_t r = ::recv(fd, (uint8_t *)&hdr + hdr_got,
+                               sizeof(hdr) - hdr_got, 0);
+            if (r <= 0) {
+                if (io_fatal(r)) {
+                    /* peer gone: fatal only if mid-message */
+                }
+                return;
+            }
+            hdr_got += (size_t)r;
+            if (hdr_got < sizeof(WireHdr)) {
+                return;
+            }
+            pay_got = 0;
+            match_header();
+        }
+        uint8_t *dst =
+            cur_recv ? cur_recv->buf : (cur_un ? cur_un->data() : nullptr);
+        size_t want = hdr.len;
+        if (cur_recv && cur_recv->len < want) {
+            want = cur_recv->len; /* truncate (caller sized exactly) */
+        }
+        while (pay_got < hdr.len) {
+            uint8_t  sink[4096];
+            uint8_t *p;
+            size_t   room;
+            if (dst && pay_got < want) {
+                p    = dst + pay_got;
+                room = want - pay_got;
+            } else {
+                p    = sink;
+                room = std::min(hdr.len - pay_got, sizeof(sink));
+            }
+            ssize_t r = ::recv(fd, p, room, 0);
+            if (r <= 0) {
+                return;
+            }
+            pay_got += (size_t)r;
+        }
+        if (cur_recv) {
+            cur_recv->done = true;
+        }
+        cur_recv = nullptr;
+        cur_un   = nullptr;
+        hdr_got  = 0;
+    }
+}
+
+/* --------------------------------------------------------- context/team */
+struct TcpAddr {
+    uint32_t ip;
+    uint16_t port;
+    uint16_t pad;
+};
+
+static uint32_t local_ip()
+{
+    uint32_t        ip = htonl(INADDR_LOOPBACK);
+    struct ifaddrs *ifs = nullptr;
+    if (getifaddrs(&ifs) == 0) {
+        for (auto *i = ifs; i; i = i->ifa_next) {
+            if (i->ifa_addr && i->ifa_addr->sa_family == AF_INET) {
+                auto *sin = (struct sockaddr_in *)i->ifa_addr;
+                if (ntohl(sin->sin_addr.s_addr) != INADDR_LOOPBACK) {
+                    ip = sin->sin_addr.s_addr;
+                    break;
+                }
+            }
+        }
+        freeifaddrs(ifs);
+    }
+    return ip;
+}
+
+static void set_nonblock(int fd)
+{
+    fcntl(fd, F_SETFL, fcntl(fd, F_GETFL, 0) | O_NONBLOCK);
+    int one = 1;
+    setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+}
+
+class TcpTl;
+
+class TcpTlContext final : public TlContext {
+  public:
+    TcpTlContext(Context *ctx, Tl *tl) : TlContext(ctx), tl_(tl)
+    {
+        lfd_ = socket(AF_INET, SOCK_STREAM, 0);
+        if (lfd_ < 0) {
+            return;
+        }
+        struct sockaddr_in a{};
+        a.sin_family      = AF_INET;
+        a.sin_addr.s_addr = htonl(INADDR_ANY);
+        a.sin_port        = 0;
+        if (bind(lfd_, (struct sockaddr *)&a, sizeof(a)) != 0 ||
+            listen(lfd_, 64) != 0) {
+            close(lfd_);
+            lfd_ = -1;
+            return;
+        }
+        socklen_t al = sizeof(a);
+        getsockname(lfd_, (struct sockaddr *)&a, &al);
+        port_ = ntohs(a.sin_port);
+        ip_   = local_ip();
+        set_nonblock(lfd_);
+    }
+    ~TcpTlContext() override
+    {
+        if (lfd_ >= 0) {
+            close(lfd_);
+        }
+    }
+    Tl *iface() override;
+
+    Tl      *tl_;
+    int      lfd_ = -1;
+    uint32_t ip_  = 0;
+    uint16_t port_ = 0;
+};
+
+class TcpTlTeam final : public TlTeam {
+  public:
+    TcpTlTeam(TlContext *tlc, Team *team) : TlTeam(tlc, team)
+    {
+        conns_.resize(team->size);
+    }
+    ~TcpTlTeam() override
+    {
+        for (auto &c : conns_) {
+            if (c.fd >= 0) {
+                close(c.fd);
+            }
+        }
+    }
+
+    size_t exchg_size() override { return sizeof(TcpAddr); }
+
+    void exchg_pack(void *buf) override
+    {
+        auto   *c = (TcpTlContext *)tlc_;
+        TcpAddr a{c->ip_, c->port_, 0};
+        memcpy(buf, &a, sizeof(a));
+    }
+
+    ucc_status_t exchg_unpack(const void *all, size_t stride) override
+    {
+        addrs_.resize(team_->size);
+        for (uint32_t r = 0; r < team_->size; r++) {
+            memcpy(&addrs_[r], (const uint8_t *)all + r * stride,
+                   sizeof(TcpAddr));
+        }
+        return UCC_OK;
+    }
+
+    ucc_status_t create_test() override
+    {
+        auto          *c  = (TcpTlContext *)tlc_;
+        const uint32_t me = team_->rank;
+        const uint32_t n  = team_->size;
+        /* accept incoming (peers with rank > me connect to me) */
+        while (true) {
+            int fd = accept(c->lfd_, nullptr, nullptr);
+            if (fd < 0) {
+                break;
+            }
+            set_nonblock(fd);
+            pending_.push_back(fd);
+        }
+        /* read 8-byte hello {team_uid_lo32, rank} from pending fds */
+        for (auto it = pending_.begin(); it != pending_.end();) {
+            uint32_t hello[2];
+            ssize_t  r = ::recv(*it, hello, sizeof(hello), MSG_PEEK);
+            if (r == (ssize_t)sizeof(hello)) {
+                ::recv(*it, hello, sizeof(hello), 0);
+                if (hello[0] == (uint32_t)team_->team_uid &&
+                    hello[1] < n) {
+                    conns_[hello[1]].fd = *it;
+                } else {
+                    close(*it);
+                }
+                it = pending_.erase(it);
+            } else if (r == 0 || (r < 0 && errno != EAGAIN)) {
+                close(*it);
+                it = pending_.erase(it);
+            } else {
+                ++it;
+            }
+        }
+        /* connect to lower ranks */
+        for (uint32_t r = 0; r < me; r++) {
+            if (conns_[r].fd >= 0 || connecting_[r] >= 0) {
+                continue;
+            }
+            int fd = socket(AF_INET, SOCK_STREAM, 0);
+            struct sockaddr_in a{};
+            a.sin_family      = AF_INET;
+            a.sin_addr.s_addr = addrs_[r].ip;
+            a.sin_port        = htons(addrs_[r].port);
+            if (connect(fd, (struct sockaddr *)&a, sizeof(a)) == 0) {
+                uint32_t hello[2] = {(uint32_t)team_->team_uid, me};
+                if (::send(fd, hello, sizeof(hello), MSG_NOSIGNAL) !=
+                    (ssize_t)sizeof(hello)) {
+                    close(fd);
+                    return UCC_ERR_NO_RESOURCE;
+                }
+                set_nonblock(fd);
+                conns_[r].fd = fd;
+            } else {
+                close(fd);
+                /* retry next create_test call */
+            }
+        }
+        for (uint32_t r = 0; r < n; r++) {
+            if (r != me && conns_[r].fd < 0) {
+                return UCC_INPROGRESS;
+            }
+        }
+        return UCC_OK;
+    }
+
+    void get_scores(Team *team, ScoreMap &map) override;
+
+    void progress()
+    {
+        for (auto &c : conns_) {
+            c.progress();
+        }
+    }
+
+    uint64_t mktag(uint64_t seq, uint32_t step)
+    {
+        return ((uint64_t)team_->id << 48) | (seq << 16) | step;
+    }
+
+    std::vector<Conn>    conns_;
+    std::vector<TcpAddr> addrs_;
+    std::vector<int>     pending_;
+    std::map<uint32_t, int> connecting_init_; /* unused placeholder */
+    int                  connecting_[64] = {
+        -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1,
+        -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1,
+        -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1,
+        -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1, -1};
+    uint64_t seq_ = 1;
+};
+
+/* ------------------------------------------------------------ TcpTask  */
+/* Base: owns send/recv op storage, helpers over the team mesh. */
+class TcpTask : public Task {
+  public:
+    TcpTask(Context *ctx, TcpTlTeam *tt, const ucc_coll_args_t &args)
+        : Task(ctx), tt_(tt), a_(args)
+    {
+    }
+    ~TcpTask() override
+    {
+        for (auto *s : sends_) {
+            delete s;
+        }
+        for (auto *r : recvs_) {
+            delete r;
+        }
+    }
+
+  protected:
+    SendOp *send_to(uint32_t peer, uint32_t step, const void *buf,
+                    size_t len)
+    {
+        auto *op = new SendOp{tt_->mktag(seq_, step), (const uint8_t *)buf,
+                              len};
+        sends_.push_back(op);
+        tt_->conns_[peer].sendq.push_back(op);
+        return op;
+    }
+    RecvOp *recv_from(uint32_t peer, uint32_t step, void *buf, size_t len)
+    {
+        auto *op = new RecvOp{tt_->mktag(seq_, step), (uint8_t *)buf, len};
+        recvs_.push_back(op);
+        Conn &c = tt_->conns_[peer];
+        /* check unexpected queue first */
+        for (auto it = c.unexp.begin(); it != c.unexp.end(); ++it) {
+            if (it->tag == op->tag) {
+                memcpy(op->buf, it->data.data(),
+                       std::min(op->len, it->data.size()));
+                op->done = true;
+                c.unexp.erase(it);
+                return op;
+            }
+        }
+        c.recvq.push_back(op);
+        return op;
+    }
+    bool ops_done()
+    {
+        tt_->progress();
+        for (auto *s : sends_) {
+            if (!s->done) {
+                return false;
+            }
+        }
+        for (auto *r : recvs_) {
+            if (!r->done) {
+                return false;
+            }
+        }
+        return true;
+    }
+    void clear_ops()
+    {
+        /* all done => no queue references remain */
+        for (auto *s : sends_) {
+            delete s;
+        }
+        for (auto *r : recvs_) {
+            delete r;
+        }
+        sends_.clear();
+        recvs_.clear();
+    }
+
+    void begin()
+    {
+        me_  = tt_->team_->rank;
+        n_   = tt_->team_->size;
+        seq_ = tt_->seq_++;
+    }
+
+    TcpTlTeam            *tt_;
+    ucc_coll_args_t       a_;
+    std::vector<SendOp *> sends_;
+    std::vector<RecvOp *> recvs_;
+    uint32_t              me_ = 0, n_ = 1;
+    uint64_t              seq_ = 0;
+    int                   phase_ = 0, round_ = 0;
+};
+
+/* ---- allreduce: recursive doubling with non-power-of-2 fold
+ * (reference tl/ucp allreduce recursive knomial role, radix 2) */
+class TcpAllreduceTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        dt_    = a_.dst.info.datatype;
+        op_    = a_.op;
+        dtsz_  = ucc_dt_size(dt_);
+        count_ = a_.dst.info.count;
+        bytes_ = count_ * dtsz_;
+        dst_   = (uint8_t *)a_.dst.info.buffer;
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        if (!inplace) {
+            ec_cpu::copy(dst_, a_.src.info.buffer, bytes_);
+        }
+        tmp_.resize(bytes_);
+        n2_ = 1;
+        while (n2_ * 2 <= n_) {
+            n2_ *= 2;
+        }
+        extras_ = n_ - n2_;
+        phase_  = 0;
+        mask_   = 1;
+        status  = UCC_INPROGRESS;
+        return progress_();
+    }
+
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    /* relabeled id among the n2 participants, or -1 */
+    int relabel() const
+    {
+        if (me_ < 2 * extras_) {
+            return (me_ & 1) ? (int)(me_ / 2) : -1;
+        }
+        return (int)(me_ - extras_);
+    }
+    uint32_t unrelabel(int id) const
+    {
+        return id < (int)extras_ ? (uint32_t)(2 * id + 1)
+                                 : (uint32_t)(id + extras_);
+    }
+
+    ucc_status_t progress_()
+    {
+        switch (phase_) {
+        case 0: /* fold: even of first 2*extras sends to odd */
+            if (extras_ > 0 && me_ < 2 * extras_) {
+                if ((me_ & 1) == 0) {
+                    send_to(me_ + 1, 0, dst_, bytes_);
+                } else {
+                    recv_from(me_ - 1, 0, tmp_.data(), bytes_);
+                }
+            }
+            phase_ = 1;
+            [[fallthrough]];
+        case 1:
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            if (extras_ > 0 && me_ < 2 * extras_ && (me_ & 1)) {
+                const void *srcs[2] = {dst_, tmp_.data()};
+                ec_cpu::reduce(dst_, srcs, 2, count_, dt_, op_);
+            }
+            clear_ops();
+            phase_ = 2;
+            [[fallthrough]];
+        case 2: { /* recursive doubling rounds */
+            int id = relabel();
+            if (id < 0) {
+                phase_ = 4; /* folded-out rank waits for result */
+                return progress_();
+            }
+            if (mask_ >= n2_) {
+                phase_ = 4;
+                return progress_();
+            }
+            uint32_t peer = unrelabel(id ^ (int)mask_);
+            send_to(peer, 16 + (uint32_t)mask_, dst_, bytes_);
+            recv_from(peer, 16 + (uint32_t)mask_, tmp_.data(), bytes_);
+            phase_ = 3;
+            [[fallthrough]];
+        }
+        case 3: {
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            const void *srcs[2] = {dst_, tmp_.data()};
+            bool last = (mask_ * 2 >= n2_);
+            ec_cpu::reduce(dst_, srcs, 2, count_, dt_, op_,
+                           last && op_ == UCC_OP_AVG ? 1.0 / n_ : 1.0);
+            clear_ops();
+            mask_ *= 2;
+            phase_ = 2;
+            return progress_();
+        }
+        case 4: /* unfold: odd of first 2*extras sends result to even */
+            if (extras_ > 0 && me_ < 2 * extras_) {
+                if (me_ & 1) {
+                    send_to(me_ - 1, 1, dst_, bytes_);
+                } else {
+                    recv_from(me_ + 1, 1, dst_, bytes_);
+                }
+            }
+            phase_ = 5;
+            [[fallthrough]];
+        case 5:
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            return UCC_OK;
+        }
+        return UCC_INPROGRESS;
+    }
+
+    ucc_datatype_t     dt_ = UCC_DT_FLOAT32;
+    ucc_reduction_op_t op_ = UCC_OP_SUM;
+    size_t             dtsz_ = 4, count_ = 0, bytes_ = 0;
+    uint8_t           *dst_ = nullptr;
+    std::vector<uint8_t> tmp_;
+    uint32_t           n2_ = 1, extras_ = 0;
+    size_t             mask_ = 1;
+};
+
+/* ---- bcast: binomial tree from root */
+class TcpBcastTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        bytes_ = a_.src.info.count * ucc_dt_size(a_.src.info.datatype);
+        buf_   = (uint8_t *)a_.src.info.buffer;
+        vr_    = (me_ + n_ - a_.root) % n_;
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    ucc_status_t progress_()
+    {
+        if (phase_ == 0) {
+            /* receive from parent (highest set bit of vr) */
+            if (vr_ != 0) {
+                uint32_t hb = 1;
+                while (hb * 2 <= vr_) {
+                    hb *= 2;
+                }
+                uint32_t parent = (vr_ - hb + a_.root) % n_;
+                recv_from(parent, 0, buf_, bytes_);
+            }
+            phase_ = 1;
+        }
+        if (phase_ == 1) {
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            /* send to children: vr + m for m > highest bit, m < n */
+            uint32_t hb = 1;
+            while (hb <= vr_) {
+                hb *= 2;
+            }
+            for (uint32_t m = hb; vr_ + m < n_; m *= 2) {
+                uint32_t child = (vr_ + m + a_.root) % n_;
+                send_to(child, 0, buf_, bytes_);
+            }
+            phase_ = 2;
+        }
+        if (!ops_done()) {
+            return UCC_INPROGRESS;
+        }
+        clear_ops();
+        return UCC_OK;
+    }
+
+    size_t   bytes_ = 0;
+    uint8_t *buf_   = nullptr;
+    uint32_t vr_    = 0;
+};
+
+/* ---- barrier / fanin / fanout: binomial fanin to 0 then fanout */
+class TcpBarrierTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        fanin_  = a_.coll_type != UCC_COLL_TYPE_FANOUT;
+        fanout_ = a_.coll_type != UCC_COLL_TYPE_FANIN;
+        phase_  = fanin_ ? 0 : 2;
+        status  = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    ucc_status_t progress_()
+    {
+        if (phase_ == 0) { /* fanin: recv children then send parent */
+            for (uint32_t m = 1; me_ + m < n_; m *= 2) {
+                if (me_ & m) {
+                    break;
+                }
+                if ((me_ | m) < n_ && (me_ & (m - 1)) == 0) {
+                    recv_from(me_ + m, 0, &token_, 1);
+                }
+            }
+            phase_ = 1;
+        }
+        if (phase_ == 1) {
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            if (me_ != 0) {
+                uint32_t m = 1;
+                while ((me_ & m) == 0) {
+                    m *= 2;
+                }
+                send_to(me_ - m, 0, &token_, 1);
+            }
+            phase_ = 2;
+        }
+        if (phase_ == 2) {
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            if (!fanout_) {
+                return UCC_OK;
+            }
+            /* fanout = bcast of the token from 0 */
+            if (me_ != 0) {
+                uint32_t hb = 1;
+                while (hb * 2 <= me_) {
+                    hb *= 2;
+                }
+                recv_from(me_ - hb, 1, &token_, 1);
+            }
+            phase_ = 3;
+        }
+        if (phase_ == 3) {
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            uint32_t hb = 1;
+            while (hb <= me_) {
+                hb *= 2;
+            }
+            for (uint32_t m = hb; me_ + m < n_; m *= 2) {
+                send_to(me_ + m, 1, &token_, 1);
+            }
+            phase_ = 4;
+        }
+        if (!ops_done()) {
+            return UCC_INPROGRESS;
+        }
+        clear_ops();
+        return UCC_OK;
+    }
+
+    uint8_t token_ = 1;
+    bool    fanin_ = true, fanout_ = true;
+};
+
+/* ---- allgather(v): ring */
+class TcpAllgatherTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        const bool is_v    = a_.coll_type == UCC_COLL_TYPE_ALLGATHERV;
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        cnt_.resize(n_);
+        dsp_.resize(n_);
+        if (is_v) {
+            size_t ds = ucc_dt_size(a_.dst.info_v.datatype);
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = ((a_.flags & UCC_COLL_ARGS_FLAG_COUNT_64BIT)
+                               ? ((const uint64_t *)a_.dst.info_v.counts)[r]
+                               : ((const uint32_t *)a_.dst.info_v.counts)[r]) *
+                          ds;
+                dsp_[r] =
+                    ((a_.flags & UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT)
+                         ? ((const uint64_t *)a_.dst.info_v.displacements)[r]
+                         : ((const uint32_t *)
+                                a_.dst.info_v.displacements)[r]) *
+                    ds;
+            }
+            buf_ = (uint8_t *)a_.dst.info_v.buffer;
+        } else {
+            size_t ds    = ucc_dt_size(a_.dst.info.datatype);
+            size_t block = a_.dst.info.count * ds / n_;
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = block;
+                dsp_[r] = r * block;
+            }
+            buf_ = (uint8_t *)a_.dst.info.buffer;
+        }
+        if (!inplace) {
+            ec_cpu::copy(buf_ + dsp_[me_], a_.src.info.buffer, cnt_[me_]);
+        }
+        round_ = 0;
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    ucc_status_t progress_()
+    {
+        const uint32_t right = (me_ + 1) % n_;
+        const uint32_t left  = (me_ + n_ - 1) % n_;
+        while (round_ < (int)n_ - 1) {
+            if (phase_ == 0) {
+                uint32_t sb = (me_ + n_ - round_) % n_;     /* send block */
+                uint32_t rb = (me_ + n_ - round_ - 1) % n_; /* recv block */
+                if (cnt_[sb]) {
+                    send_to(right, (uint32_t)round_, buf_ + dsp_[sb],
+                            cnt_[sb]);
+                }
+                if (cnt_[rb]) {
+                    recv_from(left, (uint32_t)round_, buf_ + dsp_[rb],
+                              cnt_[rb]);
+                }
+                phase_ = 1;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            phase_ = 0;
+            round_++;
+        }
+        return UCC_OK;
+    }
+
+    std::vector<size_t> cnt_, dsp_;
+    uint8_t            *buf_ = nullptr;
+};
+
+/* ---- alltoall(v): pairwise exchange */
+class TcpAlltoallTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        const bool is_v = a_.coll_type == UCC_COLL_TYPE_ALLTOALLV;
+        scnt_.resize(n_);
+        sdsp_.resize(n_);
+        rcnt_.resize(n_);
+        rdsp_.resize(n_);
+        if (is_v) {
+            size_t ss = ucc_dt_size(a_.src.info_v.datatype);
+            size_t ds = ucc_dt_size(a_.dst.info_v.datatype);
+            for (uint32_t r = 0; r < n_; r++) {
+                auto cat = [&](const void *c, uint32_t i) {
+                    return (a_.flags & UCC_COLL_ARGS_FLAG_COUNT_64BIT)
+                               ? (size_t)((const uint64_t *)c)[i]
+                               : (size_t)((const uint32_t *)c)[i];
+                };
+                auto dat = [&](const void *d, uint32_t i) {
+                    return (a_.flags &
+                            UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT)
+                               ? (size_t)((const uint64_t *)d)[i]
+                               : (size_t)((const uint32_t *)d)[i];
+                };
+                scnt_[r] = cat(a_.src.info_v.counts, r) * ss;
+                sdsp_[r] = dat(a_.src.info_v.displacements, r) * ss;
+                rcnt_[r] = cat(a_.dst.info_v.counts, r) * ds;
+                rdsp_[r] = dat(a_.dst.info_v.displacements, r) * ds;
+            }
+            sbuf_ = (const uint8_t *)a_.src.info_v.buffer;
+            dbuf_ = (uint8_t *)a_.dst.info_v.buffer;
+        } else {
+            size_t ds    = ucc_dt_size(a_.dst.info.datatype);
+            size_t block = a_.dst.info.count * ds / n_;
+            for (uint32_t r = 0; r < n_; r++) {
+                scnt_[r] = rcnt_[r] = block;
+                sdsp_[r] = rdsp_[r] = r * block;
+            }
+            sbuf_ = (const uint8_t *)a_.src.info.buffer;
+            dbuf_ = (uint8_t *)a_.dst.info.buffer;
+        }
+        ec_cpu::copy(dbuf_ + rdsp_[me_], sbuf_ + sdsp_[me_], scnt_[me_]);
+        round_ = 1;
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    ucc_status_t progress_()
+    {
+        while (round_ < (int)n_) {
+            if (phase_ == 0) {
+                uint32_t to   = (me_ + round_) % n_;
+                uint32_t from = (me_ + n_ - round_) % n_;
+                if (scnt_[to]) {
+                    send_to(to, (uint32_t)round_, sbuf_ + sdsp_[to],
+                            scnt_[to]);
+                }
+                if (rcnt_[from]) {
+                    recv_from(from, (uint32_t)round_, dbuf_ + rdsp_[from],
+                              rcnt_[from]);
+                }
+                phase_ = 1;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            phase_ = 0;
+            round_++;
+        }
+        return UCC_OK;
+    }
+
+    std::vector<size_t> scnt_, sdsp_, rcnt_, rdsp_;
+    const uint8_t      *sbuf_ = nullptr;
+    uint8_t            *dbuf_ = nullptr;
+};
+
+/* ---- reduce: linear recv+reduce at root (small n); gather/scatter(v):
+ * linear to/from root; reduce_scatter(v): reduce@0 + scatterv */
+class TcpRootedTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        ct_ = a_.coll_type;
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        root_              = (uint32_t)a_.root;
+        switch (ct_) {
+        case UCC_COLL_TYPE_REDUCE: {
+            dt_    = a_.src.info.datatype;
+            op_    = a_.op;
+            dtsz_  = ucc_dt_size(dt_);
+            count_ = a_.src.info.count;
+            bytes_ = count_ * dtsz_;
+            if (me_ == root_) {
+                dst_ = (uint8_t *)a_.dst.info.buffer;
+                if (!inplace) {
+                    ec_cpu::copy(dst_, a_.src.info.buffer, bytes_);
+                }
+                tmp_.resize(bytes_ * (n_ - 1));
+                for (uint32_t r = 0; r < n_; r++) {
+                    if (r != me_) {
+                        uint32_t slot = r < me_ ? r : r - 1;
+                        recv_from(r, 0, tmp_.data() + slot * bytes_,
+                                  bytes_);
+                    }
+                }
+            } else {
+                send_to(root_, 0, a_.src.info.buffer, bytes_);
+            }
+            break;
+        }
+        case UCC_COLL_TYPE_REDUCE_SCATTER:
+        case UCC_COLL_TYPE_REDUCE_SCATTERV: {
+            /* reduce at 0 into tmp, then scatter slices */
+            const bool is_v = ct_ == UCC_COLL_TYPE_REDUCE_SCATTERV;
+            dt_  = is_v ? a_.dst.info_v.datatype : a_.dst.info.datatype;
+            op_  = a_.op;
+            dtsz_ = ucc_dt_size(dt_);
+            cnt_.resize(n_);
+            dsp_.resize(n_);
+            if (is_v) {
+                size_t off = 0;
+                for (uint32_t r = 0; r < n_; r++) {
+                    cnt_[r] = ((a_.flags & UCC_COLL_ARGS_FLAG_COUNT_64BIT)
+                                   ? ((const uint64_t *)
+                                          a_.dst.info_v.counts)[r]
+                                   : ((const uint32_t *)
+                                          a_.dst.info_v.counts)[r]) *
+                              dtsz_;
+                    dsp_[r] = off;
+                    off += cnt_[r];
+                }
+                bytes_ = off;
+                dst_   = inplace ? (uint8_t *)a_.dst.info_v.buffer + dsp_[me_]
+                                 : (uint8_t *)a_.dst.info_v.buffer;
+                sptr_  = inplace ? (const uint8_t *)a_.dst.info_v.buffer
+                                 : (const uint8_t *)a_.src.info.buffer;
+            } else {
+                size_t out_b;
+                if (inplace) {
+                    bytes_ = a_.dst.info.count * dtsz_;
+                    out_b  = bytes_ / n_;
+                    sptr_  = (const uint8_t *)a_.dst.info.buffer;
+                    dst_   = (uint8_t *)a_.dst.info.buffer + me_ * out_b;
+                } else {
+                    out_b  = a_.dst.info.count * dtsz_;
+                    bytes_ = out_b * n_;
+                    sptr_  = (const uint8_t *)a_.src.info.buffer;
+                    dst_   = (uint8_t *)a_.dst.info.buffer;
+                }
+                for (uint32_t r = 0; r < n_; r++) {
+                    cnt_[r] = out_b;
+                    dsp_[r] = r * out_b;
+                }
+            }
+            count_ = bytes_ / dtsz_;
+            if (me_ == 0) {
+                tmp_.resize(bytes_ * n_); /* [acc][peers...] */
+                ec_cpu::copy(tmp_.data(), sptr_, bytes_);
+                for (uint32_t r = 1; r < n_; r++) {
+                    recv_from(r, 0, tmp_.data() + r * bytes_, bytes_);
+                }
+            } else {
+                send_to(0, 0, sptr_, bytes_);
+            }
+            break;
+        }
+        case UCC_COLL_TYPE_GATHER:
+        case UCC_COLL_TYPE_GATHERV: {
+            const bool is_v = ct_ == UCC_COLL_TYPE_GATHERV;
+            if (me_ == root_) {
+                dt_ = is_v ? a_.dst.info_v.datatype : a_.dst.info.datatype;
+                dtsz_ = ucc_dt_size(dt_);
+                dst_  = is_v ? (uint8_t *)a_.dst.info_v.buffer
+                             : (uint8_t *)a_.dst.info.buffer;
+                cnt_.resize(n_);
+                dsp_.resize(n_);
+                if (is_v) {
+                    for (uint32_t r = 0; r < n_; r++) {
+                        cnt_[r] =
+                            ((a_.flags & UCC_COLL_ARGS_FLAG_COUNT_64BIT)
+                                 ? ((const uint64_t *)
+                                        a_.dst.info_v.counts)[r]
+                                 : ((const uint32_t *)
+                                        a_.dst.info_v.counts)[r]) *
+                            dtsz_;
+                        dsp_[r] =
+                            ((a_.flags &
+                              UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT)
+                                 ? ((const uint64_t *)
+                                        a_.dst.info_v.displacements)[r]
+                                 : ((const uint32_t *)
+                                        a_.dst.info_v.displacements)[r]) *
+                            dtsz_;
+                    }
+                } else {
+                    size_t block = a_.dst.info.count * dtsz_ / n_;
+                    for (uint32_t r = 0; r < n_; r++) {
+                        cnt_[r] = block;
+                        dsp_[r] = r * block;
+                    }
+                }
+                if (!inplace) {
+                    ec_cpu::copy(dst_ + dsp_[me_], a_.src.info.buffer,
+                                 cnt_[me_]);
+                }
+                for (uint32_t r = 0; r < n_; r++) {
+                    if (r != me_) {
+                        recv_from(r, 0, dst_ + dsp_[r], cnt_[r]);
+                    }
+                }
+            } else {
+                size_t b = a_.src.info.count *
+                           ucc_dt_size(a_.src.info.datatype);
+                send_to(root_, 0, a_.src.info.buffer, b);
+            }
+            break;
+        }
+        case UCC_COLL_TYPE_SCATTER:
+        case UCC_COLL_TYPE_SCATTERV: {
+            const bool is_v = ct_ == UCC_COLL_TYPE_SCATTERV;
+            if (me_ == root_) {
+                dt_ = is_v ? a_.src.info_v.datatype : a_.src.info.datatype;
+                dtsz_ = ucc_dt_size(dt_);
+                auto *sb = is_v ? (const uint8_t *)a_.src.info_v.buffer
+                                : (const uint8_t *)a_.src.info.buffer;
+                cnt_.resize(n_);
+                dsp_.resize(n_);
+                if (is_v) {
+                    for (uint32_t r = 0; r < n_; r++) {
+                        cnt_[r] =
+                            ((a_.flags & UCC_COLL_ARGS_FLAG_COUNT_64BIT)
+                                 ? ((const uint64_t *)
+                                        a_.src.info_v.counts)[r]
+                                 : ((const uint32_t *)
+                                        a_.src.info_v.counts)[r]) *
+                            dtsz_;
+                        dsp_[r] =
+                            ((a_.flags &
+                              UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT)
+                                 ? ((const uint64_t *)
+                                        a_.src.info_v.displacements)[r]
+                                 : ((const uint32_t *)
+                                        a_.src.info_v.displacements)[r]) *
+                            dtsz_;
+                    }
+                } else {
+                    size_t block = a_.src.info.count * dtsz_ / n_;
+                    for (uint32_t r = 0; r < n_; r++) {
+                        cnt_[r] = block;
+                        dsp_[r] = r * block;
+                    }
+                }
+                for (uint32_t r = 0; r < n_; r++) {
+                    if (r != me_) {
+                        send_to(r, 0, sb + dsp_[r], cnt_[r]);
+                    }
+                }
+                if (!inplace && a_.dst.info.buffer) {
+                    ec_cpu::copy(a_.dst.info.buffer, sb + dsp_[me_],
+                                 cnt_[me_]);
+                }
+            } else {
+                size_t b = a_.dst.info.count *
+                           ucc_dt_size(a_.dst.info.datatype);
+                recv_from(root_, 0, a_.dst.info.buffer, b);
+            }
+            break;
+        }
+        default:
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    ucc_status_t progress_()
+    {
+        if (phase_ == 0) {
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            if (ct_ == UCC_COLL_TYPE_REDUCE && me_ == root_ && n_ > 1) {
+                std::vector<const void *> srcs;
+                srcs.push_back(dst_);
+                for (uint32_t r = 0; r + 1 < n_; r++) {
+                    srcs.push_back(tmp_.data() + r * bytes_);
+                }
+                ec_cpu::reduce(dst_, srcs.data(), (int)srcs.size(), count_,
+                               dt_, op_,
+                               op_ == UCC_OP_AVG ? 1.0 / n_ : 1.0);
+            }
+            if ((ct_ == UCC_COLL_TYPE_REDUCE_SCATTER ||
+                 ct_ == UCC_COLL_TYPE_REDUCE_SCATTERV)) {
+                if (me_ == 0) {
+                    std::vector<const void *> srcs;
+                    for (uint32_t r = 0; r < n_; r++) {
+                        srcs.push_back(tmp_.data() + r * bytes_);
+                    }
+                    ec_cpu::reduce(tmp_.data(), srcs.data(),
+                                   (int)srcs.size(), count_, dt_, op_,
+                                   op_ == UCC_OP_AVG ? 1.0 / n_ : 1.0);
+                    /* scatter slices */
+                    for (uint32_t r = 1; r < n_; r++) {
+                        if (cnt_[r]) {
+                            send_to(r, 1, tmp_.data() + dsp_[r], cnt_[r]);
+                        }
+                    }
+                    ec_cpu::copy(dst_, tmp_.data() + dsp_[0], cnt_[0]);
+                } else {
+                    if (cnt_[me_]) {
+                        recv_from(0, 1, dst_, cnt_[me_]);
+                    }
+                }
+                phase_ = 1;
+                if (!ops_done()) {
+                    return UCC_INPROGRESS;
+                }
+                clear_ops();
+                return UCC_OK;
+            }
+            return UCC_OK;
+        }
+        if (!ops_done()) {
+            return UCC_INPROGRESS;
+        }
+        clear_ops();
+        return UCC_OK;
+    }
+
+    ucc_coll_type_t    ct_ = UCC_COLL_TYPE_REDUCE;
+    ucc_datatype_t     dt_ = UCC_DT_FLOAT32;
+    ucc_reduction_op_t op_ = UCC_OP_SUM;
+    size_t             dtsz_ = 4, count_ = 0, bytes_ = 0;
+    uint32_t           root_ = 0;
+    uint8_t           *dst_  = nullptr;
+    const uint8_t     *sptr_ = nullptr;
+    std::vector<uint8_t> tmp_;
+    std::vector<size_t>  cnt_, dsp_;
+};
+
+/* --------------------------------------------------------------- iface  */
+class TcpTl final : public Tl {
+  public:
+    const char *name() const override { return "tcp"; }
+    int         default_score() const override { return 5; }
+
+    TlContext *context_create(Context *ctx) override
+    {
+        auto &cfg = Config::instance();
+        cfg.declare("TL_TCP", "ENABLE", "1",
+                    "enable the TCP host transport (inter-node fallback)");
+        if (!cfg.get_bool("TL_TCP", "ENABLE", true)) {
+            return nullptr;
+        }
+        auto *c = new TcpTlContext(ctx, this);
+        if (c->lfd_ < 0) {
+            delete c;
+            return nullptr;
+        }
+        return c;
+    }
+
+    TlTeam *team_create(TlContext *tlc, Team *team) override
+    {
+        if (team->size < 2 || team->size > 64) {
+            return nullptr;
+        }
+        /* in-process multi-rank teams: every rank has its own context and
+         * listen socket, so the mesh works; allow. */
+        return new TcpTlTeam(tlc, team);
+    }
+};
+
+static TcpTl g_tcp_tl;
+
+Tl *TcpTlContext::iface() { return &g_tcp_tl; }
+
+void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
+{
+    (void)team;
+    TcpTlTeam *self = this;
+    int sc = (int)Config::instance().get_int("TL_TCP", "SCORE",
+                                             g_tcp_tl.default_score());
+    auto add = [&](ucc_coll_type_t ct, auto maker) {
+        ScoreRange r;
+        r.start    = 0;
+        r.end      = SIZE_MAX;
+        r.score    = sc;
+        r.tl_name  = "tcp";
+        r.alg_name = "p2p";
+        r.init     = [self, maker](const ucc_coll_args_t &args, Team *t,
+                               Task **task) -> ucc_status_t {
+            *task = maker(t->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(ct, UCC_MEMORY_TYPE_HOST, r);
+    };
+    auto mk = [](auto *tag) {
+        using T = std::remove_pointer_t<decltype(tag)>;
+        return [](Context *c, TcpTlTeam *tt, const ucc_coll_args_t &a)
+                   -> Task * { return new T(c, tt, a); };
+    };
+    add(UCC_COLL_TYPE_ALLREDUCE, mk((TcpAllreduceTask *)nullptr));
+    add(UCC_COLL_TYPE_BCAST, mk((TcpBcastTask *)nullptr));
+    add(UCC_COLL_TYPE_BARRIER, mk((TcpBarrierTask *)nullptr));
+    add(UCC_COLL_TYPE_FANIN, mk((TcpBarrierTask *)nullptr));
+    add(UCC_COLL_TYPE_FANOUT, mk((TcpBarrierTask *)nullptr));
+    add(UCC_COLL_TYPE_ALLGATHER, mk((TcpAllgatherTask *)nullptr));
+    add(UCC_COLL_TYPE_ALLGATHERV, mk((TcpAllgatherTask *)nullptr));
+    add(UCC_COLL_TYPE_ALLTOALL, mk((TcpAlltoallTask *)nullptr));
+    add(UCC_COLL_TYPE_ALLTOALLV, mk((TcpAlltoallTask *)nullptr));
+    add(UCC_COLL_TYPE_REDUCE, mk((TcpRootedTask *)nullptr));
+    add(UCC_COLL_TYPE_REDUCE_SCATTER, mk((TcpRootedTask *)nullptr));
+    add(UCC_COLL_TYPE_REDUCE_SCATTERV, mk((TcpRootedTask *)nullptr));
+    add(UCC_COLL_TYPE_GATHER, mk((TcpRootedTask *)nullptr));
+    add(UCC_COLL_TYPE_GATHERV, mk((TcpRootedTask *)nullptr));
+    add(UCC_COLL_TYPE_SCATTER, mk((TcpRootedTask *)nullptr));
+    add(UCC_COLL_TYPE_SCATTERV, mk((TcpRootedTask *)nullptr));
+}
+
+} // namespace tcp
+
+Tl *tl_tcp_iface() { return &tcp::g_tcp_tl; }
+
+} // namespace ucc

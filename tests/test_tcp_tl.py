@@ -1,0 +1,99 @@
+"""tl/tcp correctness: run the in-process jig with the shm TL disabled so
+every host collective goes over the TCP mesh (127.0.0.1 sockets) — the
+inter-node transport path (reference tl/ucp role), validated on CPU."""
+
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+WORKER = r"""
+import sys
+import numpy as np
+sys.path.insert(0, %r)
+from ucc_amd import core, dtypes
+from ucc_amd.testing import LocalJob
+
+n = 5  # non-power-of-2: exercises the fold path of recursive doubling
+job = LocalJob(n)
+rng = np.random.default_rng(3)
+
+# allreduce fp32 + bf16-ish dtype via int path
+for count in (7, 1024, 100_003):
+    arrs = [(rng.random(count) - 0.5).astype(np.float32) for _ in range(n)]
+    outs = job.allreduce_np(arrs)
+    exp = np.sum(arrs, axis=0)
+    for o in outs:
+        np.testing.assert_allclose(o, exp, rtol=1e-5, atol=1e-5)
+
+# bcast
+bufs = [np.zeros(5000, np.float64) for _ in range(n)]
+bufs[2][:] = rng.random(5000)
+exp = bufs[2].copy()
+reqs = job.coll("bcast", [
+    dict(src=b.ctypes.data, dst=0, count=5000, dt=dtypes.FLOAT64, root=2)
+    for b in bufs])
+job.run(reqs)
+for b in bufs:
+    np.testing.assert_array_equal(b, exp)
+
+# alltoall
+per = 321
+srcs = [(rng.random(per * n)).astype(np.float32) for _ in range(n)]
+dsts = [np.zeros(per * n, np.float32) for _ in range(n)]
+reqs = job.coll("alltoall", [
+    dict(src=srcs[r].ctypes.data, dst=dsts[r].ctypes.data, count=per * n,
+         dt=dtypes.FLOAT32) for r in range(n)])
+job.run(reqs)
+for d in range(n):
+    for s in range(n):
+        np.testing.assert_array_equal(
+            dsts[d][s * per:(s + 1) * per], srcs[s][d * per:(d + 1) * per])
+
+# allgatherv (uneven)
+cnts = [(r + 1) * 77 for r in range(n)]
+dsps = np.concatenate([[0], np.cumsum(cnts)[:-1]]).astype(np.uint64)
+total = int(sum(cnts))
+srcs = [(rng.random(cnts[r])).astype(np.float32) for r in range(n)]
+dsts = [np.zeros(total, np.float32) for _ in range(n)]
+reqs = job.coll("allgatherv", [
+    dict(src=srcs[r].ctypes.data, dst=dsts[r].ctypes.data, count=cnts[r],
+         dt=dtypes.FLOAT32, dst_counts=cnts, dst_displs=dsps.tolist())
+    for r in range(n)])
+job.run(reqs)
+exp = np.concatenate(srcs)
+for d in dsts:
+    np.testing.assert_array_equal(d, exp)
+
+# reduce_scatter
+per = 500
+srcs = [(rng.random(per * n)).astype(np.float32) for _ in range(n)]
+dsts = [np.zeros(per, np.float32) for _ in range(n)]
+reqs = job.coll("reduce_scatter", [
+    dict(src=srcs[r].ctypes.data, dst=dsts[r].ctypes.data, count=per,
+         dt=dtypes.FLOAT32) for r in range(n)])
+job.run(reqs)
+exp = np.sum(srcs, axis=0)
+for r in range(n):
+    np.testing.assert_allclose(dsts[r], exp[r * per:(r + 1) * per],
+                               rtol=1e-5, atol=1e-5)
+
+# barrier
+reqs = job.coll("barrier", [dict(src=0, dst=0, count=0, dt=dtypes.FLOAT32)
+                            for _ in range(n)])
+job.run(reqs)
+
+print("TCP_TL_OK")
+""" % (REPO,)
+
+
+def test_tcp_tl_colls():
+    env = dict(os.environ)
+    env["UCC_TL_SHM_ENABLE"] = "0"
+    p = subprocess.run([sys.executable, "-c", WORKER], env=env,
+                       capture_output=True, text=True, timeout=300)
+    sys.stdout.write(p.stdout[-2000:])
+    sys.stderr.write(p.stderr[-2000:])
+    assert p.returncode == 0
+    assert "TCP_TL_OK" in p.stdout

@@ -19,6 +19,7 @@
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
+#include <map>
 #include <ctime>
 #include <string>
 #include <vector>
@@ -181,43 +182,52 @@ struct Rank {
     Bufs          bufs;
 };
 
-/* in-process OOB: memcpy allgather across the rank array */
+/* in-process OOB: memcpy allgather across the rank array. Rounds are
+ * keyed by per-rank sequence number (ranks advance through team-create
+ * rounds at different paces, so a single blob slot per rank would get
+ * overwritten before slower ranks copied it). */
 struct LocalOobState {
-    std::vector<std::vector<uint8_t>> blobs;
-    std::vector<int>                  seq;
-    int                               n;
+    struct Round {
+        std::vector<std::vector<uint8_t>> blobs;
+        int                               arrived = 0;
+    };
+    std::map<uint64_t, Round> rounds;
+    std::vector<uint64_t>     next_round;
+    int                       n;
 };
 static LocalOobState g_loob;
 
 struct LocalOobReq {
-    int    rank;
-    void  *recv;
-    size_t size;
-    bool   done;
+    int      rank;
+    uint64_t round;
+    void    *recv;
+    size_t   size;
 };
 
 static ucc_status_t loob_allgather(void *src, void *recv, size_t size,
                                    void *info, void **req)
 {
-    int   rank = (int)(intptr_t)info;
-    auto *r    = new LocalOobReq{rank, recv, size, false};
-    g_loob.blobs[rank].assign((uint8_t *)src, (uint8_t *)src + size);
-    g_loob.seq[rank]++;
-    *req = r;
+    int      rank  = (int)(intptr_t)info;
+    uint64_t round = g_loob.next_round[rank]++;
+    auto    &r     = g_loob.rounds[round];
+    if (r.blobs.empty()) {
+        r.blobs.resize(g_loob.n);
+    }
+    r.blobs[rank].assign((uint8_t *)src, (uint8_t *)src + size);
+    r.arrived++;
+    *req = new LocalOobReq{rank, round, recv, size};
     return UCC_OK;
 }
 static ucc_status_t loob_test(void *req)
 {
-    auto *r    = (LocalOobReq *)req;
-    int   want = g_loob.seq[r->rank];
-    for (int i = 0; i < g_loob.n; i++) {
-        if (g_loob.seq[i] < want) {
-            return UCC_INPROGRESS;
-        }
+    auto *r  = (LocalOobReq *)req;
+    auto &rd = g_loob.rounds[r->round];
+    if (rd.arrived < g_loob.n) {
+        return UCC_INPROGRESS;
     }
     for (int i = 0; i < g_loob.n; i++) {
         memcpy((uint8_t *)r->recv + (size_t)i * r->size,
-               g_loob.blobs[i].data(), r->size);
+               rd.blobs[i].data(), r->size);
     }
     return UCC_OK;
 }
@@ -477,8 +487,7 @@ static int run_inproc(const Opts &o)
 {
     int n = o.nranks;
     g_loob.n = n;
-    g_loob.blobs.resize(n);
-    g_loob.seq.assign(n, 0);
+    g_loob.next_round.assign(n, 0);
     std::vector<Rank> ranks(n);
     for (int i = 0; i < n; i++) {
         setup_rank(ranks[i], i, n, false);
