@@ -105,6 +105,56 @@ ucc_status_t fused_allreduce_graph(const GraphFusedArgs &a,
 bool dt_supported(ucc_datatype_t dt);
 bool op_supported(ucc_datatype_t dt, ucc_reduction_op_t op);
 
+/* ------------------------------------------------------------------------
+ * Device-gated staged pipeline (large-message allreduce fast path).
+ *
+ * Replaces the host-polled shm gating of the staged-linear algorithm with
+ * kernel-prologue spins on per-(slot,parity,phase) cumulative block
+ * counters in each rank's fine-grained flags buffer: the host enqueues
+ * stage/reduce/gather kernels for EVERY fragment up front (two streams)
+ * and waits for one trailing event — no host round trips inside the
+ * collective. Targets are host-computed cumulative launch counts
+ * (collectives post in the same order on every rank), x kGatedBlocks
+ * block-increments per launch.
+ *
+ * Counter layout in the flags buffer (u64 index):
+ *   kGatedCntBase + (phase * kGatedSlots + slot) * 2 + parity
+ *   phase: 0=stage 1=reduce 2=gather; slot < kGatedSlots; parity = frag&1.
+ */
+constexpr int kGatedBlocks  = 32;
+constexpr int kGatedSlots   = 8;
+constexpr int kGatedCntBase = 704; /* after graph counters (128..639) */
+
+struct GatedArgs {
+    /* data */
+    const void *src;      /* stage: user src frag; others unused        */
+    void       *dst;      /* gather: user dst frag; others unused       */
+    void       *my_in;    /* my in[slot][parity] staging area           */
+    void       *my_out;   /* my out[slot][parity] reduced-slice area    */
+    const void *peer_in[kMaxRanks];  /* peers' in areas (xGMI)          */
+    const void *peer_out[kMaxRanks]; /* peers' out areas (xGMI)         */
+    uint64_t   *local_flags;
+    uint64_t   *peer_flags[kMaxRanks];
+    uint64_t   *error_word;
+    uint64_t    len;     /* fragment bytes                              */
+    uint64_t    sl_b, sl_e; /* my reduce slice [begin,end) bytes        */
+    /* per-rank slice table for gather (byte offsets into out areas)    */
+    uint64_t    slice_b[kMaxRanks], slice_e[kMaxRanks];
+    int         rank, nranks, slot, parity;
+    ucc_datatype_t     dt;
+    ucc_reduction_op_t op;
+    float       alpha;
+    /* cumulative block-count targets (host-tracked):
+     *   stage waits  reduce_cnt >= t_prev_reduce  (parity reuse)
+     *   reduce waits stage_cnt  >= t_stage  AND gather_cnt >= t_prev_gather
+     *   gather waits reduce_cnt >= t_reduce                             */
+    uint64_t    t_prev_reduce, t_prev_gather, t_stage, t_reduce;
+};
+
+ucc_status_t staged_stage(const GatedArgs &a, hipStream_t s);
+ucc_status_t staged_reduce(const GatedArgs &a, hipStream_t s);
+ucc_status_t staged_gather(const GatedArgs &a, hipStream_t s);
+
 } // namespace ec_hip
 } // namespace ucc
 

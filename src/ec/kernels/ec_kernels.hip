@@ -385,30 +385,35 @@ __global__ void k_fused_allreduce(FusedArgs a)
      * staging counter is u64 index kMaxRanks*kMaxSlots + slot (monotone,
      * grows by nblocks per use -> target seq*nblocks). */
     uint64_t *stage_cnt = a.local_flags + 8 * kMaxRanks + a.slot;
+    __shared__ int s_err;
     if (threadIdx.x == 0) {
+        s_err = 0;
         __hip_atomic_fetch_add(stage_cnt, 1, __ATOMIC_RELAXED,
                                __HIP_MEMORY_SCOPE_AGENT);
     }
+    __syncthreads();
     if (blockIdx.x == 0) {
         if (threadIdx.x == 0) {
             uint64_t spins = 0;
             while (sys_load(stage_cnt) < a.stage_target) {
                 if (++spins > kSpinLimit) {
                     sys_store(a.error_word, 1);
-                    return;
+                    s_err = 1;
+                    break;
                 }
                 __builtin_amdgcn_s_sleep(2);
             }
         }
         __syncthreads();
-        if ((int)threadIdx.x < a.nranks) {
+        if (!s_err && (int)threadIdx.x < a.nranks) {
             sys_store(a.peer_flags[threadIdx.x] +
                           (uint64_t)a.slot * kMaxRanks + a.rank,
                       a.seq);
         }
     }
-    /* 3. all blocks wait for all ranks' arrivals */
-    if (threadIdx.x < 64) {
+    /* 3. all blocks wait for all ranks' arrivals (timeout: all threads
+     * leave together — never return while siblings sit at a barrier) */
+    if (!s_err && threadIdx.x < 64) {
         int      j     = (int)threadIdx.x;
         uint64_t spins = 0;
         if (j < a.nranks) {
@@ -417,13 +422,17 @@ __global__ void k_fused_allreduce(FusedArgs a)
             while (sys_load(f) < a.seq) {
                 if (++spins > kSpinLimit) {
                     sys_store(a.error_word, 1);
-                    return;
+                    s_err = 1;
+                    break;
                 }
                 __builtin_amdgcn_s_sleep(2);
             }
         }
     }
     __syncthreads();
+    if (s_err) {
+        return;
+    }
     __threadfence_system(); /* acquire: drop stale lines before peer reads */
     /* 4. reduce all peers' scratch into dst */
     {
@@ -501,10 +510,13 @@ __global__ void k_fused_allreduce_graph(GraphFusedArgs a)
     __syncthreads();
     /* 2. grid arrival; block 0 signals all peers with seq */
     uint64_t *stage_cnt = a.local_flags + kStageCntBase + a.slot;
+    __shared__ int s_err;
     if (threadIdx.x == 0) {
+        s_err = 0;
         __hip_atomic_fetch_add(stage_cnt, 1, __ATOMIC_RELAXED,
                                __HIP_MEMORY_SCOPE_AGENT);
     }
+    __syncthreads();
     if (blockIdx.x == 0) {
         if (threadIdx.x == 0) {
             uint64_t spins  = 0;
@@ -512,20 +524,22 @@ __global__ void k_fused_allreduce_graph(GraphFusedArgs a)
             while (sys_load(stage_cnt) < target) {
                 if (++spins > kSpinLimit) {
                     sys_store(a.error_word, 1);
-                    return;
+                    s_err = 1;
+                    break;
                 }
                 __builtin_amdgcn_s_sleep(2);
             }
         }
         __syncthreads();
-        if ((int)threadIdx.x < a.nranks) {
+        if (!s_err && (int)threadIdx.x < a.nranks) {
             sys_store(a.peer_flags[threadIdx.x] +
                           (uint64_t)a.slot * kMaxRanks + a.rank,
                       seq);
         }
     }
-    /* 3. wait all ranks arrived for this iteration */
-    if (threadIdx.x < 64) {
+    /* 3. wait all ranks arrived for this iteration (timeout: all threads
+     * leave together) */
+    if (!s_err && threadIdx.x < 64) {
         int      j     = (int)threadIdx.x;
         uint64_t spins = 0;
         if (j < a.nranks) {
@@ -534,13 +548,17 @@ __global__ void k_fused_allreduce_graph(GraphFusedArgs a)
             while (sys_load(f) < seq) {
                 if (++spins > kSpinLimit) {
                     sys_store(a.error_word, 1);
-                    return;
+                    s_err = 1;
+                    break;
                 }
                 __builtin_amdgcn_s_sleep(2);
             }
         }
     }
     __syncthreads();
+    if (s_err) {
+        return;
+    }
     __threadfence_system();
     /* 4. reduce all peers' scratch (this iteration's parity) into dst.
      * Why parity is sufficient: a peer can start staging replay i+1 while
@@ -588,6 +606,155 @@ __global__ void k_fused_allreduce_graph(GraphFusedArgs a)
             ((T *)a.dst)[t] = Cvt<T>::store(apply_alpha<A>(r, a.alpha));
         }
     }
+}
+
+/* ------------------------------------------- device-gated staged colls */
+__device__ __forceinline__ int gated_idx(int phase, int slot, int parity)
+{
+    return kGatedCntBase + (phase * kGatedSlots + slot) * 2 + parity;
+}
+
+/* all blocks wait until every rank's counter[idx] >= target; on spin
+ * timeout ALL threads leave together (no thread may return while others
+ * sit at __syncthreads — that would hang the block forever). */
+__device__ __forceinline__ bool
+gated_wait(const GatedArgs &a, int phase, uint64_t target)
+{
+    __shared__ int s_err;
+    if (threadIdx.x == 0) {
+        s_err = 0;
+    }
+    __syncthreads();
+    if (target != 0 && threadIdx.x < 64) {
+        int j = (int)threadIdx.x;
+        if (j < a.nranks) {
+            const uint64_t *f =
+                a.peer_flags[j] + gated_idx(phase, a.slot, a.parity);
+            uint64_t spins = 0;
+            while (sys_load(f) < target) {
+                if (++spins > kSpinLimit) {
+                    sys_store(a.error_word, 1);
+                    s_err = 1;
+                    break;
+                }
+                __builtin_amdgcn_s_sleep(2);
+            }
+        }
+    }
+    __syncthreads();
+    if (s_err) {
+        return false;
+    }
+    __threadfence_system(); /* acquire: see peers' payload writes */
+    return true;
+}
+
+/* per-block completion signal on my own counter */
+__device__ __forceinline__ void gated_signal(const GatedArgs &a, int phase)
+{
+    __threadfence_system();
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        __hip_atomic_fetch_add(a.local_flags +
+                                   gated_idx(phase, a.slot, a.parity),
+                               1, __ATOMIC_RELEASE,
+                               __HIP_MEMORY_SCOPE_SYSTEM);
+    }
+}
+
+__device__ __forceinline__ void d_copy_bytes(uint8_t *dst,
+                                             const uint8_t *src,
+                                             uint64_t len)
+{
+    const uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const uint64_t str = (uint64_t)gridDim.x * blockDim.x;
+    if ((((uintptr_t)dst | (uintptr_t)src) & 15) == 0) {
+        uint64_t nv = len / 16;
+        for (uint64_t i = tid; i < nv; i += str) {
+            ((uint4 *)dst)[i] = ((const uint4 *)src)[i];
+        }
+        for (uint64_t i = nv * 16 + tid; i < len; i += str) {
+            dst[i] = src[i];
+        }
+    } else {
+        for (uint64_t i = tid; i < len; i += str) {
+            dst[i] = src[i];
+        }
+    }
+}
+
+__global__ void k_staged_stage(GatedArgs a)
+{
+    if (!gated_wait(a, 1, a.t_prev_reduce)) { /* parity-area reuse */
+        return;
+    }
+    d_copy_bytes((uint8_t *)a.my_in, (const uint8_t *)a.src, a.len);
+    gated_signal(a, 0);
+}
+
+template <typename T, int OP, int VEC>
+__global__ void k_staged_reduce(GatedArgs a)
+{
+    using A = typename Cvt<T>::A;
+    if (!gated_wait(a, 0, a.t_stage) ||
+        !gated_wait(a, 2, a.t_prev_gather)) {
+        return;
+    }
+    const uint64_t cnt = (a.sl_e - a.sl_b) / sizeof(T);
+    const uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const uint64_t str = (uint64_t)gridDim.x * blockDim.x;
+    const uint64_t nv  = cnt / VEC;
+    const int      n   = a.nranks;
+    using P            = Pack<T, OP, VEC>;
+    const P *srcs[kMaxRanks];
+    for (int s = 0; s < n; s++) {
+        srcs[s] = (const P *)((const uint8_t *)a.peer_in[s] + a.sl_b);
+    }
+    P *out = (P *)a.my_out;
+    for (uint64_t i = tid; i < nv; i += str) {
+        P acc = srcs[0][i];
+        A r[VEC];
+#pragma unroll
+        for (int k = 0; k < VEC; k++) {
+            r[k] = Cvt<T>::load(acc.v[k]);
+        }
+        for (int s = 1; s < n; s++) {
+            P x = srcs[s][i];
+#pragma unroll
+            for (int k = 0; k < VEC; k++) {
+                r[k] = red<A, OP>(r[k], Cvt<T>::load(x.v[k]));
+            }
+        }
+        P o;
+#pragma unroll
+        for (int k = 0; k < VEC; k++) {
+            o.v[k] = Cvt<T>::store(apply_alpha<A>(r[k], a.alpha));
+        }
+        out[i] = o;
+    }
+    for (uint64_t t = nv * VEC + tid; t < cnt; t += str) {
+        A r = Cvt<T>::load(((const T *)srcs[0])[t]);
+        for (int s = 1; s < n; s++) {
+            r = red<A, OP>(r, Cvt<T>::load(((const T *)srcs[s])[t]));
+        }
+        ((T *)out)[t] = Cvt<T>::store(apply_alpha<A>(r, a.alpha));
+    }
+    gated_signal(a, 1);
+}
+
+__global__ void k_staged_gather(GatedArgs a)
+{
+    if (!gated_wait(a, 1, a.t_reduce)) {
+        return;
+    }
+    for (int r = 0; r < a.nranks; r++) {
+        uint64_t b = a.slice_b[r], e = a.slice_e[r];
+        if (e > b) {
+            d_copy_bytes((uint8_t *)a.dst + b,
+                         (const uint8_t *)a.peer_out[r], e - b);
+        }
+    }
+    gated_signal(a, 2);
 }
 
 /* ----------------------------------------------------------- launchers */
@@ -742,6 +909,51 @@ ucc_status_t gather_copy(const GatherArgs &a, hipStream_t s)
         blocks = 1;
     }
     hipLaunchKernelGGL(k_gather_copy, dim3(blocks), dim3(threads), 0, s, a);
+    return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
+}
+
+ucc_status_t staged_stage(const GatedArgs &a, hipStream_t s)
+{
+    hipLaunchKernelGGL(k_staged_stage, dim3(kGatedBlocks), dim3(256), 0, s,
+                       a);
+    return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
+}
+
+template <typename T, int OP>
+static ucc_status_t launch_staged_reduce(const GatedArgs &a, hipStream_t s)
+{
+    constexpr int VEC = VecOf<T>::value;
+    hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC>), dim3(kGatedBlocks),
+                       dim3(256), 0, s, a);
+    return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
+}
+
+ucc_status_t staged_reduce(const GatedArgs &a, hipStream_t s)
+{
+    ucc_reduction_op_t op = a.op;
+    switch (a.dt) {
+        UCC_DT_CASE_FLOAT(UCC_DT_BFLOAT16, bf16_t, launch_staged_reduce)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT16, fp16_t, launch_staged_reduce)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT32, float, launch_staged_reduce)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT64, double, launch_staged_reduce)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT8_E4M3, e4m3_t, launch_staged_reduce)
+        UCC_DT_CASE_FLOAT(UCC_DT_FLOAT8_E5M2, e5m2_t, launch_staged_reduce)
+        UCC_DT_CASE_INT(UCC_DT_INT8, int8_t, launch_staged_reduce)
+        UCC_DT_CASE_INT(UCC_DT_UINT8, uint8_t, launch_staged_reduce)
+        UCC_DT_CASE_INT(UCC_DT_INT16, int16_t, launch_staged_reduce)
+        UCC_DT_CASE_INT(UCC_DT_UINT16, uint16_t, launch_staged_reduce)
+        UCC_DT_CASE_INT(UCC_DT_INT32, int32_t, launch_staged_reduce)
+        UCC_DT_CASE_INT(UCC_DT_UINT32, uint32_t, launch_staged_reduce)
+        UCC_DT_CASE_INT(UCC_DT_INT64, int64_t, launch_staged_reduce)
+        UCC_DT_CASE_INT(UCC_DT_UINT64, uint64_t, launch_staged_reduce)
+    default: return UCC_ERR_NOT_SUPPORTED;
+    }
+}
+
+ucc_status_t staged_gather(const GatedArgs &a, hipStream_t s)
+{
+    hipLaunchKernelGGL(k_staged_gather, dim3(kGatedBlocks), dim3(256), 0,
+                       s, a);
     return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
 }
 

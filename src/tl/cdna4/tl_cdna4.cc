@@ -285,6 +285,10 @@ class Cdna4TlTeam final : public TlTeam {
     ShmSeg      seg_;
     std::string seg_name_;
     std::vector<bool> pslot_used_;
+    /* device-gated pipeline: cumulative kernel-launch counts per
+     * (phase, slot, parity) — identical on every rank because
+     * collectives post in the same order (targets = count x 32 blocks) */
+    uint64_t gated_launch_[3][ec_hip::kGatedSlots][2] = {};
     uint64_t    seq_ = 0;
     std::vector<uint64_t> stage_cum_; /* per-slot fused block arrivals */
     uint8_t    *scratch_ = nullptr;
@@ -1195,6 +1199,145 @@ class StagedTask final : public Cdna4Task {
     bool            copy_pending_ = false, b_launched_ = false;
 };
 
+/* Device-gated staged allreduce: the large-message fast path. All
+ * fragments' stage/reduce/gather kernels are enqueued up front (stage on
+ * the copy stream, reduce+gather on the compute stream); cross-rank and
+ * cross-fragment ordering is enforced INSIDE the kernels by prologue
+ * spins on cumulative per-(slot,parity,phase) block counters in the
+ * peers' fine-grained flag buffers (ec_hip::GatedArgs) — zero host
+ * round-trips inside the collective, one trailing event. Same xGMI
+ * traffic as the host-gated staged path (1.75 x S per rank), minus all
+ * host gating latency. */
+class GatedAllreduceTask final : public Cdna4Task {
+  public:
+    using Cdna4Task::Cdna4Task;
+
+    ucc_status_t post() override
+    {
+        begin_use();
+        if (slot_ >= (uint32_t)ec_hip::kGatedSlots) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        dt_    = a_.dst.info.datatype;
+        dtsz_  = ucc_dt_size(dt_);
+        op_    = a_.op;
+        alpha_ = 1.0f;
+        if (op_ == UCC_OP_AVG) {
+            op_    = (ucc_reduction_op_t)12;
+            alpha_ = 1.0f / (float)n_;
+        }
+        total_  = a_.dst.info.count * dtsz_;
+        dbuf_   = (uint8_t *)a_.dst.info.buffer;
+        sbuf_   = inplace ? dbuf_ : (const uint8_t *)a_.src.info.buffer;
+        nfrags_ = (total_ + tt_->cfg_.chunk - 1) / tt_->cfg_.chunk;
+        if (nfrags_ == 0) {
+            nfrags_ = 1;
+        }
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress();
+    }
+
+    ucc_status_t progress() override
+    {
+        if (phase_ == 0) { /* entry: previous use of this slot closed */
+            if (!all_ge(0)) {
+                return UCC_INPROGRESS;
+            }
+            ucc_status_t st = enqueue_all();
+            if (st != UCC_OK) {
+                return st;
+            }
+            hipEventRecord(ev(0), comp());
+            phase_ = 1;
+        }
+        hipError_t e = hipEventQuery(ev(0));
+        if (e == hipErrorNotReady) {
+            return UCC_INPROGRESS;
+        }
+        if (e != hipSuccess) {
+            return UCC_ERR_NO_RESOURCE;
+        }
+        if (*tt_->err_host_ != 0) {
+            ucc_error("gated allreduce timed out waiting for peers");
+            close_slot();
+            return UCC_ERR_TIMED_OUT;
+        }
+        close_slot();
+        return UCC_OK;
+    }
+
+  private:
+    ucc_status_t enqueue_all()
+    {
+        const size_t chunk = tt_->cfg_.chunk;
+        for (size_t f = 0; f < nfrags_; f++) {
+            const uint32_t p   = (uint32_t)(f & 1);
+            const size_t   off = f * chunk;
+            const size_t   len =
+                total_ - off < chunk ? total_ - off : chunk;
+            ec_hip::GatedArgs ga{};
+            ga.src    = sbuf_ + off;
+            ga.dst    = dbuf_ + off;
+            ga.my_in  = tt_->area(me_, slot_, p, 0);
+            ga.my_out = tt_->area(me_, slot_, p, 1);
+            for (uint32_t r = 0; r < n_; r++) {
+                ga.peer_in[r]    = tt_->area(r, slot_, p, 0);
+                ga.peer_out[r]   = tt_->area(r, slot_, p, 1);
+                ga.peer_flags[r] = tt_->peers_[r].flags;
+            }
+            ga.local_flags = tt_->flags_;
+            ga.error_word  = tt_->err_host_;
+            ga.len         = len;
+            /* byte slices, 256-aligned, tail to last rank */
+            size_t per = (len / n_) & ~(size_t)255;
+            for (uint32_t r = 0; r < n_; r++) {
+                ga.slice_b[r] = (uint64_t)r * per;
+                ga.slice_e[r] = r == n_ - 1 ? len : (uint64_t)(r + 1) * per;
+            }
+            ga.sl_b   = ga.slice_b[me_];
+            ga.sl_e   = ga.slice_e[me_];
+            ga.rank   = (int)me_;
+            ga.nranks = (int)n_;
+            ga.slot   = (int)slot_;
+            ga.parity = (int)p;
+            ga.dt     = dt_;
+            ga.op     = op_;
+            ga.alpha  = alpha_;
+            auto &L   = tt_->gated_launch_;
+            const uint64_t B = (uint64_t)ec_hip::kGatedBlocks;
+            ga.t_prev_reduce = L[1][slot_][p] * B;
+            ga.t_prev_gather = L[2][slot_][p] * B;
+            ga.t_stage       = (L[0][slot_][p] + 1) * B;
+            ga.t_reduce      = (L[1][slot_][p] + 1) * B;
+            ucc_status_t st = ec_hip::staged_stage(ga, copy_s());
+            if (st != UCC_OK) {
+                return st;
+            }
+            L[0][slot_][p]++;
+            st = ec_hip::staged_reduce(ga, comp());
+            if (st != UCC_OK) {
+                return st;
+            }
+            L[1][slot_][p]++;
+            st = ec_hip::staged_gather(ga, comp());
+            if (st != UCC_OK) {
+                return st;
+            }
+            L[2][slot_][p]++;
+        }
+        return UCC_OK;
+    }
+
+    ucc_datatype_t     dt_    = UCC_DT_FLOAT32;
+    ucc_reduction_op_t op_    = UCC_OP_SUM;
+    float              alpha_ = 1.0f;
+    size_t             dtsz_ = 4, total_ = 0, nfrags_ = 0;
+    const uint8_t     *sbuf_ = nullptr;
+    uint8_t           *dbuf_ = nullptr;
+};
+
 /* ------------------------------------------------------------ scoring  */
 class Cdna4Tl final : public Tl {
   public:
@@ -1253,6 +1396,35 @@ void Cdna4TlTeam::get_scores(Team *team, ScoreMap &map)
 {
     (void)team;
     Cdna4TlTeam *self = this;
+    auto add_gated = [&](size_t lo, size_t hi, int score) {
+        ScoreRange r;
+        r.start    = lo;
+        r.end      = hi;
+        r.score    = score;
+        r.tl_name  = "cdna4";
+        r.alg_name = "gated_pipeline";
+        r.init     = [self](const ucc_coll_args_t &args, Team *t,
+                        Task **task) -> ucc_status_t {
+            if (!ec_hip::op_supported(args.dst.info.datatype, args.op)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            /* in-process multi-rank jigs can serialize spinning kernels
+             * on one HW queue: host-gated staged path is the safe
+             * fallback there (same reasoning as the fused kernel). */
+            int same_proc = 0;
+            for (auto &p : t->procs) {
+                same_proc += (p.pid == t->ctx->proc.pid);
+            }
+            if (same_proc > 1) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new GatedAllreduceTask(t->ctx, self, args);
+            return UCC_OK;
+        };
+        for (auto mt : {UCC_MEMORY_TYPE_CUDA, UCC_MEMORY_TYPE_CUDA_MANAGED}) {
+            map.add(UCC_COLL_TYPE_ALLREDUCE, mt, r);
+        }
+    };
     auto add = [&](ucc_coll_type_t ct, size_t lo, size_t hi, int score,
                    const char *alg, bool fused) {
         ScoreRange r;
@@ -1305,6 +1477,9 @@ void Cdna4TlTeam::get_scores(Team *team, ScoreMap &map)
         }
     };
     add(UCC_COLL_TYPE_ALLREDUCE, 0, cfg_.fused_max, 100, "fused", true);
+    if (Config::instance().get_bool("TL_CDNA4", "GATED", true)) {
+        add_gated(cfg_.fused_max + 1, SIZE_MAX, 90);
+    }
     add(UCC_COLL_TYPE_ALLREDUCE, 0, SIZE_MAX, 80, "staged_linear", false);
     add(UCC_COLL_TYPE_ALLGATHER, 0, SIZE_MAX, 80, "staged_linear", false);
     add(UCC_COLL_TYPE_ALLGATHERV, 0, SIZE_MAX, 80, "staged_linear", false);

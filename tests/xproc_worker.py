@@ -79,19 +79,25 @@ def main():
                                atol=2e-1)
     results.append("fused_xproc")
 
-    # 2. large staged cross-process allreduce (fp32)
+    # 2. large cross-process allreduce (fp32): the device-gated pipeline
+    # (zero host round-trips), repeated to exercise parity/counter
+    # continuity across collectives; persistent re-post included.
     count = 20_000_000
     full = torch.randn(world, count, generator=g0)
     src = full[rank].cuda()
     dst = torch.zeros(count, device="cuda")
-    expected = full.sum(0)
     r2 = c.coll_init(team, "allreduce", src=src.data_ptr(),
                      dst=dst.data_ptr(), count=count, dt=dtypes.FLOAT32,
-                     mem_type=dtypes.MEM_CUDA)
-    wait(r2, ctx)
-    torch.cuda.synchronize()
-    torch.testing.assert_close(dst.cpu(), expected, rtol=1e-5, atol=1e-4)
-    results.append("staged_xproc")
+                     mem_type=dtypes.MEM_CUDA, flags=c.FLAG_PERSISTENT)
+    for it in range(3):
+        src.copy_(full[rank] + it)
+        torch.cuda.synchronize()
+        expected = (full + it).sum(0)
+        wait(r2, ctx)
+        torch.cuda.synchronize()
+        torch.testing.assert_close(dst.cpu(), expected, rtol=1e-5,
+                                   atol=1e-4)
+    results.append("gated_xproc")
 
     # 3. cross-process alltoallv (skewed, fp16)
     scnt = [[(r + 1) * (d + 1) * 1024 for d in range(world)]
