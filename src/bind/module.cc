@@ -313,7 +313,7 @@ PYBIND11_MODULE(_core, m)
            int mem_type, uint64_t root, uint64_t flags,
            std::vector<uint64_t> src_counts, std::vector<uint64_t> src_displs,
            std::vector<uint64_t> dst_counts,
-           std::vector<uint64_t> dst_displs) {
+           std::vector<uint64_t> dst_displs, double timeout) {
             auto r  = std::make_shared<CoreReq>();
             r->team = team;
             ucc_coll_args_t a{};
@@ -327,6 +327,10 @@ PYBIND11_MODULE(_core, m)
             a.coll_type = ct;
             a.op        = (ucc_reduction_op_t)op;
             a.root      = root;
+            if (timeout > 0) {
+                a.flags |= UCC_COLL_ARGS_FLAG_TIMEOUT;
+                a.timeout = timeout;
+            }
             bool sv = ct == UCC_COLL_TYPE_ALLTOALLV ||
                       ct == UCC_COLL_TYPE_SCATTERV;
             bool dv = ct == UCC_COLL_TYPE_ALLTOALLV ||
@@ -371,8 +375,13 @@ PYBIND11_MODULE(_core, m)
         py::arg("src_counts") = std::vector<uint64_t>(),
         py::arg("src_displs") = std::vector<uint64_t>(),
         py::arg("dst_counts") = std::vector<uint64_t>(),
-        py::arg("dst_displs") = std::vector<uint64_t>());
+        py::arg("dst_displs") = std::vector<uint64_t>(),
+        py::arg("timeout") = 0.0);
 
+    m.def("score_map_str", [](std::shared_ptr<CoreTeam> t) {
+        auto *team = reinterpret_cast<ucc::Team *>(t->team);
+        return team->score_map.to_string();
+    });
     m.def("topo_sbgps", [](std::shared_ptr<CoreTeam> t) {
         auto *team = reinterpret_cast<ucc::Team *>(t->team);
         auto  node = ucc::topo::build_sbgp(team, ucc::topo::SbgpType::NODE);

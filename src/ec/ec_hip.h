@@ -121,7 +121,7 @@ bool op_supported(ucc_datatype_t dt, ucc_reduction_op_t op);
  *   kGatedCntBase + (phase * kGatedSlots + slot) * 2 + parity
  *   phase: 0=stage 1=reduce 2=gather; slot < kGatedSlots; parity = frag&1.
  */
-constexpr int kGatedBlocks  = 32;
+constexpr int kGatedBlocks  = 64;
 constexpr int kGatedSlots   = 8;
 constexpr int kGatedCntBase = 704; /* after graph counters (128..639) */
 

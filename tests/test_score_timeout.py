@@ -1,0 +1,86 @@
+"""Algorithm-selection (coll_score) and timeout behavior.
+Reference parity: test/gtest/coll_score/ (tuning-string parse, range
+override, fallback) and test/gtest/core/test_timeout.cc."""
+
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+TUNE_WORKER = r"""
+import sys
+import numpy as np
+sys.path.insert(0, %r)
+from ucc_amd import core
+from ucc_amd.testing import LocalJob
+
+job = LocalJob(2)
+c = core()
+smap = c.score_map_str(job.teams[0])
+assert "shm" in smap, smap
+# the tuning string must have zeroed shm's allreduce score and boosted tcp
+lines = [l for l in smap.splitlines() if l.startswith("allreduce:")]
+assert lines, smap
+for l in lines:
+    if "@shm/slotted" in l:
+        assert l.rstrip().endswith(":0"), l
+# collective still completes via fallback (tcp)
+arrs = [np.ones(1000, np.float32), np.full(1000, 2.0, np.float32)]
+outs = job.allreduce_np(arrs)
+for o in outs:
+    np.testing.assert_allclose(o, np.full(1000, 3.0, np.float32))
+print("TUNE_OK")
+""" % (REPO,)
+
+
+def test_tuning_string_override():
+    env = dict(os.environ)
+    env["UCC_TUNE"] = "allreduce:@slotted:0"
+    p = subprocess.run([sys.executable, "-c", TUNE_WORKER], env=env,
+                       capture_output=True, text=True, timeout=120)
+    sys.stdout.write(p.stdout[-2000:])
+    sys.stderr.write(p.stderr[-2000:])
+    assert p.returncode == 0 and "TUNE_OK" in p.stdout
+
+
+def test_score_map_contents():
+    from ucc_amd import core
+    from ucc_amd.testing import LocalJob
+
+    job = LocalJob(2)
+    smap = core().score_map_str(job.teams[0])
+    # all 16 coll types present for host memory
+    for coll in ("allreduce", "allgather", "allgatherv", "alltoall",
+                 "alltoallv", "barrier", "bcast", "fanin", "fanout",
+                 "gather", "gatherv", "reduce", "reduce_scatter",
+                 "reduce_scatterv", "scatter", "scatterv"):
+        assert any(line.startswith(coll + ":") for line in
+                   smap.splitlines()), f"{coll} missing:\n{smap}"
+
+
+def test_timeout():
+    """A collective that cannot complete (peer never posts) must fail
+    with UCC_ERR_TIMED_OUT once its timeout elapses."""
+    from ucc_amd import core, dtypes
+    from ucc_amd.testing import LocalJob
+
+    job = LocalJob(2)
+    c = core()
+    a = np.ones(4096, np.float32)
+    o = np.zeros(4096, np.float32)
+    req = c.coll_init(job.teams[0], "allreduce", src=a.ctypes.data,
+                      dst=o.ctypes.data, count=4096, dt=dtypes.FLOAT32,
+                      timeout=0.3)
+    req.post()
+    # rank 1 never posts; progress rank 0 only
+    st = c.INPROGRESS
+    for _ in range(2_000_000):
+        st = req.test()
+        if st != c.INPROGRESS:
+            break
+        job.ctxs[0].progress()
+    assert st < 0, f"expected timeout error, got {st}"
