@@ -39,8 +39,16 @@ SHLIB  := $(BUILD)/libucc_amd.so
 
 PERFTEST := build/ucc_perftest
 INFO     := build/ucc_info
+NTESTS   := build/test_generic_dt
 
-all: $(MODULE) $(SHLIB) $(PERFTEST) $(INFO)
+all: $(MODULE) $(SHLIB) $(PERFTEST) $(INFO) $(NTESTS)
+
+$(BUILD)/tests/%.o: tests/native/%.cc
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+build/test_generic_dt: $(BUILD)/tests/test_generic_dt.o $(LIB_OBJS)
+	$(HIPCC) $^ -o $@ -lrt -L/opt/rocm/lib -lrccl -Wl,-rpath,/opt/rocm/lib
 
 $(BUILD)/tools/%.o: tools/%.cc tools/shm_oob.h
 	@mkdir -p $(dir $@)

@@ -104,6 +104,47 @@ typedef enum ucc_datatype {
 
 size_t ucc_dt_size(ucc_datatype_t dt);
 
+/* ------------------------------------------------- generic datatypes
+ * User-defined datatypes (reference ucc.h:289-433 role): registered via
+ * ucc_dt_create_generic; the returned handle is usable wherever a
+ * predefined ucc_datatype_t is. Contiguous generic types (FLAG_CONTIG +
+ * contig_size) work with every data-movement collective; a generic
+ * reduce callback (FLAG_REDUCE) enables host-path reductions. */
+typedef struct ucc_dt_generic ucc_dt_generic_t;
+
+typedef enum ucc_generic_dt_ops_flags {
+    UCC_GENERIC_DT_OPS_FLAG_CONTIG = 1u << 0,
+    UCC_GENERIC_DT_OPS_FLAG_REDUCE = 1u << 1,
+} ucc_generic_dt_ops_flags_t;
+
+typedef struct ucc_generic_dt_ops {
+    uint64_t mask;
+    uint64_t flags;       /* ucc_generic_dt_ops_flags_t */
+    size_t   contig_size; /* element bytes, with FLAG_CONTIG */
+    void    *cookie;      /* opaque user pointer passed to callbacks */
+    struct {
+        void *(*start_pack)(void *cookie, const void *buffer,
+                            size_t count);
+        void *(*start_unpack)(void *cookie, void *buffer, size_t count);
+        size_t (*packed_size)(void *object);
+        ucc_status_t (*pack)(void *object, size_t offset, void *dest,
+                             size_t *length);
+        ucc_status_t (*unpack)(void *object, size_t offset,
+                               const void *src, size_t length);
+        void (*finish)(void *object);
+    } ops;
+    /* dst[i] = reduce(src1[i], src2[i]) for count elements */
+    ucc_status_t (*reduce)(const void *src1, const void *src2, void *dst,
+                           size_t count, void *cookie);
+} ucc_generic_dt_ops_t;
+
+ucc_status_t ucc_dt_create_generic(const ucc_generic_dt_ops_t *ops,
+                                   void *cookie, ucc_datatype_t *dt);
+void         ucc_dt_destroy(ucc_datatype_t dt);
+int          ucc_dt_is_predefined(ucc_datatype_t dt);
+/* NULL if dt is predefined */
+const ucc_generic_dt_ops_t *ucc_dt_generic_ops(ucc_datatype_t dt);
+
 typedef enum ucc_reduction_op {
     UCC_OP_SUM = 0,
     UCC_OP_PROD,

@@ -338,6 +338,26 @@ ucc_status_t reduce(void *dst, const void *const *srcs, int n_srcs,
     if (n_srcs < 1) {
         return UCC_ERR_INVALID_PARAM;
     }
+    if (!ucc_dt_is_predefined(dt)) {
+        /* generic datatype: pairwise user reduce callback */
+        const ucc_generic_dt_ops_t *g = ucc_dt_generic_ops(dt);
+        if (!g || !(g->flags & UCC_GENERIC_DT_OPS_FLAG_REDUCE) ||
+            !g->reduce || alpha != 1.0) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        size_t esz = ucc_dt_size(dt);
+        if (dst != srcs[0]) {
+            memcpy(dst, srcs[0], count * esz);
+        }
+        for (int s = 1; s < n_srcs; s++) {
+            ucc_status_t st =
+                g->reduce(dst, srcs[s], dst, count, g->cookie);
+            if (st != UCC_OK) {
+                return st;
+            }
+        }
+        return UCC_OK;
+    }
     switch (dt) {
     case UCC_DT_INT8:
         return reduce_typed<Id<int8_t>, int8_t, true>(dst, srcs, n_srcs,

@@ -27,8 +27,67 @@ size_t ucc_dt_size(ucc_datatype_t dt)
     case UCC_DT_FLOAT128:
     case UCC_DT_FLOAT64_COMPLEX: return 16;
     case UCC_DT_FLOAT128_COMPLEX: return 32;
-    default: return 0;
+    default: {
+        const ucc_generic_dt_ops_t *g = ucc_dt_generic_ops(dt);
+        if (g && (g->flags & UCC_GENERIC_DT_OPS_FLAG_CONTIG)) {
+            return g->contig_size;
+        }
+        return 0;
     }
+    }
+}
+
+/* ------------------------------------------------- generic datatypes */
+/* Registered in a process-global slab; the public handle is
+ * UCC_DT_USERDEFINED_BASE + index (predefined ids stay small ints). */
+#define UCC_DT_USERDEFINED_BASE 0x2000
+
+struct ucc_dt_generic {
+    ucc_generic_dt_ops_t ops;
+    void                *cookie;
+    int                  used;
+};
+
+static struct ucc_dt_generic g_generic_dts[64];
+
+ucc_status_t ucc_dt_create_generic(const ucc_generic_dt_ops_t *ops,
+                                   void *cookie, ucc_datatype_t *dt)
+{
+    if (!ops || !dt) {
+        return UCC_ERR_INVALID_PARAM;
+    }
+    for (int i = 0; i < 64; i++) {
+        if (!g_generic_dts[i].used) {
+            g_generic_dts[i].ops    = *ops;
+            g_generic_dts[i].cookie = cookie;
+            g_generic_dts[i].used   = 1;
+            *dt = (ucc_datatype_t)(UCC_DT_USERDEFINED_BASE + i);
+            return UCC_OK;
+        }
+    }
+    return UCC_ERR_NO_RESOURCE;
+}
+
+void ucc_dt_destroy(ucc_datatype_t dt)
+{
+    int i = (int)dt - UCC_DT_USERDEFINED_BASE;
+    if (i >= 0 && i < 64) {
+        g_generic_dts[i].used = 0;
+    }
+}
+
+int ucc_dt_is_predefined(ucc_datatype_t dt)
+{
+    return (int)dt < UCC_DT_PREDEFINED_LAST;
+}
+
+const ucc_generic_dt_ops_t *ucc_dt_generic_ops(ucc_datatype_t dt)
+{
+    int i = (int)dt - UCC_DT_USERDEFINED_BASE;
+    if (i >= 0 && i < 64 && g_generic_dts[i].used) {
+        return &g_generic_dts[i].ops;
+    }
+    return 0;
 }
 
 const char *ucc_status_string(ucc_status_t status)

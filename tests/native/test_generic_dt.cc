@@ -1,0 +1,186 @@
+/* Generic user-defined datatype unit test (reference gtest dt coverage):
+ * create/query/destroy, CPU reduction via the user callback, and a full
+ * 2-rank in-process allreduce through the public C API with a generic
+ * contiguous dtype. Exits 0 on success. */
+#include <cstdio>
+#include <cstring>
+#include <map>
+#include <vector>
+
+#include "../../src/api/ucc.h"
+#include "../../src/ec/ec_cpu.h"
+
+/* element: pair of floats reduced as (min, max) */
+struct MinMax {
+    float mn, mx;
+};
+
+static ucc_status_t mm_reduce(const void *s1, const void *s2, void *dst,
+                              size_t count, void *cookie)
+{
+    (void)cookie;
+    auto *a = (const MinMax *)s1;
+    auto *b = (const MinMax *)s2;
+    auto *d = (MinMax *)dst;
+    for (size_t i = 0; i < count; i++) {
+        d[i].mn = a[i].mn < b[i].mn ? a[i].mn : b[i].mn;
+        d[i].mx = a[i].mx > b[i].mx ? a[i].mx : b[i].mx;
+    }
+    return UCC_OK;
+}
+
+/* round-keyed in-process OOB (same design as tools/perftest.cc) */
+struct Loob {
+    struct Round {
+        std::vector<std::vector<uint8_t>> blobs;
+        int                               arrived = 0;
+    };
+    std::map<uint64_t, Round> rounds;
+    uint64_t                  next[2] = {0, 0};
+};
+static Loob g_loob;
+
+struct LoobReq {
+    uint64_t round;
+    void    *recv;
+    size_t   size;
+};
+
+static ucc_status_t lo_ag(void *src, void *recv, size_t size, void *info,
+                          void **req)
+{
+    int      rank  = (int)(intptr_t)info;
+    uint64_t round = g_loob.next[rank]++;
+    auto    &r     = g_loob.rounds[round];
+    if (r.blobs.empty()) {
+        r.blobs.resize(2);
+    }
+    r.blobs[rank].assign((uint8_t *)src, (uint8_t *)src + size);
+    r.arrived++;
+    *req = new LoobReq{round, recv, size};
+    return UCC_OK;
+}
+static ucc_status_t lo_test(void *req)
+{
+    auto *r  = (LoobReq *)req;
+    auto &rd = g_loob.rounds[r->round];
+    if (rd.arrived < 2) {
+        return UCC_INPROGRESS;
+    }
+    for (int i = 0; i < 2; i++) {
+        memcpy((uint8_t *)r->recv + i * r->size, rd.blobs[i].data(),
+               r->size);
+    }
+    return UCC_OK;
+}
+static ucc_status_t lo_free(void *req)
+{
+    delete (LoobReq *)req;
+    return UCC_OK;
+}
+
+#define CHECK(x)                                                             \
+    do {                                                                     \
+        if (!(x)) {                                                          \
+            fprintf(stderr, "FAILED: %s (line %d)\n", #x, __LINE__);         \
+            return 1;                                                        \
+        }                                                                    \
+    } while (0)
+
+int main()
+{
+    /* 1. create/query/destroy */
+    ucc_generic_dt_ops_t ops{};
+    ops.flags       = UCC_GENERIC_DT_OPS_FLAG_CONTIG |
+                      UCC_GENERIC_DT_OPS_FLAG_REDUCE;
+    ops.contig_size = sizeof(MinMax);
+    ops.reduce      = mm_reduce;
+    ucc_datatype_t gdt;
+    CHECK(ucc_dt_create_generic(&ops, nullptr, &gdt) == UCC_OK);
+    CHECK(!ucc_dt_is_predefined(gdt));
+    CHECK(ucc_dt_size(gdt) == sizeof(MinMax));
+    CHECK(ucc_dt_generic_ops(gdt) != nullptr);
+    CHECK(ucc_dt_is_predefined(UCC_DT_FLOAT32));
+
+    /* 2. CPU reduce via the callback */
+    MinMax a[4] = {{1, 1}, {5, 5}, {-3, -3}, {0, 0}};
+    MinMax b[4] = {{2, 2}, {4, 4}, {-1, -9}, {7, 7}};
+    MinMax out[4];
+    const void *srcs[2] = {a, b};
+    CHECK(ucc::ec_cpu::reduce(out, srcs, 2, 4, gdt, UCC_OP_SUM) == UCC_OK);
+    CHECK(out[0].mn == 1 && out[0].mx == 2);
+    CHECK(out[2].mn == -3 && out[2].mx == -3);
+
+    /* 3. 2-rank allreduce through the public API */
+    ucc_lib_h     libs[2];
+    ucc_context_h ctxs[2];
+    ucc_team_h    teams[2];
+    for (int r = 0; r < 2; r++) {
+        ucc_lib_params_t lp{};
+        CHECK(ucc_init(&lp, nullptr, &libs[r]) == UCC_OK);
+        ucc_context_params_t cp{};
+        CHECK(ucc_context_create(libs[r], &cp, nullptr, &ctxs[r]) ==
+              UCC_OK);
+        ucc_team_params_t tp{};
+        tp.mask          = UCC_TEAM_PARAM_FIELD_OOB;
+        tp.oob.allgather = lo_ag;
+        tp.oob.req_test  = lo_test;
+        tp.oob.req_free  = lo_free;
+        tp.oob.coll_info = (void *)(intptr_t)r;
+        tp.oob.n_oob_eps = 2;
+        tp.oob.oob_ep    = r;
+        CHECK(ucc_team_create_post(&ctxs[r], 1, &tp, &teams[r]) == UCC_OK);
+    }
+    while (true) {
+        ucc_status_t s0 = ucc_team_create_test(teams[0]);
+        ucc_status_t s1 = ucc_team_create_test(teams[1]);
+        CHECK(s0 >= 0 && s1 >= 0);
+        if (s0 == UCC_OK && s1 == UCC_OK) {
+            break;
+        }
+    }
+    const size_t        n = 1000;
+    std::vector<MinMax> src0(n), src1(n), dst0(n), dst1(n);
+    for (size_t i = 0; i < n; i++) {
+        src0[i] = {(float)i, (float)i};
+        src1[i] = {(float)(n - i), (float)(n - i)};
+    }
+    ucc_coll_req_h reqs[2];
+    std::vector<MinMax> *sb[2] = {&src0, &src1}, *db[2] = {&dst0, &dst1};
+    for (int r = 0; r < 2; r++) {
+        ucc_coll_args_t args{};
+        args.mask              = UCC_COLL_ARGS_FIELD_FLAGS;
+        args.coll_type         = UCC_COLL_TYPE_ALLREDUCE;
+        args.op                = UCC_OP_SUM; /* mapped to user callback */
+        args.src.info.buffer   = sb[r]->data();
+        args.src.info.count    = n;
+        args.src.info.datatype = gdt;
+        args.src.info.mem_type = UCC_MEMORY_TYPE_HOST;
+        args.dst.info          = args.src.info;
+        args.dst.info.buffer   = db[r]->data();
+        CHECK(ucc_collective_init(&args, &reqs[r], teams[r]) == UCC_OK);
+        CHECK(ucc_collective_post(reqs[r]) == UCC_OK);
+    }
+    while (ucc_collective_test(reqs[0]) == UCC_INPROGRESS ||
+           ucc_collective_test(reqs[1]) == UCC_INPROGRESS) {
+        ucc_context_progress(ctxs[0]);
+        ucc_context_progress(ctxs[1]);
+    }
+    CHECK(ucc_collective_test(reqs[0]) == UCC_OK);
+    CHECK(ucc_collective_test(reqs[1]) == UCC_OK);
+    for (size_t i = 0; i < n; i++) {
+        float mn = src0[i].mn < src1[i].mn ? src0[i].mn : src1[i].mn;
+        float mx = src0[i].mx > src1[i].mx ? src0[i].mx : src1[i].mx;
+        CHECK(dst0[i].mn == mn && dst0[i].mx == mx);
+        CHECK(dst1[i].mn == mn && dst1[i].mx == mx);
+    }
+    for (int r = 0; r < 2; r++) {
+        ucc_collective_finalize(reqs[r]);
+        ucc_team_destroy(teams[r]);
+        ucc_context_destroy(ctxs[r]);
+        ucc_finalize(libs[r]);
+    }
+    ucc_dt_destroy(gdt);
+    printf("GENERIC_DT_OK\n");
+    return 0;
+}

@@ -84,3 +84,16 @@ def test_timeout():
             break
         job.ctxs[0].progress()
     assert st < 0, f"expected timeout error, got {st}"
+
+
+def test_generic_datatype_native():
+    """Native unit test: generic user dtype create/size/reduce-cb +
+    2-rank allreduce through the C API (build/test_generic_dt)."""
+    binary = os.path.join(REPO, "build", "test_generic_dt")
+    if not os.path.exists(binary):
+        pytest.skip("native test binary not built (run make)")
+    p = subprocess.run([binary], capture_output=True, text=True,
+                       timeout=120)
+    sys.stdout.write(p.stdout[-500:])
+    sys.stderr.write(p.stderr[-500:])
+    assert p.returncode == 0 and "GENERIC_DT_OK" in p.stdout
