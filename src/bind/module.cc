@@ -18,6 +18,7 @@
 
 #include "../api/ucc.h"
 #include "../topo/topo.h"
+#include "../mc/mc.h"
 
 namespace py = pybind11;
 
@@ -377,6 +378,50 @@ PYBIND11_MODULE(_core, m)
         py::arg("dst_counts") = std::vector<uint64_t>(),
         py::arg("dst_displs") = std::vector<uint64_t>(),
         py::arg("timeout") = 0.0);
+
+    /* ------------------------------------------------------ mem_map */
+    m.def("mem_map_export", [](uintptr_t addr, size_t len) {
+        ucc_mem_map_t        seg{(void *)addr, len};
+        ucc_mem_map_params_t p{&seg, 1};
+        size_t               sz = 0;
+        ucc_mem_map_mem_h    h  = nullptr;
+        check(ucc_mem_map(nullptr, UCC_MEM_MAP_MODE_EXPORT, &p, &sz, &h),
+              "ucc_mem_map export");
+        py::bytes out((const char *)h, sz);
+        ucc_mem_unmap(&h);
+        return out;
+    });
+    m.def("mem_map_import", [](py::bytes blob) {
+        std::string s = blob;
+        /* keep the blob alive for unmap: heap copy owned by caller via
+         * the returned handle pair (ptr to mapped base) */
+        void  *copy = malloc(s.size());
+        memcpy(copy, s.data(), s.size());
+        size_t            sz = s.size();
+        ucc_mem_map_mem_h h  = copy;
+        check(ucc_mem_map(nullptr, UCC_MEM_MAP_MODE_IMPORT, nullptr, &sz,
+                          &h),
+              "ucc_mem_map import");
+        /* first segment's mapped VA (test helper) */
+        struct SegView {
+            uint32_t magic, n;
+            int32_t  imported, pad;
+            uint64_t addr, len, base_off;
+            int32_t  mt, has_ipc;
+            uint8_t  handle[64];
+            uint64_t mapped;
+        };
+        auto *v = (SegView *)copy;
+        return (uintptr_t)(v->mapped + v->base_off);
+    });
+    m.def("mem_map_close", [](uintptr_t mapped) {
+        ucc::mc::ipc_close((void *)mapped);
+    });
+    m.def("hip_memcpy_d2d", [](uintptr_t dst, uintptr_t src, size_t len) {
+        check(ucc::mc::copy((void *)dst, UCC_MEMORY_TYPE_CUDA,
+                            (const void *)src, UCC_MEMORY_TYPE_CUDA, len),
+              "d2d copy");
+    });
 
     m.def("score_map_str", [](std::shared_ptr<CoreTeam> t) {
         auto *team = reinterpret_cast<ucc::Team *>(t->team);

@@ -3,6 +3,7 @@
  * core/ucc_team.c (nonblocking create state machine), core/ucc_coll.c
  * (score-map dispatch, zero-size fast path, persistent re-post). */
 #include "core.h"
+#include "../mc/mc.h"
 #include "../utils/profile.h"
 
 #include <cstdlib>
@@ -652,17 +653,113 @@ ucc_status_t ucc_collective_triggered_post(ucc_ee_h ee, ucc_ev_t *ev)
     return UCC_OK;
 }
 
+/* ------------------------------------------------------------- mem_map */
+/* Export: produce a transferable handle blob for the given segments
+ * (device segments carry HIP-IPC handles; host segments are recorded by
+ * address for same-node shm use). Import: open the blob received from a
+ * peer, mapping device segments into this process.
+ * Reference parity: ucc.h mem_map section + tl/cuda IPC mapping cache
+ * (tl_cuda_cache.c role, collapsed: one open per imported segment). */
+namespace {
+struct MemMapSeg {
+    uint64_t addr;   /* exporter VA            */
+    uint64_t len;
+    uint64_t base_off; /* addr - allocation base */
+    int32_t  mt;
+    int32_t  has_ipc;
+    uint8_t  handle[ucc::mc::kIpcHandleBytes];
+    uint64_t mapped; /* importer-local VA (base) — set on import */
+};
+struct MemMapBlob {
+    uint32_t  magic; /* 0x4d4d4150 "MMAP" */
+    uint32_t  n;
+    int32_t   imported;
+    int32_t   pad;
+    MemMapSeg segs[];
+};
+constexpr uint32_t kMemMapMagic = 0x4d4d4150;
+} // namespace
+
 ucc_status_t ucc_mem_map(ucc_context_h context, ucc_mem_map_flags_t flags,
                          ucc_mem_map_params_t *params, size_t *memh_size,
                          ucc_mem_map_mem_h *memh)
 {
-    (void)context; (void)flags; (void)params; (void)memh_size; (void)memh;
-    return UCC_ERR_NOT_IMPLEMENTED;
+    (void)context;
+    if (!memh || !memh_size) {
+        return UCC_ERR_INVALID_PARAM;
+    }
+    if (flags & UCC_MEM_MAP_MODE_EXPORT) {
+        if (!params) {
+            return UCC_ERR_INVALID_PARAM;
+        }
+        size_t n  = params->n_segments;
+        size_t sz = sizeof(MemMapBlob) + n * sizeof(MemMapSeg);
+        auto  *b  = (MemMapBlob *)calloc(1, sz);
+        b->magic  = kMemMapMagic;
+        b->n      = (uint32_t)n;
+        for (size_t i = 0; i < n; i++) {
+            MemMapSeg &s = b->segs[i];
+            s.addr = (uint64_t)(uintptr_t)params->segments[i].address;
+            s.len  = params->segments[i].len;
+            ucc_memory_type_t mt = UCC_MEMORY_TYPE_HOST;
+            ucc::mc::mem_query((const void *)(uintptr_t)s.addr, &mt);
+            s.mt = (int32_t)mt;
+            if (ucc::mc::is_device_mt(mt)) {
+                size_t off = 0;
+                if (ucc::mc::ipc_export((const void *)(uintptr_t)s.addr,
+                                   s.handle, &off) == UCC_OK) {
+                    s.has_ipc  = 1;
+                    s.base_off = off;
+                }
+            }
+        }
+        *memh      = b;
+        *memh_size = sz;
+        return UCC_OK;
+    }
+    if (flags & UCC_MEM_MAP_MODE_IMPORT) {
+        auto *b = (MemMapBlob *)*memh;
+        if (!b || b->magic != kMemMapMagic) {
+            return UCC_ERR_INVALID_PARAM;
+        }
+        for (uint32_t i = 0; i < b->n; i++) {
+            MemMapSeg &s = b->segs[i];
+            if (s.has_ipc) {
+                void *m = nullptr;
+                ucc_status_t st = ucc::mc::ipc_import(s.handle, &m);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                s.mapped = (uint64_t)(uintptr_t)m;
+            }
+        }
+        b->imported = 1;
+        *memh_size  = sizeof(MemMapBlob) + b->n * sizeof(MemMapSeg);
+        return UCC_OK;
+    }
+    return UCC_ERR_INVALID_PARAM;
 }
+
 ucc_status_t ucc_mem_unmap(ucc_mem_map_mem_h *memh)
 {
-    (void)memh;
-    return UCC_ERR_NOT_IMPLEMENTED;
+    if (!memh || !*memh) {
+        return UCC_ERR_INVALID_PARAM;
+    }
+    auto *b = (MemMapBlob *)*memh;
+    if (b->magic != kMemMapMagic) {
+        return UCC_ERR_INVALID_PARAM;
+    }
+    if (b->imported) {
+        for (uint32_t i = 0; i < b->n; i++) {
+            if (b->segs[i].has_ipc && b->segs[i].mapped) {
+                ucc::mc::ipc_close((void *)(uintptr_t)b->segs[i].mapped);
+            }
+        }
+    } else {
+        free(b);
+    }
+    *memh = nullptr;
+    return UCC_OK;
 }
 
 } /* extern "C" */

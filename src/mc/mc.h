@@ -25,6 +25,14 @@ ucc_status_t copy(void *dst, ucc_memory_type_t dst_mt, const void *src,
 ucc_status_t memset_(void *ptr, ucc_memory_type_t mt, int value,
                      size_t bytes);
 
+/* HIP-IPC handle export/import for ucc_mem_map (device memory). The
+ * handle blob is hipIpcMemHandle_t-sized (64B). */
+constexpr size_t kIpcHandleBytes = 64;
+ucc_status_t ipc_export(const void *ptr, void *handle_out,
+                        size_t *base_off_out);
+ucc_status_t ipc_import(const void *handle, void **mapped);
+ucc_status_t ipc_close(void *mapped);
+
 static inline bool is_device_mt(ucc_memory_type_t mt)
 {
     return mt == UCC_MEMORY_TYPE_CUDA || mt == UCC_MEMORY_TYPE_ROCM ||

@@ -89,5 +89,40 @@ ucc_status_t memset_(void *ptr, ucc_memory_type_t mt, int value, size_t bytes)
                                                       : UCC_ERR_NO_RESOURCE;
 }
 
+ucc_status_t ipc_export(const void *ptr, void *handle_out,
+                        size_t *base_off_out)
+{
+    static_assert(sizeof(hipIpcMemHandle_t) <= kIpcHandleBytes,
+                  "handle blob too small");
+    hipDeviceptr_t base  = nullptr;
+    size_t         bsize = 0;
+    if (hipMemGetAddressRange(&base, &bsize, (hipDeviceptr_t)ptr) !=
+        hipSuccess) {
+        return UCC_ERR_INVALID_PARAM;
+    }
+    hipIpcMemHandle_t h;
+    if (hipIpcGetMemHandle(&h, (void *)base) != hipSuccess) {
+        return UCC_ERR_NO_RESOURCE;
+    }
+    memcpy(handle_out, &h, sizeof(h));
+    *base_off_out = (size_t)((uintptr_t)ptr - (uintptr_t)base);
+    return UCC_OK;
+}
+
+ucc_status_t ipc_import(const void *handle, void **mapped)
+{
+    hipIpcMemHandle_t h;
+    memcpy(&h, handle, sizeof(h));
+    hipError_t e =
+        hipIpcOpenMemHandle(mapped, h, hipIpcMemLazyEnablePeerAccess);
+    return e == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
+}
+
+ucc_status_t ipc_close(void *mapped)
+{
+    return hipIpcCloseMemHandle(mapped) == hipSuccess ? UCC_OK
+                                                      : UCC_ERR_INVALID_PARAM;
+}
+
 } // namespace mc
 } // namespace ucc

@@ -128,6 +128,28 @@ def main():
         torch.testing.assert_close(got, exp)
     results.append("alltoallv_xproc")
 
+    # 3b. ucc_mem_map export/import across processes: rank 0 exports a
+    # device buffer, rank 1 imports and reads it over IPC.
+    if world == 2:
+        if rank == 0:
+            payload = torch.arange(1024, dtype=torch.float32, device="cuda")
+            blob = c.mem_map_export(payload.data_ptr(), payload.numel() * 4)
+            dist.broadcast_object_list([bytes(blob)], src=0)
+            torch.cuda.synchronize()
+            dist.barrier()
+        else:
+            holder = [None]
+            dist.broadcast_object_list(holder, src=0)
+            mapped = c.mem_map_import(holder[0])
+            got = torch.empty(1024, dtype=torch.float32, device="cuda")
+            c.hip_memcpy_d2d(got.data_ptr(), mapped, 1024 * 4)
+            torch.cuda.synchronize()
+            torch.testing.assert_close(
+                got.cpu(), torch.arange(1024, dtype=torch.float32))
+            c.mem_map_close(mapped)
+            dist.barrier()
+        results.append("mem_map")
+
     if only_basic:
         dist.barrier()
         print(f"XPROC_OK rank={rank} {'+'.join(results)}", flush=True)
