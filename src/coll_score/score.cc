@@ -5,6 +5,8 @@
 #include "../core/core.h"
 
 #include <algorithm>
+#include <cstdlib>
+#include <cstring>
 #include <sstream>
 
 namespace ucc {
@@ -87,7 +89,13 @@ ucc_status_t ScoreMap::init_coll(const ucc_coll_args_t &args, Team *team,
         }
         ucc_status_t st = r.init(args, team, task);
         if (st == UCC_OK) {
-            UCC_LOG(LogLevel::DEBUG, "score",
+            /* UCC_COLL_TRACE=1: per-collective selection print at INFO
+             * (reference UCC_COLL_TRACE, core/ucc_coll.c:329-345) */
+            static const bool trace = [] {
+                const char *e = getenv("UCC_COLL_TRACE");
+                return e && *e && strcmp(e, "0") != 0;
+            }();
+            UCC_LOG(trace ? LogLevel::INFO : LogLevel::DEBUG, "score",
                     "coll %s mem %s size %zu -> %s/%s (score %d)",
                     coll_type_name(args.coll_type), mem_type_name(mt), msgsize,
                     r.tl_name.c_str(), r.alg_name.c_str(), r.score);

@@ -222,6 +222,8 @@ ucc_status_t ucc_context_create(ucc_lib_h lib_h,
         ctx->has_oob = params->mask & UCC_CONTEXT_PARAM_FIELD_OOB;
     }
     ctx->mt   = lib->thread_mode == UCC_THREAD_MULTIPLE;
+    ctx->lock_free =
+        Config::instance().get_bool("", "LOCK_FREE_PROGRESS_Q", true);
     ctx->seq  = lib->next_ctx_seq++;
     ctx->proc = local_proc_info();
     ctx->proc.ctx_seq = ((uint64_t)ctx->proc.pid << 20) | ctx->seq;

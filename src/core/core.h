@@ -26,6 +26,7 @@
 
 #include "../api/ucc.h"
 #include "../utils/config.h"
+#include "../utils/lf_queue.h"
 #include "../utils/log.h"
 
 namespace ucc {
@@ -228,8 +229,10 @@ struct Context {
     uint64_t             seq     = 0; /* unique id of this ctx in process  */
     ProcInfo             proc;
     std::deque<Task *>   pq;
-    std::recursive_mutex pq_mtx; /* used only in THREAD_MULTIPLE           */
+    std::recursive_mutex pq_mtx; /* THREAD_MULTIPLE locked fallback        */
+    LfQueue<Task *>      lf_pq;  /* THREAD_MULTIPLE lock-free fast path    */
     bool                 mt = false;
+    bool                 lock_free = true; /* UCC_LOCK_FREE_PROGRESS_Q     */
     std::vector<std::unique_ptr<TlContext>> tl_ctxs;
     uint64_t             next_team_id = 1;
     uint32_t             n_progress_calls = 0;

@@ -19,6 +19,16 @@ void Context::pq_push(Task *t)
     if (t->in_pq) {
         return;
     }
+    if (mt && lock_free) {
+        t->in_pq = true;
+        if (lf_pq.push(t)) {
+            return;
+        }
+        /* ring full: overflow to the locked deque */
+        std::lock_guard<std::recursive_mutex> lk(pq_mtx);
+        pq.push_back(t);
+        return;
+    }
     t->in_pq = true;
     if (mt) {
         std::lock_guard<std::recursive_mutex> lk(pq_mtx);
@@ -109,7 +119,15 @@ ucc_status_t Context::progress()
 {
     n_progress_calls++;
     Task *t = nullptr;
-    {
+    if (mt && lock_free) {
+        if (!lf_pq.pop(&t)) {
+            std::lock_guard<std::recursive_mutex> lk(pq_mtx);
+            if (!pq.empty()) {
+                t = pq.front();
+                pq.pop_front();
+            }
+        }
+    } else {
         if (mt) {
             pq_mtx.lock();
         }

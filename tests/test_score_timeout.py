@@ -97,3 +97,61 @@ def test_generic_datatype_native():
     sys.stdout.write(p.stdout[-500:])
     sys.stderr.write(p.stderr[-500:])
     assert p.returncode == 0 and "GENERIC_DT_OK" in p.stdout
+
+
+def test_thread_multiple_progress():
+    """THREAD_MULTIPLE: a second thread pumping ucc_context_progress
+    concurrently with the posting thread (lock-free progress queue path,
+    reference MT progress-queue coverage)."""
+    import threading
+
+    import numpy as np
+
+    from ucc_amd import core, dtypes
+
+    c = core()
+    n = 2
+    libs = [c.Lib(thread_mode="multiple") for _ in range(n)]
+    ctxs = [c.Context(lib) for lib in libs]
+    oob = c.LocalOob(n)
+    teams = [c.team_create_post(ctxs[r], local_oob=oob, rank=r)
+             for r in range(n)]
+    while True:
+        sts = [c.team_create_test(t) for t in teams]
+        assert all(s >= 0 for s in sts)
+        if all(s == c.OK for s in sts):
+            break
+
+    stop = threading.Event()
+
+    def pump(ctx):
+        while not stop.is_set():
+            ctx.progress()
+
+    threads = [threading.Thread(target=pump, args=(ctx,)) for ctx in ctxs]
+    for t in threads:
+        t.start()
+    try:
+        for it in range(50):
+            arrs = [np.full(2048, float(r + it), np.float32)
+                    for r in range(n)]
+            outs = [np.zeros(2048, np.float32) for _ in range(n)]
+            reqs = [c.coll_init(teams[r], "allreduce",
+                                src=arrs[r].ctypes.data,
+                                dst=outs[r].ctypes.data, count=2048,
+                                dt=dtypes.FLOAT32) for r in range(n)]
+            for r in reqs:
+                r.post()
+            import time
+            deadline = time.time() + 30
+            while any(r.test() == c.INPROGRESS for r in reqs):
+                assert time.time() < deadline, "MT collective stuck"
+                for ctx in ctxs:
+                    ctx.progress()
+            expected = sum(float(r + it) for r in range(n))
+            for o in outs:
+                np.testing.assert_allclose(o, expected)
+    finally:
+        stop.set()
+        for t in threads:
+            t.join(timeout=10)
