@@ -144,11 +144,23 @@ struct GatedArgs {
     ucc_datatype_t     dt;
     ucc_reduction_op_t op;
     float       alpha;
-    /* cumulative block-count targets (host-tracked):
-     *   stage waits  reduce_cnt >= t_prev_reduce  (parity reuse)
-     *   reduce waits stage_cnt  >= t_stage  AND gather_cnt >= t_prev_gather
-     *   gather waits reduce_cnt >= t_reduce                             */
-    uint64_t    t_prev_reduce, t_prev_gather, t_stage, t_reduce;
+    /* cumulative block-count targets (host-tracked). Counters are shared
+     * by every gated collective type on a (slot,parity), so area-reuse
+     * waits cover BOTH possible consumer phases:
+     *   stage  waits reduce_cnt >= t_sw_reduce AND
+     *                gather_cnt >= t_sw_gather   (in-area reuse)
+     *   reduce waits stage_cnt  >= t_stage AND
+     *                gather_cnt >= t_prev_gather (out-area reuse)
+     *   gather waits phase gw_phase >= t_gather_wait
+     *                (1=reduce for allreduce, 0=stage for ag/alltoall) */
+    uint64_t    t_sw_reduce, t_sw_gather, t_prev_gather, t_stage,
+                t_gather_wait;
+    int         gw_phase;
+    /* per-dest cell staging (alltoall): my_in[c_dst_off[k]] <-
+     * src[c_src_off[k]], c_len[k] bytes; 0 = contiguous stage of len   */
+    int         n_cells;
+    uint64_t    c_src_off[kMaxRanks], c_dst_off[kMaxRanks],
+                c_len[kMaxRanks];
 };
 
 ucc_status_t staged_stage(const GatedArgs &a, hipStream_t s);

@@ -685,10 +685,21 @@ __device__ __forceinline__ void d_copy_bytes(uint8_t *dst,
 
 __global__ void k_staged_stage(GatedArgs a)
 {
-    if (!gated_wait(a, 1, a.t_prev_reduce)) { /* parity-area reuse */
+    if (!gated_wait(a, 1, a.t_sw_reduce) ||
+        !gated_wait(a, 2, a.t_sw_gather)) { /* in-area reuse */
         return;
     }
-    d_copy_bytes((uint8_t *)a.my_in, (const uint8_t *)a.src, a.len);
+    if (a.n_cells > 0) { /* per-dest cells (alltoall) */
+        for (int c = 0; c < a.n_cells; c++) {
+            if (a.c_len[c]) {
+                d_copy_bytes((uint8_t *)a.my_in + a.c_dst_off[c],
+                             (const uint8_t *)a.src + a.c_src_off[c],
+                             a.c_len[c]);
+            }
+        }
+    } else if (a.len) {
+        d_copy_bytes((uint8_t *)a.my_in, (const uint8_t *)a.src, a.len);
+    }
     gated_signal(a, 0);
 }
 
@@ -744,7 +755,7 @@ __global__ void k_staged_reduce(GatedArgs a)
 
 __global__ void k_staged_gather(GatedArgs a)
 {
-    if (!gated_wait(a, 1, a.t_reduce)) {
+    if (!gated_wait(a, a.gw_phase, a.t_gather_wait)) {
         return;
     }
     for (int r = 0; r < a.nranks; r++) {

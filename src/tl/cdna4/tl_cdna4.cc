@@ -1208,7 +1208,7 @@ class StagedTask final : public Cdna4Task {
  * round-trips inside the collective, one trailing event. Same xGMI
  * traffic as the host-gated staged path (1.75 x S per rank), minus all
  * host gating latency. */
-class GatedAllreduceTask final : public Cdna4Task {
+class GatedCollTask final : public Cdna4Task {
   public:
     using Cdna4Task::Cdna4Task;
 
@@ -1219,18 +1219,69 @@ class GatedAllreduceTask final : public Cdna4Task {
             return UCC_ERR_NOT_SUPPORTED;
         }
         const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
-        dt_    = a_.dst.info.datatype;
-        dtsz_  = ucc_dt_size(dt_);
-        op_    = a_.op;
-        alpha_ = 1.0f;
+        ct_     = a_.coll_type;
+        op_     = a_.op;
+        alpha_  = 1.0f;
+        gran_   = tt_->cfg_.chunk;
+        cell_   = (tt_->cfg_.chunk / n_) & ~(size_t)255;
         if (op_ == UCC_OP_AVG) {
             op_    = (ucc_reduction_op_t)12;
             alpha_ = 1.0f / (float)n_;
         }
-        total_  = a_.dst.info.count * dtsz_;
-        dbuf_   = (uint8_t *)a_.dst.info.buffer;
-        sbuf_   = inplace ? dbuf_ : (const uint8_t *)a_.src.info.buffer;
-        nfrags_ = (total_ + tt_->cfg_.chunk - 1) / tt_->cfg_.chunk;
+        switch (ct_) {
+        case UCC_COLL_TYPE_ALLREDUCE:
+            dt_    = a_.dst.info.datatype;
+            dtsz_  = ucc_dt_size(dt_);
+            total_ = a_.dst.info.count * dtsz_;
+            dbuf_  = (uint8_t *)a_.dst.info.buffer;
+            sbuf_  = inplace ? dbuf_ : (const uint8_t *)a_.src.info.buffer;
+            break;
+        case UCC_COLL_TYPE_REDUCE_SCATTER: {
+            dt_   = a_.dst.info.datatype;
+            dtsz_ = ucc_dt_size(dt_);
+            size_t out_b;
+            if (inplace) {
+                total_ = a_.dst.info.count * dtsz_;
+                out_b  = total_ / n_;
+                sbuf_  = (uint8_t *)a_.dst.info.buffer;
+                dbuf_  = (uint8_t *)a_.dst.info.buffer + me_ * out_b;
+            } else {
+                out_b  = a_.dst.info.count * dtsz_;
+                total_ = out_b * n_;
+                sbuf_  = (const uint8_t *)a_.src.info.buffer;
+                dbuf_  = (uint8_t *)a_.dst.info.buffer;
+            }
+            out_b_ = out_b;
+            break;
+        }
+        case UCC_COLL_TYPE_ALLGATHER: {
+            dt_    = a_.dst.info.datatype;
+            dtsz_  = ucc_dt_size(dt_);
+            size_t block = a_.dst.info.count * dtsz_ / n_;
+            out_b_ = block;
+            dbuf_  = (uint8_t *)a_.dst.info.buffer;
+            sbuf_  = inplace ? dbuf_ + me_ * block
+                             : (const uint8_t *)a_.src.info.buffer;
+            total_ = block; /* fragment over the per-rank block */
+            break;
+        }
+        case UCC_COLL_TYPE_ALLTOALL: {
+            if (inplace || cell_ == 0) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            dt_    = a_.dst.info.datatype;
+            dtsz_  = ucc_dt_size(dt_);
+            out_b_ = a_.dst.info.count * dtsz_ / n_; /* per-peer block */
+            sbuf_  = (const uint8_t *)a_.src.info.buffer;
+            dbuf_  = (uint8_t *)a_.dst.info.buffer;
+            total_ = out_b_;
+            gran_  = cell_;
+            break;
+        }
+        default:
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        nfrags_ = (total_ + gran_ - 1) / gran_;
         if (nfrags_ == 0) {
             nfrags_ = 1;
         }
@@ -1260,7 +1311,8 @@ class GatedAllreduceTask final : public Cdna4Task {
             return UCC_ERR_NO_RESOURCE;
         }
         if (*tt_->err_host_ != 0) {
-            ucc_error("gated allreduce timed out waiting for peers");
+            ucc_error("gated %s timed out waiting for peers",
+                      coll_type_name(ct_));
             close_slot();
             return UCC_ERR_TIMED_OUT;
         }
@@ -1271,15 +1323,14 @@ class GatedAllreduceTask final : public Cdna4Task {
   private:
     ucc_status_t enqueue_all()
     {
-        const size_t chunk = tt_->cfg_.chunk;
+        auto &L          = tt_->gated_launch_;
+        const uint64_t B = (uint64_t)ec_hip::kGatedBlocks;
         for (size_t f = 0; f < nfrags_; f++) {
             const uint32_t p   = (uint32_t)(f & 1);
-            const size_t   off = f * chunk;
+            const size_t   off = f * gran_;
             const size_t   len =
-                total_ - off < chunk ? total_ - off : chunk;
+                total_ - off < gran_ ? total_ - off : gran_;
             ec_hip::GatedArgs ga{};
-            ga.src    = sbuf_ + off;
-            ga.dst    = dbuf_ + off;
             ga.my_in  = tt_->area(me_, slot_, p, 0);
             ga.my_out = tt_->area(me_, slot_, p, 1);
             for (uint32_t r = 0; r < n_; r++) {
@@ -1290,50 +1341,132 @@ class GatedAllreduceTask final : public Cdna4Task {
             ga.local_flags = tt_->flags_;
             ga.error_word  = tt_->err_host_;
             ga.len         = len;
-            /* byte slices, 256-aligned, tail to last rank */
-            size_t per = (len / n_) & ~(size_t)255;
-            for (uint32_t r = 0; r < n_; r++) {
-                ga.slice_b[r] = (uint64_t)r * per;
-                ga.slice_e[r] = r == n_ - 1 ? len : (uint64_t)(r + 1) * per;
+            ga.rank        = (int)me_;
+            ga.nranks      = (int)n_;
+            ga.slot        = (int)slot_;
+            ga.parity      = (int)p;
+            ga.dt          = dt_;
+            ga.op          = op_;
+            ga.alpha       = alpha_;
+            ucc_status_t st = UCC_OK;
+            switch (ct_) {
+            case UCC_COLL_TYPE_ALLREDUCE: {
+                /* stage frag -> reduce my slice -> gather all slices */
+                ga.src = sbuf_ + off;
+                ga.dst = dbuf_ + off;
+                size_t per = (len / n_) & ~(size_t)255;
+                for (uint32_t r = 0; r < n_; r++) {
+                    ga.slice_b[r] = (uint64_t)r * per;
+                    ga.slice_e[r] =
+                        r == n_ - 1 ? len : (uint64_t)(r + 1) * per;
+                }
+                ga.sl_b          = ga.slice_b[me_];
+                ga.sl_e          = ga.slice_e[me_];
+                ga.t_sw_reduce   = L[1][slot_][p] * B;
+                ga.t_sw_gather   = L[2][slot_][p] * B;
+                ga.t_prev_gather = L[2][slot_][p] * B;
+                ga.t_stage       = (L[0][slot_][p] + 1) * B;
+                ga.gw_phase      = 1;
+                ga.t_gather_wait = (L[1][slot_][p] + 1) * B;
+                st = ec_hip::staged_stage(ga, copy_s());
+                L[0][slot_][p]++;
+                if (st == UCC_OK) {
+                    st = ec_hip::staged_reduce(ga, comp());
+                    L[1][slot_][p]++;
+                }
+                if (st == UCC_OK) {
+                    st = ec_hip::staged_gather(ga, comp());
+                    L[2][slot_][p]++;
+                }
+                break;
             }
-            ga.sl_b   = ga.slice_b[me_];
-            ga.sl_e   = ga.slice_e[me_];
-            ga.rank   = (int)me_;
-            ga.nranks = (int)n_;
-            ga.slot   = (int)slot_;
-            ga.parity = (int)p;
-            ga.dt     = dt_;
-            ga.op     = op_;
-            ga.alpha  = alpha_;
-            auto &L   = tt_->gated_launch_;
-            const uint64_t B = (uint64_t)ec_hip::kGatedBlocks;
-            ga.t_prev_reduce = L[1][slot_][p] * B;
-            ga.t_prev_gather = L[2][slot_][p] * B;
-            ga.t_stage       = (L[0][slot_][p] + 1) * B;
-            ga.t_reduce      = (L[1][slot_][p] + 1) * B;
-            ucc_status_t st = ec_hip::staged_stage(ga, copy_s());
+            case UCC_COLL_TYPE_REDUCE_SCATTER: {
+                /* stage frag of the whole source; reduce ONLY the
+                 * intersection with my output slice, directly into my
+                 * user dst (no gather phase) */
+                ga.src = sbuf_ + off;
+                size_t s0 = (size_t)me_ * out_b_, s1 = s0 + out_b_;
+                size_t b = s0 > off ? s0 : off;
+                size_t e = s1 < off + len ? s1 : off + len;
+                ga.sl_b   = b > e ? 0 : b - off;
+                ga.sl_e   = b > e ? 0 : e - off;
+                ga.my_out = dbuf_ + (b >= s0 && b <= s1 ? b - s0 : 0);
+                ga.t_sw_reduce   = L[1][slot_][p] * B;
+                ga.t_sw_gather   = L[2][slot_][p] * B;
+                ga.t_prev_gather = 0;
+                ga.t_stage       = (L[0][slot_][p] + 1) * B;
+                st = ec_hip::staged_stage(ga, copy_s());
+                L[0][slot_][p]++;
+                if (st == UCC_OK) {
+                    st = ec_hip::staged_reduce(ga, comp());
+                    L[1][slot_][p]++;
+                }
+                break;
+            }
+            case UCC_COLL_TYPE_ALLGATHER: {
+                /* stage my block frag; gather reads peers' STAGED data
+                 * (in areas) into dst block positions */
+                ga.src = sbuf_ + off;
+                ga.dst = dbuf_;
+                for (uint32_t r = 0; r < n_; r++) {
+                    ga.peer_out[r] = tt_->area(r, slot_, p, 0);
+                    ga.slice_b[r]  = (uint64_t)r * out_b_ + off;
+                    ga.slice_e[r]  = ga.slice_b[r] + len;
+                }
+                ga.t_sw_reduce   = L[1][slot_][p] * B;
+                ga.t_sw_gather   = L[2][slot_][p] * B;
+                ga.gw_phase      = 0;
+                ga.t_gather_wait = (L[0][slot_][p] + 1) * B;
+                st = ec_hip::staged_stage(ga, copy_s());
+                L[0][slot_][p]++;
+                if (st == UCC_OK) {
+                    st = ec_hip::staged_gather(ga, comp());
+                    L[2][slot_][p]++;
+                }
+                break;
+            }
+            case UCC_COLL_TYPE_ALLTOALL: {
+                /* stage per-dest cells; gather my cell from each peer */
+                ga.src     = sbuf_;
+                ga.dst     = dbuf_;
+                ga.n_cells = (int)n_;
+                for (uint32_t r = 0; r < n_; r++) {
+                    ga.c_src_off[r] = (uint64_t)r * out_b_ + off;
+                    ga.c_dst_off[r] = (uint64_t)r * cell_;
+                    ga.c_len[r]     = len;
+                    ga.peer_out[r] =
+                        tt_->area(r, slot_, p, 0) + me_ * cell_;
+                    ga.slice_b[r] = (uint64_t)r * out_b_ + off;
+                    ga.slice_e[r] = ga.slice_b[r] + len;
+                }
+                ga.t_sw_reduce   = L[1][slot_][p] * B;
+                ga.t_sw_gather   = L[2][slot_][p] * B;
+                ga.gw_phase      = 0;
+                ga.t_gather_wait = (L[0][slot_][p] + 1) * B;
+                st = ec_hip::staged_stage(ga, copy_s());
+                L[0][slot_][p]++;
+                if (st == UCC_OK) {
+                    st = ec_hip::staged_gather(ga, comp());
+                    L[2][slot_][p]++;
+                }
+                break;
+            }
+            default:
+                return UCC_ERR_NOT_SUPPORTED;
+            }
             if (st != UCC_OK) {
                 return st;
             }
-            L[0][slot_][p]++;
-            st = ec_hip::staged_reduce(ga, comp());
-            if (st != UCC_OK) {
-                return st;
-            }
-            L[1][slot_][p]++;
-            st = ec_hip::staged_gather(ga, comp());
-            if (st != UCC_OK) {
-                return st;
-            }
-            L[2][slot_][p]++;
         }
         return UCC_OK;
     }
 
+    ucc_coll_type_t    ct_    = UCC_COLL_TYPE_ALLREDUCE;
     ucc_datatype_t     dt_    = UCC_DT_FLOAT32;
     ucc_reduction_op_t op_    = UCC_OP_SUM;
     float              alpha_ = 1.0f;
     size_t             dtsz_ = 4, total_ = 0, nfrags_ = 0;
+    size_t             gran_ = 0, cell_ = 0, out_b_ = 0;
     const uint8_t     *sbuf_ = nullptr;
     uint8_t           *dbuf_ = nullptr;
 };
@@ -1396,7 +1529,8 @@ void Cdna4TlTeam::get_scores(Team *team, ScoreMap &map)
 {
     (void)team;
     Cdna4TlTeam *self = this;
-    auto add_gated = [&](size_t lo, size_t hi, int score) {
+    auto add_gated = [&](ucc_coll_type_t ct, size_t lo, size_t hi,
+                         int score) {
         ScoreRange r;
         r.start    = lo;
         r.end      = hi;
@@ -1405,7 +1539,9 @@ void Cdna4TlTeam::get_scores(Team *team, ScoreMap &map)
         r.alg_name = "gated_pipeline";
         r.init     = [self](const ucc_coll_args_t &args, Team *t,
                         Task **task) -> ucc_status_t {
-            if (!ec_hip::op_supported(args.dst.info.datatype, args.op)) {
+            if ((args.coll_type == UCC_COLL_TYPE_ALLREDUCE ||
+                 args.coll_type == UCC_COLL_TYPE_REDUCE_SCATTER) &&
+                !ec_hip::op_supported(args.dst.info.datatype, args.op)) {
                 return UCC_ERR_NOT_SUPPORTED;
             }
             /* in-process multi-rank jigs can serialize spinning kernels
@@ -1418,11 +1554,11 @@ void Cdna4TlTeam::get_scores(Team *team, ScoreMap &map)
             if (same_proc > 1) {
                 return UCC_ERR_NOT_SUPPORTED;
             }
-            *task = new GatedAllreduceTask(t->ctx, self, args);
+            *task = new GatedCollTask(t->ctx, self, args);
             return UCC_OK;
         };
         for (auto mt : {UCC_MEMORY_TYPE_CUDA, UCC_MEMORY_TYPE_CUDA_MANAGED}) {
-            map.add(UCC_COLL_TYPE_ALLREDUCE, mt, r);
+            map.add(ct, mt, r);
         }
     };
     auto add = [&](ucc_coll_type_t ct, size_t lo, size_t hi, int score,
@@ -1478,7 +1614,11 @@ void Cdna4TlTeam::get_scores(Team *team, ScoreMap &map)
     };
     add(UCC_COLL_TYPE_ALLREDUCE, 0, cfg_.fused_max, 100, "fused", true);
     if (Config::instance().get_bool("TL_CDNA4", "GATED", true)) {
-        add_gated(cfg_.fused_max + 1, SIZE_MAX, 90);
+        add_gated(UCC_COLL_TYPE_ALLREDUCE, cfg_.fused_max + 1, SIZE_MAX,
+                  90);
+        add_gated(UCC_COLL_TYPE_REDUCE_SCATTER, 0, SIZE_MAX, 90);
+        add_gated(UCC_COLL_TYPE_ALLGATHER, 0, SIZE_MAX, 90);
+        add_gated(UCC_COLL_TYPE_ALLTOALL, 0, SIZE_MAX, 90);
     }
     add(UCC_COLL_TYPE_ALLREDUCE, 0, SIZE_MAX, 80, "staged_linear", false);
     add(UCC_COLL_TYPE_ALLGATHER, 0, SIZE_MAX, 80, "staged_linear", false);

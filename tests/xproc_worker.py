@@ -128,6 +128,51 @@ def main():
         torch.testing.assert_close(got, exp)
     results.append("alltoallv_xproc")
 
+    # 3a2. gated reduce_scatter / allgather / alltoall (device-gated
+    # pipeline paths, cross-process only), repeated for counter
+    # continuity across mixed coll types on shared slots.
+    per = 5_000_000
+    for it in range(2):
+        full = torch.randn(world, per * world, generator=g0)
+        src = full[rank].cuda()
+        dst = torch.zeros(per, device="cuda")
+        rs = c.coll_init(team, "reduce_scatter", src=src.data_ptr(),
+                         dst=dst.data_ptr(), count=per, dt=dtypes.FLOAT32,
+                         mem_type=dtypes.MEM_CUDA)
+        wait(rs, ctx)
+        torch.cuda.synchronize()
+        exp = full.sum(0)
+        torch.testing.assert_close(
+            dst.cpu(), exp[rank * per:(rank + 1) * per], rtol=1e-5,
+            atol=1e-4)
+
+        blk = torch.randn(per, generator=g0)
+        src = blk.cuda()
+        agd = torch.zeros(per * world, device="cuda")
+        ag = c.coll_init(team, "allgather", src=src.data_ptr(),
+                         dst=agd.data_ptr(), count=per * world,
+                         dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA)
+        # every rank must contribute its own block: use rank-dependent data
+        src.copy_(blk + rank)
+        torch.cuda.synchronize()
+        wait(ag, ctx)
+        torch.cuda.synchronize()
+        exp = torch.cat([blk + r for r in range(world)])
+        torch.testing.assert_close(agd.cpu(), exp)
+
+        a2s = torch.randn(world, per * world, generator=g0)
+        src = a2s[rank].cuda()
+        a2d = torch.zeros(per * world, device="cuda")
+        a2 = c.coll_init(team, "alltoall", src=src.data_ptr(),
+                         dst=a2d.data_ptr(), count=per * world,
+                         dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA)
+        wait(a2, ctx)
+        torch.cuda.synchronize()
+        exp = torch.cat(
+            [a2s[s][rank * per:(rank + 1) * per] for s in range(world)])
+        torch.testing.assert_close(a2d.cpu(), exp)
+    results.append("gated_rs_ag_a2a")
+
     # 3b. ucc_mem_map export/import across processes: rank 0 exports a
     # device buffer, rank 1 imports and reads it over IPC.
     if world == 2:
