@@ -207,10 +207,15 @@ std::string ScoreMap::to_string() const
         for (int mi = 0; mi < UCC_MEMORY_TYPE_LAST; mi++) {
             for (const auto &r : ranges[ci][mi]) {
                 out << k_coll_names[ci] << ":"
-                    << mem_type_name((ucc_memory_type_t)mi) << ":" << r.start
-                    << "-" << (r.end == SIZE_MAX ? (size_t)0 : r.end) << ":@"
-                    << r.tl_name << "/" << r.alg_name << ":" << r.score
-                    << "\n";
+                    << mem_type_name((ucc_memory_type_t)mi) << ":"
+                    << r.start << "-";
+                if (r.end == SIZE_MAX) {
+                    out << "inf";
+                } else {
+                    out << r.end;
+                }
+                out << ":@" << r.tl_name << "/" << r.alg_name << ":"
+                    << r.score << "\n";
             }
         }
     }
