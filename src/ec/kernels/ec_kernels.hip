@@ -315,34 +315,67 @@ __global__ void k_reduce_scalar(ReduceArgs a)
 }
 
 /* -------------------------------------------------------- gather copy  */
+__device__ __forceinline__ void d_copy_bytes_part(uint8_t *dst,
+                                                  const uint8_t *src,
+                                                  uint64_t len,
+                                                  uint64_t vblock,
+                                                  uint64_t vgrid)
+{
+    const uint64_t tid = vblock * blockDim.x + threadIdx.x;
+    const uint64_t str = vgrid * blockDim.x;
+    if ((((uintptr_t)dst | (uintptr_t)src) & 15) == 0) {
+        uint64_t nv = len / 16;
+        for (uint64_t i = tid; i < nv; i += str) {
+            ((uint4 *)dst)[i] = ((const uint4 *)src)[i];
+        }
+        for (uint64_t i = nv * 16 + tid; i < len; i += str) {
+            dst[i] = src[i];
+        }
+    } else {
+        for (uint64_t i = tid; i < len; i += str) {
+            dst[i] = src[i];
+        }
+    }
+}
+
+__device__ __forceinline__ void d_copy_bytes(uint8_t *dst,
+                                             const uint8_t *src,
+                                             uint64_t len)
+{
+    d_copy_bytes_part(dst, src, len, blockIdx.x, gridDim.x);
+}
+
+
 __global__ void k_gather_copy(GatherArgs a)
 {
-    for (int s = 0; s < a.n; s++) {
-        uint8_t       *dst = (uint8_t *)a.dst_base + a.offs[s];
-        const uint8_t *src = (const uint8_t *)a.srcs[s];
-        uint64_t       len = a.lens[s];
-        const uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-        const uint64_t str = (uint64_t)gridDim.x * blockDim.x;
-        if ((((uintptr_t)dst | (uintptr_t)src) & 15) == 0) {
-            uint64_t nv = len / 16;
-            for (uint64_t i = tid; i < nv; i += str) {
-                ((uint4 *)dst)[i] = ((const uint4 *)src)[i];
-            }
-            for (uint64_t i = nv * 16 + tid; i < len; i += str) {
-                dst[i] = src[i];
-            }
-        } else if ((((uintptr_t)dst | (uintptr_t)src) & 3) == 0) {
-            uint64_t nv = len / 4;
-            for (uint64_t i = tid; i < nv; i += str) {
-                ((uint32_t *)dst)[i] = ((const uint32_t *)src)[i];
-            }
-            for (uint64_t i = nv * 4 + tid; i < len; i += str) {
-                dst[i] = src[i];
-            }
-        } else {
-            for (uint64_t i = tid; i < len; i += str) {
-                dst[i] = src[i];
-            }
+    /* blocks partitioned across sources: peers' xGMI links stream
+     * concurrently (sequential per-peer copies would serialize on one
+     * link at a time) */
+    const int grp = (int)gridDim.x / a.n > 0 ? (int)gridDim.x / a.n : 1;
+    int       s   = (int)blockIdx.x / grp;
+    int       bid = (int)blockIdx.x % grp;
+    if ((int)gridDim.x < a.n) { /* fewer blocks than sources: loop */
+        for (s = (int)blockIdx.x; s < a.n; s += gridDim.x) {
+            uint8_t       *dst = (uint8_t *)a.dst_base + a.offs[s];
+            const uint8_t *src = (const uint8_t *)a.srcs[s];
+            d_copy_bytes_part(dst, src, a.lens[s], 0, 1);
+        }
+        return;
+    }
+    if (s >= a.n) {
+        return;
+    }
+    uint8_t       *dst = (uint8_t *)a.dst_base + a.offs[s];
+    const uint8_t *src = (const uint8_t *)a.srcs[s];
+    if ((((uintptr_t)dst | (uintptr_t)src) & 3) == 0 ||
+        (((uintptr_t)dst | (uintptr_t)src) & 15) == 0) {
+        d_copy_bytes_part(dst, src, a.lens[s], (uint64_t)bid,
+                          (uint64_t)grp);
+    } else {
+        const uint64_t tid = (uint64_t)bid * blockDim.x + threadIdx.x;
+        const uint64_t str = (uint64_t)grp * blockDim.x;
+        for (uint64_t i = tid; i < a.lens[s]; i += str) {
+            dst[i] = src[i];
         }
     }
 }
@@ -662,27 +695,6 @@ __device__ __forceinline__ void gated_signal(const GatedArgs &a, int phase)
     }
 }
 
-__device__ __forceinline__ void d_copy_bytes(uint8_t *dst,
-                                             const uint8_t *src,
-                                             uint64_t len)
-{
-    const uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    const uint64_t str = (uint64_t)gridDim.x * blockDim.x;
-    if ((((uintptr_t)dst | (uintptr_t)src) & 15) == 0) {
-        uint64_t nv = len / 16;
-        for (uint64_t i = tid; i < nv; i += str) {
-            ((uint4 *)dst)[i] = ((const uint4 *)src)[i];
-        }
-        for (uint64_t i = nv * 16 + tid; i < len; i += str) {
-            dst[i] = src[i];
-        }
-    } else {
-        for (uint64_t i = tid; i < len; i += str) {
-            dst[i] = src[i];
-        }
-    }
-}
-
 __global__ void k_staged_stage(GatedArgs a)
 {
     if (!gated_wait(a, 1, a.t_sw_reduce) ||
@@ -758,11 +770,17 @@ __global__ void k_staged_gather(GatedArgs a)
     if (!gated_wait(a, a.gw_phase, a.t_gather_wait)) {
         return;
     }
-    for (int r = 0; r < a.nranks; r++) {
-        uint64_t b = a.slice_b[r], e = a.slice_e[r];
+    /* partition the grid across sources: every peer link streams
+     * concurrently instead of one-after-another (7 x 153 GB/s at once) */
+    const int grp = gridDim.x / a.nranks > 0 ? gridDim.x / a.nranks : 1;
+    const int src = (int)blockIdx.x / grp;
+    const int bid = (int)blockIdx.x % grp;
+    if (src < a.nranks) {
+        uint64_t b = a.slice_b[src], e = a.slice_e[src];
         if (e > b) {
-            d_copy_bytes((uint8_t *)a.dst + b,
-                         (const uint8_t *)a.peer_out[r], e - b);
+            d_copy_bytes_part((uint8_t *)a.dst + b,
+                              (const uint8_t *)a.peer_out[src], e - b,
+                              (uint64_t)bid, (uint64_t)grp);
         }
     }
     gated_signal(a, 2);
