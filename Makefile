@@ -20,6 +20,7 @@ LIB_SRCS := $(wildcard src/utils/*.cc) \
             $(wildcard src/schedule/*.cc) \
             $(wildcard src/coll_score/*.cc) \
             $(wildcard src/coll_patterns/*.cc) \
+            $(wildcard src/cl/*.cc) \
             $(wildcard src/mc/*.cc) \
             $(wildcard src/ec/*.cc) \
             $(wildcard src/topo/*.cc) \

@@ -1,0 +1,499 @@
+/* Hierarchical composition — the cl/hier role (reference
+ * src/components/cl/hier: allreduce RAB = node reduce -> leader
+ * allreduce -> node bcast, cl_hier.h:38-57, allreduce_rab.c),
+ * re-derived for this framework's collapsed CL/TL stack:
+ *
+ *  - After a multi-node team reaches TL_CREATE, core drives the HIER
+ *    state: two INTERNAL sub-teams are created — node_team (my node's
+ *    ranks, served by shm/cdna4) and leaders_team (lowest rank per
+ *    node, served by tcp) — through the same public team machinery.
+ *  - Sub-team bootstrap rides the PARENT team's OOB: every sub-OOB
+ *    round is one parent allgather of a fixed kPad-byte slot, so
+ *    non-members observe (contribute zeros) and the per-rank round
+ *    sequence stays positionally aligned across all ranks. Team create
+ *    is forced to exactly two OOB rounds (addr + TL exchange) for this
+ *    alignment (core pads a zero-stride TL exchange to 8 bytes).
+ *  - The hier allreduce is a 3-phase task over the sub-teams,
+ *    registered in the parent score map at score 60 for host memory
+ *    (above tcp's flat algorithms, below same-node shm which wins when
+ *    the team does not span nodes — where hier is never installed).
+ *
+ * Testability without a cluster: UCC_FAKE_NODE_SPLIT=k makes contexts
+ * hash to k pseudo-nodes (core/ucc_context_create), so the full
+ * node/leader composition runs inside one machine (tests/test_hier.py).
+ */
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <map>
+
+#include "../core/core.h"
+#include "../ec/ec_cpu.h"
+#include "../topo/topo.h"
+
+namespace ucc {
+
+static constexpr size_t kPad = 512; /* fixed parent-OOB slot per round */
+
+/* ------------------------------------------------------------- SubOob */
+/* Declared in core.h as ucc::SubOob (Team member pointers). */
+struct SubOob {
+    Team                 *parent = nullptr;
+    std::vector<uint32_t> members; /* parent ranks, sorted              */
+    int                   my_idx = -1;
+
+    struct Req {
+        SubOob              *so;
+        void                *parent_req = nullptr;
+        std::vector<uint8_t> full;   /* parent_n * kPad                 */
+        std::vector<uint8_t> send;   /* kPad                            */
+        void                *recv = nullptr;
+        size_t               size = 0;
+        bool                 done = false;
+    };
+
+    ucc_status_t start(const void *src, void *recv, size_t size,
+                       void **request)
+    {
+        if (size + 8 > kPad) {
+            return UCC_ERR_NO_RESOURCE;
+        }
+        auto *r = new Req;
+        r->so   = this;
+        r->send.assign(kPad, 0);
+        if (src && size) {
+            *(uint64_t *)r->send.data() = size;
+            memcpy(r->send.data() + 8, src, size);
+        }
+        r->full.resize((size_t)parent->size * kPad);
+        r->recv = recv;
+        r->size = size;
+        ucc_status_t st = parent->oob.allgather(
+            r->send.data(), r->full.data(), kPad, parent->oob.coll_info,
+            &r->parent_req);
+        if (st != UCC_OK) {
+            delete r;
+            return st;
+        }
+        *request = r;
+        return UCC_OK;
+    }
+
+    static ucc_status_t cb_allgather(void *src, void *recv, size_t size,
+                                     void *info, void **request)
+    {
+        return ((SubOob *)info)->start(src, recv, size, request);
+    }
+
+    static ucc_status_t cb_test(void *request)
+    {
+        auto *r = (Req *)request;
+        if (!r->done) {
+            ucc_status_t st =
+                r->so->parent->oob.req_test(r->parent_req);
+            if (st == UCC_INPROGRESS) {
+                return st;
+            }
+            r->so->parent->oob.req_free(r->parent_req);
+            r->parent_req = nullptr;
+            if (st != UCC_OK) {
+                return st;
+            }
+            if (r->recv && r->size) {
+                for (size_t i = 0; i < r->so->members.size(); i++) {
+                    memcpy((uint8_t *)r->recv + i * r->size,
+                           r->full.data() +
+                               (size_t)r->so->members[i] * kPad + 8,
+                           r->size);
+                }
+            }
+            r->done = true;
+        }
+        return UCC_OK;
+    }
+
+    static ucc_status_t cb_free(void *request)
+    {
+        delete (Req *)request;
+        return UCC_OK;
+    }
+
+    /* ------------- observer (non-member) round machinery */
+    Req *obs_req = nullptr;
+    int  obs_rounds_done = 0;
+
+    ucc_status_t observe_tick(int target_rounds)
+    {
+        if (obs_rounds_done >= target_rounds) {
+            return UCC_OK;
+        }
+        if (!obs_req) {
+            void *rq = nullptr;
+            ucc_status_t st = start(nullptr, nullptr, 0, &rq);
+            if (st != UCC_OK) {
+                return st;
+            }
+            obs_req = (Req *)rq;
+        }
+        ucc_status_t st = cb_test(obs_req);
+        if (st == UCC_INPROGRESS) {
+            return UCC_INPROGRESS;
+        }
+        cb_free(obs_req);
+        obs_req = nullptr;
+        if (st != UCC_OK) {
+            return st;
+        }
+        obs_rounds_done++;
+        return obs_rounds_done >= target_rounds ? UCC_OK : UCC_INPROGRESS;
+    }
+};
+
+namespace hier {
+
+static thread_local bool g_creating_subteam = false;
+
+bool creating_subteam() { return g_creating_subteam; }
+
+/* ------------------------------------------------------------- wanted  */
+bool wanted(Team *team)
+{
+    if (g_creating_subteam || team->is_subteam || team->size < 2 ||
+        !team->has_oob) {
+        return false;
+    }
+    if (!Config::instance().get_bool("CL_HIER", "ENABLE", true)) {
+        return false;
+    }
+    return !team->all_same_node();
+}
+
+/* ------------------------------------------------------------- setup   */
+static ucc_status_t create_subteam(Team *parent, SubOob *so,
+                                   std::unique_ptr<Team> &out)
+{
+    ucc_team_params_t tp{};
+    tp.mask          = UCC_TEAM_PARAM_FIELD_OOB;
+    tp.oob.allgather = SubOob::cb_allgather;
+    tp.oob.req_test  = SubOob::cb_test;
+    tp.oob.req_free  = SubOob::cb_free;
+    tp.oob.coll_info = so;
+    tp.oob.n_oob_eps = (uint32_t)so->members.size();
+    tp.oob.oob_ep    = (uint32_t)so->my_idx;
+    ucc_context_h ch = reinterpret_cast<ucc_context_h>(parent->ctx);
+    ucc_team_h    th = nullptr;
+    g_creating_subteam = true;
+    ucc_status_t st = ucc_team_create_post(&ch, 1, &tp, &th);
+    g_creating_subteam = false;
+    if (st != UCC_OK) {
+        return st;
+    }
+    out.reset(reinterpret_cast<Team *>(th));
+    return UCC_OK;
+}
+
+ucc_status_t setup(Team *team)
+{
+    auto node = topo::build_sbgp(team, topo::SbgpType::NODE);
+    auto ldr  = topo::build_sbgp(team, topo::SbgpType::NODE_LEADERS);
+    team->node_ranks   = node.ranks;
+    team->leader_ranks = ldr.ranks;
+    if (node.ranks.size() < 2 || ldr.ranks.size() < 2 ||
+        ldr.ranks.size() == team->size) {
+        /* every rank on its own node (1 proc/node) or trivial split:
+         * hier adds nothing over the flat inter-node path */
+        return UCC_ERR_NOT_SUPPORTED;
+    }
+    /* parent-OOB round alignment requires every node team to run the
+     * same 2 bootstrap rounds: sizes must all be >= 2 */
+    {
+        std::map<uint64_t, uint32_t> per_node;
+        for (auto &p : team->procs) {
+            per_node[p.host_hash]++;
+        }
+        for (auto &kv : per_node) {
+            if (kv.second < 2) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+        }
+    }
+    team->node_oob          = new SubOob;
+    team->node_oob->parent  = team;
+    team->node_oob->members = node.ranks;
+    team->node_oob->my_idx  = node.my_idx;
+    team->leaders_oob          = new SubOob;
+    team->leaders_oob->parent  = team;
+    team->leaders_oob->members = ldr.ranks;
+    team->leaders_oob->my_idx  = ldr.my_idx; /* -1 if not a leader */
+    team->hier_step            = 0;
+    if (node.ranks.size() == 1) {
+        /* singleton node: node team is size-1 (self TL) — skip its OOB
+         * rounds entirely is NOT possible (alignment); create it anyway
+         * via the sub-OOB (2 rounds, self TL serves colls). */
+    }
+    ucc_status_t st = create_subteam(team, team->node_oob,
+                                     team->node_team);
+    if (st != UCC_OK) {
+        return st;
+    }
+    return UCC_OK;
+}
+
+/* ------------------------------------------------------------- test    */
+ucc_status_t test(Team *team)
+{
+    const bool leader = team->leaders_oob->my_idx >= 0;
+    static const bool dbg = getenv("UCC_HIER_DEBUG") != nullptr;
+    if (dbg) {
+        fprintf(stderr, "[hier] rank %u step %d node_state %d ldr_state %d\n",
+                team->rank, team->hier_step,
+                team->node_team ? (int)team->node_team->state : -1,
+                team->leaders_team ? (int)team->leaders_team->state : -1);
+    }
+    switch (team->hier_step) {
+    case 0: { /* node team consumes its 2 parent rounds */
+        ucc_status_t st = ucc_team_create_test(
+            reinterpret_cast<ucc_team_h>(team->node_team.get()));
+        if (st < 0) {
+            return st;
+        }
+        if (team->node_team->state < Team::TL_CREATE) {
+            return UCC_INPROGRESS;
+        }
+        /* both node OOB rounds consumed: start leaders stage */
+        if (leader) {
+            ucc_status_t cs = create_subteam(team, team->leaders_oob,
+                                             team->leaders_team);
+            if (cs != UCC_OK) {
+                return cs;
+            }
+        }
+        team->hier_step = 1;
+        [[fallthrough]];
+    }
+    case 1: { /* leaders team rounds (leaders) / 2 observed (others) */
+        if (leader) {
+            ucc_status_t st = ucc_team_create_test(
+                reinterpret_cast<ucc_team_h>(team->leaders_team.get()));
+            if (st < 0) {
+                return st;
+            }
+            if (team->leaders_team->state < Team::TL_CREATE) {
+                return UCC_INPROGRESS;
+            }
+        } else {
+            ucc_status_t st = team->leaders_oob->observe_tick(2);
+            if (st == UCC_INPROGRESS) {
+                return UCC_INPROGRESS;
+            }
+            if (st != UCC_OK) {
+                return st;
+            }
+        }
+        team->hier_step = 2;
+        [[fallthrough]];
+    }
+    case 2: { /* finish both create_test phases (no parent OOB) */
+        ucc_status_t st = ucc_team_create_test(
+            reinterpret_cast<ucc_team_h>(team->node_team.get()));
+        if (st < 0) {
+            return st;
+        }
+        if (st == UCC_INPROGRESS) {
+            return UCC_INPROGRESS;
+        }
+        if (leader) {
+            st = ucc_team_create_test(
+                reinterpret_cast<ucc_team_h>(team->leaders_team.get()));
+            if (st < 0) {
+                return st;
+            }
+            if (st == UCC_INPROGRESS) {
+                return UCC_INPROGRESS;
+            }
+        }
+        return UCC_OK;
+    }
+    }
+    return UCC_ERR_INVALID_PARAM;
+}
+
+void destroy(Team *team)
+{
+    delete team->node_oob;
+    delete team->leaders_oob;
+    team->node_oob    = nullptr;
+    team->leaders_oob = nullptr;
+}
+
+/* ------------------------------------------ hier allreduce (RAB) task  */
+class HierAllreduceTask final : public Task {
+  public:
+    HierAllreduceTask(Context *ctx, Team *team,
+                      const ucc_coll_args_t &args)
+        : Task(ctx), team_(team), a_(args)
+    {
+    }
+    ~HierAllreduceTask() override
+    {
+        if (req_h_) {
+            ucc_collective_finalize(req_h_);
+        }
+    }
+
+    ucc_status_t post() override
+    {
+        phase_   = 0;
+        leader_  = team_->leaders_oob->my_idx >= 0;
+        inplace_ = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        count_   = a_.dst.info.count;
+        dt_      = a_.dst.info.datatype;
+        status   = UCC_INPROGRESS;
+        return step();
+    }
+
+    ucc_status_t progress() override { return step(); }
+
+  private:
+    ucc_status_t launch(Team *t, ucc_coll_type_t ct, uint64_t root)
+    {
+        ucc_coll_args_t sa{};
+        sa.mask      = UCC_COLL_ARGS_FIELD_FLAGS;
+        sa.flags     = a_.flags & (UCC_COLL_ARGS_FLAG_COUNT_64BIT |
+                                   UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT);
+        sa.coll_type = ct;
+        sa.root      = root;
+        sa.op = a_.op == UCC_OP_AVG ? UCC_OP_SUM : a_.op;
+        if (ct == UCC_COLL_TYPE_REDUCE) {
+            sa.src.info        = a_.src.info;
+            if (inplace_) {
+                sa.src.info.buffer   = a_.dst.info.buffer;
+                sa.src.info.count    = count_;
+                sa.src.info.datatype = dt_;
+                sa.src.info.mem_type = a_.dst.info.mem_type;
+            }
+            sa.dst.info = a_.dst.info;
+        } else if (ct == UCC_COLL_TYPE_ALLREDUCE) {
+            sa.flags |= UCC_COLL_ARGS_FLAG_IN_PLACE;
+            sa.src.info = a_.dst.info;
+            sa.dst.info = a_.dst.info;
+        } else { /* bcast */
+            sa.src.info = a_.dst.info;
+        }
+        ucc_status_t st = ucc_collective_init(
+            &sa, &req_h_, reinterpret_cast<ucc_team_h>(t));
+        if (st != UCC_OK) {
+            return st;
+        }
+        return ucc_collective_post(req_h_);
+    }
+
+    ucc_status_t step()
+    {
+        while (true) {
+            if (req_h_) {
+                ucc_status_t st = ucc_collective_test(req_h_);
+                if (st == UCC_INPROGRESS) {
+                    return UCC_INPROGRESS;
+                }
+                ucc_collective_finalize(req_h_);
+                req_h_ = nullptr;
+                if (st != UCC_OK) {
+                    return st;
+                }
+                phase_++;
+            }
+            switch (phase_) {
+            case 0: { /* node reduce to node leader (node rank 0) */
+                if (team_->node_team->size == 1) {
+                    /* singleton: local copy src->dst unless inplace */
+                    if (!inplace_) {
+                        memcpy(a_.dst.info.buffer, a_.src.info.buffer,
+                               count_ * ucc_dt_size(dt_));
+                    }
+                    phase_ = 1;
+                    continue;
+                }
+                ucc_status_t st =
+                    launch(team_->node_team.get(),
+                           UCC_COLL_TYPE_REDUCE, 0);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 1: { /* leaders allreduce (leaders only) */
+                if (!leader_) {
+                    phase_ = 2;
+                    continue;
+                }
+                ucc_status_t st = launch(team_->leaders_team.get(),
+                                         UCC_COLL_TYPE_ALLREDUCE, 0);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 2: { /* node bcast from the leader */
+                if (team_->node_team->size == 1) {
+                    phase_ = 3;
+                    continue;
+                }
+                ucc_status_t st = launch(team_->node_team.get(),
+                                         UCC_COLL_TYPE_BCAST, 0);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 3: { /* AVG: final local scale by 1/N */
+                if (a_.op == UCC_OP_AVG) {
+                    const void *srcs[1] = {a_.dst.info.buffer};
+                    ec_cpu::reduce(a_.dst.info.buffer, srcs, 1, count_,
+                                   dt_, UCC_OP_SUM,
+                                   1.0 / (double)team_->size);
+                }
+                return UCC_OK;
+            }
+            default:
+                return UCC_ERR_INVALID_PARAM;
+            }
+            /* a sub-collective was posted; poll it next iteration */
+            if (req_h_) {
+                continue;
+            }
+        }
+    }
+
+    Team           *team_;
+    ucc_coll_args_t a_;
+    ucc_coll_req_h  req_h_ = nullptr;
+    int             phase_ = 0;
+    bool            leader_ = false, inplace_ = false;
+    uint64_t        count_ = 0;
+    ucc_datatype_t  dt_ = UCC_DT_FLOAT32;
+};
+
+void add_scores(Team *team)
+{
+    ScoreRange r;
+    r.start    = 0;
+    r.end      = SIZE_MAX;
+    r.score    = 60;
+    r.tl_name  = "hier";
+    r.alg_name = "rab";
+    r.init     = [](const ucc_coll_args_t &args, Team *t,
+                Task **task) -> ucc_status_t {
+        if (args.op != UCC_OP_SUM && args.op != UCC_OP_MAX &&
+            args.op != UCC_OP_MIN && args.op != UCC_OP_PROD &&
+            args.op != UCC_OP_AVG) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        *task = new HierAllreduceTask(t->ctx, t, args);
+        return UCC_OK;
+    };
+    team->score_map.add(UCC_COLL_TYPE_ALLREDUCE, UCC_MEMORY_TYPE_HOST, r);
+}
+
+} // namespace hier
+} // namespace ucc

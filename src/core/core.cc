@@ -6,6 +6,7 @@
 #include "../mc/mc.h"
 #include "../utils/profile.h"
 
+#include <atomic>
 #include <cstdlib>
 #include <random>
 #include <unistd.h>
@@ -32,6 +33,8 @@ ProcInfo local_proc_info()
     pi.device    = -1;
     return pi;
 }
+
+Team::~Team() { hier::destroy(this); }
 
 bool Team::all_same_node() const
 {
@@ -227,6 +230,18 @@ ucc_status_t ucc_context_create(ucc_lib_h lib_h,
     ctx->seq  = lib->next_ctx_seq++;
     ctx->proc = local_proc_info();
     ctx->proc.ctx_seq = ((uint64_t)ctx->proc.pid << 20) | ctx->seq;
+    /* test hook: UCC_FAKE_NODE_SPLIT=k spreads in-process contexts over
+     * k pseudo-nodes so the hier composition is exercisable on one
+     * machine (tests/test_hier.py) */
+    if (const char *fs = getenv("UCC_FAKE_NODE_SPLIT")) {
+        int k = atoi(fs);
+        if (k > 1) {
+            static std::atomic<uint64_t> g_fake_ctx{0};
+            uint64_t idx = g_fake_ctx.fetch_add(1);
+            ctx->proc.host_hash ^=
+                0x9e3779b97f4a7c15ull * (1 + idx % (uint64_t)k);
+        }
+    }
     for (Tl *tl : tl_registry()) {
         TlContext *tlc = tl->context_create(ctx);
         if (tlc) {
@@ -270,7 +285,8 @@ ucc_status_t ucc_team_create_post(ucc_context_h *contexts,
     }
     auto *ctx  = reinterpret_cast<Context *>(contexts[0]);
     auto *team = new Team;
-    team->ctx    = ctx;
+    team->ctx        = ctx;
+    team->is_subteam = hier::creating_subteam();
     team->params = *team_params;
     team->has_oob = team_params->mask & UCC_TEAM_PARAM_FIELD_OOB;
     if (team->has_oob) {
@@ -360,11 +376,14 @@ ucc_status_t ucc_team_create_test(ucc_team_h team_h)
             team->exchg_off.push_back(off);
             off += tt->exchg_size();
         }
-        team->exchg_stride = off;
+        /* Always run the TL exchange round (pad zero strides): team
+         * create is then EXACTLY two OOB rounds on every team, which
+         * keeps parent-OOB round sequences positionally aligned for the
+         * hier sub-team bootstrap (src/cl/hier.cc). */
         if (off == 0) {
-            team->state = Team::TL_CREATE;
-            return ucc_team_create_test(team_h);
+            off = 8;
         }
+        team->exchg_stride = off;
         team->exchg_buf.assign(off, 0);
         for (size_t i = 0; i < team->tl_teams.size(); i++) {
             team->tl_teams[i]->exchg_pack(team->exchg_buf.data() +
@@ -426,6 +445,15 @@ ucc_status_t ucc_team_create_test(ucc_team_h team_h)
             team->state = Team::FAILED;
             return (team->err = UCC_ERR_NO_RESOURCE);
         }
+        if (hier::wanted(team)) {
+            ucc_status_t hs = hier::setup(team);
+            if (hs == UCC_OK) {
+                team->want_hier = true;
+                team->state     = Team::HIER_CREATE;
+                return ucc_team_create_test(team_h);
+            }
+            ucc_warn("hier setup declined (%d), flat team", hs);
+        }
         for (auto &tt : team->tl_teams) {
             tt->get_scores(team, team->score_map);
         }
@@ -436,6 +464,30 @@ ucc_status_t ucc_team_create_test(ucc_team_h team_h)
         if (log_level() >= LogLevel::INFO) {
             ucc_info("team %u size %u score map:\n%s", team->id, team->size,
                      team->score_map.to_string().c_str());
+        }
+        team->state = Team::ACTIVE;
+        return UCC_OK;
+    }
+    case Team::HIER_CREATE: {
+        ucc_status_t st = hier::test(team);
+        if (st == UCC_INPROGRESS) {
+            return UCC_INPROGRESS;
+        }
+        if (st != UCC_OK) {
+            team->state = Team::FAILED;
+            return (team->err = st);
+        }
+        for (auto &tt : team->tl_teams) {
+            tt->get_scores(team, team->score_map);
+        }
+        hier::add_scores(team);
+        std::string tune2 = Config::instance().get("", "TUNE", "");
+        if (!tune2.empty()) {
+            team->score_map.apply_str(tune2);
+        }
+        if (log_level() >= LogLevel::INFO) {
+            ucc_info("team %u size %u score map:\n%s", team->id,
+                     team->size, team->score_map.to_string().c_str());
         }
         team->state = Team::ACTIVE;
         return UCC_OK;

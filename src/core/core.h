@@ -247,6 +247,7 @@ struct Team {
         ADDR_EXCHANGE,
         TL_EXCHANGE,
         TL_CREATE,
+        HIER_CREATE, /* internal sub-team bootstrap (src/cl/hier.cc) */
         ACTIVE,
         FAILED,
     };
@@ -268,6 +269,21 @@ struct Team {
     ScoreMap            score_map;
     uint64_t            team_uid = 0; /* rank0 random, shared: shm naming  */
     uint64_t            coll_seq = 0;
+
+    /* hierarchical composition (cl/hier role): internal sub-teams built
+     * after TL_CREATE when the team spans nodes. node_team covers my
+     * node's ranks; leaders_team (leaders only) covers one rank per
+     * node. Sub-team OOB rounds ride on the parent OOB with fixed-size
+     * padded payloads so non-members can observe (see SubOob). */
+    std::unique_ptr<Team>   node_team;
+    std::unique_ptr<Team>   leaders_team;
+    struct SubOob          *node_oob    = nullptr;
+    struct SubOob          *leaders_oob = nullptr;
+    int                     hier_step   = 0;
+    bool                    want_hier   = false;
+    bool                    is_subteam  = false;
+    std::vector<uint32_t>   node_ranks, leader_ranks;
+    ~Team();
 
     bool all_same_node() const;
     bool all_have_device() const;
@@ -305,6 +321,16 @@ struct CollRequest {
  * ucc_coll_utils msgsize conventions). */
 size_t coll_args_msgsize(const ucc_coll_args_t &args, uint32_t rank,
                          uint32_t size);
+
+/* hierarchical composition hooks (src/cl/hier.cc) */
+namespace hier {
+bool         creating_subteam();
+bool         wanted(Team *team);
+ucc_status_t setup(Team *team);
+ucc_status_t test(Team *team);
+void         add_scores(Team *team);
+void         destroy(Team *team);
+} // namespace hier
 ucc_memory_type_t coll_args_mem_type(const ucc_coll_args_t &args,
                                      uint32_t rank);
 

@@ -117,7 +117,8 @@ class Cdna4TlTeam final : public TlTeam {
     {
         char buf[96];
         snprintf(buf, sizeof(buf), "/uccamd-c4-%016llx-%u",
-                 (unsigned long long)team->team_uid, team->id);
+                 (unsigned long long)team->team_uid,
+                 (unsigned)(team->team_uid >> 48));
         seg_name_ = buf;
         init_st_  = local_init();
         if (init_st_ == UCC_OK && team->rank == 0) {

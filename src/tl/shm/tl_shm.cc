@@ -84,7 +84,8 @@ class ShmTlTeam final : public TlTeam {
     {
         char buf[96];
         snprintf(buf, sizeof(buf), "/uccamd-%016llx-%u",
-                 (unsigned long long)team->team_uid, team->id);
+                 (unsigned long long)team->team_uid,
+                 (unsigned)(team->team_uid >> 48));
         name_ = buf;
         if (team->rank == 0) {
             create_st_ = seg_.create(name_, team->size, nslots_, chunk_);

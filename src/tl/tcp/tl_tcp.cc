@@ -214,6 +214,64 @@ class TcpTlContext final : public TlContext {
     int      lfd_ = -1;
     uint32_t ip_  = 0;
     uint16_t port_ = 0;
+    /* accepted-but-unclaimed connections: the ONE listener serves every
+     * team of this context (parent + hier sub-teams), so accepts are
+     * pooled here and claimed by uid from each team's create_test —
+     * never closed on mismatch (they belong to another team). */
+    struct Pending {
+        int      fd;
+        bool     have_hello = false;
+        uint32_t uid = 0, rank = 0;
+    };
+    std::vector<Pending> pending_;
+
+    void poll_accepts()
+    {
+        while (true) {
+            int fd = accept(lfd_, nullptr, nullptr);
+            if (fd < 0) {
+                break;
+            }
+            set_nonblock(fd);
+            pending_.push_back(Pending{fd});
+        }
+        for (auto &p : pending_) {
+            if (p.have_hello || p.fd < 0) {
+                continue;
+            }
+            uint32_t hello[2];
+            ssize_t  r = ::recv(p.fd, hello, sizeof(hello), MSG_PEEK);
+            if (r == (ssize_t)sizeof(hello)) {
+                ::recv(p.fd, hello, sizeof(hello), 0);
+                p.have_hello = true;
+                p.uid        = hello[0];
+                p.rank       = hello[1];
+            } else if (r == 0 || (r < 0 && errno != EAGAIN &&
+                                  errno != EWOULDBLOCK)) {
+                close(p.fd);
+                p.fd = -1;
+            }
+        }
+        pending_.erase(std::remove_if(pending_.begin(), pending_.end(),
+                                      [](const Pending &p) {
+                                          return p.fd < 0;
+                                      }),
+                       pending_.end());
+    }
+
+    /* claim a pending connection for this team uid; fills fd+rank */
+    bool claim(uint32_t uid, uint32_t n, int *fd_out, uint32_t *rank_out)
+    {
+        for (auto &p : pending_) {
+            if (p.fd >= 0 && p.have_hello && p.uid == uid && p.rank < n) {
+                *fd_out   = p.fd;
+                *rank_out = p.rank;
+                p.fd      = -1;
+                return true;
+            }
+        }
+        return false;
+    }
 };
 
 class TcpTlTeam final : public TlTeam {
@@ -255,34 +313,13 @@ class TcpTlTeam final : public TlTeam {
         auto          *c  = (TcpTlContext *)tlc_;
         const uint32_t me = team_->rank;
         const uint32_t n  = team_->size;
-        /* accept incoming (peers with rank > me connect to me) */
-        while (true) {
-            int fd = accept(c->lfd_, nullptr, nullptr);
-            if (fd < 0) {
-                break;
-            }
-            set_nonblock(fd);
-            pending_.push_back(fd);
-        }
-        /* read 8-byte hello {team_uid_lo32, rank} from pending fds */
-        for (auto it = pending_.begin(); it != pending_.end();) {
-            uint32_t hello[2];
-            ssize_t  r = ::recv(*it, hello, sizeof(hello), MSG_PEEK);
-            if (r == (ssize_t)sizeof(hello)) {
-                ::recv(*it, hello, sizeof(hello), 0);
-                if (hello[0] == (uint32_t)team_->team_uid &&
-                    hello[1] < n) {
-                    conns_[hello[1]].fd = *it;
-                } else {
-                    close(*it);
-                }
-                it = pending_.erase(it);
-            } else if (r == 0 || (r < 0 && errno != EAGAIN)) {
-                close(*it);
-                it = pending_.erase(it);
-            } else {
-                ++it;
-            }
+        /* accept incoming (peers with rank > me connect to me); the
+         * context pools accepts for all teams, claimed by uid */
+        c->poll_accepts();
+        int      cfd;
+        uint32_t crank;
+        while (c->claim((uint32_t)team_->team_uid, n, &cfd, &crank)) {
+            conns_[crank].fd = cfd;
         }
         /* connect to lower ranks */
         for (uint32_t r = 0; r < me; r++) {
@@ -327,7 +364,7 @@ class TcpTlTeam final : public TlTeam {
 
     uint64_t mktag(uint64_t seq, uint32_t step)
     {
-        return ((uint64_t)team_->id << 48) | (seq << 16) | step;
+        return ((team_->team_uid & 0xffffull) << 48) | (seq << 16) | step;
     }
 
     std::vector<Conn>    conns_;

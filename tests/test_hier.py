@@ -1,0 +1,66 @@
+"""Hierarchical composition (cl/hier role): node reduce -> leader
+allreduce -> node bcast over internal sub-teams, exercised on one
+machine via UCC_FAKE_NODE_SPLIT (contexts spread over pseudo-nodes).
+Reference parity: cl/hier allreduce RAB (allreduce_rab.c)."""
+
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+WORKER = r"""
+import sys
+import numpy as np
+sys.path.insert(0, %r)
+from ucc_amd import core, dtypes
+from ucc_amd.testing import LocalJob
+
+n = 6  # 2 pseudo-nodes x 3 ranks
+job = LocalJob(n)
+c = core()
+smap = c.score_map_str(job.teams[0])
+assert "@hier/rab" in smap, "hier not installed:\n" + smap
+
+rng = np.random.default_rng(11)
+for count in (64, 5000, 100_001):
+    arrs = [(rng.random(count) - 0.5).astype(np.float32) for _ in range(n)]
+    outs = job.allreduce_np(arrs)
+    exp = np.sum(arrs, axis=0)
+    for o in outs:
+        np.testing.assert_allclose(o, exp, rtol=1e-5, atol=1e-5)
+
+# AVG (post-scaled)
+arrs = [np.full(1000, float(r + 1), np.float32) for r in range(n)]
+outs = job.allreduce_np(arrs, op=dtypes.OP_AVG)
+exp = np.full(1000, sum(range(1, n + 1)) / n, np.float32)
+for o in outs:
+    np.testing.assert_allclose(o, exp, rtol=1e-5, atol=1e-5)
+
+# repeated (sub-team/slot reuse)
+for it in range(10):
+    arrs = [np.full(257, float(r * it + 1), np.float64) for r in range(n)]
+    outs = job.allreduce_np(arrs)
+    exp = np.sum(arrs, axis=0)
+    for o in outs:
+        np.testing.assert_allclose(o, exp)
+print("HIER_OK")
+""" % (REPO,)
+
+
+def _run(env_extra):
+    env = dict(os.environ)
+    env.update(env_extra)
+    p = subprocess.run([sys.executable, "-c", WORKER], env=env,
+                       capture_output=True, text=True, timeout=300)
+    sys.stdout.write(p.stdout[-2000:])
+    sys.stderr.write(p.stderr[-3000:])
+    assert p.returncode == 0 and "HIER_OK" in p.stdout
+
+
+def test_hier_allreduce_fake_nodes():
+    _run({"UCC_FAKE_NODE_SPLIT": "2"})
+
+
+def test_hier_allreduce_three_nodes():
+    _run({"UCC_FAKE_NODE_SPLIT": "3"})
