@@ -859,8 +859,17 @@ static ucc_status_t launch_fused_graph(const GraphFusedArgs &a,
         default: return UCC_ERR_NOT_SUPPORTED;                               \
         }
 
-ucc_status_t reduce(const ReduceArgs &a, hipStream_t s)
+ucc_status_t reduce(const ReduceArgs &a_in, hipStream_t s)
 {
+    /* complex SUM/AVG == elementwise float SUM/AVG on 2x the elements */
+    ReduceArgs a = a_in;
+    if ((a.dt == UCC_DT_FLOAT32_COMPLEX || a.dt == UCC_DT_FLOAT64_COMPLEX) &&
+        (a.op == UCC_OP_SUM || a.op == UCC_OP_AVG ||
+         (int)a.op == 12)) {
+        a.dt = a.dt == UCC_DT_FLOAT32_COMPLEX ? UCC_DT_FLOAT32
+                                              : UCC_DT_FLOAT64;
+        a.count *= 2;
+    }
     ucc_reduction_op_t op = a.op;
     switch (a.dt) {
         UCC_DT_CASE_FLOAT(UCC_DT_BFLOAT16, bf16_t, launch_reduce)
@@ -881,8 +890,17 @@ ucc_status_t reduce(const ReduceArgs &a, hipStream_t s)
     }
 }
 
-ucc_status_t fused_allreduce(const FusedArgs &a, hipStream_t s)
+ucc_status_t fused_allreduce(const FusedArgs &a_in, hipStream_t s)
 {
+    /* complex SUM/AVG == elementwise float SUM/AVG on 2x the elements */
+    FusedArgs a = a_in;
+    if ((a.dt == UCC_DT_FLOAT32_COMPLEX || a.dt == UCC_DT_FLOAT64_COMPLEX) &&
+        (a.op == UCC_OP_SUM || a.op == UCC_OP_AVG ||
+         (int)a.op == 12)) {
+        a.dt = a.dt == UCC_DT_FLOAT32_COMPLEX ? UCC_DT_FLOAT32
+                                              : UCC_DT_FLOAT64;
+        a.count *= 2;
+    }
     ucc_reduction_op_t op = a.op;
     switch (a.dt) {
         UCC_DT_CASE_FLOAT(UCC_DT_BFLOAT16, bf16_t, launch_fused)
@@ -903,8 +921,17 @@ ucc_status_t fused_allreduce(const FusedArgs &a, hipStream_t s)
     }
 }
 
-ucc_status_t fused_allreduce_graph(const GraphFusedArgs &a, hipStream_t s)
+ucc_status_t fused_allreduce_graph(const GraphFusedArgs &a_in, hipStream_t s)
 {
+    /* complex SUM/AVG == elementwise float SUM/AVG on 2x the elements */
+    GraphFusedArgs a = a_in;
+    if ((a.dt == UCC_DT_FLOAT32_COMPLEX || a.dt == UCC_DT_FLOAT64_COMPLEX) &&
+        (a.op == UCC_OP_SUM || a.op == UCC_OP_AVG ||
+         (int)a.op == 12)) {
+        a.dt = a.dt == UCC_DT_FLOAT32_COMPLEX ? UCC_DT_FLOAT32
+                                              : UCC_DT_FLOAT64;
+        a.count *= 2;
+    }
     ucc_reduction_op_t op = a.op;
     switch (a.dt) {
         UCC_DT_CASE_FLOAT(UCC_DT_BFLOAT16, bf16_t, launch_fused_graph)
@@ -957,8 +984,16 @@ static ucc_status_t launch_staged_reduce(const GatedArgs &a, hipStream_t s)
     return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
 }
 
-ucc_status_t staged_reduce(const GatedArgs &a, hipStream_t s)
+ucc_status_t staged_reduce(const GatedArgs &a_in, hipStream_t s)
 {
+    /* complex SUM/AVG == elementwise float SUM/AVG on 2x the elements */
+    GatedArgs a = a_in;
+    if ((a.dt == UCC_DT_FLOAT32_COMPLEX || a.dt == UCC_DT_FLOAT64_COMPLEX) &&
+        (a.op == UCC_OP_SUM || a.op == UCC_OP_AVG ||
+         (int)a.op == 12)) {
+        a.dt = a.dt == UCC_DT_FLOAT32_COMPLEX ? UCC_DT_FLOAT32
+                                              : UCC_DT_FLOAT64;
+    }
     ucc_reduction_op_t op = a.op;
     switch (a.dt) {
         UCC_DT_CASE_FLOAT(UCC_DT_BFLOAT16, bf16_t, launch_staged_reduce)
@@ -1009,6 +1044,9 @@ bool dt_supported(ucc_datatype_t dt)
 
 bool op_supported(ucc_datatype_t dt, ucc_reduction_op_t op)
 {
+    if (dt == UCC_DT_FLOAT32_COMPLEX || dt == UCC_DT_FLOAT64_COMPLEX) {
+        return op == UCC_OP_SUM || op == UCC_OP_AVG;
+    }
     bool is_float = dt == UCC_DT_BFLOAT16 || dt == UCC_DT_FLOAT16 ||
                     dt == UCC_DT_FLOAT32 || dt == UCC_DT_FLOAT64 ||
                     dt == UCC_DT_FLOAT8_E4M3 || dt == UCC_DT_FLOAT8_E5M2;

@@ -297,3 +297,23 @@ def test_gather_scatter_device(job):
     for r in range(n):
         torch.testing.assert_close(sdst[r].cpu(),
                                    big.cpu()[r * per:(r + 1) * per])
+
+
+def test_allreduce_device_complex(job):
+    """complex64 SUM: elementwise float pairs (reference ec complex
+    reduce role, SUM/AVG subset on device)."""
+    torch.manual_seed(20)
+    n = job.n
+    count = 4096
+    srcs = [torch.randn(count, dtype=torch.complex64, device="cuda")
+            for _ in range(n)]
+    dsts = [torch.zeros(count, dtype=torch.complex64, device="cuda")
+            for _ in range(n)]
+    expected = sum(s.cpu() for s in srcs)
+    _run_device(job, "allreduce", [
+        dict(src=srcs[r].data_ptr(), dst=dsts[r].data_ptr(), count=count,
+             dt=dtypes.FLOAT32_COMPLEX, mem_type=dtypes.MEM_CUDA)
+        for r in range(n)
+    ])
+    for d in dsts:
+        torch.testing.assert_close(d.cpu(), expected, rtol=1e-5, atol=1e-4)
