@@ -383,6 +383,16 @@ typedef struct ucc_team_attr {
     uint64_t            size;
 } ucc_team_attr_t;
 
+/* Nonblocking collective team split (reference ucc.h:1656): every rank
+ * of the parent calls with included=1 (member, my_ep = desired rank
+ * order key) or included=0 (observer). Members receive the new team;
+ * observers receive a stub team (size 0) that must still be driven with
+ * ucc_team_create_test until UCC_OK and then destroyed — it carries the
+ * observer's share of the bootstrap rounds. */
+ucc_status_t ucc_team_create_from_parent(uint64_t my_ep, uint32_t included,
+                                         ucc_team_h parent_team,
+                                         ucc_team_h *new_team);
+
 ucc_status_t ucc_team_create_post(ucc_context_h *contexts,
                                   uint32_t num_contexts,
                                   const ucc_team_params_t *team_params,

@@ -306,6 +306,17 @@ PYBIND11_MODULE(_core, m)
         return (int)ucc_team_create_test(t->team);
     });
 
+    m.def("team_create_from_parent",
+          [](std::shared_ptr<CoreTeam> parent, uint64_t my_ep,
+             uint32_t included) {
+              auto t  = std::make_shared<CoreTeam>();
+              t->ctx  = parent->ctx;
+              check(ucc_team_create_from_parent(my_ep, included,
+                                                parent->team, &t->team),
+                    "team_create_from_parent");
+              return t;
+          });
+
     /* ---------------------------------------------------- collectives */
     m.def(
         "coll_init",

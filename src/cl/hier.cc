@@ -149,6 +149,37 @@ struct SubOob {
     }
 };
 
+namespace subooob_api {
+
+SubOob *make(Team *parent, const std::vector<uint32_t> &members,
+             int my_idx)
+{
+    auto *so    = new SubOob;
+    so->parent  = parent;
+    so->members = members;
+    so->my_idx  = my_idx;
+    return so;
+}
+
+ucc_status_t observe_tick(SubOob *so, int target_rounds)
+{
+    return so->observe_tick(target_rounds);
+}
+
+void fill_oob(SubOob *so, ucc_oob_coll_t *oob)
+{
+    oob->allgather = SubOob::cb_allgather;
+    oob->req_test  = SubOob::cb_test;
+    oob->req_free  = SubOob::cb_free;
+    oob->coll_info = so;
+    oob->n_oob_eps = (uint32_t)so->members.size();
+    oob->oob_ep    = (uint32_t)so->my_idx;
+}
+
+void free_(SubOob *so) { delete so; }
+
+} // namespace subooob_api
+
 namespace hier {
 
 static thread_local bool g_creating_subteam = false;

@@ -7,6 +7,7 @@
 #include "../utils/profile.h"
 
 #include <atomic>
+#include <algorithm>
 #include <cstdlib>
 #include <random>
 #include <unistd.h>
@@ -34,7 +35,14 @@ ProcInfo local_proc_info()
     return pi;
 }
 
-Team::~Team() { hier::destroy(this); }
+Team::~Team()
+{
+    hier::destroy(this);
+    if (split_oob) {
+        subooob_api::free_(split_oob);
+        split_oob = nullptr;
+    }
+}
 
 bool Team::all_same_node() const
 {
@@ -338,12 +346,136 @@ ucc_status_t ucc_team_create_post(ucc_context_h *contexts,
     return UCC_OK;
 }
 
+/* ---- team split (reference ucc.h ucc_team_create_from_parent):
+ * round 0 exchanges {included, ep} over the parent team; members then
+ * run the normal create flow over a SubOob (parent-padded rounds);
+ * observers drive matching padded rounds and finish as a size-0 stub. */
+ucc_status_t ucc_team_create_from_parent(uint64_t my_ep, uint32_t included,
+                                         ucc_team_h parent_h,
+                                         ucc_team_h *new_team)
+{
+    auto *parent = reinterpret_cast<Team *>(parent_h);
+    if (!parent || parent->state != Team::ACTIVE || !parent->has_oob) {
+        return UCC_ERR_INVALID_PARAM;
+    }
+    auto *team           = new Team;
+    team->ctx            = parent->ctx;
+    team->is_subteam     = true;
+    team->split_parent   = parent;
+    team->split_included = included;
+    team->split_ep       = my_ep;
+    struct M {
+        uint32_t included;
+        uint32_t pad;
+        uint64_t ep;
+    } m{included, 0, my_ep};
+    team->oobr.init(parent->oob);
+    ucc_status_t st = team->oobr.start(&m, sizeof(m));
+    if (st != UCC_OK) {
+        delete team;
+        return st;
+    }
+    team->state = Team::SPLIT_MEMBERS;
+    *new_team   = reinterpret_cast<ucc_team_h>(team);
+    return UCC_OK;
+}
+
 ucc_status_t ucc_team_create_test(ucc_team_h team_h)
 {
     auto *team = reinterpret_cast<Team *>(team_h);
     auto *ctx  = team->ctx;
 
     switch (team->state) {
+    case Team::SPLIT_MEMBERS: {
+        ucc_status_t st = team->oobr.test();
+        if (st == UCC_INPROGRESS) {
+            return UCC_INPROGRESS;
+        }
+        if (st != UCC_OK) {
+            team->state = Team::FAILED;
+            return (team->err = st);
+        }
+        struct M {
+            uint32_t included;
+            uint32_t pad;
+            uint64_t ep;
+        };
+        const M *all = (const M *)team->oobr.data();
+        Team    *par = team->split_parent;
+        /* members ordered by (ep, parent rank) */
+        std::vector<std::pair<uint64_t, uint32_t>> mem;
+        for (uint32_t r = 0; r < par->size; r++) {
+            if (all[r].included) {
+                mem.push_back({all[r].ep, r});
+            }
+        }
+        std::sort(mem.begin(), mem.end());
+        std::vector<uint32_t> members;
+        int                   my_idx = -1;
+        for (size_t i = 0; i < mem.size(); i++) {
+            members.push_back(mem[i].second);
+            if (mem[i].second == par->rank) {
+                my_idx = (int)i;
+            }
+        }
+        if (!team->split_included || my_idx < 0) {
+            /* observer: mirror the members' bootstrap rounds */
+            team->split_oob = subooob_api::make(par, members, -1);
+            team->split_observe_rounds = members.size() >= 2 ? 2 : 0;
+            team->rank = 0;
+            team->size = 0;
+            team->state = Team::SPLIT_OBSERVE;
+            return ucc_team_create_test(team_h);
+        }
+        team->split_oob = subooob_api::make(par, members, my_idx);
+        subooob_api::fill_oob(team->split_oob, &team->oob);
+        team->has_oob = true;
+        team->rank    = (uint32_t)my_idx;
+        team->size    = (uint32_t)members.size();
+        team->id      = (uint16_t)(ctx->next_team_id++);
+        if (team->size == 1) {
+            team->procs.assign(1, ctx->proc);
+            team->team_uid = ctx->proc.ctx_seq ^ 0x5b17;
+            team->state    = Team::TL_CREATE;
+            for (auto &tlc : ctx->tl_ctxs) {
+                TlTeam *tt =
+                    tlc->iface()->team_create(tlc.get(), team);
+                if (tt) {
+                    team->tl_teams.emplace_back(tt);
+                }
+            }
+            return ucc_team_create_test(team_h);
+        }
+        struct R1 {
+            ProcInfo pi;
+            uint64_t uid;
+        } r1;
+        r1.pi = ctx->proc;
+        std::random_device rd;
+        r1.uid = ((uint64_t)rd() << 32) ^ rd() ^
+                 ((uint64_t)ctx->proc.pid << 16) ^ ctx->seq;
+        team->oobr.init(team->oob);
+        ucc_status_t rs = team->oobr.start(&r1, sizeof(r1));
+        if (rs != UCC_OK) {
+            team->state = Team::FAILED;
+            return (team->err = rs);
+        }
+        team->state = Team::ADDR_EXCHANGE;
+        return UCC_INPROGRESS;
+    }
+    case Team::SPLIT_OBSERVE: {
+        ucc_status_t st = subooob_api::observe_tick(
+            team->split_oob, team->split_observe_rounds);
+        if (st == UCC_INPROGRESS) {
+            return UCC_INPROGRESS;
+        }
+        if (st != UCC_OK) {
+            team->state = Team::FAILED;
+            return (team->err = st);
+        }
+        team->state = Team::ACTIVE;
+        return UCC_OK;
+    }
     case Team::ADDR_EXCHANGE: {
         ucc_status_t st = team->oobr.test();
         if (st == UCC_INPROGRESS) {

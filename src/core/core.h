@@ -244,6 +244,8 @@ struct Context {
 /* ------------------------------------------------------------------ Team */
 struct Team {
     enum State {
+        SPLIT_MEMBERS, /* team split: membership exchange round        */
+        SPLIT_OBSERVE, /* team split: non-member observing sub rounds  */
         ADDR_EXCHANGE,
         TL_EXCHANGE,
         TL_CREATE,
@@ -283,6 +285,12 @@ struct Team {
     bool                    want_hier   = false;
     bool                    is_subteam  = false;
     std::vector<uint32_t>   node_ranks, leader_ranks;
+    /* team split (ucc_team_create_from_parent) */
+    Team                   *split_parent   = nullptr;
+    struct SubOob          *split_oob      = nullptr;
+    uint32_t                split_included = 0;
+    uint64_t                split_ep       = 0;
+    int                     split_observe_rounds = 0;
     ~Team();
 
     bool all_same_node() const;
@@ -321,6 +329,16 @@ struct CollRequest {
  * ucc_coll_utils msgsize conventions). */
 size_t coll_args_msgsize(const ucc_coll_args_t &args, uint32_t rank,
                          uint32_t size);
+
+/* SubOob factory/driver (defined in src/cl/hier.cc, shared with team
+ * split) */
+namespace subooob_api {
+SubOob *make(Team *parent, const std::vector<uint32_t> &members,
+             int my_idx);
+ucc_status_t observe_tick(SubOob *so, int target_rounds);
+void         fill_oob(SubOob *so, ucc_oob_coll_t *oob);
+void         free_(SubOob *so);
+} // namespace subooob_api
 
 /* hierarchical composition hooks (src/cl/hier.cc) */
 namespace hier {

@@ -155,3 +155,44 @@ def test_thread_multiple_progress():
         stop.set()
         for t in threads:
             t.join(timeout=10)
+
+
+def test_team_split():
+    """ucc_team_create_from_parent: odd ranks form a sub-team; evens
+    observe. Collective on the sub-team validates membership/reindexing
+    (reference core/ucc_team.c team split)."""
+    import numpy as np
+
+    from ucc_amd import core, dtypes
+    from ucc_amd.testing import LocalJob
+
+    job = LocalJob(5)
+    c = core()
+    n = job.n
+    subs = []
+    for r in range(n):
+        included = 1 if r % 2 == 1 else 0
+        subs.append(c.team_create_from_parent(job.teams[r], r, included))
+    for _ in range(200000):
+        sts = [c.team_create_test(t) for t in subs]
+        assert all(s >= 0 for s in sts)
+        if all(s == c.OK for s in sts):
+            break
+    else:
+        raise TimeoutError("split did not converge")
+    members = [r for r in range(n) if r % 2 == 1]  # [1, 3]
+    arrs = {r: np.full(512, float(r), np.float32) for r in members}
+    outs = {r: np.zeros(512, np.float32) for r in members}
+    reqs = [c.coll_init(subs[r], "allreduce", src=arrs[r].ctypes.data,
+                        dst=outs[r].ctypes.data, count=512,
+                        dt=dtypes.FLOAT32) for r in members]
+    for q in reqs:
+        q.post()
+    for _ in range(2000000):
+        if all(q.test() != c.INPROGRESS for q in reqs):
+            break
+        for ctx in job.ctxs:
+            ctx.progress()
+    expected = np.full(512, float(sum(members)), np.float32)
+    for r in members:
+        np.testing.assert_allclose(outs[r], expected)
