@@ -325,7 +325,8 @@ PYBIND11_MODULE(_core, m)
            int mem_type, uint64_t root, uint64_t flags,
            std::vector<uint64_t> src_counts, std::vector<uint64_t> src_displs,
            std::vector<uint64_t> dst_counts,
-           std::vector<uint64_t> dst_displs, double timeout) {
+           std::vector<uint64_t> dst_displs, double timeout,
+           std::vector<int64_t> active_set, int tag) {
             auto r  = std::make_shared<CoreReq>();
             r->team = team;
             ucc_coll_args_t a{};
@@ -342,6 +343,16 @@ PYBIND11_MODULE(_core, m)
             if (timeout > 0) {
                 a.flags |= UCC_COLL_ARGS_FLAG_TIMEOUT;
                 a.timeout = timeout;
+            }
+            if (active_set.size() == 3) {
+                a.mask |= UCC_COLL_ARGS_FIELD_ACTIVE_SET;
+                a.active_set.start  = active_set[0];
+                a.active_set.stride = active_set[1];
+                a.active_set.size   = active_set[2];
+            }
+            if (tag >= 0) {
+                a.mask |= UCC_COLL_ARGS_FIELD_TAG;
+                a.tag = (uint16_t)tag;
             }
             bool sv = ct == UCC_COLL_TYPE_ALLTOALLV ||
                       ct == UCC_COLL_TYPE_SCATTERV;
@@ -388,7 +399,9 @@ PYBIND11_MODULE(_core, m)
         py::arg("src_displs") = std::vector<uint64_t>(),
         py::arg("dst_counts") = std::vector<uint64_t>(),
         py::arg("dst_displs") = std::vector<uint64_t>(),
-        py::arg("timeout") = 0.0);
+        py::arg("timeout") = 0.0,
+        py::arg("active_set") = std::vector<int64_t>(),
+        py::arg("tag") = -1);
 
     /* ------------------------------------------------------ mem_map */
     m.def("mem_map_export", [](uintptr_t addr, size_t len) {

@@ -121,6 +121,11 @@ class ShmTlTeam final : public TlTeam {
             r.alg_name = "slotted";
             r.init     = [self](const ucc_coll_args_t &args, Team *t,
                             Task **task) -> ucc_status_t {
+                if (args.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) {
+                    /* slot protocol needs every rank: subset colls go
+                     * to tl/tcp (score fallback) */
+                    return UCC_ERR_NOT_SUPPORTED;
+                }
                 *task = new ShmCollTask(t->ctx, self, args);
                 return UCC_OK;
             };

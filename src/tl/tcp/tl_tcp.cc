@@ -599,7 +599,11 @@ class TcpAllreduceTask final : public TcpTask {
     size_t             mask_ = 1;
 };
 
-/* ---- bcast: binomial tree from root */
+/* ---- bcast: binomial tree from root. With an ACTIVE_SET
+ * ({start, stride, size}, reference ucc.h active_set + tl/ucp active-set
+ * bcast), the tree runs over the strided subset only; the wire tag comes
+ * from args.tag (FIELD_TAG) so subset traffic cannot collide with the
+ * team's sequential collectives (non-members post nothing). */
 class TcpBcastTask final : public TcpTask {
   public:
     using TcpTask::TcpTask;
@@ -609,7 +613,43 @@ class TcpBcastTask final : public TcpTask {
         begin();
         bytes_ = a_.src.info.count * ucc_dt_size(a_.src.info.datatype);
         buf_   = (uint8_t *)a_.src.info.buffer;
-        vr_    = (me_ + n_ - a_.root) % n_;
+        if (a_.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) {
+            /* subset colls must not consume the team-wide sequence:
+             * non-members never post, so bumping it would desync tags */
+            tt_->seq_--;
+            set_.clear();
+            int64_t sz = a_.active_set.size;
+            for (int64_t i = 0; i < sz; i++) {
+                uint32_t r = (uint32_t)(((a_.active_set.start +
+                                          i * a_.active_set.stride) %
+                                             (int64_t)n_ +
+                                         (int64_t)n_) %
+                                        (int64_t)n_);
+                set_.push_back(r);
+            }
+            int me_i = -1, root_i = -1;
+            for (size_t i = 0; i < set_.size(); i++) {
+                if (set_[i] == me_) {
+                    me_i = (int)i;
+                }
+                if (set_[i] == (uint32_t)a_.root) {
+                    root_i = (int)i;
+                }
+            }
+            if (me_i < 0 || root_i < 0) {
+                return UCC_ERR_INVALID_PARAM; /* caller not in the set */
+            }
+            as_n_  = (uint32_t)set_.size();
+            vr_    = ((uint32_t)me_i + as_n_ - (uint32_t)root_i) % as_n_;
+            root_i_ = (uint32_t)root_i;
+            /* dedicated sequence space keyed by the user tag */
+            seq_ = 0x2000000000000ull |
+                   ((a_.mask & UCC_COLL_ARGS_FIELD_TAG) ? a_.tag : 0);
+        } else {
+            set_.clear();
+            as_n_ = n_;
+            vr_   = (me_ + n_ - a_.root) % n_;
+        }
         phase_ = 0;
         status = UCC_INPROGRESS;
         return progress_();
@@ -617,6 +657,12 @@ class TcpBcastTask final : public TcpTask {
     ucc_status_t progress() override { return progress_(); }
 
   private:
+    uint32_t to_team(uint32_t v) const
+    {
+        uint32_t idx = (v + root_i_) % as_n_;
+        return set_.empty() ? (v + (uint32_t)a_.root) % n_ : set_[idx];
+    }
+
     ucc_status_t progress_()
     {
         if (phase_ == 0) {
@@ -626,8 +672,7 @@ class TcpBcastTask final : public TcpTask {
                 while (hb * 2 <= vr_) {
                     hb *= 2;
                 }
-                uint32_t parent = (vr_ - hb + a_.root) % n_;
-                recv_from(parent, 0, buf_, bytes_);
+                recv_from(to_team(vr_ - hb), 0, buf_, bytes_);
             }
             phase_ = 1;
         }
@@ -641,9 +686,8 @@ class TcpBcastTask final : public TcpTask {
             while (hb <= vr_) {
                 hb *= 2;
             }
-            for (uint32_t m = hb; vr_ + m < n_; m *= 2) {
-                uint32_t child = (vr_ + m + a_.root) % n_;
-                send_to(child, 0, buf_, bytes_);
+            for (uint32_t m = hb; vr_ + m < as_n_; m *= 2) {
+                send_to(to_team(vr_ + m), 0, buf_, bytes_);
             }
             phase_ = 2;
         }
@@ -656,7 +700,8 @@ class TcpBcastTask final : public TcpTask {
 
     size_t   bytes_ = 0;
     uint8_t *buf_   = nullptr;
-    uint32_t vr_    = 0;
+    uint32_t vr_    = 0, as_n_ = 0, root_i_ = 0;
+    std::vector<uint32_t> set_;
 };
 
 /* ---- barrier / fanin / fanout: binomial fanin to 0 then fanout */
