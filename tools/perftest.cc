@@ -47,6 +47,7 @@ struct Opts {
     int         nprocs = 0; /* 0 = in-process */
     int         nranks = 2;
     bool        persistent = false, inplace = false, check = false;
+    bool        triggered = false;
 };
 
 static ucc_datatype_t dt_from_name(const std::string &s)
@@ -414,14 +415,58 @@ static int run_forked_child(const Opts &o, int rank)
             }
             continue;
         }
+#ifdef UCC_AMD_HAS_HIP
+        hipStream_t tstream = nullptr;
+        ucc_ee_h    ee      = nullptr;
+        if (o.triggered) {
+            hipStreamCreateWithFlags(&tstream, hipStreamNonBlocking);
+            ucc_ee_params_t ep{};
+            ep.ee_type         = UCC_EE_ROCM_STREAM;
+            ep.ee_context      = (void *)tstream;
+            ep.ee_context_size = sizeof(void *);
+            if (ucc_ee_create(r.team, &ep, &ee) != UCC_OK) {
+                fprintf(stderr, "ee create failed\n");
+                ucc_collective_finalize(req);
+                continue;
+            }
+        }
+#endif
+        bool trig_fail = false;
         auto iter = [&]() {
+#ifdef UCC_AMD_HAS_HIP
+            if (o.triggered) {
+                ucc_ev_t ev{};
+                ev.ev_type = UCC_EVENT_COMPUTE_COMPLETE;
+                ev.req     = req;
+                if (ucc_collective_triggered_post(ee, &ev) != UCC_OK) {
+                    trig_fail = true;
+                    return;
+                }
+                hipStreamSynchronize(tstream);
+                return;
+            }
+#endif
             ucc_collective_post(req);
             while (ucc_collective_test(req) == UCC_INPROGRESS) {
                 ucc_context_progress(r.ctx);
             }
         };
-        for (int i = 0; i < o.warmup; i++) {
+        for (int i = 0; i < o.warmup && !trig_fail; i++) {
             iter();
+        }
+        if (trig_fail) {
+            if (rank == 0) {
+                printf("%12zu        - triggered unsupported at this "
+                       "size\n", bytes);
+            }
+            ucc_collective_finalize(req);
+#ifdef UCC_AMD_HAS_HIP
+            if (ee) { ucc_ee_destroy(ee); ee = nullptr; }
+            if (tstream) { hipStreamDestroy(tstream); tstream = nullptr; }
+#endif
+            g_shm_oob.max_double(0.0);
+            g_shm_oob.max_double(0.0);
+            continue;
         }
 #ifdef UCC_AMD_HAS_HIP
         if (o.mem == "cuda") {
@@ -440,6 +485,14 @@ static int run_forked_child(const Opts &o, int rank)
 #endif
         double t  = g_shm_oob.max_double(now_s() - t0) / iters;
         ucc_collective_finalize(req);
+#ifdef UCC_AMD_HAS_HIP
+        if (ee) {
+            ucc_ee_destroy(ee);
+        }
+        if (tstream) {
+            hipStreamDestroy(tstream);
+        }
+#endif
         if (rank == 0) {
             double algbw = bytes / t / 1e9;
             double busbw = algbw * busbw_factor(ct, o.nprocs);
@@ -582,7 +635,7 @@ int main(int argc, char **argv)
 {
     Opts o;
     int  c;
-    while ((c = getopt(argc, argv, "c:b:e:n:w:m:d:o:p:j:FiCh")) != -1) {
+    while ((c = getopt(argc, argv, "c:b:e:n:w:m:d:o:p:j:FiCTh")) != -1) {
         switch (c) {
         case 'c': o.coll = optarg; break;
         case 'b': o.min_b = strtoull(optarg, nullptr, 0); break;
@@ -597,12 +650,13 @@ int main(int argc, char **argv)
         case 'F': o.persistent = true; break;
         case 'i': o.inplace = true; break;
         case 'C': o.check = true; break;
+        case 'T': o.triggered = true; o.persistent = true; break;
         case 'h':
         default:
             printf("ucc_perftest [-c coll] [-b min] [-e max] [-n iters] "
                    "[-w warmup] [-m host|cuda] [-d dtype] [-o op] "
                    "[-p nprocs(fork)] [-j inproc_ranks] [-F persistent] "
-                   "[-i inplace]\n");
+                   "[-i inplace] [-T triggered]\n");
             return c == 'h' ? 0 : 1;
         }
     }
