@@ -118,3 +118,14 @@ def test_multiproc_gpu_colls():
     if not torch.cuda.is_available():
         pytest.skip("no GPU")
     _run(_worker_gpu)
+
+
+def _worker_host_tcp(rank, world, port, q):
+    os.environ["UCC_TL_SHM_ENABLE"] = "0"  # force the tcp transport
+    _worker_host(rank, world, port, q)
+
+
+def test_multiproc_host_tcp():
+    """Real multi-process host collectives over the TCP transport
+    (inter-node path exercised across process boundaries)."""
+    _run(_worker_host_tcp, world=3)
