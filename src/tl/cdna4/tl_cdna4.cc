@@ -1487,7 +1487,7 @@ class Cdna4Tl final : public Tl {
                     "rotating in-flight collective slots");
         cfg.declare("TL_CDNA4", "PERSISTENT_SLOTS", "2",
                     "dedicated slots for persistent triggered colls");
-        cfg.declare("TL_CDNA4", "CHUNK_SIZE", "8m",
+        cfg.declare("TL_CDNA4", "CHUNK_SIZE", "32m",
                     "staging fragment bytes per slot area");
         cfg.declare("TL_CDNA4", "FUSED_MAX", "512k",
                     "max msg bytes for the fused single-kernel allreduce");
@@ -1516,7 +1516,7 @@ class Cdna4Tl final : public Tl {
         if (c.nslots + c.npers > 8) { /* flags layout: 8 slots max */
             c.npers = c.nslots < 8 ? 8 - c.nslots : 0;
         }
-        c.chunk     = cfg.get_size("TL_CDNA4", "CHUNK_SIZE", 8 * 1024 * 1024);
+        c.chunk     = cfg.get_size("TL_CDNA4", "CHUNK_SIZE", 32 * 1024 * 1024);
         c.fused_max = cfg.get_size("TL_CDNA4", "FUSED_MAX", 512 * 1024);
         return new Cdna4TlTeam(tlc, team, c);
     }

@@ -633,6 +633,7 @@ static int run_inproc(const Opts &o)
 
 int main(int argc, char **argv)
 {
+    setvbuf(stdout, nullptr, _IOLBF, 0); /* line-buffer under pipes */
     Opts o;
     int  c;
     while ((c = getopt(argc, argv, "c:b:e:n:w:m:d:o:p:j:FiCTh")) != -1) {
