@@ -395,6 +395,10 @@ static int run_forked_child(const Opts &o, int rank)
     }
     ucc_coll_type_t ct = coll_from_name(o.coll);
     r.bufs.alloc(std::max(o.max_b * 2, (size_t)4096), o.mem == "cuda");
+    /* align ranks before the first collective: team create time is
+     * rank-skewed (e.g. a several-second vendor-lib fallback on one
+     * rank) and triggered posts have no host-side entry gate */
+    g_shm_oob.max_double(0.0);
     if (rank == 0) {
         printf("# ucc_perftest  coll=%s mem=%s dt=%s op=%s procs=%d%s%s\n",
                o.coll.c_str(), o.mem.c_str(), o.dtype.c_str(),
