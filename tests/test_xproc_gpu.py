@@ -31,3 +31,22 @@ def test_xproc_device_colls():
     sys.stderr.write(proc.stderr[-3000:])
     assert proc.returncode == 0
     assert proc.stdout.count("XPROC_OK") == 2
+
+
+def test_xproc_stress():
+    """Mixed-collective cross-process stress (slot rotation, parity and
+    counter continuity across fused/gated/staged paths)."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    env = dict(os.environ)
+    env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr=127.0.0.1",
+         "--master-port=29527", os.path.join(REPO, "tests",
+                                             "stress_worker.py"), "60"],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=600)
+    sys.stdout.write(proc.stdout[-2000:])
+    sys.stderr.write(proc.stderr[-2000:])
+    assert proc.returncode == 0
+    assert proc.stdout.count("STRESS_OK") == 2
