@@ -42,7 +42,32 @@ __device__ __forceinline__ uint16_t d_f_to_bf16(float f)
     return (uint16_t)((c.u + r) >> 16);
 }
 
-/* fp8 (OCP): decode/encode via f32, same tables as the host reference. */
+/* fp8 (OCP): gfx950 has native OCP fp8<->f32 converts (VOP1/VOP3
+ * v_cvt_f32_fp8 / v_cvt_pk_fp8_f32) — use them on device; the bit-math
+ * path below stays as the portable reference (and matches the host
+ * codec bit-for-bit on finite values). */
+#if defined(__gfx950__)
+#define UCC_NATIVE_FP8 1
+#endif
+
+#ifdef UCC_NATIVE_FP8
+__device__ __forceinline__ float d_e4m3_to_f(uint8_t v)
+{
+    return __builtin_amdgcn_cvt_f32_fp8((int)v, 0);
+}
+__device__ __forceinline__ uint8_t d_f_to_e4m3(float f)
+{
+    return (uint8_t)__builtin_amdgcn_cvt_pk_fp8_f32(f, f, 0, false);
+}
+__device__ __forceinline__ float d_e5m2_to_f(uint8_t v)
+{
+    return __builtin_amdgcn_cvt_f32_bf8((int)v, 0);
+}
+__device__ __forceinline__ uint8_t d_f_to_e5m2(float f)
+{
+    return (uint8_t)__builtin_amdgcn_cvt_pk_bf8_f32(f, f, 0, false);
+}
+#else
 __device__ __forceinline__ float d_e4m3_to_f(uint8_t v)
 {
     uint32_t sign = v >> 7, exp = (v >> 3) & 0xf, man = v & 7;
@@ -138,6 +163,7 @@ __device__ __forceinline__ uint8_t d_f_to_e5m2(float f)
     }
     return sign | (uint8_t)(exp << 2) | (uint8_t)mi;
 }
+#endif /* UCC_NATIVE_FP8 */
 
 /* type traits: storage type T <-> accumulator A */
 template <typename T> struct Cvt {
