@@ -176,3 +176,50 @@ def test_multiproc_ddp_bucket_allreduce():
     """SURVEY §2.9 DP workload: bucketed persistent gradient allreduce
     (ucc_amd.parallel.ddp) matches gloo averaging over 2 processes."""
     _run(_worker_ddp)
+
+
+def _worker_moe(rank, world, port, q):
+    try:
+        import torch
+        import torch.distributed as dist
+
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from ucc_amd.parallel import Communicator
+        from ucc_amd.parallel.moe import TokenExchanger
+
+        comm = Communicator()
+        ex = TokenExchanger(comm)
+        hidden = 16
+        # skewed routing: rank r sends (r+1)*(d+2) tokens to rank d
+        send_counts = [(rank + 1) * (d + 2) for d in range(world)]
+        torch.manual_seed(300 + rank)
+        tokens = torch.randn(sum(send_counts), hidden)
+        received, recv_counts = ex.dispatch(tokens, send_counts)
+        # expected recv from rank s: (s+1)*(rank+2) rows
+        assert recv_counts == [(s + 1) * (rank + 2) for s in range(world)]
+        # verify contents: re-generate each source's tensor
+        off_r = 0
+        for s in range(world):
+            sc = [(s + 1) * (d + 2) for d in range(world)]
+            torch.manual_seed(300 + s)
+            stok = torch.randn(sum(sc), hidden)
+            soff = sum(sc[:rank])
+            exp = stok[soff:soff + sc[rank]]
+            got = received[off_r:off_r + recv_counts[s]]
+            assert torch.equal(got, exp), (s, rank)
+            off_r += recv_counts[s]
+        # combine round-trips the data
+        back = ex.combine(received, recv_counts, send_counts)
+        assert torch.equal(back, tokens)
+        q.put((rank, "ok"))
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"fail: {e!r}"))
+
+
+def test_multiproc_moe_token_exchange():
+    """SURVEY §2.9 EP workload: skewed MoE token dispatch/combine over
+    alltoallv across processes."""
+    _run(_worker_moe, world=3)
