@@ -368,6 +368,128 @@ static ucc_coll_args_t make_args(const Opts &o, Rank &r, size_t bytes,
     return a;
 }
 
+/* -C validation: run one iteration with a known fp32 pattern and check
+ * the result (allreduce/allgather/bcast/reduce_scatter). Returns false
+ * on mismatch. Works for host and device memory (staged via host). */
+static bool validate_once(const Opts &o, Rank &r, ucc_coll_req_h req,
+                          ucc_coll_type_t ct, size_t bytes, int rank,
+                          int nranks)
+{
+    if (o.dtype != "float32" || o.inplace) {
+        return true; /* only the fp32 non-inplace patterns are checked */
+    }
+    size_t count = bytes / 4;
+    if (count == 0) {
+        return true;
+    }
+    std::vector<float> h(count), out(count, 0.0f);
+    switch (ct) {
+    case UCC_COLL_TYPE_ALLREDUCE:
+        for (size_t i = 0; i < count; i++) {
+            h[i] = (float)(rank + 1) + 0.25f * (float)(i % 7);
+        }
+        break;
+    case UCC_COLL_TYPE_ALLGATHER:
+    case UCC_COLL_TYPE_REDUCE_SCATTER:
+        for (size_t i = 0; i < count; i++) {
+            h[i] = (float)(rank + 1) + 0.5f * (float)(i % 5);
+        }
+        break;
+    case UCC_COLL_TYPE_BCAST:
+        for (size_t i = 0; i < count; i++) {
+            h[i] = rank == 0 ? (float)(i % 97) : -1.0f;
+        }
+        break;
+    default:
+        return true;
+    }
+#ifdef UCC_AMD_HAS_HIP
+    if (o.mem == "cuda") {
+        hipMemcpy(r.bufs.src, h.data(), bytes, hipMemcpyHostToDevice);
+        hipMemset(r.bufs.dst, 0, bytes);
+        hipDeviceSynchronize();
+    } else
+#endif
+    {
+        memcpy(r.bufs.src, h.data(), bytes);
+        memset(r.bufs.dst, 0, bytes);
+    }
+    if (ct == UCC_COLL_TYPE_BCAST) {
+#ifdef UCC_AMD_HAS_HIP
+        if (o.mem == "cuda") {
+            hipMemcpy(r.bufs.src, h.data(), bytes,
+                      hipMemcpyHostToDevice);
+        }
+#endif
+    }
+    ucc_collective_post(req);
+    while (ucc_collective_test(req) == UCC_INPROGRESS) {
+        ucc_context_progress(r.ctx);
+    }
+#ifdef UCC_AMD_HAS_HIP
+    if (o.mem == "cuda") {
+        hipDeviceSynchronize();
+        hipMemcpy(out.data(),
+                  ct == UCC_COLL_TYPE_BCAST ? r.bufs.src : r.bufs.dst,
+                  bytes, hipMemcpyDeviceToHost);
+    } else
+#endif
+    {
+        memcpy(out.data(),
+               ct == UCC_COLL_TYPE_BCAST ? r.bufs.src : r.bufs.dst,
+               bytes);
+    }
+    auto expect_allreduce = [&](size_t i) {
+        float s = 0;
+        for (int k = 0; k < nranks; k++) {
+            s += (float)(k + 1) + 0.25f * (float)(i % 7);
+        }
+        return s;
+    };
+    size_t bad = 0;
+    if (ct == UCC_COLL_TYPE_ALLREDUCE) {
+        for (size_t i = 0; i < count; i++) {
+            if (out[i] != expect_allreduce(i)) {
+                bad++;
+            }
+        }
+    } else if (ct == UCC_COLL_TYPE_ALLGATHER) {
+        size_t per = count / nranks;
+        for (size_t i = 0; i < per * nranks; i++) {
+            int    src = (int)(i / per);
+            size_t li  = i % per;
+            float  e = (float)(src + 1) + 0.5f * (float)(li % 5);
+            if (out[i] != e) {
+                bad++;
+            }
+        }
+    } else if (ct == UCC_COLL_TYPE_REDUCE_SCATTER) {
+        size_t per = count / nranks;
+        for (size_t i = 0; i < per; i++) {
+            size_t gi = (size_t)rank * per + i;
+            float  e  = 0;
+            for (int k = 0; k < nranks; k++) {
+                e += (float)(k + 1) + 0.5f * (float)(gi % 5);
+            }
+            if (out[i] != e) {
+                bad++;
+            }
+        }
+    } else if (ct == UCC_COLL_TYPE_BCAST) {
+        for (size_t i = 0; i < count; i++) {
+            if (out[i] != (float)(i % 97)) {
+                bad++;
+            }
+        }
+    }
+    if (bad) {
+        fprintf(stderr, "[rank %d] VALIDATION FAILED: %zu/%zu bad at %zu "
+                        "bytes\n", rank, bad, count, bytes);
+        return false;
+    }
+    return true;
+}
+
 /* ---------------------------------------------------------- fork driver */
 static int run_forked_child(const Opts &o, int rank)
 {
@@ -455,6 +577,13 @@ static int run_forked_child(const Opts &o, int rank)
                 ucc_context_progress(r.ctx);
             }
         };
+        if (o.check &&
+            !validate_once(o, r, req, ct, bytes, rank, o.nprocs)) {
+            ucc_collective_finalize(req);
+            g_shm_oob.max_double(0.0);
+            g_shm_oob.max_double(0.0);
+            exit(2);
+        }
         for (int i = 0; i < o.warmup && !trig_fail; i++) {
             iter();
         }
