@@ -46,6 +46,7 @@ struct Opts {
     std::string op    = "sum";
     int         nprocs = 0; /* 0 = in-process */
     int         nranks = 2;
+    bool        skew = false; /* alltoallv: MoE-like skewed matrix */
     bool        persistent = false, inplace = false, check = false;
     bool        triggered = false;
 };
@@ -291,9 +292,10 @@ static void setup_rank(Rank &r, int rank, int nranks, bool forked)
 static ucc_coll_args_t make_args(const Opts &o, Rank &r, size_t bytes,
                                  int rank, int nranks,
                                  std::vector<uint64_t> &cnts,
-                                 std::vector<uint64_t> &dsps)
+                                 std::vector<uint64_t> &dsps,
+                                 std::vector<uint64_t> &rcnts,
+                                 std::vector<uint64_t> &rdsps)
 {
-    (void)rank;
     ucc_coll_type_t    ct = coll_from_name(o.coll);
     ucc_datatype_t     dt = dt_from_name(o.dtype);
     ucc_reduction_op_t op = op_from_name(o.op);
@@ -345,19 +347,53 @@ static ucc_coll_args_t make_args(const Opts &o, Rank &r, size_t bytes,
         break;
     }
     case UCC_COLL_TYPE_ALLTOALLV: {
-        cnts.assign(nranks, count / nranks);
+        /* -M: skewed per-peer counts (traffic-matrix generator role,
+         * reference tools/perf/generator): rank r sends to peer d a
+         * share proportional to (r+1)*(d+1), normalized to count */
+        cnts.resize(nranks);
         dsps.resize(nranks);
-        for (int i = 0; i < nranks; i++) {
-            dsps[i] = (uint64_t)i * (count / nranks);
+        if (o.skew) {
+            size_t denom = 0;
+            for (int d = 0; d < nranks; d++) {
+                denom += (size_t)(rank + 1) * (d + 1);
+            }
+            size_t off2 = 0;
+            for (int d = 0; d < nranks; d++) {
+                cnts[d] = count * ((size_t)(rank + 1) * (d + 1)) / denom;
+                dsps[d] = off2;
+                off2 += cnts[d];
+            }
+        } else {
+            for (int i = 0; i < nranks; i++) {
+                cnts[i] = count / nranks;
+                dsps[i] = (uint64_t)i * (count / nranks);
+            }
         }
         a.src.info_v.buffer        = r.bufs.src;
         a.src.info_v.counts        = cnts.data();
         a.src.info_v.displacements = dsps.data();
         a.src.info_v.datatype      = dt;
         a.src.info_v.mem_type      = mt;
+        rcnts.resize(nranks);
+        rdsps.resize(nranks);
+        if (o.skew) {
+            size_t off2 = 0;
+            for (int s = 0; s < nranks; s++) {
+                size_t denom = 0;
+                for (int d = 0; d < nranks; d++) {
+                    denom += (size_t)(s + 1) * (d + 1);
+                }
+                rcnts[s] = count * ((size_t)(s + 1) * (rank + 1)) / denom;
+                rdsps[s] = off2;
+                off2 += rcnts[s];
+            }
+        } else {
+            rcnts = cnts;
+            rdsps = dsps;
+        }
         a.dst.info_v.buffer        = r.bufs.dst;
-        a.dst.info_v.counts        = cnts.data();
-        a.dst.info_v.displacements = dsps.data();
+        a.dst.info_v.counts        = rcnts.data();
+        a.dst.info_v.displacements = rdsps.data();
         a.dst.info_v.datatype      = dt;
         a.dst.info_v.mem_type      = mt;
         break;
@@ -531,9 +567,9 @@ static int run_forked_child(const Opts &o, int rank)
     }
     for (size_t bytes = o.min_b; bytes <= o.max_b; bytes *= 2) {
         int iters = bytes >= (64 << 10) ? o.large_iters : o.iters;
-        std::vector<uint64_t> cnts, dsps;
+        std::vector<uint64_t> cnts, dsps, rcnts, rdsps;
         ucc_coll_args_t a = make_args(o, r, bytes, rank, o.nprocs, cnts,
-                                      dsps);
+                                      dsps, rcnts, rdsps);
         ucc_coll_req_h  req;
         if (ucc_collective_init(&a, &req, r.team) != UCC_OK) {
             if (rank == 0) {
@@ -704,12 +740,14 @@ static int run_inproc(const Opts &o)
            "algbw_GBps", "busbw_GBps");
     for (size_t bytes = o.min_b; bytes <= o.max_b; bytes *= 2) {
         int iters = bytes >= (64 << 10) ? o.large_iters : o.iters;
-        std::vector<std::vector<uint64_t>> cnts(n), dsps(n);
+        std::vector<std::vector<uint64_t>> cnts(n), dsps(n), rcnts(n),
+            rdsps(n);
         std::vector<ucc_coll_req_h>        reqs(n);
         bool                               ok = true;
         for (int i = 0; i < n; i++) {
-            ucc_coll_args_t a =
-                make_args(o, ranks[i], bytes, i, n, cnts[i], dsps[i]);
+            ucc_coll_args_t a = make_args(o, ranks[i], bytes, i, n,
+                                          cnts[i], dsps[i], rcnts[i],
+                                          rdsps[i]);
             if (ucc_collective_init(&a, &reqs[i], ranks[i].team) !=
                 UCC_OK) {
                 ok = false;
@@ -769,7 +807,7 @@ int main(int argc, char **argv)
     setvbuf(stdout, nullptr, _IOLBF, 0); /* line-buffer under pipes */
     Opts o;
     int  c;
-    while ((c = getopt(argc, argv, "c:b:e:n:w:m:d:o:p:j:FiCTh")) != -1) {
+    while ((c = getopt(argc, argv, "c:b:e:n:w:m:d:o:p:j:FiCTMh")) != -1) {
         switch (c) {
         case 'c': o.coll = optarg; break;
         case 'b': o.min_b = strtoull(optarg, nullptr, 0); break;
@@ -785,12 +823,13 @@ int main(int argc, char **argv)
         case 'i': o.inplace = true; break;
         case 'C': o.check = true; break;
         case 'T': o.triggered = true; o.persistent = true; break;
+        case 'M': o.skew = true; break;
         case 'h':
         default:
             printf("ucc_perftest [-c coll] [-b min] [-e max] [-n iters] "
                    "[-w warmup] [-m host|cuda] [-d dtype] [-o op] "
                    "[-p nprocs(fork)] [-j inproc_ranks] [-F persistent] "
-                   "[-i inplace] [-T triggered]\n");
+                   "[-i inplace] [-T triggered] [-M skewed-alltoallv]\n");
             return c == 'h' ? 0 : 1;
         }
     }
