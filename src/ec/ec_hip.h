@@ -77,7 +77,7 @@ ucc_status_t fused_allreduce(const FusedArgs &a, hipStream_t stream);
 constexpr int kMaxGraphBlocks = 64;
 constexpr int kStageCntBase   = 8 * kMaxRanks;      /* u64 idx of stage cnts */
 constexpr int kGraphCntBase   = 128;                /* u64 idx of blk cnts   */
-constexpr int kFlagsBytes     = 8192;               /* flags alloc size      */
+constexpr int kFlagsBytes     = 32768;              /* flags alloc size      */
 
 struct GraphFusedArgs {
     const void *src;
@@ -121,9 +121,13 @@ bool op_supported(ucc_datatype_t dt, ucc_reduction_op_t op);
  *   kGatedCntBase + (phase * kGatedSlots + slot) * 2 + parity
  *   phase: 0=stage 1=reduce 2=gather; slot < kGatedSlots; parity = frag&1.
  */
-constexpr int kGatedBlocks  = 64;
-constexpr int kGatedSlots   = 8;
-constexpr int kGatedCntBase = 704; /* after graph counters (128..639) */
+constexpr int kGatedBlocks    = 64;
+constexpr int kGatedSlots     = 8;
+constexpr int kGatedCntBase   = 704;  /* after graph counters (128..639) */
+/* per-(slot,parity,block) launch counters for the graph-replayable gated
+ * mode: u64 idx kGatedGraphBase + ((slot*2)+parity)*kGatedBlocks + block
+ * -> 1024..2047 (16 KiB) */
+constexpr int kGatedGraphBase = 1024;
 
 struct GatedArgs {
     /* data */
@@ -156,6 +160,16 @@ struct GatedArgs {
     uint64_t    t_sw_reduce, t_sw_gather, t_prev_gather, t_stage,
                 t_gather_wait;
     int         gw_phase;
+    /* Graph-replayable mode (persistent colls on a dedicated slot): when
+     * derive != 0, targets are computed ON DEVICE from per-block launch
+     * counters at kGatedGraphBase (one counter per (slot,parity,block),
+     * incremented by EVERY gated kernel of the pattern): this block's
+     * count v gives the iteration u = (v-1)/pp + 1 and the host targets
+     * above are ignored — so one capture replays forever. pp = kernels
+     * per (slot,parity) iteration (3 for allreduce, 2 for rs/ag/a2a);
+     * has_reduce/has_gather describe the pattern so waits on phases that
+     * never launch resolve to zero. */
+    int         derive, pp, has_reduce, has_gather;
     /* per-dest cell staging (alltoall): my_in[c_dst_off[k]] <-
      * src[c_src_off[k]], c_len[k] bytes; 0 = contiguous stage of len   */
     int         n_cells;

@@ -673,6 +673,37 @@ __device__ __forceinline__ int gated_idx(int phase, int slot, int parity)
     return kGatedCntBase + (phase * kGatedSlots + slot) * 2 + parity;
 }
 
+/* graph mode: this block's launch ordinal within the (slot,parity)
+ * pattern -> fills the host-target fields of `a` in place. Every gated
+ * kernel of the pattern increments the same per-block counter, so with
+ * pp kernels per iteration, counter value v maps to iteration
+ * u = (v-1)/pp + 1 on every block of the same launch (fixed grid). */
+__device__ __forceinline__ void gated_derive(GatedArgs &a, int my_ord)
+{
+    if (!a.derive) {
+        return;
+    }
+    uint64_t *cnt = a.local_flags + kGatedGraphBase +
+                    ((uint64_t)a.slot * 2 + a.parity) * kGatedBlocks +
+                    blockIdx.x;
+    __shared__ uint64_t s_u;
+    if (threadIdx.x == 0) {
+        uint64_t v = __hip_atomic_fetch_add(cnt, 1, __ATOMIC_RELAXED,
+                                            __HIP_MEMORY_SCOPE_AGENT) +
+                     1;
+        s_u = (v - 1) / (uint64_t)a.pp + 1;
+    }
+    __syncthreads();
+    const uint64_t u = s_u;
+    const uint64_t B = (uint64_t)kGatedBlocks;
+    a.t_sw_reduce   = a.has_reduce ? (u - 1) * B : 0;
+    a.t_sw_gather   = a.has_gather ? (u - 1) * B : 0;
+    a.t_prev_gather = a.has_gather ? (u - 1) * B : 0;
+    a.t_stage       = u * B;
+    a.t_gather_wait = u * B;
+    (void)my_ord;
+}
+
 /* all blocks wait until every rank's counter[idx] >= target; on spin
  * timeout ALL threads leave together (no thread may return while others
  * sit at __syncthreads — that would hang the block forever). */
@@ -723,6 +754,7 @@ __device__ __forceinline__ void gated_signal(const GatedArgs &a, int phase)
 
 __global__ void k_staged_stage(GatedArgs a)
 {
+    gated_derive(a, 0);
     if (!gated_wait(a, 1, a.t_sw_reduce) ||
         !gated_wait(a, 2, a.t_sw_gather)) { /* in-area reuse */
         return;
@@ -745,6 +777,7 @@ template <typename T, int OP, int VEC>
 __global__ void k_staged_reduce(GatedArgs a)
 {
     using A = typename Cvt<T>::A;
+    gated_derive(a, 1);
     if (!gated_wait(a, 0, a.t_stage) ||
         !gated_wait(a, 2, a.t_prev_gather)) {
         return;
@@ -793,6 +826,7 @@ __global__ void k_staged_reduce(GatedArgs a)
 
 __global__ void k_staged_gather(GatedArgs a)
 {
+    gated_derive(a, 2);
     if (!gated_wait(a, a.gw_phase, a.t_gather_wait)) {
         return;
     }

@@ -1219,6 +1219,54 @@ class GatedCollTask final : public Cdna4Task {
         if (slot_ >= (uint32_t)ec_hip::kGatedSlots) {
             return UCC_ERR_NOT_SUPPORTED;
         }
+        ucc_status_t st = setup_coll();
+        if (st != UCC_OK) {
+            return st;
+        }
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress();
+    }
+
+    /* Stream-triggered / hipGraph-capturable post: kernels derive their
+     * iteration on device (GatedArgs.derive) on a DEDICATED slot, so one
+     * capture replays as a fresh collective each time — the large-message
+     * counterpart of the fused graph kernel (docs/GATED_PIPELINE.md). */
+    ucc_status_t triggered_post(void *ee_stream) override
+    {
+        if (pslot_ < 0) {
+            pslot_ = tt_->alloc_pslot();
+            if (pslot_ < 0 || pslot_ >= ec_hip::kGatedSlots) {
+                ucc_error("no persistent slot for triggered gated coll");
+                return UCC_ERR_NO_RESOURCE;
+            }
+            me_ = tt_->team_->rank;
+            n_  = tt_->team_->size;
+            ucc_status_t st = setup_coll();
+            if (st != UCC_OK) {
+                return st;
+            }
+        }
+        slot_ = (uint32_t)pslot_;
+        ucc_status_t st = enqueue_frags((hipStream_t)ee_stream,
+                                        (hipStream_t)ee_stream, true);
+        if (st != UCC_OK) {
+            return st;
+        }
+        status = UCC_OK; /* stream-ordered completion */
+        return UCC_OK;
+    }
+
+    ~GatedCollTask() override
+    {
+        if (pslot_ >= 0) {
+            tt_->free_pslot(pslot_);
+        }
+    }
+
+  private:
+    ucc_status_t setup_coll()
+    {
         const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
         ct_     = a_.coll_type;
         op_     = a_.op;
@@ -1286,18 +1334,17 @@ class GatedCollTask final : public Cdna4Task {
         if (nfrags_ == 0) {
             nfrags_ = 1;
         }
-        phase_ = 0;
-        status = UCC_INPROGRESS;
-        return progress();
+        return UCC_OK;
     }
 
+  public:
     ucc_status_t progress() override
     {
         if (phase_ == 0) { /* entry: previous use of this slot closed */
             if (!all_ge(0)) {
                 return UCC_INPROGRESS;
             }
-            ucc_status_t st = enqueue_all();
+            ucc_status_t st = enqueue_frags(copy_s(), comp(), false);
             if (st != UCC_OK) {
                 return st;
             }
@@ -1322,7 +1369,8 @@ class GatedCollTask final : public Cdna4Task {
     }
 
   private:
-    ucc_status_t enqueue_all()
+    ucc_status_t enqueue_frags(hipStream_t stage_s, hipStream_t comp_s,
+                               bool derive)
     {
         auto &L          = tt_->gated_launch_;
         const uint64_t B = (uint64_t)ec_hip::kGatedBlocks;
@@ -1349,6 +1397,13 @@ class GatedCollTask final : public Cdna4Task {
             ga.dt          = dt_;
             ga.op          = op_;
             ga.alpha       = alpha_;
+            if (derive) {
+                ga.derive     = 1;
+                ga.pp         = ct_ == UCC_COLL_TYPE_ALLREDUCE ? 3 : 2;
+                ga.has_reduce = ct_ == UCC_COLL_TYPE_ALLREDUCE ||
+                                ct_ == UCC_COLL_TYPE_REDUCE_SCATTER;
+                ga.has_gather = ct_ != UCC_COLL_TYPE_REDUCE_SCATTER;
+            }
             ucc_status_t st = UCC_OK;
             switch (ct_) {
             case UCC_COLL_TYPE_ALLREDUCE: {
@@ -1369,15 +1424,15 @@ class GatedCollTask final : public Cdna4Task {
                 ga.t_stage       = (L[0][slot_][p] + 1) * B;
                 ga.gw_phase      = 1;
                 ga.t_gather_wait = (L[1][slot_][p] + 1) * B;
-                st = ec_hip::staged_stage(ga, copy_s());
-                L[0][slot_][p]++;
+                st = ec_hip::staged_stage(ga, stage_s);
+                if (!derive) { L[0][slot_][p]++; }
                 if (st == UCC_OK) {
-                    st = ec_hip::staged_reduce(ga, comp());
-                    L[1][slot_][p]++;
+                    st = ec_hip::staged_reduce(ga, comp_s);
+                    if (!derive) { L[1][slot_][p]++; }
                 }
                 if (st == UCC_OK) {
-                    st = ec_hip::staged_gather(ga, comp());
-                    L[2][slot_][p]++;
+                    st = ec_hip::staged_gather(ga, comp_s);
+                    if (!derive) { L[2][slot_][p]++; }
                 }
                 break;
             }
@@ -1396,11 +1451,11 @@ class GatedCollTask final : public Cdna4Task {
                 ga.t_sw_gather   = L[2][slot_][p] * B;
                 ga.t_prev_gather = 0;
                 ga.t_stage       = (L[0][slot_][p] + 1) * B;
-                st = ec_hip::staged_stage(ga, copy_s());
-                L[0][slot_][p]++;
+                st = ec_hip::staged_stage(ga, stage_s);
+                if (!derive) { L[0][slot_][p]++; }
                 if (st == UCC_OK) {
-                    st = ec_hip::staged_reduce(ga, comp());
-                    L[1][slot_][p]++;
+                    st = ec_hip::staged_reduce(ga, comp_s);
+                    if (!derive) { L[1][slot_][p]++; }
                 }
                 break;
             }
@@ -1418,11 +1473,11 @@ class GatedCollTask final : public Cdna4Task {
                 ga.t_sw_gather   = L[2][slot_][p] * B;
                 ga.gw_phase      = 0;
                 ga.t_gather_wait = (L[0][slot_][p] + 1) * B;
-                st = ec_hip::staged_stage(ga, copy_s());
-                L[0][slot_][p]++;
+                st = ec_hip::staged_stage(ga, stage_s);
+                if (!derive) { L[0][slot_][p]++; }
                 if (st == UCC_OK) {
-                    st = ec_hip::staged_gather(ga, comp());
-                    L[2][slot_][p]++;
+                    st = ec_hip::staged_gather(ga, comp_s);
+                    if (!derive) { L[2][slot_][p]++; }
                 }
                 break;
             }
@@ -1444,11 +1499,11 @@ class GatedCollTask final : public Cdna4Task {
                 ga.t_sw_gather   = L[2][slot_][p] * B;
                 ga.gw_phase      = 0;
                 ga.t_gather_wait = (L[0][slot_][p] + 1) * B;
-                st = ec_hip::staged_stage(ga, copy_s());
-                L[0][slot_][p]++;
+                st = ec_hip::staged_stage(ga, stage_s);
+                if (!derive) { L[0][slot_][p]++; }
                 if (st == UCC_OK) {
-                    st = ec_hip::staged_gather(ga, comp());
-                    L[2][slot_][p]++;
+                    st = ec_hip::staged_gather(ga, comp_s);
+                    if (!derive) { L[2][slot_][p]++; }
                 }
                 break;
             }
@@ -1470,6 +1525,7 @@ class GatedCollTask final : public Cdna4Task {
     size_t             gran_ = 0, cell_ = 0, out_b_ = 0;
     const uint8_t     *sbuf_ = nullptr;
     uint8_t           *dbuf_ = nullptr;
+    int                pslot_ = -1;
 };
 
 /* ------------------------------------------------------------ scoring  */

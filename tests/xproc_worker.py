@@ -239,6 +239,39 @@ def main():
         torch.testing.assert_close(dst.cpu(), exp, rtol=1e-5, atol=1e-4)
     results.append("hipgraph_replay")
 
+    # 5b. hipGraph capture + replay of a LARGE (gated pipeline) persistent
+    # allreduce: kernels derive their iteration on device
+    count = 12_000_000  # ~48 MB fp32 -> multi-fragment gated path
+    gfull = torch.randn(world, count, generator=g0)
+    gsrc = gfull[rank].cuda()
+    gdst = torch.zeros(count, device="cuda")
+    rg = c.coll_init(team, "allreduce", src=gsrc.data_ptr(),
+                     dst=gdst.data_ptr(), count=count, dt=dtypes.FLOAT32,
+                     mem_type=dtypes.MEM_CUDA, flags=c.FLAG_PERSISTENT)
+    # warm (non-captured triggered posts)
+    for it in range(2):
+        torch.cuda.synchronize()
+        dist.barrier()
+        c.triggered_post(ee, rg)
+        s.synchronize()
+        torch.testing.assert_close(gdst.cpu(), gfull.sum(0), rtol=1e-5,
+                                   atol=1e-4)
+    g2 = torch.cuda.CUDAGraph()
+    dist.barrier()
+    with torch.cuda.graph(g2, stream=s):
+        c.triggered_post(ee, rg)
+    dist.barrier()
+    for it in range(3):
+        gsrc.copy_(gfull[rank] * (it + 1))
+        gdst.zero_()
+        torch.cuda.synchronize()
+        dist.barrier()
+        g2.replay()
+        torch.cuda.synchronize()
+        torch.testing.assert_close(gdst.cpu(), gfull.sum(0) * (it + 1),
+                                   rtol=1e-5, atol=1e-3)
+    results.append("hipgraph_gated")
+
     c.ee_destroy(ee)
     dist.barrier()
     print(f"XPROC_OK rank={rank} {'+'.join(results)}", flush=True)
