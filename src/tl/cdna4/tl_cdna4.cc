@@ -157,6 +157,7 @@ class Cdna4TlTeam final : public TlTeam {
             (size_t)(cfg_.nslots + cfg_.npers) * 2 * 2 * cfg_.chunk;
         HIPCHK(hipMalloc((void **)&scratch_, scratch_bytes_));
         pslot_used_.assign(cfg_.npers, false);
+        pslot_kind_.assign(cfg_.npers, 0);
         /* flags: fine-grained for cross-GPU system-scope atomics */
         hipError_t e =
             hipExtMallocWithFlags((void **)&flags_, ec_hip::kFlagsBytes,
@@ -264,12 +265,25 @@ class Cdna4TlTeam final : public TlTeam {
 
     /* Persistent-slot allocator. Deterministic across ranks as long as
      * persistent triggered requests are created/finalized in the same
-     * order on every rank (collective semantics): lowest free index. */
-    int alloc_pslot()
+     * order on every rank (collective semantics): lowest matching index.
+     * A slot's device-side graph counters are NEVER reset, so a slot may
+     * only be reused by a request with the SAME kernel pattern (kind):
+     * counters then continue monotonically and stay consistent. */
+    enum PslotKind { PK_NONE = 0, PK_FUSED, PK_G_AR, PK_G_RS, PK_G_AG,
+                     PK_G_A2A };
+    int alloc_pslot(int kind)
     {
+        /* first pass: same-kind slot; second: never-used slot */
         for (uint32_t i = 0; i < cfg_.npers; i++) {
-            if (!pslot_used_[i]) {
+            if (!pslot_used_[i] && pslot_kind_[i] == kind) {
                 pslot_used_[i] = true;
+                return (int)(cfg_.nslots + i);
+            }
+        }
+        for (uint32_t i = 0; i < cfg_.npers; i++) {
+            if (!pslot_used_[i] && pslot_kind_[i] == PK_NONE) {
+                pslot_used_[i] = true;
+                pslot_kind_[i] = kind;
                 return (int)(cfg_.nslots + i);
             }
         }
@@ -281,6 +295,7 @@ class Cdna4TlTeam final : public TlTeam {
             pslot_used_[slot - cfg_.nslots] = false;
         }
     }
+    std::vector<int> pslot_kind_;
 
     Cdna4Cfg    cfg_;
     ShmSeg      seg_;
@@ -463,10 +478,11 @@ class FusedAllreduceTask final : public Cdna4Task {
             return UCC_ERR_NOT_SUPPORTED;
         }
         if (pslot_ < 0) {
-            pslot_ = tt_->alloc_pslot();
+            pslot_ = tt_->alloc_pslot(Cdna4TlTeam::PK_FUSED);
             if (pslot_ < 0) {
                 ucc_error("no free persistent slot (TL_CDNA4 "
-                          "PERSISTENT_SLOTS exhausted)");
+                          "PERSISTENT_SLOTS exhausted or pattern "
+                          "mismatch)");
                 return UCC_ERR_NO_RESOURCE;
             }
             me_ = tt_->team_->rank;
@@ -491,11 +507,9 @@ class FusedAllreduceTask final : public Cdna4Task {
         ga.slot          = pslot_;
         ga.error_word    = tt_->err_host_;
         ga.parity_stride = 2 * tt_->cfg_.chunk;
-        size_t blocks    = (bytes + 128 * 1024 - 1) / (128 * 1024);
-        ga.nblocks       = (int)(blocks < 1 ? 1
-                                 : blocks > (size_t)ec_hip::kMaxGraphBlocks
-                                     ? (size_t)ec_hip::kMaxGraphBlocks
-                                     : blocks);
+        /* FIXED grid: per-block launch counters must advance uniformly
+         * across every request that ever reuses this slot */
+        ga.nblocks = 8;
         ucc_status_t st =
             ec_hip::fused_allreduce_graph(ga, (hipStream_t)ee_stream);
         if (st != UCC_OK) {
@@ -1235,9 +1249,21 @@ class GatedCollTask final : public Cdna4Task {
     ucc_status_t triggered_post(void *ee_stream) override
     {
         if (pslot_ < 0) {
-            pslot_ = tt_->alloc_pslot();
+            ct_ = a_.coll_type == UCC_COLL_TYPE_REDUCE_SCATTERV
+                      ? UCC_COLL_TYPE_REDUCE_SCATTER
+                      : a_.coll_type;
+            int kind = ct_ == UCC_COLL_TYPE_ALLREDUCE
+                           ? Cdna4TlTeam::PK_G_AR
+                       : ct_ == UCC_COLL_TYPE_REDUCE_SCATTER
+                           ? Cdna4TlTeam::PK_G_RS
+                       : ct_ == UCC_COLL_TYPE_ALLGATHER
+                           ? Cdna4TlTeam::PK_G_AG
+                           : Cdna4TlTeam::PK_G_A2A;
+            pslot_ = tt_->alloc_pslot(kind);
             if (pslot_ < 0 || pslot_ >= ec_hip::kGatedSlots) {
-                ucc_error("no persistent slot for triggered gated coll");
+                ucc_error("no persistent slot for triggered gated coll "
+                          "(PERSISTENT_SLOTS exhausted or pattern "
+                          "mismatch)");
                 return UCC_ERR_NO_RESOURCE;
             }
             me_ = tt_->team_->rank;
