@@ -153,6 +153,9 @@ class RcclTask final : public Task {
         if (ev_) {
             hipEventDestroy(ev_);
         }
+        if (barrier_buf_) {
+            hipFree(barrier_buf_);
+        }
     }
 
     ucc_status_t post() override

@@ -389,6 +389,26 @@ class TcpTask : public Task {
     }
     ~TcpTask() override
     {
+        /* error-path teardown: ops may still sit in connection queues —
+         * unlink them before freeing or the next Conn::progress would
+         * touch freed memory */
+        for (auto &c : tt_->conns_) {
+            for (auto *s : sends_) {
+                auto it = std::find(c.sendq.begin(), c.sendq.end(), s);
+                if (it != c.sendq.end()) {
+                    c.sendq.erase(it);
+                }
+            }
+            for (auto *r : recvs_) {
+                auto it = std::find(c.recvq.begin(), c.recvq.end(), r);
+                if (it != c.recvq.end()) {
+                    c.recvq.erase(it);
+                }
+                if (c.cur_recv == r) {
+                    c.cur_recv = nullptr; /* drain remainder to sink */
+                }
+            }
+        }
         for (auto *s : sends_) {
             delete s;
         }
