@@ -1249,14 +1249,14 @@ class GatedCollTask final : public Cdna4Task {
     ucc_status_t triggered_post(void *ee_stream) override
     {
         if (pslot_ < 0) {
-            ct_ = a_.coll_type == UCC_COLL_TYPE_REDUCE_SCATTERV
-                      ? UCC_COLL_TYPE_REDUCE_SCATTER
-                      : a_.coll_type;
-            int kind = ct_ == UCC_COLL_TYPE_ALLREDUCE
+            ucc_coll_type_t k_ct = a_.coll_type;
+            int kind = k_ct == UCC_COLL_TYPE_ALLREDUCE
                            ? Cdna4TlTeam::PK_G_AR
-                       : ct_ == UCC_COLL_TYPE_REDUCE_SCATTER
+                       : (k_ct == UCC_COLL_TYPE_REDUCE_SCATTER ||
+                          k_ct == UCC_COLL_TYPE_REDUCE_SCATTERV)
                            ? Cdna4TlTeam::PK_G_RS
-                       : ct_ == UCC_COLL_TYPE_ALLGATHER
+                       : (k_ct == UCC_COLL_TYPE_ALLGATHER ||
+                          k_ct == UCC_COLL_TYPE_ALLGATHERV)
                            ? Cdna4TlTeam::PK_G_AG
                            : Cdna4TlTeam::PK_G_A2A;
             pslot_ = tt_->alloc_pslot(kind);
@@ -1334,10 +1334,61 @@ class GatedCollTask final : public Cdna4Task {
             dtsz_  = ucc_dt_size(dt_);
             size_t block = a_.dst.info.count * dtsz_ / n_;
             out_b_ = block;
+            cnt_.assign(n_, block);
+            dsp_.resize(n_);
+            for (uint32_t r = 0; r < n_; r++) {
+                dsp_[r] = (size_t)r * block;
+            }
             dbuf_  = (uint8_t *)a_.dst.info.buffer;
             sbuf_  = inplace ? dbuf_ + me_ * block
                              : (const uint8_t *)a_.src.info.buffer;
             total_ = block; /* fragment over the per-rank block */
+            break;
+        }
+        case UCC_COLL_TYPE_ALLGATHERV: {
+            /* per-rank counts are global knowledge (dst.info_v), so the
+             * symmetric-launch invariant holds: every rank launches
+             * ceil(max_block / chunk) fragments (empty tails still
+             * launch and signal) */
+            dt_   = a_.dst.info_v.datatype;
+            dtsz_ = ucc_dt_size(dt_);
+            cnt_.resize(n_);
+            dsp_.resize(n_);
+            size_t maxb = 0;
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = coll_count_at(a_, a_.dst.info_v.counts, r) *
+                          dtsz_;
+                dsp_[r] =
+                    coll_disp_at(a_, a_.dst.info_v.displacements, r) *
+                    dtsz_;
+                maxb = cnt_[r] > maxb ? cnt_[r] : maxb;
+            }
+            dbuf_  = (uint8_t *)a_.dst.info_v.buffer;
+            sbuf_  = inplace ? dbuf_ + dsp_[me_]
+                             : (const uint8_t *)a_.src.info.buffer;
+            total_ = maxb;
+            ct_    = UCC_COLL_TYPE_ALLGATHERV;
+            break;
+        }
+        case UCC_COLL_TYPE_REDUCE_SCATTERV: {
+            dt_   = a_.dst.info_v.datatype;
+            dtsz_ = ucc_dt_size(dt_);
+            cnt_.resize(n_);
+            dsp_.resize(n_);
+            size_t off = 0;
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = coll_count_at(a_, a_.dst.info_v.counts, r) *
+                          dtsz_;
+                dsp_[r] = off;
+                off += cnt_[r];
+            }
+            total_ = off;
+            sbuf_  = inplace ? (const uint8_t *)a_.dst.info_v.buffer
+                             : (const uint8_t *)a_.src.info.buffer;
+            dbuf_  = inplace
+                         ? (uint8_t *)a_.dst.info_v.buffer + dsp_[me_]
+                         : (uint8_t *)a_.dst.info_v.buffer;
+            ct_ = UCC_COLL_TYPE_REDUCE_SCATTERV;
             break;
         }
         case UCC_COLL_TYPE_ALLTOALL: {
@@ -1424,11 +1475,13 @@ class GatedCollTask final : public Cdna4Task {
             ga.op          = op_;
             ga.alpha       = alpha_;
             if (derive) {
+                const bool is_rs =
+                    ct_ == UCC_COLL_TYPE_REDUCE_SCATTER ||
+                    ct_ == UCC_COLL_TYPE_REDUCE_SCATTERV;
                 ga.derive     = 1;
                 ga.pp         = ct_ == UCC_COLL_TYPE_ALLREDUCE ? 3 : 2;
-                ga.has_reduce = ct_ == UCC_COLL_TYPE_ALLREDUCE ||
-                                ct_ == UCC_COLL_TYPE_REDUCE_SCATTER;
-                ga.has_gather = ct_ != UCC_COLL_TYPE_REDUCE_SCATTER;
+                ga.has_reduce = ct_ == UCC_COLL_TYPE_ALLREDUCE || is_rs;
+                ga.has_gather = !is_rs;
             }
             ucc_status_t st = UCC_OK;
             switch (ct_) {
@@ -1462,12 +1515,20 @@ class GatedCollTask final : public Cdna4Task {
                 }
                 break;
             }
+            case UCC_COLL_TYPE_REDUCE_SCATTERV:
             case UCC_COLL_TYPE_REDUCE_SCATTER: {
                 /* stage frag of the whole source; reduce ONLY the
                  * intersection with my output slice, directly into my
                  * user dst (no gather phase) */
                 ga.src = sbuf_ + off;
-                size_t s0 = (size_t)me_ * out_b_, s1 = s0 + out_b_;
+                size_t s0, s1;
+                if (ct_ == UCC_COLL_TYPE_REDUCE_SCATTERV) {
+                    s0 = dsp_[me_];
+                    s1 = s0 + cnt_[me_];
+                } else {
+                    s0 = (size_t)me_ * out_b_;
+                    s1 = s0 + out_b_;
+                }
                 size_t b = s0 > off ? s0 : off;
                 size_t e = s1 < off + len ? s1 : off + len;
                 ga.sl_b   = b > e ? 0 : b - off;
@@ -1485,15 +1546,25 @@ class GatedCollTask final : public Cdna4Task {
                 }
                 break;
             }
+            case UCC_COLL_TYPE_ALLGATHERV:
             case UCC_COLL_TYPE_ALLGATHER: {
                 /* stage my block frag; gather reads peers' STAGED data
-                 * (in areas) into dst block positions */
+                 * (in areas) into dst block positions. v: per-rank
+                 * lengths clipped to each block's count */
+                size_t mylen =
+                    off >= cnt_[me_] ? 0
+                    : cnt_[me_] - off < gran_ ? cnt_[me_] - off
+                                              : gran_;
                 ga.src = sbuf_ + off;
+                ga.len = mylen;
                 ga.dst = dbuf_;
                 for (uint32_t r = 0; r < n_; r++) {
                     ga.peer_out[r] = tt_->area(r, slot_, p, 0);
-                    ga.slice_b[r]  = (uint64_t)r * out_b_ + off;
-                    ga.slice_e[r]  = ga.slice_b[r] + len;
+                    size_t l = off >= cnt_[r] ? 0
+                               : cnt_[r] - off < gran_ ? cnt_[r] - off
+                                                       : gran_;
+                    ga.slice_b[r] = dsp_[r] + off;
+                    ga.slice_e[r] = ga.slice_b[r] + l;
                 }
                 ga.t_sw_reduce   = L[1][slot_][p] * B;
                 ga.t_sw_gather   = L[2][slot_][p] * B;
@@ -1551,6 +1622,7 @@ class GatedCollTask final : public Cdna4Task {
     size_t             gran_ = 0, cell_ = 0, out_b_ = 0;
     const uint8_t     *sbuf_ = nullptr;
     uint8_t           *dbuf_ = nullptr;
+    std::vector<size_t> cnt_, dsp_; /* per-rank bytes (ag(v)/rs(v)) */
     int                pslot_ = -1;
 };
 
@@ -1622,10 +1694,16 @@ void Cdna4TlTeam::get_scores(Team *team, ScoreMap &map)
         r.alg_name = "gated_pipeline";
         r.init     = [self](const ucc_coll_args_t &args, Team *t,
                         Task **task) -> ucc_status_t {
-            if ((args.coll_type == UCC_COLL_TYPE_ALLREDUCE ||
-                 args.coll_type == UCC_COLL_TYPE_REDUCE_SCATTER) &&
-                !ec_hip::op_supported(args.dst.info.datatype, args.op)) {
-                return UCC_ERR_NOT_SUPPORTED;
+            if (args.coll_type == UCC_COLL_TYPE_ALLREDUCE ||
+                args.coll_type == UCC_COLL_TYPE_REDUCE_SCATTER ||
+                args.coll_type == UCC_COLL_TYPE_REDUCE_SCATTERV) {
+                ucc_datatype_t gdt =
+                    args.coll_type == UCC_COLL_TYPE_REDUCE_SCATTERV
+                        ? args.dst.info_v.datatype
+                        : args.dst.info.datatype;
+                if (!ec_hip::op_supported(gdt, args.op)) {
+                    return UCC_ERR_NOT_SUPPORTED;
+                }
             }
             /* in-process multi-rank jigs can serialize spinning kernels
              * on one HW queue: host-gated staged path is the safe
@@ -1700,7 +1778,9 @@ void Cdna4TlTeam::get_scores(Team *team, ScoreMap &map)
         add_gated(UCC_COLL_TYPE_ALLREDUCE, cfg_.fused_max + 1, SIZE_MAX,
                   90);
         add_gated(UCC_COLL_TYPE_REDUCE_SCATTER, 0, SIZE_MAX, 90);
+        add_gated(UCC_COLL_TYPE_REDUCE_SCATTERV, 0, SIZE_MAX, 90);
         add_gated(UCC_COLL_TYPE_ALLGATHER, 0, SIZE_MAX, 90);
+        add_gated(UCC_COLL_TYPE_ALLGATHERV, 0, SIZE_MAX, 90);
         add_gated(UCC_COLL_TYPE_ALLTOALL, 0, SIZE_MAX, 90);
     }
     add(UCC_COLL_TYPE_ALLREDUCE, 0, SIZE_MAX, 80, "staged_linear", false);

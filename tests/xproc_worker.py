@@ -171,7 +171,46 @@ def main():
         exp = torch.cat(
             [a2s[s][rank * per:(rank + 1) * per] for s in range(world)])
         torch.testing.assert_close(a2d.cpu(), exp)
-    results.append("gated_rs_ag_a2a")
+        # v-variants through the gated pipeline (globally-known counts)
+        vcnts = [(r + 1) * 3_000_000 for r in range(world)]
+        vd = []
+        t0 = 0
+        for x in vcnts:
+            vd.append(t0)
+            t0 += x
+        vtotal = sum(vcnts)
+        src = torch.randn(vcnts[rank], generator=g0).cuda()
+        # regenerate deterministically per rank for expectation
+        gsrcs = []
+        gg = torch.Generator().manual_seed(4242 + it)
+        for r in range(world):
+            gsrcs.append(torch.randn(vcnts[r], generator=gg))
+        src.copy_(gsrcs[rank])
+        agd = torch.zeros(vtotal, device="cuda")
+        agv = c.coll_init(team, "allgatherv", src=src.data_ptr(),
+                          dst=agd.data_ptr(), count=vcnts[rank],
+                          dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA,
+                          dst_counts=vcnts, dst_displs=vd)
+        wait(agv, ctx)
+        torch.cuda.synchronize()
+        torch.testing.assert_close(agd.cpu(), torch.cat(gsrcs))
+
+        rsrcs = []
+        gg = torch.Generator().manual_seed(5252 + it)
+        for r in range(world):
+            rsrcs.append(torch.randn(vtotal, generator=gg))
+        rssrc = rsrcs[rank].cuda()
+        rsd = torch.zeros(vcnts[rank], device="cuda")
+        rsv = c.coll_init(team, "reduce_scatterv", src=rssrc.data_ptr(),
+                          dst=rsd.data_ptr(), count=0, dt=dtypes.FLOAT32,
+                          mem_type=dtypes.MEM_CUDA, dst_counts=vcnts)
+        wait(rsv, ctx)
+        torch.cuda.synchronize()
+        exp = sum(rsrcs)
+        torch.testing.assert_close(
+            rsd.cpu(), exp[vd[rank]:vd[rank] + vcnts[rank]], rtol=1e-5,
+            atol=1e-4)
+    results.append("gated_rs_ag_a2a_v")
 
     # 3b. ucc_mem_map export/import across processes: rank 0 exports a
     # device buffer, rank 1 imports and reads it over IPC.
