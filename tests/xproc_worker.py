@@ -179,13 +179,13 @@ def main():
             vd.append(t0)
             t0 += x
         vtotal = sum(vcnts)
-        src = torch.randn(vcnts[rank], generator=g0).cuda()
-        # regenerate deterministically per rank for expectation
+        # per-rank blocks from a local generator (must NOT consume g0:
+        # its stream is shared across ranks and vcnts differ per rank)
         gsrcs = []
         gg = torch.Generator().manual_seed(4242 + it)
         for r in range(world):
             gsrcs.append(torch.randn(vcnts[r], generator=gg))
-        src.copy_(gsrcs[rank])
+        src = gsrcs[rank].cuda()
         agd = torch.zeros(vtotal, device="cuda")
         agv = c.coll_init(team, "allgatherv", src=src.data_ptr(),
                           dst=agd.data_ptr(), count=vcnts[rank],
