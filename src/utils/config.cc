@@ -22,6 +22,18 @@ std::string Config::key(const std::string &component, const std::string &name)
 
 Config::Config()
 {
+    /* core variables always present in `ucc_info -c` dumps */
+    declare("", "LOG_LEVEL", "warn",
+            "fatal|error|warn|info|debug|trace|trace_poll");
+    declare("", "TUNE", "",
+            "score-map override string: coll:msgrange:mem:@alg:score,...");
+    declare("", "CONFIG_FILE", "ucc.conf", "ini config path");
+    declare("", "LOCK_FREE_PROGRESS_Q", "1",
+            "lock-free MPMC progress queue in THREAD_MULTIPLE");
+    declare("", "COLL_TRACE", "0",
+            "print per-collective algorithm selection at INFO");
+    declare("", "FAKE_NODE_SPLIT", "0",
+            "testing: spread contexts over k pseudo-nodes");
     /* ini file: UCC_CONFIG_FILE > ./ucc.conf ; lines "KEY = value",
      * '#'/';' comments, section headers ignored (keys are globally unique
      * via the UCC_ prefix convention). */

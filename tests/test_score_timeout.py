@@ -231,3 +231,42 @@ def test_active_set_bcast():
     outs = job.allreduce_np(arrs)
     for o in outs:
         np.testing.assert_allclose(o, np.full(100, 15.0, np.float32))
+
+
+def test_profiler_and_coll_trace(tmp_path):
+    """Aux subsystems: UCC_PROFILE_MODE event recorder output and
+    UCC_COLL_TRACE selection print (reference utils/profile +
+    UCC_COLL_TRACE parity)."""
+    prof = tmp_path / "prof.log"
+    worker = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "j = LocalJob(2)\n"
+        "j.allreduce_np([np.ones(64, np.float32)] * 2)\n"
+        "print('WORKER_OK')\n" % (REPO,))
+    env = dict(os.environ)
+    env["UCC_PROFILE_MODE"] = "log"
+    env["UCC_PROFILE_FILE"] = str(prof)
+    env["UCC_COLL_TRACE"] = "1"
+    env["UCC_LOG_LEVEL"] = "info"
+    p = subprocess.run([sys.executable, "-c", worker], env=env,
+                       capture_output=True, text=True, timeout=120)
+    assert p.returncode == 0 and "WORKER_OK" in p.stdout
+    out = p.stdout + p.stderr
+    assert "allreduce" in out and "->" in out, out[-1500:]  # coll trace
+    txt = prof.read_text()
+    assert "new allreduce" in txt and "free finalize" in txt, txt[:500]
+
+
+def test_ucc_info_tool():
+    """ucc_info: version, config dump, simulated score map."""
+    binary = os.path.join(REPO, "build", "ucc_info")
+    if not os.path.exists(binary):
+        pytest.skip("ucc_info not built")
+    p = subprocess.run([binary, "-c", "-s"], capture_output=True,
+                       text=True, timeout=120)
+    assert p.returncode == 0
+    assert "transports: self shm tcp cdna4" in p.stdout
+    assert "UCC_LOG_LEVEL" in p.stdout
+    assert "allreduce:host:" in p.stdout

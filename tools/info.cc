@@ -31,7 +31,7 @@ static void print_configs()
     auto entries = c.entries();
     for (auto &e : entries) {
         std::string key = Config::key(e.component, e.name);
-        printf("UCC_%s=%s\n", key.c_str(),
+        printf("%s=%s\n", key.c_str(),
                c.get(e.component, e.name, e.dflt).c_str());
         if (!e.doc.empty()) {
             printf("# %s\n", e.doc.c_str());
