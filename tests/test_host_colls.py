@@ -326,3 +326,72 @@ def test_topo_sbgps(job):
     assert info["node_idx"] == 0
     assert info["leaders_size"] == 1
     assert info["leaders_idx"] == 0
+
+
+@pytest.mark.parametrize("npdt,dt", [
+    (np.int32, dtypes.INT32), (np.int64, dtypes.INT64),
+    (np.uint8, dtypes.UINT8), (np.float64, dtypes.FLOAT64),
+])
+def test_allreduce_int_float_dtypes(job, npdt, dt):
+    rng = np.random.default_rng(21)
+    n = job.n
+    if np.issubdtype(npdt, np.integer):
+        arrs = [rng.integers(0, 50, 777).astype(npdt) for _ in range(n)]
+    else:
+        arrs = [(rng.random(777) - 0.5).astype(npdt) for _ in range(n)]
+    outs = [np.zeros(777, npdt) for _ in range(n)]
+    reqs = job.coll("allreduce", [
+        dict(src=arrs[r].ctypes.data, dst=outs[r].ctypes.data, count=777,
+             dt=dt) for r in range(n)])
+    job.run(reqs)
+    exp = np.sum(arrs, axis=0, dtype=npdt)
+    for o in outs:
+        if np.issubdtype(npdt, np.integer):
+            np.testing.assert_array_equal(o, exp)
+        else:
+            np.testing.assert_allclose(o, exp, rtol=1e-12, atol=1e-12)
+
+
+def test_allreduce_bf16_host(job):
+    """bf16 software reduction on the host path (reference ec/cpu bf16)."""
+    from ucc_amd import core  # noqa: F401
+    n = job.n
+    rng = np.random.default_rng(22)
+    f32 = [(rng.random(1024) - 0.5).astype(np.float32) for _ in range(n)]
+    # encode to bf16 (truncate-to-nearest-even like the library)
+    def to_bf16(a):
+        u = a.view(np.uint32)
+        r = 0x7FFF + ((u >> 16) & 1)
+        return ((u + r) >> 16).astype(np.uint16)
+    def from_bf16(b):
+        return (b.astype(np.uint32) << 16).view(np.float32)
+    srcs = [to_bf16(a) for a in f32]
+    outs = [np.zeros(1024, np.uint16) for _ in range(n)]
+    reqs = job.coll("allreduce", [
+        dict(src=srcs[r].ctypes.data, dst=outs[r].ctypes.data, count=1024,
+             dt=dtypes.BFLOAT16) for r in range(n)])
+    job.run(reqs)
+    exp = np.sum([from_bf16(s) for s in srcs], axis=0)
+    for o in outs:
+        np.testing.assert_allclose(from_bf16(o), exp, rtol=3e-2,
+                                   atol=1e-1)
+
+
+def test_allreduce_logical_bitwise(job):
+    n = job.n
+    rng = np.random.default_rng(23)
+    arrs = [rng.integers(0, 2 ** 16, 512).astype(np.uint32)
+            for _ in range(n)]
+    for op, fn in ((dtypes.OP_BAND, np.bitwise_and),
+                   (dtypes.OP_BOR, np.bitwise_or),
+                   (dtypes.OP_BXOR, np.bitwise_xor)):
+        outs = [np.zeros(512, np.uint32) for _ in range(n)]
+        reqs = job.coll("allreduce", [
+            dict(src=arrs[r].ctypes.data, dst=outs[r].ctypes.data,
+                 count=512, dt=dtypes.UINT32, op=op) for r in range(n)])
+        job.run(reqs)
+        exp = arrs[0]
+        for a in arrs[1:]:
+            exp = fn(exp, a)
+        for o in outs:
+            np.testing.assert_array_equal(o, exp)
