@@ -65,6 +65,7 @@ struct FusedArgs {
     float       alpha;
     uint64_t   *error_word; /* local fine-grained word set on spin timeout */
     int         nblocks;    /* workgroups; each handles a count slice     */
+    uint64_t    spin_limit; /* 0 = default kSpinLimit                     */
 };
 ucc_status_t fused_allreduce(const FusedArgs &a, hipStream_t stream);
 
@@ -170,6 +171,7 @@ struct GatedArgs {
      * has_reduce/has_gather describe the pattern so waits on phases that
      * never launch resolve to zero. */
     int         derive, pp, has_reduce, has_gather;
+    uint64_t    spin_limit; /* 0 = default kSpinLimit */
     /* per-dest cell staging (alltoall): my_in[c_dst_off[k]] <-
      * src[c_src_off[k]], c_len[k] bytes; 0 = contiguous stage of len   */
     int         n_cells;

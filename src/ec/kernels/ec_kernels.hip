@@ -418,6 +418,10 @@ __device__ __forceinline__ void sys_store(uint64_t *p, uint64_t v)
 }
 
 constexpr uint64_t kSpinLimit = 200u * 1000u * 1000u;
+__device__ __forceinline__ uint64_t spin_cap(uint64_t v)
+{
+    return v ? v : kSpinLimit;
+}
 
 template <typename T, int OP, int VEC>
 __global__ void k_fused_allreduce(FusedArgs a)
@@ -454,8 +458,9 @@ __global__ void k_fused_allreduce(FusedArgs a)
     if (blockIdx.x == 0) {
         if (threadIdx.x == 0) {
             uint64_t spins = 0;
+            const uint64_t cap = spin_cap(a.spin_limit);
             while (sys_load(stage_cnt) < a.stage_target) {
-                if (++spins > kSpinLimit) {
+                if (++spins > cap) {
                     sys_store(a.error_word, 1);
                     s_err = 1;
                     break;
@@ -475,11 +480,12 @@ __global__ void k_fused_allreduce(FusedArgs a)
     if (!s_err && threadIdx.x < 64) {
         int      j     = (int)threadIdx.x;
         uint64_t spins = 0;
+        const uint64_t cap = spin_cap(a.spin_limit);
         if (j < a.nranks) {
             const uint64_t *f =
                 a.local_flags + (uint64_t)a.slot * kMaxRanks + j;
             while (sys_load(f) < a.seq) {
-                if (++spins > kSpinLimit) {
+                if (++spins > cap) {
                     sys_store(a.error_word, 1);
                     s_err = 1;
                     break;
@@ -721,8 +727,9 @@ gated_wait(const GatedArgs &a, int phase, uint64_t target)
             const uint64_t *f =
                 a.peer_flags[j] + gated_idx(phase, a.slot, a.parity);
             uint64_t spins = 0;
+            const uint64_t cap = spin_cap(a.spin_limit);
             while (sys_load(f) < target) {
-                if (++spins > kSpinLimit) {
+                if (++spins > cap) {
                     sys_store(a.error_word, 1);
                     s_err = 1;
                     break;

@@ -62,6 +62,7 @@ static inline size_t coll_disp_at(const ucc_coll_args_t &a, const void *d,
     } while (0)
 
 struct Cdna4Cfg {
+    uint64_t spin_limit; /* device spin bound; 0 = kernel default */
     uint32_t nslots;
     uint32_t npers;     /* dedicated slots for persistent triggered
                            (graph-captured) collectives              */
@@ -430,6 +431,7 @@ class FusedAllreduceTask final : public Cdna4Task {
             fa.slot        = (int)slot_;
             fa.seq         = fseq_;
             fa.error_word  = tt_->err_host_;
+            fa.spin_limit  = tt_->cfg_.spin_limit;
             size_t blocks  = (bytes_ + 128 * 1024 - 1) / (128 * 1024);
             fa.nblocks     = (int)(blocks < 1 ? 1 : blocks > 16 ? 16 : blocks);
             if (tt_->stage_cum_.size() < tt_->cfg_.nslots) {
@@ -1466,6 +1468,7 @@ class GatedCollTask final : public Cdna4Task {
             }
             ga.local_flags = tt_->flags_;
             ga.error_word  = tt_->err_host_;
+            ga.spin_limit  = tt_->cfg_.spin_limit;
             ga.len         = len;
             ga.rank        = (int)me_;
             ga.nranks      = (int)n_;
@@ -1671,6 +1674,8 @@ class Cdna4Tl final : public Tl {
             c.npers = c.nslots < 8 ? 8 - c.nslots : 0;
         }
         c.chunk     = cfg.get_size("TL_CDNA4", "CHUNK_SIZE", 32 * 1024 * 1024);
+        c.spin_limit =
+            (uint64_t)cfg.get_int("TL_CDNA4", "SPIN_LIMIT", 0);
         c.fused_max = cfg.get_size("TL_CDNA4", "FUSED_MAX", 512 * 1024);
         return new Cdna4TlTeam(tlc, team, c);
     }

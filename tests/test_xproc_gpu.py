@@ -50,3 +50,23 @@ def test_xproc_stress():
     sys.stderr.write(proc.stderr[-2000:])
     assert proc.returncode == 0
     assert proc.stdout.count("STRESS_OK") == 2
+
+
+def test_xproc_spin_timeout_recovery():
+    """A dead peer degrades to UCC_ERR_TIMED_OUT via the bounded kernel
+    spin — the GPU stays usable (docs/GATED_PIPELINE.md safety)."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    env = dict(os.environ)
+    env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    env["UCC_TL_CDNA4_SPIN_LIMIT"] = "3000000"  # ~ms-scale bound
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr=127.0.0.1",
+         "--master-port=29533", os.path.join(REPO, "tests",
+                                             "timeout_worker.py")],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=300)
+    sys.stdout.write(proc.stdout[-2000:])
+    sys.stderr.write(proc.stderr[-2000:])
+    assert proc.returncode == 0
+    assert proc.stdout.count("TIMEOUT_OK") == 2
