@@ -1648,6 +1648,10 @@ class Cdna4Tl final : public Tl {
                     "staging fragment bytes per slot area");
         cfg.declare("TL_CDNA4", "FUSED_MAX", "512k",
                     "max msg bytes for the fused single-kernel allreduce");
+        cfg.declare("TL_CDNA4", "GATED", "1",
+                    "device-gated pipeline for large colls");
+        cfg.declare("TL_CDNA4", "SPIN_LIMIT", "0",
+                    "device spin bound override (0 = default ~seconds)");
         if (!cfg.get_bool("TL_CDNA4", "ENABLE", true) ||
             !mc::hip_available()) {
             return nullptr;
