@@ -679,15 +679,24 @@ __device__ __forceinline__ int gated_idx(int phase, int slot, int parity)
     return kGatedCntBase + (phase * kGatedSlots + slot) * 2 + parity;
 }
 
-/* graph mode: this block's launch ordinal within the (slot,parity)
- * pattern -> fills the host-target fields of `a` in place. Every gated
+/* Wait targets for this launch. Host-target mode reads them straight
+ * from the (const) kernarg struct; graph mode derives them from this
+ * block's launch ordinal within the (slot,parity) pattern: every gated
  * kernel of the pattern increments the same per-block counter, so with
  * pp kernels per iteration, counter value v maps to iteration
- * u = (v-1)/pp + 1 on every block of the same launch (fixed grid). */
-__device__ __forceinline__ void gated_derive(GatedArgs &a, int my_ord)
+ * u = (v-1)/pp + 1 on every block of the same launch (fixed grid).
+ * IMPORTANT: `a` stays const — mutating the by-value kernel argument
+ * would force the whole ~660-byte struct into scratch memory and put
+ * scratch reloads inside the hot loops (observed in gfx950 ISA). */
+struct GatedTargets {
+    uint64_t sw_reduce, sw_gather, prev_gather, stage, gather_wait;
+};
+
+__device__ __forceinline__ GatedTargets gated_targets(const GatedArgs &a)
 {
     if (!a.derive) {
-        return;
+        return {a.t_sw_reduce, a.t_sw_gather, a.t_prev_gather, a.t_stage,
+                a.t_gather_wait};
     }
     uint64_t *cnt = a.local_flags + kGatedGraphBase +
                     ((uint64_t)a.slot * 2 + a.parity) * kGatedBlocks +
@@ -702,12 +711,9 @@ __device__ __forceinline__ void gated_derive(GatedArgs &a, int my_ord)
     __syncthreads();
     const uint64_t u = s_u;
     const uint64_t B = (uint64_t)kGatedBlocks;
-    a.t_sw_reduce   = a.has_reduce ? (u - 1) * B : 0;
-    a.t_sw_gather   = a.has_gather ? (u - 1) * B : 0;
-    a.t_prev_gather = a.has_gather ? (u - 1) * B : 0;
-    a.t_stage       = u * B;
-    a.t_gather_wait = u * B;
-    (void)my_ord;
+    return {a.has_reduce ? (u - 1) * B : 0,
+            a.has_gather ? (u - 1) * B : 0,
+            a.has_gather ? (u - 1) * B : 0, u * B, u * B};
 }
 
 /* all blocks wait until every rank's counter[idx] >= target; on spin
@@ -759,11 +765,11 @@ __device__ __forceinline__ void gated_signal(const GatedArgs &a, int phase)
     }
 }
 
-__global__ void k_staged_stage(GatedArgs a)
+__global__ void k_staged_stage(const GatedArgs a)
 {
-    gated_derive(a, 0);
-    if (!gated_wait(a, 1, a.t_sw_reduce) ||
-        !gated_wait(a, 2, a.t_sw_gather)) { /* in-area reuse */
+    const GatedTargets t = gated_targets(a);
+    if (!gated_wait(a, 1, t.sw_reduce) ||
+        !gated_wait(a, 2, t.sw_gather)) { /* in-area reuse */
         return;
     }
     if (a.n_cells > 0) { /* per-dest cells (alltoall) */
@@ -781,12 +787,12 @@ __global__ void k_staged_stage(GatedArgs a)
 }
 
 template <typename T, int OP, int VEC>
-__global__ void k_staged_reduce(GatedArgs a)
+__global__ void k_staged_reduce(const GatedArgs a)
 {
     using A = typename Cvt<T>::A;
-    gated_derive(a, 1);
-    if (!gated_wait(a, 0, a.t_stage) ||
-        !gated_wait(a, 2, a.t_prev_gather)) {
+    const GatedTargets gt = gated_targets(a);
+    if (!gated_wait(a, 0, gt.stage) ||
+        !gated_wait(a, 2, gt.prev_gather)) {
         return;
     }
     const uint64_t cnt = (a.sl_e - a.sl_b) / sizeof(T);
@@ -795,23 +801,31 @@ __global__ void k_staged_reduce(GatedArgs a)
     const uint64_t nv  = cnt / VEC;
     const int      n   = a.nranks;
     using P            = Pack<T, OP, VEC>;
-    const P *srcs[kMaxRanks];
-    for (int s = 0; s < n; s++) {
-        srcs[s] = (const P *)((const uint8_t *)a.peer_in[s] + a.sl_b);
-    }
+    /* fully unrolled over kMaxRanks with a uniform guard: a runtime-
+     * indexed pointer array would be materialized in scratch memory and
+     * re-loaded inside the hot loop (observed in the gfx950 ISA);
+     * unrolling keeps the 8 peer base pointers in SGPRs and issues all
+     * peer loads back-to-back for xGMI latency overlap. */
     P *out = (P *)a.my_out;
     for (uint64_t i = tid; i < nv; i += str) {
-        P acc = srcs[0][i];
         A r[VEC];
-#pragma unroll
-        for (int k = 0; k < VEC; k++) {
-            r[k] = Cvt<T>::load(acc.v[k]);
-        }
-        for (int s = 1; s < n; s++) {
-            P x = srcs[s][i];
+        {
+            P acc = ((const P *)((const uint8_t *)a.peer_in[0] +
+                                 a.sl_b))[i];
 #pragma unroll
             for (int k = 0; k < VEC; k++) {
-                r[k] = red<A, OP>(r[k], Cvt<T>::load(x.v[k]));
+                r[k] = Cvt<T>::load(acc.v[k]);
+            }
+        }
+#pragma unroll
+        for (int s = 1; s < kMaxRanks; s++) {
+            if (s < n) {
+                P x = ((const P *)((const uint8_t *)a.peer_in[s] +
+                                   a.sl_b))[i];
+#pragma unroll
+                for (int k = 0; k < VEC; k++) {
+                    r[k] = red<A, OP>(r[k], Cvt<T>::load(x.v[k]));
+                }
             }
         }
         P o;
@@ -822,19 +836,26 @@ __global__ void k_staged_reduce(GatedArgs a)
         out[i] = o;
     }
     for (uint64_t t = nv * VEC + tid; t < cnt; t += str) {
-        A r = Cvt<T>::load(((const T *)srcs[0])[t]);
-        for (int s = 1; s < n; s++) {
-            r = red<A, OP>(r, Cvt<T>::load(((const T *)srcs[s])[t]));
+        A r = Cvt<T>::load(
+            ((const T *)((const uint8_t *)a.peer_in[0] + a.sl_b))[t]);
+#pragma unroll
+        for (int s = 1; s < kMaxRanks; s++) {
+            if (s < n) {
+                r = red<A, OP>(
+                    r, Cvt<T>::load(((const T *)((const uint8_t *)
+                                                     a.peer_in[s] +
+                                                 a.sl_b))[t]));
+            }
         }
         ((T *)out)[t] = Cvt<T>::store(apply_alpha<A>(r, a.alpha));
     }
     gated_signal(a, 1);
 }
 
-__global__ void k_staged_gather(GatedArgs a)
+__global__ void k_staged_gather(const GatedArgs a)
 {
-    gated_derive(a, 2);
-    if (!gated_wait(a, a.gw_phase, a.t_gather_wait)) {
+    const GatedTargets t = gated_targets(a);
+    if (!gated_wait(a, a.gw_phase, t.gather_wait)) {
         return;
     }
     /* partition the grid across sources: every peer link streams
