@@ -1239,7 +1239,12 @@ class GatedCollTask final : public Cdna4Task {
         if (st != UCC_OK) {
             return st;
         }
-        phase_ = 0;
+        if (!zc_ready_ && ct_ == UCC_COLL_TYPE_ALLREDUCE &&
+            (a_.flags & UCC_COLL_ARGS_FLAG_PERSISTENT) &&
+            Config::instance().get_bool("TL_CDNA4", "ZCOPY", true)) {
+            zc_ = true;
+        }
+        phase_ = zc_ && !zc_ready_ ? 10 : 0;
         status = UCC_INPROGRESS;
         return progress();
     }
@@ -1283,13 +1288,6 @@ class GatedCollTask final : public Cdna4Task {
         }
         status = UCC_OK; /* stream-ordered completion */
         return UCC_OK;
-    }
-
-    ~GatedCollTask() override
-    {
-        if (pslot_ >= 0) {
-            tt_->free_pslot(pslot_);
-        }
     }
 
   private:
@@ -1417,8 +1415,95 @@ class GatedCollTask final : public Cdna4Task {
     }
 
   public:
+    /* zero-copy handle exchange: my src's HIP-IPC handle travels through
+     * my scratch in-area (peers read it over the already-mapped IPC
+     * scratch), gated by the shm slot steps. One slot use, then the
+     * collective itself takes a fresh use. */
+    struct ZcBlob {
+        uint64_t          magic;
+        hipIpcMemHandle_t h;
+        uint64_t          base_off;
+        uint64_t          raw_ptr;
+        int32_t           pid;
+        int32_t           pad;
+    };
+
     ucc_status_t progress() override
     {
+        if (phase_ == 10) { /* publish my src handle */
+            if (!all_ge(0)) {
+                return UCC_INPROGRESS;
+            }
+            ZcBlob b{};
+            b.magic = 0x5a43;
+            hipDeviceptr_t base  = nullptr;
+            size_t         bsize = 0;
+            if (hipMemGetAddressRange(&base, &bsize,
+                                      (hipDeviceptr_t)sbuf_) !=
+                    hipSuccess ||
+                hipIpcGetMemHandle(&b.h, (void *)base) != hipSuccess) {
+                ucc_warn("zero-copy src export failed, using staging");
+                zc_ = false;
+                close_slot();
+                begin_use();
+                phase_ = 0;
+                return progress();
+            }
+            b.base_off = (uint64_t)((uintptr_t)sbuf_ - (uintptr_t)base);
+            b.raw_ptr  = (uint64_t)(uintptr_t)sbuf_;
+            b.pid      = (int32_t)getpid();
+            if (hipMemcpy(tt_->area(me_, slot_, 0, 0), &b, sizeof(b),
+                          hipMemcpyHostToDevice) != hipSuccess) {
+                return UCC_ERR_NO_RESOURCE;
+            }
+            publish(1);
+            phase_ = 11;
+        }
+        if (phase_ == 11) { /* open peers' handles */
+            if (!all_ge(1)) {
+                return UCC_INPROGRESS;
+            }
+            for (uint32_t r = 0; r < n_; r++) {
+                if (r == me_) {
+                    zc_peer_src_[r] = sbuf_;
+                    continue;
+                }
+                ZcBlob b{};
+                if (hipMemcpy(&b, tt_->area(r, slot_, 0, 0), sizeof(b),
+                              hipMemcpyDeviceToHost) != hipSuccess ||
+                    b.magic != 0x5a43) {
+                    ucc_warn("zero-copy import failed, using staging");
+                    zc_ = false;
+                    break;
+                }
+                if (b.pid == tt_->team_->ctx->proc.pid) {
+                    zc_peer_src_[r] =
+                        (const uint8_t *)(uintptr_t)b.raw_ptr;
+                    continue;
+                }
+                void *m = nullptr;
+                if (hipIpcOpenMemHandle(&m, b.h,
+                                        hipIpcMemLazyEnablePeerAccess) !=
+                    hipSuccess) {
+                    ucc_warn("zero-copy open failed, using staging");
+                    zc_ = false;
+                    break;
+                }
+                zc_mapped_.push_back(m);
+                zc_peer_src_[r] = (const uint8_t *)m + b.base_off;
+            }
+            publish(2);
+            phase_ = 12;
+        }
+        if (phase_ == 12) { /* all peers consumed the blobs */
+            if (!all_ge(2)) {
+                return UCC_INPROGRESS;
+            }
+            zc_ready_ = true;
+            close_slot();
+            begin_use(); /* fresh slot use for the collective itself */
+            phase_ = 0;
+        }
         if (phase_ == 0) { /* entry: previous use of this slot closed */
             if (!all_ge(0)) {
                 return UCC_INPROGRESS;
@@ -1489,9 +1574,17 @@ class GatedCollTask final : public Cdna4Task {
             ucc_status_t st = UCC_OK;
             switch (ct_) {
             case UCC_COLL_TYPE_ALLREDUCE: {
-                /* stage frag -> reduce my slice -> gather all slices */
+                /* stage frag -> reduce my slice -> gather all slices.
+                 * zero-copy: stage is a pure signal (len 0) and reduce
+                 * reads peers' USER src directly over xGMI */
                 ga.src = sbuf_ + off;
                 ga.dst = dbuf_ + off;
+                if (zc_ && zc_ready_) {
+                    ga.len = 0;
+                    for (uint32_t r = 0; r < n_; r++) {
+                        ga.peer_in[r] = zc_peer_src_[r] + off;
+                    }
+                }
                 size_t per = (len / n_) & ~(size_t)255;
                 for (uint32_t r = 0; r < n_; r++) {
                     ga.slice_b[r] = (uint64_t)r * per;
@@ -1627,6 +1720,24 @@ class GatedCollTask final : public Cdna4Task {
     uint8_t           *dbuf_ = nullptr;
     std::vector<size_t> cnt_, dsp_; /* per-rank bytes (ag(v)/rs(v)) */
     int                pslot_ = -1;
+    /* zero-copy persistent allreduce: peers' USER src buffers mapped
+     * via HIP-IPC (handles exchanged through the scratch channel at
+     * first post) — the reduce phase reads them directly, the stage
+     * kernel degenerates to a pure signal (no staging copy). */
+    bool               zc_ = false, zc_ready_ = false;
+    const uint8_t     *zc_peer_src_[ec_hip::kMaxRanks] = {};
+    std::vector<void *> zc_mapped_;
+
+  public:
+    ~GatedCollTask() override
+    {
+        if (pslot_ >= 0) {
+            tt_->free_pslot(pslot_);
+        }
+        for (void *m : zc_mapped_) {
+            hipIpcCloseMemHandle(m);
+        }
+    }
 };
 
 /* ------------------------------------------------------------ scoring  */
