@@ -171,6 +171,11 @@ struct GatedArgs {
      * has_reduce/has_gather describe the pattern so waits on phases that
      * never launch resolve to zero. */
     int         derive, pp, has_reduce, has_gather;
+    /* write-based distribution (zero-copy allreduce): reduce writes its
+     * slice DIRECTLY into every rank's user dst (peer_out[r] =
+     * mapped dst + frag_off + my_slice_off) instead of the local out
+     * area; the gather launch degenerates to a pure wait+signal. */
+    int         zc_write;
     uint64_t    spin_limit; /* 0 = default kSpinLimit */
     /* per-dest cell staging (alltoall): my_in[c_dst_off[k]] <-
      * src[c_src_off[k]], c_len[k] bytes; 0 = contiguous stage of len   */

@@ -833,7 +833,17 @@ __global__ void k_staged_reduce(const GatedArgs a)
         for (int k = 0; k < VEC; k++) {
             o.v[k] = Cvt<T>::store(apply_alpha<A>(r[k], a.alpha));
         }
-        out[i] = o;
+        if (a.zc_write) {
+            /* write my reduced slice straight into every rank's dst */
+#pragma unroll
+            for (int s = 0; s < kMaxRanks; s++) {
+                if (s < n) {
+                    ((P *)a.peer_out[s])[i] = o;
+                }
+            }
+        } else {
+            out[i] = o;
+        }
     }
     for (uint64_t t = nv * VEC + tid; t < cnt; t += str) {
         A r = Cvt<T>::load(
@@ -847,7 +857,17 @@ __global__ void k_staged_reduce(const GatedArgs a)
                                                  a.sl_b))[t]));
             }
         }
-        ((T *)out)[t] = Cvt<T>::store(apply_alpha<A>(r, a.alpha));
+        T ov = Cvt<T>::store(apply_alpha<A>(r, a.alpha));
+        if (a.zc_write) {
+#pragma unroll
+            for (int s = 0; s < kMaxRanks; s++) {
+                if (s < n) {
+                    ((T *)a.peer_out[s])[t] = ov;
+                }
+            }
+        } else {
+            ((T *)out)[t] = ov;
+        }
     }
     gated_signal(a, 1);
 }

@@ -1421,9 +1421,12 @@ class GatedCollTask final : public Cdna4Task {
      * collective itself takes a fresh use. */
     struct ZcBlob {
         uint64_t          magic;
-        hipIpcMemHandle_t h;
+        hipIpcMemHandle_t h;      /* src allocation  */
         uint64_t          base_off;
         uint64_t          raw_ptr;
+        hipIpcMemHandle_t hd;     /* dst allocation  */
+        uint64_t          d_base_off;
+        uint64_t          d_raw_ptr;
         int32_t           pid;
         int32_t           pad;
     };
@@ -1452,6 +1455,19 @@ class GatedCollTask final : public Cdna4Task {
             b.base_off = (uint64_t)((uintptr_t)sbuf_ - (uintptr_t)base);
             b.raw_ptr  = (uint64_t)(uintptr_t)sbuf_;
             b.pid      = (int32_t)getpid();
+            hipDeviceptr_t dbase  = nullptr;
+            size_t         dbsize = 0;
+            if (hipMemGetAddressRange(&dbase, &dbsize,
+                                      (hipDeviceptr_t)dbuf_) ==
+                    hipSuccess &&
+                hipIpcGetMemHandle(&b.hd, (void *)dbase) == hipSuccess) {
+                b.d_base_off =
+                    (uint64_t)((uintptr_t)dbuf_ - (uintptr_t)dbase);
+                b.d_raw_ptr = (uint64_t)(uintptr_t)dbuf_;
+                b.pad = (dbase == base) ? 1 : 0; /* same allocation */
+            } else {
+                b.magic = 0; /* dst export failed: fall back */
+            }
             if (hipMemcpy(tt_->area(me_, slot_, 0, 0), &b, sizeof(b),
                           hipMemcpyHostToDevice) != hipSuccess) {
                 return UCC_ERR_NO_RESOURCE;
@@ -1466,6 +1482,7 @@ class GatedCollTask final : public Cdna4Task {
             for (uint32_t r = 0; r < n_; r++) {
                 if (r == me_) {
                     zc_peer_src_[r] = sbuf_;
+                    zc_peer_dst_[r] = dbuf_;
                     continue;
                 }
                 ZcBlob b{};
@@ -1479,9 +1496,10 @@ class GatedCollTask final : public Cdna4Task {
                 if (b.pid == tt_->team_->ctx->proc.pid) {
                     zc_peer_src_[r] =
                         (const uint8_t *)(uintptr_t)b.raw_ptr;
+                    zc_peer_dst_[r] = (uint8_t *)(uintptr_t)b.d_raw_ptr;
                     continue;
                 }
-                void *m = nullptr;
+                void *m = nullptr, *md = nullptr;
                 if (hipIpcOpenMemHandle(&m, b.h,
                                         hipIpcMemLazyEnablePeerAccess) !=
                     hipSuccess) {
@@ -1491,6 +1509,35 @@ class GatedCollTask final : public Cdna4Task {
                 }
                 zc_mapped_.push_back(m);
                 zc_peer_src_[r] = (const uint8_t *)m + b.base_off;
+                if (hipIpcOpenMemHandle(&md, b.hd,
+                                        hipIpcMemLazyEnablePeerAccess) ==
+                    hipSuccess) {
+                    zc_mapped_.push_back(md);
+                    zc_peer_dst_[r] = (uint8_t *)md + b.d_base_off;
+                } else if (b.pad == 1) {
+                    /* dst lives in the src allocation (e.g. inplace) */
+                    zc_peer_dst_[r] =
+                        (uint8_t *)m +
+                        ((uint64_t)b.d_raw_ptr -
+                         ((uint64_t)b.raw_ptr - b.base_off));
+                } else {
+                    ucc_warn("zero-copy dst open failed, using staging");
+                    zc_ = false;
+                    break;
+                }
+            }
+            /* direct vector loads/stores need 16B-aligned user
+             * buffers on every rank */
+            if (zc_) {
+                for (uint32_t r = 0; r < n_; r++) {
+                    if (((uintptr_t)zc_peer_src_[r] & 15) ||
+                        ((uintptr_t)zc_peer_dst_[r] & 15)) {
+                        ucc_warn("zero-copy needs 16B-aligned buffers, "
+                                 "using staging");
+                        zc_ = false;
+                        break;
+                    }
+                }
             }
             publish(2);
             phase_ = 12;
@@ -1579,20 +1626,24 @@ class GatedCollTask final : public Cdna4Task {
                  * reads peers' USER src directly over xGMI */
                 ga.src = sbuf_ + off;
                 ga.dst = dbuf_ + off;
-                if (zc_ && zc_ready_) {
-                    ga.len = 0;
-                    for (uint32_t r = 0; r < n_; r++) {
-                        ga.peer_in[r] = zc_peer_src_[r] + off;
-                    }
-                }
                 size_t per = (len / n_) & ~(size_t)255;
                 for (uint32_t r = 0; r < n_; r++) {
                     ga.slice_b[r] = (uint64_t)r * per;
                     ga.slice_e[r] =
                         r == n_ - 1 ? len : (uint64_t)(r + 1) * per;
                 }
-                ga.sl_b          = ga.slice_b[me_];
-                ga.sl_e          = ga.slice_e[me_];
+                ga.sl_b = ga.slice_b[me_];
+                ga.sl_e = ga.slice_e[me_];
+                if (zc_ && zc_ready_) {
+                    /* reduce reads peers' src AND writes every rank's
+                     * dst directly; gather becomes a pure wait */
+                    ga.len      = 0;
+                    ga.zc_write = 1;
+                    for (uint32_t r = 0; r < n_; r++) {
+                        ga.peer_in[r] = zc_peer_src_[r] + off;
+                        ga.peer_out[r] = zc_peer_dst_[r] + off + ga.sl_b;
+                    }
+                }
                 ga.t_sw_reduce   = L[1][slot_][p] * B;
                 ga.t_sw_gather   = L[2][slot_][p] * B;
                 ga.t_prev_gather = L[2][slot_][p] * B;
@@ -1606,6 +1657,13 @@ class GatedCollTask final : public Cdna4Task {
                     if (!derive) { L[1][slot_][p]++; }
                 }
                 if (st == UCC_OK) {
+                    if (ga.zc_write) {
+                        /* dst already written by every rank's reduce:
+                         * the gather launch is a pure completion wait */
+                        for (uint32_t r = 0; r < n_; r++) {
+                            ga.slice_e[r] = ga.slice_b[r];
+                        }
+                    }
                     st = ec_hip::staged_gather(ga, comp_s);
                     if (!derive) { L[2][slot_][p]++; }
                 }
@@ -1726,6 +1784,7 @@ class GatedCollTask final : public Cdna4Task {
      * kernel degenerates to a pure signal (no staging copy). */
     bool               zc_ = false, zc_ready_ = false;
     const uint8_t     *zc_peer_src_[ec_hip::kMaxRanks] = {};
+    uint8_t           *zc_peer_dst_[ec_hip::kMaxRanks] = {};
     std::vector<void *> zc_mapped_;
 
   public:
