@@ -87,10 +87,23 @@ _PORT_SALT = [0]
 
 
 def _run(worker, world=2, timeout=180):
+    """Launch `world` spawned processes; one retry on a bootstrap
+    timeout (the gloo TCPStore rendezvous rarely wedges on busy CI
+    hosts — a fresh port recovers)."""
+    try:
+        return _run_once(worker, world, timeout)
+    except TimeoutError:
+        return _run_once(worker, world, timeout)
+
+
+def _run_once(worker, world=2, timeout=180):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    _PORT_SALT[0] += 7
-    port = 29531 + (os.getpid() + _PORT_SALT[0]) % 2000
+    import time
+
+    _PORT_SALT[0] += 17
+    port = 20000 + (os.getpid() * 13 + _PORT_SALT[0] +
+                    int(time.time() * 10)) % 20000
     procs = [ctx.Process(target=worker, args=(r, world, port, q))
              for r in range(world)]
     for p in procs:
