@@ -1239,7 +1239,10 @@ class GatedCollTask final : public Cdna4Task {
         if (st != UCC_OK) {
             return st;
         }
-        if (!zc_ready_ && ct_ == UCC_COLL_TYPE_ALLREDUCE &&
+        if (!zc_ready_ &&
+            (ct_ == UCC_COLL_TYPE_ALLREDUCE ||
+             ct_ == UCC_COLL_TYPE_REDUCE_SCATTER ||
+             ct_ == UCC_COLL_TYPE_ALLGATHER) &&
             (a_.flags & UCC_COLL_ARGS_FLAG_PERSISTENT) &&
             Config::instance().get_bool("TL_CDNA4", "ZCOPY", true)) {
             zc_ = true;
@@ -1688,6 +1691,13 @@ class GatedCollTask final : public Cdna4Task {
                 ga.sl_b   = b > e ? 0 : b - off;
                 ga.sl_e   = b > e ? 0 : e - off;
                 ga.my_out = dbuf_ + (b >= s0 && b <= s1 ? b - s0 : 0);
+                if (zc_ && zc_ready_) {
+                    /* reduce reads peers' user src directly */
+                    ga.len = 0;
+                    for (uint32_t r = 0; r < n_; r++) {
+                        ga.peer_in[r] = zc_peer_src_[r] + off;
+                    }
+                }
                 ga.t_sw_reduce   = L[1][slot_][p] * B;
                 ga.t_sw_gather   = L[2][slot_][p] * B;
                 ga.t_prev_gather = 0;
@@ -1712,8 +1722,14 @@ class GatedCollTask final : public Cdna4Task {
                 ga.src = sbuf_ + off;
                 ga.len = mylen;
                 ga.dst = dbuf_;
+                const bool agzc = zc_ && zc_ready_;
+                if (agzc) {
+                    ga.len = 0; /* stage = pure signal */
+                }
                 for (uint32_t r = 0; r < n_; r++) {
-                    ga.peer_out[r] = tt_->area(r, slot_, p, 0);
+                    ga.peer_out[r] =
+                        agzc ? (const void *)(zc_peer_src_[r] + off)
+                             : (const void *)tt_->area(r, slot_, p, 0);
                     size_t l = off >= cnt_[r] ? 0
                                : cnt_[r] - off < gran_ ? cnt_[r] - off
                                                        : gran_;

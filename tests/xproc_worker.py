@@ -210,7 +210,40 @@ def main():
         torch.testing.assert_close(
             rsd.cpu(), exp[vd[rank]:vd[rank] + vcnts[rank]], rtol=1e-5,
             atol=1e-4)
-    results.append("gated_rs_ag_a2a_v")
+    # 3a3. zero-copy persistent reduce_scatter / allgather (2 posts:
+    # exchange on the first, direct peer reads after)
+    per = 4_000_000
+    zfull = torch.randn(world, per * world, generator=g0)
+    zsrc = zfull[rank].cuda()
+    zdst = torch.zeros(per, device="cuda")
+    zrs = c.coll_init(team, "reduce_scatter", src=zsrc.data_ptr(),
+                      dst=zdst.data_ptr(), count=per, dt=dtypes.FLOAT32,
+                      mem_type=dtypes.MEM_CUDA, flags=c.FLAG_PERSISTENT)
+    for it in range(2):
+        zsrc.copy_(zfull[rank] * (it + 1))
+        torch.cuda.synchronize()
+        wait(zrs, ctx)
+        torch.cuda.synchronize()
+        exp = zfull.sum(0) * (it + 1)
+        torch.testing.assert_close(
+            zdst.cpu(), exp[rank * per:(rank + 1) * per], rtol=1e-5,
+            atol=1e-4)
+
+    zgsrc = torch.randn(per, generator=g0)
+    asrc = (zgsrc + rank).cuda()
+    adst = torch.zeros(per * world, device="cuda")
+    zag = c.coll_init(team, "allgather", src=asrc.data_ptr(),
+                      dst=adst.data_ptr(), count=per * world,
+                      dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA,
+                      flags=c.FLAG_PERSISTENT)
+    for it in range(2):
+        asrc.copy_(zgsrc + rank + it)
+        torch.cuda.synchronize()
+        wait(zag, ctx)
+        torch.cuda.synchronize()
+        exp = torch.cat([zgsrc + r + it for r in range(world)])
+        torch.testing.assert_close(adst.cpu(), exp)
+    results.append("gated_rs_ag_a2a_v_zc")
 
     # 3b. ucc_mem_map export/import across processes: rank 0 exports a
     # device buffer, rank 1 imports and reads it over IPC.
