@@ -695,6 +695,54 @@ ucc_status_t ucc_collective_init(ucc_coll_args_t *coll_args,
         req->args.flags = 0;
     }
     size_t msgsize = coll_args_msgsize(req->args, team->rank, team->size);
+    /* asymmetric src/dst memory spaces are not staged (reference
+     * ucc_coll.c:236-246 scratch staging): reject cleanly rather than
+     * let a device kernel touch a host pointer */
+    {
+        auto fold = [](ucc_memory_type_t m) {
+            if (m == UCC_MEMORY_TYPE_ROCM) {
+                return UCC_MEMORY_TYPE_CUDA;
+            }
+            if (m == UCC_MEMORY_TYPE_ROCM_MANAGED) {
+                return UCC_MEMORY_TYPE_CUDA_MANAGED;
+            }
+            return m;
+        };
+        const ucc_coll_args_t &a = req->args;
+        bool has_both =
+            a.coll_type != UCC_COLL_TYPE_BARRIER &&
+            a.coll_type != UCC_COLL_TYPE_FANIN &&
+            a.coll_type != UCC_COLL_TYPE_FANOUT &&
+            a.coll_type != UCC_COLL_TYPE_BCAST &&
+            !(a.mask & UCC_COLL_ARGS_FIELD_FLAGS &&
+              (a.flags & UCC_COLL_ARGS_FLAG_IN_PLACE));
+        if (has_both) {
+            bool sv = a.coll_type == UCC_COLL_TYPE_ALLTOALLV ||
+                      a.coll_type == UCC_COLL_TYPE_SCATTERV;
+            bool dv = a.coll_type == UCC_COLL_TYPE_ALLTOALLV ||
+                      a.coll_type == UCC_COLL_TYPE_ALLGATHERV ||
+                      a.coll_type == UCC_COLL_TYPE_GATHERV ||
+                      a.coll_type == UCC_COLL_TYPE_REDUCE_SCATTERV;
+            ucc_memory_type_t ms =
+                fold(sv ? a.src.info_v.mem_type : a.src.info.mem_type);
+            ucc_memory_type_t md =
+                fold(dv ? a.dst.info_v.mem_type : a.dst.info.mem_type);
+            /* rooted colls: the non-root side may pass only one valid
+             * buffer; only reject when both sides are meaningful */
+            bool rooted = a.coll_type == UCC_COLL_TYPE_REDUCE ||
+                          a.coll_type == UCC_COLL_TYPE_GATHER ||
+                          a.coll_type == UCC_COLL_TYPE_GATHERV ||
+                          a.coll_type == UCC_COLL_TYPE_SCATTER ||
+                          a.coll_type == UCC_COLL_TYPE_SCATTERV;
+            if (!rooted && ms != md) {
+                ucc_warn("asymmetric src/dst memory types (%d vs %d) "
+                         "are not supported",
+                         (int)ms, (int)md);
+                delete req;
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+        }
+    }
     /* Zero-size fast path only where a local zero implies a global zero
      * (fixed-count colls; v-variants may be locally empty but must still
      * take part in the exchange). Reference: ucc_coll.c:191-208. */
