@@ -1242,7 +1242,8 @@ class GatedCollTask final : public Cdna4Task {
         if (!zc_ready_ &&
             (ct_ == UCC_COLL_TYPE_ALLREDUCE ||
              ct_ == UCC_COLL_TYPE_REDUCE_SCATTER ||
-             ct_ == UCC_COLL_TYPE_ALLGATHER) &&
+             ct_ == UCC_COLL_TYPE_ALLGATHER ||
+             ct_ == UCC_COLL_TYPE_ALLTOALL) &&
             (a_.flags & UCC_COLL_ARGS_FLAG_PERSISTENT) &&
             Config::instance().get_bool("TL_CDNA4", "ZCOPY", true)) {
             zc_ = true;
@@ -1749,16 +1750,25 @@ class GatedCollTask final : public Cdna4Task {
                 break;
             }
             case UCC_COLL_TYPE_ALLTOALL: {
-                /* stage per-dest cells; gather my cell from each peer */
+                /* stage per-dest cells; gather my cell from each peer.
+                 * zero-copy: gather reads peers' USER src at my block's
+                 * offset directly; stage = pure signal */
                 ga.src     = sbuf_;
                 ga.dst     = dbuf_;
-                ga.n_cells = (int)n_;
+                const bool a2zc = zc_ && zc_ready_;
+                ga.n_cells = a2zc ? 0 : (int)n_;
+                if (a2zc) {
+                    ga.len = 0; /* stage = pure signal */
+                }
                 for (uint32_t r = 0; r < n_; r++) {
                     ga.c_src_off[r] = (uint64_t)r * out_b_ + off;
                     ga.c_dst_off[r] = (uint64_t)r * cell_;
-                    ga.c_len[r]     = len;
+                    ga.c_len[r]     = a2zc ? 0 : len;
                     ga.peer_out[r] =
-                        tt_->area(r, slot_, p, 0) + me_ * cell_;
+                        a2zc ? (const void *)(zc_peer_src_[r] +
+                                              me_ * out_b_ + off)
+                             : (const void *)(tt_->area(r, slot_, p, 0) +
+                                              me_ * cell_);
                     ga.slice_b[r] = (uint64_t)r * out_b_ + off;
                     ga.slice_e[r] = ga.slice_b[r] + len;
                 }

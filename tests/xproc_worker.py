@@ -243,6 +243,22 @@ def main():
         torch.cuda.synchronize()
         exp = torch.cat([zgsrc + r + it for r in range(world)])
         torch.testing.assert_close(adst.cpu(), exp)
+    # persistent alltoall through the zero-copy path
+    za2s = torch.randn(world, per * world, generator=g0)
+    zsrc2 = za2s[rank].cuda()
+    zdst2 = torch.zeros(per * world, device="cuda")
+    za2 = c.coll_init(team, "alltoall", src=zsrc2.data_ptr(),
+                      dst=zdst2.data_ptr(), count=per * world,
+                      dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA,
+                      flags=c.FLAG_PERSISTENT)
+    for it in range(2):
+        zsrc2.copy_(za2s[rank] * (it + 1))
+        torch.cuda.synchronize()
+        wait(za2, ctx)
+        torch.cuda.synchronize()
+        exp = torch.cat([za2s[s2][rank * per:(rank + 1) * per] * (it + 1)
+                         for s2 in range(world)])
+        torch.testing.assert_close(zdst2.cpu(), exp)
     results.append("gated_rs_ag_a2a_v_zc")
 
     # 3b. ucc_mem_map export/import across processes: rank 0 exports a
