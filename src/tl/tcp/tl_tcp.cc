@@ -619,6 +619,130 @@ class TcpAllreduceTask final : public TcpTask {
     size_t             mask_ = 1;
 };
 
+
+/* ---- allreduce SRA: ring reduce-scatter then ring allgather — the
+ * bandwidth-optimal large-message algorithm (reference tl/ucp
+ * allreduce_sra_knomial.c role, ring-flavoured): total wire traffic
+ * 2·S·(n-1)/n per rank vs recursive doubling's S·log2(n). Selected for
+ * messages >= UCC_TL_TCP_SRA_MIN (64 KiB default). */
+class TcpAllreduceSraTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        dt_    = a_.dst.info.datatype;
+        op_    = a_.op;
+        dtsz_  = ucc_dt_size(dt_);
+        count_ = a_.dst.info.count;
+        dst_   = (uint8_t *)a_.dst.info.buffer;
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        if (!inplace) {
+            ec_cpu::copy(dst_, a_.src.info.buffer, count_ * dtsz_);
+        }
+        /* n contiguous element-blocks (last takes the tail) */
+        per_ = count_ / n_;
+        if (per_ == 0) {
+            /* degenerate: fall back semantics = everyone owns block 0 */
+            per_ = 0;
+        }
+        tmp_.resize((per_ + count_ % n_) * dtsz_);
+        round_  = 0;
+        phase_  = 0;
+        in_rs_  = true;
+        status  = UCC_INPROGRESS;
+        return progress_();
+    }
+
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    uint64_t blk_off(uint32_t b) const { return (uint64_t)b * per_; }
+    uint64_t blk_cnt(uint32_t b) const
+    {
+        return b == n_ - 1 ? count_ - (uint64_t)(n_ - 1) * per_ : per_;
+    }
+
+    ucc_status_t progress_()
+    {
+        const uint32_t right = (me_ + 1) % n_;
+        const uint32_t left  = (me_ + n_ - 1) % n_;
+        /* phase A: ring reduce-scatter. Step k: send block
+         * (me - k), receive block (me - k - 1) and reduce into dst. */
+        while (in_rs_ && round_ < (int)n_ - 1) {
+            if (phase_ == 0) {
+                uint32_t sb = (me_ + n_ - round_) % n_;
+                uint32_t rb = (me_ + n_ - round_ - 1) % n_;
+                if (blk_cnt(sb)) {
+                    send_to(right, (uint32_t)round_,
+                            dst_ + blk_off(sb) * dtsz_,
+                            blk_cnt(sb) * dtsz_);
+                }
+                if (blk_cnt(rb)) {
+                    recv_from(left, (uint32_t)round_, tmp_.data(),
+                              blk_cnt(rb) * dtsz_);
+                }
+                phase_ = 1;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            uint32_t rb = (me_ + n_ - round_ - 1) % n_;
+            if (blk_cnt(rb)) {
+                const bool last = round_ == (int)n_ - 2;
+                const void *srcs[2] = {dst_ + blk_off(rb) * dtsz_,
+                                       tmp_.data()};
+                ec_cpu::reduce(dst_ + blk_off(rb) * dtsz_, srcs, 2,
+                               blk_cnt(rb), dt_, 
+                               op_ == UCC_OP_AVG ? UCC_OP_SUM : op_,
+                               (last && op_ == UCC_OP_AVG) ? 1.0 / n_
+                                                           : 1.0);
+            }
+            phase_ = 0;
+            round_++;
+        }
+        if (in_rs_) {
+            in_rs_ = false;
+            round_ = 0;
+        }
+        /* phase B: ring allgather of the reduced blocks */
+        while (round_ < (int)n_ - 1) {
+            if (phase_ == 0) {
+                uint32_t sb = (me_ + 1 + n_ - round_) % n_;
+                uint32_t rb = (me_ + n_ - round_) % n_;
+                if (blk_cnt(sb)) {
+                    send_to(right, 64 + (uint32_t)round_,
+                            dst_ + blk_off(sb) * dtsz_,
+                            blk_cnt(sb) * dtsz_);
+                }
+                if (blk_cnt(rb)) {
+                    recv_from(left, 64 + (uint32_t)round_,
+                              dst_ + blk_off(rb) * dtsz_,
+                              blk_cnt(rb) * dtsz_);
+                }
+                phase_ = 1;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            phase_ = 0;
+            round_++;
+        }
+        return UCC_OK;
+    }
+
+    ucc_datatype_t     dt_ = UCC_DT_FLOAT32;
+    ucc_reduction_op_t op_ = UCC_OP_SUM;
+    size_t             dtsz_ = 4;
+    uint64_t           count_ = 0, per_ = 0;
+    uint8_t           *dst_ = nullptr;
+    std::vector<uint8_t> tmp_;
+    bool               in_rs_ = true;
+};
+
 /* ---- bcast: binomial tree from root. With an ACTIVE_SET
  * ({start, stride, size}, reference ucc.h active_set + tl/ucp active-set
  * bcast), the tree runs over the strided subset only; the wire tag comes
@@ -1308,6 +1432,22 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
         return [](Context *c, TcpTlTeam *tt, const ucc_coll_args_t &a)
                    -> Task * { return new T(c, tt, a); };
     };
+    {
+        size_t sra_min = Config::instance().get_size("TL_TCP", "SRA_MIN",
+                                                     64 * 1024);
+        ScoreRange r;
+        r.start    = sra_min;
+        r.end      = SIZE_MAX;
+        r.score    = sc + 1;
+        r.tl_name  = "tcp";
+        r.alg_name = "sra_ring";
+        r.init     = [self](const ucc_coll_args_t &args, Team *t2,
+                        Task **task) -> ucc_status_t {
+            *task = new TcpAllreduceSraTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_ALLREDUCE, UCC_MEMORY_TYPE_HOST, r);
+    }
     add(UCC_COLL_TYPE_ALLREDUCE, mk((TcpAllreduceTask *)nullptr));
     add(UCC_COLL_TYPE_BCAST, mk((TcpBcastTask *)nullptr));
     add(UCC_COLL_TYPE_BARRIER, mk((TcpBarrierTask *)nullptr));

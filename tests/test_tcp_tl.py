@@ -27,6 +27,17 @@ for count in (7, 1024, 100_003):
     for o in outs:
         np.testing.assert_allclose(o, exp, rtol=1e-5, atol=1e-5)
 
+# SRA path (>=64 KiB) with AVG and MAX
+arrs = [(rng.random(120_000) - 0.5).astype(np.float32) for _ in range(n)]
+outs = job.allreduce_np(arrs, op=dtypes.OP_AVG)
+exp = np.mean(arrs, axis=0)
+for o in outs:
+    np.testing.assert_allclose(o, exp, rtol=1e-5, atol=1e-6)
+outs = job.allreduce_np(arrs, op=dtypes.OP_MAX)
+exp = np.max(arrs, axis=0)
+for o in outs:
+    np.testing.assert_array_equal(o, exp)
+
 # bcast
 bufs = [np.zeros(5000, np.float64) for _ in range(n)]
 bufs[2][:] = rng.random(5000)
