@@ -751,7 +751,11 @@ ucc_status_t ucc_collective_init(ucc_coll_args_t *coll_args,
                         req->args.coll_type == UCC_COLL_TYPE_GATHERV ||
                         req->args.coll_type == UCC_COLL_TYPE_SCATTERV ||
                         req->args.coll_type == UCC_COLL_TYPE_REDUCE_SCATTERV;
-    if (msgsize == 0 && !v_coll &&
+    const ucc_generic_dt_ops_t *ncg =
+        ucc_dt_generic_ops(req->args.src.info.datatype);
+    const bool noncontig_gen =
+        ncg && !(ncg->flags & UCC_GENERIC_DT_OPS_FLAG_CONTIG);
+    if (msgsize == 0 && !v_coll && !noncontig_gen &&
         req->args.coll_type != UCC_COLL_TYPE_BARRIER &&
         req->args.coll_type != UCC_COLL_TYPE_FANIN &&
         req->args.coll_type != UCC_COLL_TYPE_FANOUT) {

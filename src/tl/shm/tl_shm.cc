@@ -126,6 +126,12 @@ class ShmTlTeam final : public TlTeam {
                      * to tl/tcp (score fallback) */
                     return UCC_ERR_NOT_SUPPORTED;
                 }
+                const ucc_generic_dt_ops_t *g =
+                    ucc_dt_generic_ops(args.src.info.datatype);
+                if (g && !(g->flags & UCC_GENERIC_DT_OPS_FLAG_CONTIG)) {
+                    /* non-contig generic dtypes: tcp pack/unpack path */
+                    return UCC_ERR_NOT_SUPPORTED;
+                }
                 *task = new ShmCollTask(t->ctx, self, args);
                 return UCC_OK;
             };

@@ -43,6 +43,7 @@ struct RecvOp {
 struct UnexpMsg {
     uint64_t             tag;
     std::vector<uint8_t> data;
+    bool                 done = false; /* payload fully received */
 };
 
 /* One peer connection. */
@@ -62,6 +63,21 @@ class Conn {
     size_t                pay_got  = 0;
 
     void progress();
+
+    /* steal a COMPLETED unexpected message by tag (dynamic-size recv:
+     * the sender determines the length — used by the pack/unpack
+     * generic-datatype path). */
+    bool take_unexp(uint64_t tag, std::vector<uint8_t> *out)
+    {
+        for (auto it = unexp.begin(); it != unexp.end(); ++it) {
+            if (it->tag == tag && it->done) {
+                *out = std::move(it->data);
+                unexp.erase(it);
+                return true;
+            }
+        }
+        return false;
+    }
 
   private:
     void match_header();

@@ -137,6 +137,15 @@ void Conn::progress()
         if (cur_recv) {
             cur_recv->done = true;
         }
+        if (cur_un) {
+            /* mark the queue entry complete (take_unexp eligibility) */
+            for (auto &u : unexp) {
+                if (&u.data == cur_un) {
+                    u.done = true;
+                    break;
+                }
+            }
+        }
         cur_recv = nullptr;
         cur_un   = nullptr;
         hdr_got  = 0;
@@ -757,6 +766,45 @@ class TcpBcastTask final : public TcpTask {
         begin();
         bytes_ = a_.src.info.count * ucc_dt_size(a_.src.info.datatype);
         buf_   = (uint8_t *)a_.src.info.buffer;
+        /* non-contiguous generic datatype: move PACKED bytes through the
+         * tree; root packs once, every receiver unpacks and forwards the
+         * packed image (reference generic-dt pack/unpack cb semantics) */
+        gdt_ = ucc_dt_generic_ops(a_.src.info.datatype);
+        if (gdt_ && (gdt_->flags & UCC_GENERIC_DT_OPS_FLAG_CONTIG)) {
+            gdt_ = nullptr; /* contig generics act like plain bytes */
+        }
+        if (gdt_) {
+            if (!gdt_->ops.start_pack || !gdt_->ops.start_unpack ||
+                !gdt_->ops.packed_size || !gdt_->ops.pack ||
+                !gdt_->ops.unpack) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            if (me_ == (uint32_t)a_.root) {
+                void *obj = gdt_->ops.start_pack(
+                    gdt_->cookie, a_.src.info.buffer, a_.src.info.count);
+                size_t psz = gdt_->ops.packed_size(obj);
+                packed_.resize(psz);
+                size_t got = 0;
+                while (got < psz) {
+                    size_t len = psz - got;
+                    if (gdt_->ops.pack(obj, got, packed_.data() + got,
+                                       &len) != UCC_OK ||
+                        len == 0) {
+                        if (gdt_->ops.finish) {
+                            gdt_->ops.finish(obj);
+                        }
+                        return UCC_ERR_NO_MESSAGE;
+                    }
+                    got += len;
+                }
+                if (gdt_->ops.finish) {
+                    gdt_->ops.finish(obj);
+                }
+                buf_   = packed_.data();
+                bytes_ = psz;
+            }
+            /* receivers learn the size from the wire header */
+        }
         if (a_.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) {
             /* subset colls must not consume the team-wide sequence:
              * non-members never post, so bumping it would desync tags */
@@ -816,11 +864,37 @@ class TcpBcastTask final : public TcpTask {
                 while (hb * 2 <= vr_) {
                     hb *= 2;
                 }
-                recv_from(to_team(vr_ - hb), 0, buf_, bytes_);
+                if (gdt_) {
+                    gparent_ = to_team(vr_ - hb);
+                } else {
+                    recv_from(to_team(vr_ - hb), 0, buf_, bytes_);
+                }
             }
             phase_ = 1;
         }
         if (phase_ == 1) {
+            if (gdt_ && vr_ != 0 && packed_.empty()) {
+                /* dynamic-size packed recv: steal the completed
+                 * unexpected message (sender sized it) */
+                tt_->progress();
+                if (!tt_->conns_[gparent_].take_unexp(
+                        tt_->mktag(seq_, 0), &packed_)) {
+                    return UCC_INPROGRESS;
+                }
+                /* unpack into my user buffer */
+                void *obj = gdt_->ops.start_unpack(
+                    gdt_->cookie, a_.src.info.buffer, a_.src.info.count);
+                ucc_status_t us = gdt_->ops.unpack(
+                    obj, 0, packed_.data(), packed_.size());
+                if (gdt_->ops.finish) {
+                    gdt_->ops.finish(obj);
+                }
+                if (us != UCC_OK) {
+                    return us;
+                }
+                buf_   = packed_.data();
+                bytes_ = packed_.size();
+            }
             if (!ops_done()) {
                 return UCC_INPROGRESS;
             }
@@ -846,6 +920,9 @@ class TcpBcastTask final : public TcpTask {
     uint8_t *buf_   = nullptr;
     uint32_t vr_    = 0, as_n_ = 0, root_i_ = 0;
     std::vector<uint32_t> set_;
+    const ucc_generic_dt_ops_t *gdt_ = nullptr;
+    std::vector<uint8_t>        packed_;
+    uint32_t                    gparent_ = 0;
 };
 
 /* ---- barrier / fanin / fanout: binomial fanin to 0 then fanout */
@@ -1420,8 +1497,15 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
         r.score    = sc;
         r.tl_name  = "tcp";
         r.alg_name = "p2p";
-        r.init     = [self, maker](const ucc_coll_args_t &args, Team *t,
-                               Task **task) -> ucc_status_t {
+        r.init     = [self, maker, ct](const ucc_coll_args_t &args,
+                                   Team *t, Task **task) -> ucc_status_t {
+            const ucc_generic_dt_ops_t *g =
+                ucc_dt_generic_ops(args.src.info.datatype);
+            if (g && !(g->flags & UCC_GENERIC_DT_OPS_FLAG_CONTIG) &&
+                ct != UCC_COLL_TYPE_BCAST) {
+                /* pack/unpack movement implemented for bcast */
+                return UCC_ERR_NOT_SUPPORTED;
+            }
             *task = maker(t->ctx, self, args);
             return UCC_OK;
         };
