@@ -925,6 +925,94 @@ class TcpBcastTask final : public TcpTask {
     uint32_t                    gparent_ = 0;
 };
 
+
+/* ---- bcast SAG: root scatters blocks around the ring position space,
+ * then a ring allgather completes every rank — 2·S·(n-1)/n wire traffic
+ * vs the binomial tree's S·log2(n) (reference bcast sag_knomial role).
+ * Selected for messages >= UCC_TL_TCP_SAG_MIN (64 KiB default). */
+class TcpBcastSagTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        bytes_ = a_.src.info.count * ucc_dt_size(a_.src.info.datatype);
+        buf_   = (uint8_t *)a_.src.info.buffer;
+        root_  = (uint32_t)a_.root;
+        per_   = bytes_ / n_;
+        phase_ = 0;
+        round_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    uint64_t blk_off(uint32_t b) const { return (uint64_t)b * per_; }
+    uint64_t blk_len(uint32_t b) const
+    {
+        return b == n_ - 1 ? bytes_ - (uint64_t)(n_ - 1) * per_ : per_;
+    }
+
+    ucc_status_t progress_()
+    {
+        /* virtual ranks: vr 0 = root */
+        const uint32_t vr    = (me_ + n_ - root_) % n_;
+        const uint32_t right = (me_ + 1) % n_;
+        const uint32_t left  = (me_ + n_ - 1) % n_;
+        if (phase_ == 0) { /* scatter: root sends block vr to rank vr */
+            if (vr == 0) {
+                for (uint32_t v = 1; v < n_; v++) {
+                    if (blk_len(v)) {
+                        send_to((v + root_) % n_, 0, buf_ + blk_off(v),
+                                blk_len(v));
+                    }
+                }
+            } else if (blk_len(vr)) {
+                recv_from(root_, 0, buf_ + blk_off(vr), blk_len(vr));
+            }
+            phase_ = 1;
+        }
+        if (phase_ == 1) {
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            phase_ = 2;
+        }
+        /* ring allgather over virtual positions: step k, send the block
+         * I received k steps ago to the right, take one from the left */
+        while (round_ < (int)n_ - 1) {
+            if (phase_ == 2) {
+                uint32_t sb = (vr + n_ - round_) % n_;
+                uint32_t rb = (vr + n_ - round_ - 1) % n_;
+                if (blk_len(sb)) {
+                    send_to(right, 16 + (uint32_t)round_,
+                            buf_ + blk_off(sb), blk_len(sb));
+                }
+                if (blk_len(rb)) {
+                    recv_from(left, 16 + (uint32_t)round_,
+                              buf_ + blk_off(rb), blk_len(rb));
+                }
+                phase_ = 3;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            phase_ = 2;
+            round_++;
+        }
+        return UCC_OK;
+    }
+
+    size_t   bytes_ = 0;
+    uint64_t per_   = 0;
+    uint8_t *buf_   = nullptr;
+    uint32_t root_  = 0;
+};
+
 /* ---- barrier / fanin / fanout: binomial fanin to 0 then fanout */
 class TcpBarrierTask final : public TcpTask {
   public:
@@ -1010,6 +1098,125 @@ class TcpBarrierTask final : public TcpTask {
 
     uint8_t token_ = 1;
     bool    fanin_ = true, fanout_ = true;
+};
+
+
+/* ---- reduce SRG: ring reduce-scatter then linear gather to the root —
+ * large-message reduce with 2·S·(n-1)/n wire traffic (reference
+ * reduce srg_knomial role). Selected >= UCC_TL_TCP_SRG_MIN (64 KiB). */
+class TcpReduceSrgTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        dt_    = a_.src.info.datatype;
+        op_    = a_.op;
+        dtsz_  = ucc_dt_size(dt_);
+        count_ = a_.src.info.count;
+        root_  = (uint32_t)a_.root;
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        if (me_ == root_) {
+            dst_ = (uint8_t *)a_.dst.info.buffer;
+            if (!inplace) {
+                ec_cpu::copy(dst_, a_.src.info.buffer, count_ * dtsz_);
+            }
+            work_ = dst_;
+        } else {
+            /* non-roots reduce in a private copy (src is read-only) */
+            priv_.resize(count_ * dtsz_);
+            ec_cpu::copy(priv_.data(), a_.src.info.buffer,
+                         count_ * dtsz_);
+            work_ = priv_.data();
+        }
+        per_ = count_ / n_;
+        tmp_.resize((per_ + count_ % n_) * dtsz_);
+        round_ = 0;
+        phase_ = 0;
+        in_rs_ = true;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    uint64_t blk_off(uint32_t b) const { return (uint64_t)b * per_; }
+    uint64_t blk_cnt(uint32_t b) const
+    {
+        return b == n_ - 1 ? count_ - (uint64_t)(n_ - 1) * per_ : per_;
+    }
+
+    ucc_status_t progress_()
+    {
+        const uint32_t right = (me_ + 1) % n_;
+        const uint32_t left  = (me_ + n_ - 1) % n_;
+        while (in_rs_ && round_ < (int)n_ - 1) {
+            if (phase_ == 0) {
+                uint32_t sb = (me_ + n_ - round_) % n_;
+                uint32_t rb = (me_ + n_ - round_ - 1) % n_;
+                if (blk_cnt(sb)) {
+                    send_to(right, (uint32_t)round_,
+                            work_ + blk_off(sb) * dtsz_,
+                            blk_cnt(sb) * dtsz_);
+                }
+                if (blk_cnt(rb)) {
+                    recv_from(left, (uint32_t)round_, tmp_.data(),
+                              blk_cnt(rb) * dtsz_);
+                }
+                phase_ = 1;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            uint32_t rb = (me_ + n_ - round_ - 1) % n_;
+            if (blk_cnt(rb)) {
+                const bool last = round_ == (int)n_ - 2;
+                const void *srcs[2] = {work_ + blk_off(rb) * dtsz_,
+                                       tmp_.data()};
+                ec_cpu::reduce(work_ + blk_off(rb) * dtsz_, srcs, 2,
+                               blk_cnt(rb), dt_,
+                               op_ == UCC_OP_AVG ? UCC_OP_SUM : op_,
+                               (last && op_ == UCC_OP_AVG) ? 1.0 / n_
+                                                           : 1.0);
+            }
+            phase_ = 0;
+            round_++;
+        }
+        if (in_rs_) {
+            in_rs_ = false;
+            phase_ = 0;
+            /* gather: each rank owns block (me+1)%n; send it to root */
+            uint32_t own = (me_ + 1) % n_;
+            if (me_ == root_) {
+                for (uint32_t r = 0; r < n_; r++) {
+                    uint32_t b = (r + 1) % n_;
+                    if (r != me_ && blk_cnt(b)) {
+                        recv_from(r, 64, dst_ + blk_off(b) * dtsz_,
+                                  blk_cnt(b) * dtsz_);
+                    }
+                }
+            } else if (blk_cnt(own)) {
+                send_to(root_, 64, work_ + blk_off(own) * dtsz_,
+                        blk_cnt(own) * dtsz_);
+            }
+        }
+        if (!ops_done()) {
+            return UCC_INPROGRESS;
+        }
+        clear_ops();
+        return UCC_OK;
+    }
+
+    ucc_datatype_t     dt_ = UCC_DT_FLOAT32;
+    ucc_reduction_op_t op_ = UCC_OP_SUM;
+    size_t             dtsz_ = 4;
+    uint64_t           count_ = 0, per_ = 0;
+    uint32_t           root_ = 0;
+    uint8_t           *dst_ = nullptr, *work_ = nullptr;
+    std::vector<uint8_t> priv_, tmp_;
+    bool               in_rs_ = true;
 };
 
 /* ---- allgather(v): ring */
@@ -1533,6 +1740,28 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
         map.add(UCC_COLL_TYPE_ALLREDUCE, UCC_MEMORY_TYPE_HOST, r);
     }
     add(UCC_COLL_TYPE_ALLREDUCE, mk((TcpAllreduceTask *)nullptr));
+    {
+        size_t sag_min = Config::instance().get_size("TL_TCP", "SAG_MIN",
+                                                     64 * 1024);
+        ScoreRange r;
+        r.start    = sag_min;
+        r.end      = SIZE_MAX;
+        r.score    = sc + 1;
+        r.tl_name  = "tcp";
+        r.alg_name = "sag_ring";
+        r.init     = [self](const ucc_coll_args_t &args, Team *t2,
+                        Task **task) -> ucc_status_t {
+            const ucc_generic_dt_ops_t *g =
+                ucc_dt_generic_ops(args.src.info.datatype);
+            if ((args.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) ||
+                (g && !(g->flags & UCC_GENERIC_DT_OPS_FLAG_CONTIG))) {
+                return UCC_ERR_NOT_SUPPORTED; /* binomial handles these */
+            }
+            *task = new TcpBcastSagTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_BCAST, UCC_MEMORY_TYPE_HOST, r);
+    }
     add(UCC_COLL_TYPE_BCAST, mk((TcpBcastTask *)nullptr));
     add(UCC_COLL_TYPE_BARRIER, mk((TcpBarrierTask *)nullptr));
     add(UCC_COLL_TYPE_FANIN, mk((TcpBarrierTask *)nullptr));
@@ -1541,6 +1770,27 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
     add(UCC_COLL_TYPE_ALLGATHERV, mk((TcpAllgatherTask *)nullptr));
     add(UCC_COLL_TYPE_ALLTOALL, mk((TcpAlltoallTask *)nullptr));
     add(UCC_COLL_TYPE_ALLTOALLV, mk((TcpAlltoallTask *)nullptr));
+    {
+        size_t srg_min = Config::instance().get_size("TL_TCP", "SRG_MIN",
+                                                     64 * 1024);
+        ScoreRange r;
+        r.start    = srg_min;
+        r.end      = SIZE_MAX;
+        r.score    = sc + 1;
+        r.tl_name  = "tcp";
+        r.alg_name = "srg_ring";
+        r.init     = [self](const ucc_coll_args_t &args, Team *t2,
+                        Task **task) -> ucc_status_t {
+            const ucc_generic_dt_ops_t *g =
+                ucc_dt_generic_ops(args.src.info.datatype);
+            if ((args.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) || g) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpReduceSrgTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_REDUCE, UCC_MEMORY_TYPE_HOST, r);
+    }
     add(UCC_COLL_TYPE_REDUCE, mk((TcpRootedTask *)nullptr));
     add(UCC_COLL_TYPE_REDUCE_SCATTER, mk((TcpRootedTask *)nullptr));
     add(UCC_COLL_TYPE_REDUCE_SCATTERV, mk((TcpRootedTask *)nullptr));

@@ -38,6 +38,17 @@ exp = np.max(arrs, axis=0)
 for o in outs:
     np.testing.assert_array_equal(o, exp)
 
+# bcast large (SAG ring path) from a non-zero root
+big = [np.zeros(60_000, np.float64) for _ in range(n)]
+big[3][:] = rng.random(60_000)
+exp = big[3].copy()
+reqs = job.coll("bcast", [
+    dict(src=b.ctypes.data, dst=0, count=60_000, dt=dtypes.FLOAT64,
+         root=3) for b in big])
+job.run(reqs)
+for b in big:
+    np.testing.assert_array_equal(b, exp)
+
 # bcast
 bufs = [np.zeros(5000, np.float64) for _ in range(n)]
 bufs[2][:] = rng.random(5000)
