@@ -101,6 +101,17 @@ for r in range(n):
     np.testing.assert_allclose(dsts[r], exp[r * per:(r + 1) * per],
                                rtol=1e-5, atol=1e-5)
 
+# reduce large (SRG ring path) to a non-zero root
+rsrcs = [(rng.random(90_000) - 0.5).astype(np.float32) for _ in range(n)]
+rdst = np.zeros(90_000, np.float32)
+reqs = job.coll("reduce", [
+    dict(src=rsrcs[r].ctypes.data,
+         dst=rdst.ctypes.data if r == 2 else 0, count=90_000,
+         dt=dtypes.FLOAT32, root=2) for r in range(n)])
+job.run(reqs)
+np.testing.assert_allclose(rdst, np.sum(rsrcs, axis=0), rtol=1e-5,
+                           atol=1e-5)
+
 # barrier
 reqs = job.coll("barrier", [dict(src=0, dst=0, count=0, dt=dtypes.FLOAT32)
                             for _ in range(n)])
