@@ -1546,9 +1546,37 @@ class GatedCollTask final : public Cdna4Task {
             publish(2);
             phase_ = 12;
         }
-        if (phase_ == 12) { /* all peers consumed the blobs */
+        if (phase_ == 12) { /* blobs consumed -> publish my ok verdict */
             if (!all_ge(2)) {
                 return UCC_INPROGRESS;
+            }
+            /* zero-copy must be ALL-or-nothing: a rank that fell back
+             * (export/import/alignment failure) would stage while
+             * zero-copy peers read its never-staged areas. Consensus
+             * round: everyone posts ok/fail, mode = AND of all. */
+            uint64_t ok = zc_ ? 0x5a43u : 0;
+            if (hipMemcpy(tt_->area(me_, slot_, 1, 0), &ok, sizeof(ok),
+                          hipMemcpyHostToDevice) != hipSuccess) {
+                return UCC_ERR_NO_RESOURCE;
+            }
+            publish(3);
+            phase_ = 13;
+        }
+        if (phase_ == 13) { /* read all verdicts */
+            if (!all_ge(3)) {
+                return UCC_INPROGRESS;
+            }
+            for (uint32_t r = 0; r < n_ && zc_; r++) {
+                uint64_t ok = 0;
+                if (hipMemcpy(&ok, tt_->area(r, slot_, 1, 0), sizeof(ok),
+                              hipMemcpyDeviceToHost) != hipSuccess ||
+                    ok != 0x5a43u) {
+                    zc_ = false;
+                }
+            }
+            if (!zc_) {
+                ucc_warn("zero-copy disabled by team consensus, "
+                         "using the staged pipeline");
             }
             zc_ready_ = true;
             close_slot();
