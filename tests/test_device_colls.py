@@ -317,3 +317,50 @@ def test_allreduce_device_complex(job):
     ])
     for d in dsts:
         torch.testing.assert_close(d.cpu(), expected, rtol=1e-5, atol=1e-4)
+
+
+def test_allgatherv_device_zero_count(job):
+    torch.manual_seed(21)
+    n = job.n
+    cnts = [0 if r == 1 % n else (r + 1) * 500 for r in range(n)]
+    dsps = np.concatenate([[0], np.cumsum(cnts)[:-1]]).tolist()
+    total = int(sum(cnts))
+    srcs = [torch.randn(max(cnts[r], 1), device="cuda") for r in range(n)]
+    dsts = [torch.zeros(max(total, 1), device="cuda") for _ in range(n)]
+    _run_device(job, "allgatherv", [
+        dict(src=srcs[r].data_ptr(), dst=dsts[r].data_ptr(),
+             count=cnts[r], dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA,
+             dst_counts=cnts, dst_displs=dsps)
+        for r in range(n)
+    ])
+    expected = torch.cat([srcs[r].cpu()[:cnts[r]] for r in range(n)])
+    for d in dsts:
+        torch.testing.assert_close(d.cpu()[:total], expected)
+
+
+def test_alltoallv_device_zero_row(job):
+    torch.manual_seed(22)
+    n = job.n
+    scnt = [[0] * n if r == 0 else [(r) * (d + 1) * 300 for d in range(n)]
+            for r in range(n)]
+    rcnt = [[scnt[s][r] for s in range(n)] for r in range(n)]
+    sdsp = [np.concatenate([[0], np.cumsum(c)[:-1]]).tolist()
+            for c in scnt]
+    rdsp = [np.concatenate([[0], np.cumsum(c)[:-1]]).tolist()
+            for c in rcnt]
+    srcs = [torch.randn(max(sum(scnt[r]), 1), device="cuda")
+            for r in range(n)]
+    dsts = [torch.zeros(max(sum(rcnt[r]), 1), device="cuda")
+            for r in range(n)]
+    _run_device(job, "alltoallv", [
+        dict(src=srcs[r].data_ptr(), dst=dsts[r].data_ptr(), count=0,
+             dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA,
+             src_counts=scnt[r], src_displs=sdsp[r],
+             dst_counts=rcnt[r], dst_displs=rdsp[r])
+        for r in range(n)
+    ])
+    for r in range(n):
+        for s in range(n):
+            got = dsts[r].cpu()[rdsp[r][s]:rdsp[r][s] + rcnt[r][s]]
+            exp = srcs[s].cpu()[sdsp[s][r]:sdsp[s][r] + scnt[s][r]]
+            torch.testing.assert_close(got, exp)
