@@ -145,7 +145,8 @@ void ucc_lib_config_print(const ucc_lib_config_h config, void *stream,
     fprintf(f, "# %s\n", title ? title : "ucc_amd config");
     for (const auto &e : Config::instance().entries()) {
         fprintf(f, "%s=%s  # %s\n",
-                Config::key(e.component, e.name).c_str(), e.dflt.c_str(),
+                Config::key(e.component, e.name).c_str(),
+                Config::instance().get(e.component, e.name, e.dflt).c_str(),
                 e.doc.c_str());
     }
 }
@@ -186,6 +187,13 @@ ucc_status_t ucc_lib_get_attr(ucc_lib_h lib, ucc_lib_attr_t *attr)
     }
     if (attr->mask & UCC_LIB_ATTR_FIELD_COLL_TYPES) {
         attr->coll_types = UCC_COLL_TYPE_ALL;
+    }
+    if (attr->mask & UCC_LIB_ATTR_FIELD_REDUCTION_TYPES) {
+        attr->reduction_types =
+            (1u << UCC_OP_SUM) | (1u << UCC_OP_PROD) | (1u << UCC_OP_MAX) |
+            (1u << UCC_OP_MIN) | (1u << UCC_OP_LAND) | (1u << UCC_OP_LOR) |
+            (1u << UCC_OP_LXOR) | (1u << UCC_OP_BAND) |
+            (1u << UCC_OP_BOR) | (1u << UCC_OP_BXOR) | (1u << UCC_OP_AVG);
     }
     return UCC_OK;
 }
