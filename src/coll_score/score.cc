@@ -293,9 +293,11 @@ ucc_memory_type_t coll_args_mem_type(const ucc_coll_args_t &args,
         args.coll_type == UCC_COLL_TYPE_FANOUT) {
         return UCC_MEMORY_TYPE_HOST;
     }
-    if (args.coll_type == UCC_COLL_TYPE_BCAST ||
-        args.coll_type == UCC_COLL_TYPE_SCATTERV) {
-        mt = v_dst ? args.src.info_v.mem_type : args.src.info.mem_type;
+    if (args.coll_type == UCC_COLL_TYPE_SCATTERV) {
+        /* scatterv's SOURCE uses the v-union member */
+        mt = args.src.info_v.mem_type;
+    } else if (args.coll_type == UCC_COLL_TYPE_BCAST) {
+        mt = args.src.info.mem_type;
     } else {
         mt = v_dst ? args.dst.info_v.mem_type : args.dst.info.mem_type;
     }

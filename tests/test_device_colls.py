@@ -364,3 +364,34 @@ def test_alltoallv_device_zero_row(job):
             got = dsts[r].cpu()[rdsp[r][s]:rdsp[r][s] + rcnt[r][s]]
             exp = srcs[s].cpu()[sdsp[s][r]:sdsp[s][r] + scnt[s][r]]
             torch.testing.assert_close(got, exp)
+
+
+def test_gatherv_scatterv_device(job):
+    torch.manual_seed(23)
+    n, root = job.n, 0
+    cnts = [(r + 2) * 4321 for r in range(n)]
+    dsps = np.concatenate([[0], np.cumsum(cnts)[:-1]]).tolist()
+    total = int(sum(cnts))
+    srcs = [torch.randn(cnts[r], device="cuda") for r in range(n)]
+    gdst = torch.zeros(total, device="cuda")
+    _run_device(job, "gatherv", [
+        dict(src=srcs[r].data_ptr(),
+             dst=gdst.data_ptr() if r == root else 0,
+             count=cnts[r], dt=dtypes.FLOAT32, root=root,
+             mem_type=dtypes.MEM_CUDA,
+             dst_counts=cnts, dst_displs=dsps)
+        for r in range(n)
+    ])
+    expected = torch.cat([s.cpu() for s in srcs])
+    torch.testing.assert_close(gdst.cpu(), expected)
+
+    sdsts = [torch.zeros(cnts[r], device="cuda") for r in range(n)]
+    _run_device(job, "scatterv", [
+        dict(src=gdst.data_ptr() if r == root else 0,
+             dst=sdsts[r].data_ptr(), count=cnts[r], dt=dtypes.FLOAT32,
+             root=root, mem_type=dtypes.MEM_CUDA,
+             src_counts=cnts, src_displs=dsps)
+        for r in range(n)
+    ])
+    for r in range(n):
+        torch.testing.assert_close(sdsts[r].cpu(), srcs[r].cpu())
