@@ -242,7 +242,6 @@ static size_t counts_total(const ucc_coll_args_t &args, const ucc_count_t *c,
 size_t coll_args_msgsize(const ucc_coll_args_t &args, uint32_t rank,
                          uint32_t size)
 {
-    (void)rank;
     switch (args.coll_type) {
     case UCC_COLL_TYPE_BARRIER:
     case UCC_COLL_TYPE_FANIN:
@@ -264,11 +263,25 @@ size_t coll_args_msgsize(const ucc_coll_args_t &args, uint32_t rank,
         }
         return args.dst.info.count * ucc_dt_size(args.dst.info.datatype);
     case UCC_COLL_TYPE_ALLGATHERV:
+        return counts_total(args, args.dst.info_v.counts, size) *
+               ucc_dt_size(args.dst.info_v.datatype);
     case UCC_COLL_TYPE_GATHERV:
+        /* v-args are significant at the root only (UCC semantics):
+         * non-roots size by their own contiguous src */
+        if (rank != (uint32_t)args.root) {
+            return args.src.info.count *
+                   ucc_dt_size(args.src.info.datatype);
+        }
         return counts_total(args, args.dst.info_v.counts, size) *
                ucc_dt_size(args.dst.info_v.datatype);
     case UCC_COLL_TYPE_ALLTOALLV:
+        return counts_total(args, args.src.info_v.counts, size) *
+               ucc_dt_size(args.src.info_v.datatype);
     case UCC_COLL_TYPE_SCATTERV:
+        if (rank != (uint32_t)args.root) {
+            return args.dst.info.count *
+                   ucc_dt_size(args.dst.info.datatype);
+        }
         return counts_total(args, args.src.info_v.counts, size) *
                ucc_dt_size(args.src.info_v.datatype);
     case UCC_COLL_TYPE_REDUCE_SCATTERV:
@@ -281,7 +294,6 @@ size_t coll_args_msgsize(const ucc_coll_args_t &args, uint32_t rank,
 ucc_memory_type_t coll_args_mem_type(const ucc_coll_args_t &args,
                                      uint32_t rank)
 {
-    (void)rank;
     ucc_memory_type_t mt;
     bool              v_dst =
         args.coll_type == UCC_COLL_TYPE_ALLGATHERV ||
@@ -294,9 +306,14 @@ ucc_memory_type_t coll_args_mem_type(const ucc_coll_args_t &args,
         return UCC_MEMORY_TYPE_HOST;
     }
     if (args.coll_type == UCC_COLL_TYPE_SCATTERV) {
-        /* scatterv's SOURCE uses the v-union member */
-        mt = args.src.info_v.mem_type;
+        /* scatterv's SOURCE uses the v-union member — at the root;
+         * non-roots only have a plain dst */
+        mt = rank == (uint32_t)args.root ? args.src.info_v.mem_type
+                                         : args.dst.info.mem_type;
     } else if (args.coll_type == UCC_COLL_TYPE_BCAST) {
+        mt = args.src.info.mem_type;
+    } else if (args.coll_type == UCC_COLL_TYPE_GATHERV &&
+               rank != (uint32_t)args.root) {
         mt = args.src.info.mem_type;
     } else {
         mt = v_dst ? args.dst.info_v.mem_type : args.dst.info.mem_type;

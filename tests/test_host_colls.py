@@ -424,3 +424,36 @@ def test_gatherv_scatterv(job):
     job.run(reqs)
     for r in range(n):
         np.testing.assert_array_equal(sdsts[r], srcs[r])
+
+
+def test_gatherv_rootonly_vargs(job):
+    """UCC semantics: v-args are significant at the ROOT only — non-root
+    ranks pass no counts (regression: union misread crashed/flaked)."""
+    rng = np.random.default_rng(42)
+    n, root = job.n, 0
+    cnts = [(r + 1) * 21 for r in range(n)]
+    dsps = np.concatenate([[0], np.cumsum(cnts)[:-1]]).tolist()
+    total = int(sum(cnts))
+    srcs = [(rng.random(cnts[r])).astype(np.float32) for r in range(n)]
+    gdst = np.zeros(total, np.float32)
+    reqs = job.coll("gatherv", [
+        dict(src=srcs[r].ctypes.data,
+             dst=gdst.ctypes.data if r == root else 0,
+             count=cnts[r], dt=dtypes.FLOAT32, root=root,
+             **({"dst_counts": cnts, "dst_displs": dsps}
+                if r == root else {}))
+        for r in range(n)])
+    job.run(reqs)
+    np.testing.assert_array_equal(gdst, np.concatenate(srcs))
+
+    sdsts = [np.zeros(cnts[r], np.float32) for r in range(n)]
+    reqs = job.coll("scatterv", [
+        dict(src=gdst.ctypes.data if r == root else 0,
+             dst=sdsts[r].ctypes.data, count=cnts[r], dt=dtypes.FLOAT32,
+             root=root,
+             **({"src_counts": cnts, "src_displs": dsps}
+                if r == root else {}))
+        for r in range(n)])
+    job.run(reqs)
+    for r in range(n):
+        np.testing.assert_array_equal(sdsts[r], srcs[r])
