@@ -395,3 +395,25 @@ def test_gatherv_scatterv_device(job):
     ])
     for r in range(n):
         torch.testing.assert_close(sdsts[r].cpu(), srcs[r].cpu())
+
+
+def test_gatherv_device_rootonly_vargs(job):
+    """v-args significant at root only (device staged path)."""
+    torch.manual_seed(24)
+    n, root = job.n, 0
+    cnts = [(r + 1) * 777 for r in range(n)]
+    dsps = np.concatenate([[0], np.cumsum(cnts)[:-1]]).tolist()
+    total = int(sum(cnts))
+    srcs = [torch.randn(cnts[r], device="cuda") for r in range(n)]
+    gdst = torch.zeros(total, device="cuda")
+    _run_device(job, "gatherv", [
+        dict(src=srcs[r].data_ptr(),
+             dst=gdst.data_ptr() if r == root else 0,
+             count=cnts[r], dt=dtypes.FLOAT32, root=root,
+             mem_type=dtypes.MEM_CUDA,
+             **({"dst_counts": cnts, "dst_displs": dsps}
+                if r == root else {}))
+        for r in range(n)
+    ])
+    torch.testing.assert_close(gdst.cpu(),
+                               torch.cat([s.cpu() for s in srcs]))
