@@ -33,6 +33,16 @@
 namespace ucc {
 namespace {
 
+/* fire-and-forget HIP calls (cleanup paths, stream ops whose failures
+ * surface later through event queries): log-and-continue. */
+static inline void hip_warn_on_err(hipError_t e, const char *what)
+{
+    if (e != hipSuccess) {
+        ucc_warn("%s: %s", what, hipGetErrorString(e));
+    }
+}
+#define HIPWARN(expr) hip_warn_on_err((expr), #expr)
+
 using ec_hip::kMaxRanks;
 
 /* v-coll count/displacement accessors honoring the 64-bit flags
@@ -77,16 +87,16 @@ class Cdna4TlContext final : public TlContext {
     Cdna4TlContext(Context *ctx, Tl *tl, int dev) : TlContext(ctx), tl_(tl),
                                                     dev_(dev)
     {
-        hipStreamCreateWithFlags(&comp_, hipStreamNonBlocking);
-        hipStreamCreateWithFlags(&copy_, hipStreamNonBlocking);
+        HIPWARN(hipStreamCreateWithFlags(&comp_, hipStreamNonBlocking));
+        HIPWARN(hipStreamCreateWithFlags(&copy_, hipStreamNonBlocking));
     }
     ~Cdna4TlContext() override
     {
         if (comp_) {
-            hipStreamDestroy(comp_);
+            HIPWARN(hipStreamDestroy(comp_));
         }
         if (copy_) {
-            hipStreamDestroy(copy_);
+            HIPWARN(hipStreamDestroy(copy_));
         }
     }
     Tl *iface() override;
@@ -131,20 +141,20 @@ class Cdna4TlTeam final : public TlTeam {
     {
         for (uint32_t r = 0; r < (uint32_t)peers_.size(); r++) {
             if (peers_[r].ipc_s) {
-                hipIpcCloseMemHandle(peers_[r].scratch);
+                HIPWARN(hipIpcCloseMemHandle(peers_[r].scratch));
             }
             if (peers_[r].ipc_f) {
-                hipIpcCloseMemHandle(peers_[r].flags);
+                HIPWARN(hipIpcCloseMemHandle(peers_[r].flags));
             }
         }
         if (scratch_) {
-            hipFree(scratch_);
+            HIPWARN(hipFree(scratch_));
         }
         if (flags_) {
-            hipFree(flags_);
+            HIPWARN(hipFree(flags_));
         }
         if (err_host_) {
-            hipHostFree(err_host_);
+            HIPWARN(hipHostFree(err_host_));
         }
     }
 
@@ -328,7 +338,7 @@ class Cdna4Task : public Task {
     {
         for (auto ev : evs_) {
             if (ev) {
-                hipEventDestroy(ev);
+                HIPWARN(hipEventDestroy(ev));
             }
         }
     }
@@ -340,7 +350,7 @@ class Cdna4Task : public Task {
             evs_.push_back(nullptr);
         }
         if (!evs_[i]) {
-            hipEventCreateWithFlags(&evs_[i], hipEventDisableTiming);
+            HIPWARN(hipEventCreateWithFlags(&evs_[i], hipEventDisableTiming));
         }
         return evs_[i];
     }
@@ -443,7 +453,7 @@ class FusedAllreduceTask final : public Cdna4Task {
             if (st != UCC_OK) {
                 return st;
             }
-            hipEventRecord(ev(0), comp());
+            HIPWARN(hipEventRecord(ev(0), comp()));
             phase_ = 1;
         }
         if (phase_ == 1) {
@@ -844,7 +854,7 @@ class StagedTask final : public Cdna4Task {
                 }
                 bool launched = launch_compute(f, p, off, len);
                 if (launched) {
-                    hipEventRecord(ev(2), comp());
+                    HIPWARN(hipEventRecord(ev(2), comp()));
                 }
                 b_launched_ = launched;
                 phase_      = 3;
@@ -874,7 +884,7 @@ class StagedTask final : public Cdna4Task {
                     return UCC_INPROGRESS;
                 }
                 launch_gather(f, p, off, len);
-                hipEventRecord(ev(3), comp());
+                HIPWARN(hipEventRecord(ev(3), comp()));
                 phase_ = 5;
                 break;
             }
@@ -927,7 +937,7 @@ class StagedTask final : public Cdna4Task {
                 ec_hip::gather_copy(ga, copy_s());
                 copy_pending_ = true;
             }
-            hipEventRecord(ev(0 + (int)p), copy_s());
+            HIPWARN(hipEventRecord(ev(0 + (int)p), copy_s()));
             return;
         }
         case UCC_COLL_TYPE_SCATTER: {
@@ -952,7 +962,7 @@ class StagedTask final : public Cdna4Task {
                     copy_pending_ = true;
                 }
             }
-            hipEventRecord(ev(0 + (int)p), copy_s());
+            HIPWARN(hipEventRecord(ev(0 + (int)p), copy_s()));
             return;
         }
         case UCC_COLL_TYPE_ALLGATHER:
@@ -980,7 +990,7 @@ class StagedTask final : public Cdna4Task {
             hipMemcpyAsync(tt_->area(me_, slot_, p, 0), src, len,
                            hipMemcpyDeviceToDevice, copy_s());
         }
-        hipEventRecord(ev(0 + (int)p), copy_s());
+        HIPWARN(hipEventRecord(ev(0 + (int)p), copy_s()));
     }
 
     /* B compute for fragment f; returns whether a kernel was launched */
@@ -1591,7 +1601,7 @@ class GatedCollTask final : public Cdna4Task {
             if (st != UCC_OK) {
                 return st;
             }
-            hipEventRecord(ev(0), comp());
+            HIPWARN(hipEventRecord(ev(0), comp()));
             phase_ = 1;
         }
         hipError_t e = hipEventQuery(ev(0));
@@ -1848,7 +1858,7 @@ class GatedCollTask final : public Cdna4Task {
             tt_->free_pslot(pslot_);
         }
         for (void *m : zc_mapped_) {
-            hipIpcCloseMemHandle(m);
+            HIPWARN(hipIpcCloseMemHandle(m));
         }
     }
 };

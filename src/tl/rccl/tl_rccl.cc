@@ -22,6 +22,16 @@
 namespace ucc {
 namespace {
 
+/* fire-and-forget HIP calls (cleanup paths, stream ops whose failures
+ * surface later through event queries): log-and-continue. */
+static inline void hip_warn_on_err(hipError_t e, const char *what)
+{
+    if (e != hipSuccess) {
+        ucc_warn("%s: %s", what, hipGetErrorString(e));
+    }
+}
+#define HIPWARN(expr) hip_warn_on_err((expr), #expr)
+
 #define RCCLCHK(expr)                                                        \
     do {                                                                     \
         ncclResult_t _r = (expr);                                            \
@@ -37,12 +47,12 @@ class RcclTlContext final : public TlContext {
   public:
     RcclTlContext(Context *ctx, Tl *tl) : TlContext(ctx), tl_(tl)
     {
-        hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking);
+        HIPWARN(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
     }
     ~RcclTlContext() override
     {
         if (stream_) {
-            hipStreamDestroy(stream_);
+            HIPWARN(hipStreamDestroy(stream_));
         }
     }
     Tl *iface() override;
@@ -146,15 +156,15 @@ class RcclTask final : public Task {
     RcclTask(Context *ctx, RcclTlTeam *tt, const ucc_coll_args_t &args)
         : Task(ctx), tt_(tt), a_(args)
     {
-        hipEventCreateWithFlags(&ev_, hipEventDisableTiming);
+        HIPWARN(hipEventCreateWithFlags(&ev_, hipEventDisableTiming));
     }
     ~RcclTask() override
     {
         if (ev_) {
-            hipEventDestroy(ev_);
+            HIPWARN(hipEventDestroy(ev_));
         }
         if (barrier_buf_) {
-            hipFree(barrier_buf_);
+            HIPWARN(hipFree(barrier_buf_));
         }
     }
 
@@ -165,7 +175,7 @@ class RcclTask final : public Task {
             status = st;
             return st;
         }
-        hipEventRecord(ev_, stream());
+        HIPWARN(hipEventRecord(ev_, stream()));
         status = UCC_INPROGRESS;
         return UCC_OK;
     }
