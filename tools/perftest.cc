@@ -36,6 +36,18 @@
 
 using uccperf::ShmOob;
 
+#ifdef UCC_AMD_HAS_HIP
+static inline void pt_hip_warn(hipError_t e, const char *what)
+{
+    if (e != hipSuccess) {
+        fprintf(stderr, "hip warning: %s: %s\n", what,
+                hipGetErrorString(e));
+    }
+}
+#define HIPWARN(expr) pt_hip_warn((expr), #expr)
+#endif
+
+
 /* ------------------------------------------------------------------ opts */
 struct Opts {
     std::string coll = "allreduce";
@@ -148,9 +160,9 @@ struct Bufs {
                 fprintf(stderr, "hipMalloc failed\n");
                 exit(1);
             }
-            hipMemset(src, 1, bytes);
-            hipMemset(dst, 0, bytes);
-            hipDeviceSynchronize();
+            HIPWARN(hipMemset(src, 1, bytes));
+            HIPWARN(hipMemset(dst, 0, bytes));
+            HIPWARN(hipDeviceSynchronize());
 #else
             fprintf(stderr, "built without HIP\n");
             exit(1);
@@ -166,8 +178,8 @@ struct Bufs {
     {
         if (dev) {
 #ifdef UCC_AMD_HAS_HIP
-            hipFree(src);
-            hipFree(dst);
+            HIPWARN(hipFree(src));
+            HIPWARN(hipFree(dst));
 #endif
         } else {
             free(src);
@@ -441,9 +453,9 @@ static bool validate_once(const Opts &o, Rank &r, ucc_coll_req_h req,
     }
 #ifdef UCC_AMD_HAS_HIP
     if (o.mem == "cuda") {
-        hipMemcpy(r.bufs.src, h.data(), bytes, hipMemcpyHostToDevice);
-        hipMemset(r.bufs.dst, 0, bytes);
-        hipDeviceSynchronize();
+        HIPWARN(hipMemcpy(r.bufs.src, h.data(), bytes, hipMemcpyHostToDevice));
+        HIPWARN(hipMemset(r.bufs.dst, 0, bytes));
+        HIPWARN(hipDeviceSynchronize());
     } else
 #endif
     {
@@ -453,8 +465,8 @@ static bool validate_once(const Opts &o, Rank &r, ucc_coll_req_h req,
     if (ct == UCC_COLL_TYPE_BCAST) {
 #ifdef UCC_AMD_HAS_HIP
         if (o.mem == "cuda") {
-            hipMemcpy(r.bufs.src, h.data(), bytes,
-                      hipMemcpyHostToDevice);
+            HIPWARN(hipMemcpy(r.bufs.src, h.data(), bytes,
+                              hipMemcpyHostToDevice));
         }
 #endif
     }
@@ -464,10 +476,11 @@ static bool validate_once(const Opts &o, Rank &r, ucc_coll_req_h req,
     }
 #ifdef UCC_AMD_HAS_HIP
     if (o.mem == "cuda") {
-        hipDeviceSynchronize();
-        hipMemcpy(out.data(),
-                  ct == UCC_COLL_TYPE_BCAST ? r.bufs.src : r.bufs.dst,
-                  bytes, hipMemcpyDeviceToHost);
+        HIPWARN(hipDeviceSynchronize());
+        HIPWARN(hipMemcpy(out.data(),
+                          ct == UCC_COLL_TYPE_BCAST ? r.bufs.src
+                                                    : r.bufs.dst,
+                          bytes, hipMemcpyDeviceToHost));
     } else
 #endif
     {
@@ -532,9 +545,9 @@ static int run_forked_child(const Opts &o, int rank)
 #ifdef UCC_AMD_HAS_HIP
     if (o.mem == "cuda") {
         int ndev = 0;
-        hipGetDeviceCount(&ndev);
+        HIPWARN(hipGetDeviceCount(&ndev));
         if (ndev > 0) {
-            hipSetDevice(rank % ndev);
+            HIPWARN(hipSetDevice(rank % ndev));
         }
     }
 #endif
@@ -581,7 +594,7 @@ static int run_forked_child(const Opts &o, int rank)
         hipStream_t tstream = nullptr;
         ucc_ee_h    ee      = nullptr;
         if (o.triggered) {
-            hipStreamCreateWithFlags(&tstream, hipStreamNonBlocking);
+            HIPWARN(hipStreamCreateWithFlags(&tstream, hipStreamNonBlocking));
             ucc_ee_params_t ep{};
             ep.ee_type         = UCC_EE_ROCM_STREAM;
             ep.ee_context      = (void *)tstream;
@@ -604,7 +617,7 @@ static int run_forked_child(const Opts &o, int rank)
                     trig_fail = true;
                     return;
                 }
-                hipStreamSynchronize(tstream);
+                HIPWARN(hipStreamSynchronize(tstream));
                 return;
             }
 #endif
@@ -631,7 +644,7 @@ static int run_forked_child(const Opts &o, int rank)
             ucc_collective_finalize(req);
 #ifdef UCC_AMD_HAS_HIP
             if (ee) { ucc_ee_destroy(ee); ee = nullptr; }
-            if (tstream) { hipStreamDestroy(tstream); tstream = nullptr; }
+            if (tstream) { HIPWARN(hipStreamDestroy(tstream)); tstream = nullptr; }
 #endif
             g_shm_oob.max_double(0.0);
             g_shm_oob.max_double(0.0);
@@ -639,7 +652,7 @@ static int run_forked_child(const Opts &o, int rank)
         }
 #ifdef UCC_AMD_HAS_HIP
         if (o.mem == "cuda") {
-            hipDeviceSynchronize();
+            HIPWARN(hipDeviceSynchronize());
         }
 #endif
         g_shm_oob.max_double(0.0); /* barrier */
@@ -649,7 +662,7 @@ static int run_forked_child(const Opts &o, int rank)
         }
 #ifdef UCC_AMD_HAS_HIP
         if (o.mem == "cuda") {
-            hipDeviceSynchronize();
+            HIPWARN(hipDeviceSynchronize());
         }
 #endif
         double t  = g_shm_oob.max_double(now_s() - t0) / iters;
@@ -659,7 +672,7 @@ static int run_forked_child(const Opts &o, int rank)
             ucc_ee_destroy(ee);
         }
         if (tstream) {
-            hipStreamDestroy(tstream);
+            HIPWARN(hipStreamDestroy(tstream));
         }
 #endif
         if (rank == 0) {
@@ -778,7 +791,7 @@ static int run_inproc(const Opts &o)
         }
 #ifdef UCC_AMD_HAS_HIP
         if (o.mem == "cuda") {
-            hipDeviceSynchronize();
+            HIPWARN(hipDeviceSynchronize());
         }
 #endif
         double t0 = now_s();
@@ -787,7 +800,7 @@ static int run_inproc(const Opts &o)
         }
 #ifdef UCC_AMD_HAS_HIP
         if (o.mem == "cuda") {
-            hipDeviceSynchronize();
+            HIPWARN(hipDeviceSynchronize());
         }
 #endif
         double t = (now_s() - t0) / iters;
