@@ -505,6 +505,160 @@ class HierAllreduceTask final : public Task {
     ucc_datatype_t  dt_ = UCC_DT_FLOAT32;
 };
 
+/* ---- hier bcast (2step role): [root's node bcast from root] ->
+ * leaders bcast from the root's leader -> node bcast from each leader.
+ * The first hop only runs on the root's node; other nodes join at the
+ * leaders phase. */
+class HierBcastTask final : public Task {
+  public:
+    HierBcastTask(Context *ctx, Team *team, const ucc_coll_args_t &args)
+        : Task(ctx), team_(team), a_(args)
+    {
+    }
+    ~HierBcastTask() override
+    {
+        if (req_h_) {
+            ucc_collective_finalize(req_h_);
+        }
+    }
+
+    ucc_status_t post() override
+    {
+        phase_  = 0;
+        leader_ = team_->leaders_oob->my_idx >= 0;
+        root_   = (uint32_t)a_.root;
+        /* my node's members and the root's node leader */
+        on_root_node_ = false;
+        my_node_idx_  = -1;
+        for (size_t i = 0; i < team_->node_ranks.size(); i++) {
+            if (team_->node_ranks[i] == root_) {
+                on_root_node_ = true;
+            }
+            if (team_->node_ranks[i] == team_->rank) {
+                my_node_idx_ = (int)i;
+            }
+        }
+        root_leader_idx_ = -1;
+        {
+            /* the leader of the root's node = the smallest team rank on
+             * the root's host (leaders list is per-host lowest rank) */
+            uint64_t rhost = team_->procs[root_].host_hash;
+            for (size_t i = 0; i < team_->leader_ranks.size(); i++) {
+                if (team_->procs[team_->leader_ranks[i]].host_hash ==
+                    rhost) {
+                    root_leader_idx_ = (int)i;
+                    break;
+                }
+            }
+        }
+        status = UCC_INPROGRESS;
+        return step();
+    }
+
+    ucc_status_t progress() override { return step(); }
+
+  private:
+    ucc_status_t launch_bcast(Team *t, uint64_t root_in_t)
+    {
+        ucc_coll_args_t sa{};
+        sa.mask      = UCC_COLL_ARGS_FIELD_FLAGS;
+        sa.coll_type = UCC_COLL_TYPE_BCAST;
+        sa.root      = root_in_t;
+        sa.src.info  = a_.src.info;
+        ucc_status_t st = ucc_collective_init(
+            &sa, &req_h_, reinterpret_cast<ucc_team_h>(t));
+        if (st != UCC_OK) {
+            return st;
+        }
+        return ucc_collective_post(req_h_);
+    }
+
+    ucc_status_t step()
+    {
+        while (true) {
+            if (req_h_) {
+                ucc_status_t st = ucc_collective_test(req_h_);
+                if (st == UCC_INPROGRESS) {
+                    return UCC_INPROGRESS;
+                }
+                ucc_collective_finalize(req_h_);
+                req_h_ = nullptr;
+                if (st != UCC_OK) {
+                    return st;
+                }
+                phase_++;
+            }
+            switch (phase_) {
+            case 0: /* root's node: bcast from the root so the node
+                     * leader holds the data */
+                if (!on_root_node_ || team_->node_team->size == 1) {
+                    phase_ = 1;
+                    continue;
+                }
+                {
+                    /* root's index inside its node team */
+                    int ridx = -1;
+                    for (size_t i = 0; i < team_->node_ranks.size();
+                         i++) {
+                        if (team_->node_ranks[i] == root_) {
+                            ridx = (int)i;
+                        }
+                    }
+                    ucc_status_t st = launch_bcast(
+                        team_->node_team.get(), (uint64_t)ridx);
+                    if (st != UCC_OK) {
+                        return st;
+                    }
+                }
+                break;
+            case 1: /* leaders bcast from the root's leader */
+                if (!leader_) {
+                    phase_ = 2;
+                    continue;
+                }
+                {
+                    ucc_status_t st =
+                        launch_bcast(team_->leaders_team.get(),
+                                     (uint64_t)root_leader_idx_);
+                    if (st != UCC_OK) {
+                        return st;
+                    }
+                }
+                break;
+            case 2: /* every node: bcast from its leader (node rank 0);
+                     * the root's node already has the data but repeats
+                     * harmlessly (same values) to keep phases aligned */
+                if (team_->node_team->size == 1) {
+                    return UCC_OK;
+                }
+                {
+                    ucc_status_t st =
+                        launch_bcast(team_->node_team.get(), 0);
+                    if (st != UCC_OK) {
+                        return st;
+                    }
+                }
+                break;
+            case 3:
+                return UCC_OK;
+            default:
+                return UCC_ERR_INVALID_PARAM;
+            }
+            if (req_h_) {
+                continue;
+            }
+        }
+    }
+
+    Team           *team_;
+    ucc_coll_args_t a_;
+    ucc_coll_req_h  req_h_ = nullptr;
+    int             phase_ = 0;
+    bool            leader_ = false, on_root_node_ = false;
+    uint32_t        root_ = 0;
+    int             my_node_idx_ = -1, root_leader_idx_ = -1;
+};
+
 void add_scores(Team *team)
 {
     ScoreRange r;
@@ -524,6 +678,22 @@ void add_scores(Team *team)
         return UCC_OK;
     };
     team->score_map.add(UCC_COLL_TYPE_ALLREDUCE, UCC_MEMORY_TYPE_HOST, r);
+
+    ScoreRange b;
+    b.start    = 0;
+    b.end      = SIZE_MAX;
+    b.score    = 60;
+    b.tl_name  = "hier";
+    b.alg_name = "2step";
+    b.init     = [](const ucc_coll_args_t &args, Team *t,
+                Task **task) -> ucc_status_t {
+        if (args.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        *task = new HierBcastTask(t->ctx, t, args);
+        return UCC_OK;
+    };
+    team->score_map.add(UCC_COLL_TYPE_BCAST, UCC_MEMORY_TYPE_HOST, b);
 }
 
 } // namespace hier

@@ -37,6 +37,18 @@ exp = np.full(1000, sum(range(1, n + 1)) / n, np.float32)
 for o in outs:
     np.testing.assert_allclose(o, exp, rtol=1e-5, atol=1e-5)
 
+# hier 2step bcast from a non-leader root
+root = 4  # rank 4: non-leader in both 2- and 3-way splits (n=6)
+bufs = [np.zeros(30_000, np.float32) for _ in range(n)]
+bufs[root][:] = rng.random(30_000).astype(np.float32)
+exp2 = bufs[root].copy()
+reqs = job.coll("bcast", [
+    dict(src=b.ctypes.data, dst=0, count=30_000, dt=dtypes.FLOAT32,
+         root=root) for b in bufs])
+job.run(reqs)
+for b in bufs:
+    np.testing.assert_array_equal(b, exp2)
+
 # repeated (sub-team/slot reuse)
 for it in range(10):
     arrs = [np.full(257, float(r * it + 1), np.float64) for r in range(n)]
