@@ -9,7 +9,7 @@ PY_EXT   := $(shell $(PYTHON) -c "import sysconfig;print(sysconfig.get_config_va
 PY_INC   := $(shell $(PYTHON) -m pybind11 --includes)
 
 CXXFLAGS := --offload-arch=$(ARCH) -O3 -std=c++17 -fPIC -Wall -Wextra \
-            -Wno-unused-parameter -DUCC_AMD_HAS_HIP -DUCC_AMD_HAS_TL_CDNA4 -DUCC_AMD_HAS_TL_RCCL
+            -Wno-unused-parameter -MMD -MP -DUCC_AMD_HAS_HIP -DUCC_AMD_HAS_TL_CDNA4 -DUCC_AMD_HAS_TL_RCCL
 LDFLAGS  := -shared -fPIC -L/opt/rocm/lib -lrccl -Wl,-rpath,/opt/rocm/lib
 
 BUILD := build
@@ -83,3 +83,6 @@ clean:
 	rm -rf $(BUILD) ucc_amd/_core*.so
 
 .PHONY: all clean
+
+# auto-generated header dependencies (-MMD)
+-include $(LIB_OBJS:.o=.d) $(BIND_OBJ:.o=.d)

@@ -257,6 +257,39 @@ ucc_status_t setup(Team *team)
     team->leaders_oob->members = ldr.ranks;
     team->leaders_oob->my_idx  = ldr.my_idx; /* -1 if not a leader */
     team->hier_step            = 0;
+    /* rails (split_rail): rank p of every node forms rail p — needs
+     * UNIFORM node sizes. Node order = leader order. */
+    {
+        std::map<uint64_t, std::vector<uint32_t>> per_node;
+        for (uint32_t r = 0; r < team->size; r++) {
+            per_node[team->procs[r].host_hash].push_back(r);
+        }
+        size_t nsz = per_node.begin()->second.size();
+        bool uniform = true;
+        for (auto &kv : per_node) {
+            if (kv.second.size() != nsz) {
+                uniform = false;
+            }
+        }
+        if (uniform && node.my_idx >= 0 && (size_t)node.my_idx < nsz) {
+            std::vector<uint32_t> rail;
+            for (uint32_t lr : ldr.ranks) { /* node order by leader */
+                uint64_t hh = team->procs[lr].host_hash;
+                rail.push_back(per_node[hh][node.my_idx]);
+            }
+            int my_idx = -1;
+            for (size_t i = 0; i < rail.size(); i++) {
+                if (rail[i] == team->rank) {
+                    my_idx = (int)i;
+                }
+            }
+            team->rail_oob          = new SubOob;
+            team->rail_oob->parent  = team;
+            team->rail_oob->members = rail;
+            team->rail_oob->my_idx  = my_idx;
+            team->rails_ok          = my_idx >= 0;
+        }
+    }
     if (node.ranks.size() == 1) {
         /* singleton node: node team is size-1 (self TL) — skip its OOB
          * rounds entirely is NOT possible (alignment); create it anyway
@@ -321,6 +354,29 @@ ucc_status_t test(Team *team)
                 return st;
             }
         }
+        if (team->rails_ok) { /* every rank drives its own rail: the
+                               * per-rank parent round sequence stays
+                               * aligned (uniform node sizes) */
+            ucc_status_t cs =
+                create_subteam(team, team->rail_oob, team->rail_team);
+            if (cs != UCC_OK) {
+                team->rails_ok = false;
+            }
+        }
+        team->hier_step = 4;
+        [[fallthrough]];
+    }
+    case 4: { /* rail team rounds (all ranks in lockstep) */
+        if (team->rails_ok) {
+            ucc_status_t st = ucc_team_create_test(
+                reinterpret_cast<ucc_team_h>(team->rail_team.get()));
+            if (st < 0) {
+                return st;
+            }
+            if (team->rail_team->state < Team::TL_CREATE) {
+                return UCC_INPROGRESS;
+            }
+        }
         team->hier_step = 2;
         [[fallthrough]];
     }
@@ -343,6 +399,16 @@ ucc_status_t test(Team *team)
                 return UCC_INPROGRESS;
             }
         }
+        if (team->rails_ok) {
+            st = ucc_team_create_test(
+                reinterpret_cast<ucc_team_h>(team->rail_team.get()));
+            if (st < 0) {
+                return st;
+            }
+            if (st == UCC_INPROGRESS) {
+                return UCC_INPROGRESS;
+            }
+        }
         return UCC_OK;
     }
     }
@@ -353,8 +419,10 @@ void destroy(Team *team)
 {
     delete team->node_oob;
     delete team->leaders_oob;
+    delete team->rail_oob;
     team->node_oob    = nullptr;
     team->leaders_oob = nullptr;
+    team->rail_oob    = nullptr;
 }
 
 /* ------------------------------------------ hier allreduce (RAB) task  */
@@ -503,6 +571,170 @@ class HierAllreduceTask final : public Task {
     bool            leader_ = false, inplace_ = false;
     uint64_t        count_ = 0;
     ucc_datatype_t  dt_ = UCC_DT_FLOAT32;
+};
+
+/* ---- hier allreduce (split_rail role): node reduce_scatterv ->
+ * per-rail allreduce (rank p of every node forms rail p, so ALL ranks
+ * drive inter-node traffic concurrently instead of just the leaders) ->
+ * node allgatherv. Requires uniform node sizes (team->rails_ok).
+ * Reference parity: cl/hier allreduce split_rail
+ * (components/cl/hier/allreduce/allreduce_split_rail.c) — re-derived:
+ * slices are packed into the user dst and every sub-step runs in-place,
+ * so the inter-node phase moves exactly count/node_size elements per
+ * rank with no repack. */
+class SplitRailAllreduceTask final : public Task {
+  public:
+    SplitRailAllreduceTask(Context *ctx, Team *team,
+                           const ucc_coll_args_t &args)
+        : Task(ctx), team_(team), a_(args)
+    {
+    }
+    ~SplitRailAllreduceTask() override
+    {
+        if (req_h_) {
+            ucc_collective_finalize(req_h_);
+        }
+    }
+
+    ucc_status_t post() override
+    {
+        phase_   = 0;
+        inplace_ = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        count_   = a_.dst.info.count;
+        dt_      = a_.dst.info.datatype;
+        dtsz_    = ucc_dt_size(dt_);
+        const uint32_t nn = (uint32_t)team_->node_ranks.size();
+        int my_nidx = -1;
+        for (size_t i = 0; i < team_->node_ranks.size(); i++) {
+            if (team_->node_ranks[i] == team_->rank) {
+                my_nidx = (int)i;
+            }
+        }
+        if (my_nidx < 0 || !team_->rails_ok || !team_->rail_team) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        /* packed per-node-rank slices of the dst vector */
+        cnt_.resize(nn);
+        dsp_.resize(nn);
+        uint64_t per = count_ / nn, rem = count_ % nn, off = 0;
+        for (uint32_t r = 0; r < nn; r++) {
+            cnt_[r] = per + (r < rem ? 1 : 0);
+            dsp_[r] = off;
+            off += cnt_[r];
+        }
+        my_off_ = dsp_[(size_t)my_nidx];
+        my_cnt_ = cnt_[(size_t)my_nidx];
+        if (!inplace_) { /* all sub-steps run in-place on dst */
+            memcpy(a_.dst.info.buffer, a_.src.info.buffer,
+                   count_ * dtsz_);
+        }
+        status = UCC_INPROGRESS;
+        return step();
+    }
+
+    ucc_status_t progress() override { return step(); }
+
+  private:
+    ucc_status_t step()
+    {
+        while (true) {
+            if (req_h_) {
+                ucc_status_t st = ucc_collective_test(req_h_);
+                if (st == UCC_INPROGRESS) {
+                    return UCC_INPROGRESS;
+                }
+                ucc_collective_finalize(req_h_);
+                req_h_ = nullptr;
+                if (st != UCC_OK) {
+                    return st;
+                }
+                phase_++;
+            }
+            ucc_coll_args_t sa{};
+            sa.mask  = UCC_COLL_ARGS_FIELD_FLAGS;
+            sa.flags = UCC_COLL_ARGS_FLAG_IN_PLACE |
+                       UCC_COLL_ARGS_FLAG_COUNT_64BIT |
+                       UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT;
+            sa.op    = a_.op == UCC_OP_AVG ? UCC_OP_SUM : a_.op;
+            switch (phase_) {
+            case 0: { /* node reduce_scatterv: my packed slice lands at
+                       * dst + my_off (in-place RSV convention) */
+                sa.coll_type           = UCC_COLL_TYPE_REDUCE_SCATTERV;
+                sa.dst.info_v.buffer   = a_.dst.info.buffer;
+                sa.dst.info_v.counts   = (ucc_count_t *)cnt_.data();
+                sa.dst.info_v.datatype = dt_;
+                sa.dst.info_v.mem_type = a_.dst.info.mem_type;
+                ucc_status_t st        = launch(team_->node_team.get(), sa);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 1: { /* rail allreduce of my slice (inter-node) */
+                if (my_cnt_ == 0) {
+                    phase_ = 2;
+                    continue;
+                }
+                sa.coll_type         = UCC_COLL_TYPE_ALLREDUCE;
+                sa.dst.info.buffer   = (uint8_t *)a_.dst.info.buffer +
+                                     my_off_ * dtsz_;
+                sa.dst.info.count    = my_cnt_;
+                sa.dst.info.datatype = dt_;
+                sa.dst.info.mem_type = a_.dst.info.mem_type;
+                sa.src.info          = sa.dst.info;
+                ucc_status_t st      = launch(team_->rail_team.get(), sa);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 2: { /* AVG: scale just my slice before the allgather */
+                if (a_.op == UCC_OP_AVG && my_cnt_) {
+                    uint8_t    *b = (uint8_t *)a_.dst.info.buffer +
+                                 my_off_ * dtsz_;
+                    const void *srcs[1] = {b};
+                    ec_cpu::reduce(b, srcs, 1, my_cnt_, dt_, UCC_OP_SUM,
+                                   1.0 / (double)team_->size);
+                }
+                sa.coll_type                = UCC_COLL_TYPE_ALLGATHERV;
+                sa.dst.info_v.buffer        = a_.dst.info.buffer;
+                sa.dst.info_v.counts        = (ucc_count_t *)cnt_.data();
+                sa.dst.info_v.displacements = (ucc_aint_t *)dsp_.data();
+                sa.dst.info_v.datatype      = dt_;
+                sa.dst.info_v.mem_type      = a_.dst.info.mem_type;
+                ucc_status_t st = launch(team_->node_team.get(), sa);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 3:
+                return UCC_OK;
+            default:
+                return UCC_ERR_INVALID_PARAM;
+            }
+        }
+    }
+
+    ucc_status_t launch(Team *t, ucc_coll_args_t &sa)
+    {
+        ucc_status_t st = ucc_collective_init(
+            &sa, &req_h_, reinterpret_cast<ucc_team_h>(t));
+        if (st != UCC_OK) {
+            return st;
+        }
+        return ucc_collective_post(req_h_);
+    }
+
+    Team                 *team_;
+    ucc_coll_args_t       a_;
+    ucc_coll_req_h        req_h_ = nullptr;
+    int                   phase_ = 0;
+    bool                  inplace_ = false;
+    uint64_t              count_ = 0, my_off_ = 0, my_cnt_ = 0;
+    size_t                dtsz_ = 4;
+    ucc_datatype_t        dt_ = UCC_DT_FLOAT32;
+    std::vector<uint64_t> cnt_, dsp_;
 };
 
 /* ---- hier bcast (2step role): [root's node bcast from root] ->
@@ -678,6 +910,30 @@ void add_scores(Team *team)
         return UCC_OK;
     };
     team->score_map.add(UCC_COLL_TYPE_ALLREDUCE, UCC_MEMORY_TYPE_HOST, r);
+
+    if (team->rails_ok && team->rail_team) {
+        /* split_rail beats RAB on large vectors: every rank carries
+         * 1/node_size of the inter-node traffic instead of leaders
+         * carrying all of it */
+        ScoreRange s;
+        s.start    = 64 * 1024;
+        s.end      = SIZE_MAX;
+        s.score    = 61;
+        s.tl_name  = "hier";
+        s.alg_name = "split_rail";
+        s.init     = [](const ucc_coll_args_t &args, Team *t,
+                    Task **task) -> ucc_status_t {
+            if (args.op != UCC_OP_SUM && args.op != UCC_OP_MAX &&
+                args.op != UCC_OP_MIN && args.op != UCC_OP_PROD &&
+                args.op != UCC_OP_AVG) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new SplitRailAllreduceTask(t->ctx, t, args);
+            return UCC_OK;
+        };
+        team->score_map.add(UCC_COLL_TYPE_ALLREDUCE, UCC_MEMORY_TYPE_HOST,
+                            s);
+    }
 
     ScoreRange b;
     b.start    = 0;

@@ -279,8 +279,11 @@ struct Team {
      * padded payloads so non-members can observe (see SubOob). */
     std::unique_ptr<Team>   node_team;
     std::unique_ptr<Team>   leaders_team;
+    std::unique_ptr<Team>   rail_team; /* my node-position across nodes */
     struct SubOob          *node_oob    = nullptr;
     struct SubOob          *leaders_oob = nullptr;
+    struct SubOob          *rail_oob    = nullptr;
+    bool                    rails_ok    = false; /* uniform node sizes */
     int                     hier_step   = 0;
     bool                    want_hier   = false;
     bool                    is_subteam  = false;

@@ -195,8 +195,8 @@ class Cdna4TlTeam final : public TlTeam {
         b.scratch_ptr = (uint64_t)(uintptr_t)scratch_;
         b.flags_ptr   = (uint64_t)(uintptr_t)flags_;
         if (init_st_ == UCC_OK) {
-            hipIpcGetMemHandle(&b.sh, scratch_);
-            hipIpcGetMemHandle(&b.fh, flags_);
+            HIPWARN(hipIpcGetMemHandle(&b.sh, scratch_));
+            HIPWARN(hipIpcGetMemHandle(&b.fh, flags_));
         }
         memcpy(buf, &b, sizeof(b));
     }
@@ -987,8 +987,8 @@ class StagedTask final : public Cdna4Task {
         }
         copy_pending_ = len > 0;
         if (len > 0) {
-            hipMemcpyAsync(tt_->area(me_, slot_, p, 0), src, len,
-                           hipMemcpyDeviceToDevice, copy_s());
+            HIPWARN(hipMemcpyAsync(tt_->area(me_, slot_, p, 0), src, len,
+                                   hipMemcpyDeviceToDevice, copy_s()));
         }
         HIPWARN(hipEventRecord(ev(0 + (int)p), copy_s()));
     }

@@ -21,6 +21,7 @@ job = LocalJob(n)
 c = core()
 smap = c.score_map_str(job.teams[0])
 assert "@hier/rab" in smap, "hier not installed:\n" + smap
+assert "@hier/split_rail" in smap, "split_rail not installed:\n" + smap
 
 rng = np.random.default_rng(11)
 for count in (64, 5000, 100_001):
@@ -48,6 +49,17 @@ reqs = job.coll("bcast", [
 job.run(reqs)
 for b in bufs:
     np.testing.assert_array_equal(b, exp2)
+
+# split_rail explicitly (>=64KB auto-picks it; also force via tuning
+# check on a large in-place vector)
+arrs = [(rng.random(300_000) - 0.5).astype(np.float32) for _ in range(n)]
+exp3 = np.sum(arrs, axis=0)
+reqs = job.coll("allreduce", [
+    dict(src=0, dst=a.ctypes.data, count=300_000, dt=dtypes.FLOAT32,
+         flags=c.FLAG_IN_PLACE) for a in arrs])
+job.run(reqs)
+for a in arrs:
+    np.testing.assert_allclose(a, exp3, rtol=1e-5, atol=1e-4)
 
 # repeated (sub-team/slot reuse)
 for it in range(10):
