@@ -1259,6 +1259,9 @@ class GatedCollTask final : public Cdna4Task {
             zc_ = true;
         }
         phase_ = zc_ && !zc_ready_ ? 10 : 0;
+        if (ct_ == UCC_COLL_TYPE_ALLTOALLV && !a2av_ready_) {
+            phase_ = 20; /* global-max exchange first */
+        }
         status = UCC_INPROGRESS;
         return progress();
     }
@@ -1269,6 +1272,11 @@ class GatedCollTask final : public Cdna4Task {
      * counterpart of the fused graph kernel (docs/GATED_PIPELINE.md). */
     ucc_status_t triggered_post(void *ee_stream) override
     {
+        if (a_.coll_type == UCC_COLL_TYPE_ALLTOALLV) {
+            /* needs a host-side global-max exchange before enqueue:
+             * not expressible as a pure stream capture */
+            return UCC_ERR_NOT_SUPPORTED;
+        }
         if (pslot_ < 0) {
             ucc_coll_type_t k_ct = a_.coll_type;
             int kind = k_ct == UCC_COLL_TYPE_ALLREDUCE
@@ -1418,6 +1426,45 @@ class GatedCollTask final : public Cdna4Task {
             gran_  = cell_;
             break;
         }
+        case UCC_COLL_TYPE_ALLTOALLV: {
+            /* Per-peer send row (src.info_v) and recv column
+             * (dst.info_v) are LOCAL knowledge, but the symmetric-launch
+             * invariant needs the GLOBAL max pair length — exchanged
+             * host-side through the scratch channel (phases 20/21)
+             * before any fragment is enqueued. Rank i's send cell to j
+             * and its gather of j's cell both clip to the locally-known
+             * lengths; empty tails still launch and signal. */
+            if (inplace || cell_ == 0) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            dt_          = a_.src.info_v.datatype;
+            dtsz_        = ucc_dt_size(dt_);
+            size_t rdtsz = ucc_dt_size(a_.dst.info_v.datatype);
+            cnt_.resize(n_);
+            dsp_.resize(n_);
+            rcnt_.resize(n_);
+            rdsp_.resize(n_);
+            size_t lmax = 0;
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = coll_count_at(a_, a_.src.info_v.counts, r) *
+                          dtsz_;
+                dsp_[r] =
+                    coll_disp_at(a_, a_.src.info_v.displacements, r) *
+                    dtsz_;
+                rcnt_[r] = coll_count_at(a_, a_.dst.info_v.counts, r) *
+                           rdtsz;
+                rdsp_[r] =
+                    coll_disp_at(a_, a_.dst.info_v.displacements, r) *
+                    rdtsz;
+                lmax = cnt_[r] > lmax ? cnt_[r] : lmax;
+                lmax = rcnt_[r] > lmax ? rcnt_[r] : lmax;
+            }
+            sbuf_  = (const uint8_t *)a_.src.info_v.buffer;
+            dbuf_  = (uint8_t *)a_.dst.info_v.buffer;
+            total_ = lmax; /* provisional; replaced by the global max */
+            gran_  = cell_;
+            break;
+        }
         default:
             return UCC_ERR_NOT_SUPPORTED;
         }
@@ -1447,6 +1494,42 @@ class GatedCollTask final : public Cdna4Task {
 
     ucc_status_t progress() override
     {
+        if (phase_ == 20) { /* a2av: publish my local max pair length */
+            if (!all_ge(0)) {
+                return UCC_INPROGRESS;
+            }
+            uint64_t lmax = (uint64_t)total_;
+            if (hipMemcpy(tt_->area(me_, slot_, 0, 0), &lmax,
+                          sizeof(lmax), hipMemcpyHostToDevice) !=
+                hipSuccess) {
+                return UCC_ERR_NO_RESOURCE;
+            }
+            publish(1);
+            phase_ = 21;
+        }
+        if (phase_ == 21) { /* a2av: global max = max of all locals */
+            if (!all_ge(1)) {
+                return UCC_INPROGRESS;
+            }
+            uint64_t gmax = 0;
+            for (uint32_t r = 0; r < n_; r++) {
+                uint64_t v = 0;
+                if (hipMemcpy(&v, tt_->area(r, slot_, 0, 0), sizeof(v),
+                              hipMemcpyDeviceToHost) != hipSuccess) {
+                    return UCC_ERR_NO_RESOURCE;
+                }
+                gmax = v > gmax ? v : gmax;
+            }
+            total_  = gmax;
+            nfrags_ = (total_ + gran_ - 1) / gran_;
+            if (nfrags_ == 0) {
+                nfrags_ = 1;
+            }
+            a2av_ready_ = true;
+            close_slot();
+            begin_use(); /* fresh slot use for the collective itself */
+            phase_ = 0;
+        }
         if (phase_ == 10) { /* publish my src handle */
             if (!all_ge(0)) {
                 return UCC_INPROGRESS;
@@ -1787,6 +1870,43 @@ class GatedCollTask final : public Cdna4Task {
                 }
                 break;
             }
+            case UCC_COLL_TYPE_ALLTOALLV: {
+                /* per-dest cells with per-pair lengths: send cells clip
+                 * to my send row, gather slices clip to my recv column.
+                 * Fragment count is the exchanged GLOBAL max, so every
+                 * rank launches identically (ledger stays in sync). */
+                ga.src     = sbuf_;
+                ga.dst     = dbuf_;
+                ga.n_cells = (int)n_;
+                for (uint32_t r = 0; r < n_; r++) {
+                    size_t sl = off >= cnt_[r] ? 0
+                                : cnt_[r] - off < gran_ ? cnt_[r] - off
+                                                        : gran_;
+                    size_t rl = off >= rcnt_[r] ? 0
+                                : rcnt_[r] - off < gran_
+                                    ? rcnt_[r] - off
+                                    : gran_;
+                    ga.c_src_off[r] = dsp_[r] + off;
+                    ga.c_dst_off[r] = (uint64_t)r * cell_;
+                    ga.c_len[r]     = sl;
+                    ga.peer_out[r] =
+                        (const void *)(tt_->area(r, slot_, p, 0) +
+                                       me_ * cell_);
+                    ga.slice_b[r] = rdsp_[r] + off;
+                    ga.slice_e[r] = ga.slice_b[r] + rl;
+                }
+                ga.t_sw_reduce   = L[1][slot_][p] * B;
+                ga.t_sw_gather   = L[2][slot_][p] * B;
+                ga.gw_phase      = 0;
+                ga.t_gather_wait = (L[0][slot_][p] + 1) * B;
+                st = ec_hip::staged_stage(ga, stage_s);
+                if (!derive) { L[0][slot_][p]++; }
+                if (st == UCC_OK) {
+                    st = ec_hip::staged_gather(ga, comp_s);
+                    if (!derive) { L[2][slot_][p]++; }
+                }
+                break;
+            }
             case UCC_COLL_TYPE_ALLTOALL: {
                 /* stage per-dest cells; gather my cell from each peer.
                  * zero-copy: gather reads peers' USER src at my block's
@@ -1840,7 +1960,9 @@ class GatedCollTask final : public Cdna4Task {
     size_t             gran_ = 0, cell_ = 0, out_b_ = 0;
     const uint8_t     *sbuf_ = nullptr;
     uint8_t           *dbuf_ = nullptr;
-    std::vector<size_t> cnt_, dsp_; /* per-rank bytes (ag(v)/rs(v)) */
+    std::vector<size_t> cnt_, dsp_;   /* send row bytes (ag/rs/a2av)  */
+    std::vector<size_t> rcnt_, rdsp_; /* recv column bytes (a2av)     */
+    bool                a2av_ready_ = false; /* global max exchanged  */
     int                pslot_ = -1;
     /* zero-copy persistent allreduce: peers' USER src buffers mapped
      * via HIP-IPC (handles exchanged through the scratch channel at
@@ -2025,6 +2147,7 @@ void Cdna4TlTeam::get_scores(Team *team, ScoreMap &map)
         add_gated(UCC_COLL_TYPE_ALLGATHER, 0, SIZE_MAX, 90);
         add_gated(UCC_COLL_TYPE_ALLGATHERV, 0, SIZE_MAX, 90);
         add_gated(UCC_COLL_TYPE_ALLTOALL, 0, SIZE_MAX, 90);
+        add_gated(UCC_COLL_TYPE_ALLTOALLV, 0, SIZE_MAX, 90);
     }
     add(UCC_COLL_TYPE_ALLREDUCE, 0, SIZE_MAX, 80, "staged_linear", false);
     add(UCC_COLL_TYPE_ALLGATHER, 0, SIZE_MAX, 80, "staged_linear", false);

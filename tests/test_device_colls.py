@@ -228,6 +228,36 @@ def test_alltoallv_device_skewed(job):
             torch.testing.assert_close(got, exp)
 
 
+def test_alltoallv_device_multifrag(job):
+    """Gated a2av crossing the staging-cell boundary: every pair moves
+    more than one cell (chunk/n bytes), so the global-max exchange and
+    the multi-fragment pipeline both run (>1 fragment per pair)."""
+    torch.manual_seed(29)
+    n = job.n
+    cell_elems = ((32 * 1024 * 1024 // n) & ~255) // 4
+    scnt = [[cell_elems + (d * 7919 + r * 101) % 50_000
+             for d in range(n)] for r in range(n)]
+    rcnt = [[scnt[s][r] for s in range(n)] for r in range(n)]
+    sdsp = [np.concatenate([[0], np.cumsum(c)[:-1]]).tolist() for c in scnt]
+    rdsp = [np.concatenate([[0], np.cumsum(c)[:-1]]).tolist() for c in rcnt]
+    srcs = [torch.randn(sum(scnt[r]), device="cuda") for r in range(n)]
+    dsts = [torch.zeros(sum(rcnt[r]), device="cuda") for r in range(n)]
+    _run_device(job, "alltoallv", [
+        dict(src=srcs[r].data_ptr(), dst=dsts[r].data_ptr(), count=0,
+             dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA,
+             src_counts=scnt[r], src_displs=sdsp[r],
+             dst_counts=rcnt[r], dst_displs=rdsp[r])
+        for r in range(n)
+    ])
+    for r in range(n):
+        for s in range(n):
+            got = dsts[r].cpu()[rdsp[r][s]:rdsp[r][s] + rcnt[r][s]]
+            exp = srcs[s].cpu()[sdsp[s][r]:sdsp[s][r] + scnt[s][r]]
+            torch.testing.assert_close(got, exp)
+    del srcs, dsts
+    torch.cuda.empty_cache()
+
+
 def test_allgatherv_device(job):
     torch.manual_seed(17)
     n = job.n
