@@ -737,6 +737,265 @@ class SplitRailAllreduceTask final : public Task {
     std::vector<uint64_t> cnt_, dsp_;
 };
 
+/* ---- hier reduce (2step role): node reduce -> leaders reduce to the
+ * root's node leader -> [node bcast delivers to a non-leader root].
+ * Reference parity: cl/hier reduce 2step
+ * (components/cl/hier/reduce/reduce_2step.c) — re-derived. Intermediate
+ * partials live in task-owned host scratch; only the root's user dst is
+ * ever written. */
+class HierReduceTask final : public Task {
+  public:
+    HierReduceTask(Context *ctx, Team *team, const ucc_coll_args_t &args)
+        : Task(ctx), team_(team), a_(args)
+    {
+    }
+    ~HierReduceTask() override
+    {
+        if (req_h_) {
+            ucc_collective_finalize(req_h_);
+        }
+    }
+
+    ucc_status_t post() override
+    {
+        phase_   = 0;
+        leader_  = team_->leaders_oob->my_idx >= 0;
+        root_    = (uint32_t)a_.root;
+        is_root_ = team_->rank == root_;
+        count_   = a_.src.info.count;
+        dt_      = a_.src.info.datatype;
+        if (is_root_) {
+            count_ = a_.dst.info.count;
+            dt_    = a_.dst.info.datatype;
+        }
+        dtsz_ = ucc_dt_size(dt_);
+        on_root_node_ = false;
+        for (uint32_t nr : team_->node_ranks) {
+            if (nr == root_) {
+                on_root_node_ = true;
+            }
+        }
+        root_leader_idx_ = -1;
+        {
+            uint64_t rhost = team_->procs[root_].host_hash;
+            for (size_t i = 0; i < team_->leader_ranks.size(); i++) {
+                if (team_->procs[team_->leader_ranks[i]].host_hash ==
+                    rhost) {
+                    root_leader_idx_ = (int)i;
+                    break;
+                }
+            }
+        }
+        root_is_leader_ =
+            root_leader_idx_ >= 0 &&
+            team_->leader_ranks[(size_t)root_leader_idx_] == root_;
+        if (leader_ || on_root_node_) {
+            scratch_.resize(count_ * dtsz_);
+        }
+        status = UCC_INPROGRESS;
+        return step();
+    }
+
+    ucc_status_t progress() override { return step(); }
+
+  private:
+    ucc_status_t step()
+    {
+        while (true) {
+            if (req_h_) {
+                ucc_status_t st = ucc_collective_test(req_h_);
+                if (st == UCC_INPROGRESS) {
+                    return UCC_INPROGRESS;
+                }
+                ucc_collective_finalize(req_h_);
+                req_h_ = nullptr;
+                if (st != UCC_OK) {
+                    return st;
+                }
+                phase_++;
+            }
+            ucc_coll_args_t sa{};
+            sa.mask  = UCC_COLL_ARGS_FIELD_FLAGS;
+            sa.flags = 0;
+            sa.op    = a_.op == UCC_OP_AVG ? UCC_OP_SUM : a_.op;
+            switch (phase_) {
+            case 0: { /* node reduce to my node's leader (node rank 0) */
+                sa.coll_type         = UCC_COLL_TYPE_REDUCE;
+                sa.root              = 0;
+                sa.src.info          = a_.src.info;
+                sa.src.info.count    = count_;
+                sa.src.info.datatype = dt_;
+                if (is_root_ && (a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE)) {
+                    sa.src.info.buffer   = a_.dst.info.buffer;
+                    sa.src.info.mem_type = a_.dst.info.mem_type;
+                }
+                sa.dst.info          = sa.src.info;
+                sa.dst.info.buffer   = scratch_.data(); /* leader only */
+                ucc_status_t st = launch(team_->node_team.get(), sa);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 1: { /* leaders reduce to the root's node leader */
+                if (!leader_) {
+                    phase_ = 2;
+                    continue;
+                }
+                sa.coll_type         = UCC_COLL_TYPE_REDUCE;
+                sa.root              = (uint64_t)root_leader_idx_;
+                sa.src.info.buffer   = scratch_.data();
+                sa.src.info.count    = count_;
+                sa.src.info.datatype = dt_;
+                sa.src.info.mem_type = UCC_MEMORY_TYPE_HOST;
+                sa.dst.info          = sa.src.info;
+                if (team_->leaders_oob->my_idx == root_leader_idx_) {
+                    sa.dst.info.buffer = root_is_leader_
+                                             ? a_.dst.info.buffer
+                                             : scratch_.data();
+                    if (root_is_leader_) {
+                        sa.dst.info.mem_type = a_.dst.info.mem_type;
+                    }
+                }
+                ucc_status_t st = launch(team_->leaders_team.get(), sa);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 2: { /* non-leader root: node bcast delivers the total
+                       * (non-root node members receive into scratch) */
+                if (!on_root_node_ || root_is_leader_) {
+                    phase_ = 3;
+                    continue;
+                }
+                sa.coll_type         = UCC_COLL_TYPE_BCAST;
+                sa.root              = 0;
+                sa.src.info.buffer   = is_root_ ? a_.dst.info.buffer
+                                                : scratch_.data();
+                sa.src.info.count    = count_;
+                sa.src.info.datatype = dt_;
+                sa.src.info.mem_type = is_root_ ? a_.dst.info.mem_type
+                                                : UCC_MEMORY_TYPE_HOST;
+                ucc_status_t st = launch(team_->node_team.get(), sa);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 3: { /* AVG post-scale at the root */
+                if (is_root_ && a_.op == UCC_OP_AVG) {
+                    const void *srcs[1] = {a_.dst.info.buffer};
+                    ec_cpu::reduce(a_.dst.info.buffer, srcs, 1, count_,
+                                   dt_, UCC_OP_SUM,
+                                   1.0 / (double)team_->size);
+                }
+                return UCC_OK;
+            }
+            default:
+                return UCC_ERR_INVALID_PARAM;
+            }
+        }
+    }
+
+    ucc_status_t launch(Team *t, ucc_coll_args_t &sa)
+    {
+        ucc_status_t st = ucc_collective_init(
+            &sa, &req_h_, reinterpret_cast<ucc_team_h>(t));
+        if (st != UCC_OK) {
+            return st;
+        }
+        return ucc_collective_post(req_h_);
+    }
+
+    Team                *team_;
+    ucc_coll_args_t      a_;
+    ucc_coll_req_h       req_h_ = nullptr;
+    int                  phase_ = 0;
+    bool                 leader_ = false, is_root_ = false;
+    bool                 on_root_node_ = false, root_is_leader_ = false;
+    uint32_t             root_ = 0;
+    int                  root_leader_idx_ = -1;
+    uint64_t             count_ = 0;
+    size_t               dtsz_ = 4;
+    ucc_datatype_t       dt_ = UCC_DT_FLOAT32;
+    std::vector<uint8_t> scratch_;
+};
+
+/* ---- hier barrier: node fanin -> leaders barrier -> node fanout.     */
+class HierBarrierTask final : public Task {
+  public:
+    HierBarrierTask(Context *ctx, Team *team, const ucc_coll_args_t &args)
+        : Task(ctx), team_(team)
+    {
+        (void)args;
+    }
+    ~HierBarrierTask() override
+    {
+        if (req_h_) {
+            ucc_collective_finalize(req_h_);
+        }
+    }
+
+    ucc_status_t post() override
+    {
+        phase_  = 0;
+        leader_ = team_->leaders_oob->my_idx >= 0;
+        status  = UCC_INPROGRESS;
+        return step();
+    }
+
+    ucc_status_t progress() override { return step(); }
+
+  private:
+    ucc_status_t step()
+    {
+        while (true) {
+            if (req_h_) {
+                ucc_status_t st = ucc_collective_test(req_h_);
+                if (st == UCC_INPROGRESS) {
+                    return UCC_INPROGRESS;
+                }
+                ucc_collective_finalize(req_h_);
+                req_h_ = nullptr;
+                if (st != UCC_OK) {
+                    return st;
+                }
+                phase_++;
+            }
+            static const ucc_coll_type_t kPhase[3] = {
+                UCC_COLL_TYPE_FANIN, UCC_COLL_TYPE_BARRIER,
+                UCC_COLL_TYPE_FANOUT};
+            if (phase_ >= 3) {
+                return UCC_OK;
+            }
+            if (phase_ == 1 && !leader_) {
+                phase_ = 2;
+                continue;
+            }
+            ucc_coll_args_t sa{};
+            sa.coll_type = kPhase[phase_];
+            sa.root      = 0;
+            Team *t = phase_ == 1 ? team_->leaders_team.get()
+                                  : team_->node_team.get();
+            ucc_status_t st = ucc_collective_init(
+                &sa, &req_h_, reinterpret_cast<ucc_team_h>(t));
+            if (st != UCC_OK) {
+                return st;
+            }
+            st = ucc_collective_post(req_h_);
+            if (st != UCC_OK) {
+                return st;
+            }
+        }
+    }
+
+    Team          *team_;
+    ucc_coll_req_h req_h_ = nullptr;
+    int            phase_ = 0;
+    bool           leader_ = false;
+};
+
 /* ---- hier bcast (2step role): [root's node bcast from root] ->
  * leaders bcast from the root's leader -> node bcast from each leader.
  * The first hop only runs on the root's node; other nodes join at the
@@ -950,6 +1209,41 @@ void add_scores(Team *team)
         return UCC_OK;
     };
     team->score_map.add(UCC_COLL_TYPE_BCAST, UCC_MEMORY_TYPE_HOST, b);
+
+    ScoreRange rd;
+    rd.start    = 0;
+    rd.end      = SIZE_MAX;
+    rd.score    = 60;
+    rd.tl_name  = "hier";
+    rd.alg_name = "2step";
+    rd.init     = [](const ucc_coll_args_t &args, Team *t,
+                 Task **task) -> ucc_status_t {
+        if ((args.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) ||
+            !ucc_dt_is_predefined(args.src.info.datatype)) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        if (args.op != UCC_OP_SUM && args.op != UCC_OP_MAX &&
+            args.op != UCC_OP_MIN && args.op != UCC_OP_PROD &&
+            args.op != UCC_OP_AVG) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        *task = new HierReduceTask(t->ctx, t, args);
+        return UCC_OK;
+    };
+    team->score_map.add(UCC_COLL_TYPE_REDUCE, UCC_MEMORY_TYPE_HOST, rd);
+
+    ScoreRange ba;
+    ba.start    = 0;
+    ba.end      = SIZE_MAX;
+    ba.score    = 60;
+    ba.tl_name  = "hier";
+    ba.alg_name = "fanin_fanout";
+    ba.init     = [](const ucc_coll_args_t &args, Team *t,
+                 Task **task) -> ucc_status_t {
+        *task = new HierBarrierTask(t->ctx, t, args);
+        return UCC_OK;
+    };
+    team->score_map.add(UCC_COLL_TYPE_BARRIER, UCC_MEMORY_TYPE_HOST, ba);
 }
 
 } // namespace hier

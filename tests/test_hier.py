@@ -61,6 +61,42 @@ job.run(reqs)
 for a in arrs:
     np.testing.assert_allclose(a, exp3, rtol=1e-5, atol=1e-4)
 
+# hier 2step reduce: leader root, non-leader root, AVG, in-place
+for root, op in ((0, dtypes.OP_SUM), (4, dtypes.OP_SUM),
+                 (4, dtypes.OP_AVG)):
+    srcs = [rng.random(12_345).astype(np.float32) for _ in range(n)]
+    dst = np.zeros(12_345, np.float32)
+    exp4 = np.sum(srcs, axis=0)
+    if op == dtypes.OP_AVG:
+        exp4 = exp4 / n
+    reqs = job.coll("reduce", [
+        dict(src=srcs[r].ctypes.data, dst=(dst.ctypes.data if r == root
+                                           else 0),
+             count=12_345, dt=dtypes.FLOAT32, root=root, op=op)
+        for r in range(n)])
+    job.run(reqs)
+    np.testing.assert_allclose(dst, exp4, rtol=1e-5, atol=1e-5)
+
+# in-place reduce at root (root's contribution read from dst)
+srcs = [rng.random(999).astype(np.float32) for _ in range(n)]
+root = 3
+dst = srcs[root].copy()
+exp5 = np.sum(srcs, axis=0)
+reqs = job.coll("reduce", [
+    dict(src=(0 if r == root else srcs[r].ctypes.data),
+         dst=(dst.ctypes.data if r == root else 0), count=999,
+         dt=dtypes.FLOAT32, root=root,
+         flags=(c.FLAG_IN_PLACE if r == root else 0))
+    for r in range(n)])
+job.run(reqs)
+np.testing.assert_allclose(dst, exp5, rtol=1e-5, atol=1e-5)
+
+# hier barrier (fanin -> leaders barrier -> fanout), repeated
+for _ in range(3):
+    reqs = job.coll("barrier", [dict(src=0, dst=0, count=0, dt=dtypes.INT8)
+                              for _ in range(n)])
+    job.run(reqs)
+
 # repeated (sub-team/slot reuse)
 for it in range(10):
     arrs = [np.full(257, float(r * it + 1), np.float64) for r in range(n)]
