@@ -737,6 +737,219 @@ class SplitRailAllreduceTask final : public Task {
     std::vector<uint64_t> cnt_, dsp_;
 };
 
+/* ---- hier allgatherv: node gatherv to the leader (packed) -> leaders
+ * allgatherv of node-aggregated blocks -> node bcast of the full packed
+ * vector -> local unpack into user dst positions. All per-rank counts
+ * are global knowledge (dst.info_v), so every phase's geometry is
+ * locally computable. Reference parity: cl/hier allgatherv
+ * (components/cl/hier/allgatherv/) — re-derived with node-major packed
+ * relay and a final local unpack. */
+class HierAllgathervTask final : public Task {
+  public:
+    HierAllgathervTask(Context *ctx, Team *team,
+                       const ucc_coll_args_t &args)
+        : Task(ctx), team_(team), a_(args)
+    {
+    }
+    ~HierAllgathervTask() override
+    {
+        if (req_h_) {
+            ucc_collective_finalize(req_h_);
+        }
+    }
+
+    static uint64_t cnt_at(const ucc_coll_args_t &a, const void *counts,
+                           uint32_t r)
+    {
+        return (a.flags & UCC_COLL_ARGS_FLAG_COUNT_64BIT)
+                   ? ((const uint64_t *)counts)[r]
+                   : ((const uint32_t *)counts)[r];
+    }
+    static uint64_t dsp_at(const ucc_coll_args_t &a, const void *displs,
+                           uint32_t r)
+    {
+        return (a.flags & UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT)
+                   ? ((const uint64_t *)displs)[r]
+                   : ((const uint32_t *)displs)[r];
+    }
+
+    ucc_status_t post() override
+    {
+        phase_   = 0;
+        leader_  = team_->leaders_oob->my_idx >= 0;
+        inplace_ = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        dt_      = a_.dst.info_v.datatype;
+        dtsz_    = ucc_dt_size(dt_);
+        const uint32_t n = team_->size;
+        /* node-major packed layout: nodes in leader order, members in
+         * team-rank order within each node */
+        cnt_.resize(n);
+        udsp_.resize(n);
+        pdsp_.resize(n); /* packed offset of rank r's block            */
+        node_bytes_.assign(team_->leader_ranks.size(), 0);
+        node_poff_.assign(team_->leader_ranks.size(), 0);
+        for (uint32_t r = 0; r < n; r++) {
+            cnt_[r]  = cnt_at(a_, a_.dst.info_v.counts, r) * dtsz_;
+            udsp_[r] = dsp_at(a_, a_.dst.info_v.displacements, r) * dtsz_;
+        }
+        size_t off = 0;
+        my_node_idx_ = -1;
+        for (size_t k = 0; k < team_->leader_ranks.size(); k++) {
+            uint64_t h = team_->procs[team_->leader_ranks[k]].host_hash;
+            node_poff_[k] = off;
+            for (uint32_t r = 0; r < n; r++) {
+                if (team_->procs[r].host_hash == h) {
+                    pdsp_[r] = off;
+                    off += cnt_[r];
+                    node_bytes_[k] += cnt_[r];
+                    if (r == team_->rank) {
+                        my_node_idx_ = (int)k;
+                    }
+                }
+            }
+        }
+        total_ = off;
+        if (my_node_idx_ < 0) {
+            return UCC_ERR_INVALID_PARAM;
+        }
+        packed_.resize(total_);
+        status = UCC_INPROGRESS;
+        return step();
+    }
+
+    ucc_status_t progress() override { return step(); }
+
+  private:
+    ucc_status_t step()
+    {
+        while (true) {
+            if (req_h_) {
+                ucc_status_t st = ucc_collective_test(req_h_);
+                if (st == UCC_INPROGRESS) {
+                    return UCC_INPROGRESS;
+                }
+                ucc_collective_finalize(req_h_);
+                req_h_ = nullptr;
+                if (st != UCC_OK) {
+                    return st;
+                }
+                phase_++;
+            }
+            ucc_coll_args_t sa{};
+            sa.mask  = UCC_COLL_ARGS_FIELD_FLAGS;
+            sa.flags = UCC_COLL_ARGS_FLAG_COUNT_64BIT |
+                       UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT;
+            const uint8_t *my_src =
+                inplace_ ? (const uint8_t *)a_.dst.info_v.buffer +
+                               udsp_[team_->rank]
+                         : (const uint8_t *)a_.src.info.buffer;
+            switch (phase_) {
+            case 0: { /* node gatherv -> leader's packed block */
+                sa.coll_type         = UCC_COLL_TYPE_GATHERV;
+                sa.root              = 0; /* node leader = lowest rank */
+                sa.src.info.buffer   = (void *)my_src;
+                sa.src.info.count    = cnt_[team_->rank] / dtsz_;
+                sa.src.info.datatype = dt_;
+                sa.src.info.mem_type = UCC_MEMORY_TYPE_HOST;
+                if (leader_) {
+                    const auto  &nr = team_->node_ranks;
+                    sub_cnt_.resize(nr.size());
+                    sub_dsp_.resize(nr.size());
+                    size_t base = node_poff_[(size_t)my_node_idx_];
+                    for (size_t j = 0; j < nr.size(); j++) {
+                        sub_cnt_[j] = cnt_[nr[j]] / dtsz_;
+                        sub_dsp_[j] = (pdsp_[nr[j]] - base) / dtsz_;
+                    }
+                    sa.dst.info_v.buffer = packed_.data() + base;
+                    sa.dst.info_v.counts = (ucc_count_t *)sub_cnt_.data();
+                    sa.dst.info_v.displacements =
+                        (ucc_aint_t *)sub_dsp_.data();
+                    sa.dst.info_v.datatype = dt_;
+                    sa.dst.info_v.mem_type = UCC_MEMORY_TYPE_HOST;
+                }
+                ucc_status_t st = launch(team_->node_team.get(), sa);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 1: { /* leaders allgatherv of node blocks (in-place:
+                       * my node's block is already at its packed spot) */
+                if (!leader_) {
+                    phase_ = 2;
+                    continue;
+                }
+                const size_t nl = team_->leader_ranks.size();
+                sub_cnt_.resize(nl);
+                sub_dsp_.resize(nl);
+                for (size_t k = 0; k < nl; k++) {
+                    sub_cnt_[k] = node_bytes_[k] / dtsz_;
+                    sub_dsp_[k] = node_poff_[k] / dtsz_;
+                }
+                sa.flags |= UCC_COLL_ARGS_FLAG_IN_PLACE;
+                sa.coll_type         = UCC_COLL_TYPE_ALLGATHERV;
+                sa.dst.info_v.buffer = packed_.data();
+                sa.dst.info_v.counts = (ucc_count_t *)sub_cnt_.data();
+                sa.dst.info_v.displacements = (ucc_aint_t *)sub_dsp_.data();
+                sa.dst.info_v.datatype = dt_;
+                sa.dst.info_v.mem_type = UCC_MEMORY_TYPE_HOST;
+                ucc_status_t st = launch(team_->leaders_team.get(), sa);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 2: { /* node bcast of the full packed vector */
+                sa.coll_type         = UCC_COLL_TYPE_BCAST;
+                sa.root              = 0;
+                sa.src.info.buffer   = packed_.data();
+                sa.src.info.count    = total_ / dtsz_;
+                sa.src.info.datatype = dt_;
+                sa.src.info.mem_type = UCC_MEMORY_TYPE_HOST;
+                ucc_status_t st = launch(team_->node_team.get(), sa);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 3: { /* local unpack into user dst positions */
+                uint8_t *dst = (uint8_t *)a_.dst.info_v.buffer;
+                for (uint32_t r = 0; r < team_->size; r++) {
+                    if (cnt_[r]) {
+                        memcpy(dst + udsp_[r], packed_.data() + pdsp_[r],
+                               cnt_[r]);
+                    }
+                }
+                return UCC_OK;
+            }
+            default:
+                return UCC_ERR_INVALID_PARAM;
+            }
+        }
+    }
+
+    ucc_status_t launch(Team *t, ucc_coll_args_t &sa)
+    {
+        ucc_status_t st = ucc_collective_init(
+            &sa, &req_h_, reinterpret_cast<ucc_team_h>(t));
+        if (st != UCC_OK) {
+            return st;
+        }
+        return ucc_collective_post(req_h_);
+    }
+
+    Team                 *team_;
+    ucc_coll_args_t       a_;
+    ucc_coll_req_h        req_h_ = nullptr;
+    int                   phase_ = 0, my_node_idx_ = -1;
+    bool                  leader_ = false, inplace_ = false;
+    size_t                dtsz_ = 4, total_ = 0;
+    ucc_datatype_t        dt_ = UCC_DT_FLOAT32;
+    std::vector<size_t>   cnt_, udsp_, pdsp_, node_bytes_, node_poff_;
+    std::vector<uint64_t> sub_cnt_, sub_dsp_;
+    std::vector<uint8_t>  packed_;
+};
+
 /* ---- hier reduce (2step role): node reduce -> leaders reduce to the
  * root's node leader -> [node bcast delivers to a non-leader root].
  * Reference parity: cl/hier reduce 2step
@@ -1209,6 +1422,24 @@ void add_scores(Team *team)
         return UCC_OK;
     };
     team->score_map.add(UCC_COLL_TYPE_BCAST, UCC_MEMORY_TYPE_HOST, b);
+
+    ScoreRange ag;
+    ag.start    = 0;
+    ag.end      = SIZE_MAX;
+    ag.score    = 60;
+    ag.tl_name  = "hier";
+    ag.alg_name = "node_packed";
+    ag.init     = [](const ucc_coll_args_t &args, Team *t,
+                 Task **task) -> ucc_status_t {
+        if ((args.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) ||
+            !ucc_dt_is_predefined(args.dst.info_v.datatype)) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        *task = new HierAllgathervTask(t->ctx, t, args);
+        return UCC_OK;
+    };
+    team->score_map.add(UCC_COLL_TYPE_ALLGATHERV, UCC_MEMORY_TYPE_HOST,
+                        ag);
 
     ScoreRange rd;
     rd.start    = 0;

@@ -91,6 +91,30 @@ reqs = job.coll("reduce", [
 job.run(reqs)
 np.testing.assert_allclose(dst, exp5, rtol=1e-5, atol=1e-5)
 
+# hier allgatherv (node-packed relay + unpack), ragged counts incl. 0
+assert "@hier/node_packed" in smap, smap
+cnts = [137, 0, 5001, 64, 2999, 1][:n]
+while len(cnts) < n:
+    cnts.append(71 * len(cnts))
+dsps = []
+off = 17  # leading gap: displacements need not be packed
+for cq in cnts:
+    dsps.append(off)
+    off += cq + 3  # gaps between blocks
+tot = off + 5
+srcs6 = [rng.random(max(cnts[r], 1)).astype(np.float64)[:cnts[r]]
+         for r in range(n)]
+dsts6 = [np.zeros(tot, np.float64) for _ in range(n)]
+reqs = job.coll("allgatherv", [
+    dict(src=(srcs6[r].ctypes.data if cnts[r] else 0),
+         dst=dsts6[r].ctypes.data, count=cnts[r], dt=dtypes.FLOAT64,
+         dst_counts=cnts, dst_displs=dsps) for r in range(n)])
+job.run(reqs)
+for d in dsts6:
+    for r in range(n):
+        np.testing.assert_array_equal(
+            d[dsps[r]:dsps[r] + cnts[r]], srcs6[r])
+
 # hier barrier (fanin -> leaders barrier -> fanout), repeated
 for _ in range(3):
     reqs = job.coll("barrier", [dict(src=0, dst=0, count=0, dt=dtypes.INT8)
