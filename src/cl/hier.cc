@@ -737,6 +737,390 @@ class SplitRailAllreduceTask final : public Task {
     std::vector<uint64_t> cnt_, dsp_;
 };
 
+static uint64_t v_cnt_at(const ucc_coll_args_t &a, const void *counts,
+                         uint32_t r)
+{
+    return (a.flags & UCC_COLL_ARGS_FLAG_COUNT_64BIT)
+               ? ((const uint64_t *)counts)[r]
+               : ((const uint32_t *)counts)[r];
+}
+static uint64_t v_dsp_at(const ucc_coll_args_t &a, const void *displs,
+                         uint32_t r)
+{
+    return (a.flags & UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT)
+               ? ((const uint64_t *)displs)[r]
+               : ((const uint32_t *)displs)[r];
+}
+
+/* ---- hier alltoallv (node-aggregated): members pack their send data
+ * and gather it (plus their count row+column) to the node leader;
+ * leaders run ONE aggregated alltoallv (per-pair payload ordered
+ * dst-member-major, src-member asc); leaders repack per dst member and
+ * scatterv; members unpack by the node-major src order. Aggregation
+ * collapses nranks^2 small inter-node messages into nleaders^2 — the
+ * reference's a2av_node_thresh role (components/cl/hier/alltoallv/
+ * alltoallv.c), re-derived. Small-message path: registered below
+ * CL_HIER_A2AV_NODE_THRESH only. */
+class HierAlltoallvTask final : public Task {
+  public:
+    HierAlltoallvTask(Context *ctx, Team *team,
+                      const ucc_coll_args_t &args)
+        : Task(ctx), team_(team), a_(args)
+    {
+    }
+    ~HierAlltoallvTask() override
+    {
+        if (req_h_) {
+            ucc_collective_finalize(req_h_);
+        }
+    }
+
+    ucc_status_t post() override
+    {
+        phase_  = 0;
+        leader_ = team_->leaders_oob->my_idx >= 0;
+        if (a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        const uint32_t n = team_->size;
+        sdtsz_ = ucc_dt_size(a_.src.info_v.datatype);
+        rdtsz_ = ucc_dt_size(a_.dst.info_v.datatype);
+        row_.resize(n);
+        col_.resize(n);
+        for (uint32_t j = 0; j < n; j++) {
+            row_[j] = v_cnt_at(a_, a_.src.info_v.counts, j) * sdtsz_;
+            col_[j] = v_cnt_at(a_, a_.dst.info_v.counts, j) * rdtsz_;
+        }
+        /* node-major global src order (nodes in leader order, members in
+         * team-rank order) — the unpack order every rank agrees on */
+        nodes_.assign(team_->leader_ranks.size(),
+                      std::vector<uint32_t>());
+        for (size_t k = 0; k < team_->leader_ranks.size(); k++) {
+            uint64_t h = team_->procs[team_->leader_ranks[k]].host_hash;
+            for (uint32_t r = 0; r < n; r++) {
+                if (team_->procs[r].host_hash == h) {
+                    nodes_[k].push_back(r);
+                }
+            }
+        }
+        /* pack my send data: dst team-rank ascending */
+        size_t tot_s = 0, tot_r = 0;
+        for (uint32_t j = 0; j < n; j++) {
+            tot_s += row_[j];
+            tot_r += col_[j];
+        }
+        send_pk_.resize(tot_s);
+        recv_pk_.resize(tot_r);
+        {
+            const uint8_t *src = (const uint8_t *)a_.src.info_v.buffer;
+            size_t         off = 0;
+            for (uint32_t j = 0; j < n; j++) {
+                if (row_[j]) {
+                    memcpy(send_pk_.data() + off,
+                           src + v_dsp_at(a_,
+                                          a_.src.info_v.displacements,
+                                          j) *
+                                     sdtsz_,
+                           row_[j]);
+                    off += row_[j];
+                }
+            }
+        }
+        if (leader_) {
+            meta_.assign((size_t)nodes_[my_node_()].size() * 2 * n, 0);
+        }
+        my_meta_.resize(2 * n);
+        for (uint32_t j = 0; j < n; j++) {
+            my_meta_[j]     = row_[j];
+            my_meta_[n + j] = col_[j];
+        }
+        status = UCC_INPROGRESS;
+        return step();
+    }
+
+    ucc_status_t progress() override { return step(); }
+
+  private:
+    size_t my_node_() const
+    {
+        for (size_t k = 0; k < nodes_.size(); k++) {
+            for (uint32_t r : nodes_[k]) {
+                if (r == team_->rank) {
+                    return k;
+                }
+            }
+        }
+        return 0;
+    }
+
+    /* s_{i -> j} in bytes, from gathered node metadata (leader only;
+     * i must be a member of my node) */
+    uint64_t s_row(size_t member_idx, uint32_t j) const
+    {
+        return meta_[member_idx * 2 * team_->size + j];
+    }
+    /* r_{j <- i} = s_{i -> j} for my member j, any global i */
+    uint64_t r_col(size_t member_idx, uint32_t i) const
+    {
+        return meta_[member_idx * 2 * team_->size + team_->size + i];
+    }
+
+    ucc_status_t step()
+    {
+        const uint32_t n  = team_->size;
+        const size_t   nl = nodes_.size();
+        while (true) {
+            if (req_h_) {
+                ucc_status_t st = ucc_collective_test(req_h_);
+                if (st == UCC_INPROGRESS) {
+                    return UCC_INPROGRESS;
+                }
+                ucc_collective_finalize(req_h_);
+                req_h_ = nullptr;
+                if (st != UCC_OK) {
+                    return st;
+                }
+                phase_++;
+            }
+            ucc_coll_args_t sa{};
+            sa.mask  = UCC_COLL_ARGS_FIELD_FLAGS;
+            sa.flags = UCC_COLL_ARGS_FLAG_COUNT_64BIT |
+                       UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT;
+            switch (phase_) {
+            case 0: { /* node gather of [row|column] metadata */
+                sa.coll_type         = UCC_COLL_TYPE_GATHER;
+                sa.root              = 0;
+                sa.src.info.buffer   = my_meta_.data();
+                sa.src.info.count    = 2 * n;
+                sa.src.info.datatype = UCC_DT_UINT64;
+                sa.src.info.mem_type = UCC_MEMORY_TYPE_HOST;
+                sa.dst.info          = sa.src.info;
+                sa.dst.info.count    = (uint64_t)2 * n *
+                                    nodes_[my_node_()].size();
+                if (leader_) {
+                    sa.dst.info.buffer = meta_.data();
+                }
+                ucc_status_t st = launch(team_->node_team.get(), sa);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 1: { /* node gatherv of packed send payloads */
+                const auto &mem = nodes_[my_node_()];
+                sa.coll_type         = UCC_COLL_TYPE_GATHERV;
+                sa.root              = 0;
+                sa.src.info.buffer   = send_pk_.data();
+                sa.src.info.count    = send_pk_.size();
+                sa.src.info.datatype = UCC_DT_UINT8;
+                sa.src.info.mem_type = UCC_MEMORY_TYPE_HOST;
+                if (leader_) {
+                    sub_cnt_.resize(mem.size());
+                    sub_dsp_.resize(mem.size());
+                    size_t off = 0;
+                    for (size_t m = 0; m < mem.size(); m++) {
+                        uint64_t t = 0;
+                        for (uint32_t j = 0; j < n; j++) {
+                            t += s_row(m, j);
+                        }
+                        sub_cnt_[m] = t;
+                        sub_dsp_[m] = off;
+                        off += t;
+                    }
+                    gath_.resize(off);
+                    sa.dst.info_v.buffer        = gath_.data();
+                    sa.dst.info_v.counts = (ucc_count_t *)sub_cnt_.data();
+                    sa.dst.info_v.displacements =
+                        (ucc_aint_t *)sub_dsp_.data();
+                    sa.dst.info_v.datatype = UCC_DT_UINT8;
+                    sa.dst.info_v.mem_type = UCC_MEMORY_TYPE_HOST;
+                }
+                ucc_status_t st = launch(team_->node_team.get(), sa);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 2: { /* leaders aggregated alltoallv */
+                if (!leader_) {
+                    phase_ = 3;
+                    continue;
+                }
+                const auto &mem = nodes_[my_node_()];
+                /* send buf to leader l: for dst member j of node l (in
+                 * node order), for src member m of my node: s_m[j] */
+                a2a_scnt_.assign(nl, 0);
+                a2a_sdsp_.assign(nl, 0);
+                a2a_rcnt_.assign(nl, 0);
+                a2a_rdsp_.assign(nl, 0);
+                size_t tot = 0;
+                for (size_t l = 0; l < nl; l++) {
+                    a2a_sdsp_[l] = tot;
+                    for (uint32_t j : nodes_[l]) {
+                        for (size_t m = 0; m < mem.size(); m++) {
+                            a2a_scnt_[l] += s_row(m, j);
+                        }
+                    }
+                    tot += a2a_scnt_[l];
+                }
+                a2a_send_.resize(tot);
+                /* member m's packed block starts at sub_dsp_[m]; within
+                 * it, dst j's piece is at the row prefix sum */
+                {
+                    std::vector<uint64_t> moff(mem.size());
+                    size_t w = 0;
+                    for (size_t l = 0; l < nl; l++) {
+                        for (uint32_t j : nodes_[l]) {
+                            for (size_t m = 0; m < mem.size(); m++) {
+                                uint64_t pre = 0;
+                                for (uint32_t j2 = 0; j2 < j; j2++) {
+                                    pre += s_row(m, j2);
+                                }
+                                uint64_t len = s_row(m, j);
+                                if (len) {
+                                    memcpy(a2a_send_.data() + w,
+                                           gath_.data() + sub_dsp_[m] +
+                                               pre,
+                                           len);
+                                }
+                                w += len;
+                            }
+                        }
+                    }
+                    (void)moff;
+                }
+                size_t rtot = 0;
+                for (size_t l = 0; l < nl; l++) {
+                    a2a_rdsp_[l] = rtot;
+                    for (size_t m = 0; m < mem.size(); m++) {
+                        for (uint32_t i : nodes_[l]) {
+                            a2a_rcnt_[l] += r_col(m, i);
+                        }
+                    }
+                    rtot += a2a_rcnt_[l];
+                }
+                a2a_recv_.resize(rtot);
+                sa.coll_type                = UCC_COLL_TYPE_ALLTOALLV;
+                sa.src.info_v.buffer        = a2a_send_.data();
+                sa.src.info_v.counts        = (ucc_count_t *)a2a_scnt_.data();
+                sa.src.info_v.displacements = (ucc_aint_t *)a2a_sdsp_.data();
+                sa.src.info_v.datatype      = UCC_DT_UINT8;
+                sa.src.info_v.mem_type      = UCC_MEMORY_TYPE_HOST;
+                sa.dst.info_v.buffer        = a2a_recv_.data();
+                sa.dst.info_v.counts        = (ucc_count_t *)a2a_rcnt_.data();
+                sa.dst.info_v.displacements = (ucc_aint_t *)a2a_rdsp_.data();
+                sa.dst.info_v.datatype      = UCC_DT_UINT8;
+                sa.dst.info_v.mem_type      = UCC_MEMORY_TYPE_HOST;
+                ucc_status_t st = launch(team_->leaders_team.get(), sa);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 3: { /* node scatterv of per-dst-member blocks, ordered
+                       * by src in node-major global order */
+                const auto &mem = nodes_[my_node_()];
+                sa.coll_type = UCC_COLL_TYPE_SCATTERV;
+                sa.root      = 0;
+                if (leader_) {
+                    /* repack: arriving block from node l is
+                     * (dst member-major, src member asc); target is
+                     * (dst member-major over ALL srcs node-major) */
+                    sub_cnt_.assign(mem.size(), 0);
+                    sub_dsp_.assign(mem.size(), 0);
+                    size_t tot = 0;
+                    for (size_t m = 0; m < mem.size(); m++) {
+                        sub_dsp_[m] = tot;
+                        for (uint32_t i = 0; i < n; i++) {
+                            sub_cnt_[m] += r_col(m, i);
+                        }
+                        tot += sub_cnt_[m];
+                    }
+                    scat_.resize(tot);
+                    std::vector<uint64_t> roff(nl);
+                    for (size_t l = 0; l < nl; l++) {
+                        roff[l] = a2a_rdsp_[l];
+                    }
+                    for (size_t m = 0; m < mem.size(); m++) {
+                        size_t w = sub_dsp_[m];
+                        for (size_t l = 0; l < nl; l++) {
+                            for (uint32_t i : nodes_[l]) {
+                                uint64_t len = r_col(m, i);
+                                if (len) {
+                                    memcpy(scat_.data() + w,
+                                           a2a_recv_.data() + roff[l],
+                                           len);
+                                }
+                                roff[l] += len;
+                                w += len;
+                            }
+                        }
+                    }
+                    sa.src.info_v.buffer        = scat_.data();
+                    sa.src.info_v.counts = (ucc_count_t *)sub_cnt_.data();
+                    sa.src.info_v.displacements =
+                        (ucc_aint_t *)sub_dsp_.data();
+                    sa.src.info_v.datatype = UCC_DT_UINT8;
+                    sa.src.info_v.mem_type = UCC_MEMORY_TYPE_HOST;
+                }
+                sa.dst.info.buffer   = recv_pk_.data();
+                sa.dst.info.count    = recv_pk_.size();
+                sa.dst.info.datatype = UCC_DT_UINT8;
+                sa.dst.info.mem_type = UCC_MEMORY_TYPE_HOST;
+                ucc_status_t st = launch(team_->node_team.get(), sa);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                break;
+            }
+            case 4: { /* unpack: src blocks arrive node-major */
+                uint8_t *dst = (uint8_t *)a_.dst.info_v.buffer;
+                size_t   off = 0;
+                for (size_t l = 0; l < nl; l++) {
+                    for (uint32_t i : nodes_[l]) {
+                        if (col_[i]) {
+                            memcpy(dst + v_dsp_at(
+                                             a_,
+                                             a_.dst.info_v.displacements,
+                                             i) *
+                                             rdtsz_,
+                                   recv_pk_.data() + off, col_[i]);
+                            off += col_[i];
+                        }
+                    }
+                }
+                return UCC_OK;
+            }
+            default:
+                return UCC_ERR_INVALID_PARAM;
+            }
+        }
+    }
+
+    ucc_status_t launch(Team *t, ucc_coll_args_t &sa)
+    {
+        ucc_status_t st = ucc_collective_init(
+            &sa, &req_h_, reinterpret_cast<ucc_team_h>(t));
+        if (st != UCC_OK) {
+            return st;
+        }
+        return ucc_collective_post(req_h_);
+    }
+
+    Team                              *team_;
+    ucc_coll_args_t                    a_;
+    ucc_coll_req_h                     req_h_ = nullptr;
+    int                                phase_ = 0;
+    bool                               leader_ = false;
+    size_t                             sdtsz_ = 1, rdtsz_ = 1;
+    std::vector<uint64_t>              row_, col_, my_meta_, meta_;
+    std::vector<std::vector<uint32_t>> nodes_;
+    std::vector<uint64_t> sub_cnt_, sub_dsp_, a2a_scnt_, a2a_sdsp_,
+        a2a_rcnt_, a2a_rdsp_;
+    std::vector<uint8_t> send_pk_, recv_pk_, gath_, a2a_send_, a2a_recv_,
+        scat_;
+};
+
 /* ---- hier allgatherv: node gatherv to the leader (packed) -> leaders
  * allgatherv of node-aggregated blocks -> node bcast of the full packed
  * vector -> local unpack into user dst positions. All per-rank counts
@@ -758,21 +1142,6 @@ class HierAllgathervTask final : public Task {
         }
     }
 
-    static uint64_t cnt_at(const ucc_coll_args_t &a, const void *counts,
-                           uint32_t r)
-    {
-        return (a.flags & UCC_COLL_ARGS_FLAG_COUNT_64BIT)
-                   ? ((const uint64_t *)counts)[r]
-                   : ((const uint32_t *)counts)[r];
-    }
-    static uint64_t dsp_at(const ucc_coll_args_t &a, const void *displs,
-                           uint32_t r)
-    {
-        return (a.flags & UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT)
-                   ? ((const uint64_t *)displs)[r]
-                   : ((const uint32_t *)displs)[r];
-    }
-
     ucc_status_t post() override
     {
         phase_   = 0;
@@ -789,8 +1158,8 @@ class HierAllgathervTask final : public Task {
         node_bytes_.assign(team_->leader_ranks.size(), 0);
         node_poff_.assign(team_->leader_ranks.size(), 0);
         for (uint32_t r = 0; r < n; r++) {
-            cnt_[r]  = cnt_at(a_, a_.dst.info_v.counts, r) * dtsz_;
-            udsp_[r] = dsp_at(a_, a_.dst.info_v.displacements, r) * dtsz_;
+            cnt_[r]  = v_cnt_at(a_, a_.dst.info_v.counts, r) * dtsz_;
+            udsp_[r] = v_dsp_at(a_, a_.dst.info_v.displacements, r) * dtsz_;
         }
         size_t off = 0;
         my_node_idx_ = -1;
@@ -1422,6 +1791,33 @@ void add_scores(Team *team)
         return UCC_OK;
     };
     team->score_map.add(UCC_COLL_TYPE_BCAST, UCC_MEMORY_TYPE_HOST, b);
+
+    /* node-aggregated a2av pays off while per-pair messages are small:
+     * nleaders^2 sockets instead of nranks^2. Above the threshold the
+     * flat pairwise path wins (no triple copy of big payloads). */
+    size_t a2av_thresh = Config::instance().get_size(
+        "CL_HIER", "A2AV_NODE_THRESH", 256 * 1024);
+    if (a2av_thresh > 0) {
+        ScoreRange av;
+        av.start    = 0;
+        av.end      = a2av_thresh;
+        av.score    = 60;
+        av.tl_name  = "hier";
+        av.alg_name = "node_aggregated";
+        av.init     = [](const ucc_coll_args_t &args, Team *t,
+                     Task **task) -> ucc_status_t {
+            if ((args.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) ||
+                (args.flags & UCC_COLL_ARGS_FLAG_IN_PLACE) ||
+                !ucc_dt_is_predefined(args.src.info_v.datatype) ||
+                !ucc_dt_is_predefined(args.dst.info_v.datatype)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new HierAlltoallvTask(t->ctx, t, args);
+            return UCC_OK;
+        };
+        team->score_map.add(UCC_COLL_TYPE_ALLTOALLV,
+                            UCC_MEMORY_TYPE_HOST, av);
+    }
 
     ScoreRange ag;
     ag.start    = 0;

@@ -115,6 +115,34 @@ for d in dsts6:
         np.testing.assert_array_equal(
             d[dsps[r]:dsps[r] + cnts[r]], srcs6[r])
 
+# hier node-aggregated alltoallv: skewed counts incl. zero pairs
+assert "@hier/node_aggregated" in smap, smap
+scnt = [[((r * 3 + d * 5) %% 7) * 23 + (0 if (r + d) %% 4 == 0 else 11)
+         for d in range(n)] for r in range(n)]
+rcnt = [[scnt[s][r] for s in range(n)] for r in range(n)]
+def _dsp(cs):
+    out, off = [], 2  # leading gap
+    for cq in cs:
+        out.append(off)
+        off += cq + 1  # inter-block gaps
+    return out, off
+sdsp = [_dsp(c)[0] for c in scnt]
+rdsp = [_dsp(c)[0] for c in rcnt]
+stot = [_dsp(c)[1] for c in scnt]
+rtot = [_dsp(c)[1] for c in rcnt]
+srcs7 = [rng.random(stot[r]).astype(np.float32) for r in range(n)]
+dsts7 = [np.zeros(rtot[r], np.float32) for r in range(n)]
+reqs = job.coll("alltoallv", [
+    dict(src=srcs7[r].ctypes.data, dst=dsts7[r].ctypes.data, count=0,
+         dt=dtypes.FLOAT32, src_counts=scnt[r], src_displs=sdsp[r],
+         dst_counts=rcnt[r], dst_displs=rdsp[r]) for r in range(n)])
+job.run(reqs)
+for r in range(n):
+    for s in range(n):
+        np.testing.assert_array_equal(
+            dsts7[r][rdsp[r][s]:rdsp[r][s] + rcnt[r][s]],
+            srcs7[s][sdsp[s][r]:sdsp[s][r] + scnt[s][r]])
+
 # hier barrier (fanin -> leaders barrier -> fanout), repeated
 for _ in range(3):
     reqs = job.coll("barrier", [dict(src=0, dst=0, count=0, dt=dtypes.INT8)
