@@ -326,7 +326,8 @@ PYBIND11_MODULE(_core, m)
            std::vector<uint64_t> src_counts, std::vector<uint64_t> src_displs,
            std::vector<uint64_t> dst_counts,
            std::vector<uint64_t> dst_displs, double timeout,
-           std::vector<int64_t> active_set, int tag) {
+           std::vector<int64_t> active_set, int tag, int src_mem_type) {
+            int s_mt = src_mem_type >= 0 ? src_mem_type : mem_type;
             auto r  = std::make_shared<CoreReq>();
             r->team = team;
             ucc_coll_args_t a{};
@@ -369,12 +370,12 @@ PYBIND11_MODULE(_core, m)
                 a.src.info_v.counts        = r->counts_s.data();
                 a.src.info_v.displacements = r->displs_s.data();
                 a.src.info_v.datatype      = (ucc_datatype_t)dt;
-                a.src.info_v.mem_type      = (ucc_memory_type_t)mem_type;
+                a.src.info_v.mem_type      = (ucc_memory_type_t)s_mt;
             } else {
                 a.src.info.buffer   = (void *)src;
                 a.src.info.count    = count;
                 a.src.info.datatype = (ucc_datatype_t)dt;
-                a.src.info.mem_type = (ucc_memory_type_t)mem_type;
+                a.src.info.mem_type = (ucc_memory_type_t)s_mt;
             }
             if (dv && !r->counts_d.empty()) {
                 a.dst.info_v.buffer        = (void *)dst;
@@ -401,7 +402,7 @@ PYBIND11_MODULE(_core, m)
         py::arg("dst_displs") = std::vector<uint64_t>(),
         py::arg("timeout") = 0.0,
         py::arg("active_set") = std::vector<int64_t>(),
-        py::arg("tag") = -1);
+        py::arg("tag") = -1, py::arg("src_mem_type") = -1);
 
     /* ------------------------------------------------------ mem_map */
     m.def("mem_map_export", [](uintptr_t addr, size_t len) {

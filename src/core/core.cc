@@ -736,18 +736,78 @@ ucc_status_t ucc_collective_init(ucc_coll_args_t *coll_args,
             ucc_memory_type_t md =
                 fold(dv ? a.dst.info_v.mem_type : a.dst.info.mem_type);
             /* rooted colls: the non-root side may pass only one valid
-             * buffer; only reject when both sides are meaningful */
+             * buffer; only the root's pair is meaningful */
             bool rooted = a.coll_type == UCC_COLL_TYPE_REDUCE ||
                           a.coll_type == UCC_COLL_TYPE_GATHER ||
                           a.coll_type == UCC_COLL_TYPE_GATHERV ||
                           a.coll_type == UCC_COLL_TYPE_SCATTER ||
                           a.coll_type == UCC_COLL_TYPE_SCATTERV;
-            if (!rooted && ms != md) {
-                ucc_warn("asymmetric src/dst memory types (%d vs %d) "
-                         "are not supported",
-                         (int)ms, (int)md);
-                delete req;
-                return UCC_ERR_NOT_SUPPORTED;
+            bool meaningful = !rooted || a.root == team->rank;
+            if (meaningful && ms != md &&
+                ucc_dt_is_predefined(sv ? a.src.info_v.datatype
+                                        : a.src.info.datatype)) {
+                /* stage src into a scratch of the dst's memtype, copied
+                 * at every post (reference ucc_coll.c:236-246 asymmetric
+                 * staging, re-derived: scratch lives dst-side so no
+                 * copy-back is needed and device TLs stay eligible) */
+                size_t sbytes = 0;
+                size_t dtsz   = ucc_dt_size(sv ? a.src.info_v.datatype
+                                               : a.src.info.datatype);
+                if (sv) { /* displacement layout: copy the full extent */
+                    for (uint32_t r = 0; r < team->size; r++) {
+                        uint64_t cn =
+                            ((a.flags & UCC_COLL_ARGS_FLAG_COUNT_64BIT)
+                                 ? ((const uint64_t *)
+                                        a.src.info_v.counts)[r]
+                                 : ((const uint32_t *)
+                                        a.src.info_v.counts)[r]);
+                        uint64_t dp =
+                            ((a.flags &
+                              UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT)
+                                 ? ((const uint64_t *)
+                                        a.src.info_v.displacements)[r]
+                                 : ((const uint32_t *)
+                                        a.src.info_v.displacements)[r]);
+                        size_t e = (size_t)(dp + cn) * dtsz;
+                        sbytes   = e > sbytes ? e : sbytes;
+                    }
+                } else {
+                    sbytes = (size_t)a.src.info.count * dtsz;
+                    if (a.coll_type == UCC_COLL_TYPE_ALLGATHER ||
+                        a.coll_type == UCC_COLL_TYPE_GATHER) {
+                        /* src side is one block; tolerate callers that
+                         * set src.count to the gathered total */
+                        size_t blk =
+                            ((size_t)a.dst.info.count / team->size) *
+                            dtsz;
+                        sbytes = blk < sbytes ? blk : sbytes;
+                    }
+                }
+                void *scratch = nullptr;
+                if (sbytes &&
+                    mc::alloc(&scratch, sbytes, md) != UCC_OK) {
+                    delete req;
+                    return UCC_ERR_NO_MEMORY;
+                }
+                req->asymm_scratch = scratch;
+                req->asymm_mt      = md;
+                const void *usrc   = sv ? a.src.info_v.buffer
+                                        : a.src.info.buffer;
+                ucc_memory_type_t umt = ms;
+                req->pre_post = [scratch, usrc, sbytes, md, umt]() {
+                    return sbytes ? mc::copy(scratch, md, usrc, umt,
+                                             sbytes)
+                                  : UCC_OK;
+                };
+                if (sv) {
+                    req->args.src.info_v.buffer   = scratch;
+                    req->args.src.info_v.mem_type =
+                        a.dst.info_v.mem_type;
+                } else {
+                    req->args.src.info.buffer   = scratch;
+                    req->args.src.info.mem_type =
+                        dv ? a.dst.info_v.mem_type : a.dst.info.mem_type;
+                }
             }
         }
     }
@@ -798,6 +858,13 @@ ucc_status_t ucc_collective_post(ucc_coll_req_h request)
     req->seq          = req->team->coll_seq++;
     req->super.status = UCC_INPROGRESS;
     UCC_PROFILE_REQUEST_EVENT("post", (uintptr_t)req);
+    if (req->pre_post) {
+        ucc_status_t pst = req->pre_post();
+        if (pst != UCC_OK) {
+            req->super.status = pst;
+            return pst;
+        }
+    }
     task_start(req->task);
     if (req->task->status != UCC_INPROGRESS) {
         req->super.status = req->task->status;
@@ -815,6 +882,9 @@ ucc_status_t ucc_collective_finalize(ucc_coll_req_h request)
 {
     auto *req = reinterpret_cast<CollRequest *>(request);
     UCC_PROFILE_REQUEST_FREE("finalize", (uintptr_t)req);
+    if (req->asymm_scratch) {
+        mc::mem_free(req->asymm_scratch, req->asymm_mt);
+    }
     delete req->task;
     delete req;
     return UCC_OK;
@@ -886,6 +956,13 @@ ucc_status_t ucc_collective_triggered_post(ucc_ee_h ee, ucc_ev_t *ev)
     auto *req = reinterpret_cast<CollRequest *>(ev->req);
     if (req->posted && !req->persistent) {
         return UCC_ERR_INVALID_PARAM;
+    }
+    if (req->pre_post) { /* asymm staging copy: synchronous (the user's
+                          * src must be final before a triggered post) */
+        ucc_status_t pst = req->pre_post();
+        if (pst != UCC_OK) {
+            return pst;
+        }
     }
     ucc_status_t st = req->task->triggered_post(e->stream);
     if (st != UCC_OK && st != UCC_INPROGRESS) {

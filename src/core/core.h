@@ -326,6 +326,11 @@ struct CollRequest {
     bool            persistent = false;
     bool            posted     = false;
     uint64_t        seq        = 0;
+    /* asymmetric src/dst memtype staging (reference ucc_coll.c:236-246):
+     * src is copied into a scratch of the dst's memtype at every post */
+    std::function<ucc_status_t()> pre_post;
+    void             *asymm_scratch = nullptr;
+    ucc_memory_type_t asymm_mt      = UCC_MEMORY_TYPE_HOST;
 };
 
 /* msgsize used for score-map range selection (bytes, per reference

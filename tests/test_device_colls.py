@@ -228,6 +228,51 @@ def test_alltoallv_device_skewed(job):
             torch.testing.assert_close(got, exp)
 
 
+def test_asymm_memtype_allreduce(job):
+    """Host src + device dst: core stages src into a device scratch at
+    post (reference ucc_coll.c:236-246 role) and the device TL runs."""
+    torch.manual_seed(31)
+    n = job.n
+    count = 40_000
+    srcs = [np.random.default_rng(r).standard_normal(count)
+            .astype(np.float32) for r in range(n)]
+    dsts = [torch.zeros(count, device="cuda") for _ in range(n)]
+    expected = torch.from_numpy(np.sum(srcs, axis=0))
+    _run_device(job, "allreduce", [
+        dict(src=srcs[r].ctypes.data, dst=dsts[r].data_ptr(), count=count,
+             dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA,
+             src_mem_type=dtypes.MEM_HOST)
+        for r in range(n)
+    ])
+    for d in dsts:
+        torch.testing.assert_close(d.cpu(), expected, rtol=1e-5,
+                                   atol=1e-4)
+
+
+def test_asymm_memtype_gather_root(job):
+    """Rooted asymm: the root's src is host while everyone's data side is
+    device — only the root stages, keeping TL choice consistent."""
+    torch.manual_seed(32)
+    n, per, root = job.n, 5000, 1
+    srcs_dev = [torch.randn(per, device="cuda") for _ in range(n)]
+    src_host = srcs_dev[root].cpu().numpy()
+    dst = torch.zeros(per * n, device="cuda")
+    args = []
+    for r in range(n):
+        if r == root:
+            args.append(dict(src=src_host.ctypes.data,
+                             dst=dst.data_ptr(), count=per * n,
+                             dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA,
+                             src_mem_type=dtypes.MEM_HOST, root=root))
+        else:
+            args.append(dict(src=srcs_dev[r].data_ptr(), dst=0,
+                             count=per, dt=dtypes.FLOAT32,
+                             mem_type=dtypes.MEM_CUDA, root=root))
+    _run_device(job, "gather", args)
+    expected = torch.cat([s.cpu() for s in srcs_dev])
+    torch.testing.assert_close(dst.cpu(), expected)
+
+
 def test_alltoallv_device_multifrag(job):
     """Gated a2av crossing the staging-cell boundary: every pair moves
     more than one cell (chunk/n bytes), so the global-max exchange and
