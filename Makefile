@@ -40,7 +40,7 @@ SHLIB  := $(BUILD)/libucc_amd.so
 
 PERFTEST := build/ucc_perftest
 INFO     := build/ucc_info
-NTESTS   := build/test_generic_dt
+NTESTS   := build/test_generic_dt build/test_obj_size
 
 all: $(MODULE) $(SHLIB) $(PERFTEST) $(INFO) $(NTESTS)
 
@@ -49,6 +49,9 @@ $(BUILD)/tests/%.o: tests/native/%.cc
 	$(HIPCC) $(CXXFLAGS) -c $< -o $@
 
 build/test_generic_dt: $(BUILD)/tests/test_generic_dt.o $(LIB_OBJS)
+	$(HIPCC) $^ -o $@ -lrt -L/opt/rocm/lib -lrccl -Wl,-rpath,/opt/rocm/lib
+
+build/test_obj_size: $(BUILD)/tests/test_obj_size.o $(LIB_OBJS)
 	$(HIPCC) $^ -o $@ -lrt -L/opt/rocm/lib -lrccl -Wl,-rpath,/opt/rocm/lib
 
 $(BUILD)/tools/%.o: tools/%.cc tools/shm_oob.h

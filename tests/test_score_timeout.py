@@ -99,6 +99,18 @@ def test_generic_datatype_native():
     assert p.returncode == 0 and "GENERIC_DT_OK" in p.stdout
 
 
+def test_native_obj_size():
+    """Native object-size regression guard (reference test_obj_size.cc
+    role): progress-engine objects stay cache-friendly."""
+    binary = os.path.join(REPO, "build", "test_obj_size")
+    if not os.path.exists(binary):
+        pytest.skip("native test binary not built")
+    p = subprocess.run([binary], capture_output=True, text=True,
+                       timeout=60)
+    sys.stdout.write(p.stdout[-500:])
+    assert p.returncode == 0 and "OBJ_SIZE_OK" in p.stdout
+
+
 def test_thread_multiple_progress():
     """THREAD_MULTIPLE: a second thread pumping ucc_context_progress
     concurrently with the posting thread (lock-free progress queue path,
@@ -270,3 +282,88 @@ def test_ucc_info_tool():
     assert "transports: self shm tcp cdna4" in p.stdout
     assert "UCC_LOG_LEVEL" in p.stdout
     assert "allreduce:host:" in p.stdout
+
+
+def test_concurrent_multi_team():
+    """THREAD_MULTIPLE with TWO teams over the same contexts: one thread
+    per team drives independent collectives concurrently, stressing
+    team-tag isolation in the shm slot segments and the MT progress
+    path. Reference MT multi-team coverage (test_mt.cc role)."""
+    import threading
+
+    import numpy as np
+
+    from ucc_amd import core, dtypes
+
+    c = core()
+    n = 4
+    libs = [c.Lib(thread_mode="multiple") for _ in range(n)]
+    ctxs = [c.Context(lib) for lib in libs]
+    oob1 = c.LocalOob(n)
+    oob2 = c.LocalOob(n)
+    teams1 = [c.team_create_post(ctxs[r], local_oob=oob1, rank=r)
+              for r in range(n)]
+    while True:
+        sts = [c.team_create_test(t) for t in teams1]
+        assert all(s >= 0 for s in sts)
+        if all(s == c.OK for s in sts):
+            break
+    teams2 = [c.team_create_post(ctxs[r], local_oob=oob2, rank=r)
+              for r in range(n)]
+    while True:
+        sts = [c.team_create_test(t) for t in teams2]
+        assert all(s >= 0 for s in sts)
+        if all(s == c.OK for s in sts):
+            break
+
+    errors = []
+
+    def drive(teams, coll, iters, seed):
+        try:
+            rng = np.random.default_rng(seed)
+            for it in range(iters):
+                if coll == "allreduce":
+                    arrs = [rng.standard_normal(1500).astype(np.float32)
+                            for _ in range(n)]
+                    outs = [np.zeros(1500, np.float32) for _ in range(n)]
+                    reqs = [c.coll_init(teams[r], "allreduce",
+                                        src=arrs[r].ctypes.data,
+                                        dst=outs[r].ctypes.data,
+                                        count=1500, dt=dtypes.FLOAT32)
+                            for r in range(n)]
+                else:
+                    root = it % n
+                    arrs = [np.zeros(900, np.float64) for _ in range(n)]
+                    arrs[root][:] = rng.standard_normal(900)
+                    outs = [arrs[root].copy()]
+                    reqs = [c.coll_init(teams[r], "bcast",
+                                        src=arrs[r].ctypes.data, dst=0,
+                                        count=900, dt=dtypes.FLOAT64,
+                                        root=root) for r in range(n)]
+                for rq in reqs:
+                    rq.post()
+                spins = 0
+                while any(rq.test() == c.INPROGRESS for rq in reqs):
+                    for ctx in ctxs:
+                        ctx.progress()
+                    spins += 1
+                    assert spins < 40_000_000
+                if coll == "allreduce":
+                    exp = np.sum(arrs, axis=0)
+                    for o in outs:
+                        np.testing.assert_allclose(o, exp, rtol=1e-5,
+                                                   atol=1e-4)
+                else:
+                    for a in arrs:
+                        np.testing.assert_array_equal(a, outs[0])
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    t1 = threading.Thread(target=drive, args=(teams1, "allreduce", 25, 5))
+    t2 = threading.Thread(target=drive, args=(teams2, "bcast", 25, 9))
+    t1.start()
+    t2.start()
+    t1.join(180)
+    t2.join(180)
+    assert not t1.is_alive() and not t2.is_alive()
+    assert not errors, errors
