@@ -187,6 +187,18 @@ class RcclTask final : public Task {
         }
         hipError_t e = hipEventQuery(ev_);
         if (e == hipErrorNotReady) {
+            /* failure detection: a comm that hit a transport error never
+             * completes the event — surface the async error instead of
+             * spinning forever (reference tl_nccl optional
+             * ncclCommGetAsyncError polling) */
+            ncclResult_t ar = ncclSuccess;
+            if (ncclCommGetAsyncError(tt_->comm_, &ar) == ncclSuccess &&
+                ar != ncclSuccess && ar != ncclInProgress) {
+                ucc_error("rccl async error: %s",
+                          ncclGetErrorString(ar));
+                status = UCC_ERR_NO_RESOURCE;
+                return status;
+            }
             return UCC_INPROGRESS;
         }
         status = e == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
