@@ -278,8 +278,9 @@ class RcclTask final : public Task {
         }
         case UCC_COLL_TYPE_BARRIER: {
             /* 4-byte allreduce (reference tl_nccl barrier) */
-            if (!barrier_buf_) {
-                hipMalloc(&barrier_buf_, 4);
+            if (!barrier_buf_ &&
+                hipMalloc(&barrier_buf_, 4) != hipSuccess) {
+                return UCC_ERR_NO_MEMORY;
             }
             RCCLCHK(ncclAllReduce(barrier_buf_, barrier_buf_, 1,
                                   ncclFloat32, ncclSum, comm, s));
