@@ -823,6 +823,141 @@ ucc_status_t ucc_collective_init(ucc_coll_args_t *coll_args,
         ucc_dt_generic_ops(req->args.src.info.datatype);
     const bool noncontig_gen =
         ncg && !(ncg->flags & UCC_GENERIC_DT_OPS_FLAG_CONTIG);
+    /* Non-contiguous generic dt, data-movement colls: pack the local
+     * contribution into a packed byte image, run the collective on
+     * plain bytes, unpack at completion (reference generic-dt
+     * pack/unpack role generalized from the tl path; assumes the type's
+     * packed size is content-independent — homogeneous elements — so
+     * every rank computes identical packed block sizes). Host memory
+     * only (pack callbacks are host functions). Bcast keeps the tl/tcp
+     * packed-image tree (handles size-dynamic packs); reduce-family
+     * colls cannot run on packed bytes. */
+    if (noncontig_gen &&
+        (req->args.coll_type == UCC_COLL_TYPE_ALLGATHER ||
+         req->args.coll_type == UCC_COLL_TYPE_ALLTOALL ||
+         req->args.coll_type == UCC_COLL_TYPE_GATHER ||
+         req->args.coll_type == UCC_COLL_TYPE_SCATTER) &&
+        req->args.src.info.mem_type == UCC_MEMORY_TYPE_HOST &&
+        ncg->ops.start_pack && ncg->ops.start_unpack &&
+        ncg->ops.packed_size && ncg->ops.pack && ncg->ops.unpack) {
+        auto pack_all = [ncg](const void *buf, size_t count, uint8_t *out,
+                              size_t psz) -> ucc_status_t {
+            void *obj = ncg->ops.start_pack(ncg->cookie, buf, count);
+            size_t got = 0;
+            while (got < psz) {
+                size_t len = psz - got;
+                if (ncg->ops.pack(obj, got, out + got, &len) != UCC_OK ||
+                    len == 0) {
+                    if (ncg->ops.finish) {
+                        ncg->ops.finish(obj);
+                    }
+                    return UCC_ERR_NO_MESSAGE;
+                }
+                got += len;
+            }
+            if (ncg->ops.finish) {
+                ncg->ops.finish(obj);
+            }
+            return UCC_OK;
+        };
+        auto unpack_all = [ncg](void *buf, size_t count,
+                                const uint8_t *in,
+                                size_t psz) -> ucc_status_t {
+            void *obj = ncg->ops.start_unpack(ncg->cookie, buf, count);
+            ucc_status_t st = ncg->ops.unpack(obj, 0, in, psz);
+            if (ncg->ops.finish) {
+                ncg->ops.finish(obj);
+            }
+            return st;
+        };
+        const ucc_coll_args_t &a    = req->args;
+        const uint32_t         nt   = team->size;
+        const bool             root = a.root == team->rank;
+        ucc_coll_type_t        ct   = a.coll_type;
+        /* per-role element geometry (count conventions: rooted colls
+         * carry the total at the root, one block elsewhere) */
+        size_t blk_el = 0, pack_el = 0, unp_el = 0;
+        bool   i_pack = false, i_unpack = false;
+        switch (ct) {
+        case UCC_COLL_TYPE_ALLGATHER:
+            blk_el  = (size_t)a.dst.info.count / nt;
+            pack_el = blk_el;
+            unp_el  = a.dst.info.count;
+            i_pack = i_unpack = true;
+            break;
+        case UCC_COLL_TYPE_ALLTOALL:
+            blk_el  = (size_t)a.dst.info.count / nt;
+            pack_el = a.dst.info.count;
+            unp_el  = a.dst.info.count;
+            i_pack = i_unpack = true;
+            break;
+        case UCC_COLL_TYPE_GATHER:
+            blk_el  = root ? (size_t)a.dst.info.count / nt
+                           : (size_t)a.src.info.count;
+            pack_el = blk_el;
+            unp_el  = root ? a.dst.info.count : 0;
+            i_pack  = true;
+            i_unpack = root;
+            break;
+        case UCC_COLL_TYPE_SCATTER:
+            blk_el  = root ? (size_t)a.src.info.count / nt
+                           : (size_t)a.dst.info.count;
+            pack_el = root ? a.src.info.count : 0;
+            unp_el  = blk_el;
+            i_pack  = root;
+            i_unpack = true;
+            break;
+        default:
+            break;
+        }
+        if (blk_el == 0) {
+            delete req;
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        /* packed bytes of one block, probed locally (content-
+         * independent size assumption) */
+        const void *probe_buf =
+            i_pack ? a.src.info.buffer : a.dst.info.buffer;
+        void  *pobj  = ncg->ops.start_pack(ncg->cookie, probe_buf,
+                                           blk_el);
+        size_t blk_p = ncg->ops.packed_size(pobj);
+        if (ncg->ops.finish) {
+            ncg->ops.finish(pobj);
+        }
+        size_t pack_p = pack_el / blk_el * blk_p;
+        size_t unp_p  = unp_el / blk_el * blk_p;
+        if (i_pack) {
+            req->gdt_send.resize(pack_p);
+            const void *ub = a.src.info.buffer;
+            req->pre_post  = [pack_all, ub, pack_el,
+                             &buf = req->gdt_send, pack_p]() {
+                return pack_all(ub, pack_el, buf.data(), pack_p);
+            };
+        }
+        if (i_unpack) {
+            req->gdt_recv.resize(unp_p);
+            void *ub           = a.dst.info.buffer;
+            req->post_complete = [unpack_all, ub, unp_el,
+                                  &buf = req->gdt_recv, unp_p]() {
+                return unpack_all(ub, unp_el, buf.data(), unp_p);
+            };
+        }
+        /* rewrite to plain bytes, preserving each role's count
+         * convention (counts scale by blk_p/blk_el) */
+        req->args.src.info.count =
+            (size_t)a.src.info.count / blk_el * blk_p;
+        req->args.dst.info.count =
+            (size_t)a.dst.info.count / blk_el * blk_p;
+        if (i_pack) {
+            req->args.src.info.buffer = req->gdt_send.data();
+        }
+        if (i_unpack) {
+            req->args.dst.info.buffer = req->gdt_recv.data();
+        }
+        req->args.src.info.datatype = UCC_DT_UINT8;
+        req->args.dst.info.datatype = UCC_DT_UINT8;
+        msgsize = coll_args_msgsize(req->args, team->rank, team->size);
+    }
     if (msgsize == 0 && !v_coll && !noncontig_gen &&
         req->args.coll_type != UCC_COLL_TYPE_BARRIER &&
         req->args.coll_type != UCC_COLL_TYPE_FANIN &&

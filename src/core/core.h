@@ -331,6 +331,10 @@ struct CollRequest {
     std::function<ucc_status_t()> pre_post;
     void             *asymm_scratch = nullptr;
     ucc_memory_type_t asymm_mt      = UCC_MEMORY_TYPE_HOST;
+    /* non-contig generic dt: runs after the inner (packed-byte)
+     * collective completes OK, before user-visible completion */
+    std::function<ucc_status_t()> post_complete;
+    std::vector<uint8_t>          gdt_send, gdt_recv;
 };
 
 /* msgsize used for score-map range selection (bytes, per reference

@@ -54,6 +54,12 @@ void task_completed(Task *t)
 {
     t->on_complete();
     if (t->req_) {
+        if (t->status == UCC_OK && t->req_->post_complete) {
+            ucc_status_t fs = t->req_->post_complete();
+            if (fs != UCC_OK) {
+                t->status = fs;
+            }
+        }
         t->req_->super.status = t->status;
         if ((t->req_->args.mask & UCC_COLL_ARGS_FIELD_CB) &&
             t->req_->args.cb.cb) {
