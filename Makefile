@@ -40,7 +40,7 @@ SHLIB  := $(BUILD)/libucc_amd.so
 
 PERFTEST := build/ucc_perftest
 INFO     := build/ucc_info
-NTESTS   := build/test_generic_dt build/test_obj_size
+NTESTS   := build/test_generic_dt build/test_obj_size build/test_mt
 
 all: $(MODULE) $(SHLIB) $(PERFTEST) $(INFO) $(NTESTS)
 
@@ -52,6 +52,9 @@ build/test_generic_dt: $(BUILD)/tests/test_generic_dt.o $(LIB_OBJS)
 	$(HIPCC) $^ -o $@ -lrt -L/opt/rocm/lib -lrccl -Wl,-rpath,/opt/rocm/lib
 
 build/test_obj_size: $(BUILD)/tests/test_obj_size.o $(LIB_OBJS)
+	$(HIPCC) $^ -o $@ -lrt -L/opt/rocm/lib -lrccl -Wl,-rpath,/opt/rocm/lib
+
+build/test_mt: $(BUILD)/tests/test_mt.o $(LIB_OBJS)
 	$(HIPCC) $^ -o $@ -lrt -L/opt/rocm/lib -lrccl -Wl,-rpath,/opt/rocm/lib
 
 $(BUILD)/tools/%.o: tools/%.cc tools/shm_oob.h
@@ -131,3 +134,29 @@ asan: build/asan_generic_dt build/asan_obj_size build/asan_perftest
 	  ASAN_OPTIONS=detect_leaks=1 ./build/asan_perftest -c $$cl -j 3 -b 8 -e 65536 -n 2 -w 1 || exit 1; done
 	for cl in allreduce gather allgatherv; do \
 	  ASAN_OPTIONS=detect_leaks=1 UCC_FAKE_NODE_SPLIT=2 ./build/asan_perftest -c $$cl -j 4 -b 8 -e 16384 -n 2 -w 1 || exit 1; done
+
+# ---------------------------------------------------------------- TSAN
+# Host ThreadSanitizer build + the native MT test (lock-free progress
+# queue, shm slot counters, concurrent multi-team driving).
+TSAN_FLAGS := -O1 -g -fsanitize=thread -fno-omit-frame-pointer
+TSAN_OBJS  := $(patsubst %.cc,$(BUILD)/tsan/%.o,$(LIB_SRCS)) \
+              $(patsubst %.hip,$(BUILD)/tsan/%.o,$(KERNEL_SRCS))
+
+$(BUILD)/tsan/%.o: %.cc
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) $(TSAN_FLAGS) -Wno-option-ignored -c $< -o $@
+
+$(BUILD)/tsan/%.o: %.hip
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) $(TSAN_FLAGS) -Wno-option-ignored -c $< -o $@
+
+$(BUILD)/tsan/tests/%.o: tests/native/%.cc
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) $(TSAN_FLAGS) -Wno-option-ignored -c $< -o $@
+
+build/tsan_mt: $(BUILD)/tsan/tests/test_mt.o $(TSAN_OBJS)
+	$(HIPCC) -fsanitize=thread $^ -o $@ -lrt -L/opt/rocm/lib -lrccl -Wl,-rpath,/opt/rocm/lib
+
+.PHONY: tsan
+tsan: build/tsan_mt
+	TSAN_OPTIONS="halt_on_error=1" ./build/tsan_mt

@@ -976,7 +976,7 @@ ucc_status_t ucc_collective_init(ucc_coll_args_t *coll_args,
         (req->args.flags & UCC_COLL_ARGS_FLAG_TIMEOUT)) {
         req->task->timeout = req->args.timeout;
     }
-    req->super.status = UCC_OPERATION_INITIALIZED;
+    req_status_store(&req->super, UCC_OPERATION_INITIALIZED);
     *request          = &req->super;
     UCC_PROFILE_REQUEST_NEW(coll_type_name(req->args.coll_type),
                             (uintptr_t)req);
@@ -991,26 +991,27 @@ ucc_status_t ucc_collective_post(ucc_coll_req_h request)
     }
     req->posted       = true;
     req->seq          = req->team->coll_seq++;
-    req->super.status = UCC_INPROGRESS;
+    req_status_store(&req->super, UCC_INPROGRESS);
     UCC_PROFILE_REQUEST_EVENT("post", (uintptr_t)req);
     if (req->pre_post) {
         ucc_status_t pst = req->pre_post();
         if (pst != UCC_OK) {
-            req->super.status = pst;
+            req_status_store(&req->super, pst);
             return pst;
         }
     }
     task_start(req->task);
-    if (req->task->status != UCC_INPROGRESS) {
-        req->super.status = req->task->status;
+    ucc_status_t ts = req->task->status.load();
+    if (ts != UCC_INPROGRESS) {
+        req_status_store(&req->super, ts);
     }
-    return req->task->status < 0 ? req->task->status : UCC_OK;
+    return ts < 0 ? ts : UCC_OK;
 }
 
 ucc_status_t ucc_collective_test(ucc_coll_req_h request)
 {
     auto *req = reinterpret_cast<CollRequest *>(request);
-    return req->super.status;
+    return req_status_load(&req->super);
 }
 
 ucc_status_t ucc_collective_finalize(ucc_coll_req_h request)
@@ -1104,7 +1105,7 @@ ucc_status_t ucc_collective_triggered_post(ucc_ee_h ee, ucc_ev_t *ev)
         return st;
     }
     req->posted        = true;
-    req->super.status  = req->task->status;
+    req_status_store(&req->super, req->task->status.load());
     ucc_ev_t done      = *ev;
     done.ev_type       = UCC_EVENT_COLLECTIVE_POST;
     e->events.push_back(done);

@@ -15,6 +15,7 @@
 #ifndef UCC_AMD_CORE_H_
 #define UCC_AMD_CORE_H_
 
+#include <atomic>
 #include <cstdint>
 #include <cstring>
 #include <deque>
@@ -65,7 +66,11 @@ class Task {
     /* Called exactly once when the task reaches a terminal status. */
     virtual void on_complete() {}
 
-    ucc_status_t status = UCC_OPERATION_INITIALIZED;
+    /* written by the progress thread, read by user threads through
+     * ucc_collective_test / re-post checks: atomic for THREAD_MULTIPLE
+     * (tsan-verified, `make tsan`). Relaxed-by-default seq_cst is fine:
+     * per-collective cost is nanoseconds. */
+    std::atomic<ucc_status_t> status{UCC_OPERATION_INITIALIZED};
     Context     *ctx_;
     CollRequest *req_        = nullptr; /* set on the root task           */
     Schedule    *sched       = nullptr; /* parent schedule, if any        */
@@ -74,7 +79,7 @@ class Task {
     std::vector<Task *> dependents; /* started when I complete            */
     double       start_time = 0;
     double       timeout    = 0; /* seconds; 0 = none                     */
-    bool         in_pq      = false;
+    std::atomic<bool> in_pq{false};
 
     void depends_on(Task *producer)
     {
@@ -318,6 +323,19 @@ struct Ee {
 };
 
 /* ----------------------------------------------------------- CollRequest */
+/* super.status is a plain C field read by user threads while the
+ * progress thread completes the task: access through atomic builtins
+ * (THREAD_MULTIPLE; tsan-verified). */
+static inline void req_status_store(ucc_coll_req_t *r, ucc_status_t s)
+{
+    __atomic_store_n((int *)&r->status, (int)s, __ATOMIC_RELEASE);
+}
+static inline ucc_status_t req_status_load(const ucc_coll_req_t *r)
+{
+    return (ucc_status_t)__atomic_load_n((const int *)&r->status,
+                                         __ATOMIC_ACQUIRE);
+}
+
 struct CollRequest {
     ucc_coll_req_t  super; /* must be first: public handle casts here      */
     ucc_coll_args_t args{};

@@ -53,19 +53,16 @@ void task_start(Task *t)
 void task_completed(Task *t)
 {
     t->on_complete();
-    if (t->req_) {
-        if (t->status == UCC_OK && t->req_->post_complete) {
-            ucc_status_t fs = t->req_->post_complete();
-            if (fs != UCC_OK) {
-                t->status = fs;
-            }
-        }
-        t->req_->super.status = t->status;
-        if ((t->req_->args.mask & UCC_COLL_ARGS_FIELD_CB) &&
-            t->req_->args.cb.cb) {
-            t->req_->args.cb.cb(t->req_->args.cb.data, t->status);
+    if (t->req_ && t->status == UCC_OK && t->req_->post_complete) {
+        ucc_status_t fs = t->req_->post_complete();
+        if (fs != UCC_OK) {
+            t->status = fs;
         }
     }
+    /* everything that dereferences t must happen BEFORE completion is
+     * published: the moment super.status flips, a THREAD_MULTIPLE user
+     * may call ucc_collective_finalize and free the task under us
+     * (tsan-caught lifetime race). */
     if (t->sched) {
         t->sched->subtask_completed(t);
     }
@@ -73,6 +70,20 @@ void task_completed(Task *t)
         d->n_satisfied++;
         if (d->n_satisfied == d->n_deps) {
             task_start(d);
+        }
+    }
+    if (t->req_) {
+        ucc_coll_req_t *pub = &t->req_->super;
+        ucc_status_t    st  = t->status.load();
+        ucc_coll_callback_t cb{};
+        bool has_cb = (t->req_->args.mask & UCC_COLL_ARGS_FIELD_CB) &&
+                      t->req_->args.cb.cb;
+        if (has_cb) {
+            cb = t->req_->args.cb;
+        }
+        req_status_store(pub, st); /* last access to t/req             */
+        if (has_cb) {
+            cb.cb(cb.data, st);
         }
     }
 }
@@ -105,7 +116,7 @@ ucc_status_t Schedule::post()
 void Schedule::subtask_completed(Task *t)
 {
     if (t->status < 0 && status >= 0) {
-        status = t->status; /* propagate first error */
+        status = t->status.load(); /* propagate first error */
         if (!posting_) {
             task_completed(this);
         }

@@ -111,6 +111,18 @@ def test_native_obj_size():
     assert p.returncode == 0 and "OBJ_SIZE_OK" in p.stdout
 
 
+def test_native_mt():
+    """Native THREAD_MULTIPLE driver (also the `make tsan` payload):
+    dedicated progress threads + two teams driven concurrently."""
+    binary = os.path.join(REPO, "build", "test_mt")
+    if not os.path.exists(binary):
+        pytest.skip("native test binary not built")
+    p = subprocess.run([binary], capture_output=True, text=True,
+                       timeout=120)
+    sys.stdout.write(p.stdout[-500:])
+    assert p.returncode == 0 and "MT_OK" in p.stdout
+
+
 def test_thread_multiple_progress():
     """THREAD_MULTIPLE: a second thread pumping ucc_context_progress
     concurrently with the posting thread (lock-free progress queue path,
