@@ -23,6 +23,9 @@
  * allgather(v), reduce_scatter(v), bcast, reduce. Host-memory colls and
  * everything else fall back per the score map.
  */
+#include <map>
+#include <mutex>
+
 #include <hip/hip_runtime.h>
 
 #include "../../core/core.h"
@@ -137,8 +140,35 @@ class Cdna4TlTeam final : public TlTeam {
         }
     }
 
+    /* team-lifetime IPC import cache: user-buffer handles opened for
+     * zero-copy stay mapped until the team dies, so re-created
+     * persistent colls on the same tensors (e.g. a rebuilt DDP bucket)
+     * skip the hipIpcOpenMemHandle on every rank (reference mem_map /
+     * rcache role for the xGMI path). */
+    void *ipc_open_cached(const hipIpcMemHandle_t &h)
+    {
+        std::string k((const char *)&h, sizeof(h));
+        std::lock_guard<std::mutex> g(ipc_mu_);
+        auto it = ipc_cache_.find(k);
+        if (it != ipc_cache_.end()) {
+            return it->second;
+        }
+        void *m = nullptr;
+        if (hipIpcOpenMemHandle(&m, h, hipIpcMemLazyEnablePeerAccess) !=
+            hipSuccess) {
+            return nullptr;
+        }
+        ipc_cache_[k] = m;
+        return m;
+    }
+    std::map<std::string, void *> ipc_cache_;
+    std::mutex                    ipc_mu_;
+
     ~Cdna4TlTeam() override
     {
+        for (auto &kv : ipc_cache_) {
+            HIPWARN(hipIpcCloseMemHandle(kv.second));
+        }
         for (uint32_t r = 0; r < (uint32_t)peers_.size(); r++) {
             if (peers_[r].ipc_s) {
                 HIPWARN(hipIpcCloseMemHandle(peers_[r].scratch));
@@ -1596,20 +1626,15 @@ class GatedCollTask final : public Cdna4Task {
                     zc_peer_dst_[r] = (uint8_t *)(uintptr_t)b.d_raw_ptr;
                     continue;
                 }
-                void *m = nullptr, *md = nullptr;
-                if (hipIpcOpenMemHandle(&m, b.h,
-                                        hipIpcMemLazyEnablePeerAccess) !=
-                    hipSuccess) {
+                void *m = tt_->ipc_open_cached(b.h), *md = nullptr;
+                if (!m) {
                     ucc_warn("zero-copy open failed, using staging");
                     zc_ = false;
                     break;
                 }
-                zc_mapped_.push_back(m);
                 zc_peer_src_[r] = (const uint8_t *)m + b.base_off;
-                if (hipIpcOpenMemHandle(&md, b.hd,
-                                        hipIpcMemLazyEnablePeerAccess) ==
-                    hipSuccess) {
-                    zc_mapped_.push_back(md);
+                md              = tt_->ipc_open_cached(b.hd);
+                if (md) {
                     zc_peer_dst_[r] = (uint8_t *)md + b.d_base_off;
                 } else if (b.pad == 1) {
                     /* dst lives in the src allocation (e.g. inplace) */
@@ -1971,7 +1996,6 @@ class GatedCollTask final : public Cdna4Task {
     bool               zc_ = false, zc_ready_ = false;
     const uint8_t     *zc_peer_src_[ec_hip::kMaxRanks] = {};
     uint8_t           *zc_peer_dst_[ec_hip::kMaxRanks] = {};
-    std::vector<void *> zc_mapped_;
 
   public:
     ~GatedCollTask() override
@@ -1979,9 +2003,7 @@ class GatedCollTask final : public Cdna4Task {
         if (pslot_ >= 0) {
             tt_->free_pslot(pslot_);
         }
-        for (void *m : zc_mapped_) {
-            HIPWARN(hipIpcCloseMemHandle(m));
-        }
+        /* zero-copy imports stay in the team cache (ipc_open_cached) */
     }
 };
 
