@@ -1284,9 +1284,20 @@ class GatedCollTask final : public Cdna4Task {
              ct_ == UCC_COLL_TYPE_REDUCE_SCATTER ||
              ct_ == UCC_COLL_TYPE_ALLGATHER ||
              ct_ == UCC_COLL_TYPE_ALLTOALL) &&
-            (a_.flags & UCC_COLL_ARGS_FLAG_PERSISTENT) &&
             Config::instance().get_bool("TL_CDNA4", "ZCOPY", true)) {
-            zc_ = true;
+            if (a_.flags & UCC_COLL_ARGS_FLAG_PERSISTENT) {
+                zc_ = true; /* exchange amortized over re-posts */
+            } else {
+                /* one-shot: the 4 host-gated exchange rounds (team
+                 * IPC-import cache makes the opens free after first
+                 * contact) pay off once the staging copy they remove
+                 * is big enough */
+                size_t min_b = Config::instance().get_size(
+                    "TL_CDNA4", "ZCOPY_ONESHOT_MIN", 16 * 1024 * 1024);
+                if (min_b > 0 && total_ >= min_b) {
+                    zc_ = true;
+                }
+            }
         }
         phase_ = zc_ && !zc_ready_ ? 10 : 0;
         if (ct_ == UCC_COLL_TYPE_ALLTOALLV && !a2av_ready_) {

@@ -228,6 +228,30 @@ def test_alltoallv_device_skewed(job):
             torch.testing.assert_close(got, exp)
 
 
+def test_oneshot_zerocopy_allreduce(job):
+    """Non-persistent allreduce >= ZCOPY_ONESHOT_MIN (16 MiB): the gated
+    pipeline runs its zero-copy exchange per post (imports served by the
+    team IPC cache) and reduces straight from user buffers."""
+    torch.manual_seed(33)
+    n = job.n
+    count = 6 * 1024 * 1024  # 24 MiB fp32 > one-shot threshold
+    srcs = [torch.randn(count, device="cuda") for _ in range(n)]
+    dsts = [torch.zeros(count, device="cuda") for _ in range(n)]
+    expected = sum(s.cpu() for s in srcs)
+    for _ in range(2):  # second pass hits the IPC-import cache
+        _run_device(job, "allreduce", [
+            dict(src=srcs[r].data_ptr(), dst=dsts[r].data_ptr(),
+                 count=count, dt=dtypes.FLOAT32,
+                 mem_type=dtypes.MEM_CUDA)
+            for r in range(n)
+        ])
+    for d in dsts:
+        torch.testing.assert_close(d.cpu(), expected, rtol=1e-5,
+                                   atol=1e-3 * n)
+    del srcs, dsts
+    torch.cuda.empty_cache()
+
+
 def test_asymm_memtype_allreduce(job):
     """Host src + device dst: core stages src into a device scratch at
     post (reference ucc_coll.c:236-246 role) and the device TL runs."""
