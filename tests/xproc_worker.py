@@ -99,6 +99,23 @@ def main():
                                    atol=1e-4)
     results.append("gated_xproc")
 
+    # 2b. one-shot (non-persistent) large allreduce: takes the
+    # zero-copy path per post (>= ZCOPY_ONESHOT_MIN); the second call
+    # re-imports peers' buffers from the team IPC cache.
+    for it in (7, 8):
+        src.copy_(full[rank] + it)
+        torch.cuda.synchronize()
+        expected = (full + it).sum(0)
+        r2b = c.coll_init(team, "allreduce", src=src.data_ptr(),
+                          dst=dst.data_ptr(), count=count,
+                          dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA)
+        wait(r2b, ctx)
+        torch.cuda.synchronize()
+        torch.testing.assert_close(dst.cpu(), expected, rtol=1e-5,
+                                   atol=1e-4)
+        del r2b
+    results.append("oneshot_zc_xproc")
+
     # 3. cross-process alltoallv (skewed, fp16)
     scnt = [[(r + 1) * (d + 1) * 1024 for d in range(world)]
             for r in range(world)]
