@@ -65,7 +65,7 @@ def main():
             (torch.bfloat16, dtypes.BFLOAT16, 3e-2, 5e-1),
             (torch.float16, dtypes.FLOAT16, 2e-2, 3e-1)]
     colls = ["allreduce", "reduce_scatter", "allgather", "alltoall",
-             "bcast", "reduce"]
+             "alltoallv", "bcast", "reduce"]
     for it in range(iters):
         coll = rng.choice(colls)
         tdt, dt, rtol, atol = rng.choice(tdts)
@@ -119,6 +119,37 @@ def main():
             exp = torch.cat([full[s].to(tdt)[rank * per:(rank + 1) * per]
                              for s in range(world)])
             torch.testing.assert_close(dst.cpu(), exp)
+        elif coll == "alltoallv":
+            # skewed pairs (some zero) through the gated a2av path
+            base = min(per, 500_000)
+            scnt = [[((r * 5 + d * 3 + it) % 4) * (base // 3)
+                     for d in range(world)] for r in range(world)]
+            rcnt = [scnt[s][rank] for s in range(world)]
+            sd, off = [], 0
+            for cq in scnt[rank]:
+                sd.append(off)
+                off += cq
+            rd, roff = [], 0
+            for cq in rcnt:
+                rd.append(roff)
+                roff += cq
+            full = [torch.randn(sum(scnt[r]), generator=g).to(tdt)
+                    for r in range(world)]
+            src = full[rank].cuda() if sum(scnt[rank]) else \
+                torch.zeros(1, dtype=tdt, device="cuda")
+            dst = torch.zeros(max(roff, 1), dtype=tdt, device="cuda")
+            req = c.coll_init(team, "alltoallv", src=src.data_ptr(),
+                              dst=dst.data_ptr(), count=0, dt=dt,
+                              mem_type=dtypes.MEM_CUDA,
+                              src_counts=scnt[rank], src_displs=sd,
+                              dst_counts=rcnt, dst_displs=rd)
+            wait(req, ctx)
+            torch.cuda.synchronize()
+            for s in range(world):
+                got = dst.cpu()[rd[s]:rd[s] + rcnt[s]]
+                sod = sum(scnt[s][:rank])
+                exp = full[s][sod:sod + rcnt[s]]
+                torch.testing.assert_close(got, exp)
         elif coll == "bcast":
             root = it % world
             full = torch.randn(per, generator=g).to(tdt)
