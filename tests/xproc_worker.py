@@ -145,6 +145,42 @@ def main():
         torch.testing.assert_close(got, exp)
     results.append("alltoallv_xproc")
 
+    # 3c. multi-fragment gated alltoallv: per-pair length crosses the
+    # staging cell (chunk/world), so the device-gated a2av runs its
+    # host global-max exchange AND >1 pipelined fragment per pair.
+    cell_elems = ((32 * 1024 * 1024 // world) & ~255) // 4
+    mf_scnt = [[cell_elems + 30_000 + 1000 * (r + d)
+                for d in range(world)] for r in range(world)]
+    mf_rcnt = [mf_scnt[s][rank] for s in range(world)]
+    mf_sd, off = [], 0
+    for cq in mf_scnt[rank]:
+        mf_sd.append(off)
+        off += cq
+    mf_rd, roff = [], 0
+    for cq in mf_rcnt:
+        mf_rd.append(roff)
+        roff += cq
+    g3 = torch.Generator().manual_seed(777)
+    mf_full = [torch.randn(sum(mf_scnt[r]), generator=g3)
+               for r in range(world)]
+    mf_src = mf_full[rank].cuda()
+    mf_dst = torch.zeros(roff, device="cuda")
+    r3c = c.coll_init(team, "alltoallv", src=mf_src.data_ptr(),
+                      dst=mf_dst.data_ptr(), count=0, dt=dtypes.FLOAT32,
+                      mem_type=dtypes.MEM_CUDA,
+                      src_counts=mf_scnt[rank], src_displs=mf_sd,
+                      dst_counts=mf_rcnt, dst_displs=mf_rd)
+    wait(r3c, ctx)
+    torch.cuda.synchronize()
+    for s in range(world):
+        got = mf_dst.cpu()[mf_rd[s]:mf_rd[s] + mf_rcnt[s]]
+        sod = sum(mf_scnt[s][:rank])
+        exp = mf_full[s][sod:sod + mf_rcnt[s]]
+        torch.testing.assert_close(got, exp)
+    del mf_src, mf_dst, mf_full
+    torch.cuda.empty_cache()
+    results.append("a2av_multifrag_gated")
+
     # 3a2. gated reduce_scatter / allgather / alltoall (device-gated
     # pipeline paths, cross-process only), repeated for counter
     # continuity across mixed coll types on shared slots.
