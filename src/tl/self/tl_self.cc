@@ -35,7 +35,10 @@ class SelfTask final : public Task {
         switch (a_.coll_type) {
         case UCC_COLL_TYPE_BARRIER:
         case UCC_COLL_TYPE_FANIN:
-        case UCC_COLL_TYPE_FANOUT: return UCC_OK;
+        case UCC_COLL_TYPE_FANOUT:
+        case UCC_COLL_TYPE_BCAST: /* single buffer: root already has it
+                                   * (dst is unused/null for bcast) */
+            return UCC_OK;
         case UCC_COLL_TYPE_ALLTOALLV: {
             if (inplace) {
                 return UCC_OK;

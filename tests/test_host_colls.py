@@ -13,7 +13,7 @@ from ucc_amd import dtypes
 from ucc_amd.testing import LocalJob
 
 
-SIZES = [2, 3, 8]
+SIZES = [1, 2, 3, 8]  # 1 = self TL (zero-transport rank)
 COUNTS = [1, 7, 1024, 70000]
 
 

@@ -6,6 +6,7 @@ capture/replay — the production topology bench.py uses at N>1."""
 import os
 import subprocess
 import sys
+import time
 
 import pytest
 
@@ -16,17 +17,33 @@ pytestmark = pytest.mark.gpu
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
+def _torchrun(script_args, timeout=600, env_extra=None):
+    """torchrun with a time-salted rendezvous port and one retry: the
+    elastic agent occasionally fails its local TCPStore bind on a busy
+    box, which is unrelated to the code under test."""
+    last = None
+    for attempt in range(2):
+        port = 29000 + (int(time.time() * 7) + attempt * 131) % 2000
+        env = dict(os.environ)
+        env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+        if env_extra:
+            env.update(env_extra)
+        proc = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node=2", "--master-addr=127.0.0.1",
+             f"--master-port={port}"] + script_args,
+            cwd=REPO, env=env, capture_output=True, text=True,
+            timeout=timeout)
+        last = proc
+        if proc.returncode == 0:
+            return proc
+    return last
+
+
 def test_xproc_device_colls():
     if not torch.cuda.is_available():
         pytest.skip("no GPU")
-    env = dict(os.environ)
-    env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
-    proc = subprocess.run(
-        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-         "--nproc-per-node=2", "--master-addr=127.0.0.1",
-         "--master-port=29517", os.path.join(REPO, "tests",
-                                             "xproc_worker.py")],
-        cwd=REPO, env=env, capture_output=True, text=True, timeout=600)
+    proc = _torchrun([os.path.join(REPO, "tests", "xproc_worker.py")])
     sys.stdout.write(proc.stdout[-3000:])
     sys.stderr.write(proc.stderr[-3000:])
     assert proc.returncode == 0
@@ -38,14 +55,8 @@ def test_xproc_stress():
     counter continuity across fused/gated/staged paths)."""
     if not torch.cuda.is_available():
         pytest.skip("no GPU")
-    env = dict(os.environ)
-    env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
-    proc = subprocess.run(
-        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-         "--nproc-per-node=2", "--master-addr=127.0.0.1",
-         "--master-port=29527", os.path.join(REPO, "tests",
-                                             "stress_worker.py"), "60"],
-        cwd=REPO, env=env, capture_output=True, text=True, timeout=600)
+    proc = _torchrun([os.path.join(REPO, "tests", "stress_worker.py"),
+                      "60"])
     sys.stdout.write(proc.stdout[-2000:])
     sys.stderr.write(proc.stderr[-2000:])
     assert proc.returncode == 0
@@ -57,15 +68,9 @@ def test_xproc_spin_timeout_recovery():
     spin — the GPU stays usable (docs/GATED_PIPELINE.md safety)."""
     if not torch.cuda.is_available():
         pytest.skip("no GPU")
-    env = dict(os.environ)
-    env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
-    env["UCC_TL_CDNA4_SPIN_LIMIT"] = "3000000"  # ~ms-scale bound
-    proc = subprocess.run(
-        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-         "--nproc-per-node=2", "--master-addr=127.0.0.1",
-         "--master-port=29533", os.path.join(REPO, "tests",
-                                             "timeout_worker.py")],
-        cwd=REPO, env=env, capture_output=True, text=True, timeout=300)
+    proc = _torchrun([os.path.join(REPO, "tests", "timeout_worker.py")],
+                     timeout=300,
+                     env_extra={"UCC_TL_CDNA4_SPIN_LIMIT": "3000000"})
     sys.stdout.write(proc.stdout[-2000:])
     sys.stderr.write(proc.stderr[-2000:])
     assert proc.returncode == 0
