@@ -480,6 +480,11 @@ typedef struct ucc_coll_args {
 ucc_status_t ucc_collective_init(ucc_coll_args_t *coll_args,
                                  ucc_coll_req_h *request, ucc_team_h team);
 ucc_status_t ucc_collective_post(ucc_coll_req_h request);
+/* Initialize AND post in one call (reference ucc.h:1984-1995 role):
+ * on success the request is in flight; on failure nothing to free. */
+ucc_status_t ucc_collective_init_and_post(ucc_coll_args_t *coll_args,
+                                          ucc_coll_req_h *request,
+                                          ucc_team_h team);
 ucc_status_t ucc_collective_test(ucc_coll_req_h request);
 ucc_status_t ucc_collective_finalize(ucc_coll_req_h request);
 

@@ -1008,6 +1008,22 @@ ucc_status_t ucc_collective_post(ucc_coll_req_h request)
     return ts < 0 ? ts : UCC_OK;
 }
 
+ucc_status_t ucc_collective_init_and_post(ucc_coll_args_t *coll_args,
+                                          ucc_coll_req_h *request,
+                                          ucc_team_h team)
+{
+    ucc_status_t st = ucc_collective_init(coll_args, request, team);
+    if (st != UCC_OK) {
+        return st;
+    }
+    st = ucc_collective_post(*request);
+    if (st < 0) {
+        ucc_collective_finalize(*request);
+        *request = nullptr;
+    }
+    return st;
+}
+
 ucc_status_t ucc_collective_test(ucc_coll_req_h request)
 {
     auto *req = reinterpret_cast<CollRequest *>(request);

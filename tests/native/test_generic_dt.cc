@@ -300,15 +300,16 @@ int main()
 
         auto run2 = [&](ucc_coll_args_t args0, ucc_coll_args_t args1) {
             ucc_coll_req_h rq[2];
-            if (ucc_collective_init(&args0, &rq[0], teams[0]) != UCC_OK) {
+            /* one rank uses the combined init_and_post entry point */
+            if (ucc_collective_init_and_post(&args0, &rq[0], teams[0]) <
+                0) {
                 return false;
             }
             if (ucc_collective_init(&args1, &rq[1], teams[1]) != UCC_OK) {
                 ucc_collective_finalize(rq[0]);
                 return false;
             }
-            if (ucc_collective_post(rq[0]) != UCC_OK ||
-                ucc_collective_post(rq[1]) != UCC_OK) {
+            if (ucc_collective_post(rq[1]) != UCC_OK) {
                 return false;
             }
             while (ucc_collective_test(rq[0]) == UCC_INPROGRESS ||
