@@ -21,8 +21,9 @@ def _torchrun(script_args, timeout=600, env_extra=None):
     """torchrun with a time-salted rendezvous port and one retry: the
     elastic agent occasionally fails its local TCPStore bind on a busy
     box, which is unrelated to the code under test."""
+    print("ndev=", torch.cuda.device_count())
     last = None
-    for attempt in range(2):
+    for attempt in range(3):
         port = 29000 + (int(time.time() * 7) + attempt * 131) % 2000
         env = dict(os.environ)
         env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
