@@ -149,6 +149,20 @@ for _ in range(3):
                               for _ in range(n)])
     job.run(reqs)
 
+# persistent allreduce over hier (re-post re-runs the composition)
+pa = [np.zeros(4000, np.float32) for _ in range(n)]
+pout = [np.zeros(4000, np.float32) for _ in range(n)]
+preq = job.coll("allreduce", [
+    dict(src=pa[r].ctypes.data, dst=pout[r].ctypes.data, count=4000,
+         dt=dtypes.FLOAT32, flags=c.FLAG_PERSISTENT) for r in range(n)])
+for it in range(3):
+    for r in range(n):
+        pa[r][:] = rng.random(4000).astype(np.float32) + it
+    expp = np.sum(pa, axis=0)
+    job.run(preq)
+    for o in pout:
+        np.testing.assert_allclose(o, expp, rtol=1e-5, atol=1e-4)
+
 # repeated (sub-team/slot reuse)
 for it in range(10):
     arrs = [np.full(257, float(r * it + 1), np.float64) for r in range(n)]

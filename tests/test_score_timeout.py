@@ -379,3 +379,24 @@ def test_concurrent_multi_team():
     t2.join(180)
     assert not t1.is_alive() and not t2.is_alive()
     assert not errors, errors
+
+
+def test_tuning_string_malformed():
+    """Malformed UCC_TUNE fragments must not crash team creation; valid
+    fragments in the same string still apply."""
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "from ucc_amd import core\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "job = LocalJob(2)\n"
+        "smap = core().score_map_str(job.teams[0])\n"
+        "assert 'allreduce' in smap\n"
+        "print('MALFORMED_OK')\n" % (REPO,))
+    for tune in ("garbage", "allreduce", ":::", "allreduce:@x:notanum",
+                 "nosuchcoll:@a:5,allreduce:@slotted:7", ","):
+        env = dict(os.environ)
+        env["UCC_TUNE"] = tune
+        p = subprocess.run([sys.executable, "-c", code], env=env,
+                           capture_output=True, text=True, timeout=120)
+        assert p.returncode == 0 and "MALFORMED_OK" in p.stdout, (
+            tune, p.stdout[-500:], p.stderr[-1000:])
