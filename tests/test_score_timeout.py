@@ -400,3 +400,63 @@ def test_tuning_string_malformed():
                            capture_output=True, text=True, timeout=120)
         assert p.returncode == 0 and "MALFORMED_OK" in p.stdout, (
             tune, p.stdout[-500:], p.stderr[-1000:])
+
+
+def test_oob_failure_surfaces_cleanly():
+    """An OOB allgather that fails mid-bootstrap must surface a negative
+    status from ucc_team_create_test — no crash, no hang. Two ranks run
+    in threads with a barrier-synchronized blocking OOB (the py OOB is
+    synchronous); both ranks' round-2 allgather raises."""
+    import threading
+
+    from ucc_amd import core
+
+    c = core()
+    libs = [c.Lib(thread_mode="multiple") for _ in range(2)]
+    ctxs = [c.Context(l) for l in libs]
+    blobs = {}
+    barrier = threading.Barrier(2, timeout=60)
+    lock = threading.Lock()
+
+    def mk(rank):
+        state = {"next": 0}
+
+        def ag(data: bytes):
+            rnd = state["next"]
+            state["next"] += 1
+            if rnd >= 1:  # team create uses exactly 2 rounds; fail the
+                raise RuntimeError("injected OOB failure")  # second
+            with lock:
+                blobs.setdefault(rnd, {})[rank] = data
+            barrier.wait()  # both contributions present
+            with lock:
+                out = [blobs[rnd][0], blobs[rnd][1]]
+            barrier.wait()  # both read before next round reuses dict
+            return out
+
+        return ag
+
+    results = {}
+
+    def driver(rank):
+        try:
+            team = c.team_create_post(ctxs[rank], py_allgather=mk(rank),
+                                      rank=rank, n_ranks=2)
+            st = c.INPROGRESS
+            for _ in range(500_000):
+                st = c.team_create_test(team)
+                if st != c.INPROGRESS:
+                    break
+            results[rank] = st
+        except Exception as e:
+            results[rank] = repr(e)
+
+    th = [threading.Thread(target=driver, args=(r,)) for r in range(2)]
+    for x in th:
+        x.start()
+    for x in th:
+        x.join(120)
+    assert all(not x.is_alive() for x in th)
+    assert len(results) == 2, results
+    for r, st in results.items():
+        assert isinstance(st, int) and st < 0, results  # error surfaced
