@@ -395,6 +395,24 @@ def test_allreduce_logical_bitwise(job):
             exp = fn(exp, a)
         for o in outs:
             np.testing.assert_array_equal(o, exp)
+    # logical ops (land/lor/lxor): C truth semantics on integers
+    arrs = [rng.integers(0, 2, 512).astype(np.int32) for _ in range(n)]
+    for op, fn in ((dtypes.OP_LAND, lambda a, b:
+                    np.logical_and(a, b).astype(np.int32)),
+                   (dtypes.OP_LOR, lambda a, b:
+                    np.logical_or(a, b).astype(np.int32)),
+                   (dtypes.OP_LXOR, lambda a, b:
+                    np.logical_xor(a, b).astype(np.int32))):
+        outs = [np.zeros(512, np.int32) for _ in range(n)]
+        reqs = job.coll("allreduce", [
+            dict(src=arrs[r].ctypes.data, dst=outs[r].ctypes.data,
+                 count=512, dt=dtypes.INT32, op=op) for r in range(n)])
+        job.run(reqs)
+        exp = arrs[0].copy()
+        for a in arrs[1:]:
+            exp = fn(exp, a)
+        for o in outs:
+            np.testing.assert_array_equal(o, exp)
 
 
 def test_gatherv_scatterv(job):
