@@ -1383,6 +1383,93 @@ class TcpAlltoallTask final : public TcpTask {
     uint8_t            *dbuf_ = nullptr;
 };
 
+/* ---- Bruck alltoall for small messages: ceil(log2 n) rounds of
+ * aggregated exchanges instead of n-1 pairwise sends, so the per-round
+ * socket latency is paid log(n) times. Reference parity: the Bruck
+ * pattern (coll_patterns/bruck_alltoall.h + tl_ucp alltoall bruck),
+ * re-derived from the standard formulation:
+ *   1. tmp[j]   = src[(me + j) mod n]
+ *   2. step s=1,2,4..: send blocks {j: j&s} to (me+s) mod n, receive
+ *      the same positions from (me-s) mod n
+ *   3. dst[(me - j) mod n] = tmp[j]
+ * Fixed block size only (non-v); registered below TCP_BRUCK_MAX. */
+class TcpAlltoallBruckTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        size_t ds = ucc_dt_size(a_.dst.info.datatype);
+        blk_      = a_.dst.info.count * ds / n_;
+        sbuf_     = (const uint8_t *)a_.src.info.buffer;
+        dbuf_     = (uint8_t *)a_.dst.info.buffer;
+        tmp_.resize((size_t)n_ * blk_);
+        xs_.resize(((size_t)n_ / 2 + 1) * blk_);
+        xr_.resize(((size_t)n_ / 2 + 1) * blk_);
+        for (uint32_t j = 0; j < n_; j++) {
+            memcpy(tmp_.data() + (size_t)j * blk_,
+                   sbuf_ + (size_t)((me_ + j) % n_) * blk_, blk_);
+        }
+        step_  = 1;
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    ucc_status_t progress_()
+    {
+        while (step_ < n_) {
+            if (phase_ == 0) { /* pack + exchange */
+                nblk_ = 0;
+                for (uint32_t j = 0; j < n_; j++) {
+                    if (j & step_) {
+                        memcpy(xs_.data() + (size_t)nblk_ * blk_,
+                               tmp_.data() + (size_t)j * blk_, blk_);
+                        nblk_++;
+                    }
+                }
+                uint32_t to   = (me_ + step_) % n_;
+                uint32_t from = (me_ + n_ - step_) % n_;
+                if (nblk_ && blk_) {
+                    send_to(to, 0x4000u + step_, xs_.data(),
+                            (size_t)nblk_ * blk_);
+                    recv_from(from, 0x4000u + step_, xr_.data(),
+                              (size_t)nblk_ * blk_);
+                }
+                phase_ = 1;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            uint32_t i = 0;
+            for (uint32_t j = 0; j < n_; j++) {
+                if (j & step_) {
+                    memcpy(tmp_.data() + (size_t)j * blk_,
+                           xr_.data() + (size_t)i * blk_, blk_);
+                    i++;
+                }
+            }
+            phase_ = 0;
+            step_ <<= 1;
+        }
+        for (uint32_t j = 0; j < n_; j++) {
+            memcpy(dbuf_ + (size_t)((me_ + n_ - j) % n_) * blk_,
+                   tmp_.data() + (size_t)j * blk_, blk_);
+        }
+        return UCC_OK;
+    }
+
+    size_t               blk_ = 0;
+    uint32_t             step_ = 1, nblk_ = 0;
+    const uint8_t       *sbuf_ = nullptr;
+    uint8_t             *dbuf_ = nullptr;
+    std::vector<uint8_t> tmp_, xs_, xr_;
+};
+
 /* ---- reduce: linear recv+reduce at root (small n); gather/scatter(v):
  * linear to/from root; reduce_scatter(v): reduce@0 + scatterv */
 class TcpRootedTask final : public TcpTask {
@@ -1769,6 +1856,33 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
     add(UCC_COLL_TYPE_ALLGATHER, mk((TcpAllgatherTask *)nullptr));
     add(UCC_COLL_TYPE_ALLGATHERV, mk((TcpAllgatherTask *)nullptr));
     add(UCC_COLL_TYPE_ALLTOALL, mk((TcpAlltoallTask *)nullptr));
+    {
+        /* Bruck wins while per-round latency dominates: log2(n) rounds
+         * vs n-1, at the cost of 2x data volume (each block moves
+         * ~log(n)/2 times) */
+        size_t bruck_max = Config::instance().get_size(
+            "TL_TCP", "BRUCK_MAX", 64 * 1024);
+        if (bruck_max > 0) {
+            ScoreRange r;
+            r.start    = 0;
+            r.end      = bruck_max;
+            r.score    = sc + 1;
+            r.tl_name  = "tcp";
+            r.alg_name = "bruck";
+            r.init     = [self](const ucc_coll_args_t &args, Team *t2,
+                            Task **task) -> ucc_status_t {
+                const ucc_generic_dt_ops_t *g =
+                    ucc_dt_generic_ops(args.dst.info.datatype);
+                if ((args.flags & UCC_COLL_ARGS_FLAG_IN_PLACE) ||
+                    (g && !(g->flags & UCC_GENERIC_DT_OPS_FLAG_CONTIG))) {
+                    return UCC_ERR_NOT_SUPPORTED;
+                }
+                *task = new TcpAlltoallBruckTask(t2->ctx, self, args);
+                return UCC_OK;
+            };
+            map.add(UCC_COLL_TYPE_ALLTOALL, UCC_MEMORY_TYPE_HOST, r);
+        }
+    }
     add(UCC_COLL_TYPE_ALLTOALLV, mk((TcpAlltoallTask *)nullptr));
     {
         size_t srg_min = Config::instance().get_size("TL_TCP", "SRG_MIN",

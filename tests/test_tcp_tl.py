@@ -130,3 +130,40 @@ def test_tcp_tl_colls():
     sys.stderr.write(p.stderr[-2000:])
     assert p.returncode == 0
     assert "TCP_TL_OK" in p.stdout
+
+
+def test_alltoall_bruck():
+    """Bruck alltoall (log2 n aggregated rounds) picked for small
+    messages over tcp; correctness incl. non-power-of-two ranks."""
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "from ucc_amd import core, dtypes\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "for n in (3, 5, 8):\n"
+        "    job = LocalJob(n)\n"
+        "    c = core()\n"
+        "    assert '@tcp/bruck' in c.score_map_str(job.teams[0])\n"
+        "    rng = np.random.default_rng(n)\n"
+        "    for per in (1, 33, 500):\n"
+        "        srcs = [rng.standard_normal(per * n).astype(np.float32)\n"
+        "                for _ in range(n)]\n"
+        "        dsts = [np.zeros(per * n, np.float32) for _ in range(n)]\n"
+        "        reqs = job.coll('alltoall', [\n"
+        "            dict(src=srcs[r].ctypes.data, dst=dsts[r].ctypes.data,\n"
+        "                 count=per * n, dt=dtypes.FLOAT32)\n"
+        "            for r in range(n)])\n"
+        "        job.run(reqs)\n"
+        "        for r in range(n):\n"
+        "            for s in range(n):\n"
+        "                np.testing.assert_array_equal(\n"
+        "                    dsts[r][s * per:(s + 1) * per],\n"
+        "                    srcs[s][r * per:(r + 1) * per])\n"
+        "print('BRUCK_OK')\n" % (REPO,))
+    env = dict(os.environ)
+    env["UCC_TL_SHM_ENABLE"] = "0"
+    p = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=300)
+    sys.stdout.write(p.stdout[-1000:])
+    sys.stderr.write(p.stderr[-2000:])
+    assert p.returncode == 0 and "BRUCK_OK" in p.stdout
