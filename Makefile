@@ -130,7 +130,7 @@ build/asan_obj_size: $(BUILD)/asan/tests/test_obj_size.o $(ASAN_OBJS)
 asan: build/asan_generic_dt build/asan_obj_size build/asan_perftest
 	ASAN_OPTIONS=detect_leaks=1 ./build/asan_obj_size
 	ASAN_OPTIONS=detect_leaks=1 ./build/asan_generic_dt
-	for cl in allreduce bcast alltoallv reduce_scatter barrier; do \
+	for cl in allreduce bcast alltoall alltoallv reduce_scatter barrier; do \
 	  ASAN_OPTIONS=detect_leaks=1 ./build/asan_perftest -c $$cl -j 3 -b 8 -e 65536 -n 2 -w 1 || exit 1; done
 	for cl in allreduce gather allgatherv; do \
 	  ASAN_OPTIONS=detect_leaks=1 UCC_FAKE_NODE_SPLIT=2 ./build/asan_perftest -c $$cl -j 4 -b 8 -e 16384 -n 2 -w 1 || exit 1; done
