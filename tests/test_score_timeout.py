@@ -460,3 +460,29 @@ def test_oob_failure_surfaces_cleanly():
     assert len(results) == 2, results
     for r, st in results.items():
         assert isinstance(st, int) and st < 0, results  # error surfaced
+
+
+def test_config_file_ini(tmp_path):
+    """UCC_CONFIG_FILE ini values apply (env still wins over file)."""
+    ini = tmp_path / "ucc.conf"
+    ini.write_text(
+        "# comment\n"
+        "UCC_TL_TCP_BRUCK_MAX = 1234\n"
+        "UCC_TL_CDNA4_CHUNK_SIZE = 8m\n")
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "from ucc_amd import core\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "job = LocalJob(2)\n"
+        "smap = core().score_map_str(job.teams[0])\n"
+        "line = [l for l in smap.splitlines() if 'bruck' in l][0]\n"
+        "assert '0-1234' in line, line  # ini-driven range\n"
+        "print('INI_OK')\n" % (REPO,))
+    env = dict(os.environ)
+    env["UCC_CONFIG_FILE"] = str(ini)
+    env["UCC_TL_SHM_ENABLE"] = "0"
+    p = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=120)
+    sys.stdout.write(p.stdout[-500:])
+    sys.stderr.write(p.stderr[-1000:])
+    assert p.returncode == 0 and "INI_OK" in p.stdout
