@@ -486,3 +486,46 @@ def test_config_file_ini(tmp_path):
     sys.stdout.write(p.stdout[-500:])
     sys.stderr.write(p.stderr[-1000:])
     assert p.returncode == 0 and "INI_OK" in p.stdout
+
+
+def test_team_churn():
+    """Repeated create/run/destroy cycles on fresh teams over the same
+    contexts: shm segment names, team ids and fds must all recycle
+    cleanly (leak class the reference covers with create/destroy
+    loops)."""
+    import numpy as np
+
+    from ucc_amd import core, dtypes
+
+    c = core()
+    n = 3
+    libs = [c.Lib() for _ in range(n)]
+    ctxs = [c.Context(l) for l in libs]
+    for cycle in range(8):
+        oob = c.LocalOob(n)
+        teams = [c.team_create_post(ctxs[r], local_oob=oob, rank=r)
+                 for r in range(n)]
+        while True:
+            sts = [c.team_create_test(t) for t in teams]
+            assert all(s >= 0 for s in sts), (cycle, sts)
+            if all(s == c.OK for s in sts):
+                break
+        arrs = [np.full(512, float(r + 1 + cycle), np.float32)
+                for r in range(n)]
+        outs = [np.zeros(512, np.float32) for _ in range(n)]
+        reqs = [c.coll_init(teams[r], "allreduce",
+                            src=arrs[r].ctypes.data,
+                            dst=outs[r].ctypes.data, count=512,
+                            dt=dtypes.FLOAT32) for r in range(n)]
+        for rq in reqs:
+            rq.post()
+        spins = 0
+        while any(rq.test() == c.INPROGRESS for rq in reqs):
+            for ctx in ctxs:
+                ctx.progress()
+            spins += 1
+            assert spins < 20_000_000, cycle
+        exp = np.full(512, sum(range(1, n + 1)) + n * cycle, np.float32)
+        for o in outs:
+            np.testing.assert_allclose(o, exp)
+        del reqs, teams, oob  # destroy before the next cycle
