@@ -190,3 +190,76 @@ def test_hier_allreduce_fake_nodes():
 
 def test_hier_allreduce_three_nodes():
     _run({"UCC_FAKE_NODE_SPLIT": "3"})
+
+
+NONUNIFORM_WORKER = r"""
+import sys
+import numpy as np
+sys.path.insert(0, %r)
+from ucc_amd import core, dtypes
+from ucc_amd.testing import LocalJob
+
+n = 5  # split 2 -> pseudo-nodes of 3 and 2 (NON-uniform)
+job = LocalJob(n)
+c = core()
+smap = c.score_map_str(job.teams[0])
+assert "@hier/rab" in smap, smap
+assert "@hier/split_rail" not in smap, smap  # needs uniform nodes
+assert "@hier/node_aggregated" in smap, smap
+
+rng = np.random.default_rng(21)
+# RAB allreduce
+arrs = [(rng.random(40_000) - 0.5).astype(np.float32) for _ in range(n)]
+outs = job.allreduce_np(arrs)
+exp = np.sum(arrs, axis=0)
+for o in outs:
+    np.testing.assert_allclose(o, exp, rtol=1e-5, atol=1e-5)
+
+# node-aggregated a2av across non-uniform nodes
+scnt = [[(r + 2 * d + 1) %% 5 * 13 for d in range(n)] for r in range(n)]
+rcnt = [[scnt[s][r] for s in range(n)] for r in range(n)]
+def _d(cs):
+    out, off = [], 0
+    for cq in cs:
+        out.append(off)
+        off += cq
+    return out, off
+sdsp = [_d(cq)[0] for cq in scnt]
+rdsp = [_d(cq)[0] for cq in rcnt]
+srcs = [rng.random(max(_d(scnt[r])[1], 1)).astype(np.float32)
+        for r in range(n)]
+dsts = [np.zeros(max(_d(rcnt[r])[1], 1), np.float32) for r in range(n)]
+reqs = job.coll("alltoallv", [
+    dict(src=srcs[r].ctypes.data, dst=dsts[r].ctypes.data, count=0,
+         dt=dtypes.FLOAT32, src_counts=scnt[r], src_displs=sdsp[r],
+         dst_counts=rcnt[r], dst_displs=rdsp[r]) for r in range(n)])
+job.run(reqs)
+for r in range(n):
+    for s in range(n):
+        np.testing.assert_array_equal(
+            dsts[r][rdsp[r][s]:rdsp[r][s] + rcnt[r][s]],
+            srcs[s][sdsp[s][r]:sdsp[s][r] + scnt[s][r]])
+
+# 2step reduce with a non-leader root on the smaller node
+srcs = [rng.random(7_777).astype(np.float32) for _ in range(n)]
+dst = np.zeros(7_777, np.float32)
+root = n - 1
+reqs = job.coll("reduce", [
+    dict(src=srcs[r].ctypes.data,
+         dst=(dst.ctypes.data if r == root else 0), count=7_777,
+         dt=dtypes.FLOAT32, root=root) for r in range(n)])
+job.run(reqs)
+np.testing.assert_allclose(dst, np.sum(srcs, axis=0), rtol=1e-5,
+                           atol=1e-5)
+print("HIER_NU_OK")
+""" % (REPO,)
+
+
+def test_hier_nonuniform_nodes():
+    env = dict(os.environ)
+    env["UCC_FAKE_NODE_SPLIT"] = "2"
+    p = subprocess.run([sys.executable, "-c", NONUNIFORM_WORKER], env=env,
+                       capture_output=True, text=True, timeout=300)
+    sys.stdout.write(p.stdout[-2000:])
+    sys.stderr.write(p.stderr[-3000:])
+    assert p.returncode == 0 and "HIER_NU_OK" in p.stdout
