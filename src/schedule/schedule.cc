@@ -16,11 +16,12 @@ double time_sec()
 
 void Context::pq_push(Task *t)
 {
-    if (t->in_pq) {
+    /* atomic claim: two concurrent pushers must not both enqueue (a
+     * double enqueue means double progress()/task_completed()) */
+    if (t->in_pq.exchange(true, std::memory_order_acq_rel)) {
         return;
     }
     if (mt && lock_free) {
-        t->in_pq = true;
         if (lf_pq.push(t)) {
             return;
         }
@@ -29,7 +30,6 @@ void Context::pq_push(Task *t)
         pq.push_back(t);
         return;
     }
-    t->in_pq = true;
     if (mt) {
         std::lock_guard<std::recursive_mutex> lk(pq_mtx);
         pq.push_back(t);

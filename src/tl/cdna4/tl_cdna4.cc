@@ -1583,28 +1583,34 @@ class GatedCollTask final : public Cdna4Task {
                                       (hipDeviceptr_t)sbuf_) !=
                     hipSuccess ||
                 hipIpcGetMemHandle(&b.h, (void *)base) != hipSuccess) {
+                /* MUST stay in the consensus rounds: a unilateral jump
+                 * to staging would satisfy peers' all_ge() gates via
+                 * close_slot() while they read a stale blob from a
+                 * prior use of this rotating slot — silent wrong mode.
+                 * Publish a magic=0 blob so the AND-consensus in
+                 * phases 11-13 forces every rank to staging. */
                 ucc_warn("zero-copy src export failed, using staging");
                 zc_ = false;
-                close_slot();
-                begin_use();
-                phase_ = 0;
-                return progress();
-            }
-            b.base_off = (uint64_t)((uintptr_t)sbuf_ - (uintptr_t)base);
-            b.raw_ptr  = (uint64_t)(uintptr_t)sbuf_;
-            b.pid      = (int32_t)getpid();
-            hipDeviceptr_t dbase  = nullptr;
-            size_t         dbsize = 0;
-            if (hipMemGetAddressRange(&dbase, &dbsize,
-                                      (hipDeviceptr_t)dbuf_) ==
-                    hipSuccess &&
-                hipIpcGetMemHandle(&b.hd, (void *)dbase) == hipSuccess) {
-                b.d_base_off =
-                    (uint64_t)((uintptr_t)dbuf_ - (uintptr_t)dbase);
-                b.d_raw_ptr = (uint64_t)(uintptr_t)dbuf_;
-                b.pad = (dbase == base) ? 1 : 0; /* same allocation */
+                b   = ZcBlob{}; /* magic = 0 */
             } else {
-                b.magic = 0; /* dst export failed: fall back */
+                b.base_off =
+                    (uint64_t)((uintptr_t)sbuf_ - (uintptr_t)base);
+                b.raw_ptr = (uint64_t)(uintptr_t)sbuf_;
+                b.pid     = (int32_t)getpid();
+                hipDeviceptr_t dbase  = nullptr;
+                size_t         dbsize = 0;
+                if (hipMemGetAddressRange(&dbase, &dbsize,
+                                          (hipDeviceptr_t)dbuf_) ==
+                        hipSuccess &&
+                    hipIpcGetMemHandle(&b.hd, (void *)dbase) ==
+                        hipSuccess) {
+                    b.d_base_off =
+                        (uint64_t)((uintptr_t)dbuf_ - (uintptr_t)dbase);
+                    b.d_raw_ptr = (uint64_t)(uintptr_t)dbuf_;
+                    b.pad = (dbase == base) ? 1 : 0; /* same alloc */
+                } else {
+                    b.magic = 0; /* dst export failed: fall back */
+                }
             }
             if (hipMemcpy(tt_->area(me_, slot_, 0, 0), &b, sizeof(b),
                           hipMemcpyHostToDevice) != hipSuccess) {
@@ -1617,7 +1623,7 @@ class GatedCollTask final : public Cdna4Task {
             if (!all_ge(1)) {
                 return UCC_INPROGRESS;
             }
-            for (uint32_t r = 0; r < n_; r++) {
+            for (uint32_t r = 0; r < n_ && zc_; r++) {
                 if (r == me_) {
                     zc_peer_src_[r] = sbuf_;
                     zc_peer_dst_[r] = dbuf_;

@@ -14,6 +14,7 @@
 #include <cstdint>
 #include <cstring>
 #include <deque>
+#include <list>
 #include <map>
 #include <vector>
 
@@ -54,7 +55,10 @@ class Conn {
     std::deque<SendOp *> sendq;
     /* ---- recv side */
     std::deque<RecvOp *>  recvq;     /* posted expectations (FIFO/tag)  */
-    std::deque<UnexpMsg>  unexp;
+    /* std::list (not deque): take_unexp() erases from the middle while
+     * cur_un may point into another element's payload vector — list
+     * erase leaves every other element (and cur_un) stable. */
+    std::list<UnexpMsg>   unexp;
     /* in-flight incoming message state */
     WireHdr               hdr{};
     size_t                hdr_got  = 0;

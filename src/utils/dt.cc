@@ -2,6 +2,8 @@
  * (src/ucc/api/ucc.h:203-221) and ucc_status.h strings. */
 #include "../api/ucc.h"
 
+#include <atomic>
+
 extern "C" {
 
 size_t ucc_dt_size(ucc_datatype_t dt)
@@ -45,7 +47,11 @@ size_t ucc_dt_size(ucc_datatype_t dt)
 struct ucc_dt_generic {
     ucc_generic_dt_ops_t ops;
     void                *cookie;
-    int                  used;
+    /* 0 = free, 1 = claimed (being filled), 2 = live. Slots are claimed
+     * with CAS so THREAD_MULTIPLE creators never share a slot, and ops
+     * are published with a release store so readers doing an acquire
+     * load of 'used' see a fully-written struct. */
+    std::atomic<int>     used;
 };
 
 static struct ucc_dt_generic g_generic_dts[64];
@@ -57,10 +63,12 @@ ucc_status_t ucc_dt_create_generic(const ucc_generic_dt_ops_t *ops,
         return UCC_ERR_INVALID_PARAM;
     }
     for (int i = 0; i < 64; i++) {
-        if (!g_generic_dts[i].used) {
+        int expect = 0;
+        if (g_generic_dts[i].used.compare_exchange_strong(
+                expect, 1, std::memory_order_acq_rel)) {
             g_generic_dts[i].ops    = *ops;
             g_generic_dts[i].cookie = cookie;
-            g_generic_dts[i].used   = 1;
+            g_generic_dts[i].used.store(2, std::memory_order_release);
             *dt = (ucc_datatype_t)(UCC_DT_USERDEFINED_BASE + i);
             return UCC_OK;
         }
@@ -72,7 +80,7 @@ void ucc_dt_destroy(ucc_datatype_t dt)
 {
     int i = (int)dt - UCC_DT_USERDEFINED_BASE;
     if (i >= 0 && i < 64) {
-        g_generic_dts[i].used = 0;
+        g_generic_dts[i].used.store(0, std::memory_order_release);
     }
 }
 
@@ -84,7 +92,8 @@ int ucc_dt_is_predefined(ucc_datatype_t dt)
 const ucc_generic_dt_ops_t *ucc_dt_generic_ops(ucc_datatype_t dt)
 {
     int i = (int)dt - UCC_DT_USERDEFINED_BASE;
-    if (i >= 0 && i < 64 && g_generic_dts[i].used) {
+    if (i >= 0 && i < 64 &&
+        g_generic_dts[i].used.load(std::memory_order_acquire) == 2) {
         return &g_generic_dts[i].ops;
     }
     return 0;
