@@ -78,7 +78,8 @@ ucc_status_t fused_allreduce(const FusedArgs &a, hipStream_t stream);
 constexpr int kMaxGraphBlocks = 64;
 constexpr int kStageCntBase   = 8 * kMaxRanks;      /* u64 idx of stage cnts */
 constexpr int kGraphCntBase   = 128;                /* u64 idx of blk cnts   */
-constexpr int kFlagsBytes     = 32768;              /* flags alloc size      */
+constexpr int kFlagsBytes     = 131072;             /* flags alloc size
+                                    (>= 8*(kGatedMirrorBase + 3*8*2*8)) */
 
 struct GraphFusedArgs {
     const void *src;
@@ -122,13 +123,21 @@ bool op_supported(ucc_datatype_t dt, ucc_reduction_op_t op);
  *   kGatedCntBase + (phase * kGatedSlots + slot) * 2 + parity
  *   phase: 0=stage 1=reduce 2=gather; slot < kGatedSlots; parity = frag&1.
  */
-constexpr int kGatedBlocks    = 64;
+constexpr int kGatedBlocks    = 64;  /* default grid (UCC_TL_CDNA4_GATED_BLOCKS) */
+constexpr int kGatedMaxBlocks = 512; /* flag-layout bound for the knob   */
 constexpr int kGatedSlots     = 8;
 constexpr int kGatedCntBase   = 704;  /* after graph counters (128..639) */
 /* per-(slot,parity,block) launch counters for the graph-replayable gated
- * mode: u64 idx kGatedGraphBase + ((slot*2)+parity)*kGatedBlocks + block
- * -> 1024..2047 (16 KiB) */
+ * mode: u64 idx kGatedGraphBase + ((slot*2)+parity)*kGatedMaxBlocks + block
+ * -> 1024..9215 */
 constexpr int kGatedGraphBase = 1024;
+/* push-model mirrors: rank r's phase counter value, WRITTEN BY RANK r
+ * over xGMI into every peer's flag buffer when its last block arrives,
+ * so waiters poll LOCAL memory only (B x n remote pollers would steal
+ * link bandwidth - the fused kernel already works this way).
+ * u64 idx kGatedMirrorBase + ((phase*kGatedSlots+slot)*2+parity)*kMaxRanks
+ *         + src_rank -> 9216..9599 */
+constexpr int kGatedMirrorBase = 9216;
 
 struct GatedArgs {
     /* data */
@@ -176,6 +185,9 @@ struct GatedArgs {
      * mapped dst + frag_off + my_slice_off) instead of the local out
      * area; the gather launch degenerates to a pure wait+signal. */
     int         zc_write;
+    int         nblocks;    /* gated grid size (team-lifetime constant,
+                               same on every rank; <= kGatedMaxBlocks;
+                               0 = kGatedBlocks default) */
     uint64_t    spin_limit; /* 0 = default kSpinLimit */
     /* per-dest cell staging (alltoall): my_in[c_dst_off[k]] <-
      * src[c_src_off[k]], c_len[k] bytes; 0 = contiguous stage of len   */

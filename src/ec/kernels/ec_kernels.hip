@@ -351,11 +351,24 @@ __device__ __forceinline__ void d_copy_bytes_part(uint8_t *dst,
     const uint64_t str = vgrid * blockDim.x;
     if ((((uintptr_t)dst | (uintptr_t)src) & 15) == 0) {
         uint64_t nv = len / 16;
-        for (uint64_t i = tid; i < nv; i += str) {
+        uint64_t i  = tid;
+        /* 4-deep load batching: all four 16B loads issue before the
+         * first dependent store (64B outstanding per lane — G7). */
+        for (; i + 3 * str < nv; i += 4 * str) {
+            uint4 v0 = ((const uint4 *)src)[i];
+            uint4 v1 = ((const uint4 *)src)[i + str];
+            uint4 v2 = ((const uint4 *)src)[i + 2 * str];
+            uint4 v3 = ((const uint4 *)src)[i + 3 * str];
+            ((uint4 *)dst)[i]           = v0;
+            ((uint4 *)dst)[i + str]     = v1;
+            ((uint4 *)dst)[i + 2 * str] = v2;
+            ((uint4 *)dst)[i + 3 * str] = v3;
+        }
+        for (; i < nv; i += str) {
             ((uint4 *)dst)[i] = ((const uint4 *)src)[i];
         }
-        for (uint64_t i = nv * 16 + tid; i < len; i += str) {
-            dst[i] = src[i];
+        for (uint64_t t = nv * 16 + tid; t < len; t += str) {
+            dst[t] = src[t];
         }
     } else {
         for (uint64_t i = tid; i < len; i += str) {
@@ -679,6 +692,13 @@ __device__ __forceinline__ int gated_idx(int phase, int slot, int parity)
     return kGatedCntBase + (phase * kGatedSlots + slot) * 2 + parity;
 }
 
+__device__ __forceinline__ int gated_mirror_idx(int phase, int slot,
+                                                int parity, int src)
+{
+    return kGatedMirrorBase +
+           ((phase * kGatedSlots + slot) * 2 + parity) * kMaxRanks + src;
+}
+
 /* Wait targets for this launch. Host-target mode reads them straight
  * from the (const) kernarg struct; graph mode derives them from this
  * block's launch ordinal within the (slot,parity) pattern: every gated
@@ -699,7 +719,7 @@ __device__ __forceinline__ GatedTargets gated_targets(const GatedArgs &a)
                 a.t_gather_wait};
     }
     uint64_t *cnt = a.local_flags + kGatedGraphBase +
-                    ((uint64_t)a.slot * 2 + a.parity) * kGatedBlocks +
+                    ((uint64_t)a.slot * 2 + a.parity) * kGatedMaxBlocks +
                     blockIdx.x;
     __shared__ uint64_t s_u;
     if (threadIdx.x == 0) {
@@ -710,15 +730,20 @@ __device__ __forceinline__ GatedTargets gated_targets(const GatedArgs &a)
     }
     __syncthreads();
     const uint64_t u = s_u;
-    const uint64_t B = (uint64_t)kGatedBlocks;
+    const uint64_t B = (uint64_t)(a.nblocks ? a.nblocks : kGatedBlocks);
     return {a.has_reduce ? (u - 1) * B : 0,
             a.has_gather ? (u - 1) * B : 0,
             a.has_gather ? (u - 1) * B : 0, u * B, u * B};
 }
 
-/* all blocks wait until every rank's counter[idx] >= target; on spin
- * timeout ALL threads leave together (no thread may return while others
- * sit at __syncthreads — that would hang the block forever). */
+/* all blocks wait until every rank reached counter[idx] >= target; on
+ * spin timeout ALL threads leave together (no thread may return while
+ * others sit at __syncthreads — that would hang the block forever).
+ * PUSH protocol: ranks publish their phase counters into every peer's
+ * mirror slots (gated_signal), so this wait polls LOCAL memory only —
+ * remote xGMI reads during the wait would steal link bandwidth from the
+ * concurrently-running data phases (B x n remote pollers; see
+ * MI355X_MICROARCH.md "polling-cost"). */
 __device__ __forceinline__ bool
 gated_wait(const GatedArgs &a, int phase, uint64_t target)
 {
@@ -731,7 +756,8 @@ gated_wait(const GatedArgs &a, int phase, uint64_t target)
         int j = (int)threadIdx.x;
         if (j < a.nranks) {
             const uint64_t *f =
-                a.peer_flags[j] + gated_idx(phase, a.slot, a.parity);
+                a.local_flags +
+                gated_mirror_idx(phase, a.slot, a.parity, j);
             uint64_t spins = 0;
             const uint64_t cap = spin_cap(a.spin_limit);
             while (sys_load(f) < target) {
@@ -752,16 +778,38 @@ gated_wait(const GatedArgs &a, int phase, uint64_t target)
     return true;
 }
 
-/* per-block completion signal on my own counter */
+/* per-block completion signal on my own counter; the LAST arriving block
+ * of this launch pushes the new cumulative value into every rank's
+ * mirror slot (n system-scope stores over xGMI — the only remote flag
+ * traffic in the whole pipeline). acq_rel on the counter makes every
+ * earlier block's release (and thus its payload writes, xGMI ones
+ * included) visible-before the mirror store. */
 __device__ __forceinline__ void gated_signal(const GatedArgs &a, int phase)
 {
     __threadfence_system();
     __syncthreads();
     if (threadIdx.x == 0) {
-        __hip_atomic_fetch_add(a.local_flags +
-                                   gated_idx(phase, a.slot, a.parity),
-                               1, __ATOMIC_RELEASE,
-                               __HIP_MEMORY_SCOPE_SYSTEM);
+        const uint64_t B = (uint64_t)(a.nblocks ? a.nblocks
+                                                : kGatedBlocks);
+        uint64_t v = __hip_atomic_fetch_add(
+                         a.local_flags +
+                             gated_idx(phase, a.slot, a.parity),
+                         1, __ATOMIC_ACQ_REL,
+                         __HIP_MEMORY_SCOPE_SYSTEM) +
+                     1;
+        if (v % B == 0) { /* last arriver of this launch */
+            /* guard against the ROCm 7.2 dropped-vmcnt hazard
+             * (MI355X_MICROARCH.md "Compiler hazard"): the release
+             * write-back must drain before the mirror stores leave */
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            for (int j = 0; j < a.nranks; j++) {
+                __hip_atomic_store(
+                    a.peer_flags[j] +
+                        gated_mirror_idx(phase, a.slot, a.parity,
+                                         a.rank),
+                    v, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_SYSTEM);
+            }
+        }
     }
 }
 
@@ -807,7 +855,58 @@ __global__ void k_staged_reduce(const GatedArgs a)
      * unrolling keeps the 8 peer base pointers in SGPRs and issues all
      * peer loads back-to-back for xGMI latency overlap. */
     P *out = (P *)a.my_out;
-    for (uint64_t i = tid; i < nv; i += str) {
+    /* 2-deep batching: both 16B vectors' peer loads are issued before
+     * any reduction consumes them, doubling outstanding loads per lane
+     * (2n x 16B; HBM-miss latency ~900 cyc and xGMI more — G7 ILP). */
+    uint64_t i = tid;
+    for (; i + str < nv; i += 2 * str) {
+        P x0[kMaxRanks], x1[kMaxRanks];
+#pragma unroll
+        for (int s = 0; s < kMaxRanks; s++) {
+            if (s < n) {
+                const P *src =
+                    (const P *)((const uint8_t *)a.peer_in[s] + a.sl_b);
+                x0[s] = src[i];
+                x1[s] = src[i + str];
+            }
+        }
+        A r0[VEC], r1[VEC];
+#pragma unroll
+        for (int k = 0; k < VEC; k++) {
+            r0[k] = Cvt<T>::load(x0[0].v[k]);
+            r1[k] = Cvt<T>::load(x1[0].v[k]);
+        }
+#pragma unroll
+        for (int s = 1; s < kMaxRanks; s++) {
+            if (s < n) {
+#pragma unroll
+                for (int k = 0; k < VEC; k++) {
+                    r0[k] = red<A, OP>(r0[k], Cvt<T>::load(x0[s].v[k]));
+                    r1[k] = red<A, OP>(r1[k], Cvt<T>::load(x1[s].v[k]));
+                }
+            }
+        }
+        P o0, o1;
+#pragma unroll
+        for (int k = 0; k < VEC; k++) {
+            o0.v[k] = Cvt<T>::store(apply_alpha<A>(r0[k], a.alpha));
+            o1.v[k] = Cvt<T>::store(apply_alpha<A>(r1[k], a.alpha));
+        }
+        if (a.zc_write) {
+            /* write my reduced slice straight into every rank's dst */
+#pragma unroll
+            for (int s = 0; s < kMaxRanks; s++) {
+                if (s < n) {
+                    ((P *)a.peer_out[s])[i]       = o0;
+                    ((P *)a.peer_out[s])[i + str] = o1;
+                }
+            }
+        } else {
+            out[i]       = o0;
+            out[i + str] = o1;
+        }
+    }
+    for (; i < nv; i += str) {
         A r[VEC];
         {
             P acc = ((const P *)((const uint8_t *)a.peer_in[0] +
@@ -834,7 +933,6 @@ __global__ void k_staged_reduce(const GatedArgs a)
             o.v[k] = Cvt<T>::store(apply_alpha<A>(r[k], a.alpha));
         }
         if (a.zc_write) {
-            /* write my reduced slice straight into every rank's dst */
 #pragma unroll
             for (int s = 0; s < kMaxRanks; s++) {
                 if (s < n) {
@@ -1076,10 +1174,16 @@ ucc_status_t gather_copy(const GatherArgs &a, hipStream_t s)
     return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
 }
 
+static inline int gated_grid(const GatedArgs &a)
+{
+    int b = a.nblocks ? a.nblocks : kGatedBlocks;
+    return b > kGatedMaxBlocks ? kGatedMaxBlocks : b;
+}
+
 ucc_status_t staged_stage(const GatedArgs &a, hipStream_t s)
 {
-    hipLaunchKernelGGL(k_staged_stage, dim3(kGatedBlocks), dim3(256), 0, s,
-                       a);
+    hipLaunchKernelGGL(k_staged_stage, dim3(gated_grid(a)), dim3(256), 0,
+                       s, a);
     return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
 }
 
@@ -1087,7 +1191,7 @@ template <typename T, int OP>
 static ucc_status_t launch_staged_reduce(const GatedArgs &a, hipStream_t s)
 {
     constexpr int VEC = VecOf<T>::value;
-    hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC>), dim3(kGatedBlocks),
+    hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC>), dim3(gated_grid(a)),
                        dim3(256), 0, s, a);
     return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
 }
@@ -1124,7 +1228,7 @@ ucc_status_t staged_reduce(const GatedArgs &a_in, hipStream_t s)
 
 ucc_status_t staged_gather(const GatedArgs &a, hipStream_t s)
 {
-    hipLaunchKernelGGL(k_staged_gather, dim3(kGatedBlocks), dim3(256), 0,
+    hipLaunchKernelGGL(k_staged_gather, dim3(gated_grid(a)), dim3(256), 0,
                        s, a);
     return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
 }

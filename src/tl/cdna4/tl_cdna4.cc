@@ -81,6 +81,7 @@ struct Cdna4Cfg {
                            (graph-captured) collectives              */
     size_t   chunk;     /* staging fragment bytes                     */
     size_t   fused_max; /* msg sizes <= this take the fused kernel    */
+    int      gated_blocks; /* gated kernel grid size (team constant)  */
 };
 
 class Cdna4Tl;
@@ -1751,7 +1752,10 @@ class GatedCollTask final : public Cdna4Task {
                                bool derive)
     {
         auto &L          = tt_->gated_launch_;
-        const uint64_t B = (uint64_t)ec_hip::kGatedBlocks;
+        const int nblk   = tt_->cfg_.gated_blocks
+                               ? tt_->cfg_.gated_blocks
+                               : ec_hip::kGatedBlocks;
+        const uint64_t B = (uint64_t)nblk;
         for (size_t f = 0; f < nfrags_; f++) {
             const uint32_t p   = (uint32_t)(f & 1);
             const size_t   off = f * gran_;
@@ -1768,6 +1772,7 @@ class GatedCollTask final : public Cdna4Task {
             ga.local_flags = tt_->flags_;
             ga.error_word  = tt_->err_host_;
             ga.spin_limit  = tt_->cfg_.spin_limit;
+            ga.nblocks     = nblk;
             ga.len         = len;
             ga.rank        = (int)me_;
             ga.nranks      = (int)n_;
@@ -2047,6 +2052,9 @@ class Cdna4Tl final : public Tl {
                     "device-gated pipeline for large colls");
         cfg.declare("TL_CDNA4", "SPIN_LIMIT", "0",
                     "device spin bound override (0 = default ~seconds)");
+        cfg.declare("TL_CDNA4", "GATED_BLOCKS", "0",
+                    "gated-pipeline kernel grid size (workgroups of 256; "
+                    "0 = built-in default; must match on every rank)");
         if (!cfg.get_bool("TL_CDNA4", "ENABLE", true) ||
             !mc::hip_available()) {
             return nullptr;
@@ -2076,6 +2084,14 @@ class Cdna4Tl final : public Tl {
         c.spin_limit =
             (uint64_t)cfg.get_int("TL_CDNA4", "SPIN_LIMIT", 0);
         c.fused_max = cfg.get_size("TL_CDNA4", "FUSED_MAX", 512 * 1024);
+        c.gated_blocks =
+            (int)cfg.get_int("TL_CDNA4", "GATED_BLOCKS", 0);
+        if (c.gated_blocks < 0) {
+            c.gated_blocks = 0;
+        }
+        if (c.gated_blocks > ec_hip::kGatedMaxBlocks) {
+            c.gated_blocks = ec_hip::kGatedMaxBlocks;
+        }
         return new Cdna4TlTeam(tlc, team, c);
     }
 };
