@@ -509,4 +509,9 @@ PYBIND11_MODULE(_core, m)
     m.def("version", []() { return std::string(ucc_get_version_string()); });
     m.def("hip_device_count",
           []() { return ucc_amd_hip_device_count_c(); });
+    /* raw (non-cache-hit) allocations done by the mc scratch mpool —
+     * regression hook: repeated asymmetric-staging colls must hit the
+     * pool, not hipMalloc (ref mc_rocm.c:97-108) */
+    m.def("scratch_raw_allocs",
+          []() { return (size_t)ucc::mc::scratch_raw_allocs(); });
 }

@@ -785,11 +785,12 @@ ucc_status_t ucc_collective_init(ucc_coll_args_t *coll_args,
                 }
                 void *scratch = nullptr;
                 if (sbytes &&
-                    mc::alloc(&scratch, sbytes, md) != UCC_OK) {
+                    mc::scratch_alloc(&scratch, sbytes, md) != UCC_OK) {
                     delete req;
                     return UCC_ERR_NO_MEMORY;
                 }
                 req->asymm_scratch = scratch;
+                req->asymm_bytes   = sbytes;
                 req->asymm_mt      = md;
                 const void *usrc   = sv ? a.src.info_v.buffer
                                         : a.src.info.buffer;
@@ -1035,7 +1036,8 @@ ucc_status_t ucc_collective_finalize(ucc_coll_req_h request)
     auto *req = reinterpret_cast<CollRequest *>(request);
     UCC_PROFILE_REQUEST_FREE("finalize", (uintptr_t)req);
     if (req->asymm_scratch) {
-        mc::mem_free(req->asymm_scratch, req->asymm_mt);
+        mc::scratch_free(req->asymm_scratch, req->asymm_bytes,
+                         req->asymm_mt);
     }
     delete req->task;
     delete req;

@@ -348,6 +348,7 @@ struct CollRequest {
      * src is copied into a scratch of the dst's memtype at every post */
     std::function<ucc_status_t()> pre_post;
     void             *asymm_scratch = nullptr;
+    size_t            asymm_bytes   = 0;
     ucc_memory_type_t asymm_mt      = UCC_MEMORY_TYPE_HOST;
     /* non-contig generic dt: runs after the inner (packed-byte)
      * collective completes OK, before user-visible completion */

@@ -19,6 +19,18 @@ int  hip_device_count();
 ucc_status_t mem_query(const void *ptr, ucc_memory_type_t *mt);
 ucc_status_t alloc(void **ptr, size_t size, ucc_memory_type_t mt);
 ucc_status_t mem_free(void *ptr, ucc_memory_type_t mt);
+
+/* Scratch mpool (parity: reference mc_rocm.c:97-108, 1MB elems / max 8
+ * cached): short-lived staging buffers for asymmetric-memtype and pack
+ * paths. Requests <= the elem size are served from a per-memtype cache
+ * of pre-sized buffers (hipMalloc in a collective's init path is a
+ * latency hazard); larger requests fall through to plain alloc.
+ * scratch_free() returns pooled buffers to the cache. Thread-safe. */
+ucc_status_t scratch_alloc(void **ptr, size_t size, ucc_memory_type_t mt);
+void         scratch_free(void *ptr, size_t size, ucc_memory_type_t mt);
+/* observability for tests: number of raw device/host allocations done
+ * by scratch_alloc since process start (cache hits do not count) */
+size_t       scratch_raw_allocs();
 /* Synchronous copy between any host/device combination. */
 ucc_status_t copy(void *dst, ucc_memory_type_t dst_mt, const void *src,
                   ucc_memory_type_t src_mt, size_t bytes);

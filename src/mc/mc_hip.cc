@@ -1,7 +1,11 @@
 #include "mc.h"
 #include "../utils/log.h"
 
+#include <atomic>
 #include <cstring>
+#include <mutex>
+#include <vector>
+
 #include <hip/hip_runtime.h>
 
 namespace ucc {
@@ -87,6 +91,64 @@ ucc_status_t memset_(void *ptr, ucc_memory_type_t mt, int value, size_t bytes)
     }
     return hipMemset(ptr, value, bytes) == hipSuccess ? UCC_OK
                                                       : UCC_ERR_NO_RESOURCE;
+}
+
+/* ----------------------------------------------------- scratch mpool */
+namespace {
+constexpr size_t kScratchElem = 1 << 20; /* 1 MiB (ref mc_rocm.c:20-25) */
+constexpr size_t kScratchMax  = 8;       /* cached elems per pool       */
+
+struct ScratchPool {
+    std::mutex          mtx;
+    std::vector<void *> free_list;
+};
+ScratchPool g_pool_dev, g_pool_host;
+std::atomic<size_t> g_raw_allocs{0};
+
+ScratchPool &pool_for(ucc_memory_type_t mt)
+{
+    return is_device_mt(mt) ? g_pool_dev : g_pool_host;
+}
+} // namespace
+
+ucc_status_t scratch_alloc(void **ptr, size_t size, ucc_memory_type_t mt)
+{
+    if (size > kScratchElem) {
+        g_raw_allocs.fetch_add(1, std::memory_order_relaxed);
+        return alloc(ptr, size, mt);
+    }
+    ScratchPool &p = pool_for(mt);
+    {
+        std::lock_guard<std::mutex> lk(p.mtx);
+        if (!p.free_list.empty()) {
+            *ptr = p.free_list.back();
+            p.free_list.pop_back();
+            return UCC_OK;
+        }
+    }
+    g_raw_allocs.fetch_add(1, std::memory_order_relaxed);
+    return alloc(ptr, kScratchElem, mt);
+}
+
+void scratch_free(void *ptr, size_t size, ucc_memory_type_t mt)
+{
+    if (!ptr) {
+        return;
+    }
+    if (size <= kScratchElem) { /* pooled size class */
+        ScratchPool &p = pool_for(mt);
+        std::lock_guard<std::mutex> lk(p.mtx);
+        if (p.free_list.size() < kScratchMax) {
+            p.free_list.push_back(ptr);
+            return;
+        }
+    }
+    (void)mem_free(ptr, mt);
+}
+
+size_t scratch_raw_allocs()
+{
+    return g_raw_allocs.load(std::memory_order_relaxed);
 }
 
 ucc_status_t ipc_export(const void *ptr, void *handle_out,

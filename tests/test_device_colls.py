@@ -9,7 +9,7 @@ import pytest
 
 torch = pytest.importorskip("torch")
 
-from ucc_amd import dtypes
+from ucc_amd import core, dtypes
 from ucc_amd.testing import LocalJob
 
 pytestmark = pytest.mark.gpu
@@ -268,6 +268,40 @@ def test_asymm_memtype_allreduce(job):
              src_mem_type=dtypes.MEM_HOST)
         for r in range(n)
     ])
+    for d in dsts:
+        torch.testing.assert_close(d.cpu(), expected, rtol=1e-5,
+                                   atol=1e-4)
+
+
+def test_asymm_scratch_mpool_reuse(job):
+    """Repeated asymmetric-staging colls must serve their device scratch
+    from the mc mpool (reference mc_rocm.c:97-108), not a fresh
+    hipMalloc per collective: after a warm first pass, raw allocation
+    count stays flat across further passes."""
+    n = job.n
+    count = 30_000  # 120 KB < the 1 MiB pool elem
+    srcs = [np.random.default_rng(100 + r).standard_normal(count)
+            .astype(np.float32) for r in range(n)]
+    dsts = [torch.zeros(count, device="cuda") for _ in range(n)]
+    expected = torch.from_numpy(np.sum(srcs, axis=0))
+
+    def one_pass():
+        _run_device(job, "allreduce", [
+            dict(src=srcs[r].ctypes.data, dst=dsts[r].data_ptr(),
+                 count=count, dt=dtypes.FLOAT32,
+                 mem_type=dtypes.MEM_CUDA,
+                 src_mem_type=dtypes.MEM_HOST)
+            for r in range(n)
+        ])
+
+    one_pass()  # warm the pool
+    before = core().scratch_raw_allocs()
+    for _ in range(4):
+        one_pass()
+    after = core().scratch_raw_allocs()
+    assert after == before, (
+        f"scratch mpool miss: {after - before} raw allocs in 4 warm "
+        "passes")
     for d in dsts:
         torch.testing.assert_close(d.cpu(), expected, rtol=1e-5,
                                    atol=1e-4)
