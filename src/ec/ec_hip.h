@@ -199,6 +199,10 @@ struct GatedArgs {
 ucc_status_t staged_stage(const GatedArgs &a, hipStream_t s);
 ucc_status_t staged_reduce(const GatedArgs &a, hipStream_t s);
 ucc_status_t staged_gather(const GatedArgs &a, hipStream_t s);
+/* copy-engine gating (SDMA data path): pure wait on gw_phase/t_gather_wait
+ * (1 block), and signal-phase-2-then-wait-team (full gated grid). */
+ucc_status_t gated_wait_only(const GatedArgs &a, hipStream_t s);
+ucc_status_t gated_done(const GatedArgs &a, hipStream_t s);
 
 } // namespace ec_hip
 } // namespace ucc

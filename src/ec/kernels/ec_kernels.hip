@@ -992,6 +992,42 @@ __global__ void k_staged_gather(const GatedArgs a)
     gated_signal(a, 2);
 }
 
+static inline int gated_grid(const GatedArgs &a)
+{
+    int b = a.nblocks ? a.nblocks : kGatedBlocks;
+    return b > kGatedMaxBlocks ? kGatedMaxBlocks : b;
+}
+
+/* Copy-engine collectives (reference alltoallv_ce.c:50,218-226 role):
+ * data moves by hipMemcpyAsync on SDMA engines, kernels only gate.
+ * k_gated_wait_only orders the memcpys after every rank's entry signal;
+ * k_gated_done publishes my-copies-complete and waits the team. */
+__global__ void k_gated_wait_only(const GatedArgs a)
+{
+    (void)gated_wait(a, a.gw_phase, a.t_gather_wait);
+}
+
+__global__ void k_gated_done(const GatedArgs a)
+{
+    gated_signal(a, 2);
+    (void)gated_wait(a, 2, a.t_gather_wait);
+}
+
+ucc_status_t gated_wait_only(const GatedArgs &a, hipStream_t s)
+{
+    /* pure wait: one block is enough (it adds no signal counts) */
+    hipLaunchKernelGGL(k_gated_wait_only, dim3(1), dim3(256), 0, s, a);
+    return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
+}
+
+ucc_status_t gated_done(const GatedArgs &a, hipStream_t s)
+{
+    /* must launch the full gated grid: signal arithmetic is counts*B */
+    hipLaunchKernelGGL(k_gated_done, dim3(gated_grid(a)), dim3(256),
+                       0, s, a);
+    return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
+}
+
 /* ----------------------------------------------------------- launchers */
 static bool aligned16(const void *p) { return (((uintptr_t)p) & 15) == 0; }
 
@@ -1172,12 +1208,6 @@ ucc_status_t gather_copy(const GatherArgs &a, hipStream_t s)
     }
     hipLaunchKernelGGL(k_gather_copy, dim3(blocks), dim3(threads), 0, s, a);
     return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
-}
-
-static inline int gated_grid(const GatedArgs &a)
-{
-    int b = a.nblocks ? a.nblocks : kGatedBlocks;
-    return b > kGatedMaxBlocks ? kGatedMaxBlocks : b;
 }
 
 ucc_status_t staged_stage(const GatedArgs &a, hipStream_t s)

@@ -82,6 +82,7 @@ struct Cdna4Cfg {
     size_t   chunk;     /* staging fragment bytes                     */
     size_t   fused_max; /* msg sizes <= this take the fused kernel    */
     int      gated_blocks; /* gated kernel grid size (team constant)  */
+    bool     ce_alltoall;  /* SDMA copy-engine alltoall data path     */
 };
 
 class Cdna4Tl;
@@ -102,12 +103,32 @@ class Cdna4TlContext final : public TlContext {
         if (copy_) {
             HIPWARN(hipStreamDestroy(copy_));
         }
+        for (auto &s : ce_) {
+            if (s) {
+                HIPWARN(hipStreamDestroy(s));
+            }
+        }
     }
     Tl *iface() override;
+
+    /* copy-engine streams: concurrent hipMemcpyAsync lanes so per-peer
+     * SDMA transfers overlap (one stream would serialize the blits) */
+    static constexpr int kNumCe = 4;
+    hipStream_t ce(int i)
+    {
+        i &= kNumCe - 1;
+        if (!ce_[i] &&
+            hipStreamCreateWithFlags(&ce_[i], hipStreamNonBlocking) !=
+                hipSuccess) {
+            ce_[i] = nullptr;
+        }
+        return ce_[i];
+    }
 
     Tl         *tl_;
     int         dev_;
     hipStream_t comp_ = nullptr, copy_ = nullptr;
+    hipStream_t ce_[kNumCe] = {};
 };
 
 struct PeerRes {
@@ -1748,9 +1769,90 @@ class GatedCollTask final : public Cdna4Task {
     }
 
   private:
+    /* SDMA copy-engine alltoall (reference alltoallv_ce.c role): data
+     * moves entirely by hipMemcpyAsync pulls over the zero-copy-mapped
+     * peer src buffers — the copy engines carry the bytes while the CUs
+     * stay free for overlapped compute. Kernels only gate:
+     *   [entry signal+reuse waits] -> [all-entered wait] ->
+     *   n SDMA pulls on kNumCe streams -> [done signal + team wait]. */
+    ucc_status_t enqueue_ce_a2a(hipStream_t comp_s)
+    {
+        auto *ctx        = (Cdna4TlContext *)tt_->tlc_;
+        auto &L          = tt_->gated_launch_;
+        const int nblk   = tt_->cfg_.gated_blocks
+                               ? tt_->cfg_.gated_blocks
+                               : ec_hip::kGatedBlocks;
+        const uint64_t B = (uint64_t)nblk;
+        const uint32_t p = 0;
+        ec_hip::GatedArgs ga{};
+        ga.local_flags = tt_->flags_;
+        ga.error_word  = tt_->err_host_;
+        ga.spin_limit  = tt_->cfg_.spin_limit;
+        ga.nblocks     = nblk;
+        ga.rank        = (int)me_;
+        ga.nranks      = (int)n_;
+        ga.slot        = (int)slot_;
+        ga.parity      = (int)p;
+        for (uint32_t r = 0; r < n_; r++) {
+            ga.peer_flags[r] = tt_->peers_[r].flags;
+        }
+        ga.len     = 0; /* entry stage is a pure signal */
+        ga.n_cells = 0;
+        ga.t_sw_reduce = L[1][slot_][p] * B;
+        ga.t_sw_gather = L[2][slot_][p] * B;
+        ucc_status_t st = ec_hip::staged_stage(ga, comp_s);
+        L[0][slot_][p]++;
+        if (st != UCC_OK) {
+            return st;
+        }
+        /* no peer buffer is written until ITS owner posted (entered) */
+        ga.gw_phase      = 0;
+        ga.t_gather_wait = L[0][slot_][p] * B;
+        st = ec_hip::gated_wait_only(ga, comp_s);
+        if (st != UCC_OK) {
+            return st;
+        }
+        HIPCHK(hipEventRecord(ev(8), comp_s));
+        bool used[Cdna4TlContext::kNumCe] = {};
+        for (uint32_t r = 0; r < n_; r++) {
+            /* stagger start peer per rank so SDMA queues don't all hit
+             * the same xGMI link first */
+            uint32_t rr = (r + me_ + 1) % n_;
+            int      si = (int)(r % Cdna4TlContext::kNumCe);
+            hipStream_t cs = ctx->ce(si);
+            if (!cs) {
+                return UCC_ERR_NO_RESOURCE;
+            }
+            if (!used[si]) {
+                HIPCHK(hipStreamWaitEvent(cs, ev(8), 0));
+                used[si] = true;
+            }
+            HIPCHK(hipMemcpyAsync(dbuf_ + (size_t)rr * out_b_,
+                                  zc_peer_src_[rr] +
+                                      (size_t)me_ * out_b_,
+                                  out_b_, hipMemcpyDeviceToDevice, cs));
+        }
+        for (int si = 0; si < Cdna4TlContext::kNumCe; si++) {
+            if (used[si]) {
+                HIPCHK(hipEventRecord(ev(9 + si), ctx->ce(si)));
+                HIPCHK(hipStreamWaitEvent(comp_s, ev(9 + si), 0));
+            }
+        }
+        /* my pulls done -> publish; complete when every rank's pulls
+         * (including reads of MY src) are done */
+        ga.t_gather_wait = (L[2][slot_][p] + 1) * B;
+        st = ec_hip::gated_done(ga, comp_s);
+        L[2][slot_][p]++;
+        return st;
+    }
+
     ucc_status_t enqueue_frags(hipStream_t stage_s, hipStream_t comp_s,
                                bool derive)
     {
+        if (ct_ == UCC_COLL_TYPE_ALLTOALL && zc_ && zc_ready_ &&
+            !derive && tt_->cfg_.ce_alltoall) {
+            return enqueue_ce_a2a(comp_s);
+        }
         auto &L          = tt_->gated_launch_;
         const int nblk   = tt_->cfg_.gated_blocks
                                ? tt_->cfg_.gated_blocks
@@ -2055,6 +2157,9 @@ class Cdna4Tl final : public Tl {
         cfg.declare("TL_CDNA4", "GATED_BLOCKS", "0",
                     "gated-pipeline kernel grid size (workgroups of 256; "
                     "0 = built-in default; must match on every rank)");
+        cfg.declare("TL_CDNA4", "CE_ALLTOALL", "1",
+                    "move zero-copy alltoall data on SDMA copy engines "
+                    "(hipMemcpyAsync) instead of gather kernels");
         if (!cfg.get_bool("TL_CDNA4", "ENABLE", true) ||
             !mc::hip_available()) {
             return nullptr;
@@ -2092,6 +2197,7 @@ class Cdna4Tl final : public Tl {
         if (c.gated_blocks > ec_hip::kGatedMaxBlocks) {
             c.gated_blocks = ec_hip::kGatedMaxBlocks;
         }
+        c.ce_alltoall = cfg.get_bool("TL_CDNA4", "CE_ALLTOALL", true);
         return new Cdna4TlTeam(tlc, team, c);
     }
 };
