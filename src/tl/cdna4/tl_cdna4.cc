@@ -83,6 +83,7 @@ struct Cdna4Cfg {
     size_t   fused_max; /* msg sizes <= this take the fused kernel    */
     int      gated_blocks; /* gated kernel grid size (team constant)  */
     bool     ce_alltoall;  /* SDMA copy-engine alltoall data path     */
+    size_t   ce_alltoall_min; /* min total msg bytes for the CE path  */
 };
 
 class Cdna4Tl;
@@ -1850,8 +1851,19 @@ class GatedCollTask final : public Cdna4Task {
                                bool derive)
     {
         if (ct_ == UCC_COLL_TYPE_ALLTOALL && zc_ && zc_ready_ &&
-            !derive && tt_->cfg_.ce_alltoall) {
+            !derive && tt_->cfg_.ce_alltoall &&
+            out_b_ * n_ >= tt_->cfg_.ce_alltoall_min) {
             return enqueue_ce_a2a(comp_s);
+        }
+        if (zc_ && zc_ready_ && nfrags_ > 1) {
+            /* zero-copy reads/writes USER buffers only — no staging
+             * area is touched, so the chunk-size bound (the only reason
+             * to fragment) does not apply: one fragment = one
+             * stage-signal + reduce + gather per iteration instead of
+             * 3 x nfrags launches and their gate waits. Symmetric on
+             * every rank (zc is team-consensus). */
+            gran_   = total_;
+            nfrags_ = 1;
         }
         auto &L          = tt_->gated_launch_;
         const int nblk   = tt_->cfg_.gated_blocks
@@ -2160,6 +2172,10 @@ class Cdna4Tl final : public Tl {
         cfg.declare("TL_CDNA4", "CE_ALLTOALL", "1",
                     "move zero-copy alltoall data on SDMA copy engines "
                     "(hipMemcpyAsync) instead of gather kernels");
+        cfg.declare("TL_CDNA4", "CE_ALLTOALL_MIN", "64m",
+                    "min total message bytes for the SDMA alltoall "
+                    "(measured crossover vs the gather kernel, "
+                    "profiles/rocprof_kernels_r02.md)");
         if (!cfg.get_bool("TL_CDNA4", "ENABLE", true) ||
             !mc::hip_available()) {
             return nullptr;
@@ -2198,6 +2214,8 @@ class Cdna4Tl final : public Tl {
             c.gated_blocks = ec_hip::kGatedMaxBlocks;
         }
         c.ce_alltoall = cfg.get_bool("TL_CDNA4", "CE_ALLTOALL", true);
+        c.ce_alltoall_min =
+            cfg.get_size("TL_CDNA4", "CE_ALLTOALL_MIN", 64 * 1024 * 1024);
         return new Cdna4TlTeam(tlc, team, c);
     }
 };
