@@ -193,7 +193,16 @@ bool wanted(Team *team)
         !team->has_oob) {
         return false;
     }
-    if (!Config::instance().get_bool("CL_HIER", "ENABLE", true)) {
+    auto &cfg = Config::instance();
+    cfg.declare("CL_HIER", "FRAG_SIZE", "4m",
+                "fragment bytes for pipelined hier allreduce "
+                "(0 = monolithic; pipelined when msg >= 2*FRAG_SIZE)");
+    cfg.declare("CL_HIER", "PIPELINE_DEPTH", "2",
+                "fragments in flight for pipelined hier collectives");
+    cfg.declare("CL_HIER", "PIPELINE_TRACE", "",
+                "append 'P/C frag stage' pipeline events to this file "
+                "(testing hook)");
+    if (!cfg.get_bool("CL_HIER", "ENABLE", true)) {
         return false;
     }
     return !team->all_same_node();
@@ -570,6 +579,120 @@ class HierAllreduceTask final : public Task {
     int             phase_ = 0;
     bool            leader_ = false, inplace_ = false;
     uint64_t        count_ = 0;
+    ucc_datatype_t  dt_ = UCC_DT_FLOAT32;
+};
+
+/* ---- RAB over the generic fragment pipeline: large messages split
+ * into frags so the node-reduce of frag f+1 overlaps the leader
+ * allreduce of frag f and the node-bcast of f-1 (reference
+ * cl_hier.h:47-57 pipeline configs + ucc_schedule_pipelined.h). */
+class HierAllreducePipeTask final : public PipelineTask {
+  public:
+    HierAllreducePipeTask(Context *ctx, Team *team,
+                          const ucc_coll_args_t &args, size_t frag_elems,
+                          size_t depth)
+        : PipelineTask(ctx), team_(team), a_(args)
+    {
+        count_   = a_.dst.info.count;
+        dt_      = a_.dst.info.datatype;
+        dtsz_    = ucc_dt_size(dt_);
+        inplace_ = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        fc_      = frag_elems ? frag_elems : count_;
+        n_frags  = (count_ + fc_ - 1) / fc_;
+        n_stages = a_.op == UCC_OP_AVG ? 4 : 3;
+        pdepth   = depth ? depth : 2;
+        stage_post = [this](size_t f, size_t s, ucc_coll_req_h *r) {
+            return do_stage(f, s, r);
+        };
+    }
+
+    ucc_status_t post() override
+    {
+        leader_ = team_->leaders_oob->my_idx >= 0;
+        return PipelineTask::post();
+    }
+
+  private:
+    ucc_status_t do_stage(size_t f, size_t s, ucc_coll_req_h *req)
+    {
+        *req         = nullptr;
+        size_t   b   = f * fc_;
+        size_t   cnt = count_ - b < fc_ ? count_ - b : fc_;
+        uint8_t *dstp =
+            (uint8_t *)a_.dst.info.buffer + b * dtsz_;
+        const uint8_t *srcp =
+            inplace_ ? dstp
+                     : (const uint8_t *)a_.src.info.buffer + b * dtsz_;
+        ucc_coll_args_t sa{};
+        sa.mask  = UCC_COLL_ARGS_FIELD_FLAGS;
+        sa.flags = a_.flags & (UCC_COLL_ARGS_FLAG_COUNT_64BIT |
+                               UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT);
+        sa.op    = a_.op == UCC_OP_AVG ? UCC_OP_SUM : a_.op;
+        sa.root  = 0;
+        Team *st_team = nullptr;
+        switch (s) {
+        case 0: /* node reduce of the slice to the node leader */
+            if (team_->node_team->size == 1) {
+                if (!inplace_) {
+                    memcpy(dstp, srcp, cnt * dtsz_);
+                }
+                return UCC_OK;
+            }
+            sa.coll_type         = UCC_COLL_TYPE_REDUCE;
+            sa.src.info          = a_.src.info;
+            sa.src.info.buffer   = (void *)srcp;
+            sa.src.info.count    = cnt;
+            sa.src.info.datatype = dt_;
+            sa.src.info.mem_type = a_.dst.info.mem_type;
+            sa.dst.info          = a_.dst.info;
+            sa.dst.info.buffer   = dstp;
+            sa.dst.info.count    = cnt;
+            st_team              = team_->node_team.get();
+            break;
+        case 1: /* leaders allreduce of the slice (leaders only) */
+            if (!leader_) {
+                return UCC_OK;
+            }
+            sa.coll_type       = UCC_COLL_TYPE_ALLREDUCE;
+            sa.flags          |= UCC_COLL_ARGS_FLAG_IN_PLACE;
+            sa.src.info        = a_.dst.info;
+            sa.src.info.buffer = dstp;
+            sa.src.info.count  = cnt;
+            sa.dst.info        = sa.src.info;
+            st_team            = team_->leaders_team.get();
+            break;
+        case 2: /* node bcast of the reduced slice */
+            if (team_->node_team->size == 1) {
+                return UCC_OK;
+            }
+            sa.coll_type       = UCC_COLL_TYPE_BCAST;
+            sa.src.info        = a_.dst.info;
+            sa.src.info.buffer = dstp;
+            sa.src.info.count  = cnt;
+            st_team            = team_->node_team.get();
+            break;
+        case 3: { /* AVG: scale the completed slice by 1/N */
+            const void *sp[1] = {dstp};
+            ec_cpu::reduce(dstp, sp, 1, cnt, dt_, UCC_OP_SUM,
+                           1.0 / (double)team_->size);
+            return UCC_OK;
+        }
+        default:
+            return UCC_ERR_INVALID_PARAM;
+        }
+        ucc_status_t st = ucc_collective_init(
+            &sa, req, reinterpret_cast<ucc_team_h>(st_team));
+        if (st != UCC_OK) {
+            return st;
+        }
+        return ucc_collective_post(*req);
+    }
+
+    Team           *team_;
+    ucc_coll_args_t a_;
+    bool            leader_ = false, inplace_ = false;
+    uint64_t        count_ = 0;
+    size_t          fc_ = 0, dtsz_ = 4;
     ucc_datatype_t  dt_ = UCC_DT_FLOAT32;
 };
 
@@ -1746,6 +1869,30 @@ void add_scores(Team *team)
             args.op != UCC_OP_MIN && args.op != UCC_OP_PROD &&
             args.op != UCC_OP_AVG) {
             return UCC_ERR_NOT_SUPPORTED;
+        }
+        auto  &cfg = Config::instance();
+        size_t fb  = cfg.get_size("CL_HIER", "FRAG_SIZE",
+                                  4 * 1024 * 1024);
+        size_t dtsz = ucc_dt_size(args.dst.info.datatype);
+        size_t msg  = (size_t)args.dst.info.count * dtsz;
+        if (fb > 0 && dtsz > 0 && msg >= 2 * fb) {
+            size_t depth = (size_t)cfg.get_int(
+                "CL_HIER", "PIPELINE_DEPTH", 2);
+            auto *pt = new HierAllreducePipeTask(
+                t->ctx, t, args, fb / dtsz, depth);
+            std::string tr = cfg.get("CL_HIER", "PIPELINE_TRACE", "");
+            if (!tr.empty()) {
+                std::string path = tr;
+                pt->trace = [path](char ev, size_t f, size_t s) {
+                    FILE *fp = fopen(path.c_str(), "a");
+                    if (fp) {
+                        fprintf(fp, "%c %zu %zu\n", ev, f, s);
+                        fclose(fp);
+                    }
+                };
+            }
+            *task = pt;
+            return UCC_OK;
         }
         *task = new HierAllreduceTask(t->ctx, t, args);
         return UCC_OK;

@@ -117,6 +117,48 @@ class Schedule : public Task {
     bool                posting_     = false;
 };
 
+/* --------------------------------------------------------- PipelineTask */
+/* Generic fragment pipeline (reference ucc_schedule_pipelined.h:36-78
+ * role, re-derived): a collective is split into n_frags fragments, each
+ * running the same n_stages-stage chain; at most pdepth fragments are in
+ * flight, so stage s+1 of fragment f overlaps stage s of fragment f+1.
+ * Stages are sub-collective requests produced by stage_post(frag, stage)
+ * — posting happens in FRAGMENT ORDER per stage on every rank, which is
+ * the matching invariant our TLs need (tl/tcp tags and tl/shm slots are
+ * assigned by per-team post sequence). A stage may "skip" by returning
+ * UCC_OK with *req = nullptr. */
+class PipelineTask : public Task {
+  public:
+    using StagePost = std::function<ucc_status_t(
+        size_t frag, size_t stage, ucc_coll_req_h *req)>;
+
+    explicit PipelineTask(Context *ctx) : Task(ctx) {}
+    ~PipelineTask() override;
+
+    size_t    n_frags  = 1;
+    size_t    n_stages = 1;
+    size_t    pdepth   = 2;
+    StagePost stage_post;
+    /* optional observer for tests/tracing: ev 'P' = stage posted,
+     * 'C' = stage completed */
+    std::function<void(char ev, size_t frag, size_t stage)> trace;
+
+    ucc_status_t post() override;
+    ucc_status_t progress() override { return drive(); }
+
+  private:
+    struct Flight {
+        size_t         frag = 0, stage = 0;
+        ucc_coll_req_h req    = nullptr;
+        bool           posted = false, active = false;
+    };
+    ucc_status_t drive();
+
+    std::vector<Flight> fl_;
+    std::vector<size_t> ord_;  /* per-stage posted-fragment count */
+    size_t              next_ = 0, done_ = 0;
+};
+
 /* ------------------------------------------------------------- ScoreMap  */
 using CollInitFn =
     std::function<ucc_status_t(const ucc_coll_args_t &, Team *, Task **)>;

@@ -132,6 +132,92 @@ void Schedule::subtask_completed(Task *t)
     }
 }
 
+/* ----------------------------------------------------- PipelineTask */
+PipelineTask::~PipelineTask()
+{
+    for (auto &f : fl_) {
+        if (f.req) {
+            ucc_collective_finalize(f.req);
+        }
+    }
+}
+
+ucc_status_t PipelineTask::post()
+{
+    done_ = 0;
+    next_ = 0;
+    ord_.assign(n_stages, 0);
+    size_t d = pdepth < n_frags ? pdepth : n_frags;
+    if (d == 0) {
+        status = UCC_OK;
+        return UCC_OK;
+    }
+    fl_.assign(d, Flight{});
+    for (size_t i = 0; i < d; i++) {
+        fl_[i].frag   = next_++;
+        fl_[i].active = true;
+    }
+    status = UCC_INPROGRESS;
+    return drive();
+}
+
+ucc_status_t PipelineTask::drive()
+{
+    bool moved = true;
+    while (moved) {
+        moved = false;
+        for (auto &f : fl_) {
+            if (!f.active) {
+                continue;
+            }
+            if (!f.posted) {
+                if (ord_[f.stage] != f.frag) {
+                    continue; /* stage s posts in fragment order */
+                }
+                ucc_coll_req_h r  = nullptr;
+                ucc_status_t   st = stage_post(f.frag, f.stage, &r);
+                if (st != UCC_OK) {
+                    return st;
+                }
+                f.req    = r;
+                f.posted = true;
+                ord_[f.stage]++;
+                if (trace) {
+                    trace('P', f.frag, f.stage);
+                }
+                moved = true;
+            }
+            ucc_status_t st = f.req ? ucc_collective_test(f.req) : UCC_OK;
+            if (st == UCC_INPROGRESS) {
+                continue;
+            }
+            if (f.req) {
+                ucc_collective_finalize(f.req);
+                f.req = nullptr;
+            }
+            if (st != UCC_OK) {
+                return st;
+            }
+            if (trace) {
+                trace('C', f.frag, f.stage);
+            }
+            f.posted = false;
+            f.stage++;
+            moved = true;
+            if (f.stage == n_stages) {
+                done_++;
+                if (next_ < n_frags) {
+                    f.frag  = next_++;
+                    f.stage = 0;
+                } else {
+                    f.active = false;
+                }
+            }
+        }
+    }
+    return done_ == n_frags ? UCC_OK : UCC_INPROGRESS;
+}
+
 ucc_status_t Context::progress()
 {
     n_progress_calls++;

@@ -1855,7 +1855,8 @@ class GatedCollTask final : public Cdna4Task {
             out_b_ * n_ >= tt_->cfg_.ce_alltoall_min) {
             return enqueue_ce_a2a(comp_s);
         }
-        if (zc_ && zc_ready_ && nfrags_ > 1) {
+        if (zc_ && zc_ready_ && nfrags_ > 1 &&
+            Config::instance().get_bool("TL_CDNA4", "ZC_DEFRAG", true)) {
             /* zero-copy reads/writes USER buffers only — no staging
              * area is touched, so the chunk-size bound (the only reason
              * to fragment) does not apply: one fragment = one
