@@ -188,6 +188,8 @@ struct GatedArgs {
     int         nblocks;    /* gated grid size (team-lifetime constant,
                                same on every rank; <= kGatedMaxBlocks;
                                0 = kGatedBlocks default) */
+    int         pull_wait;  /* 1 = poll peers' counters remotely instead
+                               of local mirrors (debug fallback) */
     uint64_t    spin_limit; /* 0 = default kSpinLimit */
     /* per-dest cell staging (alltoall): my_in[c_dst_off[k]] <-
      * src[c_src_off[k]], c_len[k] bytes; 0 = contiguous stage of len   */

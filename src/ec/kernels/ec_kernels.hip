@@ -756,8 +756,11 @@ gated_wait(const GatedArgs &a, int phase, uint64_t target)
         int j = (int)threadIdx.x;
         if (j < a.nranks) {
             const uint64_t *f =
-                a.local_flags +
-                gated_mirror_idx(phase, a.slot, a.parity, j);
+                a.pull_wait
+                    ? a.peer_flags[j] +
+                          gated_idx(phase, a.slot, a.parity)
+                    : a.local_flags +
+                          gated_mirror_idx(phase, a.slot, a.parity, j);
             uint64_t spins = 0;
             const uint64_t cap = spin_cap(a.spin_limit);
             while (sys_load(f) < target) {

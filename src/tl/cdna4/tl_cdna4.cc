@@ -167,30 +167,58 @@ class Cdna4TlTeam final : public TlTeam {
      * zero-copy stay mapped until the team dies, so re-created
      * persistent colls on the same tensors (e.g. a rebuilt DDP bucket)
      * skip the hipIpcOpenMemHandle on every rank (reference mem_map /
-     * rcache role for the xGMI path). */
-    void *ipc_open_cached(const hipIpcMemHandle_t &h)
+     * rcache role for the xGMI path).
+     *
+     * Lifecycle (reference tl_cuda_cache.c:113-140 invalidate role):
+     * keyed by (exporter pid, exporter base VA). A hit re-validates the
+     * HANDLE BYTES — if the exporter freed and a new allocation landed
+     * on the same VA, the handle differs and the stale mapping is
+     * closed and re-opened. Conversely a NEW (pid,va) whose handle
+     * bytes match a cached entry at a DIFFERENT va means the driver
+     * recycled the handle identity after a free: the old entry is
+     * stale and must be closed BEFORE the open (an open of the same
+     * handle can return the old refcounted mapping). */
+    void *ipc_open_cached(int32_t pid, uint64_t remote_base,
+                          const hipIpcMemHandle_t &h)
     {
-        std::string k((const char *)&h, sizeof(h));
+        std::string hb((const char *)&h, sizeof(h));
+        auto        key = std::make_pair(pid, remote_base);
         std::lock_guard<std::mutex> g(ipc_mu_);
-        auto it = ipc_cache_.find(k);
+        auto it = ipc_cache_.find(key);
         if (it != ipc_cache_.end()) {
-            return it->second;
+            if (it->second.handle == hb) {
+                return it->second.mapped;
+            }
+            HIPWARN(hipIpcCloseMemHandle(it->second.mapped));
+            ipc_cache_.erase(it);
+        }
+        for (auto jt = ipc_cache_.begin(); jt != ipc_cache_.end();) {
+            if (jt->second.handle == hb) {
+                HIPWARN(hipIpcCloseMemHandle(jt->second.mapped));
+                jt = ipc_cache_.erase(jt);
+            } else {
+                ++jt;
+            }
         }
         void *m = nullptr;
         if (hipIpcOpenMemHandle(&m, h, hipIpcMemLazyEnablePeerAccess) !=
             hipSuccess) {
             return nullptr;
         }
-        ipc_cache_[k] = m;
+        ipc_cache_[key] = IpcEnt{hb, m};
         return m;
     }
-    std::map<std::string, void *> ipc_cache_;
+    struct IpcEnt {
+        std::string handle;
+        void       *mapped;
+    };
+    std::map<std::pair<int32_t, uint64_t>, IpcEnt> ipc_cache_;
     std::mutex                    ipc_mu_;
 
     ~Cdna4TlTeam() override
     {
         for (auto &kv : ipc_cache_) {
-            HIPWARN(hipIpcCloseMemHandle(kv.second));
+            HIPWARN(hipIpcCloseMemHandle(kv.second.mapped));
         }
         for (uint32_t r = 0; r < (uint32_t)peers_.size(); r++) {
             if (peers_[r].ipc_s) {
@@ -1666,14 +1694,17 @@ class GatedCollTask final : public Cdna4Task {
                     zc_peer_dst_[r] = (uint8_t *)(uintptr_t)b.d_raw_ptr;
                     continue;
                 }
-                void *m = tt_->ipc_open_cached(b.h), *md = nullptr;
+                void *m = tt_->ipc_open_cached(
+                    b.pid, b.raw_ptr - b.base_off, b.h);
+                void *md = nullptr;
                 if (!m) {
                     ucc_warn("zero-copy open failed, using staging");
                     zc_ = false;
                     break;
                 }
                 zc_peer_src_[r] = (const uint8_t *)m + b.base_off;
-                md              = tt_->ipc_open_cached(b.hd);
+                md              = tt_->ipc_open_cached(
+                    b.pid, b.d_raw_ptr - b.d_base_off, b.hd);
                 if (md) {
                     zc_peer_dst_[r] = (uint8_t *)md + b.d_base_off;
                 } else if (b.pad == 1) {
@@ -1790,6 +1821,8 @@ class GatedCollTask final : public Cdna4Task {
         ga.error_word  = tt_->err_host_;
         ga.spin_limit  = tt_->cfg_.spin_limit;
         ga.nblocks     = nblk;
+        ga.pull_wait =
+            !Config::instance().get_bool("TL_CDNA4", "PUSH", true);
         ga.rank        = (int)me_;
         ga.nranks      = (int)n_;
         ga.slot        = (int)slot_;
@@ -1888,6 +1921,8 @@ class GatedCollTask final : public Cdna4Task {
             ga.error_word  = tt_->err_host_;
             ga.spin_limit  = tt_->cfg_.spin_limit;
             ga.nblocks     = nblk;
+            ga.pull_wait   = !Config::instance().get_bool(
+                "TL_CDNA4", "PUSH", true);
             ga.len         = len;
             ga.rank        = (int)me_;
             ga.nranks      = (int)n_;
