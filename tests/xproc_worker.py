@@ -116,6 +116,28 @@ def main():
         del r2b
     results.append("oneshot_zc_xproc")
 
+    # 2c. exporter-side free + realloc (reference tl_cuda_cache.c
+    # invalidate-on-overlap role): empty_cache() releases the backing
+    # allocations, and the next allocations may land on the same VA or
+    # recycle the IPC handle identity — the team import cache must
+    # re-validate instead of serving a stale mapping (r02 regression:
+    # garbage peer blocks in the first zero-copy coll after realloc).
+    del r2, src, dst
+    torch.cuda.empty_cache()
+    count2 = 18_000_000
+    full2 = torch.randn(world, count2, generator=g0)
+    src = full2[rank].cuda()
+    dst = torch.zeros(count2, device="cuda")
+    r2c = c.coll_init(team, "allreduce", src=src.data_ptr(),
+                      dst=dst.data_ptr(), count=count2,
+                      dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA)
+    wait(r2c, ctx)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(dst.cpu(), full2.sum(0), rtol=1e-5,
+                               atol=1e-4)
+    del r2c
+    results.append("zc_realloc_xproc")
+
     # 3. cross-process alltoallv (skewed, fp16)
     scnt = [[(r + 1) * (d + 1) * 1024 for d in range(world)]
             for r in range(world)]
