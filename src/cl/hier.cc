@@ -1883,10 +1883,11 @@ void add_scores(Team *team)
             std::string tr = cfg.get("CL_HIER", "PIPELINE_TRACE", "");
             if (!tr.empty()) {
                 std::string path = tr;
-                pt->trace = [path](char ev, size_t f, size_t s) {
+                void       *id   = (void *)pt;
+                pt->trace = [path, id](char ev, size_t f, size_t s) {
                     FILE *fp = fopen(path.c_str(), "a");
                     if (fp) {
-                        fprintf(fp, "%c %zu %zu\n", ev, f, s);
+                        fprintf(fp, "%c %p %zu %zu\n", ev, id, f, s);
                         fclose(fp);
                     }
                 };

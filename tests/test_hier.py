@@ -263,3 +263,100 @@ def test_hier_nonuniform_nodes():
     sys.stdout.write(p.stdout[-2000:])
     sys.stderr.write(p.stderr[-3000:])
     assert p.returncode == 0 and "HIER_NU_OK" in p.stdout
+
+
+PIPELINE_WORKER = r"""
+import os
+import sys
+import numpy as np
+sys.path.insert(0, %r)
+from ucc_amd import core, dtypes
+from ucc_amd.testing import LocalJob
+
+n = 6
+job = LocalJob(n)
+c = core()
+
+rng = np.random.default_rng(31)
+count = 200_000  # 800 KB fp32, FRAG_SIZE=64k -> 13 fragments
+arrs = [(rng.random(count) - 0.5).astype(np.float32) for _ in range(n)]
+outs = job.allreduce_np(arrs)
+exp = np.sum(arrs, axis=0)
+for o in outs:
+    np.testing.assert_allclose(o, exp, rtol=1e-5, atol=1e-5)
+
+# AVG through the pipelined path (per-fragment post-scale)
+arrs = [np.full(count, float(r + 1), np.float32) for r in range(n)]
+outs = job.allreduce_np(arrs, op=dtypes.OP_AVG)
+expa = np.full(count, sum(range(1, n + 1)) / n, np.float32)
+for o in outs:
+    np.testing.assert_allclose(o, expa, rtol=1e-5, atol=1e-5)
+print("HIER_PIPE_OK")
+"""  % (REPO,)
+
+
+def test_hier_pipelined_allreduce_interleaves(tmp_path):
+    """fragments of a composed hier collective must
+    OVERLAP stages — stage-2 (node bcast) of fragment 0 must run before
+    stage-1 (leader allreduce) of the last fragment completes, per
+    pipeline task (reference ucc_schedule_pipelined.h:36-78 role)."""
+    trace = tmp_path / "pipe_trace.txt"
+    env = dict(os.environ)
+    env.update({
+        "UCC_FAKE_NODE_SPLIT": "2",
+        "UCC_CL_HIER_FRAG_SIZE": "65536",
+        "UCC_CL_HIER_PIPELINE_TRACE": str(trace),
+        # split_rail outranks RAB >=64KB; pin RAB (the pipelined alg)
+        "UCC_TUNE": "allreduce:@rab:99",
+    })
+    p = subprocess.run([sys.executable, "-c", PIPELINE_WORKER], env=env,
+                       capture_output=True, text=True, timeout=300)
+    sys.stdout.write(p.stdout[-2000:])
+    sys.stderr.write(p.stderr[-3000:])
+    assert p.returncode == 0 and "HIER_PIPE_OK" in p.stdout
+    assert trace.exists(), "pipeline trace file missing: pipelined path" \
+                           " was not taken"
+    # group events by task instance; each line: "P|C <taskid> frag stage"
+    per_task = {}
+    gen = {}
+    for ln in trace.read_text().splitlines():
+        ev, tid, f, s = ln.split()
+        f, s = int(f), int(s)
+        # task addresses are reused across collectives: a P(0,0) event
+        # opens a fresh stream for that address
+        if ev == "P" and f == 0 and s == 0:
+            gen[tid] = gen.get(tid, 0) + 1
+        key = (tid, gen.get(tid, 0))
+        per_task.setdefault(key, []).append((ev, f, s))
+    checked = 0
+    for tid, evs in per_task.items():
+        frags = {f for _, f, _ in evs}
+        if len(frags) < 3:
+            continue
+        last = max(frags)
+        # index of stage-2 post of frag 0 and stage-1 completion of the
+        # last frag within THIS task's event order
+        i_p02 = next(i for i, (ev, f, s) in enumerate(evs)
+                     if ev == "P" and f == 0 and s == 2)
+        i_cl1 = next(i for i, (ev, f, s) in enumerate(evs)
+                     if ev == "C" and f == last and s == 1)
+        assert i_p02 < i_cl1, (
+            f"no interleave in task {tid}: P(0,2)@{i_p02} "
+            f">= C({last},1)@{i_cl1}")
+        # pipeline depth respected: a frag is "open" from its first post
+        # until its own final-stage completion; never more than pdepth=2
+        # open at once
+        final_stage = {}
+        for ev, f, s in evs:
+            final_stage[f] = max(final_stage.get(f, 0), s)
+        open_frags = set()
+        max_open = 0
+        for ev, f, s in evs:
+            if ev == "P":
+                open_frags.add(f)
+            if ev == "C" and s == final_stage[f]:
+                open_frags.discard(f)
+            max_open = max(max_open, len(open_frags))
+        assert max_open <= 2, f"pdepth exceeded: {max_open}"
+        checked += 1
+    assert checked >= 1, "no multi-fragment pipeline task traced"
