@@ -860,6 +860,141 @@ class SplitRailAllreduceTask final : public Task {
     std::vector<uint64_t> cnt_, dsp_;
 };
 
+/* ---- split_rail over the fragment pipeline: the node-RSV of frag f+1
+ * overlaps the rail allreduce of frag f and the node-AGV of f-1, so the
+ * inter-node rail phase streams instead of idling the node links
+ * (reference cl_hier.h:47-57 split_rail pipeline config role). */
+class HierSplitRailPipeTask final : public PipelineTask {
+  public:
+    HierSplitRailPipeTask(Context *ctx, Team *team,
+                          const ucc_coll_args_t &args, size_t frag_elems,
+                          size_t depth)
+        : PipelineTask(ctx), team_(team), a_(args)
+    {
+        count_   = a_.dst.info.count;
+        dt_      = a_.dst.info.datatype;
+        dtsz_    = ucc_dt_size(dt_);
+        inplace_ = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        fc_      = frag_elems ? frag_elems : count_;
+        n_frags  = (count_ + fc_ - 1) / fc_;
+        n_stages = 3;
+        pdepth   = depth ? depth : 2;
+        stage_post = [this](size_t f, size_t s, ucc_coll_req_h *r) {
+            return do_stage(f, s, r);
+        };
+    }
+
+    ucc_status_t post() override
+    {
+        const uint32_t nn = (uint32_t)team_->node_ranks.size();
+        int            my_nidx = -1;
+        for (size_t i = 0; i < team_->node_ranks.size(); i++) {
+            if (team_->node_ranks[i] == team_->rank) {
+                my_nidx = (int)i;
+            }
+        }
+        if (my_nidx < 0 || !team_->rails_ok || !team_->rail_team) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        nidx_ = (uint32_t)my_nidx;
+        nn_   = nn;
+        /* per-fragment packed slice tables (pointers into these arrays
+         * must outlive the in-flight sub-collectives) */
+        fcnt_.assign(n_frags, {});
+        fdsp_.assign(n_frags, {});
+        for (size_t f = 0; f < n_frags; f++) {
+            uint64_t b   = f * fc_;
+            uint64_t cnt = count_ - b < fc_ ? count_ - b : fc_;
+            auto    &cv  = fcnt_[f];
+            auto    &dv  = fdsp_[f];
+            cv.resize(nn);
+            dv.resize(nn);
+            uint64_t per = cnt / nn, rem = cnt % nn, off = 0;
+            for (uint32_t r = 0; r < nn; r++) {
+                cv[r] = per + (r < rem ? 1 : 0);
+                dv[r] = off;
+                off += cv[r];
+            }
+        }
+        if (!inplace_) { /* all sub-steps run in-place on dst */
+            memcpy(a_.dst.info.buffer, a_.src.info.buffer,
+                   count_ * dtsz_);
+        }
+        return PipelineTask::post();
+    }
+
+  private:
+    ucc_status_t do_stage(size_t f, size_t s, ucc_coll_req_h *req)
+    {
+        *req          = nullptr;
+        uint64_t b    = f * fc_;
+        uint8_t *base = (uint8_t *)a_.dst.info.buffer + b * dtsz_;
+        uint64_t moff = fdsp_[f][nidx_], mcnt = fcnt_[f][nidx_];
+        ucc_coll_args_t sa{};
+        sa.mask  = UCC_COLL_ARGS_FIELD_FLAGS;
+        sa.flags = UCC_COLL_ARGS_FLAG_IN_PLACE |
+                   UCC_COLL_ARGS_FLAG_COUNT_64BIT |
+                   UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT;
+        sa.op    = a_.op == UCC_OP_AVG ? UCC_OP_SUM : a_.op;
+        Team *st_team = nullptr;
+        switch (s) {
+        case 0: /* node RSV of the fragment slice */
+            sa.coll_type           = UCC_COLL_TYPE_REDUCE_SCATTERV;
+            sa.dst.info_v.buffer   = base;
+            sa.dst.info_v.counts   = (ucc_count_t *)fcnt_[f].data();
+            sa.dst.info_v.datatype = dt_;
+            sa.dst.info_v.mem_type = a_.dst.info.mem_type;
+            st_team                = team_->node_team.get();
+            break;
+        case 1: /* rail allreduce of my sub-slice */
+            if (mcnt == 0) {
+                return UCC_OK;
+            }
+            sa.coll_type         = UCC_COLL_TYPE_ALLREDUCE;
+            sa.dst.info.buffer   = base + moff * dtsz_;
+            sa.dst.info.count    = mcnt;
+            sa.dst.info.datatype = dt_;
+            sa.dst.info.mem_type = a_.dst.info.mem_type;
+            sa.src.info          = sa.dst.info;
+            st_team              = team_->rail_team.get();
+            break;
+        case 2: { /* AVG scale my sub-slice, then node AGV */
+            if (a_.op == UCC_OP_AVG && mcnt) {
+                uint8_t    *p       = base + moff * dtsz_;
+                const void *srcs[1] = {p};
+                ec_cpu::reduce(p, srcs, 1, mcnt, dt_, UCC_OP_SUM,
+                               1.0 / (double)team_->size);
+            }
+            sa.coll_type                = UCC_COLL_TYPE_ALLGATHERV;
+            sa.dst.info_v.buffer        = base;
+            sa.dst.info_v.counts        = (ucc_count_t *)fcnt_[f].data();
+            sa.dst.info_v.displacements = (ucc_aint_t *)fdsp_[f].data();
+            sa.dst.info_v.datatype      = dt_;
+            sa.dst.info_v.mem_type      = a_.dst.info.mem_type;
+            st_team                     = team_->node_team.get();
+            break;
+        }
+        default:
+            return UCC_ERR_INVALID_PARAM;
+        }
+        ucc_status_t st = ucc_collective_init(
+            &sa, req, reinterpret_cast<ucc_team_h>(st_team));
+        if (st != UCC_OK) {
+            return st;
+        }
+        return ucc_collective_post(*req);
+    }
+
+    Team           *team_;
+    ucc_coll_args_t a_;
+    bool            inplace_ = false;
+    uint64_t        count_ = 0;
+    size_t          fc_ = 0, dtsz_ = 4;
+    uint32_t        nidx_ = 0, nn_ = 1;
+    ucc_datatype_t  dt_ = UCC_DT_FLOAT32;
+    std::vector<std::vector<uint64_t>> fcnt_, fdsp_;
+};
+
 static uint64_t v_cnt_at(const ucc_coll_args_t &a, const void *counts,
                          uint32_t r)
 {
@@ -1916,6 +2051,34 @@ void add_scores(Team *team)
                 args.op != UCC_OP_MIN && args.op != UCC_OP_PROD &&
                 args.op != UCC_OP_AVG) {
                 return UCC_ERR_NOT_SUPPORTED;
+            }
+            auto  &cfg = Config::instance();
+            size_t fb  = cfg.get_size("CL_HIER", "FRAG_SIZE",
+                                      4 * 1024 * 1024);
+            size_t dtsz = ucc_dt_size(args.dst.info.datatype);
+            size_t msg  = (size_t)args.dst.info.count * dtsz;
+            if (fb > 0 && dtsz > 0 && msg >= 2 * fb) {
+                size_t depth = (size_t)cfg.get_int(
+                    "CL_HIER", "PIPELINE_DEPTH", 2);
+                auto *pt = new HierSplitRailPipeTask(
+                    t->ctx, t, args, fb / dtsz, depth);
+                std::string tr =
+                    cfg.get("CL_HIER", "PIPELINE_TRACE", "");
+                if (!tr.empty()) {
+                    std::string path = tr;
+                    void       *id   = (void *)pt;
+                    pt->trace = [path, id](char ev, size_t f,
+                                           size_t s2) {
+                        FILE *fp = fopen(path.c_str(), "a");
+                        if (fp) {
+                            fprintf(fp, "%c %p %zu %zu\n", ev, id, f,
+                                    s2);
+                            fclose(fp);
+                        }
+                    };
+                }
+                *task = pt;
+                return UCC_OK;
             }
             *task = new SplitRailAllreduceTask(t->ctx, t, args);
             return UCC_OK;

@@ -295,19 +295,23 @@ print("HIER_PIPE_OK")
 """  % (REPO,)
 
 
-def test_hier_pipelined_allreduce_interleaves(tmp_path):
+import pytest
+
+
+@pytest.mark.parametrize("alg", ["rab", "split_rail"])
+def test_hier_pipelined_allreduce_interleaves(tmp_path, alg):
     """fragments of a composed hier collective must
-    OVERLAP stages — stage-2 (node bcast) of fragment 0 must run before
-    stage-1 (leader allreduce) of the last fragment completes, per
-    pipeline task (reference ucc_schedule_pipelined.h:36-78 role)."""
+    OVERLAP stages — stage-2 (node bcast / node allgatherv) of fragment
+    0 must run before stage-1 (inter-node phase) of the last fragment
+    completes, per pipeline task (reference
+    ucc_schedule_pipelined.h:36-78 role)."""
     trace = tmp_path / "pipe_trace.txt"
     env = dict(os.environ)
     env.update({
         "UCC_FAKE_NODE_SPLIT": "2",
         "UCC_CL_HIER_FRAG_SIZE": "65536",
         "UCC_CL_HIER_PIPELINE_TRACE": str(trace),
-        # split_rail outranks RAB >=64KB; pin RAB (the pipelined alg)
-        "UCC_TUNE": "allreduce:@rab:99",
+        "UCC_TUNE": f"allreduce:@{alg}:99",
     })
     p = subprocess.run([sys.executable, "-c", PIPELINE_WORKER], env=env,
                        capture_output=True, text=True, timeout=300)
