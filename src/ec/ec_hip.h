@@ -139,6 +139,11 @@ constexpr int kGatedGraphBase = 1024;
  *         + src_rank -> 9216..9599 */
 constexpr int kGatedMirrorBase = 9216;
 
+/* default device spin bound (iterations of the bounded wait loops);
+ * host-side team setup scales it by the xGMI hop count of the farthest
+ * peer before passing it down via GatedArgs/FusedArgs.spin_limit */
+constexpr uint64_t kDefaultSpinLimit = 200u * 1000u * 1000u;
+
 struct GatedArgs {
     /* data */
     const void *src;      /* stage: user src frag; others unused        */

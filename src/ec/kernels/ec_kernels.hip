@@ -430,7 +430,7 @@ __device__ __forceinline__ void sys_store(uint64_t *p, uint64_t v)
     __hip_atomic_store(p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
 }
 
-constexpr uint64_t kSpinLimit = 200u * 1000u * 1000u;
+constexpr uint64_t kSpinLimit = kDefaultSpinLimit;
 __device__ __forceinline__ uint64_t spin_cap(uint64_t v)
 {
     return v ? v : kSpinLimit;

@@ -30,6 +30,7 @@
 
 #include "../../core/core.h"
 #include "../../ec/ec_hip.h"
+#include "../../topo/topo.h"
 #include "../../mc/mc.h"
 #include "../shm/slot_seg.h"
 
@@ -290,6 +291,40 @@ class Cdna4TlTeam final : public TlTeam {
         const uint32_t n  = team_->size;
         const int32_t  me = (int32_t)getpid();
         peers_.resize(n);
+        dev_map_.resize(n);
+        /* Topology gate (reference tl_cuda_team_topo.c:56-100 role):
+         * every pair must be xGMI/peer-access reachable or same-device,
+         * or this TL refuses the team and the score fallback chain
+         * (rccl, shm staging) takes over instead of hanging on an
+         * unreachable IPC read. Spin bounds scale with the worst hop
+         * count so remote-peer latency is not misread as a timeout. */
+        {
+            int mydev = ((Cdna4TlContext *)tlc_)->dev_;
+            const topo::GpuLinks &gl = topo::gpu_links();
+            max_hops_ = 1;
+            for (uint32_t r = 0; r < n; r++) {
+                ExchgBlob b;
+                memcpy(&b, (const uint8_t *)all + r * stride, sizeof(b));
+                dev_map_[r] = b.device;
+                if (b.device == mydev || r == team_->rank) {
+                    continue;
+                }
+                if (b.device < gl.ndev && mydev < gl.ndev &&
+                    b.device >= 0) {
+                    if (!gl.peer[mydev][b.device]) {
+                        ucc_warn("cdna4: no peer access dev%d->dev%d "
+                                 "(rank %u); dropping cdna4 team "
+                                 "(fallback TLs take over)",
+                                 mydev, b.device, r);
+                        return UCC_ERR_NOT_SUPPORTED;
+                    }
+                    int h = gl.hops[mydev][b.device];
+                    if (h > (int)max_hops_) {
+                        max_hops_ = (uint32_t)h;
+                    }
+                }
+            }
+        }
         for (uint32_t r = 0; r < n; r++) {
             ExchgBlob b;
             memcpy(&b, (const uint8_t *)all + r * stride, sizeof(b));
@@ -397,6 +432,19 @@ class Cdna4TlTeam final : public TlTeam {
      * (phase, slot, parity) — identical on every rank because
      * collectives post in the same order (targets = count x 32 blocks) */
     uint64_t gated_launch_[3][ec_hip::kGatedSlots][2] = {};
+    std::vector<int> dev_map_;  /* rank -> hip device (topology gate) */
+    uint32_t         max_hops_ = 1;
+    /* spin bound scaled by the farthest peer's hop count (user
+     * override wins) */
+    uint64_t spin_limit() const
+    {
+        if (cfg_.spin_limit) {
+            return cfg_.spin_limit;
+        }
+        return max_hops_ > 1
+                   ? ec_hip::kDefaultSpinLimit * (uint64_t)max_hops_
+                   : 0; /* 0 = kernel default */
+    }
     uint64_t    seq_ = 0;
     std::vector<uint64_t> stage_cum_; /* per-slot fused block arrivals */
     uint8_t    *scratch_ = nullptr;
@@ -522,7 +570,7 @@ class FusedAllreduceTask final : public Cdna4Task {
             fa.slot        = (int)slot_;
             fa.seq         = fseq_;
             fa.error_word  = tt_->err_host_;
-            fa.spin_limit  = tt_->cfg_.spin_limit;
+            fa.spin_limit  = tt_->spin_limit();
             size_t blocks  = (bytes_ + 128 * 1024 - 1) / (128 * 1024);
             fa.nblocks     = (int)(blocks < 1 ? 1 : blocks > 16 ? 16 : blocks);
             if (tt_->stage_cum_.size() < tt_->cfg_.nslots) {
@@ -1819,7 +1867,7 @@ class GatedCollTask final : public Cdna4Task {
         ec_hip::GatedArgs ga{};
         ga.local_flags = tt_->flags_;
         ga.error_word  = tt_->err_host_;
-        ga.spin_limit  = tt_->cfg_.spin_limit;
+        ga.spin_limit  = tt_->spin_limit();
         ga.nblocks     = nblk;
         ga.pull_wait =
             !Config::instance().get_bool("TL_CDNA4", "PUSH", true);
@@ -1919,7 +1967,7 @@ class GatedCollTask final : public Cdna4Task {
             }
             ga.local_flags = tt_->flags_;
             ga.error_word  = tt_->err_host_;
-            ga.spin_limit  = tt_->cfg_.spin_limit;
+            ga.spin_limit  = tt_->spin_limit();
             ga.nblocks     = nblk;
             ga.pull_wait   = !Config::instance().get_bool(
                 "TL_CDNA4", "PUSH", true);
