@@ -174,6 +174,13 @@ struct GatedArgs {
      *                (1=reduce for allreduce, 0=stage for ag/alltoall) */
     uint64_t    t_sw_reduce, t_sw_gather, t_prev_gather, t_stage,
                 t_gather_wait;
+    /* post-launch cumulative counter values for THIS launch's signal
+     * phase (stage=0/reduce=1/gather=2): the block whose increment
+     * reaches the value is the last arriver and pushes the mirrors.
+     * Counters accumulate BLOCK counts, so collectives with different
+     * grid sizes compose on the same (slot,parity). Derive mode
+     * computes these on device as u*B instead. */
+    uint64_t    t_sig_stage, t_sig_reduce, t_sig_gather;
     int         gw_phase;
     /* Graph-replayable mode (persistent colls on a dedicated slot): when
      * derive != 0, targets are computed ON DEVICE from per-block launch

@@ -710,13 +710,15 @@ __device__ __forceinline__ int gated_mirror_idx(int phase, int slot,
  * scratch reloads inside the hot loops (observed in gfx950 ISA). */
 struct GatedTargets {
     uint64_t sw_reduce, sw_gather, prev_gather, stage, gather_wait;
+    uint64_t sig_stage, sig_reduce, sig_gather;
 };
 
 __device__ __forceinline__ GatedTargets gated_targets(const GatedArgs &a)
 {
     if (!a.derive) {
-        return {a.t_sw_reduce, a.t_sw_gather, a.t_prev_gather, a.t_stage,
-                a.t_gather_wait};
+        return {a.t_sw_reduce,  a.t_sw_gather,  a.t_prev_gather,
+                a.t_stage,      a.t_gather_wait,
+                a.t_sig_stage,  a.t_sig_reduce, a.t_sig_gather};
     }
     uint64_t *cnt = a.local_flags + kGatedGraphBase +
                     ((uint64_t)a.slot * 2 + a.parity) * kGatedMaxBlocks +
@@ -733,7 +735,8 @@ __device__ __forceinline__ GatedTargets gated_targets(const GatedArgs &a)
     const uint64_t B = (uint64_t)(a.nblocks ? a.nblocks : kGatedBlocks);
     return {a.has_reduce ? (u - 1) * B : 0,
             a.has_gather ? (u - 1) * B : 0,
-            a.has_gather ? (u - 1) * B : 0, u * B, u * B};
+            a.has_gather ? (u - 1) * B : 0, u * B, u * B,
+            u * B, u * B, u * B};
 }
 
 /* all blocks wait until every rank reached counter[idx] >= target; on
@@ -787,20 +790,19 @@ gated_wait(const GatedArgs &a, int phase, uint64_t target)
  * traffic in the whole pipeline). acq_rel on the counter makes every
  * earlier block's release (and thus its payload writes, xGMI ones
  * included) visible-before the mirror store. */
-__device__ __forceinline__ void gated_signal(const GatedArgs &a, int phase)
+__device__ __forceinline__ void gated_signal(const GatedArgs &a, int phase,
+                                             uint64_t sig_target)
 {
     __threadfence_system();
     __syncthreads();
     if (threadIdx.x == 0) {
-        const uint64_t B = (uint64_t)(a.nblocks ? a.nblocks
-                                                : kGatedBlocks);
         uint64_t v = __hip_atomic_fetch_add(
                          a.local_flags +
                              gated_idx(phase, a.slot, a.parity),
                          1, __ATOMIC_ACQ_REL,
                          __HIP_MEMORY_SCOPE_SYSTEM) +
                      1;
-        if (v % B == 0) { /* last arriver of this launch */
+        if (v == sig_target) { /* last arriver of this launch */
             /* guard against the ROCm 7.2 dropped-vmcnt hazard
              * (MI355X_MICROARCH.md "Compiler hazard"): the release
              * write-back must drain before the mirror stores leave */
@@ -834,7 +836,7 @@ __global__ void k_staged_stage(const GatedArgs a)
     } else if (a.len) {
         d_copy_bytes((uint8_t *)a.my_in, (const uint8_t *)a.src, a.len);
     }
-    gated_signal(a, 0);
+    gated_signal(a, 0, t.sig_stage);
 }
 
 template <typename T, int OP, int VEC>
@@ -970,7 +972,7 @@ __global__ void k_staged_reduce(const GatedArgs a)
             ((T *)out)[t] = ov;
         }
     }
-    gated_signal(a, 1);
+    gated_signal(a, 1, gt.sig_reduce);
 }
 
 __global__ void k_staged_gather(const GatedArgs a)
@@ -992,7 +994,7 @@ __global__ void k_staged_gather(const GatedArgs a)
                               (uint64_t)bid, (uint64_t)grp);
         }
     }
-    gated_signal(a, 2);
+    gated_signal(a, 2, t.sig_gather);
 }
 
 static inline int gated_grid(const GatedArgs &a)
@@ -1012,7 +1014,7 @@ __global__ void k_gated_wait_only(const GatedArgs a)
 
 __global__ void k_gated_done(const GatedArgs a)
 {
-    gated_signal(a, 2);
+    gated_signal(a, 2, a.t_sig_gather);
     (void)gated_wait(a, 2, a.t_gather_wait);
 }
 

@@ -1880,16 +1880,18 @@ class GatedCollTask final : public Cdna4Task {
         }
         ga.len     = 0; /* entry stage is a pure signal */
         ga.n_cells = 0;
-        ga.t_sw_reduce = L[1][slot_][p] * B;
-        ga.t_sw_gather = L[2][slot_][p] * B;
+        ga.t_sw_reduce  = L[1][slot_][p];
+        ga.t_sw_gather  = L[2][slot_][p];
+        ga.t_sig_stage  = L[0][slot_][p] + B;
+        ga.t_sig_gather = L[2][slot_][p] + B;
         ucc_status_t st = ec_hip::staged_stage(ga, comp_s);
-        L[0][slot_][p]++;
+        L[0][slot_][p] += B;
         if (st != UCC_OK) {
             return st;
         }
         /* no peer buffer is written until ITS owner posted (entered) */
         ga.gw_phase      = 0;
-        ga.t_gather_wait = L[0][slot_][p] * B;
+        ga.t_gather_wait = L[0][slot_][p];
         st = ec_hip::gated_wait_only(ga, comp_s);
         if (st != UCC_OK) {
             return st;
@@ -1922,9 +1924,9 @@ class GatedCollTask final : public Cdna4Task {
         }
         /* my pulls done -> publish; complete when every rank's pulls
          * (including reads of MY src) are done */
-        ga.t_gather_wait = (L[2][slot_][p] + 1) * B;
+        ga.t_gather_wait = L[2][slot_][p] + B;
         st = ec_hip::gated_done(ga, comp_s);
-        L[2][slot_][p]++;
+        L[2][slot_][p] += B;
         return st;
     }
 
@@ -1948,9 +1950,17 @@ class GatedCollTask final : public Cdna4Task {
             nfrags_ = 1;
         }
         auto &L          = tt_->gated_launch_;
-        const int nblk   = tt_->cfg_.gated_blocks
-                               ? tt_->cfg_.gated_blocks
-                               : ec_hip::kGatedBlocks;
+        /* grid size: measured crossover on the 2-proc rig
+         * (profiles/rocprof_kernels_r02.md): 64 blocks win below
+         * ~160 MiB, 96 above (single-frag zc reduce fills more CUs on
+         * big slices). The launch ledger accumulates BLOCK counts, so
+         * per-collective grids compose on shared (slot,parity)
+         * counters. */
+        const int nblk =
+            tt_->cfg_.gated_blocks
+                ? tt_->cfg_.gated_blocks
+                : (nfrags_ == 1 && total_ >= 160u * 1024 * 1024 ? 96
+                                                  : ec_hip::kGatedBlocks);
         const uint64_t B = (uint64_t)nblk;
         for (size_t f = 0; f < nfrags_; f++) {
             const uint32_t p   = (uint32_t)(f & 1);
@@ -2014,17 +2024,20 @@ class GatedCollTask final : public Cdna4Task {
                         ga.peer_out[r] = zc_peer_dst_[r] + off + ga.sl_b;
                     }
                 }
-                ga.t_sw_reduce   = L[1][slot_][p] * B;
-                ga.t_sw_gather   = L[2][slot_][p] * B;
-                ga.t_prev_gather = L[2][slot_][p] * B;
-                ga.t_stage       = (L[0][slot_][p] + 1) * B;
+                ga.t_sw_reduce   = L[1][slot_][p];
+                ga.t_sw_gather   = L[2][slot_][p];
+                ga.t_prev_gather = L[2][slot_][p];
+                ga.t_stage       = L[0][slot_][p] + B;
                 ga.gw_phase      = 1;
-                ga.t_gather_wait = (L[1][slot_][p] + 1) * B;
+                ga.t_gather_wait = L[1][slot_][p] + B;
+                ga.t_sig_stage   = L[0][slot_][p] + B;
+                ga.t_sig_reduce  = L[1][slot_][p] + B;
+                ga.t_sig_gather  = L[2][slot_][p] + B;
                 st = ec_hip::staged_stage(ga, stage_s);
-                if (!derive) { L[0][slot_][p]++; }
+                if (!derive) { L[0][slot_][p] += B; }
                 if (st == UCC_OK) {
                     st = ec_hip::staged_reduce(ga, comp_s);
-                    if (!derive) { L[1][slot_][p]++; }
+                    if (!derive) { L[1][slot_][p] += B; }
                 }
                 if (st == UCC_OK) {
                     if (ga.zc_write) {
@@ -2035,7 +2048,7 @@ class GatedCollTask final : public Cdna4Task {
                         }
                     }
                     st = ec_hip::staged_gather(ga, comp_s);
-                    if (!derive) { L[2][slot_][p]++; }
+                    if (!derive) { L[2][slot_][p] += B; }
                 }
                 break;
             }
@@ -2065,15 +2078,17 @@ class GatedCollTask final : public Cdna4Task {
                         ga.peer_in[r] = zc_peer_src_[r] + off;
                     }
                 }
-                ga.t_sw_reduce   = L[1][slot_][p] * B;
-                ga.t_sw_gather   = L[2][slot_][p] * B;
+                ga.t_sw_reduce   = L[1][slot_][p];
+                ga.t_sw_gather   = L[2][slot_][p];
                 ga.t_prev_gather = 0;
-                ga.t_stage       = (L[0][slot_][p] + 1) * B;
+                ga.t_stage       = L[0][slot_][p] + B;
+                ga.t_sig_stage   = L[0][slot_][p] + B;
+                ga.t_sig_reduce  = L[1][slot_][p] + B;
                 st = ec_hip::staged_stage(ga, stage_s);
-                if (!derive) { L[0][slot_][p]++; }
+                if (!derive) { L[0][slot_][p] += B; }
                 if (st == UCC_OK) {
                     st = ec_hip::staged_reduce(ga, comp_s);
-                    if (!derive) { L[1][slot_][p]++; }
+                    if (!derive) { L[1][slot_][p] += B; }
                 }
                 break;
             }
@@ -2103,15 +2118,17 @@ class GatedCollTask final : public Cdna4Task {
                     ga.slice_b[r] = dsp_[r] + off;
                     ga.slice_e[r] = ga.slice_b[r] + l;
                 }
-                ga.t_sw_reduce   = L[1][slot_][p] * B;
-                ga.t_sw_gather   = L[2][slot_][p] * B;
+                ga.t_sw_reduce   = L[1][slot_][p];
+                ga.t_sw_gather   = L[2][slot_][p];
                 ga.gw_phase      = 0;
-                ga.t_gather_wait = (L[0][slot_][p] + 1) * B;
+                ga.t_gather_wait = L[0][slot_][p] + B;
+                ga.t_sig_stage   = L[0][slot_][p] + B;
+                ga.t_sig_gather  = L[2][slot_][p] + B;
                 st = ec_hip::staged_stage(ga, stage_s);
-                if (!derive) { L[0][slot_][p]++; }
+                if (!derive) { L[0][slot_][p] += B; }
                 if (st == UCC_OK) {
                     st = ec_hip::staged_gather(ga, comp_s);
-                    if (!derive) { L[2][slot_][p]++; }
+                    if (!derive) { L[2][slot_][p] += B; }
                 }
                 break;
             }
@@ -2140,15 +2157,17 @@ class GatedCollTask final : public Cdna4Task {
                     ga.slice_b[r] = rdsp_[r] + off;
                     ga.slice_e[r] = ga.slice_b[r] + rl;
                 }
-                ga.t_sw_reduce   = L[1][slot_][p] * B;
-                ga.t_sw_gather   = L[2][slot_][p] * B;
+                ga.t_sw_reduce   = L[1][slot_][p];
+                ga.t_sw_gather   = L[2][slot_][p];
                 ga.gw_phase      = 0;
-                ga.t_gather_wait = (L[0][slot_][p] + 1) * B;
+                ga.t_gather_wait = L[0][slot_][p] + B;
+                ga.t_sig_stage   = L[0][slot_][p] + B;
+                ga.t_sig_gather  = L[2][slot_][p] + B;
                 st = ec_hip::staged_stage(ga, stage_s);
-                if (!derive) { L[0][slot_][p]++; }
+                if (!derive) { L[0][slot_][p] += B; }
                 if (st == UCC_OK) {
                     st = ec_hip::staged_gather(ga, comp_s);
-                    if (!derive) { L[2][slot_][p]++; }
+                    if (!derive) { L[2][slot_][p] += B; }
                 }
                 break;
             }
@@ -2175,15 +2194,17 @@ class GatedCollTask final : public Cdna4Task {
                     ga.slice_b[r] = (uint64_t)r * out_b_ + off;
                     ga.slice_e[r] = ga.slice_b[r] + len;
                 }
-                ga.t_sw_reduce   = L[1][slot_][p] * B;
-                ga.t_sw_gather   = L[2][slot_][p] * B;
+                ga.t_sw_reduce   = L[1][slot_][p];
+                ga.t_sw_gather   = L[2][slot_][p];
                 ga.gw_phase      = 0;
-                ga.t_gather_wait = (L[0][slot_][p] + 1) * B;
+                ga.t_gather_wait = L[0][slot_][p] + B;
+                ga.t_sig_stage   = L[0][slot_][p] + B;
+                ga.t_sig_gather  = L[2][slot_][p] + B;
                 st = ec_hip::staged_stage(ga, stage_s);
-                if (!derive) { L[0][slot_][p]++; }
+                if (!derive) { L[0][slot_][p] += B; }
                 if (st == UCC_OK) {
                     st = ec_hip::staged_gather(ga, comp_s);
-                    if (!derive) { L[2][slot_][p]++; }
+                    if (!derive) { L[2][slot_][p] += B; }
                 }
                 break;
             }
@@ -2256,7 +2277,7 @@ class Cdna4Tl final : public Tl {
         cfg.declare("TL_CDNA4", "CE_ALLTOALL", "1",
                     "move zero-copy alltoall data on SDMA copy engines "
                     "(hipMemcpyAsync) instead of gather kernels");
-        cfg.declare("TL_CDNA4", "CE_ALLTOALL_MIN", "64m",
+        cfg.declare("TL_CDNA4", "CE_ALLTOALL_MIN", "192m",
                     "min total message bytes for the SDMA alltoall "
                     "(measured crossover vs the gather kernel, "
                     "profiles/rocprof_kernels_r02.md)");
@@ -2299,7 +2320,8 @@ class Cdna4Tl final : public Tl {
         }
         c.ce_alltoall = cfg.get_bool("TL_CDNA4", "CE_ALLTOALL", true);
         c.ce_alltoall_min =
-            cfg.get_size("TL_CDNA4", "CE_ALLTOALL_MIN", 64 * 1024 * 1024);
+            cfg.get_size("TL_CDNA4", "CE_ALLTOALL_MIN",
+                         192 * 1024 * 1024);
         return new Cdna4TlTeam(tlc, team, c);
     }
 };
