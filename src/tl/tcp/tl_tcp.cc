@@ -1013,6 +1013,443 @@ class TcpBcastSagTask final : public TcpTask {
     uint32_t root_  = 0;
 };
 
+/* ---- bcast: recursive k-nomial tree with a radix knob
+ * (reference coll_patterns/recursive_knomial.h:13-17 role, re-derived).
+ * Virtual rank vr = (me - root) mod n; vr's parent clears its lowest
+ * nonzero base-k digit; after receiving, vr sends to children
+ * vr + i*q for every digit level q below its own, i = 1..k-1. Radix
+ * k=2 degenerates to the binomial tree; larger k trades tree depth
+ * (latency) for root fan-out. UCC_TL_TCP_KN_RADIX. */
+class TcpBcastKnomialTask final : public TcpTask {
+  public:
+    TcpBcastKnomialTask(Context *ctx, TcpTlTeam *tt,
+                        const ucc_coll_args_t &args, uint32_t radix)
+        : TcpTask(ctx, tt, args), k_(radix < 2 ? 2 : radix)
+    {
+    }
+
+    ucc_status_t post() override
+    {
+        begin();
+        bytes_ = a_.src.info.count * ucc_dt_size(a_.src.info.datatype);
+        buf_   = (uint8_t *)a_.src.info.buffer;
+        vr_    = (me_ + n_ - (uint32_t)a_.root) % n_;
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    uint32_t to_team(uint32_t v) const
+    {
+        return (v + (uint32_t)a_.root) % n_;
+    }
+
+    ucc_status_t progress_()
+    {
+        if (phase_ == 0) {
+            /* my receive level = lowest nonzero base-k digit of vr */
+            plevel_ = 1;
+            if (vr_ != 0) {
+                uint64_t q = 1;
+                while ((vr_ / q) % k_ == 0) {
+                    q *= k_;
+                }
+                plevel_ = q * k_; /* children live strictly below q*k */
+                uint32_t digit  = (uint32_t)((vr_ / q) % k_);
+                uint32_t parent = vr_ - digit * (uint32_t)q;
+                recv_from(to_team(parent), 0, buf_, bytes_);
+                plevel_ = q; /* I fan out at levels < q */
+            } else {
+                uint64_t q = 1;
+                while (q < n_) {
+                    q *= k_;
+                }
+                plevel_ = q;
+            }
+            phase_ = 1;
+        }
+        if (phase_ == 1) { /* wait for the parent data */
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            /* fan out: highest level first (big subtrees start early) */
+            for (uint64_t q = plevel_ / k_; q >= 1; q /= k_) {
+                for (uint32_t i = 1; i < k_; i++) {
+                    uint64_t child = vr_ + (uint64_t)i * q;
+                    if (child < n_) {
+                        send_to(to_team((uint32_t)child), 0, buf_,
+                                bytes_);
+                    }
+                }
+                if (q == 1) {
+                    break;
+                }
+            }
+            phase_ = 2;
+        }
+        if (!ops_done()) {
+            return UCC_INPROGRESS;
+        }
+        clear_ops();
+        return UCC_OK;
+    }
+
+    uint32_t k_ = 2, vr_ = 0;
+    uint64_t plevel_ = 1;
+    size_t   bytes_ = 0;
+    uint8_t *buf_   = nullptr;
+};
+
+/* ---- double binary tree (DBT) bcast/reduce
+ * (reference coll_patterns/double_binary_tree.h:51-224 role,
+ * re-derived as the two-tree scheme): the message splits in halves;
+ * half A flows through an in-order balanced BST over virtual ranks,
+ * half B through the same tree shifted by one position, so (for even
+ * n) a rank that is internal in one tree is a leaf in the other and
+ * per-rank send load stays ~1x the message across both trees —
+ * latency*bandwidth-balanced for medium messages. The collective root
+ * bridges to/from each tree's BST root with one extra hop when they
+ * differ. */
+struct DbtNode {
+    int parent = -1, left = -1, right = -1;
+};
+
+/* node of the in-order balanced BST over positions [0, n) */
+static inline DbtNode dbt_node(uint32_t n, uint32_t v)
+{
+    DbtNode  r;
+    uint32_t lo = 0, hi = n;
+    while (true) {
+        uint32_t mid = lo + (hi - lo) / 2;
+        if (v == mid) {
+            if (mid > lo) {
+                r.left = (int)(lo + (mid - lo) / 2);
+            }
+            if (mid + 1 < hi) {
+                r.right = (int)(mid + 1 + (hi - mid - 1) / 2);
+            }
+            return r;
+        }
+        r.parent = (int)mid;
+        if (v < mid) {
+            hi = mid;
+        } else {
+            lo = mid + 1;
+        }
+    }
+}
+
+static inline uint32_t dbt_root_pos(uint32_t n) { return n / 2; }
+
+class TcpBcastDbtTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        bytes_ = a_.src.info.count * ucc_dt_size(a_.src.info.datatype);
+        buf_   = (uint8_t *)a_.src.info.buffer;
+        halfa_ = bytes_ / 2;
+        if (n_ < 2 || halfa_ == 0) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        /* positions: p1 maps the collective root onto the BST root */
+        const uint32_t M = dbt_root_pos(n_);
+        p1_ = (me_ + n_ - (uint32_t)a_.root + M) % n_;
+        p2_ = (p1_ + 1) % n_;
+        t1_ = dbt_node(n_, p1_);
+        t2_ = dbt_node(n_, p2_);
+        /* rank holding tree-2's BST root */
+        r2_ = from_p1((M + n_ - 1) % n_);
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    /* team rank from a p1 position */
+    uint32_t from_p1(uint32_t p) const
+    {
+        const uint32_t M = dbt_root_pos(n_);
+        return (p + n_ - M + (uint32_t)a_.root) % n_;
+    }
+    uint32_t t1_rank(int pos) const { return from_p1((uint32_t)pos); }
+    uint32_t t2_rank(int pos) const
+    {
+        /* p2 = (p1+1) mod n  =>  p1 = (p2-1) mod n */
+        return from_p1(((uint32_t)pos + n_ - 1) % n_);
+    }
+
+    ucc_status_t progress_()
+    {
+        const bool     is_root = me_ == (uint32_t)a_.root;
+        const uint32_t M       = dbt_root_pos(n_);
+        uint8_t       *bufb    = buf_ + halfa_;
+        const size_t   lenb    = bytes_ - halfa_;
+        if (phase_ == 0) {
+            /* half A through tree 1 (root IS tree-1's BST root) */
+            if (p1_ != M) {
+                recv_from(t1_rank(t1_.parent), 1, buf_, halfa_);
+            }
+            /* half B through tree 2; bridge root -> r2 if distinct */
+            if (is_root && r2_ != me_) {
+                send_to(r2_, 2, bufb, lenb);
+            }
+            if (p2_ == M) { /* tree-2 BST root */
+                if (!is_root) {
+                    recv_from((uint32_t)a_.root, 2, bufb, lenb);
+                }
+            } else if (is_root) {
+                /* the root already has half B, but its T2 parent still
+                 * sends one copy — drain it into scratch so the
+                 * connection's tag space stays clean */
+                discard_.resize(lenb);
+                recv_from(t2_rank(t2_.parent), 3, discard_.data(),
+                          lenb);
+            } else {
+                recv_from(t2_rank(t2_.parent), 3, bufb, lenb);
+            }
+            phase_   = 1;
+            senta_   = false;
+            sentb_   = false;
+            have_a_  = is_root; /* root == T1's BST root by mapping */
+            have_b_  = is_root;
+        }
+        /* forward each half as soon as it lands (the two trees make
+         * independent progress — that is the DBT bandwidth trick) */
+        tt_->progress();
+        if (phase_ == 1) {
+            if (!senta_ && (have_a_ || half_done(0))) {
+                if (t1_.left >= 0) {
+                    send_to(t1_rank(t1_.left), 1, buf_, halfa_);
+                }
+                if (t1_.right >= 0) {
+                    send_to(t1_rank(t1_.right), 1, buf_, halfa_);
+                }
+                senta_ = true;
+            }
+            bool b_in = have_b_ || half_done(1);
+            if (!sentb_ && p2_ == M && !is_root) {
+                /* tree-2 root got the bridge copy */
+                b_in = b_in || bridge_done();
+            }
+            if (!sentb_ && b_in) {
+                if (t2_.left >= 0) {
+                    send_to(t2_rank(t2_.left), 3, bufb, lenb);
+                }
+                if (t2_.right >= 0) {
+                    send_to(t2_rank(t2_.right), 3, bufb, lenb);
+                }
+                sentb_ = true;
+            }
+            if (senta_ && sentb_ && ops_done()) {
+                clear_ops();
+                return UCC_OK;
+            }
+            return UCC_INPROGRESS;
+        }
+        return UCC_INPROGRESS;
+    }
+
+    /* recv completion helpers: recvs_ were posted in phase 0 in a
+     * known order; scan for the op with the matching buffer */
+    bool half_done(int which) const
+    {
+        const uint8_t *b = which == 0 ? buf_ : buf_ + halfa_;
+        for (auto *r : recvs_) {
+            if (r->buf == b && r->done) {
+                return true;
+            }
+        }
+        return false;
+    }
+    bool bridge_done() const { return half_done(1); }
+
+    size_t               bytes_ = 0, halfa_ = 0;
+    uint8_t             *buf_   = nullptr;
+    uint32_t             p1_ = 0, p2_ = 0, r2_ = 0;
+    DbtNode              t1_, t2_;
+    std::vector<uint8_t> discard_;
+    bool                 senta_ = false, sentb_ = false;
+    bool                 have_a_ = false, have_b_ = false;
+};
+
+/* ---- DBT reduce: the bcast flow reversed — children's partial sums
+ * of each half arrive, are combined with my contribution, and flow to
+ * the tree parent; tree roots bridge to the collective root. */
+class TcpReduceDbtTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        dt_    = a_.src.info.datatype;
+        op_    = a_.op == UCC_OP_AVG ? UCC_OP_SUM : a_.op;
+        dtsz_  = ucc_dt_size(dt_);
+        count_ = a_.src.info.count;
+        bytes_ = count_ * dtsz_;
+        cnta_  = count_ / 2;
+        if (n_ < 2 || cnta_ == 0 || count_ == cnta_) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        const bool is_root = me_ == (uint32_t)a_.root;
+        /* working copy of my contribution */
+        work_.resize(bytes_);
+        memcpy(work_.data(),
+               is_root && inplace ? a_.dst.info.buffer
+                                  : a_.src.info.buffer,
+               bytes_);
+        const uint32_t M = dbt_root_pos(n_);
+        p1_ = (me_ + n_ - (uint32_t)a_.root + M) % n_;
+        p2_ = (p1_ + 1) % n_;
+        t1_ = dbt_node(n_, p1_);
+        t2_ = dbt_node(n_, p2_);
+        r2_ = from_p1((M + n_ - 1) % n_);
+        /* post child recvs for both halves */
+        na_ = nb_ = 0;
+        if (t1_.left >= 0) {
+            child_recv(0, t1_rank(t1_.left), 1, true);
+        }
+        if (t1_.right >= 0) {
+            child_recv(1, t1_rank(t1_.right), 1, true);
+        }
+        if (t2_.left >= 0) {
+            child_recv(2, t2_rank(t2_.left), 3, false);
+        }
+        if (t2_.right >= 0) {
+            child_recv(3, t2_rank(t2_.right), 3, false);
+        }
+        phase_ = 0;
+        reda_ = redb_ = false;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    uint32_t from_p1(uint32_t p) const
+    {
+        const uint32_t M = dbt_root_pos(n_);
+        return (p + n_ - M + (uint32_t)a_.root) % n_;
+    }
+    uint32_t t1_rank(int pos) const { return from_p1((uint32_t)pos); }
+    uint32_t t2_rank(int pos) const
+    {
+        return from_p1(((uint32_t)pos + n_ - 1) % n_);
+    }
+
+    void child_recv(int slot, uint32_t rank, uint32_t tag, bool half_a)
+    {
+        size_t len = half_a ? cnta_ * dtsz_ : bytes_ - cnta_ * dtsz_;
+        tmp_[slot].resize(len);
+        cr_[slot] = recv_from(rank, tag, tmp_[slot].data(), len);
+        if (half_a) {
+            na_++;
+        } else {
+            nb_++;
+        }
+    }
+
+    bool kids_done(bool half_a) const
+    {
+        for (int i = half_a ? 0 : 2; i < (half_a ? 2 : 4); i++) {
+            if (cr_[i] && !cr_[i]->done) {
+                return false;
+            }
+        }
+        return true;
+    }
+
+    ucc_status_t progress_()
+    {
+        tt_->progress();
+        const bool     is_root = me_ == (uint32_t)a_.root;
+        const uint32_t M       = dbt_root_pos(n_);
+        uint8_t       *wa      = work_.data();
+        uint8_t       *wb      = work_.data() + cnta_ * dtsz_;
+        const uint64_t cb      = count_ - cnta_;
+        if (!reda_ && kids_done(true)) {
+            /* fold children's half-A partials into mine, pass up T1 */
+            for (int i = 0; i < 2; i++) {
+                if (cr_[i]) {
+                    const void *srcs[2] = {wa, tmp_[i].data()};
+                    ec_cpu::reduce(wa, srcs, 2, cnta_, dt_, op_);
+                }
+            }
+            if (p1_ != M) {
+                send_to(t1_rank(t1_.parent), 1, wa, cnta_ * dtsz_);
+            }
+            reda_ = true;
+        }
+        if (!redb_ && kids_done(false)) {
+            for (int i = 2; i < 4; i++) {
+                if (cr_[i]) {
+                    const void *srcs[2] = {wb, tmp_[i].data()};
+                    ec_cpu::reduce(wb, srcs, 2, cb, dt_, op_);
+                }
+            }
+            if (p2_ != M) {
+                send_to(t2_rank(t2_.parent), 3, wb,
+                        bytes_ - cnta_ * dtsz_);
+            } else if (!is_root) {
+                /* tree-2 root bridges half B to the collective root */
+                send_to((uint32_t)a_.root, 2, wb,
+                        bytes_ - cnta_ * dtsz_);
+            }
+            redb_ = true;
+        }
+        if (!(reda_ && redb_)) {
+            return UCC_INPROGRESS;
+        }
+        if (is_root && !rootrecv_) {
+            /* collective root: half A arrives via T1 (I am its BST
+             * root); half B via the bridge unless I am also T2 root */
+            if (r2_ != me_) {
+                br_.resize(bytes_ - cnta_ * dtsz_);
+                brop_ = recv_from(r2_, 2, br_.data(), br_.size());
+            }
+            rootrecv_ = true;
+        }
+        if (!ops_done()) {
+            return UCC_INPROGRESS;
+        }
+        if (is_root) {
+            uint8_t *dst = (uint8_t *)a_.dst.info.buffer;
+            if (brop_) {
+                memcpy(wb, br_.data(), br_.size());
+            }
+            memcpy(dst, work_.data(), bytes_);
+            if (a_.op == UCC_OP_AVG) {
+                const void *srcs[1] = {dst};
+                ec_cpu::reduce(dst, srcs, 1, count_, dt_, UCC_OP_SUM,
+                               1.0 / (double)n_);
+            }
+        }
+        clear_ops();
+        return UCC_OK;
+    }
+
+    ucc_datatype_t       dt_ = UCC_DT_FLOAT32;
+    ucc_reduction_op_t   op_ = UCC_OP_SUM;
+    size_t               dtsz_ = 4, bytes_ = 0;
+    uint64_t             count_ = 0, cnta_ = 0;
+    uint32_t             p1_ = 0, p2_ = 0, r2_ = 0;
+    DbtNode              t1_, t2_;
+    std::vector<uint8_t> work_, br_;
+    std::vector<uint8_t> tmp_[4];
+    RecvOp              *cr_[4] = {};
+    RecvOp              *brop_  = nullptr;
+    int                  na_ = 0, nb_ = 0;
+    bool reda_ = false, redb_ = false, rootrecv_ = false;
+};
+
 /* ---- barrier / fanin / fanout: binomial fanin to 0 then fanout */
 class TcpBarrierTask final : public TcpTask {
   public:
@@ -1848,6 +2285,71 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
             return UCC_OK;
         };
         map.add(UCC_COLL_TYPE_BCAST, UCC_MEMORY_TYPE_HOST, r);
+    }
+    {
+        /* k-nomial bcast with a radix knob (small-medium latency band)
+         * + DBT bcast/reduce (medium bandwidth*latency band) —
+         * reference recursive_knomial.h / double_binary_tree.h */
+        uint32_t radix = (uint32_t)Config::instance().get_int(
+            "TL_TCP", "KN_RADIX", 4);
+        size_t dbt_min = Config::instance().get_size("TL_TCP",
+                                                     "DBT_MIN", 8192);
+        size_t dbt_max = Config::instance().get_size(
+            "TL_TCP", "DBT_MAX", 4 * 1024 * 1024);
+        ScoreRange kr;
+        kr.start    = 0;
+        kr.end      = dbt_min;
+        kr.score    = sc + 1;
+        kr.tl_name  = "tcp";
+        kr.alg_name = "knomial";
+        kr.init     = [self, radix](const ucc_coll_args_t &args, Team *t2,
+                                Task **task) -> ucc_status_t {
+            const ucc_generic_dt_ops_t *g =
+                ucc_dt_generic_ops(args.src.info.datatype);
+            if ((args.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) ||
+                (g && !(g->flags & UCC_GENERIC_DT_OPS_FLAG_CONTIG))) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpBcastKnomialTask(t2->ctx, self, args, radix);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_BCAST, UCC_MEMORY_TYPE_HOST, kr);
+
+        ScoreRange db;
+        db.start    = dbt_min;
+        db.end      = dbt_max;
+        db.score    = sc + 2;
+        db.tl_name  = "tcp";
+        db.alg_name = "dbt";
+        db.init     = [self](const ucc_coll_args_t &args, Team *t2,
+                         Task **task) -> ucc_status_t {
+            const ucc_generic_dt_ops_t *g =
+                ucc_dt_generic_ops(args.src.info.datatype);
+            if ((args.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) ||
+                (g && !(g->flags & UCC_GENERIC_DT_OPS_FLAG_CONTIG))) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpBcastDbtTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_BCAST, UCC_MEMORY_TYPE_HOST, db);
+
+        ScoreRange dr;
+        dr.start    = dbt_min;
+        dr.end      = dbt_max;
+        dr.score    = sc + 2;
+        dr.tl_name  = "tcp";
+        dr.alg_name = "dbt";
+        dr.init     = [self](const ucc_coll_args_t &args, Team *t2,
+                         Task **task) -> ucc_status_t {
+            if ((args.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) ||
+                !ucc_dt_is_predefined(args.src.info.datatype)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpReduceDbtTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_REDUCE, UCC_MEMORY_TYPE_HOST, dr);
     }
     add(UCC_COLL_TYPE_BCAST, mk((TcpBcastTask *)nullptr));
     add(UCC_COLL_TYPE_BARRIER, mk((TcpBarrierTask *)nullptr));

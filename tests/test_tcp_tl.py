@@ -167,3 +167,95 @@ def test_alltoall_bruck():
     sys.stdout.write(p.stdout[-1000:])
     sys.stderr.write(p.stderr[-2000:])
     assert p.returncode == 0 and "BRUCK_OK" in p.stdout
+
+
+def test_bcast_knomial_and_dbt():
+    """k-nomial radix-k bcast (reference recursive_knomial.h role) and
+    double-binary-tree bcast/reduce (double_binary_tree.h role): all
+    roots x odd/even team sizes x sizes straddling the score bands,
+    validated against numpy references over the tcp mesh."""
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "from ucc_amd import core, dtypes\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "for n in (2, 3, 6, 8):\n"
+        "    job = LocalJob(n)\n"
+        "    c = core()\n"
+        "    smap = c.score_map_str(job.teams[0])\n"
+        "    assert '@tcp/knomial' in smap, smap\n"
+        "    assert '@tcp/dbt' in smap, smap\n"
+        "    rng = np.random.default_rng(n)\n"
+        "    # knomial band (<8KB) and dbt band (8KB..4MB)\n"
+        "    for count in (17, 1000, 5000, 60_000):\n"
+        "        for root in range(n):\n"
+        "            bufs = [np.zeros(count, np.float32)\n"
+        "                    for _ in range(n)]\n"
+        "            bufs[root][:] = rng.standard_normal(count)\n"
+        "            exp = bufs[root].copy()\n"
+        "            reqs = job.coll('bcast', [\n"
+        "                dict(src=b.ctypes.data, dst=0, count=count,\n"
+        "                     dt=dtypes.FLOAT32, root=root)\n"
+        "                for b in bufs])\n"
+        "            job.run(reqs)\n"
+        "            for b in bufs:\n"
+        "                np.testing.assert_array_equal(b, exp)\n"
+        "    # dbt reduce: every root, odd count (uneven halves)\n"
+        "    for count in (4097, 30_001):\n"
+        "        for root in range(n):\n"
+        "            srcs = [rng.standard_normal(count)\n"
+        "                    .astype(np.float32) for _ in range(n)]\n"
+        "            dst = np.zeros(count, np.float32)\n"
+        "            reqs = job.coll('reduce', [\n"
+        "                dict(src=srcs[r].ctypes.data,\n"
+        "                     dst=dst.ctypes.data if r == root else 0,\n"
+        "                     count=count, dt=dtypes.FLOAT32,\n"
+        "                     root=root) for r in range(n)])\n"
+        "            job.run(reqs)\n"
+        "            np.testing.assert_allclose(\n"
+        "                dst, np.sum(srcs, axis=0), rtol=1e-5,\n"
+        "                atol=1e-4)\n"
+        "print('KNOM_DBT_OK')\n"
+    ) % (REPO,)
+    env = dict(os.environ)
+    env["UCC_TL_SHM_ENABLE"] = "0"
+    p = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=600)
+    sys.stdout.write(p.stdout[-2000:])
+    sys.stderr.write(p.stderr[-3000:])
+    assert p.returncode == 0 and "KNOM_DBT_OK" in p.stdout
+
+
+def test_bcast_knomial_radix_sweep():
+    """The radix knob changes the tree shape, not the result."""
+    for radix in (2, 3, 8):
+        code = (
+            "import sys; sys.path.insert(0, %r)\n"
+            "import numpy as np\n"
+            "from ucc_amd import core, dtypes\n"
+            "from ucc_amd.testing import LocalJob\n"
+            "n = 7\n"
+            "job = LocalJob(n)\n"
+            "rng = np.random.default_rng(5)\n"
+            "for root in (0, 3, 6):\n"
+            "    bufs = [np.zeros(999, np.float32) for _ in range(n)]\n"
+            "    bufs[root][:] = rng.standard_normal(999)\n"
+            "    exp = bufs[root].copy()\n"
+            "    reqs = job.coll('bcast', [\n"
+            "        dict(src=b.ctypes.data, dst=0, count=999,\n"
+            "             dt=dtypes.FLOAT32, root=root) for b in bufs])\n"
+            "    job.run(reqs)\n"
+            "    for b in bufs:\n"
+            "        np.testing.assert_array_equal(b, exp)\n"
+            "print('RADIX_OK')\n"
+        ) % (REPO,)
+        env = dict(os.environ)
+        env["UCC_TL_SHM_ENABLE"] = "0"
+        env["UCC_TL_TCP_KN_RADIX"] = str(radix)
+        env["UCC_TUNE"] = "bcast:@knomial:99"
+        p = subprocess.run([sys.executable, "-c", code], env=env,
+                           capture_output=True, text=True, timeout=300)
+        sys.stdout.write(p.stdout[-500:])
+        sys.stderr.write(p.stderr[-2000:])
+        assert p.returncode == 0 and "RADIX_OK" in p.stdout, \
+            f"radix {radix} failed"
