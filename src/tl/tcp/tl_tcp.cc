@@ -752,6 +752,216 @@ class TcpAllreduceSraTask final : public TcpTask {
     bool               in_rs_ = true;
 };
 
+/* ---- sliding-window allreduce (reference tl/ucp
+ * allreduce_sliding_window.c:23-45 role, re-derived for the socket
+ * p2p layer): the message is cut into windows; each window runs the
+ * SRA ring independently and up to DEPTH windows are in flight, so
+ * the reduce-scatter rounds of window w overlap the allgather rounds
+ * of window w-1 — bounded-buffer overlap for huge host messages
+ * without the O(message) idle time of a monolithic ring. Window/round
+ * pairs are disambiguated by the 16-bit tag step (w*128 + round), so
+ * no cross-rank posting-order constraint exists. */
+class TcpAllreduceSlidingTask final : public TcpTask {
+  public:
+    TcpAllreduceSlidingTask(Context *ctx, TcpTlTeam *tt,
+                            const ucc_coll_args_t &args,
+                            size_t win_bytes, int depth)
+        : TcpTask(ctx, tt, args), win_bytes_(win_bytes),
+          depth_(depth < 1 ? 1 : depth)
+    {
+    }
+
+    ucc_status_t post() override
+    {
+        begin();
+        dt_    = a_.dst.info.datatype;
+        op_    = a_.op;
+        dtsz_  = ucc_dt_size(dt_);
+        count_ = a_.dst.info.count;
+        dst_   = (uint8_t *)a_.dst.info.buffer;
+        if (!(a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE)) {
+            ec_cpu::copy(dst_, a_.src.info.buffer, count_ * dtsz_);
+        }
+        welems_ = win_bytes_ / dtsz_;
+        if (welems_ == 0) {
+            welems_ = 1;
+        }
+        /* tag step field fits 512 windows of 128 steps */
+        uint64_t maxw = 448;
+        if ((count_ + welems_ - 1) / welems_ > maxw) {
+            welems_ = (count_ + maxw - 1) / maxw;
+        }
+        nwin_ = (count_ + welems_ - 1) / welems_;
+        size_t d = std::min<size_t>((size_t)depth_, nwin_);
+        fl_.assign(d, Win{});
+        next_ = 0;
+        done_ = 0;
+        for (auto &w : fl_) {
+            start_window(w);
+        }
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    struct Win {
+        uint64_t idx = 0, base = 0, cnt = 0, per = 0;
+        int      round = 0, phase = 0;
+        bool     in_rs = true, active = false;
+        std::vector<uint8_t>  tmp;
+        std::vector<SendOp *> s_ops;
+        std::vector<RecvOp *> r_ops;
+    };
+
+    void start_window(Win &w)
+    {
+        if (next_ >= nwin_) {
+            w.active = false;
+            return;
+        }
+        w.idx    = next_++;
+        w.base   = w.idx * welems_;
+        w.cnt    = std::min<uint64_t>(welems_, count_ - w.base);
+        w.per    = w.cnt / n_;
+        w.round  = 0;
+        w.phase  = 0;
+        w.in_rs  = true;
+        w.active = true;
+        w.tmp.resize((w.per + w.cnt % n_) * dtsz_);
+        w.s_ops.clear();
+        w.r_ops.clear();
+    }
+
+    uint64_t boff(const Win &w, uint32_t b) const
+    {
+        return w.base + (uint64_t)b * w.per;
+    }
+    uint64_t bcnt(const Win &w, uint32_t b) const
+    {
+        return b == n_ - 1 ? w.cnt - (uint64_t)(n_ - 1) * w.per : w.per;
+    }
+    bool wops_done(const Win &w) const
+    {
+        for (auto *s : w.s_ops) {
+            if (!s->done) {
+                return false;
+            }
+        }
+        for (auto *r : w.r_ops) {
+            if (!r->done) {
+                return false;
+            }
+        }
+        return true;
+    }
+
+    ucc_status_t progress_()
+    {
+        const uint32_t right = (me_ + 1) % n_;
+        const uint32_t left  = (me_ + n_ - 1) % n_;
+        tt_->progress();
+        bool moved = true;
+        while (moved) {
+            moved = false;
+            for (auto &w : fl_) {
+                if (!w.active) {
+                    continue;
+                }
+                const uint32_t tag0 = (uint32_t)(w.idx * 128);
+                if (w.in_rs && w.round >= (int)n_ - 1) {
+                    w.in_rs = false;
+                    w.round = 0;
+                }
+                if (w.in_rs) { /* reduce-scatter rounds */
+                    if (w.phase == 0) {
+                        uint32_t sb = (me_ + n_ - w.round) % n_;
+                        uint32_t rb = (me_ + n_ - w.round - 1) % n_;
+                        w.s_ops.clear();
+                        w.r_ops.clear();
+                        if (bcnt(w, sb)) {
+                            w.s_ops.push_back(send_to(
+                                right, tag0 + (uint32_t)w.round,
+                                dst_ + boff(w, sb) * dtsz_,
+                                bcnt(w, sb) * dtsz_));
+                        }
+                        if (bcnt(w, rb)) {
+                            w.r_ops.push_back(recv_from(
+                                left, tag0 + (uint32_t)w.round,
+                                w.tmp.data(), bcnt(w, rb) * dtsz_));
+                        }
+                        w.phase = 1;
+                        moved   = true;
+                    }
+                    if (w.phase == 1 && wops_done(w)) {
+                        uint32_t rb = (me_ + n_ - w.round - 1) % n_;
+                        if (bcnt(w, rb)) {
+                            const bool last = w.round == (int)n_ - 2;
+                            const void *srcs[2] = {
+                                dst_ + boff(w, rb) * dtsz_,
+                                w.tmp.data()};
+                            ec_cpu::reduce(
+                                dst_ + boff(w, rb) * dtsz_, srcs, 2,
+                                bcnt(w, rb), dt_,
+                                op_ == UCC_OP_AVG ? UCC_OP_SUM : op_,
+                                (last && op_ == UCC_OP_AVG)
+                                    ? 1.0 / n_
+                                    : 1.0);
+                        }
+                        w.phase = 0;
+                        w.round++;
+                        moved = true;
+                        if (w.round >= (int)n_ - 1) {
+                            w.in_rs = false;
+                            w.round = 0;
+                        }
+                    }
+                } else if (w.round < (int)n_ - 1) { /* allgather */
+                    if (w.phase == 0) {
+                        uint32_t sb = (me_ + 1 + n_ - w.round) % n_;
+                        uint32_t rb = (me_ + n_ - w.round) % n_;
+                        w.s_ops.clear();
+                        w.r_ops.clear();
+                        if (bcnt(w, sb)) {
+                            w.s_ops.push_back(send_to(
+                                right, tag0 + 64 + (uint32_t)w.round,
+                                dst_ + boff(w, sb) * dtsz_,
+                                bcnt(w, sb) * dtsz_));
+                        }
+                        if (bcnt(w, rb)) {
+                            w.r_ops.push_back(recv_from(
+                                left, tag0 + 64 + (uint32_t)w.round,
+                                dst_ + boff(w, rb) * dtsz_,
+                                bcnt(w, rb) * dtsz_));
+                        }
+                        w.phase = 1;
+                        moved   = true;
+                    }
+                    if (w.phase == 1 && wops_done(w)) {
+                        w.phase = 0;
+                        w.round++;
+                        moved = true;
+                    }
+                } else { /* window complete */
+                    done_++;
+                    start_window(w);
+                    moved = w.active;
+                }
+            }
+        }
+        return done_ == nwin_ ? UCC_OK : UCC_INPROGRESS;
+    }
+
+    size_t   win_bytes_;
+    int      depth_;
+    ucc_datatype_t     dt_ = UCC_DT_FLOAT32;
+    ucc_reduction_op_t op_ = UCC_OP_SUM;
+    size_t   dtsz_ = 4;
+    uint64_t count_ = 0, welems_ = 0, nwin_ = 0, next_ = 0, done_ = 0;
+    uint8_t *dst_ = nullptr;
+    std::vector<Win> fl_;
+};
+
 /* ---- bcast: binomial tree from root. With an ACTIVE_SET
  * ({start, stride, size}, reference ucc.h active_set + tl/ucp active-set
  * bcast), the tree runs over the strided subset only; the wire tag comes
@@ -2259,6 +2469,33 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
         r.init     = [self](const ucc_coll_args_t &args, Team *t2,
                         Task **task) -> ucc_status_t {
             *task = new TcpAllreduceSraTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_ALLREDUCE, UCC_MEMORY_TYPE_HOST, r);
+    }
+    {
+        /* sliding-window overlap for huge host messages (reference
+         * allreduce_sliding_window.c role) */
+        size_t sw_min = Config::instance().get_size(
+            "TL_TCP", "SLIDING_MIN", 64 * 1024 * 1024);
+        size_t sw_win = Config::instance().get_size(
+            "TL_TCP", "SLIDING_WINDOW", 8 * 1024 * 1024);
+        int sw_depth = (int)Config::instance().get_int(
+            "TL_TCP", "SLIDING_DEPTH", 2);
+        ScoreRange r;
+        r.start    = sw_min;
+        r.end      = SIZE_MAX;
+        r.score    = sc + 2;
+        r.tl_name  = "tcp";
+        r.alg_name = "sliding_window";
+        r.init     = [self, sw_win, sw_depth](
+                     const ucc_coll_args_t &args, Team *t2,
+                     Task **task) -> ucc_status_t {
+            if (!ucc_dt_is_predefined(args.dst.info.datatype)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpAllreduceSlidingTask(t2->ctx, self, args,
+                                                sw_win, sw_depth);
             return UCC_OK;
         };
         map.add(UCC_COLL_TYPE_ALLREDUCE, UCC_MEMORY_TYPE_HOST, r);

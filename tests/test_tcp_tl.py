@@ -259,3 +259,46 @@ def test_bcast_knomial_radix_sweep():
         sys.stderr.write(p.stderr[-2000:])
         assert p.returncode == 0 and "RADIX_OK" in p.stdout, \
             f"radix {radix} failed"
+
+
+def test_allreduce_sliding_window():
+    """Sliding-window allreduce (reference allreduce_sliding_window.c
+    role): windows of the message run SRA rings concurrently with a
+    bounded depth. Threshold lowered so the CPU test exercises many
+    windows; AVG covers the per-window final-round scale."""
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "from ucc_amd import core, dtypes\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "for n in (2, 5):\n"
+        "    job = LocalJob(n)\n"
+        "    c = core()\n"
+        "    assert '@tcp/sliding_window' in "
+        "c.score_map_str(job.teams[0])\n"
+        "    rng = np.random.default_rng(n)\n"
+        "    count = 700_001  # ~2.8MB fp32 -> ~11 windows of 256KB\n"
+        "    arrs = [(rng.random(count) - 0.5).astype(np.float32)\n"
+        "            for _ in range(n)]\n"
+        "    outs = job.allreduce_np(arrs)\n"
+        "    exp = np.sum(arrs, axis=0)\n"
+        "    for o in outs:\n"
+        "        np.testing.assert_allclose(o, exp, rtol=1e-5,\n"
+        "                                   atol=1e-4)\n"
+        "    outs = job.allreduce_np(arrs, op=dtypes.OP_AVG)\n"
+        "    exp = np.mean(arrs, axis=0)\n"
+        "    for o in outs:\n"
+        "        np.testing.assert_allclose(o, exp, rtol=1e-5,\n"
+        "                                   atol=1e-5)\n"
+        "print('SLIDING_OK')\n"
+    ) % (REPO,)
+    env = dict(os.environ)
+    env["UCC_TL_SHM_ENABLE"] = "0"
+    env["UCC_TL_TCP_SLIDING_MIN"] = "1048576"
+    env["UCC_TL_TCP_SLIDING_WINDOW"] = "262144"
+    env["UCC_TL_TCP_SLIDING_DEPTH"] = "3"
+    p = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=600)
+    sys.stdout.write(p.stdout[-1000:])
+    sys.stderr.write(p.stderr[-3000:])
+    assert p.returncode == 0 and "SLIDING_OK" in p.stdout
