@@ -25,6 +25,7 @@ namespace py = pybind11;
 extern "C" ucc_status_t ucc_amd_coll_from_name_c(const char *,
                                                  ucc_coll_type_t *);
 extern "C" int ucc_amd_hip_device_count_c();
+extern "C" uint64_t ucc_amd_cdna4_memh_uses();
 
 /* ------------------------------------------------------------- LocalOob */
 struct LocalOobShared {
@@ -326,7 +327,8 @@ PYBIND11_MODULE(_core, m)
            std::vector<uint64_t> src_counts, std::vector<uint64_t> src_displs,
            std::vector<uint64_t> dst_counts,
            std::vector<uint64_t> dst_displs, double timeout,
-           std::vector<int64_t> active_set, int tag, int src_mem_type) {
+           std::vector<int64_t> active_set, int tag, int src_mem_type,
+           uintptr_t src_memh, uintptr_t dst_memh) {
             int s_mt = src_mem_type >= 0 ? src_mem_type : mem_type;
             auto r  = std::make_shared<CoreReq>();
             r->team = team;
@@ -354,6 +356,14 @@ PYBIND11_MODULE(_core, m)
             if (tag >= 0) {
                 a.mask |= UCC_COLL_ARGS_FIELD_TAG;
                 a.tag = (uint16_t)tag;
+            }
+            if (src_memh) {
+                a.mask |= UCC_COLL_ARGS_FIELD_MEM_MAP_SRC_MEMH;
+                a.src_memh = (ucc_mem_map_mem_h)src_memh;
+            }
+            if (dst_memh) {
+                a.mask |= UCC_COLL_ARGS_FIELD_MEM_MAP_DST_MEMH;
+                a.dst_memh = (ucc_mem_map_mem_h)dst_memh;
             }
             bool sv = ct == UCC_COLL_TYPE_ALLTOALLV ||
                       ct == UCC_COLL_TYPE_SCATTERV;
@@ -402,7 +412,8 @@ PYBIND11_MODULE(_core, m)
         py::arg("dst_displs") = std::vector<uint64_t>(),
         py::arg("timeout") = 0.0,
         py::arg("active_set") = std::vector<int64_t>(),
-        py::arg("tag") = -1, py::arg("src_mem_type") = -1);
+        py::arg("tag") = -1, py::arg("src_mem_type") = -1,
+        py::arg("src_memh") = 0, py::arg("dst_memh") = 0);
 
     /* ------------------------------------------------------ mem_map */
     m.def("mem_map_export", [](uintptr_t addr, size_t len) {
@@ -439,6 +450,23 @@ PYBIND11_MODULE(_core, m)
         auto *v = (SegView *)copy;
         return (uintptr_t)(v->mapped + v->base_off);
     });
+    /* live exported handle for coll-args src_memh/dst_memh (the
+     * bytes-returning export above frees its handle immediately) */
+    m.def("mem_map_export_keep", [](uintptr_t addr, size_t len) {
+        ucc_mem_map_t        seg{(void *)addr, len};
+        ucc_mem_map_params_t p{&seg, 1};
+        size_t               sz = 0;
+        ucc_mem_map_mem_h    h  = nullptr;
+        check(ucc_mem_map(nullptr, UCC_MEM_MAP_MODE_EXPORT, &p, &sz, &h),
+              "ucc_mem_map export");
+        return (uintptr_t)h;
+    });
+    m.def("mem_unmap_handle", [](uintptr_t h) {
+        ucc_mem_map_mem_h hh = (ucc_mem_map_mem_h)h;
+        ucc_mem_unmap(&hh);
+    });
+    m.def("cdna4_memh_uses",
+          []() { return (uint64_t)ucc_amd_cdna4_memh_uses(); });
     m.def("mem_map_close", [](uintptr_t mapped) {
         ucc::mc::ipc_close((void *)mapped);
     });

@@ -1217,6 +1217,30 @@ ucc_status_t ucc_mem_map(ucc_context_h context, ucc_mem_map_flags_t flags,
     return UCC_ERR_INVALID_PARAM;
 }
 
+/* TL helper (reference alltoall_onesided.c src/dst memh role): look up
+ * a registered EXPORTED segment covering [addr, addr+len) and hand back
+ * its IPC handle + the offset of addr from the exporting allocation's
+ * base. Lets collectives on registered buffers skip the per-post
+ * hipIpcGetMemHandle. */
+int ucc_memh_lookup(ucc_mem_map_mem_h memh, const void *addr, size_t len,
+                    void *handle_out, uint64_t *alloc_off)
+{
+    auto *b = (MemMapBlob *)memh;
+    if (!b || b->magic != kMemMapMagic || b->imported) {
+        return 0;
+    }
+    uint64_t a = (uint64_t)(uintptr_t)addr;
+    for (uint32_t i = 0; i < b->n; i++) {
+        MemMapSeg &s = b->segs[i];
+        if (s.has_ipc && a >= s.addr && a + len <= s.addr + s.len) {
+            memcpy(handle_out, s.handle, ucc::mc::kIpcHandleBytes);
+            *alloc_off = s.base_off + (a - s.addr);
+            return 1;
+        }
+    }
+    return 0;
+}
+
 ucc_status_t ucc_mem_unmap(ucc_mem_map_mem_h *memh)
 {
     if (!memh || !*memh) {

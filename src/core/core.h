@@ -427,4 +427,13 @@ ucc_memory_type_t coll_args_mem_type(const ucc_coll_args_t &args,
 
 } // namespace ucc
 
+/* mem_map handle lookup for TLs (defined in core.cc, C linkage): finds
+ * a registered device segment covering [addr, addr+len) inside an
+ * EXPORTED ucc_mem_map handle; copies its IPC handle (64B) and sets
+ * the offset of addr from the exporting allocation base. Returns 1 on
+ * hit. */
+extern "C" int ucc_memh_lookup(ucc_mem_map_mem_h memh, const void *addr,
+                               size_t len, void *handle_out,
+                               uint64_t *alloc_off);
+
 #endif

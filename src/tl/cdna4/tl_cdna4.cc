@@ -23,6 +23,7 @@
  * allgather(v), reduce_scatter(v), bcast, reduce. Host-memory colls and
  * everything else fall back per the score map.
  */
+#include <atomic>
 #include <map>
 #include <mutex>
 
@@ -64,6 +65,14 @@ static inline size_t coll_disp_at(const ucc_coll_args_t &a, const void *d,
     return (a.flags & UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT)
                ? (size_t)((const uint64_t *)d)[i]
                : (size_t)((const uint32_t *)d)[i];
+}
+
+/* observability: zc exchanges that consumed user-registered mem_map
+ * handles (honored src_memh/dst_memh) — regression hook for tests */
+static std::atomic<uint64_t> g_memh_uses{0};
+extern "C" uint64_t ucc_amd_cdna4_memh_uses()
+{
+    return g_memh_uses.load(std::memory_order_relaxed);
 }
 
 #define HIPCHK(expr)                                                         \
@@ -1676,9 +1685,55 @@ class GatedCollTask final : public Cdna4Task {
             }
             ZcBlob b{};
             b.magic = 0x5a43;
+            /* registered-buffer fast path (reference
+             * alltoall_onesided.c src_memh/dst_memh role): handles
+             * come from the user's ucc_mem_map export instead of a
+             * per-post hipIpcGetMemHandle */
+            uint64_t moff = 0;
+            bool     src_reg =
+                (a_.mask & UCC_COLL_ARGS_FIELD_MEM_MAP_SRC_MEMH) &&
+                a_.src_memh &&
+                ucc_memh_lookup(a_.src_memh, sbuf_, total_, &b.h,
+                                &moff);
+            if (src_reg) {
+                g_memh_uses.fetch_add(1, std::memory_order_relaxed);
+                b.base_off = moff;
+                b.raw_ptr  = (uint64_t)(uintptr_t)sbuf_;
+                b.pid      = (int32_t)getpid();
+                uint64_t doff = 0;
+                if ((a_.mask &
+                     UCC_COLL_ARGS_FIELD_MEM_MAP_DST_MEMH) &&
+                    a_.dst_memh &&
+                    ucc_memh_lookup(a_.dst_memh, dbuf_, total_, &b.hd,
+                                    &doff)) {
+                    b.d_base_off = doff;
+                    b.d_raw_ptr  = (uint64_t)(uintptr_t)dbuf_;
+                    b.pad        = 0;
+                } else {
+                    /* dst not registered: fall through to the runtime
+                     * export for the dst side only */
+                    hipDeviceptr_t dbase  = nullptr;
+                    size_t         dbsize = 0;
+                    if (hipMemGetAddressRange(&dbase, &dbsize,
+                                              (hipDeviceptr_t)dbuf_) ==
+                            hipSuccess &&
+                        hipIpcGetMemHandle(&b.hd, (void *)dbase) ==
+                            hipSuccess) {
+                        b.d_base_off = (uint64_t)((uintptr_t)dbuf_ -
+                                                  (uintptr_t)dbase);
+                        b.d_raw_ptr  = (uint64_t)(uintptr_t)dbuf_;
+                        b.pad        = 0;
+                    } else {
+                        b.magic = 0;
+                        zc_     = false;
+                    }
+                }
+            }
             hipDeviceptr_t base  = nullptr;
             size_t         bsize = 0;
-            if (hipMemGetAddressRange(&base, &bsize,
+            if (src_reg) {
+                /* blob already filled from the registered handles */
+            } else if (hipMemGetAddressRange(&base, &bsize,
                                       (hipDeviceptr_t)sbuf_) !=
                     hipSuccess ||
                 hipIpcGetMemHandle(&b.h, (void *)base) != hipSuccess) {

@@ -138,6 +138,35 @@ def main():
     del r2c
     results.append("zc_realloc_xproc")
 
+    # 2d. registered-buffer (onesided-role) collective: src/dst memh
+    # from ucc_mem_map are honored — the zc exchange consumes the
+    # pre-exported handles instead of per-post hipIpcGetMemHandle
+    # (reference tl/ucp alltoall_onesided src_memh/dst_memh role).
+    count3 = 12_000_000
+    full3 = torch.randn(world, count3, generator=g0)
+    src3 = full3[rank].cuda()
+    dst3 = torch.zeros(count3, device="cuda")
+    sh = c.mem_map_export_keep(src3.data_ptr(), count3 * 4)
+    dh = c.mem_map_export_keep(dst3.data_ptr(), count3 * 4)
+    uses0 = c.cdna4_memh_uses()
+    r2d = c.coll_init(team, "allreduce", src=src3.data_ptr(),
+                      dst=dst3.data_ptr(), count=count3,
+                      dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA,
+                      flags=c.FLAG_PERSISTENT, src_memh=sh, dst_memh=dh)
+    for it in range(2):
+        src3.copy_(full3[rank] * (it + 1))
+        torch.cuda.synchronize()
+        wait(r2d, ctx)
+        torch.cuda.synchronize()
+        torch.testing.assert_close(dst3.cpu(), full3.sum(0) * (it + 1),
+                                   rtol=1e-5, atol=1e-4)
+    assert c.cdna4_memh_uses() > uses0, \
+        "registered handles were not consumed by the zc exchange"
+    del r2d
+    c.mem_unmap_handle(sh)
+    c.mem_unmap_handle(dh)
+    results.append("memh_onesided_xproc")
+
     # 3. cross-process alltoallv (skewed, fp16)
     scnt = [[(r + 1) * (d + 1) * 1024 for d in range(world)]
             for r in range(world)]
