@@ -490,6 +490,21 @@ PYBIND11_MODULE(_core, m)
         d["node_idx"]     = node.my_idx;
         d["leaders_size"] = ldr.ranks.size();
         d["leaders_idx"]  = ldr.my_idx;
+        auto sock = ucc::topo::build_sbgp(team,
+                                          ucc::topo::SbgpType::SOCKET);
+        auto numa = ucc::topo::build_sbgp(team,
+                                          ucc::topo::SbgpType::NUMA);
+        auto sldr = ucc::topo::build_sbgp(
+            team, ucc::topo::SbgpType::SOCKET_LEADERS);
+        auto nldr = ucc::topo::build_sbgp(
+            team, ucc::topo::SbgpType::NUMA_LEADERS);
+        d["socket_size"]         = sock.ranks.size();
+        d["socket_idx"]          = sock.my_idx;
+        d["numa_size"]           = numa.ranks.size();
+        d["numa_idx"]            = numa.my_idx;
+        d["socket_leaders_size"] = sldr.ranks.size();
+        d["numa_leaders_size"]   = nldr.ranks.size();
+        d["same_cpu"] = ucc::topo::team_same_cpu(team);
         return d;
     });
     m.def("gpu_link_hops", []() {

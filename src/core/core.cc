@@ -24,6 +24,19 @@ static uint64_t str_hash(const char *s)
     return h;
 }
 
+/* read one small integer file (sysfs); -1 on failure */
+static int read_int_file(const char *path)
+{
+    FILE *f = fopen(path, "r");
+    if (!f) {
+        return -1;
+    }
+    int  v  = -1;
+    int  ok = fscanf(f, "%d", &v);
+    fclose(f);
+    return ok == 1 ? v : -1;
+}
+
 ProcInfo local_proc_info()
 {
     ProcInfo pi;
@@ -32,6 +45,66 @@ ProcInfo local_proc_info()
     pi.host_hash = str_hash(host);
     pi.pid       = (int32_t)getpid();
     pi.device    = -1;
+    /* socket/NUMA of the cpu this process currently runs on
+     * (reference ucc_proc_info sysfs role). Best effort: stays -1
+     * where sysfs is unavailable. */
+    int cpu = sched_getcpu();
+    if (cpu >= 0) {
+        char path[128];
+        snprintf(path, sizeof(path),
+                 "/sys/devices/system/cpu/cpu%d/topology/"
+                 "physical_package_id", cpu);
+        pi.socket_id = (int16_t)read_int_file(path);
+        for (int nd = 0; nd < 64; nd++) {
+            snprintf(path, sizeof(path),
+                     "/sys/devices/system/node/node%d/cpumap", nd);
+            FILE *f = fopen(path, "r");
+            if (!f) {
+                break;
+            }
+            /* cpumap is a comma-separated hex mask, lowest word last */
+            char mask[512] = {0};
+            if (fgets(mask, sizeof(mask), f)) {
+                /* walk 32-bit words from the right */
+                int         bit   = 0;
+                bool        found = false;
+                const char *end   = mask + strlen(mask);
+                const char *q     = end;
+                while (q > mask && !found) {
+                    const char *comma = q;
+                    while (comma > mask && comma[-1] != ',') {
+                        comma--;
+                    }
+                    uint32_t w = (uint32_t)strtoul(comma, nullptr, 16);
+                    if (cpu >= bit && cpu < bit + 32 &&
+                        (w >> (cpu - bit)) & 1u) {
+                        pi.numa_id = (int16_t)nd;
+                        found      = true;
+                    }
+                    bit += 32;
+                    q = comma > mask ? comma - 1 : mask;
+                }
+            }
+            fclose(f);
+            if (pi.numa_id >= 0) {
+                break;
+            }
+        }
+    }
+    /* CPU model consensus hash (vendor+model from /proc/cpuinfo) */
+    FILE *ci = fopen("/proc/cpuinfo", "r");
+    if (ci) {
+        char line[256];
+        while (fgets(line, sizeof(line), ci)) {
+            if (!strncmp(line, "model name", 10) ||
+                !strncmp(line, "vendor_id", 9)) {
+                pi.cpu_hash =
+                    (uint32_t)(str_hash(line) ^ (pi.cpu_hash * 31));
+                break;
+            }
+        }
+        fclose(ci);
+    }
     return pi;
 }
 

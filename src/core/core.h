@@ -197,6 +197,12 @@ struct ProcInfo {
     int32_t  pid       = 0;
     int32_t  device    = -1; /* hip device id, -1 = none                   */
     uint64_t ctx_seq   = 0;  /* unique per (pid, context)                  */
+    /* placement within the node (reference ucc_proc_info socket/numa
+     * via sysfs) and a CPU model hash for symmetric-tuning consensus
+     * (reference ucc_topo.h:88-95) */
+    int16_t  socket_id = -1;
+    int16_t  numa_id   = -1;
+    uint32_t cpu_hash  = 0;
 };
 ProcInfo local_proc_info();
 

@@ -36,6 +36,51 @@ Sbgp build_sbgp(const Team *team, SbgpType type)
         }
         break;
     }
+    case SbgpType::SOCKET:
+    case SbgpType::NUMA: {
+        /* intra-node grouping by socket/NUMA id; unknown ids (-1)
+         * degrade to one NODE-wide group (correct, untuned) */
+        uint64_t my_host = procs[team->rank].host_hash;
+        auto     key     = [&](uint32_t r) {
+            return type == SbgpType::SOCKET ? procs[r].socket_id
+                                                : procs[r].numa_id;
+        };
+        int my_key = key(team->rank);
+        for (uint32_t r = 0; r < n; r++) {
+            if (procs[r].host_hash == my_host && key(r) == my_key) {
+                if (r == team->rank) {
+                    s.my_idx = (int)s.ranks.size();
+                }
+                s.ranks.push_back(r);
+            }
+        }
+        break;
+    }
+    case SbgpType::SOCKET_LEADERS:
+    case SbgpType::NUMA_LEADERS: {
+        /* lowest rank per distinct (socket|numa) id ON MY NODE */
+        uint64_t my_host = procs[team->rank].host_hash;
+        auto     key     = [&](uint32_t r) {
+            return type == SbgpType::SOCKET_LEADERS
+                           ? procs[r].socket_id
+                           : procs[r].numa_id;
+        };
+        std::vector<int> seen;
+        for (uint32_t r = 0; r < n; r++) {
+            if (procs[r].host_hash != my_host) {
+                continue;
+            }
+            int k = key(r);
+            if (std::find(seen.begin(), seen.end(), k) == seen.end()) {
+                seen.push_back(k);
+                if (r == team->rank) {
+                    s.my_idx = (int)s.ranks.size();
+                }
+                s.ranks.push_back(r);
+            }
+        }
+        break;
+    }
     case SbgpType::NODE_LEADERS: {
         /* lowest team rank per distinct host hash, in rank order */
         std::vector<uint64_t> seen;
@@ -53,6 +98,16 @@ Sbgp build_sbgp(const Team *team, SbgpType type)
     }
     }
     return s;
+}
+
+bool team_same_cpu(const Team *team)
+{
+    for (const auto &p : team->procs) {
+        if (p.cpu_hash != team->procs[0].cpu_hash) {
+            return false;
+        }
+    }
+    return true;
 }
 
 const GpuLinks &gpu_links()

@@ -20,10 +20,19 @@ namespace ucc {
 namespace topo {
 
 enum class SbgpType {
-    NODE,         /* my node's ranks                       */
-    NODE_LEADERS, /* lowest rank of each node              */
+    NODE,           /* my node's ranks                       */
+    NODE_LEADERS,   /* lowest rank of each node              */
+    SOCKET,         /* my node's ranks on my CPU socket      */
+    SOCKET_LEADERS, /* lowest rank of each socket on my node */
+    NUMA,           /* my node's ranks on my NUMA domain     */
+    NUMA_LEADERS,   /* lowest rank of each NUMA on my node   */
     FULL,
 };
+
+/* CPU-model consensus across the team (reference ucc_topo.h:88-95):
+ * symmetric tuning defaults are only safe when every rank runs on the
+ * same CPU model. */
+bool team_same_cpu(const Team *team);
 
 struct Sbgp {
     SbgpType              type;

@@ -326,6 +326,18 @@ def test_topo_sbgps(job):
     assert info["node_idx"] == 0
     assert info["leaders_size"] == 1
     assert info["leaders_idx"] == 0
+    # SOCKET/NUMA sbgps (reference ucc_sbgp.h:10-41 kinds): the whole
+    # in-process jig runs on one machine, so the subgroup sizes sum to
+    # the node and my_idx is a valid membership; unknown ids (-1) must
+    # degrade to a full-node group, never an empty one.
+    assert 1 <= info["socket_size"] <= job.n
+    assert 0 <= info["socket_idx"] < info["socket_size"]
+    assert 1 <= info["numa_size"] <= job.n
+    assert 0 <= info["numa_idx"] < info["numa_size"]
+    assert info["socket_leaders_size"] >= 1
+    assert info["numa_leaders_size"] >= 1
+    # every rank of the jig is the same process => same CPU model
+    assert info["same_cpu"] is True
 
 
 @pytest.mark.parametrize("npdt,dt", [
