@@ -524,6 +524,21 @@ PYBIND11_MODULE(_core, m)
     });
     m.def("ee_destroy",
           [](uintptr_t ee) { ucc_ee_destroy((ucc_ee_h)ee); });
+    /* pop one event from the EE queue; returns ev_type or -1 if empty
+     * (reference ucc.h event flow: POST at launch, COLLECTIVE_COMPLETE
+     * when the stream work finishes) */
+    m.def("ee_pop_event", [](uintptr_t ee) {
+        ucc_ev_t *ev = nullptr;
+        if (ucc_ee_get_event((ucc_ee_h)ee, &ev) != UCC_OK) {
+            return -1;
+        }
+        int t = (int)ev->ev_type;
+        ucc_ee_ack_event((ucc_ee_h)ee, ev);
+        return t;
+    });
+    m.attr("EVENT_COLLECTIVE_POST") = (int)UCC_EVENT_COLLECTIVE_POST;
+    m.attr("EVENT_COLLECTIVE_COMPLETE") =
+        (int)UCC_EVENT_COLLECTIVE_COMPLETE;
     m.def("triggered_post", [](uintptr_t ee, std::shared_ptr<CoreReq> r) {
         ucc_ev_t ev{};
         ev.ev_type = UCC_EVENT_COMPUTE_COMPLETE;

@@ -1145,9 +1145,30 @@ ucc_status_t ucc_ee_destroy(ucc_ee_h ee)
     delete reinterpret_cast<Ee *>(ee);
     return UCC_OK;
 }
+/* drain fired stream events into the user-visible queue */
+static void ee_poll_pending(Ee *e)
+{
+    while (!e->pending.empty()) {
+        auto &p = e->pending.front();
+        int   q = mc::event_query(p.hip_ev);
+        if (q == 0) {
+            break; /* stream work still running (FIFO per stream) */
+        }
+        ucc_ev_t done = p.ev;
+        done.ev_type  = UCC_EVENT_COLLECTIVE_COMPLETE;
+        if (q < 0 && done.req) {
+            req_status_store(done.req, UCC_ERR_NO_RESOURCE);
+        }
+        e->events.push_back(done);
+        mc::event_free(p.hip_ev);
+        e->pending.pop_front();
+    }
+}
+
 ucc_status_t ucc_ee_get_event(ucc_ee_h ee, ucc_ev_t **ev)
 {
     auto *e = reinterpret_cast<Ee *>(ee);
+    ee_poll_pending(e);
     if (e->events.empty()) {
         return UCC_ERR_NOT_FOUND;
     }
@@ -1199,8 +1220,21 @@ ucc_status_t ucc_collective_triggered_post(ucc_ee_h ee, ucc_ev_t *ev)
     req_status_store(&req->super, req->task->status.load());
     ucc_ev_t done      = *ev;
     done.ev_type       = UCC_EVENT_COLLECTIVE_POST;
+    done.req           = &req->super;
     e->events.push_back(done);
+    /* completion event: fires when the stream-ordered work finishes */
+    void *hev = nullptr;
+    if (mc::stream_event_record(e->stream, &hev) == UCC_OK) {
+        e->pending.push_back(Ee::Pending{hev, done});
+    }
     return UCC_OK;
+}
+
+Ee::~Ee()
+{
+    for (auto &p : pending) {
+        mc::event_free(p.hip_ev);
+    }
 }
 
 /* ------------------------------------------------------------- mem_map */

@@ -368,6 +368,16 @@ struct Ee {
     ucc_ee_params_t  params{};
     void            *stream = nullptr; /* hipStream_t */
     std::deque<ucc_ev_t> events;
+    /* triggered collectives whose stream work is in flight: a HIP
+     * event recorded after the launch + the COLLECTIVE_COMPLETE ev to
+     * publish once it fires (reference ucc.h event flow: POST at
+     * launch, COMPLETE when the device work is done) */
+    struct Pending {
+        void     *hip_ev;
+        ucc_ev_t  ev;
+    };
+    std::deque<Pending> pending;
+    ~Ee();
 };
 
 /* ----------------------------------------------------------- CollRequest */

@@ -45,6 +45,13 @@ ucc_status_t ipc_export(const void *ptr, void *handle_out,
 ucc_status_t ipc_import(const void *handle, void **mapped);
 ucc_status_t ipc_close(void *mapped);
 
+/* HIP event helpers for the EE completion-event flow (gated: no-ops
+ * without a device). ev handles are opaque (hipEvent_t). */
+ucc_status_t stream_event_record(void *stream, void **ev_out);
+/* 1 = complete, 0 = pending, <0 = error */
+int          event_query(void *ev);
+void         event_free(void *ev);
+
 static inline bool is_device_mt(ucc_memory_type_t mt)
 {
     return mt == UCC_MEMORY_TYPE_CUDA || mt == UCC_MEMORY_TYPE_ROCM ||

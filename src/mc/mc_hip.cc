@@ -151,6 +151,44 @@ size_t scratch_raw_allocs()
     return g_raw_allocs.load(std::memory_order_relaxed);
 }
 
+ucc_status_t stream_event_record(void *stream, void **ev_out)
+{
+    if (!hip_available()) {
+        *ev_out = nullptr;
+        return UCC_ERR_NOT_SUPPORTED;
+    }
+    hipEvent_t e = nullptr;
+    if (hipEventCreateWithFlags(&e, hipEventDisableTiming) !=
+            hipSuccess ||
+        hipEventRecord(e, (hipStream_t)stream) != hipSuccess) {
+        if (e) {
+            (void)hipEventDestroy(e);
+        }
+        return UCC_ERR_NO_RESOURCE;
+    }
+    *ev_out = (void *)e;
+    return UCC_OK;
+}
+
+int event_query(void *ev)
+{
+    hipError_t e = hipEventQuery((hipEvent_t)ev);
+    if (e == hipSuccess) {
+        return 1;
+    }
+    if (e == hipErrorNotReady) {
+        return 0;
+    }
+    return -1;
+}
+
+void event_free(void *ev)
+{
+    if (ev) {
+        (void)hipEventDestroy((hipEvent_t)ev);
+    }
+}
+
 ucc_status_t ipc_export(const void *ptr, void *handle_out,
                         size_t *base_off_out)
 {

@@ -412,6 +412,18 @@ def main():
         s.synchronize()
         torch.testing.assert_close(dst.cpu(), expected, rtol=1e-5,
                                    atol=1e-4)
+        # EE event-queue flow (reference ucc.h:2050-2260): a POST
+        # event at launch and a COLLECTIVE_COMPLETE once the stream
+        # work finished (the stream is synchronized here, so both
+        # must be available).
+        evs = []
+        while True:
+            t = c.ee_pop_event(ee)
+            if t < 0:
+                break
+            evs.append(t)
+        assert c.EVENT_COLLECTIVE_POST in evs, evs
+        assert c.EVENT_COLLECTIVE_COMPLETE in evs, evs
     results.append("triggered")
 
     # 5. hipGraph capture + replay of the triggered fused allreduce
