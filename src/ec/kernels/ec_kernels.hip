@@ -839,7 +839,7 @@ __global__ void k_staged_stage(const GatedArgs a)
     gated_signal(a, 0, t.sig_stage);
 }
 
-template <typename T, int OP, int VEC>
+template <typename T, int OP, int VEC, int U>
 __global__ void k_staged_reduce(const GatedArgs a)
 {
     using A = typename Cvt<T>::A;
@@ -860,55 +860,76 @@ __global__ void k_staged_reduce(const GatedArgs a)
      * unrolling keeps the 8 peer base pointers in SGPRs and issues all
      * peer loads back-to-back for xGMI latency overlap. */
     P *out = (P *)a.my_out;
-    /* 2-deep batching: both 16B vectors' peer loads are issued before
-     * any reduction consumes them, doubling outstanding loads per lane
-     * (2n x 16B; HBM-miss latency ~900 cyc and xGMI more — G7 ILP). */
+    /* U-deep batching: all U x n peer loads are issued before any
+     * reduction consumes them. The per-lane bytes-in-flight must cover
+     * the bandwidth-latency product (≈6.3 TB/s x ~600 ns needs ~4 MB
+     * chip-wide): U is chosen at launch so U*n*16B stays ~128B/lane
+     * without spilling the pack registers (U=4 at n<=2, 2 at n<=4,
+     * 1 at n>=5 where the peer unroll already provides the depth). */
     uint64_t i = tid;
-    for (; i + str < nv; i += 2 * str) {
-        P x0[kMaxRanks], x1[kMaxRanks];
-#pragma unroll
-        for (int s = 0; s < kMaxRanks; s++) {
-            if (s < n) {
-                const P *src =
-                    (const P *)((const uint8_t *)a.peer_in[s] + a.sl_b);
-                x0[s] = src[i];
-                x1[s] = src[i + str];
-            }
-        }
-        A r0[VEC], r1[VEC];
-#pragma unroll
-        for (int k = 0; k < VEC; k++) {
-            r0[k] = Cvt<T>::load(x0[0].v[k]);
-            r1[k] = Cvt<T>::load(x1[0].v[k]);
-        }
-#pragma unroll
-        for (int s = 1; s < kMaxRanks; s++) {
-            if (s < n) {
-#pragma unroll
-                for (int k = 0; k < VEC; k++) {
-                    r0[k] = red<A, OP>(r0[k], Cvt<T>::load(x0[s].v[k]));
-                    r1[k] = red<A, OP>(r1[k], Cvt<T>::load(x1[s].v[k]));
-                }
-            }
-        }
-        P o0, o1;
-#pragma unroll
-        for (int k = 0; k < VEC; k++) {
-            o0.v[k] = Cvt<T>::store(apply_alpha<A>(r0[k], a.alpha));
-            o1.v[k] = Cvt<T>::store(apply_alpha<A>(r1[k], a.alpha));
-        }
-        if (a.zc_write) {
-            /* write my reduced slice straight into every rank's dst */
+    if (U > 1) {
+        for (; i + (U - 1) * str < nv; i += U * str) {
+            P x[U][kMaxRanks];
 #pragma unroll
             for (int s = 0; s < kMaxRanks; s++) {
                 if (s < n) {
-                    ((P *)a.peer_out[s])[i]       = o0;
-                    ((P *)a.peer_out[s])[i + str] = o1;
+                    const P *src =
+                        (const P *)((const uint8_t *)a.peer_in[s] +
+                                    a.sl_b);
+#pragma unroll
+                    for (int u = 0; u < U; u++) {
+                        x[u][s] = src[i + (uint64_t)u * str];
+                    }
                 }
             }
-        } else {
-            out[i]       = o0;
-            out[i + str] = o1;
+            A r[U][VEC];
+#pragma unroll
+            for (int u = 0; u < U; u++) {
+#pragma unroll
+                for (int k = 0; k < VEC; k++) {
+                    r[u][k] = Cvt<T>::load(x[u][0].v[k]);
+                }
+            }
+#pragma unroll
+            for (int s = 1; s < kMaxRanks; s++) {
+                if (s < n) {
+#pragma unroll
+                    for (int u = 0; u < U; u++) {
+#pragma unroll
+                        for (int k = 0; k < VEC; k++) {
+                            r[u][k] = red<A, OP>(
+                                r[u][k], Cvt<T>::load(x[u][s].v[k]));
+                        }
+                    }
+                }
+            }
+            P o[U];
+#pragma unroll
+            for (int u = 0; u < U; u++) {
+#pragma unroll
+                for (int k = 0; k < VEC; k++) {
+                    o[u].v[k] =
+                        Cvt<T>::store(apply_alpha<A>(r[u][k], a.alpha));
+                }
+            }
+            if (a.zc_write) {
+                /* write my reduced slice into every rank's dst */
+#pragma unroll
+                for (int s = 0; s < kMaxRanks; s++) {
+                    if (s < n) {
+#pragma unroll
+                        for (int u = 0; u < U; u++) {
+                            ((P *)a.peer_out[s])[i + (uint64_t)u * str] =
+                                o[u];
+                        }
+                    }
+                }
+            } else {
+#pragma unroll
+                for (int u = 0; u < U; u++) {
+                    out[i + (uint64_t)u * str] = o[u];
+                }
+            }
         }
     }
     for (; i < nv; i += str) {
@@ -1226,8 +1247,18 @@ template <typename T, int OP>
 static ucc_status_t launch_staged_reduce(const GatedArgs &a, hipStream_t s)
 {
     constexpr int VEC = VecOf<T>::value;
-    hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC>), dim3(gated_grid(a)),
-                       dim3(256), 0, s, a);
+    /* batching depth by rank count (bytes-in-flight per lane vs pack
+     * register budget — see the kernel comment) */
+    if (a.nranks <= 2) {
+        hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC, 4>),
+                           dim3(gated_grid(a)), dim3(256), 0, s, a);
+    } else if (a.nranks <= 4) {
+        hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC, 2>),
+                           dim3(gated_grid(a)), dim3(256), 0, s, a);
+    } else {
+        hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC, 1>),
+                           dim3(gated_grid(a)), dim3(256), 0, s, a);
+    }
     return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
 }
 
