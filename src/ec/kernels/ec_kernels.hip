@@ -1248,14 +1248,27 @@ static ucc_status_t launch_staged_reduce(const GatedArgs &a, hipStream_t s)
 {
     constexpr int VEC = VecOf<T>::value;
     /* batching depth by rank count (bytes-in-flight per lane vs pack
-     * register budget — see the kernel comment) */
-    if (a.nranks <= 2) {
+     * register budget — see the kernel comment); UCC_EC_REDUCE_DEPTH
+     * overrides for tuning (0 = auto) */
+    static int ov = [] {
+        const char *e = getenv("UCC_EC_REDUCE_DEPTH");
+        return e ? atoi(e) : 0;
+    }();
+    int depth = ov ? ov : (a.nranks <= 2 ? 4 : a.nranks <= 4 ? 2 : 1);
+    switch (depth) {
+    case 8:
+        hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC, 8>),
+                           dim3(gated_grid(a)), dim3(256), 0, s, a);
+        break;
+    case 4:
         hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC, 4>),
                            dim3(gated_grid(a)), dim3(256), 0, s, a);
-    } else if (a.nranks <= 4) {
+        break;
+    case 2:
         hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC, 2>),
                            dim3(gated_grid(a)), dim3(256), 0, s, a);
-    } else {
+        break;
+    default:
         hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC, 1>),
                            dim3(gated_grid(a)), dim3(256), 0, s, a);
     }
