@@ -1660,6 +1660,77 @@ class TcpReduceDbtTask final : public TcpTask {
     bool reda_ = false, redb_ = false, rootrecv_ = false;
 };
 
+/* ---- Bruck allgather (reference tl/ucp allgather bruck +
+ * coll_patterns/bruck role, re-derived): ceil(log2 n) rounds; after
+ * round k each rank owns 2^k consecutive blocks starting from its own,
+ * sends them to (me - 2^k) and receives the next run from (me + 2^k);
+ * a final local rotation lands blocks at their team positions. The
+ * latency-optimal allgather for small blocks (vs the ring's n-1
+ * rounds); any n. */
+class TcpAllgatherBruckTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        dtsz_  = ucc_dt_size(a_.dst.info.datatype);
+        total_ = a_.dst.info.count * dtsz_;
+        blk_   = total_ / n_;
+        dst_   = (uint8_t *)a_.dst.info.buffer;
+        if (blk_ == 0 || blk_ * n_ != total_) {
+            return UCC_ERR_NOT_SUPPORTED; /* ragged: ring handles it */
+        }
+        work_.resize(total_);
+        memcpy(work_.data(),
+               inplace ? dst_ + me_ * blk_
+                       : (const uint8_t *)a_.src.info.buffer,
+               blk_);
+        have_  = 1;
+        round_ = 0;
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    ucc_status_t progress_()
+    {
+        while (have_ < n_) {
+            if (phase_ == 0) {
+                uint64_t cnt = have_ < n_ - have_ ? have_ : n_ - have_;
+                uint32_t to   = (me_ + n_ - (uint32_t)have_) % n_;
+                uint32_t from = (me_ + (uint32_t)have_) % n_;
+                send_to(to, (uint32_t)round_, work_.data(), cnt * blk_);
+                recv_from(from, (uint32_t)round_,
+                          work_.data() + have_ * blk_, cnt * blk_);
+                phase_ = 1;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            uint64_t cnt = have_ < n_ - have_ ? have_ : n_ - have_;
+            have_ += cnt;
+            round_++;
+            phase_ = 0;
+        }
+        /* local rotation: work[j] is block (me + j) mod n */
+        for (uint32_t j = 0; j < n_; j++) {
+            memcpy(dst_ + ((me_ + j) % n_) * blk_,
+                   work_.data() + (uint64_t)j * blk_, blk_);
+        }
+        return UCC_OK;
+    }
+
+    size_t               dtsz_ = 4;
+    uint64_t             total_ = 0, blk_ = 0, have_ = 1;
+    uint8_t             *dst_ = nullptr;
+    std::vector<uint8_t> work_;
+};
+
 /* ---- barrier / fanin / fanout: binomial fanin to 0 then fanout */
 class TcpBarrierTask final : public TcpTask {
   public:
@@ -2592,6 +2663,26 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
     add(UCC_COLL_TYPE_BARRIER, mk((TcpBarrierTask *)nullptr));
     add(UCC_COLL_TYPE_FANIN, mk((TcpBarrierTask *)nullptr));
     add(UCC_COLL_TYPE_FANOUT, mk((TcpBarrierTask *)nullptr));
+    {
+        /* Bruck allgather: log-latency for small blocks */
+        size_t bmax = Config::instance().get_size(
+            "TL_TCP", "AG_BRUCK_MAX", 64 * 1024);
+        ScoreRange r;
+        r.start    = 0;
+        r.end      = bmax;
+        r.score    = sc + 1;
+        r.tl_name  = "tcp";
+        r.alg_name = "bruck";
+        r.init     = [self](const ucc_coll_args_t &args, Team *t2,
+                        Task **task) -> ucc_status_t {
+            if (!ucc_dt_is_predefined(args.dst.info.datatype)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpAllgatherBruckTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_ALLGATHER, UCC_MEMORY_TYPE_HOST, r);
+    }
     add(UCC_COLL_TYPE_ALLGATHER, mk((TcpAllgatherTask *)nullptr));
     add(UCC_COLL_TYPE_ALLGATHERV, mk((TcpAllgatherTask *)nullptr));
     add(UCC_COLL_TYPE_ALLTOALL, mk((TcpAlltoallTask *)nullptr));
