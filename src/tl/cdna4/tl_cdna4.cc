@@ -1678,6 +1678,17 @@ class GatedCollTask final : public Cdna4Task {
             close_slot();
             begin_use(); /* fresh slot use for the collective itself */
             phase_ = 0;
+            /* CE a2av (reference alltoallv_ce.c role): symmetric
+             * decision from the GLOBAL max pair length — every rank
+             * computes the same verdict. The zc exchange then maps
+             * peers' src buffers and their send tables so each rank
+             * PULLS its column with SDMA memcpys. */
+            if (!zc_ready_ && tt_->cfg_.ce_alltoall &&
+                Config::instance().get_bool("TL_CDNA4", "ZCOPY", true) &&
+                total_ * n_ >= tt_->cfg_.ce_alltoall_min) {
+                zc_    = true;
+                phase_ = 10;
+            }
         }
         if (phase_ == 10) { /* publish my src handle */
             if (!all_ge(0)) {
@@ -1753,7 +1764,10 @@ class GatedCollTask final : public Cdna4Task {
                 b.pid     = (int32_t)getpid();
                 hipDeviceptr_t dbase  = nullptr;
                 size_t         dbsize = 0;
-                if (hipMemGetAddressRange(&dbase, &dbsize,
+                if (ct_ == UCC_COLL_TYPE_ALLTOALLV) {
+                    /* CE a2av pulls from peers' src into LOCAL dst:
+                     * no dst export needed */
+                } else if (hipMemGetAddressRange(&dbase, &dbsize,
                                           (hipDeviceptr_t)dbuf_) ==
                         hipSuccess &&
                     hipIpcGetMemHandle(&b.hd, (void *)dbase) ==
@@ -1770,6 +1784,20 @@ class GatedCollTask final : public Cdna4Task {
                           hipMemcpyHostToDevice) != hipSuccess) {
                 return UCC_ERR_NO_RESOURCE;
             }
+            if (ct_ == UCC_COLL_TYPE_ALLTOALLV) {
+                /* publish my send table (byte counts + displs) after
+                 * the blob so peers can pull their columns */
+                uint64_t tbl[2 * ec_hip::kMaxRanks] = {};
+                for (uint32_t r = 0; r < n_; r++) {
+                    tbl[r]                      = cnt_[r];
+                    tbl[ec_hip::kMaxRanks + r] = dsp_[r];
+                }
+                if (hipMemcpy(tt_->area(me_, slot_, 0, 0) + 256, tbl,
+                              sizeof(tbl),
+                              hipMemcpyHostToDevice) != hipSuccess) {
+                    return UCC_ERR_NO_RESOURCE;
+                }
+            }
             publish(1);
             phase_ = 11;
         }
@@ -1777,7 +1805,19 @@ class GatedCollTask final : public Cdna4Task {
             if (!all_ge(1)) {
                 return UCC_INPROGRESS;
             }
+            const bool a2av = ct_ == UCC_COLL_TYPE_ALLTOALLV;
             for (uint32_t r = 0; r < n_ && zc_; r++) {
+                if (a2av) { /* read peer r's send table */
+                    uint64_t tbl[2 * ec_hip::kMaxRanks];
+                    if (hipMemcpy(tbl, tt_->area(r, slot_, 0, 0) + 256,
+                                  sizeof(tbl),
+                                  hipMemcpyDeviceToHost) != hipSuccess) {
+                        zc_ = false;
+                        break;
+                    }
+                    zc_scol_[r] = tbl[me_];
+                    zc_sdsp_[r] = tbl[ec_hip::kMaxRanks + me_];
+                }
                 if (r == me_) {
                     zc_peer_src_[r] = sbuf_;
                     zc_peer_dst_[r] = dbuf_;
@@ -1806,6 +1846,9 @@ class GatedCollTask final : public Cdna4Task {
                     break;
                 }
                 zc_peer_src_[r] = (const uint8_t *)m + b.base_off;
+                if (a2av) {
+                    continue; /* pulls land in LOCAL dst */
+                }
                 md              = tt_->ipc_open_cached(
                     b.pid, b.d_raw_ptr - b.d_base_off, b.hd);
                 if (md) {
@@ -1823,8 +1866,9 @@ class GatedCollTask final : public Cdna4Task {
                 }
             }
             /* direct vector loads/stores need 16B-aligned user
-             * buffers on every rank */
-            if (zc_) {
+             * buffers on every rank (CE a2av moves bytes with
+             * memcpy: exempt) */
+            if (zc_ && !a2av) {
                 for (uint32_t r = 0; r < n_; r++) {
                     if (((uintptr_t)zc_peer_src_[r] & 15) ||
                         ((uintptr_t)zc_peer_dst_[r] & 15)) {
@@ -1985,6 +2029,83 @@ class GatedCollTask final : public Cdna4Task {
         return st;
     }
 
+    /* CE alltoallv: identical gating to enqueue_ce_a2a; per-peer pull
+     * lengths come from the exchanged send tables (len clipped to my
+     * recv column, matching the gated a2av semantics). */
+    ucc_status_t enqueue_ce_a2av(hipStream_t comp_s)
+    {
+        auto *ctx        = (Cdna4TlContext *)tt_->tlc_;
+        auto &L          = tt_->gated_launch_;
+        const int nblk   = tt_->cfg_.gated_blocks
+                               ? tt_->cfg_.gated_blocks
+                               : ec_hip::kGatedBlocks;
+        const uint64_t B = (uint64_t)nblk;
+        const uint32_t p = 0;
+        ec_hip::GatedArgs ga{};
+        ga.local_flags = tt_->flags_;
+        ga.error_word  = tt_->err_host_;
+        ga.spin_limit  = tt_->spin_limit();
+        ga.nblocks     = nblk;
+        ga.pull_wait =
+            !Config::instance().get_bool("TL_CDNA4", "PUSH", true);
+        ga.rank   = (int)me_;
+        ga.nranks = (int)n_;
+        ga.slot   = (int)slot_;
+        ga.parity = (int)p;
+        for (uint32_t r = 0; r < n_; r++) {
+            ga.peer_flags[r] = tt_->peers_[r].flags;
+        }
+        ga.len         = 0;
+        ga.n_cells     = 0;
+        ga.t_sw_reduce  = L[1][slot_][p];
+        ga.t_sw_gather  = L[2][slot_][p];
+        ga.t_sig_stage  = L[0][slot_][p] + B;
+        ga.t_sig_gather = L[2][slot_][p] + B;
+        ucc_status_t st = ec_hip::staged_stage(ga, comp_s);
+        L[0][slot_][p] += B;
+        if (st != UCC_OK) {
+            return st;
+        }
+        ga.gw_phase      = 0;
+        ga.t_gather_wait = L[0][slot_][p];
+        st = ec_hip::gated_wait_only(ga, comp_s);
+        if (st != UCC_OK) {
+            return st;
+        }
+        HIPCHK(hipEventRecord(ev(8), comp_s));
+        bool used[Cdna4TlContext::kNumCe] = {};
+        for (uint32_t r = 0; r < n_; r++) {
+            uint32_t rr  = (r + me_ + 1) % n_;
+            uint64_t len = zc_scol_[rr] < rcnt_[rr] ? zc_scol_[rr]
+                                                    : rcnt_[rr];
+            if (len == 0) {
+                continue;
+            }
+            int         si = (int)(r % Cdna4TlContext::kNumCe);
+            hipStream_t cs = ctx->ce(si);
+            if (!cs) {
+                return UCC_ERR_NO_RESOURCE;
+            }
+            if (!used[si]) {
+                HIPCHK(hipStreamWaitEvent(cs, ev(8), 0));
+                used[si] = true;
+            }
+            HIPCHK(hipMemcpyAsync(dbuf_ + rdsp_[rr],
+                                  zc_peer_src_[rr] + zc_sdsp_[rr], len,
+                                  hipMemcpyDeviceToDevice, cs));
+        }
+        for (int si = 0; si < Cdna4TlContext::kNumCe; si++) {
+            if (used[si]) {
+                HIPCHK(hipEventRecord(ev(9 + si), ctx->ce(si)));
+                HIPCHK(hipStreamWaitEvent(comp_s, ev(9 + si), 0));
+            }
+        }
+        ga.t_gather_wait = L[2][slot_][p] + B;
+        st = ec_hip::gated_done(ga, comp_s);
+        L[2][slot_][p] += B;
+        return st;
+    }
+
     ucc_status_t enqueue_frags(hipStream_t stage_s, hipStream_t comp_s,
                                bool derive)
     {
@@ -1992,6 +2113,10 @@ class GatedCollTask final : public Cdna4Task {
             !derive && tt_->cfg_.ce_alltoall &&
             out_b_ * n_ >= tt_->cfg_.ce_alltoall_min) {
             return enqueue_ce_a2a(comp_s);
+        }
+        if (ct_ == UCC_COLL_TYPE_ALLTOALLV && zc_ && zc_ready_ &&
+            !derive) {
+            return enqueue_ce_a2av(comp_s);
         }
         if (zc_ && zc_ready_ && nfrags_ > 1 &&
             Config::instance().get_bool("TL_CDNA4", "ZC_DEFRAG", true)) {
@@ -2292,6 +2417,9 @@ class GatedCollTask final : public Cdna4Task {
     bool               zc_ = false, zc_ready_ = false;
     const uint8_t     *zc_peer_src_[ec_hip::kMaxRanks] = {};
     uint8_t           *zc_peer_dst_[ec_hip::kMaxRanks] = {};
+    /* CE a2av: peer r's send-to-me byte count / displacement */
+    uint64_t           zc_scol_[ec_hip::kMaxRanks] = {};
+    uint64_t           zc_sdsp_[ec_hip::kMaxRanks] = {};
 
   public:
     ~GatedCollTask() override

@@ -94,3 +94,17 @@ def test_xproc_multi_device():
     sys.stderr.write(proc.stderr[-3000:])
     assert proc.returncode == 0
     assert proc.stdout.count("XPROC_OK") == min(ndev, 8)
+
+
+def test_xproc_ce_alltoall_paths():
+    """Force every alltoall/alltoallv through the SDMA copy-engine path
+    (threshold floored): the whole xproc suite's a2a/a2av assertions
+    then validate the CE gating + pulled-table correctness."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    proc = _torchrun([os.path.join(REPO, "tests", "xproc_worker.py")],
+                     env_extra={"UCC_TL_CDNA4_CE_ALLTOALL_MIN": "4096"})
+    sys.stdout.write(proc.stdout[-3000:])
+    sys.stderr.write(proc.stderr[-3000:])
+    assert proc.returncode == 0
+    assert proc.stdout.count("XPROC_OK") == 2
