@@ -1266,7 +1266,6 @@ class TcpBcastKnomialTask final : public TcpTask {
                 while ((vr_ / q) % k_ == 0) {
                     q *= k_;
                 }
-                plevel_ = q * k_; /* children live strictly below q*k */
                 uint32_t digit  = (uint32_t)((vr_ / q) % k_);
                 uint32_t parent = vr_ - digit * (uint32_t)q;
                 recv_from(to_team(parent), 0, buf_, bytes_);
