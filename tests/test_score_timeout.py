@@ -475,7 +475,8 @@ def test_config_file_ini(tmp_path):
         "from ucc_amd.testing import LocalJob\n"
         "job = LocalJob(2)\n"
         "smap = core().score_map_str(job.teams[0])\n"
-        "line = [l for l in smap.splitlines() if 'bruck' in l][0]\n"
+        "line = [l for l in smap.splitlines()\n"
+            "        if 'bruck' in l and l.startswith('alltoall:')][0]\n"
         "assert '0-1234' in line, line  # ini-driven range\n"
         "print('INI_OK')\n" % (REPO,))
     env = dict(os.environ)
