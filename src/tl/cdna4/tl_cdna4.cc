@@ -2460,6 +2460,11 @@ class Cdna4Tl final : public Tl {
         cfg.declare("TL_CDNA4", "CE_ALLTOALL", "1",
                     "move zero-copy alltoall data on SDMA copy engines "
                     "(hipMemcpyAsync) instead of gather kernels");
+        cfg.declare("TL_CDNA4", "ZC_DEFRAG", "1",
+                    "single-fragment zero-copy collectives (debug off)");
+        cfg.declare("TL_CDNA4", "PUSH", "1",
+                    "push-model gated flags; 0 = remote-poll fallback "
+                    "(debug)");
         cfg.declare("TL_CDNA4", "CE_ALLTOALL_MIN", "192m",
                     "min total message bytes for the SDMA alltoall "
                     "(measured crossover vs the gather kernel, "

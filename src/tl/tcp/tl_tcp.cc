@@ -2469,6 +2469,20 @@ class TcpTl final : public Tl {
         auto &cfg = Config::instance();
         cfg.declare("TL_TCP", "ENABLE", "1",
                     "enable the TCP host transport (inter-node fallback)");
+        cfg.declare("TL_TCP", "KN_RADIX", "4",
+                    "k-nomial bcast tree radix");
+        cfg.declare("TL_TCP", "DBT_MIN", "8192",
+                    "double-binary-tree bcast/reduce lower bound bytes");
+        cfg.declare("TL_TCP", "DBT_MAX", "4m",
+                    "double-binary-tree bcast/reduce upper bound bytes");
+        cfg.declare("TL_TCP", "SLIDING_MIN", "64m",
+                    "sliding-window allreduce threshold bytes");
+        cfg.declare("TL_TCP", "SLIDING_WINDOW", "8m",
+                    "sliding-window allreduce window bytes");
+        cfg.declare("TL_TCP", "SLIDING_DEPTH", "2",
+                    "sliding-window allreduce windows in flight");
+        cfg.declare("TL_TCP", "AG_BRUCK_MAX", "64k",
+                    "Bruck allgather upper bound bytes");
         if (!cfg.get_bool("TL_TCP", "ENABLE", true)) {
             return nullptr;
         }
