@@ -2016,6 +2016,189 @@ class TcpAllgatherTask final : public TcpTask {
 };
 
 /* ---- alltoall(v): pairwise exchange */
+/* ---- alltoallv hybrid (reference tl/ucp alltoallv_hybrid.c role,
+ * re-derived): pairs >= thr move pairwise-direct; pairs < thr travel
+ * as [src,dest,len,bytes] envelopes through a Bruck digit exchange —
+ * at round b every rank sends ONE aggregated message (its pending
+ * envelopes whose remaining cyclic distance has bit b set) to
+ * me+2^b, collapsing up to n-1 tiny messages per rank into
+ * ceil(log2 n) aggregated ones. The dominant regime is skewed MoE
+ * dispatch where most pairs are small and a few are huge. */
+class TcpAlltoallvHybridTask final : public TcpTask {
+  public:
+    TcpAlltoallvHybridTask(Context *ctx, TcpTlTeam *tt,
+                           const ucc_coll_args_t &args, size_t thr)
+        : TcpTask(ctx, tt, args), thr_(thr)
+    {
+    }
+
+    ucc_status_t post() override
+    {
+        begin();
+        scnt_.resize(n_);
+        sdsp_.resize(n_);
+        rcnt_.resize(n_);
+        rdsp_.resize(n_);
+        size_t ss = ucc_dt_size(a_.src.info_v.datatype);
+        size_t ds = ucc_dt_size(a_.dst.info_v.datatype);
+        for (uint32_t r = 0; r < n_; r++) {
+            auto cat = [&](const void *c, uint32_t i) {
+                return (a_.flags & UCC_COLL_ARGS_FLAG_COUNT_64BIT)
+                           ? (size_t)((const uint64_t *)c)[i]
+                           : (size_t)((const uint32_t *)c)[i];
+            };
+            auto dat = [&](const void *d, uint32_t i) {
+                return (a_.flags &
+                        UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT)
+                           ? (size_t)((const uint64_t *)d)[i]
+                           : (size_t)((const uint32_t *)d)[i];
+            };
+            scnt_[r] = cat(a_.src.info_v.counts, r) * ss;
+            sdsp_[r] = dat(a_.src.info_v.displacements, r) * ss;
+            rcnt_[r] = cat(a_.dst.info_v.counts, r) * ds;
+            rdsp_[r] = dat(a_.dst.info_v.displacements, r) * ds;
+        }
+        sbuf_ = (const uint8_t *)a_.src.info_v.buffer;
+        dbuf_ = (uint8_t *)a_.dst.info_v.buffer;
+        ec_cpu::copy(dbuf_ + rdsp_[me_], sbuf_ + sdsp_[me_],
+                     std::min(scnt_[me_], rcnt_[me_]));
+        /* large pairs: direct, one message per pair (step tag 1) */
+        for (uint32_t r = 1; r < n_; r++) {
+            uint32_t to   = (me_ + r) % n_;
+            uint32_t from = (me_ + n_ - r) % n_;
+            if (scnt_[to] >= thr_ && scnt_[to]) {
+                send_to(to, 1, sbuf_ + sdsp_[to], scnt_[to]);
+            }
+            if (rcnt_[from] >= thr_ && rcnt_[from]) {
+                recv_from(from, 1, dbuf_ + rdsp_[from], rcnt_[from]);
+            }
+        }
+        /* small pairs become envelopes */
+        pend_.clear();
+        for (uint32_t r = 1; r < n_; r++) {
+            uint32_t d = (me_ + r) % n_;
+            if (scnt_[d] < thr_ && scnt_[d]) {
+                Env e;
+                e.src  = me_;
+                e.dest = d;
+                e.data.assign(sbuf_ + sdsp_[d],
+                              sbuf_ + sdsp_[d] + scnt_[d]);
+                pend_.push_back(std::move(e));
+            }
+        }
+        nrounds_ = 0;
+        while ((1u << nrounds_) < n_) {
+            nrounds_++;
+        }
+        sendbufs_.assign(nrounds_, {});
+        round_ = 0;
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    struct Env {
+        uint32_t             src, dest;
+        std::vector<uint8_t> data;
+    };
+
+    void deliver(const Env &e)
+    {
+        size_t len = std::min((size_t)e.data.size(), rcnt_[e.src]);
+        memcpy(dbuf_ + rdsp_[e.src], e.data.data(), len);
+    }
+
+    ucc_status_t progress_()
+    {
+        while (round_ < (int)nrounds_) {
+            const uint32_t p    = 1u << round_;
+            const uint32_t to   = (me_ + p) % n_;
+            const uint32_t from = (me_ + n_ - p) % n_;
+            if (phase_ == 0) {
+                /* serialize pending envelopes with bit p set in their
+                 * remaining distance; keep the rest */
+                auto &sb = sendbufs_[round_];
+                sb.assign(4, 0); /* u32 count header */
+                uint32_t            cnt = 0;
+                std::vector<Env>    keep;
+                for (auto &e : pend_) {
+                    uint32_t dist = (e.dest + n_ - me_) % n_;
+                    if (dist & p) {
+                        uint32_t hdr[3] = {e.src, e.dest,
+                                           (uint32_t)e.data.size()};
+                        sb.insert(sb.end(), (uint8_t *)hdr,
+                                  (uint8_t *)hdr + sizeof(hdr));
+                        sb.insert(sb.end(), e.data.begin(),
+                                  e.data.end());
+                        cnt++;
+                    } else {
+                        keep.push_back(std::move(e));
+                    }
+                }
+                pend_ = std::move(keep);
+                memcpy(sb.data(), &cnt, 4);
+                rsend_ = send_to(to, 0x40 + (uint32_t)round_, sb.data(),
+                                 sb.size());
+                phase_ = 1;
+            }
+            /* wait my send + steal the aggregated unexpected message */
+            tt_->progress();
+            if (!rsend_->done) {
+                return UCC_INPROGRESS;
+            }
+            std::vector<uint8_t> rb;
+            if (!tt_->conns_[from].take_unexp(
+                    tt_->mktag(seq_, 0x40 + (uint32_t)round_), &rb)) {
+                return UCC_INPROGRESS;
+            }
+            uint32_t cnt = 0;
+            if (rb.size() >= 4) {
+                memcpy(&cnt, rb.data(), 4);
+            }
+            size_t off = 4;
+            for (uint32_t i = 0; i < cnt; i++) {
+                if (off + 12 > rb.size()) {
+                    return UCC_ERR_NO_MESSAGE;
+                }
+                uint32_t hdr[3];
+                memcpy(hdr, rb.data() + off, 12);
+                off += 12;
+                if (off + hdr[2] > rb.size()) {
+                    return UCC_ERR_NO_MESSAGE;
+                }
+                Env e;
+                e.src  = hdr[0];
+                e.dest = hdr[1];
+                e.data.assign(rb.data() + off, rb.data() + off + hdr[2]);
+                off += hdr[2];
+                if (e.dest == me_) {
+                    deliver(e);
+                } else {
+                    pend_.push_back(std::move(e));
+                }
+            }
+            round_++;
+            phase_ = 0;
+        }
+        if (!ops_done()) { /* direct large-pair messages */
+            return UCC_INPROGRESS;
+        }
+        clear_ops();
+        return UCC_OK;
+    }
+
+    size_t                            thr_;
+    std::vector<size_t>               scnt_, sdsp_, rcnt_, rdsp_;
+    const uint8_t                    *sbuf_ = nullptr;
+    uint8_t                          *dbuf_ = nullptr;
+    std::vector<Env>                  pend_;
+    std::vector<std::vector<uint8_t>> sendbufs_;
+    SendOp                           *rsend_   = nullptr;
+    uint32_t                          nrounds_ = 0;
+};
+
 class TcpAlltoallTask final : public TcpTask {
   public:
     using TcpTask::TcpTask;
@@ -2483,6 +2666,9 @@ class TcpTl final : public Tl {
                     "sliding-window allreduce windows in flight");
         cfg.declare("TL_TCP", "AG_BRUCK_MAX", "64k",
                     "Bruck allgather upper bound bytes");
+        cfg.declare("TL_TCP", "A2AV_HYBRID_THRESH", "4096",
+                    "alltoallv hybrid: pairs below this ride the Bruck "
+                    "digit exchange (0 disables the hybrid alg)");
         if (!cfg.get_bool("TL_TCP", "ENABLE", true)) {
             return nullptr;
         }
@@ -2724,6 +2910,32 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
                 return UCC_OK;
             };
             map.add(UCC_COLL_TYPE_ALLTOALL, UCC_MEMORY_TYPE_HOST, r);
+        }
+    }
+    {
+        /* hybrid a2av: Bruck digit-exchange for small pairs + direct
+         * pairwise for large (skewed MoE dispatch regime) */
+        size_t thr = Config::instance().get_size(
+            "TL_TCP", "A2AV_HYBRID_THRESH", 4096);
+        if (thr > 0) {
+            ScoreRange r;
+            r.start    = 0;
+            r.end      = SIZE_MAX;
+            r.score    = sc + 1;
+            r.tl_name  = "tcp";
+            r.alg_name = "hybrid";
+            r.init     = [self, thr](const ucc_coll_args_t &args, Team *t2,
+                                 Task **task) -> ucc_status_t {
+                if ((args.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) ||
+                    !ucc_dt_is_predefined(args.src.info_v.datatype) ||
+                    (args.flags & UCC_COLL_ARGS_FLAG_IN_PLACE)) {
+                    return UCC_ERR_NOT_SUPPORTED;
+                }
+                *task = new TcpAlltoallvHybridTask(t2->ctx, self, args,
+                                                   thr);
+                return UCC_OK;
+            };
+            map.add(UCC_COLL_TYPE_ALLTOALLV, UCC_MEMORY_TYPE_HOST, r);
         }
     }
     add(UCC_COLL_TYPE_ALLTOALLV, mk((TcpAlltoallTask *)nullptr));

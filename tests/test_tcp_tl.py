@@ -341,3 +341,60 @@ def test_allgather_bruck():
     sys.stdout.write(p.stdout[-500:])
     sys.stderr.write(p.stderr[-2000:])
     assert p.returncode == 0 and "AG_BRUCK_OK" in p.stdout
+
+
+def test_alltoallv_hybrid():
+    """Hybrid a2av (reference alltoallv_hybrid.c role): small pairs
+    aggregate through the Bruck digit exchange, large pairs go direct.
+    Skewed counts straddling the threshold, zero pairs, odd/even n."""
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "from ucc_amd import core, dtypes\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "for n in (2, 3, 6, 8):\n"
+        "    job = LocalJob(n)\n"
+        "    c = core()\n"
+        "    assert '@tcp/hybrid' in c.score_map_str(job.teams[0])\n"
+        "    rng = np.random.default_rng(n)\n"
+        "    # counts: mix of 0, tiny (<thr=1024B), and big (>thr)\n"
+        "    scnt = [[0 if (r + d) %% 5 == 0 else\n"
+        "             (7 + r + d if (r * d) %% 3 else 900 + 100 * d)\n"
+        "             for d in range(n)] for r in range(n)]\n"
+        "    rcnt = [[scnt[s][r] for s in range(n)]\n"
+        "            for r in range(n)]\n"
+        "    def dsp(cs):\n"
+        "        out, off = [], 3\n"
+        "        for cq in cs:\n"
+        "            out.append(off)\n"
+        "            off += cq + 2\n"
+        "        return out, off\n"
+        "    sdsp = [dsp(cq)[0] for cq in scnt]\n"
+        "    rdsp = [dsp(cq)[0] for cq in rcnt]\n"
+        "    stot = [dsp(cq)[1] for cq in scnt]\n"
+        "    rtot = [dsp(cq)[1] for cq in rcnt]\n"
+        "    srcs = [rng.standard_normal(stot[r]).astype(np.float32)\n"
+        "            for r in range(n)]\n"
+        "    dsts = [np.zeros(rtot[r], np.float32) for r in range(n)]\n"
+        "    reqs = job.coll('alltoallv', [\n"
+        "        dict(src=srcs[r].ctypes.data, dst=dsts[r].ctypes.data,\n"
+        "             count=0, dt=dtypes.FLOAT32, src_counts=scnt[r],\n"
+        "             src_displs=sdsp[r], dst_counts=rcnt[r],\n"
+        "             dst_displs=rdsp[r]) for r in range(n)])\n"
+        "    job.run(reqs)\n"
+        "    for r in range(n):\n"
+        "        for s in range(n):\n"
+        "            np.testing.assert_array_equal(\n"
+        "                dsts[r][rdsp[r][s]:rdsp[r][s] + rcnt[r][s]],\n"
+        "                srcs[s][sdsp[s][r]:sdsp[s][r] + scnt[s][r]])\n"
+        "print('HYBRID_OK')\n"
+    ) % (REPO,)
+    env = dict(os.environ)
+    env["UCC_TL_SHM_ENABLE"] = "0"
+    env["UCC_TL_TCP_A2AV_HYBRID_THRESH"] = "1024"
+    env["UCC_TUNE"] = "alltoallv:@hybrid:99"
+    p = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=300)
+    sys.stdout.write(p.stdout[-500:])
+    sys.stderr.write(p.stderr[-3000:])
+    assert p.returncode == 0 and "HYBRID_OK" in p.stdout
