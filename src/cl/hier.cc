@@ -29,6 +29,8 @@
 
 #include "../core/core.h"
 #include "../ec/ec_cpu.h"
+#include "../ec/ec_hip.h"
+#include "../mc/mc.h"
 #include "../topo/topo.h"
 
 namespace ucc {
@@ -597,18 +599,46 @@ class HierAllreducePipeTask final : public PipelineTask {
         dt_      = a_.dst.info.datatype;
         dtsz_    = ucc_dt_size(dt_);
         inplace_ = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        dev_     = mc::is_device_mt(a_.dst.info.mem_type);
         fc_      = frag_elems ? frag_elems : count_;
+        if (fc_ == 0 || fc_ > count_) {
+            fc_ = count_ ? count_ : 1;
+        }
         n_frags  = (count_ + fc_ - 1) / fc_;
         n_stages = a_.op == UCC_OP_AVG ? 4 : 3;
         pdepth   = depth ? depth : 2;
         stage_post = [this](size_t f, size_t s, ucc_coll_req_h *r) {
             return do_stage(f, s, r);
         };
+        /* device memory: the inter-node leader phase runs on HOST
+         * buffers (tcp leaders team) — one staging buffer per in-flight
+         * fragment, D2H before the leader allreduce and H2D after
+         * (stage_done hook). Node phases stay on device (cdna4). */
+        if (dev_) {
+            stage_done = [this](size_t f, size_t s) -> ucc_status_t {
+                if (s == 1 && leader_) {
+                    size_t b   = f * fc_;
+                    size_t cnt = count_ - b < fc_ ? count_ - b : fc_;
+                    return mc::copy((uint8_t *)a_.dst.info.buffer +
+                                        b * dtsz_,
+                                    a_.dst.info.mem_type,
+                                    hstg_[f % pdepth].data(),
+                                    UCC_MEMORY_TYPE_HOST, cnt * dtsz_);
+                }
+                return UCC_OK;
+            };
+        }
     }
 
     ucc_status_t post() override
     {
         leader_ = team_->leaders_oob->my_idx >= 0;
+        if (dev_ && leader_ && hstg_.empty()) {
+            hstg_.resize(pdepth);
+            for (auto &v : hstg_) {
+                v.resize(fc_ * dtsz_);
+            }
+        }
         return PipelineTask::post();
     }
 
@@ -634,7 +664,8 @@ class HierAllreducePipeTask final : public PipelineTask {
         case 0: /* node reduce of the slice to the node leader */
             if (team_->node_team->size == 1) {
                 if (!inplace_) {
-                    memcpy(dstp, srcp, cnt * dtsz_);
+                    return mc::copy(dstp, a_.dst.info.mem_type, srcp,
+                                    a_.dst.info.mem_type, cnt * dtsz_);
                 }
                 return UCC_OK;
             }
@@ -658,6 +689,18 @@ class HierAllreducePipeTask final : public PipelineTask {
             sa.src.info        = a_.dst.info;
             sa.src.info.buffer = dstp;
             sa.src.info.count  = cnt;
+            if (dev_) {
+                /* inter-node over host transports: stage D2H */
+                uint8_t *h = hstg_[f % pdepth].data();
+                ucc_status_t cs =
+                    mc::copy(h, UCC_MEMORY_TYPE_HOST, dstp,
+                             a_.dst.info.mem_type, cnt * dtsz_);
+                if (cs != UCC_OK) {
+                    return cs;
+                }
+                sa.src.info.buffer   = h;
+                sa.src.info.mem_type = UCC_MEMORY_TYPE_HOST;
+            }
             sa.dst.info        = sa.src.info;
             st_team            = team_->leaders_team.get();
             break;
@@ -672,6 +715,21 @@ class HierAllreducePipeTask final : public PipelineTask {
             st_team            = team_->node_team.get();
             break;
         case 3: { /* AVG: scale the completed slice by 1/N */
+            if (dev_) {
+                ec_hip::ReduceArgs ra{};
+                ra.dst     = dstp;
+                ra.srcs[0] = dstp;
+                ra.n_srcs  = 1;
+                ra.count   = cnt;
+                ra.dt      = dt_;
+                ra.op      = (ucc_reduction_op_t)12; /* SUM w/ alpha */
+                ra.alpha   = 1.0f / (float)team_->size;
+                ucc_status_t rs = ec_hip::reduce(ra, nullptr);
+                if (rs != UCC_OK) {
+                    return rs;
+                }
+                return mc::device_sync();
+            }
             const void *sp[1] = {dstp};
             ec_cpu::reduce(dstp, sp, 1, cnt, dt_, UCC_OP_SUM,
                            1.0 / (double)team_->size);
@@ -690,10 +748,11 @@ class HierAllreducePipeTask final : public PipelineTask {
 
     Team           *team_;
     ucc_coll_args_t a_;
-    bool            leader_ = false, inplace_ = false;
+    bool            leader_ = false, inplace_ = false, dev_ = false;
     uint64_t        count_ = 0;
     size_t          fc_ = 0, dtsz_ = 4;
     ucc_datatype_t  dt_ = UCC_DT_FLOAT32;
+    std::vector<std::vector<uint8_t>> hstg_;
 };
 
 /* ---- hier allreduce (split_rail role): node reduce_scatterv ->
@@ -2034,6 +2093,42 @@ void add_scores(Team *team)
         return UCC_OK;
     };
     team->score_map.add(UCC_COLL_TYPE_ALLREDUCE, UCC_MEMORY_TYPE_HOST, r);
+
+    {
+        /* device-memory hier allreduce: node phases run on the device
+         * TL (cdna4); the inter-node leader phase stages through host
+         * buffers over the host transports (the multi-node GPU
+         * composition the reference reaches via tl/ucp GPU-aware UCX —
+         * re-derived as explicit D2H/H2D staging in the pipeline). */
+        ScoreRange dr;
+        dr.start    = 0;
+        dr.end      = SIZE_MAX;
+        dr.score    = 60;
+        dr.tl_name  = "hier";
+        dr.alg_name = "rab_dev";
+        dr.init     = [](const ucc_coll_args_t &args, Team *t,
+                     Task **task) -> ucc_status_t {
+            if (args.op != UCC_OP_SUM && args.op != UCC_OP_MAX &&
+                args.op != UCC_OP_MIN && args.op != UCC_OP_PROD &&
+                args.op != UCC_OP_AVG) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            auto  &cfg = Config::instance();
+            size_t fb  = cfg.get_size("CL_HIER", "FRAG_SIZE",
+                                      4 * 1024 * 1024);
+            size_t dtsz = ucc_dt_size(args.dst.info.datatype);
+            if (dtsz == 0) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            size_t depth = (size_t)cfg.get_int(
+                "CL_HIER", "PIPELINE_DEPTH", 2);
+            *task = new HierAllreducePipeTask(
+                t->ctx, t, args, fb ? fb / dtsz : 0, depth);
+            return UCC_OK;
+        };
+        team->score_map.add(UCC_COLL_TYPE_ALLREDUCE,
+                            UCC_MEMORY_TYPE_CUDA, dr);
+    }
 
     if (team->rails_ok && team->rail_team) {
         /* split_rail beats RAB on large vectors: every rank carries

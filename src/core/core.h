@@ -139,6 +139,9 @@ class PipelineTask : public Task {
     size_t    n_stages = 1;
     size_t    pdepth   = 2;
     StagePost stage_post;
+    /* optional epilogue run when (frag, stage) completes — e.g. the
+     * H2D copy-back after a host-staged leader phase */
+    std::function<ucc_status_t(size_t frag, size_t stage)> stage_done;
     /* optional observer for tests/tracing: ev 'P' = stage posted,
      * 'C' = stage completed */
     std::function<void(char ev, size_t frag, size_t stage)> trace;

@@ -45,6 +45,9 @@ ucc_status_t ipc_export(const void *ptr, void *handle_out,
 ucc_status_t ipc_import(const void *handle, void **mapped);
 ucc_status_t ipc_close(void *mapped);
 
+/* Synchronize the device's default stream (no-op without a device). */
+ucc_status_t device_sync();
+
 /* HIP event helpers for the EE completion-event flow (gated: no-ops
  * without a device). ev handles are opaque (hipEvent_t). */
 ucc_status_t stream_event_record(void *stream, void **ev_out);

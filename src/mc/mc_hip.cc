@@ -151,6 +151,16 @@ size_t scratch_raw_allocs()
     return g_raw_allocs.load(std::memory_order_relaxed);
 }
 
+ucc_status_t device_sync()
+{
+    if (!hip_available()) {
+        return UCC_OK;
+    }
+    return hipStreamSynchronize(nullptr) == hipSuccess
+               ? UCC_OK
+               : UCC_ERR_NO_RESOURCE;
+}
+
 ucc_status_t stream_event_record(void *stream, void **ev_out)
 {
     if (!hip_available()) {

@@ -198,6 +198,12 @@ ucc_status_t PipelineTask::drive()
             if (st != UCC_OK) {
                 return st;
             }
+            if (stage_done) {
+                st = stage_done(f.frag, f.stage);
+                if (st != UCC_OK) {
+                    return st;
+                }
+            }
             if (trace) {
                 trace('C', f.frag, f.stage);
             }

@@ -364,3 +364,66 @@ def test_hier_pipelined_allreduce_interleaves(tmp_path, alg):
         assert max_open <= 2, f"pdepth exceeded: {max_open}"
         checked += 1
     assert checked >= 1, "no multi-fragment pipeline task traced"
+
+
+DEVICE_WORKER = r"""
+import sys
+import numpy as np
+sys.path.insert(0, %r)
+import torch
+from ucc_amd import core, dtypes
+from ucc_amd.testing import LocalJob
+
+torch.cuda.set_device(0)
+n = 4  # 2 pseudo-nodes x 2 ranks
+job = LocalJob(n)
+c = core()
+smap = c.score_map_str(job.teams[0])
+assert "@hier/rab_dev" in smap, smap
+
+torch.manual_seed(77)
+# multi-fragment device RAB: node phases on cdna4, leader phase staged
+# D2H over the host transports
+for count in (5000, 3_000_000):
+    srcs = [torch.randn(count, device="cuda") for _ in range(n)]
+    dsts = [torch.zeros(count, device="cuda") for _ in range(n)]
+    exp = sum(s.cpu() for s in srcs)
+    reqs = job.coll("allreduce", [
+        dict(src=srcs[r].data_ptr(), dst=dsts[r].data_ptr(),
+             count=count, dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA)
+        for r in range(n)])
+    job.run(reqs)
+    torch.cuda.synchronize()
+    for d in dsts:
+        torch.testing.assert_close(d.cpu(), exp, rtol=1e-5, atol=1e-4)
+
+# AVG on device (stage-3 device scale)
+srcs = [torch.full((40_000,), float(r + 1), device="cuda")
+        for r in range(n)]
+dsts = [torch.zeros(40_000, device="cuda") for _ in range(n)]
+exp = torch.full((40_000,), sum(range(1, n + 1)) / n)
+reqs = job.coll("allreduce", [
+    dict(src=srcs[r].data_ptr(), dst=dsts[r].data_ptr(), count=40_000,
+         dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA, op=dtypes.OP_AVG)
+    for r in range(n)])
+job.run(reqs)
+torch.cuda.synchronize()
+for d in dsts:
+    torch.testing.assert_close(d.cpu(), exp, rtol=1e-5, atol=1e-5)
+print("HIER_DEV_OK")
+""" % (REPO,)
+
+
+@pytest.mark.gpu
+def test_hier_device_allreduce_fake_nodes():
+    """Device-memory hier RAB: node phases over the cdna4 TL, leader
+    phase staged D2H over host transports (multi-node GPU composition,
+    testable on one box via UCC_FAKE_NODE_SPLIT)."""
+    env = dict(os.environ)
+    env.update({"UCC_FAKE_NODE_SPLIT": "2",
+                "UCC_CL_HIER_FRAG_SIZE": "1048576"})
+    p = subprocess.run([sys.executable, "-c", DEVICE_WORKER], env=env,
+                       capture_output=True, text=True, timeout=300)
+    sys.stdout.write(p.stdout[-2000:])
+    sys.stderr.write(p.stderr[-3000:])
+    assert p.returncode == 0 and "HIER_DEV_OK" in p.stdout
