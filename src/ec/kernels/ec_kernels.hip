@@ -51,6 +51,39 @@ __device__ __forceinline__ uint16_t d_f_to_bf16(float f)
 #endif
 
 #ifdef UCC_NATIVE_FP8
+/* paired converts: one VOP handles 2 elements (v_cvt_pk_f32_fp8 /
+ * v_cvt_pk_fp8_f32) — halves the VALU cost of the fp8 reduce, which is
+ * 2x the elements per byte of bf16 */
+__device__ __forceinline__ void d_e4m3x2_to_f(uint8_t lo, uint8_t hi,
+                                              float &a, float &b)
+{
+    int packed = (int)lo | ((int)hi << 8);
+    auto v2    = __builtin_amdgcn_cvt_pk_f32_fp8(packed, false);
+    a          = v2[0];
+    b          = v2[1];
+}
+__device__ __forceinline__ void d_f_to_e4m3x2(float a, float b,
+                                              uint8_t &lo, uint8_t &hi)
+{
+    int w = __builtin_amdgcn_cvt_pk_fp8_f32(a, b, 0, false);
+    lo    = (uint8_t)w;
+    hi    = (uint8_t)(w >> 8);
+}
+__device__ __forceinline__ void d_e5m2x2_to_f(uint8_t lo, uint8_t hi,
+                                              float &a, float &b)
+{
+    int packed = (int)lo | ((int)hi << 8);
+    auto v2    = __builtin_amdgcn_cvt_pk_f32_bf8(packed, false);
+    a          = v2[0];
+    b          = v2[1];
+}
+__device__ __forceinline__ void d_f_to_e5m2x2(float a, float b,
+                                              uint8_t &lo, uint8_t &hi)
+{
+    int w = __builtin_amdgcn_cvt_pk_bf8_f32(a, b, 0, false);
+    lo    = (uint8_t)w;
+    hi    = (uint8_t)(w >> 8);
+}
 __device__ __forceinline__ float d_e4m3_to_f(uint8_t v)
 {
     return __builtin_amdgcn_cvt_f32_fp8((int)v, 0);
@@ -283,6 +316,90 @@ template <typename T, int OP, int VEC>
 struct __align__(16) Pack {
     T v[VEC];
 };
+
+#ifdef UCC_NATIVE_FP8
+template <typename T>
+inline constexpr bool is_fp8_v = std::is_same<T, e4m3_t>::value ||
+                                 std::is_same<T, e5m2_t>::value;
+
+template <typename T>
+__device__ __forceinline__ void cvt2_load(T lo, T hi, float &a, float &b)
+{
+    if constexpr (std::is_same<T, e4m3_t>::value) {
+        d_e4m3x2_to_f(lo.v, hi.v, a, b);
+    } else {
+        d_e5m2x2_to_f(lo.v, hi.v, a, b);
+    }
+}
+template <typename T>
+__device__ __forceinline__ void cvt2_store(float a, float b, T &lo, T &hi)
+{
+    if constexpr (std::is_same<T, e4m3_t>::value) {
+        d_f_to_e4m3x2(a, b, lo.v, hi.v);
+    } else {
+        d_f_to_e5m2x2(a, b, lo.v, hi.v);
+    }
+}
+#else
+template <typename T> inline constexpr bool is_fp8_v = false;
+template <typename T>
+__device__ __forceinline__ void cvt2_load(T, T, float &, float &) {}
+template <typename T>
+__device__ __forceinline__ void cvt2_store(float, float, T &, T &) {}
+#endif
+
+/* vector-wide load/accumulate/store with paired fp8 fast paths */
+template <typename T, int OP, int VEC, typename A, typename P>
+__device__ __forceinline__ void vload(const P &x, A (&r)[VEC])
+{
+    if constexpr (is_fp8_v<T> && std::is_same<A, float>::value) {
+#pragma unroll
+        for (int k = 0; k < VEC; k += 2) {
+            cvt2_load<T>(x.v[k], x.v[k + 1], r[k], r[k + 1]);
+        }
+    } else {
+#pragma unroll
+        for (int k = 0; k < VEC; k++) {
+            r[k] = Cvt<T>::load(x.v[k]);
+        }
+    }
+}
+template <typename T, int OP, int VEC, typename A, typename P>
+__device__ __forceinline__ void vaccum(A (&r)[VEC], const P &x)
+{
+    if constexpr (is_fp8_v<T> && std::is_same<A, float>::value) {
+#pragma unroll
+        for (int k = 0; k < VEC; k += 2) {
+            float a, b;
+            cvt2_load<T>(x.v[k], x.v[k + 1], a, b);
+            r[k]     = red<A, OP>(r[k], a);
+            r[k + 1] = red<A, OP>(r[k + 1], b);
+        }
+    } else {
+#pragma unroll
+        for (int k = 0; k < VEC; k++) {
+            r[k] = red<A, OP>(r[k], Cvt<T>::load(x.v[k]));
+        }
+    }
+}
+template <typename T, int OP, int VEC, typename A, typename P>
+__device__ __forceinline__ void vstore(P &o, const A (&r)[VEC],
+                                       float alpha)
+{
+    if constexpr (is_fp8_v<T> && std::is_same<A, float>::value) {
+#pragma unroll
+        for (int k = 0; k < VEC; k += 2) {
+            cvt2_store<T>(apply_alpha<A>(r[k], alpha),
+                          apply_alpha<A>(r[k + 1], alpha), o.v[k],
+                          o.v[k + 1]);
+        }
+    } else {
+#pragma unroll
+        for (int k = 0; k < VEC; k++) {
+            o.v[k] = Cvt<T>::store(apply_alpha<A>(r[k], alpha));
+        }
+    }
+}
 
 template <typename T, int OP, int VEC>
 __global__ void k_reduce(ReduceArgs a)
@@ -885,32 +1002,21 @@ __global__ void k_staged_reduce(const GatedArgs a)
             A r[U][VEC];
 #pragma unroll
             for (int u = 0; u < U; u++) {
-#pragma unroll
-                for (int k = 0; k < VEC; k++) {
-                    r[u][k] = Cvt<T>::load(x[u][0].v[k]);
-                }
+                vload<T, OP, VEC, A>(x[u][0], r[u]);
             }
 #pragma unroll
             for (int s = 1; s < kMaxRanks; s++) {
                 if (s < n) {
 #pragma unroll
                     for (int u = 0; u < U; u++) {
-#pragma unroll
-                        for (int k = 0; k < VEC; k++) {
-                            r[u][k] = red<A, OP>(
-                                r[u][k], Cvt<T>::load(x[u][s].v[k]));
-                        }
+                        vaccum<T, OP, VEC, A>(r[u], x[u][s]);
                     }
                 }
             }
             P o[U];
 #pragma unroll
             for (int u = 0; u < U; u++) {
-#pragma unroll
-                for (int k = 0; k < VEC; k++) {
-                    o[u].v[k] =
-                        Cvt<T>::store(apply_alpha<A>(r[u][k], a.alpha));
-                }
+                vstore<T, OP, VEC, A>(o[u], r[u], a.alpha);
             }
             if (a.zc_write) {
                 /* write my reduced slice into every rank's dst */
@@ -937,27 +1043,18 @@ __global__ void k_staged_reduce(const GatedArgs a)
         {
             P acc = ((const P *)((const uint8_t *)a.peer_in[0] +
                                  a.sl_b))[i];
-#pragma unroll
-            for (int k = 0; k < VEC; k++) {
-                r[k] = Cvt<T>::load(acc.v[k]);
-            }
+            vload<T, OP, VEC, A>(acc, r);
         }
 #pragma unroll
         for (int s = 1; s < kMaxRanks; s++) {
             if (s < n) {
                 P x = ((const P *)((const uint8_t *)a.peer_in[s] +
                                    a.sl_b))[i];
-#pragma unroll
-                for (int k = 0; k < VEC; k++) {
-                    r[k] = red<A, OP>(r[k], Cvt<T>::load(x.v[k]));
-                }
+                vaccum<T, OP, VEC, A>(r, x);
             }
         }
         P o;
-#pragma unroll
-        for (int k = 0; k < VEC; k++) {
-            o.v[k] = Cvt<T>::store(apply_alpha<A>(r[k], a.alpha));
-        }
+        vstore<T, OP, VEC, A>(o, r, a.alpha);
         if (a.zc_write) {
 #pragma unroll
             for (int s = 0; s < kMaxRanks; s++) {
