@@ -607,6 +607,8 @@ class HierAllreducePipeTask final : public PipelineTask {
         n_frags  = (count_ + fc_ - 1) / fc_;
         n_stages = a_.op == UCC_OP_AVG ? 4 : 3;
         pdepth   = depth ? depth : 2;
+        pair_early = 0; /* node reduce and node bcast share node_team */
+        pair_late  = 2;
         stage_post = [this](size_t f, size_t s, ucc_coll_req_h *r) {
             return do_stage(f, s, r);
         };
@@ -938,6 +940,8 @@ class HierSplitRailPipeTask final : public PipelineTask {
         n_frags  = (count_ + fc_ - 1) / fc_;
         n_stages = 3;
         pdepth   = depth ? depth : 2;
+        pair_early = 0; /* node RSV and node AGV share node_team */
+        pair_late  = 2;
         stage_post = [this](size_t f, size_t s, ucc_coll_req_h *r) {
             return do_stage(f, s, r);
         };

@@ -139,6 +139,18 @@ class PipelineTask : public Task {
     size_t    n_stages = 1;
     size_t    pdepth   = 2;
     StagePost stage_post;
+    /* Two stages that post on the SAME sub-team must publish their
+     * posts in an order that is identical on every rank (sub-collective
+     * matching is per-team post-sequence based). Setting pair_early /
+     * pair_late pins the canonical interleave
+     *   E(0), E(1), L(0), E(2), L(1), ... , L(n-1)
+     * by gating L(f) on E(min(f+1, last)) having POSTED; E(f) after
+     * L(f-2) holds automatically through flight recycling. Without the
+     * pair, completion-timing differences between ranks can reorder the
+     * shared team's posts (observed: cdna4 node-team desync in the
+     * device hier RAB). */
+    size_t pair_early = SIZE_MAX;
+    size_t pair_late  = SIZE_MAX;
     /* optional epilogue run when (frag, stage) completes — e.g. the
      * H2D copy-back after a host-staged leader phase */
     std::function<ucc_status_t(size_t frag, size_t stage)> stage_done;

@@ -3,6 +3,7 @@
  * manager + ucc_schedule_t) and core/ucc_progress_queue*. */
 #include "../core/core.h"
 
+#include <algorithm>
 #include <ctime>
 
 namespace ucc {
@@ -173,6 +174,11 @@ ucc_status_t PipelineTask::drive()
             if (!f.posted) {
                 if (ord_[f.stage] != f.frag) {
                     continue; /* stage s posts in fragment order */
+                }
+                if (f.stage == pair_late &&
+                    ord_[pair_early] <
+                        std::min(f.frag + 2, n_frags)) {
+                    continue; /* canonical shared-team interleave */
                 }
                 ucc_coll_req_h r  = nullptr;
                 ucc_status_t   st = stage_post(f.frag, f.stage, &r);
