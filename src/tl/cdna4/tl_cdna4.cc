@@ -1647,27 +1647,31 @@ class GatedCollTask final : public Cdna4Task {
             if (!all_ge(0)) {
                 return UCC_INPROGRESS;
             }
-            uint64_t lmax = (uint64_t)total_;
-            if (hipMemcpy(tt_->area(me_, slot_, 0, 0), &lmax,
-                          sizeof(lmax), hipMemcpyHostToDevice) !=
-                hipSuccess) {
+            uint64_t mm[2] = {(uint64_t)total_, 0};
+            for (uint32_t r = 0; r < n_; r++) {
+                mm[1] += cnt_[r]; /* my total send bytes */
+            }
+            if (hipMemcpy(tt_->area(me_, slot_, 0, 0), mm, sizeof(mm),
+                          hipMemcpyHostToDevice) != hipSuccess) {
                 return UCC_ERR_NO_RESOURCE;
             }
             publish(1);
             phase_ = 21;
         }
-        if (phase_ == 21) { /* a2av: global max = max of all locals */
+        if (phase_ == 21) { /* a2av: global max + global byte sum */
             if (!all_ge(1)) {
                 return UCC_INPROGRESS;
             }
             uint64_t gmax = 0;
+            a2av_gsum_    = 0;
             for (uint32_t r = 0; r < n_; r++) {
-                uint64_t v = 0;
-                if (hipMemcpy(&v, tt_->area(r, slot_, 0, 0), sizeof(v),
+                uint64_t v[2] = {0, 0};
+                if (hipMemcpy(v, tt_->area(r, slot_, 0, 0), sizeof(v),
                               hipMemcpyDeviceToHost) != hipSuccess) {
                     return UCC_ERR_NO_RESOURCE;
                 }
-                gmax = v > gmax ? v : gmax;
+                gmax = v[0] > gmax ? v[0] : gmax;
+                a2av_gsum_ += v[1];
             }
             total_  = gmax;
             nfrags_ = (total_ + gran_ - 1) / gran_;
@@ -1679,13 +1683,17 @@ class GatedCollTask final : public Cdna4Task {
             begin_use(); /* fresh slot use for the collective itself */
             phase_ = 0;
             /* CE a2av (reference alltoallv_ce.c role): symmetric
-             * decision from the GLOBAL max pair length — every rank
-             * computes the same verdict. The zc exchange then maps
-             * peers' src buffers and their send tables so each rank
-             * PULLS its column with SDMA memcpys. */
+             * decision from the GLOBAL byte sum (exchanged with the
+             * max), so skewed MoE matrices — whose per-rank totals
+             * differ — still get one team-wide verdict. The zc
+             * exchange then maps peers' src buffers and their send
+             * tables so each rank PULLS its column with SDMA memcpys.
+             * SDMA also sidesteps the gated path's worst skew cost:
+             * fragment count there scales with the MAX pair length. */
             if (!zc_ready_ && tt_->cfg_.ce_alltoall &&
                 Config::instance().get_bool("TL_CDNA4", "ZCOPY", true) &&
-                total_ * n_ >= tt_->cfg_.ce_alltoall_min) {
+                a2av_gsum_ / (n_ ? n_ : 1) >=
+                    tt_->cfg_.ce_alltoall_min / 4) {
                 zc_    = true;
                 phase_ = 10;
             }
@@ -2409,6 +2417,7 @@ class GatedCollTask final : public Cdna4Task {
     std::vector<size_t> cnt_, dsp_;   /* send row bytes (ag/rs/a2av)  */
     std::vector<size_t> rcnt_, rdsp_; /* recv column bytes (a2av)     */
     bool                a2av_ready_ = false; /* global max exchanged  */
+    uint64_t            a2av_gsum_  = 0;     /* global send-byte sum  */
     int                pslot_ = -1;
     /* zero-copy persistent allreduce: peers' USER src buffers mapped
      * via HIP-IPC (handles exchanged through the scratch channel at
