@@ -297,3 +297,42 @@ def test_multiproc_moe_token_exchange():
     """SURVEY §2.9 EP workload: skewed MoE token dispatch/combine over
     alltoallv across processes."""
     _run(_worker_moe, world=3)
+
+
+def _worker_ulysses(rank, world, port, q):
+    try:
+        import torch
+        import torch.distributed as dist
+
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from ucc_amd.parallel import Communicator
+        from ucc_amd.parallel.ulysses import HeadSeqExchanger
+
+        comm = Communicator()
+        ex = HeadSeqExchanger(comm)
+        s_local, H, dh = 6, 4 * world, 8
+        # global sequence, deterministic on every rank
+        torch.manual_seed(777)
+        full = torch.randn(world * s_local, H, dh)
+        x = full[rank * s_local:(rank + 1) * s_local]
+        # seq-sharded -> head-sharded: full sequence, my head slice
+        y = ex.seq_to_heads(x.contiguous())
+        hp = H // world
+        exp = full[:, rank * hp:(rank + 1) * hp, :]
+        assert torch.equal(y, exp), "seq_to_heads mismatch"
+        # inverse restores the original shard exactly
+        back = ex.heads_to_seq(y)
+        assert torch.equal(back, x), "heads_to_seq mismatch"
+        q.put((rank, "ok"))
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"fail: {e!r}"))
+
+
+def test_multiproc_ulysses_head_seq_exchange():
+    """SURVEY §2.9 SP/Ulysses workload: the head<->sequence alltoall
+    switch for sequence-parallel attention, verified against the
+    logical relayout of a reference full tensor."""
+    _run(_worker_ulysses, world=4)
