@@ -13,17 +13,23 @@ namespace mc {
 
 int hip_device_count()
 {
-    static int count = []() {
-        int        n = 0;
-        hipError_t e = hipGetDeviceCount(&n);
-        if (e != hipSuccess) {
-            UCC_LOG(LogLevel::DEBUG, "mc", "hipGetDeviceCount: %s",
-                    hipGetErrorString(e));
-            return 0;
-        }
-        return n;
-    }();
-    return count;
+    /* cache positive answers only: a failed early probe (e.g. before
+     * the process's primary HIP runtime finished loading) must not
+     * latch the library into host-only mode forever */
+    static std::atomic<int> cached{-1};
+    int c = cached.load(std::memory_order_relaxed);
+    if (c >= 0) {
+        return c;
+    }
+    int        n = 0;
+    hipError_t e = hipGetDeviceCount(&n);
+    if (e != hipSuccess) {
+        UCC_LOG(LogLevel::DEBUG, "mc", "hipGetDeviceCount: %s",
+                hipGetErrorString(e));
+        return 0; /* not cached */
+    }
+    cached.store(n, std::memory_order_relaxed);
+    return n;
 }
 
 extern "C" int ucc_amd_hip_device_count_c() { return hip_device_count(); }
