@@ -347,6 +347,20 @@ def test_hier_pipelined_allreduce_interleaves(tmp_path, alg):
         assert i_p02 < i_cl1, (
             f"no interleave in task {tid}: P(0,2)@{i_p02} "
             f">= C({last},1)@{i_cl1}")
+        # shared-team post order must be CANONICAL (rank-deterministic):
+        # stages 0 and 2 both post on the node team; required sequence
+        # is E(0), E(1), L(0), E(2), L(1), ..., L(last)
+        node_posts = [(f, st) for ev, f, st in evs
+                      if ev == "P" and st in (0, 2)]
+        nf = last + 1
+        pos = {}
+        for i2, fs in enumerate(node_posts):
+            pos.setdefault(fs, i2)
+        for f in range(nf):
+            need = (min(f + 1, nf - 1), 0)
+            assert pos[(f, 2)] > pos[need], (
+                f"canonical interleave violated: L({f}) before "
+                f"E({need[0]}) in {node_posts}")
         # pipeline depth respected: a frag is "open" from its first post
         # until its own final-stage completion; never more than pdepth=2
         # open at once
