@@ -2212,17 +2212,29 @@ class GatedCollTask final : public Cdna4Task {
                         ga.peer_out[r] = zc_peer_dst_[r] + off + ga.sl_b;
                     }
                 }
+                /* zero-copy: stage and gather are pure signal/wait —
+                 * a full grid of spinning blocks would only steal CUs
+                 * from the reduce; the cumulative-block ledger lets
+                 * them launch with ONE block (derive mode keeps the
+                 * uniform grid its device-side arithmetic needs). */
+                const uint64_t Bs =
+                    (!derive && ga.zc_write) ? 1 : B;
+                const uint64_t Bg = Bs;
                 ga.t_sw_reduce   = L[1][slot_][p];
                 ga.t_sw_gather   = L[2][slot_][p];
                 ga.t_prev_gather = L[2][slot_][p];
-                ga.t_stage       = L[0][slot_][p] + B;
+                ga.t_stage       = L[0][slot_][p] + Bs;
                 ga.gw_phase      = 1;
                 ga.t_gather_wait = L[1][slot_][p] + B;
-                ga.t_sig_stage   = L[0][slot_][p] + B;
+                ga.t_sig_stage   = L[0][slot_][p] + Bs;
                 ga.t_sig_reduce  = L[1][slot_][p] + B;
-                ga.t_sig_gather  = L[2][slot_][p] + B;
-                st = ec_hip::staged_stage(ga, stage_s);
-                if (!derive) { L[0][slot_][p] += B; }
+                ga.t_sig_gather  = L[2][slot_][p] + Bg;
+                {
+                    ec_hip::GatedArgs gs = ga;
+                    gs.nblocks           = (int)Bs;
+                    st = ec_hip::staged_stage(gs, stage_s);
+                }
+                if (!derive) { L[0][slot_][p] += Bs; }
                 if (st == UCC_OK) {
                     st = ec_hip::staged_reduce(ga, comp_s);
                     if (!derive) { L[1][slot_][p] += B; }
@@ -2235,8 +2247,10 @@ class GatedCollTask final : public Cdna4Task {
                             ga.slice_e[r] = ga.slice_b[r];
                         }
                     }
-                    st = ec_hip::staged_gather(ga, comp_s);
-                    if (!derive) { L[2][slot_][p] += B; }
+                    ec_hip::GatedArgs gg = ga;
+                    gg.nblocks           = (int)Bg;
+                    st = ec_hip::staged_gather(gg, comp_s);
+                    if (!derive) { L[2][slot_][p] += Bg; }
                 }
                 break;
             }
@@ -2266,14 +2280,20 @@ class GatedCollTask final : public Cdna4Task {
                         ga.peer_in[r] = zc_peer_src_[r] + off;
                     }
                 }
+                const uint64_t Bs =
+                    (!derive && zc_ && zc_ready_) ? 1 : B;
                 ga.t_sw_reduce   = L[1][slot_][p];
                 ga.t_sw_gather   = L[2][slot_][p];
                 ga.t_prev_gather = 0;
-                ga.t_stage       = L[0][slot_][p] + B;
-                ga.t_sig_stage   = L[0][slot_][p] + B;
+                ga.t_stage       = L[0][slot_][p] + Bs;
+                ga.t_sig_stage   = L[0][slot_][p] + Bs;
                 ga.t_sig_reduce  = L[1][slot_][p] + B;
-                st = ec_hip::staged_stage(ga, stage_s);
-                if (!derive) { L[0][slot_][p] += B; }
+                {
+                    ec_hip::GatedArgs gs = ga;
+                    gs.nblocks           = (int)Bs;
+                    st = ec_hip::staged_stage(gs, stage_s);
+                }
+                if (!derive) { L[0][slot_][p] += Bs; }
                 if (st == UCC_OK) {
                     st = ec_hip::staged_reduce(ga, comp_s);
                     if (!derive) { L[1][slot_][p] += B; }
@@ -2306,14 +2326,20 @@ class GatedCollTask final : public Cdna4Task {
                     ga.slice_b[r] = dsp_[r] + off;
                     ga.slice_e[r] = ga.slice_b[r] + l;
                 }
+                const uint64_t Bs =
+                    (!derive && agzc) ? 1 : B;
                 ga.t_sw_reduce   = L[1][slot_][p];
                 ga.t_sw_gather   = L[2][slot_][p];
                 ga.gw_phase      = 0;
-                ga.t_gather_wait = L[0][slot_][p] + B;
-                ga.t_sig_stage   = L[0][slot_][p] + B;
+                ga.t_gather_wait = L[0][slot_][p] + Bs;
+                ga.t_sig_stage   = L[0][slot_][p] + Bs;
                 ga.t_sig_gather  = L[2][slot_][p] + B;
-                st = ec_hip::staged_stage(ga, stage_s);
-                if (!derive) { L[0][slot_][p] += B; }
+                {
+                    ec_hip::GatedArgs gs = ga;
+                    gs.nblocks           = (int)Bs;
+                    st = ec_hip::staged_stage(gs, stage_s);
+                }
+                if (!derive) { L[0][slot_][p] += Bs; }
                 if (st == UCC_OK) {
                     st = ec_hip::staged_gather(ga, comp_s);
                     if (!derive) { L[2][slot_][p] += B; }
@@ -2382,14 +2408,20 @@ class GatedCollTask final : public Cdna4Task {
                     ga.slice_b[r] = (uint64_t)r * out_b_ + off;
                     ga.slice_e[r] = ga.slice_b[r] + len;
                 }
+                const uint64_t Bs =
+                    (!derive && a2zc) ? 1 : B;
                 ga.t_sw_reduce   = L[1][slot_][p];
                 ga.t_sw_gather   = L[2][slot_][p];
                 ga.gw_phase      = 0;
-                ga.t_gather_wait = L[0][slot_][p] + B;
-                ga.t_sig_stage   = L[0][slot_][p] + B;
+                ga.t_gather_wait = L[0][slot_][p] + Bs;
+                ga.t_sig_stage   = L[0][slot_][p] + Bs;
                 ga.t_sig_gather  = L[2][slot_][p] + B;
-                st = ec_hip::staged_stage(ga, stage_s);
-                if (!derive) { L[0][slot_][p] += B; }
+                {
+                    ec_hip::GatedArgs gs = ga;
+                    gs.nblocks           = (int)Bs;
+                    st = ec_hip::staged_stage(gs, stage_s);
+                }
+                if (!derive) { L[0][slot_][p] += Bs; }
                 if (st == UCC_OK) {
                     st = ec_hip::staged_gather(ga, comp_s);
                     if (!derive) { L[2][slot_][p] += B; }
