@@ -2147,8 +2147,10 @@ class GatedCollTask final : public Cdna4Task {
         const int nblk =
             tt_->cfg_.gated_blocks
                 ? tt_->cfg_.gated_blocks
-                : (nfrags_ == 1 && total_ >= 160u * 1024 * 1024 ? 96
-                                                  : ec_hip::kGatedBlocks);
+                : (nfrags_ == 1 && total_ >= 160u * 1024 * 1024
+                       ? 128 /* re-measured after 1-block gates:
+                                profiles/rocprof_kernels_r02.md */
+                       : ec_hip::kGatedBlocks);
         const uint64_t B = (uint64_t)nblk;
         for (size_t f = 0; f < nfrags_; f++) {
             const uint32_t p   = (uint32_t)(f & 1);
