@@ -1351,23 +1351,32 @@ static ucc_status_t launch_staged_reduce(const GatedArgs &a, hipStream_t s)
         const char *e = getenv("UCC_EC_REDUCE_DEPTH");
         return e ? atoi(e) : 0;
     }();
+    static int thr = [] {
+        const char *e = getenv("UCC_EC_REDUCE_THREADS");
+        int         v = e ? atoi(e) : 256;
+        return (v == 128 || v == 512 || v == 1024) ? v : 256;
+    }();
     int depth = ov ? ov : (a.nranks <= 2 ? 4 : a.nranks <= 4 ? 2 : 1);
+    /* the protocol counts BLOCKS, so the grid is fixed by the ledger;
+     * threads-per-block is a free tuning axis (the wait/signal logic
+     * only needs >=64 threads) */
+    int blocks = gated_grid(a), threads = thr;
     switch (depth) {
     case 8:
         hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC, 8>),
-                           dim3(gated_grid(a)), dim3(256), 0, s, a);
+                           dim3(blocks), dim3(threads), 0, s, a);
         break;
     case 4:
         hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC, 4>),
-                           dim3(gated_grid(a)), dim3(256), 0, s, a);
+                           dim3(blocks), dim3(threads), 0, s, a);
         break;
     case 2:
         hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC, 2>),
-                           dim3(gated_grid(a)), dim3(256), 0, s, a);
+                           dim3(blocks), dim3(threads), 0, s, a);
         break;
     default:
         hipLaunchKernelGGL((k_staged_reduce<T, OP, VEC, 1>),
-                           dim3(gated_grid(a)), dim3(256), 0, s, a);
+                           dim3(blocks), dim3(threads), 0, s, a);
     }
     return hipGetLastError() == hipSuccess ? UCC_OK : UCC_ERR_NO_RESOURCE;
 }
