@@ -25,7 +25,21 @@ static void print_version()
 
 static void print_configs()
 {
-    /* touch the main tables so defaults are declared */
+    /* create a lib+context so every TL's context_create runs its
+     * declare() calls (the dump should list ALL knobs, not just the
+     * core table) */
+    {
+        ucc_lib_h        lib = nullptr;
+        ucc_lib_params_t lp{};
+        if (ucc_init(&lp, nullptr, &lib) == UCC_OK) {
+            ucc_context_h        ctx = nullptr;
+            ucc_context_params_t cp{};
+            if (ucc_context_create(lib, &cp, nullptr, &ctx) == UCC_OK) {
+                ucc_context_destroy(ctx);
+            }
+            ucc_finalize(lib);
+        }
+    }
     Config &c = Config::instance();
     printf("#\n# configuration variables (UCC_<COMPONENT>_<NAME>)\n#\n");
     auto entries = c.entries();
