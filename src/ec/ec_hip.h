@@ -66,6 +66,14 @@ struct FusedArgs {
     uint64_t   *error_word; /* local fine-grained word set on spin timeout */
     int         nblocks;    /* workgroups; each handles a count slice     */
     uint64_t    spin_limit; /* 0 = default kSpinLimit                     */
+    /* host-pinned completion flag: the LAST block (cumulative arrival
+     * counter at flags[kFusedDoneBase + slot] reaching done_target)
+     * release-stores done_seq here — the host polls plain memory
+     * instead of hipEventRecord+Query (µs-class latency saving on the
+     * small-message path). */
+    uint64_t   *done_host;
+    uint64_t    done_seq;
+    uint64_t    done_target;
 };
 ucc_status_t fused_allreduce(const FusedArgs &a, hipStream_t stream);
 
@@ -138,6 +146,9 @@ constexpr int kGatedGraphBase = 1024;
  * u64 idx kGatedMirrorBase + ((phase*kGatedSlots+slot)*2+parity)*kMaxRanks
  *         + src_rank -> 9216..9599 */
 constexpr int kGatedMirrorBase = 9216;
+/* fused-kernel completion arrival counters: u64 idx
+ * kFusedDoneBase + slot -> 9600..9607 */
+constexpr int kFusedDoneBase = 9600;
 
 /* default device spin bound (iterations of the bounded wait loops);
  * host-side team setup scales it by the xGMI hop count of the farthest

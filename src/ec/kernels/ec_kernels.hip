@@ -666,6 +666,28 @@ __global__ void k_fused_allreduce(FusedArgs a)
             ((T *)a.dst)[t] = Cvt<T>::store(apply_alpha<A>(r, a.alpha));
         }
     }
+    /* 5. host-visible completion: the last arriving block publishes
+     * done_seq to the pinned word — the host polls plain memory
+     * instead of event record+query (see FusedArgs.done_host). dst
+     * stores are fenced before the arrival add; the release store
+     * makes them visible to the host with the flag. */
+    if (a.done_host) {
+        __threadfence_system();
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            uint64_t v = __hip_atomic_fetch_add(
+                             a.local_flags + kFusedDoneBase + a.slot, 1,
+                             __ATOMIC_ACQ_REL,
+                             __HIP_MEMORY_SCOPE_AGENT) +
+                         1;
+            if (v == a.done_target) {
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                __hip_atomic_store(a.done_host, a.done_seq,
+                                   __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_SYSTEM);
+            }
+        }
+    }
 }
 
 /* ------------------------------------- graph-capturable fused allreduce */

@@ -270,8 +270,10 @@ class Cdna4TlTeam final : public TlTeam {
             HIPCHK(hipMalloc((void **)&flags_, ec_hip::kFlagsBytes));
         }
         HIPCHK(hipMemset(flags_, 0, ec_hip::kFlagsBytes));
-        HIPCHK(hipHostMalloc((void **)&err_host_, 64, hipHostMallocDefault));
-        *err_host_ = 0;
+        HIPCHK(hipHostMalloc((void **)&err_host_, 128,
+                             hipHostMallocDefault));
+        memset((void *)err_host_, 0, 128);
+        done_host_ = err_host_ + 8; /* [8..15]: per-slot fused flags */
         HIPCHK(hipDeviceSynchronize());
         return UCC_OK;
     }
@@ -459,7 +461,9 @@ class Cdna4TlTeam final : public TlTeam {
     uint8_t    *scratch_ = nullptr;
     size_t      scratch_bytes_ = 0;
     uint64_t   *flags_ = nullptr;  /* device fine-grained               */
-    uint64_t   *err_host_ = nullptr; /* host-pinned, device-writable    */
+    uint64_t   *err_host_  = nullptr; /* host-pinned, device-writable   */
+    uint64_t   *done_host_ = nullptr; /* per-slot fused completion flags */
+    std::vector<uint64_t> done_cum_;  /* cumulative completion arrivals  */
     std::vector<PeerRes> peers_;
     ucc_status_t init_st_ = UCC_OK;
     bool         attached_ = false;
@@ -584,28 +588,31 @@ class FusedAllreduceTask final : public Cdna4Task {
             fa.nblocks     = (int)(blocks < 1 ? 1 : blocks > 16 ? 16 : blocks);
             if (tt_->stage_cum_.size() < tt_->cfg_.nslots) {
                 tt_->stage_cum_.assign(tt_->cfg_.nslots, 0);
+                tt_->done_cum_.assign(tt_->cfg_.nslots, 0);
             }
             tt_->stage_cum_[slot_] += (uint64_t)fa.nblocks;
             fa.stage_target = tt_->stage_cum_[slot_];
+            /* host-pinned completion flag instead of event
+             * record+query (µs-class small-message latency saving) */
+            tt_->done_cum_[slot_] += (uint64_t)fa.nblocks;
+            fa.done_target = tt_->done_cum_[slot_];
+            fa.done_seq    = fseq_;
+            fa.done_host   = tt_->done_host_ + slot_;
             ucc_status_t st = ec_hip::fused_allreduce(fa, comp());
             if (st != UCC_OK) {
                 return st;
             }
-            HIPWARN(hipEventRecord(ev(0), comp()));
             phase_ = 1;
         }
         if (phase_ == 1) {
-            hipError_t e = hipEventQuery(ev(0));
-            if (e == hipErrorNotReady) {
-                return UCC_INPROGRESS;
-            }
-            if (e != hipSuccess) {
-                return UCC_ERR_NO_RESOURCE;
-            }
             if (*tt_->err_host_ != 0) {
                 ucc_error("fused allreduce timed out waiting for peers");
                 close_slot();
                 return UCC_ERR_TIMED_OUT;
+            }
+            if (__atomic_load_n(tt_->done_host_ + slot_,
+                                __ATOMIC_ACQUIRE) < fseq_) {
+                return UCC_INPROGRESS;
             }
             close_slot();
             return UCC_OK;
