@@ -213,6 +213,12 @@ struct GatedArgs {
                                0 = kGatedBlocks default) */
     int         pull_wait;  /* 1 = poll peers' counters remotely instead
                                of local mirrors (debug fallback) */
+    /* host-pinned completion (set ONLY on the final kernel of a
+     * host-posted collective): the last arriving block release-stores
+     * done_seq after its signal push — host polls plain memory instead
+     * of a trailing hipEvent. */
+    uint64_t   *done_host;
+    uint64_t    done_seq;
     uint64_t    spin_limit; /* 0 = default kSpinLimit */
     /* per-dest cell staging (alltoall): my_in[c_dst_off[k]] <-
      * src[c_src_off[k]], c_len[k] bytes; 0 = contiguous stage of len   */

@@ -930,7 +930,8 @@ gated_wait(const GatedArgs &a, int phase, uint64_t target)
  * earlier block's release (and thus its payload writes, xGMI ones
  * included) visible-before the mirror store. */
 __device__ __forceinline__ void gated_signal(const GatedArgs &a, int phase,
-                                             uint64_t sig_target)
+                                             uint64_t sig_target,
+                                             bool pub_done = true)
 {
     __threadfence_system();
     __syncthreads();
@@ -952,6 +953,12 @@ __device__ __forceinline__ void gated_signal(const GatedArgs &a, int phase,
                         gated_mirror_idx(phase, a.slot, a.parity,
                                          a.rank),
                     v, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_SYSTEM);
+            }
+            if (pub_done && a.done_host) {
+                /* final kernel of a host-posted collective */
+                __hip_atomic_store(a.done_host, a.done_seq,
+                                   __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_SYSTEM);
             }
         }
     }
@@ -1149,13 +1156,27 @@ static inline int gated_grid(const GatedArgs &a)
  * k_gated_done publishes my-copies-complete and waits the team. */
 __global__ void k_gated_wait_only(const GatedArgs a)
 {
-    (void)gated_wait(a, a.gw_phase, a.t_gather_wait);
+    if (!gated_wait(a, a.gw_phase, a.t_gather_wait)) {
+        return;
+    }
+    if (a.done_host && blockIdx.x == 0 && threadIdx.x == 0) {
+        /* completion point: the team's final phase is confirmed, so
+         * every peer has stopped reading this rank's buffers */
+        __hip_atomic_store(a.done_host, a.done_seq, __ATOMIC_RELEASE,
+                           __HIP_MEMORY_SCOPE_SYSTEM);
+    }
 }
 
 __global__ void k_gated_done(const GatedArgs a)
 {
-    gated_signal(a, 2, a.t_sig_gather);
-    (void)gated_wait(a, 2, a.t_gather_wait);
+    gated_signal(a, 2, a.t_sig_gather, /*pub_done=*/false);
+    if (!gated_wait(a, 2, a.t_gather_wait)) {
+        return;
+    }
+    if (a.done_host && blockIdx.x == 0 && threadIdx.x == 0) {
+        __hip_atomic_store(a.done_host, a.done_seq, __ATOMIC_RELEASE,
+                           __HIP_MEMORY_SCOPE_SYSTEM);
+    }
 }
 
 ucc_status_t gated_wait_only(const GatedArgs &a, hipStream_t s)

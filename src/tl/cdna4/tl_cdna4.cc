@@ -270,10 +270,12 @@ class Cdna4TlTeam final : public TlTeam {
             HIPCHK(hipMalloc((void **)&flags_, ec_hip::kFlagsBytes));
         }
         HIPCHK(hipMemset(flags_, 0, ec_hip::kFlagsBytes));
-        HIPCHK(hipHostMalloc((void **)&err_host_, 128,
+        HIPCHK(hipHostMalloc((void **)&err_host_, 192,
                              hipHostMallocDefault));
-        memset((void *)err_host_, 0, 128);
-        done_host_ = err_host_ + 8; /* [8..15]: per-slot fused flags */
+        memset((void *)err_host_, 0, 192);
+        done_host_  = err_host_ + 8;  /* [8..15]: per-slot fused flags */
+        gdone_host_ = err_host_ + 16; /* [16..23]: per-slot gated flags */
+        gdone_seq_.assign(ec_hip::kGatedSlots, 0);
         HIPCHK(hipDeviceSynchronize());
         return UCC_OK;
     }
@@ -463,7 +465,9 @@ class Cdna4TlTeam final : public TlTeam {
     uint64_t   *flags_ = nullptr;  /* device fine-grained               */
     uint64_t   *err_host_  = nullptr; /* host-pinned, device-writable   */
     uint64_t   *done_host_ = nullptr; /* per-slot fused completion flags */
+    uint64_t   *gdone_host_ = nullptr; /* per-slot gated completion flags */
     std::vector<uint64_t> done_cum_;  /* cumulative completion arrivals  */
+    std::vector<uint64_t> gdone_seq_; /* per-slot gated completion seqs  */
     std::vector<PeerRes> peers_;
     ucc_status_t init_st_ = UCC_OK;
     bool         attached_ = false;
@@ -1938,25 +1942,25 @@ class GatedCollTask final : public Cdna4Task {
             if (!all_ge(0)) {
                 return UCC_INPROGRESS;
             }
+            done_seq_       = ++tt_->gdone_seq_[slot_];
             ucc_status_t st = enqueue_frags(copy_s(), comp(), false);
             if (st != UCC_OK) {
                 return st;
             }
-            HIPWARN(hipEventRecord(ev(0), comp()));
             phase_ = 1;
         }
-        hipError_t e = hipEventQuery(ev(0));
-        if (e == hipErrorNotReady) {
-            return UCC_INPROGRESS;
-        }
-        if (e != hipSuccess) {
-            return UCC_ERR_NO_RESOURCE;
-        }
+        /* host-pinned completion flag written by the final kernel once
+         * the TEAM's final phase is confirmed (every peer has stopped
+         * reading this rank's buffers) — no trailing hipEvent */
         if (*tt_->err_host_ != 0) {
             ucc_error("gated %s timed out waiting for peers",
                       coll_type_name(ct_));
             close_slot();
             return UCC_ERR_TIMED_OUT;
+        }
+        if (__atomic_load_n(tt_->gdone_host_ + slot_,
+                            __ATOMIC_ACQUIRE) < done_seq_) {
+            return UCC_INPROGRESS;
         }
         close_slot();
         return UCC_OK;
@@ -2039,6 +2043,8 @@ class GatedCollTask final : public Cdna4Task {
         /* my pulls done -> publish; complete when every rank's pulls
          * (including reads of MY src) are done */
         ga.t_gather_wait = L[2][slot_][p] + B;
+        ga.done_host     = tt_->gdone_host_ + slot_;
+        ga.done_seq      = done_seq_;
         st = ec_hip::gated_done(ga, comp_s);
         L[2][slot_][p] += B;
         return st;
@@ -2116,6 +2122,8 @@ class GatedCollTask final : public Cdna4Task {
             }
         }
         ga.t_gather_wait = L[2][slot_][p] + B;
+        ga.done_host     = tt_->gdone_host_ + slot_;
+        ga.done_seq      = done_seq_;
         st = ec_hip::gated_done(ga, comp_s);
         L[2][slot_][p] += B;
         return st;
@@ -2258,6 +2266,10 @@ class GatedCollTask final : public Cdna4Task {
                     }
                     ec_hip::GatedArgs gg = ga;
                     gg.nblocks           = (int)Bg;
+                    if (!derive && f == nfrags_ - 1) {
+                        gg.done_host = tt_->gdone_host_ + slot_;
+                        gg.done_seq  = done_seq_;
+                    }
                     st = ec_hip::staged_gather(gg, comp_s);
                     if (!derive) { L[2][slot_][p] += Bg; }
                 }
@@ -2307,6 +2319,17 @@ class GatedCollTask final : public Cdna4Task {
                     st = ec_hip::staged_reduce(ga, comp_s);
                     if (!derive) { L[1][slot_][p] += B; }
                 }
+                if (st == UCC_OK && !derive && f == nfrags_ - 1) {
+                    /* completion = the TEAM's reduces done: a peer's
+                     * zero-copy reduce may still be reading MY src
+                     * when my own reduce retires */
+                    ec_hip::GatedArgs gw = ga;
+                    gw.gw_phase          = 1;
+                    gw.t_gather_wait     = L[1][slot_][p];
+                    gw.done_host         = tt_->gdone_host_ + slot_;
+                    gw.done_seq          = done_seq_;
+                    st = ec_hip::gated_wait_only(gw, comp_s);
+                }
                 break;
             }
             case UCC_COLL_TYPE_ALLGATHERV:
@@ -2353,6 +2376,14 @@ class GatedCollTask final : public Cdna4Task {
                     st = ec_hip::staged_gather(ga, comp_s);
                     if (!derive) { L[2][slot_][p] += B; }
                 }
+                if (st == UCC_OK && !derive && f == nfrags_ - 1) {
+                    ec_hip::GatedArgs gw = ga;
+                    gw.gw_phase          = 2;
+                    gw.t_gather_wait     = L[2][slot_][p];
+                    gw.done_host         = tt_->gdone_host_ + slot_;
+                    gw.done_seq          = done_seq_;
+                    st = ec_hip::gated_wait_only(gw, comp_s);
+                }
                 break;
             }
             case UCC_COLL_TYPE_ALLTOALLV: {
@@ -2391,6 +2422,14 @@ class GatedCollTask final : public Cdna4Task {
                 if (st == UCC_OK) {
                     st = ec_hip::staged_gather(ga, comp_s);
                     if (!derive) { L[2][slot_][p] += B; }
+                }
+                if (st == UCC_OK && !derive && f == nfrags_ - 1) {
+                    ec_hip::GatedArgs gw = ga;
+                    gw.gw_phase          = 2;
+                    gw.t_gather_wait     = L[2][slot_][p];
+                    gw.done_host         = tt_->gdone_host_ + slot_;
+                    gw.done_seq          = done_seq_;
+                    st = ec_hip::gated_wait_only(gw, comp_s);
                 }
                 break;
             }
@@ -2435,6 +2474,14 @@ class GatedCollTask final : public Cdna4Task {
                     st = ec_hip::staged_gather(ga, comp_s);
                     if (!derive) { L[2][slot_][p] += B; }
                 }
+                if (st == UCC_OK && !derive && f == nfrags_ - 1) {
+                    ec_hip::GatedArgs gw = ga;
+                    gw.gw_phase          = 2;
+                    gw.t_gather_wait     = L[2][slot_][p];
+                    gw.done_host         = tt_->gdone_host_ + slot_;
+                    gw.done_seq          = done_seq_;
+                    st = ec_hip::gated_wait_only(gw, comp_s);
+                }
                 break;
             }
             default:
@@ -2459,6 +2506,7 @@ class GatedCollTask final : public Cdna4Task {
     std::vector<size_t> rcnt_, rdsp_; /* recv column bytes (a2av)     */
     bool                a2av_ready_ = false; /* global max exchanged  */
     uint64_t            a2av_gsum_  = 0;     /* global send-byte sum  */
+    uint64_t            done_seq_   = 0;     /* host completion seq   */
     int                pslot_ = -1;
     /* zero-copy persistent allreduce: peers' USER src buffers mapped
      * via HIP-IPC (handles exchanged through the scratch channel at
