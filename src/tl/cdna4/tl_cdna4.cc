@@ -2546,8 +2546,10 @@ class Cdna4Tl final : public Tl {
                     "dedicated slots for persistent triggered colls");
         cfg.declare("TL_CDNA4", "CHUNK_SIZE", "32m",
                     "staging fragment bytes per slot area");
-        cfg.declare("TL_CDNA4", "FUSED_MAX", "512k",
-                    "max msg bytes for the fused single-kernel allreduce");
+        cfg.declare("TL_CDNA4", "FUSED_MAX", "4m",
+                    "max msg bytes for the fused single-kernel allreduce "
+                    "(measured crossover vs the gated pipeline, "
+                    "profiles/rocprof_kernels_r02.md)");
         cfg.declare("TL_CDNA4", "GATED", "1",
                     "device-gated pipeline for large colls");
         cfg.declare("TL_CDNA4", "SPIN_LIMIT", "0",
@@ -2595,7 +2597,8 @@ class Cdna4Tl final : public Tl {
         c.chunk     = cfg.get_size("TL_CDNA4", "CHUNK_SIZE", 32 * 1024 * 1024);
         c.spin_limit =
             (uint64_t)cfg.get_int("TL_CDNA4", "SPIN_LIMIT", 0);
-        c.fused_max = cfg.get_size("TL_CDNA4", "FUSED_MAX", 512 * 1024);
+        c.fused_max = cfg.get_size("TL_CDNA4", "FUSED_MAX",
+                                   4 * 1024 * 1024);
         c.gated_blocks =
             (int)cfg.get_int("TL_CDNA4", "GATED_BLOCKS", 0);
         if (c.gated_blocks < 0) {
