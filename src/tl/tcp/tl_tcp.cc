@@ -2372,6 +2372,202 @@ class TcpAlltoallBruckTask final : public TcpTask {
 
 /* ---- reduce: linear recv+reduce at root (small n); gather/scatter(v):
  * linear to/from root; reduce_scatter(v): reduce@0 + scatterv */
+/* ---- k-nomial gather/scatter (reference tl/ucp gather/scatter
+ * knomial role, re-derived): blocks travel in VIRTUAL-rank order
+ * (vr = (rank - root) mod n) so every subtree owns a contiguous range
+ * [vr, vr + q) and each tree edge moves ONE contiguous message.
+ * Scatter: the root lays the team blocks out in virtual order in a
+ * scratch buffer, then each parent forwards each child its subtree
+ * range; ranks keep their own first block. Gather is the exact
+ * reverse (children aggregate their subtree ranges up; the root
+ * unpacks virtual order back to team order). log_k(n) latency vs the
+ * linear task's root fan of n-1 messages. */
+class TcpGatherScatterKnTask final : public TcpTask {
+  public:
+    TcpGatherScatterKnTask(Context *ctx, TcpTlTeam *tt,
+                           const ucc_coll_args_t &args, uint32_t radix)
+        : TcpTask(ctx, tt, args), k_(radix < 2 ? 2 : radix)
+    {
+    }
+
+    ucc_status_t post() override
+    {
+        begin();
+        gather_ = a_.coll_type == UCC_COLL_TYPE_GATHER;
+        root_   = (uint32_t)a_.root;
+        vr_     = (me_ + n_ - root_) % n_;
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        if (gather_) {
+            dtsz_ = ucc_dt_size(a_.src.info.datatype);
+            /* root convention: count is the gathered TOTAL (dst side);
+             * leaves pass their single block count */
+            blk_ = me_ == root_ ? a_.dst.info.count * dtsz_ / n_
+                                : a_.src.info.count * dtsz_;
+        } else {
+            dtsz_ = ucc_dt_size(me_ == root_ ? a_.src.info.datatype
+                                             : a_.dst.info.datatype);
+            blk_  = me_ == root_
+                        ? a_.src.info.count * dtsz_ / n_
+                        : a_.dst.info.count * dtsz_;
+        }
+        if (blk_ == 0) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        /* subtree size: q = my fan-out level (lowest nonzero base-k
+         * digit of vr; n for the root) */
+        sub_ = 1;
+        if (vr_ == 0) {
+            while (sub_ < n_) {
+                sub_ *= k_;
+            }
+        } else {
+            uint64_t q = 1;
+            while ((vr_ / q) % k_ == 0) {
+                q *= k_;
+            }
+            sub_ = q;
+        }
+        nsub_ = std::min<uint64_t>(sub_, n_ - vr_);
+        work_.resize(nsub_ * blk_);
+        if (gather_) {
+            /* my own block at virtual offset 0 */
+            const void *mysrc =
+                (me_ == root_ && inplace)
+                    ? (const uint8_t *)a_.dst.info.buffer + me_ * blk_
+                    : a_.src.info.buffer;
+            memcpy(work_.data(), mysrc, blk_);
+            /* children push their subtree ranges up */
+            for (uint64_t q = sub_ / k_; q >= 1; q /= k_) {
+                for (uint32_t i = 1; i < k_; i++) {
+                    uint64_t c = vr_ + (uint64_t)i * q;
+                    if (c < n_) {
+                        uint64_t cn = std::min<uint64_t>(q, n_ - c);
+                        recv_from(to_team((uint32_t)c), 0,
+                                  work_.data() + (c - vr_) * blk_,
+                                  cn * blk_);
+                    }
+                }
+                if (q == 1) {
+                    break;
+                }
+            }
+            phase_ = 0;
+        } else { /* scatter */
+            if (vr_ == 0) {
+                /* root: virtual-order layout */
+                const uint8_t *src =
+                    (const uint8_t *)(inplace ? a_.dst.info.buffer
+                                              : a_.src.info.buffer);
+                for (uint64_t v = 0; v < n_; v++) {
+                    memcpy(work_.data() + v * blk_,
+                           src + ((v + root_) % n_) * blk_, blk_);
+                }
+                send_children();
+                phase_ = 1;
+            } else {
+                uint32_t digit  = 0;
+                uint64_t q      = 1;
+                while ((vr_ / q) % k_ == 0) {
+                    q *= k_;
+                }
+                digit = (uint32_t)((vr_ / q) % k_);
+                recv_from(to_team(vr_ - digit * (uint32_t)q), 0,
+                          work_.data(), nsub_ * blk_);
+                phase_ = 0;
+            }
+        }
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    uint32_t to_team(uint32_t v) const { return (v + root_) % n_; }
+
+    void send_children()
+    {
+        for (uint64_t q = sub_ / k_; q >= 1; q /= k_) {
+            for (uint32_t i = 1; i < k_; i++) {
+                uint64_t c = vr_ + (uint64_t)i * q;
+                if (c < n_) {
+                    uint64_t cn = std::min<uint64_t>(q, n_ - c);
+                    send_to(to_team((uint32_t)c), 0,
+                            work_.data() + (c - vr_) * blk_,
+                            cn * blk_);
+                }
+            }
+            if (q == 1) {
+                break;
+            }
+        }
+    }
+
+    ucc_status_t progress_()
+    {
+        if (gather_) {
+            if (phase_ == 0) { /* all children arrived? */
+                if (!ops_done()) {
+                    return UCC_INPROGRESS;
+                }
+                clear_ops();
+                if (vr_ != 0) { /* push my aggregated range up */
+                    uint64_t q = sub_;
+                    uint32_t digit =
+                        (uint32_t)((vr_ / q) % k_);
+                    send_to(to_team(vr_ - digit * (uint32_t)q), 0,
+                            work_.data(), nsub_ * blk_);
+                }
+                phase_ = 1;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            if (vr_ == 0) { /* unpack virtual -> team order */
+                uint8_t *dst = (uint8_t *)a_.dst.info.buffer;
+                for (uint64_t v = 0; v < n_; v++) {
+                    memcpy(dst + ((v + root_) % n_) * blk_,
+                           work_.data() + v * blk_, blk_);
+                }
+            }
+            return UCC_OK;
+        }
+        /* scatter */
+        if (phase_ == 0) { /* wait my range, then forward children */
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            send_children();
+            phase_ = 1;
+        }
+        if (!ops_done()) {
+            return UCC_INPROGRESS;
+        }
+        clear_ops();
+        if (me_ == root_) {
+            /* root's own block: already in place when IN_PLACE; the
+             * root's dst may legitimately be null (send-only root) */
+            if (!(a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE) &&
+                a_.dst.info.buffer) {
+                memcpy(a_.dst.info.buffer,
+                       (const uint8_t *)a_.src.info.buffer +
+                           me_ * blk_,
+                       blk_);
+            }
+        } else {
+            memcpy(a_.dst.info.buffer, work_.data(), blk_);
+        }
+        return UCC_OK;
+    }
+
+    uint32_t k_ = 2, root_ = 0, vr_ = 0;
+    bool     gather_ = false;
+    uint64_t sub_ = 1, nsub_ = 1;
+    size_t   dtsz_ = 4, blk_ = 0;
+    std::vector<uint8_t> work_;
+};
+
 class TcpRootedTask final : public TcpTask {
   public:
     using TcpTask::TcpTask;
@@ -2963,6 +3159,36 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
     add(UCC_COLL_TYPE_REDUCE, mk((TcpRootedTask *)nullptr));
     add(UCC_COLL_TYPE_REDUCE_SCATTER, mk((TcpRootedTask *)nullptr));
     add(UCC_COLL_TYPE_REDUCE_SCATTERV, mk((TcpRootedTask *)nullptr));
+    {
+        uint32_t radix = (uint32_t)Config::instance().get_int(
+            "TL_TCP", "KN_RADIX", 4);
+        for (auto ct : {UCC_COLL_TYPE_GATHER, UCC_COLL_TYPE_SCATTER}) {
+            ScoreRange r;
+            r.start    = 0;
+            r.end      = SIZE_MAX;
+            r.score    = sc + 1;
+            r.tl_name  = "tcp";
+            r.alg_name = "knomial";
+            r.init     = [self, radix, ct](const ucc_coll_args_t &args,
+                                       Team *t2,
+                                       Task **task) -> ucc_status_t {
+                const bool   root = t2->rank == (uint32_t)args.root;
+                ucc_datatype_t dt =
+                    ct == UCC_COLL_TYPE_GATHER
+                        ? args.src.info.datatype
+                        : (root ? args.src.info.datatype
+                                : args.dst.info.datatype);
+                if ((args.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) ||
+                    !ucc_dt_is_predefined(dt)) {
+                    return UCC_ERR_NOT_SUPPORTED;
+                }
+                *task = new TcpGatherScatterKnTask(t2->ctx, self, args,
+                                                   radix);
+                return UCC_OK;
+            };
+            map.add(ct, UCC_MEMORY_TYPE_HOST, r);
+        }
+    }
     add(UCC_COLL_TYPE_GATHER, mk((TcpRootedTask *)nullptr));
     add(UCC_COLL_TYPE_GATHERV, mk((TcpRootedTask *)nullptr));
     add(UCC_COLL_TYPE_SCATTER, mk((TcpRootedTask *)nullptr));
