@@ -752,6 +752,149 @@ class TcpAllreduceSraTask final : public TcpTask {
     bool               in_rs_ = true;
 };
 
+/* ---- k-nomial (radix-k) allreduce (reference tl/ucp
+ * allreduce_knomial + recursive_knomial.h PROXY/EXTRA role,
+ * re-derived): n2 = largest power of k <= n; EXTRA ranks (vr >= n2)
+ * fold their vector into a PROXY (vr - n2) first; proxies run
+ * ceil(log_k n2) rounds where each k-group exchanges full vectors and
+ * reduces; proxies return the result to their extras. Radix k trades
+ * (k-1) messages per round for fewer rounds — the small-message
+ * latency knob (UCC_TL_TCP_KN_RADIX). */
+class TcpAllreduceKnomialTask final : public TcpTask {
+  public:
+    TcpAllreduceKnomialTask(Context *ctx, TcpTlTeam *tt,
+                            const ucc_coll_args_t &args, uint32_t radix)
+        : TcpTask(ctx, tt, args), k_(radix < 2 ? 2 : radix)
+    {
+    }
+
+    ucc_status_t post() override
+    {
+        begin();
+        dt_    = a_.dst.info.datatype;
+        op_    = a_.op == UCC_OP_AVG ? UCC_OP_SUM : a_.op;
+        dtsz_  = ucc_dt_size(dt_);
+        count_ = a_.dst.info.count;
+        bytes_ = count_ * dtsz_;
+        dst_   = (uint8_t *)a_.dst.info.buffer;
+        if (!(a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE)) {
+            ec_cpu::copy(dst_, a_.src.info.buffer, bytes_);
+        }
+        vr_ = me_; /* allreduce is symmetric: no root shift needed */
+        n2_ = 1;
+        while (n2_ * k_ <= n_) {
+            n2_ *= k_;
+        }
+        extra_ = vr_ >= n2_;
+        /* a proxy may serve SEVERAL extras when n > 2*n2 (extras fold
+         * modulo n2 — reference recursive_knomial PROXY role) */
+        nex_ = 0;
+        if (!extra_) {
+            for (uint64_t e = vr_ + n2_; e < n_; e += n2_) {
+                nex_++;
+            }
+        }
+        proxy_       = nex_ > 0;
+        size_t slots = k_ - 1 > nex_ ? k_ - 1 : nex_;
+        tmp_.resize(slots * bytes_);
+        phase_ = 0;
+        q_     = 1;
+        if (extra_) { /* fold into my proxy first */
+            send_to((uint32_t)(vr_ % n2_), 63, dst_, bytes_);
+        } else {
+            uint32_t slot = 0;
+            for (uint64_t e = vr_ + n2_; e < n_; e += n2_) {
+                recv_from((uint32_t)e, 63,
+                          tmp_.data() + (size_t)slot * bytes_, bytes_);
+                slot++;
+            }
+        }
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    ucc_status_t progress_()
+    {
+        if (phase_ == 0) { /* extra fold */
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            for (uint32_t j = 0; j < nex_; j++) {
+                const void *srcs[2] = {dst_,
+                                       tmp_.data() + (size_t)j * bytes_};
+                ec_cpu::reduce(dst_, srcs, 2, count_, dt_, op_);
+            }
+            if (extra_) { /* wait for the final result */
+                recv_from((uint32_t)(vr_ % n2_), 62, dst_, bytes_);
+                phase_ = 3;
+                return progress_();
+            }
+            phase_ = 1;
+        }
+        while (q_ < n2_) {
+            if (phase_ == 1) { /* post this round's k-group exchange */
+                uint32_t digit = (uint32_t)((vr_ / q_) % k_);
+                uint32_t base  = vr_ - digit * (uint32_t)q_;
+                int      slot  = 0;
+                for (uint32_t j = 0; j < k_; j++) {
+                    if (j == digit) {
+                        continue;
+                    }
+                    uint32_t peer = base + j * (uint32_t)q_;
+                    send_to(peer, (uint32_t)round_, dst_, bytes_);
+                    recv_from(peer, (uint32_t)round_,
+                              tmp_.data() + (size_t)slot * bytes_,
+                              bytes_);
+                    slot++;
+                }
+                phase_ = 2;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            for (uint32_t j = 0; j + 1 < k_; j++) {
+                const void *srcs[2] = {dst_,
+                                       tmp_.data() + (size_t)j * bytes_};
+                ec_cpu::reduce(dst_, srcs, 2, count_, dt_, op_);
+            }
+            q_ *= k_;
+            round_++;
+            phase_ = 1;
+        }
+        if (phase_ != 3 && proxy_) { /* return the result to my extras */
+            for (uint64_t e = vr_ + n2_; e < n_; e += n2_) {
+                send_to((uint32_t)e, 62, dst_, bytes_);
+            }
+            phase_ = 3;
+        }
+        if (!ops_done()) {
+            return UCC_INPROGRESS;
+        }
+        clear_ops();
+        if (a_.op == UCC_OP_AVG) {
+            const void *srcs[1] = {dst_};
+            ec_cpu::reduce(dst_, srcs, 1, count_, dt_, UCC_OP_SUM,
+                           1.0 / (double)n_);
+        }
+        return UCC_OK;
+    }
+
+    uint32_t k_ = 2, vr_ = 0;
+    uint64_t n2_ = 1, q_ = 1;
+    uint32_t nex_ = 0;
+    bool     extra_ = false, proxy_ = false;
+    ucc_datatype_t     dt_ = UCC_DT_FLOAT32;
+    ucc_reduction_op_t op_ = UCC_OP_SUM;
+    size_t   dtsz_ = 4, bytes_ = 0;
+    uint64_t count_ = 0;
+    uint8_t *dst_ = nullptr;
+    std::vector<uint8_t> tmp_;
+};
+
 /* ---- sliding-window allreduce (reference tl/ucp
  * allreduce_sliding_window.c:23-45 role, re-derived for the socket
  * p2p layer): the message is cut into windows; each window runs the
@@ -2938,6 +3081,32 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
             return UCC_OK;
         };
         map.add(UCC_COLL_TYPE_ALLREDUCE, UCC_MEMORY_TYPE_HOST, r);
+    }
+    {
+        /* radix-k knomial allreduce for the small-message band */
+        uint32_t radix = (uint32_t)Config::instance().get_int(
+            "TL_TCP", "KN_RADIX", 4);
+        size_t kn_max = Config::instance().get_size(
+            "TL_TCP", "KN_AR_MAX", 64 * 1024);
+        if (radix > 2 && kn_max > 0) {
+            ScoreRange r;
+            r.start    = 0;
+            r.end      = kn_max;
+            r.score    = sc + 1;
+            r.tl_name  = "tcp";
+            r.alg_name = "knomial";
+            r.init     = [self, radix](const ucc_coll_args_t &args,
+                                   Team *t2,
+                                   Task **task) -> ucc_status_t {
+                if (!ucc_dt_is_predefined(args.dst.info.datatype)) {
+                    return UCC_ERR_NOT_SUPPORTED;
+                }
+                *task = new TcpAllreduceKnomialTask(t2->ctx, self,
+                                                    args, radix);
+                return UCC_OK;
+            };
+            map.add(UCC_COLL_TYPE_ALLREDUCE, UCC_MEMORY_TYPE_HOST, r);
+        }
     }
     {
         /* sliding-window overlap for huge host messages (reference

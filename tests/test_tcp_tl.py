@@ -459,3 +459,50 @@ def test_gather_scatter_knomial():
         sys.stderr.write(p.stderr[-3000:])
         assert p.returncode == 0 and "KN_GS_OK" in p.stdout, \
             f"radix {radix}"
+
+
+def test_allreduce_knomial_radix():
+    """Radix-k knomial allreduce (reference allreduce_knomial +
+    recursive_knomial.h PROXY/EXTRA role): non-power-of-k team sizes
+    fold extras into proxies; result identical to the reference sum."""
+    for radix in (3, 4):
+        code = (
+            "import sys; sys.path.insert(0, %r)\n"
+            "import numpy as np\n"
+            "from ucc_amd import core, dtypes\n"
+            "from ucc_amd.testing import LocalJob\n"
+            "c = core()\n"
+            "for n in (2, 3, 5, 7, 9):\n"
+            "    job = LocalJob(n)\n"
+            "    assert '@tcp/knomial' in c.score_map_str(\n"
+            "        job.teams[0]), n\n"
+            "    rng = np.random.default_rng(n)\n"
+            "    for cnt in (1, 501, 9_001):\n"
+            "        arrs = [(rng.random(cnt) - 0.5)\n"
+            "                .astype(np.float32) for _ in range(n)]\n"
+            "        outs = job.allreduce_np(arrs)\n"
+            "        exp = np.sum(arrs, axis=0)\n"
+            "        for o in outs:\n"
+            "            np.testing.assert_allclose(\n"
+            "                o, exp, rtol=1e-5, atol=1e-5)\n"
+            "    # AVG through the knomial path\n"
+            "    arrs = [np.full(777, float(r + 1), np.float32)\n"
+            "            for r in range(n)]\n"
+            "    outs = job.allreduce_np(arrs, op=dtypes.OP_AVG)\n"
+            "    exp = np.full(777, sum(range(1, n + 1)) / n,\n"
+            "                  np.float32)\n"
+            "    for o in outs:\n"
+            "        np.testing.assert_allclose(o, exp, rtol=1e-5,\n"
+            "                                   atol=1e-5)\n"
+            "print('KN_AR_OK')\n"
+        ) % (REPO,)
+        env = dict(os.environ)
+        env["UCC_TL_SHM_ENABLE"] = "0"
+        env["UCC_TL_TCP_KN_RADIX"] = str(radix)
+        env["UCC_TUNE"] = "allreduce:@knomial:99"
+        p = subprocess.run([sys.executable, "-c", code], env=env,
+                           capture_output=True, text=True, timeout=600)
+        sys.stdout.write(p.stdout[-500:])
+        sys.stderr.write(p.stderr[-3000:])
+        assert p.returncode == 0 and "KN_AR_OK" in p.stdout, \
+            f"radix {radix}"
