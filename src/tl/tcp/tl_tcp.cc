@@ -2515,6 +2515,93 @@ class TcpAlltoallBruckTask final : public TcpTask {
 
 /* ---- reduce: linear recv+reduce at root (small n); gather/scatter(v):
  * linear to/from root; reduce_scatter(v): reduce@0 + scatterv */
+/* ---- ring reduce-scatter (reference tl/ucp reduce_scatter ring /
+ * recursive-halving role, re-derived as the bandwidth-optimal ring):
+ * n-1 rounds over a work copy; at round r, send block (me-1-r) to the
+ * right, receive block (me-2-r) from the left and reduce — the fully
+ * reduced block landing at rank me after the last round is block me,
+ * which goes to dst. (n-1)/n * S bytes moved per rank. */
+class TcpReduceScatterRingTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        dt_   = a_.dst.info.datatype;
+        op_   = a_.op == UCC_OP_AVG ? UCC_OP_SUM : a_.op;
+        dtsz_ = ucc_dt_size(dt_);
+        if (inplace) {
+            total_ = a_.dst.info.count;
+            per_   = total_ / n_;
+        } else {
+            per_   = a_.dst.info.count;
+            total_ = per_ * n_;
+        }
+        if (per_ == 0 || per_ * n_ != total_) {
+            return UCC_ERR_NOT_SUPPORTED; /* ragged: linear handles */
+        }
+        work_.resize(total_ * dtsz_);
+        memcpy(work_.data(),
+               inplace ? a_.dst.info.buffer : a_.src.info.buffer,
+               total_ * dtsz_);
+        tmp_.resize(per_ * dtsz_);
+        round_ = 0;
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    uint8_t *blk(uint32_t b) { return work_.data() + (size_t)b * per_ * dtsz_; }
+
+    ucc_status_t progress_()
+    {
+        const uint32_t right = (me_ + 1) % n_;
+        const uint32_t left  = (me_ + n_ - 1) % n_;
+        while (round_ < (int)n_ - 1) {
+            if (phase_ == 0) {
+                uint32_t sb = (me_ + 2 * n_ - 1 - (uint32_t)round_) % n_;
+                uint32_t rb = (me_ + 2 * n_ - 2 - (uint32_t)round_) % n_;
+                (void)rb;
+                send_to(right, (uint32_t)round_, blk(sb), per_ * dtsz_);
+                recv_from(left, (uint32_t)round_, tmp_.data(),
+                          per_ * dtsz_);
+                phase_ = 1;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            uint32_t rb = (me_ + 2 * n_ - 2 - (uint32_t)round_) % n_;
+            const void *srcs[2] = {blk(rb), tmp_.data()};
+            ec_cpu::reduce(blk(rb), srcs, 2, per_, dt_, op_);
+            phase_ = 0;
+            round_++;
+        }
+        uint8_t *dst = (uint8_t *)a_.dst.info.buffer;
+        if (a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE) {
+            dst = (uint8_t *)a_.dst.info.buffer + (size_t)me_ * per_ *
+                                                      dtsz_;
+        }
+        memcpy(dst, blk(me_), per_ * dtsz_);
+        if (a_.op == UCC_OP_AVG) {
+            const void *srcs[1] = {dst};
+            ec_cpu::reduce(dst, srcs, 1, per_, dt_, UCC_OP_SUM,
+                           1.0 / (double)n_);
+        }
+        return UCC_OK;
+    }
+
+    ucc_datatype_t     dt_ = UCC_DT_FLOAT32;
+    ucc_reduction_op_t op_ = UCC_OP_SUM;
+    size_t   dtsz_ = 4;
+    uint64_t total_ = 0, per_ = 0;
+    std::vector<uint8_t> work_, tmp_;
+};
+
 /* ---- k-nomial gather/scatter (reference tl/ucp gather/scatter
  * knomial role, re-derived): blocks travel in VIRTUAL-rank order
  * (vr = (rank - root) mod n) so every subtree owns a contiguous range
@@ -3326,6 +3413,25 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
         map.add(UCC_COLL_TYPE_REDUCE, UCC_MEMORY_TYPE_HOST, r);
     }
     add(UCC_COLL_TYPE_REDUCE, mk((TcpRootedTask *)nullptr));
+    {
+        size_t rs_min = Config::instance().get_size(
+            "TL_TCP", "RS_RING_MIN", 64 * 1024);
+        ScoreRange r;
+        r.start    = rs_min;
+        r.end      = SIZE_MAX;
+        r.score    = sc + 1;
+        r.tl_name  = "tcp";
+        r.alg_name = "ring";
+        r.init     = [self](const ucc_coll_args_t &args, Team *t2,
+                        Task **task) -> ucc_status_t {
+            if (!ucc_dt_is_predefined(args.dst.info.datatype)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpReduceScatterRingTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_REDUCE_SCATTER, UCC_MEMORY_TYPE_HOST, r);
+    }
     add(UCC_COLL_TYPE_REDUCE_SCATTER, mk((TcpRootedTask *)nullptr));
     add(UCC_COLL_TYPE_REDUCE_SCATTERV, mk((TcpRootedTask *)nullptr));
     {
