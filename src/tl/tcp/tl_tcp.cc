@@ -2171,7 +2171,8 @@ class TcpAlltoallvHybridTask final : public TcpTask {
   public:
     TcpAlltoallvHybridTask(Context *ctx, TcpTlTeam *tt,
                            const ucc_coll_args_t &args, size_t thr)
-        : TcpTask(ctx, tt, args), thr_(thr)
+        : TcpTask(ctx, tt, args),
+          thr_(thr > UINT32_MAX ? (size_t)UINT32_MAX : thr)
     {
     }
 
@@ -2308,8 +2309,9 @@ class TcpAlltoallvHybridTask final : public TcpTask {
                 uint32_t hdr[3];
                 memcpy(hdr, rb.data() + off, 12);
                 off += 12;
-                if (off + hdr[2] > rb.size()) {
-                    return UCC_ERR_NO_MESSAGE;
+                if (off + hdr[2] > rb.size() || hdr[0] >= n_ ||
+                    hdr[1] >= n_) {
+                    return UCC_ERR_NO_MESSAGE; /* malformed envelope */
                 }
                 Env e;
                 e.src  = hdr[0];
@@ -2564,8 +2566,6 @@ class TcpReduceScatterRingTask final : public TcpTask {
         while (round_ < (int)n_ - 1) {
             if (phase_ == 0) {
                 uint32_t sb = (me_ + 2 * n_ - 1 - (uint32_t)round_) % n_;
-                uint32_t rb = (me_ + 2 * n_ - 2 - (uint32_t)round_) % n_;
-                (void)rb;
                 send_to(right, (uint32_t)round_, blk(sb), per_ * dtsz_);
                 recv_from(left, (uint32_t)round_, tmp_.data(),
                           per_ * dtsz_);
