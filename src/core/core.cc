@@ -1192,8 +1192,25 @@ ucc_status_t ucc_ee_set_event(ucc_ee_h ee, ucc_ev_t *ev)
 }
 ucc_status_t ucc_ee_wait(ucc_ee_h ee, ucc_ev_t *ev)
 {
-    (void)ee; (void)ev;
-    return UCC_ERR_NOT_SUPPORTED; /* host-side wait: use stream sync */
+    /* block until the NEXT event is available (reference ucc_ee.c
+     * role): drain pending stream completions while spinning. The
+     * event is returned in *ev and stays at the queue head until
+     * acked. */
+    auto *e = reinterpret_cast<Ee *>(ee);
+    if (!e || !ev) {
+        return UCC_ERR_INVALID_PARAM;
+    }
+    for (;;) {
+        ee_poll_pending(e);
+        if (!e->events.empty()) {
+            *ev = e->events.front();
+            return UCC_OK;
+        }
+        if (e->pending.empty()) {
+            return UCC_ERR_NOT_FOUND; /* nothing will ever arrive */
+        }
+        sched_yield();
+    }
 }
 ucc_status_t ucc_collective_triggered_post(ucc_ee_h ee, ucc_ev_t *ev)
 {
