@@ -1657,6 +1657,10 @@ class HierReduceTask final : public Task {
         if (req_h_) {
             ucc_collective_finalize(req_h_);
         }
+        if (dscr_) {
+            mc::scratch_free(dscr_, count_ * dtsz_,
+                             UCC_MEMORY_TYPE_CUDA);
+        }
     }
 
     ucc_status_t post() override
@@ -1692,8 +1696,23 @@ class HierReduceTask final : public Task {
         root_is_leader_ =
             root_leader_idx_ >= 0 &&
             team_->leader_ranks[(size_t)root_leader_idx_] == root_;
+        dev_ = mc::is_device_mt(is_root_ ? a_.dst.info.mem_type
+                                         : a_.src.info.mem_type);
         if (leader_ || on_root_node_) {
-            scratch_.resize(count_ * dtsz_);
+            if (dev_) {
+                /* node/delivery phases stay on a DEVICE scratch; only
+                 * the inter-node leader phase stages through the host */
+                if (!dscr_ &&
+                    mc::scratch_alloc(&dscr_, count_ * dtsz_,
+                                      UCC_MEMORY_TYPE_CUDA) != UCC_OK) {
+                    return UCC_ERR_NO_MEMORY;
+                }
+                if (leader_) {
+                    scratch_.resize(count_ * dtsz_);
+                }
+            } else {
+                scratch_.resize(count_ * dtsz_);
+            }
         }
         status = UCC_INPROGRESS;
         return step();
@@ -1715,6 +1734,17 @@ class HierReduceTask final : public Task {
                 if (st != UCC_OK) {
                     return st;
                 }
+                if (unstage_) { /* H2D: leaders result back to device */
+                    unstage_ = false;
+                    void *dv = root_is_leader_ ? a_.dst.info.buffer
+                                               : dscr_;
+                    ucc_status_t cs = mc::copy(
+                        dv, UCC_MEMORY_TYPE_CUDA, scratch_.data(),
+                        UCC_MEMORY_TYPE_HOST, count_ * dtsz_);
+                    if (cs != UCC_OK) {
+                        return cs;
+                    }
+                }
                 phase_++;
             }
             ucc_coll_args_t sa{};
@@ -1733,7 +1763,8 @@ class HierReduceTask final : public Task {
                     sa.src.info.mem_type = a_.dst.info.mem_type;
                 }
                 sa.dst.info          = sa.src.info;
-                sa.dst.info.buffer   = scratch_.data(); /* leader only */
+                sa.dst.info.buffer   = dev_ ? dscr_
+                                            : scratch_.data();
                 ucc_status_t st = launch(team_->node_team.get(), sa);
                 if (st != UCC_OK) {
                     return st;
@@ -1745,6 +1776,14 @@ class HierReduceTask final : public Task {
                     phase_ = 2;
                     continue;
                 }
+                if (dev_) { /* stage the node partial D2H */
+                    ucc_status_t cs = mc::copy(
+                        scratch_.data(), UCC_MEMORY_TYPE_HOST, dscr_,
+                        UCC_MEMORY_TYPE_CUDA, count_ * dtsz_);
+                    if (cs != UCC_OK) {
+                        return cs;
+                    }
+                }
                 sa.coll_type         = UCC_COLL_TYPE_REDUCE;
                 sa.root              = (uint64_t)root_leader_idx_;
                 sa.src.info.buffer   = scratch_.data();
@@ -1753,12 +1792,14 @@ class HierReduceTask final : public Task {
                 sa.src.info.mem_type = UCC_MEMORY_TYPE_HOST;
                 sa.dst.info          = sa.src.info;
                 if (team_->leaders_oob->my_idx == root_leader_idx_) {
-                    sa.dst.info.buffer = root_is_leader_
-                                             ? a_.dst.info.buffer
-                                             : scratch_.data();
-                    if (root_is_leader_) {
+                    sa.dst.info.buffer =
+                        (root_is_leader_ && !dev_)
+                            ? a_.dst.info.buffer
+                            : scratch_.data();
+                    if (root_is_leader_ && !dev_) {
                         sa.dst.info.mem_type = a_.dst.info.mem_type;
                     }
+                    unstage_ = dev_; /* H2D the result on completion */
                 }
                 ucc_status_t st = launch(team_->leaders_team.get(), sa);
                 if (st != UCC_OK) {
@@ -1775,11 +1816,14 @@ class HierReduceTask final : public Task {
                 sa.coll_type         = UCC_COLL_TYPE_BCAST;
                 sa.root              = 0;
                 sa.src.info.buffer   = is_root_ ? a_.dst.info.buffer
+                                      : dev_    ? dscr_
                                                 : scratch_.data();
                 sa.src.info.count    = count_;
                 sa.src.info.datatype = dt_;
-                sa.src.info.mem_type = is_root_ ? a_.dst.info.mem_type
-                                                : UCC_MEMORY_TYPE_HOST;
+                sa.src.info.mem_type =
+                    is_root_ ? a_.dst.info.mem_type
+                    : dev_   ? UCC_MEMORY_TYPE_CUDA
+                             : UCC_MEMORY_TYPE_HOST;
                 ucc_status_t st = launch(team_->node_team.get(), sa);
                 if (st != UCC_OK) {
                     return st;
@@ -1788,10 +1832,29 @@ class HierReduceTask final : public Task {
             }
             case 3: { /* AVG post-scale at the root */
                 if (is_root_ && a_.op == UCC_OP_AVG) {
-                    const void *srcs[1] = {a_.dst.info.buffer};
-                    ec_cpu::reduce(a_.dst.info.buffer, srcs, 1, count_,
-                                   dt_, UCC_OP_SUM,
-                                   1.0 / (double)team_->size);
+                    if (dev_) {
+                        ec_hip::ReduceArgs ra{};
+                        ra.dst     = a_.dst.info.buffer;
+                        ra.srcs[0] = a_.dst.info.buffer;
+                        ra.n_srcs  = 1;
+                        ra.count   = count_;
+                        ra.dt      = dt_;
+                        ra.op      = (ucc_reduction_op_t)12;
+                        ra.alpha   = 1.0f / (float)team_->size;
+                        ucc_status_t rs = ec_hip::reduce(ra, nullptr);
+                        if (rs != UCC_OK) {
+                            return rs;
+                        }
+                        ucc_status_t ds = mc::device_sync();
+                        if (ds != UCC_OK) {
+                            return ds;
+                        }
+                    } else {
+                        const void *srcs[1] = {a_.dst.info.buffer};
+                        ec_cpu::reduce(a_.dst.info.buffer, srcs, 1,
+                                       count_, dt_, UCC_OP_SUM,
+                                       1.0 / (double)team_->size);
+                    }
                 }
                 return UCC_OK;
             }
@@ -1822,6 +1885,8 @@ class HierReduceTask final : public Task {
     uint64_t             count_ = 0;
     size_t               dtsz_ = 4;
     ucc_datatype_t       dt_ = UCC_DT_FLOAT32;
+    bool                 dev_ = false, unstage_ = false;
+    void                *dscr_ = nullptr;
     std::vector<uint8_t> scratch_;
 };
 
@@ -1952,13 +2017,34 @@ class HierBcastTask final : public Task {
     ucc_status_t progress() override { return step(); }
 
   private:
-    ucc_status_t launch_bcast(Team *t, uint64_t root_in_t)
+    ucc_status_t launch_bcast(Team *t, uint64_t root_in_t,
+                              bool host_staged = false)
     {
         ucc_coll_args_t sa{};
         sa.mask      = UCC_COLL_ARGS_FIELD_FLAGS;
         sa.coll_type = UCC_COLL_TYPE_BCAST;
         sa.root      = root_in_t;
         sa.src.info  = a_.src.info;
+        if (host_staged) {
+            /* device memory crossing nodes: the leader phase runs on a
+             * host staging copy (D2H here, H2D when it completes) */
+            size_t bytes = a_.src.info.count *
+                           ucc_dt_size(a_.src.info.datatype);
+            hstg_.resize(bytes);
+            if ((int)root_in_t ==
+                team_->leaders_oob->my_idx) { /* I hold the data */
+                ucc_status_t cs =
+                    mc::copy(hstg_.data(), UCC_MEMORY_TYPE_HOST,
+                             a_.src.info.buffer, a_.src.info.mem_type,
+                             bytes);
+                if (cs != UCC_OK) {
+                    return cs;
+                }
+            }
+            sa.src.info.buffer   = hstg_.data();
+            sa.src.info.mem_type = UCC_MEMORY_TYPE_HOST;
+            staged_ = true;
+        }
         ucc_status_t st = ucc_collective_init(
             &sa, &req_h_, reinterpret_cast<ucc_team_h>(t));
         if (st != UCC_OK) {
@@ -1979,6 +2065,18 @@ class HierBcastTask final : public Task {
                 req_h_ = nullptr;
                 if (st != UCC_OK) {
                     return st;
+                }
+                if (staged_) { /* leaders phase on a host copy: H2D */
+                    staged_ = false;
+                    size_t bytes = a_.src.info.count *
+                                   ucc_dt_size(a_.src.info.datatype);
+                    ucc_status_t cs =
+                        mc::copy(a_.src.info.buffer,
+                                 a_.src.info.mem_type, hstg_.data(),
+                                 UCC_MEMORY_TYPE_HOST, bytes);
+                    if (cs != UCC_OK) {
+                        return cs;
+                    }
                 }
                 phase_++;
             }
@@ -2011,9 +2109,10 @@ class HierBcastTask final : public Task {
                     continue;
                 }
                 {
-                    ucc_status_t st =
-                        launch_bcast(team_->leaders_team.get(),
-                                     (uint64_t)root_leader_idx_);
+                    ucc_status_t st = launch_bcast(
+                        team_->leaders_team.get(),
+                        (uint64_t)root_leader_idx_,
+                        mc::is_device_mt(a_.src.info.mem_type));
                     if (st != UCC_OK) {
                         return st;
                     }
@@ -2049,6 +2148,8 @@ class HierBcastTask final : public Task {
     ucc_coll_req_h  req_h_ = nullptr;
     int             phase_ = 0;
     bool            leader_ = false, on_root_node_ = false;
+    bool            staged_ = false;
+    std::vector<uint8_t> hstg_;
     uint32_t        root_ = 0;
     int             my_node_idx_ = -1, root_leader_idx_ = -1;
 };
@@ -2201,6 +2302,10 @@ void add_scores(Team *team)
         return UCC_OK;
     };
     team->score_map.add(UCC_COLL_TYPE_BCAST, UCC_MEMORY_TYPE_HOST, b);
+    /* device-memory 2step bcast: node phases on the device TL, the
+     * leaders hop staged through host buffers (same pattern as
+     * rab_dev) */
+    team->score_map.add(UCC_COLL_TYPE_BCAST, UCC_MEMORY_TYPE_CUDA, b);
 
     /* node-aggregated a2av pays off while per-pair messages are small:
      * nleaders^2 sockets instead of nranks^2. Above the threshold the
@@ -2268,6 +2373,7 @@ void add_scores(Team *team)
         return UCC_OK;
     };
     team->score_map.add(UCC_COLL_TYPE_REDUCE, UCC_MEMORY_TYPE_HOST, rd);
+    team->score_map.add(UCC_COLL_TYPE_REDUCE, UCC_MEMORY_TYPE_CUDA, rd);
 
     ScoreRange ba;
     ba.start    = 0;

@@ -424,6 +424,37 @@ job.run(reqs)
 torch.cuda.synchronize()
 for d in dsts:
     torch.testing.assert_close(d.cpu(), exp, rtol=1e-5, atol=1e-5)
+# 2step bcast on device (root-node dev bcast -> staged leaders hop ->
+# node dev bcast); non-leader root
+root = 3
+bufs = [torch.zeros(123_456, device="cuda") for _ in range(n)]
+bufs[root].normal_()
+exp_b = bufs[root].cpu().clone()
+reqs = job.coll("bcast", [
+    dict(src=b.data_ptr(), dst=0, count=123_456, dt=dtypes.FLOAT32,
+         mem_type=dtypes.MEM_CUDA, root=root) for b in bufs])
+job.run(reqs)
+torch.cuda.synchronize()
+for b in bufs:
+    torch.testing.assert_close(b.cpu(), exp_b)
+
+# 2step reduce on device: leader and non-leader roots, AVG
+for root, op in ((0, dtypes.OP_SUM), (3, dtypes.OP_SUM),
+                 (3, dtypes.OP_AVG)):
+    srcs = [torch.randn(77_777, device="cuda") for _ in range(n)]
+    dst = torch.zeros(77_777, device="cuda")
+    exp_r = sum(s2.cpu() for s2 in srcs)
+    if op == dtypes.OP_AVG:
+        exp_r = exp_r / n
+    reqs = job.coll("reduce", [
+        dict(src=srcs[r].data_ptr(),
+             dst=(dst.data_ptr() if r == root else 0), count=77_777,
+             dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA, root=root,
+             op=op) for r in range(n)])
+    job.run(reqs)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(dst.cpu(), exp_r, rtol=1e-5, atol=1e-4)
+
 print("HIER_DEV_OK")
 """ % (REPO,)
 
