@@ -1873,6 +1873,105 @@ class TcpAllgatherBruckTask final : public TcpTask {
     std::vector<uint8_t> work_;
 };
 
+/* ---- neighbor-exchange allgather (reference tl/ucp allgather
+ * neighbor role, Chan et al., re-derived): EVEN n only. Round 0 pairs
+ * (2i, 2i+1) swap their own blocks; every later round alternates
+ * direction and swaps the CONTIGUOUS 2-block group acquired in the
+ * previous round — n/2 rounds total with 2-block messages, vs the
+ * ring's n-1 single-block rounds. The group locations follow a
+ * data-independent pattern, so every rank simulates all ranks' group
+ * state g[p] locally (O(n) per round) and always knows what its
+ * partner is sending. */
+class TcpAllgatherNeighborTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        if (n_ % 2 != 0 || n_ < 4) {
+            return UCC_ERR_NOT_SUPPORTED; /* odd n: bruck/ring */
+        }
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        dtsz_ = ucc_dt_size(a_.dst.info.datatype);
+        blk_  = a_.dst.info.count * dtsz_ / n_;
+        dst_  = (uint8_t *)a_.dst.info.buffer;
+        if (blk_ == 0 ||
+            blk_ * n_ != a_.dst.info.count * dtsz_) {
+            return UCC_ERR_NOT_SUPPORTED;
+        }
+        if (!inplace) {
+            memcpy(dst_ + me_ * blk_, a_.src.info.buffer, blk_);
+        }
+        g_.resize(n_);
+        round_ = 0;
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    uint32_t partner(uint32_t p, int r) const
+    {
+        const bool even = (p % 2) == 0;
+        if (r == 0) {
+            return even ? p + 1 : p - 1;
+        }
+        if (even == ((r % 2) == 1)) {
+            return (p + n_ - 1) % n_;
+        }
+        return (p + 1) % n_;
+    }
+
+    ucc_status_t progress_()
+    {
+        const int nr = (int)(n_ / 2);
+        while (round_ < nr) {
+            if (phase_ == 0) {
+                uint32_t peer = partner(me_, round_);
+                if (round_ == 0) {
+                    send_to(peer, 0, dst_ + me_ * blk_, blk_);
+                    recv_from(peer, 0, dst_ + peer * blk_, blk_);
+                } else {
+                    uint32_t sb = g_[me_], rb = g_[peer];
+                    for (uint32_t j = 0; j < 2; j++) {
+                        send_to(peer, (uint32_t)round_ * 4 + j,
+                                dst_ + ((sb + j) % n_) * blk_, blk_);
+                        recv_from(peer, (uint32_t)round_ * 4 + j,
+                                  dst_ + ((rb + j) % n_) * blk_,
+                                  blk_);
+                    }
+                }
+                phase_ = 1;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            /* advance every rank's group state */
+            if (round_ == 0) {
+                for (uint32_t q = 0; q < n_; q++) {
+                    g_[q] = q & ~1u; /* the (2i,2i+1) pair each holds */
+                }
+            } else {
+                std::vector<uint32_t> ng(n_);
+                for (uint32_t q = 0; q < n_; q++) {
+                    ng[q] = g_[partner(q, round_)];
+                }
+                g_ = std::move(ng);
+            }
+            phase_ = 0;
+            round_++;
+        }
+        return UCC_OK;
+    }
+
+    size_t   dtsz_ = 4, blk_ = 0;
+    uint8_t *dst_ = nullptr;
+    std::vector<uint32_t> g_;
+};
+
 /* ---- barrier / fanin / fanout: binomial fanin to 0 then fanout */
 class TcpBarrierTask final : public TcpTask {
   public:
@@ -3092,6 +3191,9 @@ class TcpTl final : public Tl {
                     "sliding-window allreduce windows in flight");
         cfg.declare("TL_TCP", "AG_BRUCK_MAX", "64k",
                     "Bruck allgather upper bound bytes");
+        cfg.declare("TL_TCP", "AG_NEIGHBOR_MIN", "256k",
+                    "neighbor-exchange allgather lower bound bytes "
+                    "(even team sizes only)");
         cfg.declare("TL_TCP", "A2AV_HYBRID_THRESH", "4096",
                     "alltoallv hybrid: pairs below this ride the Bruck "
                     "digit exchange (0 disables the hybrid alg)");
@@ -3330,6 +3432,29 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
                 return UCC_ERR_NOT_SUPPORTED;
             }
             *task = new TcpAllgatherBruckTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_ALLGATHER, UCC_MEMORY_TYPE_HOST, r);
+    }
+    {
+        /* neighbor exchange: n/2 rounds of 2-block swaps (even n) —
+         * halves the per-round latency count vs the ring at the same
+         * wire volume; wins for mid/large blocks on even teams */
+        size_t nmin = Config::instance().get_size(
+            "TL_TCP", "AG_NEIGHBOR_MIN", 256 * 1024);
+        ScoreRange r;
+        r.start    = nmin;
+        r.end      = SIZE_MAX;
+        r.score    = sc + 1;
+        r.tl_name  = "tcp";
+        r.alg_name = "neighbor";
+        r.init     = [self](const ucc_coll_args_t &args, Team *t2,
+                        Task **task) -> ucc_status_t {
+            if (t2->size % 2 != 0 || t2->size < 4 ||
+                !ucc_dt_is_predefined(args.dst.info.datatype)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpAllgatherNeighborTask(t2->ctx, self, args);
             return UCC_OK;
         };
         map.add(UCC_COLL_TYPE_ALLGATHER, UCC_MEMORY_TYPE_HOST, r);

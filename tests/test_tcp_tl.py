@@ -343,6 +343,47 @@ def test_allgather_bruck():
     assert p.returncode == 0 and "AG_BRUCK_OK" in p.stdout
 
 
+def test_allgather_neighbor():
+    """Neighbor-exchange allgather (reference tl/ucp allgather neighbor
+    role): n/2 rounds of 2-block swaps with alternating direction, even
+    n only; odd n must fall through to ring/bruck."""
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "from ucc_amd import core, dtypes\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "for n in (4, 6, 8, 3, 5):\n"  # odd n: selection falls back
+        "    job = LocalJob(n)\n"
+        "    if n %% 2 == 0:\n"
+        "        smap = core().score_map_str(job.teams[0])\n"
+        "        assert '@tcp/neighbor' in smap, smap\n"
+        "    rng = np.random.default_rng(n)\n"
+        "    for per in (1, 77, 5000):\n"
+        "        blks = [rng.standard_normal(per).astype(np.float32)\n"
+        "                for _ in range(n)]\n"
+        "        dsts = [np.zeros(per * n, np.float32)\n"
+        "                for _ in range(n)]\n"
+        "        reqs = job.coll('allgather', [\n"
+        "            dict(src=blks[r].ctypes.data,\n"
+        "                 dst=dsts[r].ctypes.data, count=per * n,\n"
+        "                 dt=dtypes.FLOAT32) for r in range(n)])\n"
+        "        job.run(reqs)\n"
+        "        exp = np.concatenate(blks)\n"
+        "        for d in dsts:\n"
+        "            np.testing.assert_array_equal(d, exp)\n"
+        "print('AG_NBR_OK')\n"
+    ) % (REPO,)
+    env = dict(os.environ)
+    env["UCC_TL_SHM_ENABLE"] = "0"
+    env["UCC_TUNE"] = "allgather:@neighbor:99"
+    env["UCC_TL_TCP_AG_NEIGHBOR_MIN"] = "0"
+    p = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=300)
+    sys.stdout.write(p.stdout[-500:])
+    sys.stderr.write(p.stderr[-2000:])
+    assert p.returncode == 0 and "AG_NBR_OK" in p.stdout
+
+
 def test_alltoallv_hybrid():
     """Hybrid a2av (reference alltoallv_hybrid.c role): small pairs
     aggregate through the Bruck digit exchange, large pairs go direct.
