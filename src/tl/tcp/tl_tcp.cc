@@ -1972,6 +1972,83 @@ class TcpAllgatherNeighborTask final : public TcpTask {
     std::vector<uint32_t> g_;
 };
 
+/* ---- sparbit-role allgather (reference tl/ucp allgather sparbit,
+ * re-derived): ceil(log2 n) rounds, DATA-ORDERED — every block lands
+ * directly at its absolute dst position, so there is no work buffer
+ * and no final rotation (Bruck pays both). Round k: rank p owns the
+ * circular run [p, p+2^k) of blocks, sends its first min(2^k, n-2^k)
+ * to (p-2^k) and receives the run [p+2^k, ...) from (p+2^k); runs that
+ * wrap the buffer end split into two ops. Any n. */
+class TcpAllgatherSparbitTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        dtsz_ = ucc_dt_size(a_.dst.info.datatype);
+        blk_  = a_.dst.info.count * dtsz_ / n_;
+        dst_  = (uint8_t *)a_.dst.info.buffer;
+        if (blk_ == 0 ||
+            blk_ * n_ != a_.dst.info.count * dtsz_) {
+            return UCC_ERR_NOT_SUPPORTED; /* ragged: ring handles it */
+        }
+        if (!inplace) {
+            memcpy(dst_ + me_ * blk_, a_.src.info.buffer, blk_);
+        }
+        round_ = 0;
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    /* post run [first, first+cnt) of blocks (mod n) as 1-2 wire ops */
+    void post_run(uint32_t peer, uint32_t tag, uint64_t first,
+                  uint64_t cnt, bool is_send)
+    {
+        uint64_t head = std::min(cnt, (uint64_t)n_ - first);
+        if (is_send) {
+            send_to(peer, tag, dst_ + first * blk_, head * blk_);
+            if (cnt > head) {
+                send_to(peer, tag + 1, dst_, (cnt - head) * blk_);
+            }
+        } else {
+            recv_from(peer, tag, dst_ + first * blk_, head * blk_);
+            if (cnt > head) {
+                recv_from(peer, tag + 1, dst_, (cnt - head) * blk_);
+            }
+        }
+    }
+
+    ucc_status_t progress_()
+    {
+        while ((1ull << round_) < n_) {
+            uint64_t d = 1ull << round_;
+            if (phase_ == 0) {
+                uint64_t cnt  = std::min(d, (uint64_t)n_ - d);
+                uint32_t to   = (uint32_t)((me_ + n_ - d) % n_);
+                uint32_t from = (uint32_t)((me_ + d) % n_);
+                post_run(to, (uint32_t)round_ * 4, me_, cnt, true);
+                post_run(from, (uint32_t)round_ * 4, from, cnt, false);
+                phase_ = 1;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            phase_ = 0;
+            round_++;
+        }
+        return UCC_OK;
+    }
+
+    size_t   dtsz_ = 4, blk_ = 0;
+    uint8_t *dst_ = nullptr;
+};
+
 /* ---- barrier / fanin / fanout: binomial fanin to 0 then fanout */
 class TcpBarrierTask final : public TcpTask {
   public:
@@ -3191,6 +3268,9 @@ class TcpTl final : public Tl {
                     "sliding-window allreduce windows in flight");
         cfg.declare("TL_TCP", "AG_BRUCK_MAX", "64k",
                     "Bruck allgather upper bound bytes");
+        cfg.declare("TL_TCP", "AG_SPARBIT_MAX", "64k",
+                    "sparbit (data-ordered log-round) allgather upper "
+                    "bound bytes");
         cfg.declare("TL_TCP", "AG_NEIGHBOR_MIN", "256k",
                     "neighbor-exchange allgather lower bound bytes "
                     "(even team sizes only)");
@@ -3432,6 +3512,27 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
                 return UCC_ERR_NOT_SUPPORTED;
             }
             *task = new TcpAllgatherBruckTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_ALLGATHER, UCC_MEMORY_TYPE_HOST, r);
+    }
+    {
+        /* sparbit role: log-round AND data-ordered (no work buffer, no
+         * rotation) — preferred over bruck in the small-block band */
+        size_t smax = Config::instance().get_size(
+            "TL_TCP", "AG_SPARBIT_MAX", 64 * 1024);
+        ScoreRange r;
+        r.start    = 0;
+        r.end      = smax;
+        r.score    = sc + 2;
+        r.tl_name  = "tcp";
+        r.alg_name = "sparbit";
+        r.init     = [self](const ucc_coll_args_t &args, Team *t2,
+                        Task **task) -> ucc_status_t {
+            if (!ucc_dt_is_predefined(args.dst.info.datatype)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpAllgatherSparbitTask(t2->ctx, self, args);
             return UCC_OK;
         };
         map.add(UCC_COLL_TYPE_ALLGATHER, UCC_MEMORY_TYPE_HOST, r);

@@ -343,6 +343,45 @@ def test_allgather_bruck():
     assert p.returncode == 0 and "AG_BRUCK_OK" in p.stdout
 
 
+def test_allgather_sparbit():
+    """Sparbit-role allgather (reference tl/ucp allgather sparbit):
+    ceil(log2 n) rounds, data-ordered (blocks land at absolute dst
+    positions, no work buffer or rotation), any n including the wrap
+    split of circular runs."""
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "from ucc_amd import core, dtypes\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "for n in (2, 3, 5, 6, 8, 11):\n"
+        "    job = LocalJob(n)\n"
+        "    smap = core().score_map_str(job.teams[0])\n"
+        "    assert 'allgather:host:0-65536:@tcp/sparbit' in smap, smap\n"
+        "    rng = np.random.default_rng(n)\n"
+        "    for per in (1, 77, 4000):\n"
+        "        blks = [rng.standard_normal(per).astype(np.float32)\n"
+        "                for _ in range(n)]\n"
+        "        dsts = [np.zeros(per * n, np.float32)\n"
+        "                for _ in range(n)]\n"
+        "        reqs = job.coll('allgather', [\n"
+        "            dict(src=blks[r].ctypes.data,\n"
+        "                 dst=dsts[r].ctypes.data, count=per * n,\n"
+        "                 dt=dtypes.FLOAT32) for r in range(n)])\n"
+        "        job.run(reqs)\n"
+        "        exp = np.concatenate(blks)\n"
+        "        for d in dsts:\n"
+        "            np.testing.assert_array_equal(d, exp)\n"
+        "print('AG_SPARBIT_OK')\n"
+    ) % (REPO,)
+    env = dict(os.environ)
+    env["UCC_TL_SHM_ENABLE"] = "0"
+    p = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=300)
+    sys.stdout.write(p.stdout[-500:])
+    sys.stderr.write(p.stderr[-2000:])
+    assert p.returncode == 0 and "AG_SPARBIT_OK" in p.stdout
+
+
 def test_allgather_neighbor():
     """Neighbor-exchange allgather (reference tl/ucp allgather neighbor
     role): n/2 rounds of 2-block swaps with alternating direction, even
