@@ -2778,6 +2778,201 @@ class TcpReduceScatterRingTask final : public TcpTask {
     std::vector<uint8_t> work_, tmp_;
 };
 
+/* ---- recursive-halving reduce_scatter (reference tl/ucp
+ * reduce_scatter knomial / Rabenseifner role, re-derived): log2(m)
+ * rounds; at each round the active group and its block range halve,
+ * partners exchange the half each keeps and reduce — (m-1)/m * S
+ * bytes moved in log rounds, vs the ring's n-1 rounds. Non-power-of-2
+ * n folds the first n-m odd ranks into even proxies up front
+ * (m = largest power of two <= n); because m may not divide the n
+ * output blocks, a final redistribution sends each block from the
+ * active that finished owning it to the rank the API assigns it to
+ * (a no-op when n is a power of two). */
+class TcpReduceScatterHalvingTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        dt_   = a_.dst.info.datatype;
+        op_   = a_.op == UCC_OP_AVG ? UCC_OP_SUM : a_.op;
+        dtsz_ = ucc_dt_size(dt_);
+        if (inplace) {
+            total_ = a_.dst.info.count;
+            per_   = total_ / n_;
+        } else {
+            per_   = a_.dst.info.count;
+            total_ = per_ * n_;
+        }
+        if (per_ == 0 || per_ * n_ != total_ || n_ < 2) {
+            return UCC_ERR_NOT_SUPPORTED; /* ragged: linear handles */
+        }
+        m_ = 1;
+        while (m_ * 2 <= n_) {
+            m_ *= 2;
+        }
+        nex_ = n_ - m_; /* extras: odd ranks < 2*nex_ */
+        work_.resize((size_t)total_ * dtsz_);
+        memcpy(work_.data(),
+               inplace ? a_.dst.info.buffer : a_.src.info.buffer,
+               (size_t)total_ * dtsz_);
+        tmp_.resize((size_t)total_ * dtsz_);
+        extra_  = me_ < 2 * nex_ && (me_ % 2) == 1;
+        vr_     = extra_ ? UINT32_MAX
+                         : (me_ < 2 * nex_ ? me_ / 2 : me_ - nex_);
+        lo_     = 0;
+        hi_     = n_;
+        vlo_    = 0;
+        vn_     = m_;
+        round_  = 0;
+        phase_  = 0;
+        status  = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    uint8_t *blk(uint64_t b) { return work_.data() + b * per_ * dtsz_; }
+    uint32_t real_of(uint32_t v) const
+    {
+        return v < nex_ ? 2 * v : v + nex_;
+    }
+    /* the active virtual rank that ends the halving owning block b */
+    uint32_t owner_of(uint64_t b) const
+    {
+        uint64_t lo = 0, hi = n_;
+        uint32_t vlo = 0, vn = m_;
+        while (vn > 1) {
+            uint64_t mid = lo + (hi - lo) / 2;
+            if (b < mid) {
+                hi = mid;
+            } else {
+                lo  = mid;
+                vlo += vn / 2;
+            }
+            vn /= 2;
+        }
+        return vlo;
+    }
+    void scale_avg(uint8_t *p)
+    {
+        if (a_.op == UCC_OP_AVG) {
+            const void *s[1] = {p};
+            ec_cpu::reduce(p, s, 1, per_, dt_, UCC_OP_SUM,
+                           1.0 / (double)n_);
+        }
+    }
+
+    ucc_status_t progress_()
+    {
+        /* phase A: fold extras into their even proxies */
+        if (round_ == 0 && phase_ == 0) {
+            if (extra_) {
+                send_to(me_ - 1, 1000, work_.data(),
+                        (size_t)total_ * dtsz_);
+            } else if (vr_ < nex_) {
+                recv_from(me_ + 1, 1000, tmp_.data(),
+                          (size_t)total_ * dtsz_);
+            }
+            phase_ = 1;
+        }
+        if (round_ == 0 && phase_ == 1) {
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            if (!extra_ && vr_ < nex_) {
+                const void *srcs[2] = {work_.data(), tmp_.data()};
+                ec_cpu::reduce(work_.data(), srcs, 2, total_, dt_, op_);
+            }
+            round_ = 1;
+            phase_ = 0;
+        }
+        /* phase B: recursive halving among the m_ actives */
+        if (!extra_) {
+            while (vn_ > 1) {
+                uint64_t mid    = lo_ + (hi_ - lo_) / 2;
+                uint32_t half_v = vn_ / 2;
+                bool     low    = vr_ < vlo_ + half_v;
+                uint32_t peer   = real_of(low ? vr_ + half_v
+                                              : vr_ - half_v);
+                uint64_t kb = low ? lo_ : mid;   /* keep   */
+                uint64_t ke = low ? mid : hi_;
+                uint64_t sb = low ? mid : lo_;   /* send   */
+                uint64_t se = low ? hi_ : mid;
+                if (phase_ == 0) {
+                    send_to(peer, (uint32_t)round_ * 4, blk(sb),
+                            (se - sb) * per_ * dtsz_);
+                    recv_from(peer, (uint32_t)round_ * 4, tmp_.data(),
+                              (ke - kb) * per_ * dtsz_);
+                    phase_ = 1;
+                }
+                if (!ops_done()) {
+                    return UCC_INPROGRESS;
+                }
+                clear_ops();
+                const void *srcs[2] = {blk(kb), tmp_.data()};
+                ec_cpu::reduce(blk(kb), srcs, 2,
+                               (ke - kb) * per_, dt_, op_);
+                lo_ = kb;
+                hi_ = ke;
+                if (!low) {
+                    vlo_ += half_v;
+                }
+                vn_ = half_v;
+                round_++;
+                phase_ = 0;
+            }
+        }
+        /* phase C: redistribute finished blocks to their API owners */
+        if (phase_ == 0) {
+            uint8_t *dst = (uint8_t *)a_.dst.info.buffer;
+            if (a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE) {
+                dst += (size_t)me_ * per_ * dtsz_;
+            }
+            if (!extra_) {
+                for (uint64_t b = lo_; b < hi_; b++) {
+                    if ((uint32_t)b == me_) {
+                        memcpy(dst, blk(b), per_ * dtsz_);
+                        scale_avg(dst);
+                    } else {
+                        send_to((uint32_t)b, 2000, blk(b),
+                                per_ * dtsz_);
+                    }
+                }
+            }
+            uint32_t own = real_of(owner_of(me_));
+            if (own != me_) {
+                recv_from(own, 2000, dst, per_ * dtsz_);
+                need_scale_ = true;
+            }
+            phase_ = 2;
+        }
+        if (!ops_done()) {
+            return UCC_INPROGRESS;
+        }
+        clear_ops();
+        if (need_scale_) {
+            uint8_t *dst = (uint8_t *)a_.dst.info.buffer;
+            if (a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE) {
+                dst += (size_t)me_ * per_ * dtsz_;
+            }
+            scale_avg(dst);
+        }
+        return UCC_OK;
+    }
+
+    ucc_datatype_t     dt_ = UCC_DT_FLOAT32;
+    ucc_reduction_op_t op_ = UCC_OP_SUM;
+    size_t   dtsz_ = 4;
+    uint64_t total_ = 0, per_ = 0, lo_ = 0, hi_ = 0;
+    uint32_t m_ = 1, nex_ = 0, vr_ = 0, vlo_ = 0, vn_ = 1;
+    bool     extra_ = false, need_scale_ = false;
+    std::vector<uint8_t> work_, tmp_;
+};
+
 /* ---- k-nomial gather/scatter (reference tl/ucp gather/scatter
  * knomial role, re-derived): blocks travel in VIRTUAL-rank order
  * (vr = (rank - root) mod n) so every subtree owns a contiguous range
@@ -3657,6 +3852,24 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
             return UCC_OK;
         };
         map.add(UCC_COLL_TYPE_REDUCE_SCATTER, UCC_MEMORY_TYPE_HOST, r);
+        /* recursive halving below the ring band: log2 rounds beat the
+         * ring's n-1 while the message sizes still fit latency-bound
+         * traffic (reference reduce_scatter knomial role) */
+        ScoreRange h;
+        h.start    = 0;
+        h.end      = rs_min;
+        h.score    = sc + 1;
+        h.tl_name  = "tcp";
+        h.alg_name = "knomial";
+        h.init     = [self](const ucc_coll_args_t &args, Team *t2,
+                        Task **task) -> ucc_status_t {
+            if (!ucc_dt_is_predefined(args.dst.info.datatype)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpReduceScatterHalvingTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_REDUCE_SCATTER, UCC_MEMORY_TYPE_HOST, h);
     }
     add(UCC_COLL_TYPE_REDUCE_SCATTER, mk((TcpRootedTask *)nullptr));
     add(UCC_COLL_TYPE_REDUCE_SCATTERV, mk((TcpRootedTask *)nullptr));

@@ -588,6 +588,63 @@ def test_allreduce_knomial_radix():
             f"radix {radix}"
 
 
+def test_reduce_scatter_halving():
+    """Recursive-halving reduce_scatter (reference tl/ucp rs knomial /
+    Rabenseifner role): log2(m) halving rounds with a fold for
+    non-power-of-2 n and a final block redistribution; sum + AVG."""
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "from ucc_amd import core, dtypes\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "c = core()\n"
+        "for n in (2, 3, 5, 6, 8, 11):\n"
+        "    job = LocalJob(n)\n"
+        "    smap = c.score_map_str(job.teams[0])\n"
+        "    assert '@tcp/knomial' in smap, smap\n"
+        "    rng = np.random.default_rng(n)\n"
+        "    for per in (1, 501, 4001):\n"
+        "        srcs = [(rng.random(per * n) - 0.5).astype(np.float32)\n"
+        "                for _ in range(n)]\n"
+        "        dsts = [np.zeros(per, np.float32) for _ in range(n)]\n"
+        "        reqs = job.coll('reduce_scatter', [\n"
+        "            dict(src=srcs[r].ctypes.data,\n"
+        "                 dst=dsts[r].ctypes.data,\n"
+        "                 count=per, dt=dtypes.FLOAT32)\n"
+        "            for r in range(n)])\n"
+        "        job.run(reqs)\n"
+        "        exp = np.sum(srcs, axis=0)\n"
+        "        for r in range(n):\n"
+        "            np.testing.assert_allclose(\n"
+        "                dsts[r], exp[r * per:(r + 1) * per],\n"
+        "                rtol=1e-5, atol=1e-5)\n"
+        "    # AVG (receiver-side 1/n scaling)\n"
+        "    per = 777\n"
+        "    srcs = [(rng.random(per * n) - 0.5).astype(np.float32)\n"
+        "            for _ in range(n)]\n"
+        "    dsts = [np.zeros(per, np.float32) for _ in range(n)]\n"
+        "    reqs = job.coll('reduce_scatter', [\n"
+        "        dict(src=srcs[r].ctypes.data, dst=dsts[r].ctypes.data,\n"
+        "             count=per, dt=dtypes.FLOAT32, op=dtypes.OP_AVG)\n"
+        "        for r in range(n)])\n"
+        "    job.run(reqs)\n"
+        "    exp = np.sum(srcs, axis=0)\n"
+        "    for r in range(n):\n"
+        "        np.testing.assert_allclose(\n"
+        "            dsts[r], exp[r * per:(r + 1) * per] / n,\n"
+        "            rtol=1e-5, atol=1e-5)\n"
+        "print('RS_HALVING_OK')\n"
+    ) % (REPO,)
+    env = dict(os.environ)
+    env["UCC_TL_SHM_ENABLE"] = "0"
+    env["UCC_TUNE"] = "reduce_scatter:@knomial:99"
+    p = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=600)
+    sys.stdout.write(p.stdout[-500:])
+    sys.stderr.write(p.stderr[-2000:])
+    assert p.returncode == 0 and "RS_HALVING_OK" in p.stdout
+
+
 def test_reduce_scatter_ring():
     """Ring reduce-scatter (reference tl/ucp rs ring role): n-1 rounds,
     each rank ends owning its fully reduced block; in-place + AVG."""
