@@ -2334,6 +2334,103 @@ class TcpAllgatherTask final : public TcpTask {
     uint8_t            *buf_ = nullptr;
 };
 
+/* ---- linear batched allgather(v) (reference tl/ucp allgather linear
+ * / allgather_batched_num_posts role, re-derived): every rank sends
+ * its own block DIRECTLY to all n-1 peers and receives each peer's
+ * block into its final position — one hop per block (the ring
+ * forwards each block n-1 times through intermediate ranks), at the
+ * cost of n-1 simultaneous connections. AG_LINEAR_NUM_POSTS throttles
+ * how many peers are in flight (0 = all at once). */
+class TcpAllgatherLinearTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+
+    ucc_status_t post() override
+    {
+        begin();
+        const bool is_v    = a_.coll_type == UCC_COLL_TYPE_ALLGATHERV;
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        cnt_.resize(n_);
+        dsp_.resize(n_);
+        if (is_v) {
+            size_t ds = ucc_dt_size(a_.dst.info_v.datatype);
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = ((a_.flags & UCC_COLL_ARGS_FLAG_COUNT_64BIT)
+                               ? ((const uint64_t *)
+                                      a_.dst.info_v.counts)[r]
+                               : ((const uint32_t *)
+                                      a_.dst.info_v.counts)[r]) *
+                          ds;
+                dsp_[r] =
+                    ((a_.flags & UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT)
+                         ? ((const uint64_t *)
+                                a_.dst.info_v.displacements)[r]
+                         : ((const uint32_t *)
+                                a_.dst.info_v.displacements)[r]) *
+                    ds;
+            }
+            buf_ = (uint8_t *)a_.dst.info_v.buffer;
+        } else {
+            size_t ds    = ucc_dt_size(a_.dst.info.datatype);
+            size_t block = a_.dst.info.count * ds / n_;
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = block;
+                dsp_[r] = r * block;
+            }
+            buf_ = (uint8_t *)a_.dst.info.buffer;
+        }
+        if (!inplace) {
+            ec_cpu::copy(buf_ + dsp_[me_], a_.src.info.buffer,
+                         cnt_[me_]);
+        }
+        round_ = 1;
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    ucc_status_t progress_()
+    {
+        const uint32_t np = (uint32_t)Config::instance().get_int(
+            "TL_TCP", "AG_LINEAR_NUM_POSTS", 0);
+        const uint32_t w = np == 0 ? n_ : (np < 1 ? 1 : np);
+        while (round_ < (int)n_) {
+            if (phase_ == 0) {
+                uint32_t batch = 0;
+                while (round_ + (int)batch < (int)n_ && batch < w) {
+                    int      rd   = round_ + (int)batch;
+                    uint32_t to   = (me_ + rd) % n_;
+                    uint32_t from = (me_ + n_ - rd) % n_;
+                    if (cnt_[me_]) {
+                        send_to(to, (uint32_t)rd, buf_ + dsp_[me_],
+                                cnt_[me_]);
+                    }
+                    if (cnt_[from]) {
+                        recv_from(from, (uint32_t)rd,
+                                  buf_ + dsp_[from], cnt_[from]);
+                    }
+                    batch++;
+                }
+                batch_ = (int)batch;
+                phase_ = 1;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            phase_ = 0;
+            round_ += batch_;
+        }
+        return UCC_OK;
+    }
+
+    std::vector<size_t> cnt_, dsp_;
+    uint8_t            *buf_ = nullptr;
+    int                 batch_ = 0;
+};
+
 /* ---- alltoall(v): pairwise exchange */
 /* ---- alltoallv hybrid (reference tl/ucp alltoallv_hybrid.c role,
  * re-derived): pairs >= thr move pairwise-direct; pairs < thr travel
@@ -2575,18 +2672,32 @@ class TcpAlltoallTask final : public TcpTask {
   private:
     ucc_status_t progress_()
     {
+        /* num_posts batching (reference tl/ucp alltoall pairwise
+         * num_posts role): post up to W peer pairs' sends+recvs before
+         * waiting — data moves straight between user buffers (no
+         * staging), so overlapping rounds costs no memory and hides
+         * per-peer latency. 0 = post all n-1 rounds at once. */
+        const uint32_t np = (uint32_t)Config::instance().get_int(
+            "TL_TCP", "A2A_NUM_POSTS", 0);
+        const uint32_t w = np == 0 ? n_ : (np < 1 ? 1 : np);
         while (round_ < (int)n_) {
             if (phase_ == 0) {
-                uint32_t to   = (me_ + round_) % n_;
-                uint32_t from = (me_ + n_ - round_) % n_;
-                if (scnt_[to]) {
-                    send_to(to, (uint32_t)round_, sbuf_ + sdsp_[to],
-                            scnt_[to]);
+                uint32_t batch = 0;
+                while (round_ + (int)batch < (int)n_ && batch < w) {
+                    int      rd   = round_ + (int)batch;
+                    uint32_t to   = (me_ + rd) % n_;
+                    uint32_t from = (me_ + n_ - rd) % n_;
+                    if (scnt_[to]) {
+                        send_to(to, (uint32_t)rd, sbuf_ + sdsp_[to],
+                                scnt_[to]);
+                    }
+                    if (rcnt_[from]) {
+                        recv_from(from, (uint32_t)rd,
+                                  dbuf_ + rdsp_[from], rcnt_[from]);
+                    }
+                    batch++;
                 }
-                if (rcnt_[from]) {
-                    recv_from(from, (uint32_t)round_, dbuf_ + rdsp_[from],
-                              rcnt_[from]);
-                }
+                batch_ = (int)batch;
                 phase_ = 1;
             }
             if (!ops_done()) {
@@ -2594,7 +2705,7 @@ class TcpAlltoallTask final : public TcpTask {
             }
             clear_ops();
             phase_ = 0;
-            round_++;
+            round_ += batch_;
         }
         return UCC_OK;
     }
@@ -2602,6 +2713,7 @@ class TcpAlltoallTask final : public TcpTask {
     std::vector<size_t> scnt_, sdsp_, rcnt_, rdsp_;
     const uint8_t      *sbuf_ = nullptr;
     uint8_t            *dbuf_ = nullptr;
+    int                 batch_ = 0;
 };
 
 /* ---- Bruck alltoall for small messages: ceil(log2 n) rounds of
@@ -3469,6 +3581,16 @@ class TcpTl final : public Tl {
         cfg.declare("TL_TCP", "AG_NEIGHBOR_MIN", "256k",
                     "neighbor-exchange allgather lower bound bytes "
                     "(even team sizes only)");
+        cfg.declare("TL_TCP", "AG_LINEAR_MIN", "64k",
+                    "linear direct allgather lower bound bytes");
+        cfg.declare("TL_TCP", "AG_LINEAR_MAX", "256k",
+                    "linear direct allgather upper bound bytes");
+        cfg.declare("TL_TCP", "AG_LINEAR_NUM_POSTS", "0",
+                    "linear allgather: peers posted before waiting "
+                    "(0 = all at once)");
+        cfg.declare("TL_TCP", "A2A_NUM_POSTS", "0",
+                    "pairwise alltoall(v): peer pairs posted before "
+                    "waiting (0 = all at once)");
         cfg.declare("TL_TCP", "A2AV_HYBRID_THRESH", "4096",
                     "alltoallv hybrid: pairs below this ride the Bruck "
                     "digit exchange (0 disables the hybrid alg)");
@@ -3731,6 +3853,34 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
             return UCC_OK;
         };
         map.add(UCC_COLL_TYPE_ALLGATHER, UCC_MEMORY_TYPE_HOST, r);
+    }
+    {
+        /* linear direct: one hop per block, all peers in flight —
+         * wins the band between the log-round and ring regimes */
+        size_t lmin = Config::instance().get_size(
+            "TL_TCP", "AG_LINEAR_MIN", 64 * 1024);
+        size_t lmax = Config::instance().get_size(
+            "TL_TCP", "AG_LINEAR_MAX", 256 * 1024);
+        ScoreRange r;
+        r.start    = lmin;
+        r.end      = lmax;
+        r.score    = sc + 1;
+        r.tl_name  = "tcp";
+        r.alg_name = "linear";
+        r.init     = [self](const ucc_coll_args_t &args, Team *t2,
+                        Task **task) -> ucc_status_t {
+            ucc_datatype_t dt =
+                args.coll_type == UCC_COLL_TYPE_ALLGATHERV
+                    ? args.dst.info_v.datatype
+                    : args.dst.info.datatype;
+            if (!ucc_dt_is_predefined(dt)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpAllgatherLinearTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_ALLGATHER, UCC_MEMORY_TYPE_HOST, r);
+        map.add(UCC_COLL_TYPE_ALLGATHERV, UCC_MEMORY_TYPE_HOST, r);
     }
     {
         /* neighbor exchange: n/2 rounds of 2-block swaps (even n) —

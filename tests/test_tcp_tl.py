@@ -423,6 +423,79 @@ def test_allgather_neighbor():
     assert p.returncode == 0 and "AG_NBR_OK" in p.stdout
 
 
+def test_allgather_linear_batched():
+    """Linear batched allgather(v) (reference tl/ucp allgather linear /
+    batched_num_posts role): direct one-hop sends to all peers with a
+    posting-window throttle; also exercises the A2A_NUM_POSTS window
+    on the pairwise alltoall."""
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "from ucc_amd import core, dtypes\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "for n in (3, 5, 8):\n"
+        "    job = LocalJob(n)\n"
+        "    smap = core().score_map_str(job.teams[0])\n"
+        "    assert '@tcp/linear' in smap, smap\n"
+        "    rng = np.random.default_rng(n)\n"
+        "    for per in (77, 4000):\n"
+        "        blks = [rng.standard_normal(per).astype(np.float32)\n"
+        "                for _ in range(n)]\n"
+        "        dsts = [np.zeros(per * n, np.float32)\n"
+        "                for _ in range(n)]\n"
+        "        reqs = job.coll('allgather', [\n"
+        "            dict(src=blks[r].ctypes.data,\n"
+        "                 dst=dsts[r].ctypes.data, count=per * n,\n"
+        "                 dt=dtypes.FLOAT32) for r in range(n)])\n"
+        "        job.run(reqs)\n"
+        "        exp = np.concatenate(blks)\n"
+        "        for d in dsts:\n"
+        "            np.testing.assert_array_equal(d, exp)\n"
+        "    # ragged allgatherv through the same linear task\n"
+        "    cnts = [13 * (r + 1) for r in range(n)]\n"
+        "    dspl = np.cumsum([0] + cnts[:-1]).tolist()\n"
+        "    tot = sum(cnts)\n"
+        "    blks = [rng.standard_normal(cnts[r]).astype(np.float32)\n"
+        "            for r in range(n)]\n"
+        "    dsts = [np.zeros(tot, np.float32) for _ in range(n)]\n"
+        "    reqs = job.coll('allgatherv', [\n"
+        "        dict(src=blks[r].ctypes.data, dst=dsts[r].ctypes.data,\n"
+        "             count=cnts[r], dt=dtypes.FLOAT32,\n"
+        "             dst_counts=cnts, dst_displs=dspl)\n"
+        "        for r in range(n)])\n"
+        "    job.run(reqs)\n"
+        "    exp = np.concatenate(blks)\n"
+        "    for d in dsts:\n"
+        "        np.testing.assert_array_equal(d, exp)\n"
+        "    # pairwise alltoall with a narrow posting window\n"
+        "    per = 501\n"
+        "    srcs = [rng.standard_normal(per * n).astype(np.float32)\n"
+        "            for _ in range(n)]\n"
+        "    dsts = [np.zeros(per * n, np.float32) for _ in range(n)]\n"
+        "    reqs = job.coll('alltoall', [\n"
+        "        dict(src=srcs[r].ctypes.data, dst=dsts[r].ctypes.data,\n"
+        "             count=per * n, dt=dtypes.FLOAT32)\n"
+        "        for r in range(n)])\n"
+        "    job.run(reqs)\n"
+        "    for d in range(n):\n"
+        "        exp = np.concatenate([\n"
+        "            srcs[s][d * per:(d + 1) * per] for s in range(n)])\n"
+        "        np.testing.assert_array_equal(dsts[d], exp)\n"
+        "print('AG_LINEAR_OK')\n"
+    ) % (REPO,)
+    env = dict(os.environ)
+    env["UCC_TL_SHM_ENABLE"] = "0"
+    env["UCC_TUNE"] = "allgather:@linear:99,allgatherv:@linear:99"
+    env["UCC_TL_TCP_AG_LINEAR_NUM_POSTS"] = "2"
+    env["UCC_TL_TCP_A2A_NUM_POSTS"] = "2"
+    env["UCC_TL_TCP_BRUCK_MAX"] = "0"
+    p = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=300)
+    sys.stdout.write(p.stdout[-500:])
+    sys.stderr.write(p.stderr[-2000:])
+    assert p.returncode == 0 and "AG_LINEAR_OK" in p.stdout
+
+
 def test_alltoallv_hybrid():
     """Hybrid a2av (reference alltoallv_hybrid.c role): small pairs
     aggregate through the Bruck digit exchange, large pairs go direct.
