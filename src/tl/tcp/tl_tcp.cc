@@ -1455,6 +1455,126 @@ class TcpBcastKnomialTask final : public TcpTask {
     uint8_t *buf_   = nullptr;
 };
 
+/* ---- k-nomial reduce (reference tl/ucp reduce knomial role,
+ * re-derived as the exact reverse of the k-nomial bcast tree): every
+ * rank receives each child subtree's partial sum (children of virtual
+ * rank v are v + i*q for level q < fan bound, i in [1,k)), reduces
+ * them with its own contribution, and forwards one message to the
+ * parent given by clearing the lowest nonzero base-k digit —
+ * log_k(n) depth vs the linear task's n-1 root fan. */
+class TcpReduceKnomialTask final : public TcpTask {
+  public:
+    TcpReduceKnomialTask(Context *ctx, TcpTlTeam *tt,
+                         const ucc_coll_args_t &args, uint32_t radix)
+        : TcpTask(ctx, tt, args), k_(radix < 2 ? 2 : radix)
+    {
+    }
+
+    ucc_status_t post() override
+    {
+        begin();
+        dt_    = a_.src.info.datatype;
+        op_    = a_.op == UCC_OP_AVG ? UCC_OP_SUM : a_.op;
+        count_ = a_.src.info.count;
+        dtsz_  = ucc_dt_size(dt_);
+        root_  = (uint32_t)a_.root;
+        vr_    = (me_ + n_ - root_) % n_;
+        work_.resize(count_ * dtsz_);
+        const void *src = a_.src.info.buffer;
+        if (me_ == root_ && (a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE)) {
+            src = a_.dst.info.buffer;
+        }
+        memcpy(work_.data(), src, count_ * dtsz_);
+        /* fan bound: lowest nonzero base-k digit position (the root
+         * fans at every level) — identical to the bcast tree */
+        uint64_t q = 1;
+        if (vr_ != 0) {
+            while ((vr_ / q) % k_ == 0) {
+                q *= k_;
+            }
+            uint32_t digit = (uint32_t)((vr_ / q) % k_);
+            parent_        = (vr_ - digit * (uint32_t)q + root_) % n_;
+        } else {
+            while (q < n_) {
+                q *= k_;
+            }
+            parent_ = UINT32_MAX;
+        }
+        plevel_ = q;
+        nch_    = 0;
+        for (uint64_t lv = 1; lv * k_ <= plevel_; lv *= k_) {
+            for (uint32_t i = 1; i < k_; i++) {
+                if (vr_ + i * lv < n_) {
+                    nch_++;
+                }
+            }
+        }
+        chbuf_.resize((size_t)nch_ * count_ * dtsz_);
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    ucc_status_t progress_()
+    {
+        if (phase_ == 0) { /* post all child receives */
+            uint32_t c = 0;
+            for (uint64_t lv = 1; lv * k_ <= plevel_; lv *= k_) {
+                for (uint32_t i = 1; i < k_; i++) {
+                    uint64_t child = vr_ + (uint64_t)i * lv;
+                    if (child < n_) {
+                        recv_from((uint32_t)((child + root_) % n_), 0,
+                                  chbuf_.data() +
+                                      (size_t)c * count_ * dtsz_,
+                                  count_ * dtsz_);
+                        c++;
+                    }
+                }
+            }
+            phase_ = 1;
+        }
+        if (phase_ == 1) { /* reduce children, forward to parent */
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            for (uint32_t c = 0; c < nch_; c++) {
+                const void *srcs[2] = {
+                    work_.data(),
+                    chbuf_.data() + (size_t)c * count_ * dtsz_};
+                ec_cpu::reduce(work_.data(), srcs, 2, count_, dt_, op_);
+            }
+            if (parent_ != UINT32_MAX) {
+                send_to(parent_, 0, work_.data(), count_ * dtsz_);
+            }
+            phase_ = 2;
+        }
+        if (!ops_done()) {
+            return UCC_INPROGRESS;
+        }
+        clear_ops();
+        if (me_ == root_) {
+            void *dst = a_.dst.info.buffer;
+            memcpy(dst, work_.data(), count_ * dtsz_);
+            if (a_.op == UCC_OP_AVG) {
+                const void *s[1] = {dst};
+                ec_cpu::reduce(dst, s, 1, count_, dt_, UCC_OP_SUM,
+                               1.0 / (double)n_);
+            }
+        }
+        return UCC_OK;
+    }
+
+    uint32_t k_ = 2, vr_ = 0, root_ = 0, parent_ = UINT32_MAX, nch_ = 0;
+    uint64_t plevel_ = 1, count_ = 0;
+    size_t   dtsz_ = 4;
+    ucc_datatype_t     dt_ = UCC_DT_FLOAT32;
+    ucc_reduction_op_t op_ = UCC_OP_SUM;
+    std::vector<uint8_t> work_, chbuf_;
+};
+
 /* ---- double binary tree (DBT) bcast/reduce
  * (reference coll_patterns/double_binary_tree.h:51-224 role,
  * re-derived as the two-tree scheme): the message splits in halves;
@@ -3808,6 +3928,25 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
             return UCC_OK;
         };
         map.add(UCC_COLL_TYPE_REDUCE, UCC_MEMORY_TYPE_HOST, dr);
+
+        /* k-nomial reduce below the DBT band: log_k(n) hops instead
+         * of the linear task's n-1 root fan */
+        ScoreRange rk;
+        rk.start    = 0;
+        rk.end      = dbt_min;
+        rk.score    = sc + 1;
+        rk.tl_name  = "tcp";
+        rk.alg_name = "knomial";
+        rk.init     = [self, radix](const ucc_coll_args_t &args,
+                                Team *t2, Task **task) -> ucc_status_t {
+            if ((args.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) ||
+                !ucc_dt_is_predefined(args.src.info.datatype)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpReduceKnomialTask(t2->ctx, self, args, radix);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_REDUCE, UCC_MEMORY_TYPE_HOST, rk);
     }
     add(UCC_COLL_TYPE_BCAST, mk((TcpBcastTask *)nullptr));
     add(UCC_COLL_TYPE_BARRIER, mk((TcpBarrierTask *)nullptr));

@@ -614,6 +614,57 @@ def test_gather_scatter_knomial():
             f"radix {radix}"
 
 
+def test_reduce_knomial_radix():
+    """Radix-k knomial reduce (reference tl/ucp reduce knomial role):
+    children's subtree sums climb the reversed bcast tree; every root,
+    radices 2/3/4, odd/even n, AVG and in-place root."""
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "from ucc_amd import core, dtypes\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "for n in (2, 5, 7, 8):\n"
+        "    job = LocalJob(n)\n"
+        "    rng = np.random.default_rng(n)\n"
+        "    count = 1003\n"
+        "    for root in range(n):\n"
+        "        srcs = [(rng.random(count) - 0.5).astype(np.float32)\n"
+        "                for _ in range(n)]\n"
+        "        dst = np.zeros(count, np.float32)\n"
+        "        reqs = job.coll('reduce', [\n"
+        "            dict(src=srcs[r].ctypes.data,\n"
+        "                 dst=dst.ctypes.data if r == root else 0,\n"
+        "                 count=count, dt=dtypes.FLOAT32, root=root)\n"
+        "            for r in range(n)])\n"
+        "        job.run(reqs)\n"
+        "        np.testing.assert_allclose(dst, np.sum(srcs, axis=0),\n"
+        "                                   rtol=1e-5, atol=1e-5)\n"
+        "    # AVG at root 1\n"
+        "    srcs = [(rng.random(count) - 0.5).astype(np.float32)\n"
+        "            for _ in range(n)]\n"
+        "    dst = np.zeros(count, np.float32)\n"
+        "    reqs = job.coll('reduce', [\n"
+        "        dict(src=srcs[r].ctypes.data,\n"
+        "             dst=dst.ctypes.data if r == 1 %% n else 0,\n"
+        "             count=count, dt=dtypes.FLOAT32, root=1 %% n,\n"
+        "             op=dtypes.OP_AVG) for r in range(n)])\n"
+        "    job.run(reqs)\n"
+        "    np.testing.assert_allclose(dst, np.mean(srcs, axis=0),\n"
+        "                               rtol=1e-5, atol=1e-6)\n"
+        "print('REDUCE_KN_OK')\n"
+    ) % (REPO,)
+    for radix in ("2", "3", "4"):
+        env = dict(os.environ)
+        env["UCC_TL_SHM_ENABLE"] = "0"
+        env["UCC_TUNE"] = "reduce:@knomial:99"
+        env["UCC_TL_TCP_KN_RADIX"] = radix
+        p = subprocess.run([sys.executable, "-c", code], env=env,
+                           capture_output=True, text=True, timeout=300)
+        sys.stdout.write(p.stdout[-300:])
+        sys.stderr.write(p.stderr[-2000:])
+        assert p.returncode == 0 and "REDUCE_KN_OK" in p.stdout, radix
+
+
 def test_allreduce_knomial_radix():
     """Radix-k knomial allreduce (reference allreduce_knomial +
     recursive_knomial.h PROXY/EXTRA role): non-power-of-k team sizes
