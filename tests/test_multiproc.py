@@ -336,3 +336,68 @@ def test_multiproc_ulysses_head_seq_exchange():
     switch for sequence-parallel attention, verified against the
     logical relayout of a reference full tensor."""
     _run(_worker_ulysses, world=4)
+
+
+def _worker_pp_ring(rank, world, port, q):
+    try:
+        import torch
+        import torch.distributed as dist
+
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from ucc_amd.parallel import Communicator
+        from ucc_amd.parallel.pp import PipelineStage
+        from ucc_amd.parallel.ring_attention import RingKV
+
+        comm = Communicator()
+        # --- PP: activations forward, grads backward, two microbatches
+        pp = PipelineStage(comm)
+        for mb in range(2):
+            acts = torch.full((64,), float(rank * 10 + mb))
+            pp.recv_forward(acts)
+            if rank > 0:
+                exp = float((rank - 1) * 10 + mb)
+                assert torch.all(acts == exp), (rank, mb, acts[0])
+            acts = torch.full((64,), float(rank * 10 + mb))
+            pp.send_forward(acts)
+            grads = torch.full((32,), float(rank * 100 + mb))
+            pp.recv_backward(grads)
+            if rank < world - 1:
+                exp = float((rank + 1) * 100 + mb)
+                assert torch.all(grads == exp), (rank, mb, grads[0])
+            grads = torch.full((32,), float(rank * 100 + mb))
+            pp.send_backward(grads)
+        # stage-weight bcast from rank 0
+        w = torch.full((16,), float(rank))
+        pp.broadcast_stage_weights([w], root=0)
+        assert torch.all(w == 0.0)
+        # --- ring attention: full KV sweep visits every shard once
+        ring = RingKV(comm)
+        kv = torch.full((128,), float(rank))
+        seen = []
+        ring.ring_steps(kv.clone(), lambda s, owner: seen.append(
+            (owner, float(s[0]), bool(torch.all(s == s[0])))))
+        assert len(seen) == world
+        for owner, val, uniform in seen:
+            assert uniform and val == float(owner), seen
+        # displacement=2 rotation
+        out = ring.rotate(kv, displacement=2)
+        assert torch.all(out == float((rank - 2) % world)), out[0]
+        # CP gather
+        full = ring.gather(kv)
+        exp = torch.cat([torch.full((128,), float(r))
+                         for r in range(world)])
+        assert torch.equal(full, exp)
+        q.put((rank, "ok"))
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        q.put((rank, f"fail: {e!r}"))
+
+
+def test_multiproc_pp_and_ring_attention():
+    """SURVEY §2.9 PP + CP/ring-attention workloads: p2p stage edges
+    over active-set bcast (forward/backward microbatches), stage-weight
+    bcast, and ring-attention KV rotation as sparse alltoallv hops plus
+    the CP allgather, over 4 gloo-bootstrapped processes."""
+    _run(_worker_pp_ring, world=4)
