@@ -1461,6 +1461,9 @@ class HierAllgathervTask final : public Task {
         if (req_h_) {
             ucc_collective_finalize(req_h_);
         }
+        if (dscr_) {
+            mc::scratch_free(dscr_, total_, UCC_MEMORY_TYPE_CUDA);
+        }
     }
 
     ucc_status_t post() override
@@ -1470,6 +1473,7 @@ class HierAllgathervTask final : public Task {
         inplace_ = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
         dt_      = a_.dst.info_v.datatype;
         dtsz_    = ucc_dt_size(dt_);
+        dev_     = mc::is_device_mt(a_.dst.info_v.mem_type);
         const uint32_t n = team_->size;
         /* node-major packed layout: nodes in leader order, members in
          * team-rank order within each node */
@@ -1503,6 +1507,17 @@ class HierAllgathervTask final : public Task {
             return UCC_ERR_INVALID_PARAM;
         }
         packed_.resize(total_);
+        if (dev_) { /* device run: packed layout lives in device
+                     * scratch; the host packed_ stages only the
+                     * leaders' inter-node hop (rab_dev pattern) */
+            void *p = nullptr;
+            ucc_status_t st =
+                mc::scratch_alloc(&p, total_, UCC_MEMORY_TYPE_CUDA);
+            if (st != UCC_OK) {
+                return st;
+            }
+            dscr_ = (uint8_t *)p;
+        }
         status = UCC_INPROGRESS;
         return step();
     }
@@ -1540,7 +1555,8 @@ class HierAllgathervTask final : public Task {
                 sa.src.info.buffer   = (void *)my_src;
                 sa.src.info.count    = cnt_[team_->rank] / dtsz_;
                 sa.src.info.datatype = dt_;
-                sa.src.info.mem_type = UCC_MEMORY_TYPE_HOST;
+                sa.src.info.mem_type =
+                    dev_ ? UCC_MEMORY_TYPE_CUDA : UCC_MEMORY_TYPE_HOST;
                 if (leader_) {
                     const auto  &nr = team_->node_ranks;
                     sub_cnt_.resize(nr.size());
@@ -1550,12 +1566,14 @@ class HierAllgathervTask final : public Task {
                         sub_cnt_[j] = cnt_[nr[j]] / dtsz_;
                         sub_dsp_[j] = (pdsp_[nr[j]] - base) / dtsz_;
                     }
-                    sa.dst.info_v.buffer = packed_.data() + base;
+                    sa.dst.info_v.buffer =
+                        (dev_ ? dscr_ : packed_.data()) + base;
                     sa.dst.info_v.counts = (ucc_count_t *)sub_cnt_.data();
                     sa.dst.info_v.displacements =
                         (ucc_aint_t *)sub_dsp_.data();
                     sa.dst.info_v.datatype = dt_;
-                    sa.dst.info_v.mem_type = UCC_MEMORY_TYPE_HOST;
+                    sa.dst.info_v.mem_type = dev_ ? UCC_MEMORY_TYPE_CUDA
+                                                  : UCC_MEMORY_TYPE_HOST;
                 }
                 ucc_status_t st = launch(team_->node_team.get(), sa);
                 if (st != UCC_OK) {
@@ -1568,6 +1586,16 @@ class HierAllgathervTask final : public Task {
                 if (!leader_) {
                     phase_ = 2;
                     continue;
+                }
+                if (dev_) { /* stage my node's packed block D2H */
+                    size_t       base = node_poff_[(size_t)my_node_idx_];
+                    ucc_status_t cs   = mc::copy(
+                        packed_.data() + base, UCC_MEMORY_TYPE_HOST,
+                        dscr_ + base, UCC_MEMORY_TYPE_CUDA,
+                        node_bytes_[(size_t)my_node_idx_]);
+                    if (cs != UCC_OK) {
+                        return cs;
+                    }
                 }
                 const size_t nl = team_->leader_ranks.size();
                 sub_cnt_.resize(nl);
@@ -1590,12 +1618,21 @@ class HierAllgathervTask final : public Task {
                 break;
             }
             case 2: { /* node bcast of the full packed vector */
+                if (dev_ && leader_) { /* unstage the gathered total */
+                    ucc_status_t cs = mc::copy(
+                        dscr_, UCC_MEMORY_TYPE_CUDA, packed_.data(),
+                        UCC_MEMORY_TYPE_HOST, total_);
+                    if (cs != UCC_OK) {
+                        return cs;
+                    }
+                }
                 sa.coll_type         = UCC_COLL_TYPE_BCAST;
                 sa.root              = 0;
-                sa.src.info.buffer   = packed_.data();
+                sa.src.info.buffer   = dev_ ? dscr_ : packed_.data();
                 sa.src.info.count    = total_ / dtsz_;
                 sa.src.info.datatype = dt_;
-                sa.src.info.mem_type = UCC_MEMORY_TYPE_HOST;
+                sa.src.info.mem_type =
+                    dev_ ? UCC_MEMORY_TYPE_CUDA : UCC_MEMORY_TYPE_HOST;
                 ucc_status_t st = launch(team_->node_team.get(), sa);
                 if (st != UCC_OK) {
                     return st;
@@ -1605,7 +1642,18 @@ class HierAllgathervTask final : public Task {
             case 3: { /* local unpack into user dst positions */
                 uint8_t *dst = (uint8_t *)a_.dst.info_v.buffer;
                 for (uint32_t r = 0; r < team_->size; r++) {
-                    if (cnt_[r]) {
+                    if (!cnt_[r]) {
+                        continue;
+                    }
+                    if (dev_) {
+                        ucc_status_t cs = mc::copy(
+                            dst + udsp_[r], UCC_MEMORY_TYPE_CUDA,
+                            dscr_ + pdsp_[r], UCC_MEMORY_TYPE_CUDA,
+                            cnt_[r]);
+                        if (cs != UCC_OK) {
+                            return cs;
+                        }
+                    } else {
                         memcpy(dst + udsp_[r], packed_.data() + pdsp_[r],
                                cnt_[r]);
                     }
@@ -1632,7 +1680,9 @@ class HierAllgathervTask final : public Task {
     ucc_coll_args_t       a_;
     ucc_coll_req_h        req_h_ = nullptr;
     int                   phase_ = 0, my_node_idx_ = -1;
-    bool                  leader_ = false, inplace_ = false;
+    bool                  leader_ = false, inplace_ = false,
+                          dev_ = false;
+    uint8_t              *dscr_ = nullptr;
     size_t                dtsz_ = 4, total_ = 0;
     ucc_datatype_t        dt_ = UCC_DT_FLOAT32;
     std::vector<size_t>   cnt_, udsp_, pdsp_, node_bytes_, node_poff_;
@@ -2350,6 +2400,12 @@ void add_scores(Team *team)
         return UCC_OK;
     };
     team->score_map.add(UCC_COLL_TYPE_ALLGATHERV, UCC_MEMORY_TYPE_HOST,
+                        ag);
+    /* device-memory allgatherv: node gatherv + node bcast on the
+     * device TL with the packed layout in device scratch; only the
+     * leaders' inter-node allgatherv hop stages D2H/H2D (same shape
+     * as rab_dev / the 2step device variants) */
+    team->score_map.add(UCC_COLL_TYPE_ALLGATHERV, UCC_MEMORY_TYPE_CUDA,
                         ag);
 
     ScoreRange rd;

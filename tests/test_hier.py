@@ -455,6 +455,27 @@ for root, op in ((0, dtypes.OP_SUM), (3, dtypes.OP_SUM),
     torch.cuda.synchronize()
     torch.testing.assert_close(dst.cpu(), exp_r, rtol=1e-5, atol=1e-4)
 
+# device allgatherv (node-packed): ragged counts incl. a zero block;
+# node gatherv + node bcast run on the device TL, the leaders hop is
+# staged D2H/H2D
+cnts = [1000 * (r + 1) for r in range(n)]
+cnts[1] = 0
+dsps = np.concatenate([[0], np.cumsum(cnts)[:-1]]).astype(np.uint64)
+total = int(sum(cnts))
+g_srcs = [torch.randn(max(cnts[r], 1), device="cuda")
+          for r in range(n)]
+g_dsts = [torch.zeros(total, device="cuda") for _ in range(n)]
+reqs = job.coll("allgatherv", [
+    dict(src=g_srcs[r].data_ptr(), dst=g_dsts[r].data_ptr(),
+         count=cnts[r], dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA,
+         dst_counts=cnts, dst_displs=dsps.tolist())
+    for r in range(n)])
+job.run(reqs)
+torch.cuda.synchronize()
+exp_g = torch.cat([g_srcs[r][:cnts[r]].cpu() for r in range(n)])
+for d in g_dsts:
+    torch.testing.assert_close(d.cpu(), exp_g)
+
 print("HIER_DEV_OK")
 """ % (REPO,)
 
