@@ -331,6 +331,17 @@ ucc_status_t ucc_context_create(ucc_lib_h lib_h,
                 0x9e3779b97f4a7c15ull * (1 + idx % (uint64_t)k);
         }
     }
+    /* test hook: UCC_FAKE_SOCKET_SPLIT=k spreads in-process contexts
+     * over k pseudo-sockets of one node, so socket-aware staging (shm
+     * two-level bcast, SOCKET sbgps) is exercisable anywhere */
+    if (const char *ss = getenv("UCC_FAKE_SOCKET_SPLIT")) {
+        int k = atoi(ss);
+        if (k > 1) {
+            static std::atomic<uint64_t> g_fake_sck{0};
+            ctx->proc.socket_id =
+                (int16_t)(g_fake_sck.fetch_add(1) % (uint64_t)k);
+        }
+    }
     for (Tl *tl : tl_registry()) {
         TlContext *tlc = tl->context_create(ctx);
         if (tlc) {

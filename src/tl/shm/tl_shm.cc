@@ -74,6 +74,9 @@ class ShmCollTask final : public Task {
     size_t   round_ = 0, nrounds_ = 0, cell_ = 0;
     int      phase_ = 0;
     double   alpha_ = 1.0;
+    /* socket-aware bcast staging */
+    bool     sck_ = false, sck_leader_ = false, root_sck_ = true;
+    uint32_t sck_ldr_ = 0;
 };
 
 /* -------------------------------------------------------------- TL team */
@@ -246,6 +249,35 @@ ucc_status_t ShmCollTask::post()
         dbuf_    = (uint8_t *)a_.src.info.buffer;
         sbuf_    = dbuf_;
         nrounds_ = (total_ + chunk - 1) / chunk;
+        /* socket-aware two-level staging (reference ucc_sbgp SOCKET
+         * kind consumption): when the node team spans CPU sockets,
+         * one leader per non-root socket copies each chunk of the
+         * root's result area into its own data area and its socket
+         * peers read THAT — one cross-socket transfer per socket per
+         * chunk instead of one per remote reader. */
+        if (Config::instance().get_bool("TL_SHM", "SOCKET_STAGING",
+                                        true)) {
+            const auto &procs = team->procs;
+            int16_t     rsck  = procs[a_.root].socket_id;
+            bool        known = rsck >= 0;
+            bool        multi = false;
+            sck_ldr_          = me_;
+            for (uint32_t r = 0; r < n_ && known; r++) {
+                if (procs[r].socket_id < 0) {
+                    known = false;
+                }
+                if (procs[r].socket_id != rsck) {
+                    multi = true;
+                }
+                if (procs[r].socket_id == procs[me_].socket_id &&
+                    r < sck_ldr_) {
+                    sck_ldr_ = r;
+                }
+            }
+            sck_        = known && multi;
+            root_sck_   = procs[me_].socket_id == rsck;
+            sck_leader_ = sck_ && !root_sck_ && sck_ldr_ == me_;
+        }
         break;
     case UCC_COLL_TYPE_BARRIER:
     case UCC_COLL_TYPE_FANIN:
@@ -660,6 +692,16 @@ ucc_status_t ShmCollTask::prog_bcast()
                 if (j >= 2 && !all_ge(2 * (j - 2) + 2)) {
                     return UCC_INPROGRESS;
                 }
+                if (sck_leader_) {
+                    /* relay: ONE cross-socket read of the root chunk,
+                     * written to my dst and my data area for my
+                     * socket's peers (they poll my 2j+1 publication) */
+                    if (!root_ge(2 * j + 1)) {
+                        return UCC_INPROGRESS;
+                    }
+                    memcpy(dbuf_ + off, seg.result(slot_, p), len);
+                    memcpy(seg.data(slot_, p, me_), dbuf_ + off, len);
+                }
             }
             publish(2 * j + 1);
             phase_ = 1;
@@ -669,6 +711,14 @@ ucc_status_t ShmCollTask::prog_bcast()
                 if (!all_ge(2 * j + 1)) {
                     return UCC_INPROGRESS;
                 }
+            } else if (sck_ && !root_sck_) {
+                if (!sck_leader_) { /* read my socket leader's relay */
+                    if (!rank_ge(sck_ldr_, 2 * j + 1)) {
+                        return UCC_INPROGRESS;
+                    }
+                    memcpy(dbuf_ + off, seg.data(slot_, p, sck_ldr_),
+                           len);
+                } /* leader already copied in phase 0 */
             } else {
                 if (!root_ge(2 * j + 1)) {
                     return UCC_INPROGRESS;
@@ -886,7 +936,17 @@ class ShmTl final : public Tl {
     int         default_score() const override { return 40; }
     TlContext  *context_create(Context *ctx) override
     {
-        if (!Config::instance().get_bool("TL_SHM", "ENABLE", true)) {
+        auto &cfg = Config::instance();
+        cfg.declare("TL_SHM", "ENABLE", "1", "enable the shm TL");
+        cfg.declare("TL_SHM", "CHUNK_SIZE", "512k",
+                    "shm staging chunk bytes per round");
+        cfg.declare("TL_SHM", "MAX_CONCURRENT", "4",
+                    "concurrent shm collective slots");
+        cfg.declare("TL_SHM", "SOCKET_STAGING", "1",
+                    "bcast: relay chunks through one leader per "
+                    "non-root CPU socket (one cross-socket read per "
+                    "socket instead of per rank)");
+        if (!cfg.get_bool("TL_SHM", "ENABLE", true)) {
             return nullptr;
         }
         return new ShmTlContext(ctx, this);

@@ -6,11 +6,16 @@ parameterized over team sizes x dtypes x ops, validated against host
 golden buffers.
 """
 
+import os
+import sys
+
 import numpy as np
 import pytest
 
 from ucc_amd import dtypes
 from ucc_amd.testing import LocalJob
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 SIZES = [1, 2, 3, 8]  # 1 = self TL (zero-transport rank)
@@ -487,3 +492,46 @@ def test_gatherv_rootonly_vargs(job):
     job.run(reqs)
     for r in range(n):
         np.testing.assert_array_equal(sdsts[r], srcs[r])
+
+
+def test_shm_socket_staged_bcast():
+    """Socket-aware shm bcast (reference ucc_sbgp SOCKET consumption):
+    with UCC_FAKE_SOCKET_SPLIT the node team spans pseudo-sockets, so
+    chunks relay through one leader per non-root socket. Correctness
+    across roots, sizes spanning multiple chunks, and a mid-team
+    leader; plus a run with staging disabled for identical results."""
+    import subprocess
+
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "from ucc_amd import core, dtypes\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "n = 6\n"
+        "job = LocalJob(n)\n"
+        "info = core().topo_sbgps(job.teams[0])\n"
+        "assert info['socket_leaders_size'] == 3, info\n"
+        "rng = np.random.default_rng(5)\n"
+        "for root in (0, 1, 4):\n"
+        "    for count in (63, 5000, 300_000):\n"
+        "        bufs = [np.zeros(count, np.float64) for _ in range(n)]\n"
+        "        bufs[root][:] = rng.random(count)\n"
+        "        exp = bufs[root].copy()\n"
+        "        reqs = job.coll('bcast', [\n"
+        "            dict(src=b.ctypes.data, dst=0, count=count,\n"
+        "                 dt=dtypes.FLOAT64, root=root) for b in bufs])\n"
+        "        job.run(reqs)\n"
+        "        for b in bufs:\n"
+        "            np.testing.assert_array_equal(b, exp)\n"
+        "print('SCK_BCAST_OK')\n"
+    ) % (REPO,)
+    for staging in ("1", "0"):
+        env = dict(os.environ)
+        env["UCC_FAKE_SOCKET_SPLIT"] = "3"
+        env["UCC_TL_SHM_SOCKET_STAGING"] = staging
+        env["UCC_TL_SHM_CHUNK_SIZE"] = "65536"
+        p = subprocess.run([sys.executable, "-c", code], env=env,
+                           capture_output=True, text=True, timeout=300)
+        sys.stdout.write(p.stdout[-300:])
+        sys.stderr.write(p.stderr[-2000:])
+        assert p.returncode == 0 and "SCK_BCAST_OK" in p.stdout, staging
