@@ -3695,16 +3695,17 @@ class TcpTl final : public Tl {
                     "sliding-window allreduce windows in flight");
         cfg.declare("TL_TCP", "AG_BRUCK_MAX", "64k",
                     "Bruck allgather upper bound bytes");
-        cfg.declare("TL_TCP", "AG_SPARBIT_MAX", "64k",
+        cfg.declare("TL_TCP", "AG_SPARBIT_MAX", "512k",
                     "sparbit (data-ordered log-round) allgather upper "
                     "bound bytes");
         cfg.declare("TL_TCP", "AG_NEIGHBOR_MIN", "256k",
                     "neighbor-exchange allgather lower bound bytes "
                     "(even team sizes only)");
-        cfg.declare("TL_TCP", "AG_LINEAR_MIN", "64k",
+        cfg.declare("TL_TCP", "AG_LINEAR_MIN", "512k",
                     "linear direct allgather lower bound bytes");
-        cfg.declare("TL_TCP", "AG_LINEAR_MAX", "256k",
-                    "linear direct allgather upper bound bytes");
+        cfg.declare("TL_TCP", "AG_LINEAR_MAX", "0",
+                    "linear direct allgather upper bound bytes "
+                    "(0 = unbounded)");
         cfg.declare("TL_TCP", "AG_LINEAR_NUM_POSTS", "0",
                     "linear allgather: peers posted before waiting "
                     "(0 = all at once)");
@@ -3959,12 +3960,21 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
         ScoreRange r;
         r.start    = 0;
         r.end      = bmax;
-        r.score    = sc + 1;
+        r.score    = sc + 3; /* measured: beats sparbit below 64k
+                                (profiles/tcp_host_crossovers_r02.md) */
         r.tl_name  = "tcp";
         r.alg_name = "bruck";
         r.init     = [self](const ucc_coll_args_t &args, Team *t2,
                         Task **task) -> ucc_status_t {
             if (!ucc_dt_is_predefined(args.dst.info.datatype)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            /* equal blocks required: ragged totals fall back at
+             * INIT time (post-time rejection has no fallback chain) */
+            if ((args.dst.info.count *
+                 ucc_dt_size(args.dst.info.datatype)) %
+                    t2->size !=
+                0) {
                 return UCC_ERR_NOT_SUPPORTED;
             }
             *task = new TcpAllgatherBruckTask(t2->ctx, self, args);
@@ -3976,16 +3986,24 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
         /* sparbit role: log-round AND data-ordered (no work buffer, no
          * rotation) — preferred over bruck in the small-block band */
         size_t smax = Config::instance().get_size(
-            "TL_TCP", "AG_SPARBIT_MAX", 64 * 1024);
+            "TL_TCP", "AG_SPARBIT_MAX", 512 * 1024);
         ScoreRange r;
         r.start    = 0;
         r.end      = smax;
-        r.score    = sc + 2;
+        r.score    = sc + 2; /* wins 64k-512k in the measured sweep */
         r.tl_name  = "tcp";
         r.alg_name = "sparbit";
         r.init     = [self](const ucc_coll_args_t &args, Team *t2,
                         Task **task) -> ucc_status_t {
             if (!ucc_dt_is_predefined(args.dst.info.datatype)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            /* equal blocks required: ragged totals fall back at
+             * INIT time (post-time rejection has no fallback chain) */
+            if ((args.dst.info.count *
+                 ucc_dt_size(args.dst.info.datatype)) %
+                    t2->size !=
+                0) {
                 return UCC_ERR_NOT_SUPPORTED;
             }
             *task = new TcpAllgatherSparbitTask(t2->ctx, self, args);
@@ -3997,13 +4015,17 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
         /* linear direct: one hop per block, all peers in flight —
          * wins the band between the log-round and ring regimes */
         size_t lmin = Config::instance().get_size(
-            "TL_TCP", "AG_LINEAR_MIN", 64 * 1024);
+            "TL_TCP", "AG_LINEAR_MIN", 512 * 1024);
         size_t lmax = Config::instance().get_size(
-            "TL_TCP", "AG_LINEAR_MAX", 256 * 1024);
+            "TL_TCP", "AG_LINEAR_MAX", 0); /* 0 = unbounded */
+        if (lmax == 0) {
+            lmax = SIZE_MAX;
+        }
         ScoreRange r;
         r.start    = lmin;
         r.end      = lmax;
-        r.score    = sc + 1;
+        r.score    = sc + 2; /* one hop per block: measured 2-2.7x over
+                                the ring forwarding path >= 1 MiB */
         r.tl_name  = "tcp";
         r.alg_name = "linear";
         r.init     = [self](const ucc_coll_args_t &args, Team *t2,
@@ -4014,6 +4036,11 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
                     : args.dst.info.datatype;
             if (!ucc_dt_is_predefined(dt)) {
                 return UCC_ERR_NOT_SUPPORTED;
+            }
+            if (args.coll_type == UCC_COLL_TYPE_ALLGATHER &&
+                (args.dst.info.count * ucc_dt_size(dt)) % t2->size !=
+                    0) {
+                return UCC_ERR_NOT_SUPPORTED; /* ragged: default task */
             }
             *task = new TcpAllgatherLinearTask(t2->ctx, self, args);
             return UCC_OK;
@@ -4037,6 +4064,14 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
                         Task **task) -> ucc_status_t {
             if (t2->size % 2 != 0 || t2->size < 4 ||
                 !ucc_dt_is_predefined(args.dst.info.datatype)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            /* equal blocks required: ragged totals fall back at
+             * INIT time (post-time rejection has no fallback chain) */
+            if ((args.dst.info.count *
+                 ucc_dt_size(args.dst.info.datatype)) %
+                    t2->size !=
+                0) {
                 return UCC_ERR_NOT_SUPPORTED;
             }
             *task = new TcpAllgatherNeighborTask(t2->ctx, self, args);

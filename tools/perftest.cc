@@ -315,6 +315,25 @@ static ucc_coll_args_t make_args(const Opts &o, Rank &r, size_t bytes,
     size_t             count = bytes / ds;
     ucc_memory_type_t  mt = o.mem == "cuda" ? UCC_MEMORY_TYPE_CUDA
                                             : UCC_MEMORY_TYPE_HOST;
+    /* block colls need count divisible by nranks (equal blocks is the
+     * allgather/alltoall contract; reference perftest sizes the same
+     * way) — round down, keeping at least one element per rank */
+    switch (ct) {
+    case UCC_COLL_TYPE_ALLGATHER:
+    case UCC_COLL_TYPE_ALLTOALL:
+    case UCC_COLL_TYPE_GATHER:
+    case UCC_COLL_TYPE_SCATTER:
+    case UCC_COLL_TYPE_REDUCE_SCATTER:
+    case UCC_COLL_TYPE_ALLGATHERV:
+    case UCC_COLL_TYPE_ALLTOALLV:
+        count -= count % (size_t)nranks;
+        if (count == 0) {
+            count = (size_t)nranks;
+        }
+        break;
+    default:
+        break;
+    }
     ucc_coll_args_t a{};
     a.mask      = UCC_COLL_ARGS_FIELD_FLAGS;
     a.flags     = (o.persistent ? UCC_COLL_ARGS_FLAG_PERSISTENT : 0) |
@@ -621,9 +640,16 @@ static int run_forked_child(const Opts &o, int rank)
                 return;
             }
 #endif
-            ucc_collective_post(req);
-            while (ucc_collective_test(req) == UCC_INPROGRESS) {
+            if (ucc_collective_post(req) != UCC_OK) {
+                trig_fail = true; /* reuse the unsupported-row path */
+                return;
+            }
+            ucc_status_t ts;
+            while ((ts = ucc_collective_test(req)) == UCC_INPROGRESS) {
                 ucc_context_progress(r.ctx);
+            }
+            if (ts != UCC_OK) {
+                trig_fail = true;
             }
         };
         if (o.check &&
@@ -638,8 +664,8 @@ static int run_forked_child(const Opts &o, int rank)
         }
         if (trig_fail) {
             if (rank == 0) {
-                printf("%12zu        - triggered unsupported at this "
-                       "size\n", bytes);
+                printf("%12zu        - unsupported at this size\n",
+                       bytes);
             }
             ucc_collective_finalize(req);
 #ifdef UCC_AMD_HAS_HIP
@@ -666,6 +692,13 @@ static int run_forked_child(const Opts &o, int rank)
         }
 #endif
         double t  = g_shm_oob.max_double(now_s() - t0) / iters;
+        if (trig_fail) { /* a timed iteration errored: no number */
+            if (rank == 0) {
+                printf("%12zu        - failed during timing\n", bytes);
+            }
+            ucc_collective_finalize(req);
+            continue;
+        }
         ucc_collective_finalize(req);
 #ifdef UCC_AMD_HAS_HIP
         if (ee) {
@@ -771,23 +804,36 @@ static int run_inproc(const Opts &o)
             printf("%12zu        - coll_init unsupported\n", bytes);
             continue;
         }
+        bool coll_fail = false;
         auto iter = [&]() {
             for (int i = 0; i < n; i++) {
-                ucc_collective_post(reqs[i]);
+                if (ucc_collective_post(reqs[i]) != UCC_OK) {
+                    coll_fail = true;
+                }
             }
             bool done = false;
             while (!done) {
                 done = true;
                 for (int i = 0; i < n; i++) {
-                    if (ucc_collective_test(reqs[i]) == UCC_INPROGRESS) {
+                    ucc_status_t ts = ucc_collective_test(reqs[i]);
+                    if (ts == UCC_INPROGRESS) {
                         done = false;
                         ucc_context_progress(ranks[i].ctx);
+                    } else if (ts != UCC_OK) {
+                        coll_fail = true;
                     }
                 }
             }
         };
-        for (int i = 0; i < o.warmup; i++) {
+        for (int i = 0; i < o.warmup && !coll_fail; i++) {
             iter();
+        }
+        if (coll_fail) {
+            printf("%12zu        - unsupported at this size\n", bytes);
+            for (int i = 0; i < n; i++) {
+                ucc_collective_finalize(reqs[i]);
+            }
+            continue;
         }
 #ifdef UCC_AMD_HAS_HIP
         if (o.mem == "cuda") {
@@ -806,6 +852,10 @@ static int run_inproc(const Opts &o)
         double t = (now_s() - t0) / iters;
         for (int i = 0; i < n; i++) {
             ucc_collective_finalize(reqs[i]);
+        }
+        if (coll_fail) {
+            printf("%12zu        - failed during timing\n", bytes);
+            continue;
         }
         double algbw = bytes / t / 1e9;
         double busbw = algbw * busbw_factor(ct, n);
