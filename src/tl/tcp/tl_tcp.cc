@@ -183,6 +183,16 @@ static void set_nonblock(int fd)
     fcntl(fd, F_SETFL, fcntl(fd, F_GETFL, 0) | O_NONBLOCK);
     int one = 1;
     setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+    /* large-message throughput: the ~208KB kernel default stalls the
+     * single-pass progress loop once per buffer-full (reference UCX
+     * sockcm sizes buffers up the same way); knob in bytes, 0 keeps
+     * the kernel default */
+    int sb = (int)Config::instance().get_size("TL_TCP", "SOCKBUF",
+                                              4 * 1024 * 1024);
+    if (sb > 0) {
+        setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &sb, sizeof(sb));
+        setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &sb, sizeof(sb));
+    }
 }
 
 class TcpTl;
@@ -3701,6 +3711,8 @@ class TcpTl final : public Tl {
         cfg.declare("TL_TCP", "AG_NEIGHBOR_MIN", "256k",
                     "neighbor-exchange allgather lower bound bytes "
                     "(even team sizes only)");
+        cfg.declare("TL_TCP", "SOCKBUF", "4m",
+                    "SO_SNDBUF/SO_RCVBUF bytes (0 = kernel default)");
         cfg.declare("TL_TCP", "AG_LINEAR_MIN", "512k",
                     "linear direct allgather lower bound bytes");
         cfg.declare("TL_TCP", "AG_LINEAR_MAX", "0",
