@@ -550,3 +550,30 @@ def test_gatherv_device_rootonly_vargs(job):
     ])
     torch.testing.assert_close(gdst.cpu(),
                                torch.cat([s.cpu() for s in srcs]))
+
+
+def test_device_size1_team():
+    """Size-1 team on device memory: the self TL serves every coll as a
+    device copy/no-op (reference tl/self role for n=1 GPU jobs)."""
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    torch.cuda.set_device(0)
+    j1 = LocalJob(1)
+    smap = core().score_map_str(j1.teams[0])
+    assert "@self" in smap, smap
+    src = torch.randn(4096, device="cuda")
+    dst = torch.zeros(4096, device="cuda")
+    for coll in ("allreduce", "allgather", "alltoall",
+                 "reduce_scatter"):
+        dst.zero_()
+        _run_device(j1, coll, [
+            dict(src=src.data_ptr(), dst=dst.data_ptr(), count=4096,
+                 dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA)])
+        torch.testing.assert_close(dst, src)
+    # bcast: in-place single rank, must be a no-op that completes
+    _run_device(j1, "bcast", [
+        dict(src=src.data_ptr(), dst=0, count=4096, dt=dtypes.FLOAT32,
+             mem_type=dtypes.MEM_CUDA)])
+    # barrier
+    _run_device(j1, "barrier", [
+        dict(src=0, dst=0, count=0, dt=dtypes.FLOAT32)])
