@@ -320,6 +320,114 @@ class RcclTask final : public Task {
             RCCLCHK(ncclGroupEnd());
             return UCC_OK;
         }
+        case UCC_COLL_TYPE_ALLGATHERV: {
+            /* no NCCL primitive: grouped p2p ring of own-block sends
+             * (reference tl_nccl allgatherv role) */
+            size_t ds  = ucc_dt_size(a_.dst.info_v.datatype);
+            auto  *db  = (uint8_t *)a_.dst.info_v.buffer;
+            size_t mine = vcnt(a_, a_.dst.info_v.counts, me) * ds;
+            const void *src =
+                inplace ? db + vdsp(a_, a_.dst.info_v.displacements, me) *
+                                   ds
+                        : a_.src.info.buffer;
+            RCCLCHK(ncclGroupStart());
+            for (uint32_t r = 0; r < n; r++) {
+                RCCLCHK(ncclSend(src, mine, ncclInt8, (int)r, comm, s));
+                RCCLCHK(ncclRecv(
+                    db + vdsp(a_, a_.dst.info_v.displacements, r) * ds,
+                    vcnt(a_, a_.dst.info_v.counts, r) * ds, ncclInt8,
+                    (int)r, comm, s));
+            }
+            RCCLCHK(ncclGroupEnd());
+            return UCC_OK;
+        }
+        case UCC_COLL_TYPE_GATHER:
+        case UCC_COLL_TYPE_GATHERV: {
+            const bool is_v = a_.coll_type == UCC_COLL_TYPE_GATHERV;
+            size_t     ds   = ucc_dt_size(
+                me == a_.root && !is_v ? a_.dst.info.datatype
+                                           : a_.src.info.datatype);
+            RCCLCHK(ncclGroupStart());
+            if (me == a_.root) {
+                auto *db = (uint8_t *)(is_v ? a_.dst.info_v.buffer
+                                            : a_.dst.info.buffer);
+                size_t per =
+                    is_v ? 0 : a_.dst.info.count / n * ds;
+                for (uint32_t r = 0; r < n; r++) {
+                    uint8_t *dst =
+                        is_v ? db + vdsp(a_, a_.dst.info_v.displacements,
+                                         r) *
+                                        ds
+                             : db + (size_t)r * per;
+                    size_t len =
+                        is_v ? vcnt(a_, a_.dst.info_v.counts, r) * ds
+                             : per;
+                    RCCLCHK(ncclRecv(dst, len, ncclInt8, (int)r, comm,
+                                     s));
+                }
+            }
+            {
+                const void *src =
+                    (inplace && me == a_.root)
+                        ? nullptr /* in-place root: block already home */
+                        : a_.src.info.buffer;
+                size_t len = a_.src.info.count * ds;
+                if (inplace && me == a_.root) {
+                    /* self send still required to satisfy the posted
+                     * recv: source it from the dst block */
+                    auto *db = (uint8_t *)(is_v ? a_.dst.info_v.buffer
+                                                : a_.dst.info.buffer);
+                    src = is_v ? db + vdsp(a_,
+                                           a_.dst.info_v.displacements,
+                                           me) *
+                                          ds
+                               : db + (size_t)me *
+                                          (a_.dst.info.count / n) * ds;
+                    len = is_v ? vcnt(a_, a_.dst.info_v.counts, me) * ds
+                               : a_.dst.info.count / n * ds;
+                }
+                RCCLCHK(ncclSend(src, len, ncclInt8, (int)a_.root, comm,
+                                 s));
+            }
+            RCCLCHK(ncclGroupEnd());
+            return UCC_OK;
+        }
+        case UCC_COLL_TYPE_SCATTER:
+        case UCC_COLL_TYPE_SCATTERV: {
+            const bool is_v = a_.coll_type == UCC_COLL_TYPE_SCATTERV;
+            size_t     ds   = ucc_dt_size(
+                me == a_.root && !is_v ? a_.src.info.datatype
+                                           : a_.dst.info.datatype);
+            RCCLCHK(ncclGroupStart());
+            if (me == a_.root) {
+                auto *sb = (const uint8_t *)(is_v ? a_.src.info_v.buffer
+                                                  : a_.src.info.buffer);
+                size_t per = is_v ? 0 : a_.src.info.count / n * ds;
+                for (uint32_t r = 0; r < n; r++) {
+                    if (inplace && r == me) {
+                        continue; /* root's block stays in place */
+                    }
+                    const uint8_t *src =
+                        is_v ? sb + vdsp(a_, a_.src.info_v.displacements,
+                                         r) *
+                                        ds
+                             : sb + (size_t)r * per;
+                    size_t len =
+                        is_v ? vcnt(a_, a_.src.info_v.counts, r) * ds
+                             : per;
+                    RCCLCHK(ncclSend(src, len, ncclInt8, (int)r, comm,
+                                     s));
+                }
+            }
+            if (!(inplace && me == a_.root)) {
+                void  *dst = a_.dst.info.buffer;
+                size_t len = a_.dst.info.count * ds;
+                RCCLCHK(ncclRecv(dst, len, ncclInt8, (int)a_.root, comm,
+                                 s));
+            }
+            RCCLCHK(ncclGroupEnd());
+            return UCC_OK;
+        }
         default:
             return UCC_ERR_NOT_SUPPORTED;
         }
@@ -402,6 +510,11 @@ void RcclTlTeam::get_scores(Team *team, ScoreMap &map)
     add(UCC_COLL_TYPE_BARRIER);
     add(UCC_COLL_TYPE_ALLTOALL);
     add(UCC_COLL_TYPE_ALLTOALLV);
+    add(UCC_COLL_TYPE_ALLGATHERV);
+    add(UCC_COLL_TYPE_GATHER);
+    add(UCC_COLL_TYPE_GATHERV);
+    add(UCC_COLL_TYPE_SCATTER);
+    add(UCC_COLL_TYPE_SCATTERV);
 }
 
 } // namespace
