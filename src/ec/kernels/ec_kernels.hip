@@ -84,6 +84,46 @@ __device__ __forceinline__ void d_f_to_e5m2x2(float a, float b,
     lo    = (uint8_t)w;
     hi    = (uint8_t)(w >> 8);
 }
+/* dword-wide converts: the pk builtins take a word-select operand, so
+ * one dword (4 fp8) costs exactly 2 VOPs each way with NO byte
+ * repacking VALU (the paired path above still assembles a 16-bit
+ * packed operand from two byte registers per pair). */
+__device__ __forceinline__ void d_e4m3x4_to_f(uint32_t w, float &a,
+                                              float &b, float &c,
+                                              float &d)
+{
+    auto lo = __builtin_amdgcn_cvt_pk_f32_fp8((int)w, false);
+    auto hi = __builtin_amdgcn_cvt_pk_f32_fp8((int)w, true);
+    a       = lo[0];
+    b       = lo[1];
+    c       = hi[0];
+    d       = hi[1];
+}
+__device__ __forceinline__ uint32_t d_f_to_e4m3x4(float a, float b,
+                                                  float c, float d)
+{
+    int w = __builtin_amdgcn_cvt_pk_fp8_f32(a, b, 0, false);
+    w     = __builtin_amdgcn_cvt_pk_fp8_f32(c, d, w, true);
+    return (uint32_t)w;
+}
+__device__ __forceinline__ void d_e5m2x4_to_f(uint32_t w, float &a,
+                                              float &b, float &c,
+                                              float &d)
+{
+    auto lo = __builtin_amdgcn_cvt_pk_f32_bf8((int)w, false);
+    auto hi = __builtin_amdgcn_cvt_pk_f32_bf8((int)w, true);
+    a       = lo[0];
+    b       = lo[1];
+    c       = hi[0];
+    d       = hi[1];
+}
+__device__ __forceinline__ uint32_t d_f_to_e5m2x4(float a, float b,
+                                                  float c, float d)
+{
+    int w = __builtin_amdgcn_cvt_pk_bf8_f32(a, b, 0, false);
+    w     = __builtin_amdgcn_cvt_pk_bf8_f32(c, d, w, true);
+    return (uint32_t)w;
+}
 __device__ __forceinline__ float d_e4m3_to_f(uint8_t v)
 {
     return __builtin_amdgcn_cvt_f32_fp8((int)v, 0);
@@ -340,22 +380,60 @@ __device__ __forceinline__ void cvt2_store(float a, float b, T &lo, T &hi)
         d_f_to_e5m2x2(a, b, lo.v, hi.v);
     }
 }
+template <typename T>
+__device__ __forceinline__ void cvt4_load(uint32_t w, float &a, float &b,
+                                          float &c, float &d)
+{
+    if constexpr (std::is_same<T, e4m3_t>::value) {
+        d_e4m3x4_to_f(w, a, b, c, d);
+    } else {
+        d_e5m2x4_to_f(w, a, b, c, d);
+    }
+}
+template <typename T>
+__device__ __forceinline__ uint32_t cvt4_store(float a, float b, float c,
+                                               float d)
+{
+    if constexpr (std::is_same<T, e4m3_t>::value) {
+        return d_f_to_e4m3x4(a, b, c, d);
+    }
+    return d_f_to_e5m2x4(a, b, c, d);
+}
 #else
 template <typename T> inline constexpr bool is_fp8_v = false;
 template <typename T>
 __device__ __forceinline__ void cvt2_load(T, T, float &, float &) {}
 template <typename T>
 __device__ __forceinline__ void cvt2_store(float, float, T &, T &) {}
+template <typename T>
+__device__ __forceinline__ void cvt4_load(uint32_t, float &, float &,
+                                          float &, float &)
+{
+}
+template <typename T>
+__device__ __forceinline__ uint32_t cvt4_store(float, float, float,
+                                               float)
+{
+    return 0;
+}
 #endif
 
-/* vector-wide load/accumulate/store with paired fp8 fast paths */
+/* vector-wide load/accumulate/store with dword-wide fp8 fast paths
+ * (4 elements per word-selected pk convert pair, zero repack VALU) */
+template <int N> struct __align__(16) PackW {
+    uint32_t w[N];
+};
+
 template <typename T, int OP, int VEC, typename A, typename P>
 __device__ __forceinline__ void vload(const P &x, A (&r)[VEC])
 {
-    if constexpr (is_fp8_v<T> && std::is_same<A, float>::value) {
+    if constexpr (is_fp8_v<T> && std::is_same<A, float>::value &&
+                  VEC % 4 == 0) {
+        auto pw = __builtin_bit_cast(PackW<VEC / 4>, x);
 #pragma unroll
-        for (int k = 0; k < VEC; k += 2) {
-            cvt2_load<T>(x.v[k], x.v[k + 1], r[k], r[k + 1]);
+        for (int j = 0; j < VEC / 4; j++) {
+            cvt4_load<T>(pw.w[j], r[4 * j], r[4 * j + 1], r[4 * j + 2],
+                         r[4 * j + 3]);
         }
     } else {
 #pragma unroll
@@ -367,13 +445,17 @@ __device__ __forceinline__ void vload(const P &x, A (&r)[VEC])
 template <typename T, int OP, int VEC, typename A, typename P>
 __device__ __forceinline__ void vaccum(A (&r)[VEC], const P &x)
 {
-    if constexpr (is_fp8_v<T> && std::is_same<A, float>::value) {
+    if constexpr (is_fp8_v<T> && std::is_same<A, float>::value &&
+                  VEC % 4 == 0) {
+        auto pw = __builtin_bit_cast(PackW<VEC / 4>, x);
 #pragma unroll
-        for (int k = 0; k < VEC; k += 2) {
-            float a, b;
-            cvt2_load<T>(x.v[k], x.v[k + 1], a, b);
-            r[k]     = red<A, OP>(r[k], a);
-            r[k + 1] = red<A, OP>(r[k + 1], b);
+        for (int j = 0; j < VEC / 4; j++) {
+            float a, b, c, d;
+            cvt4_load<T>(pw.w[j], a, b, c, d);
+            r[4 * j]     = red<A, OP>(r[4 * j], a);
+            r[4 * j + 1] = red<A, OP>(r[4 * j + 1], b);
+            r[4 * j + 2] = red<A, OP>(r[4 * j + 2], c);
+            r[4 * j + 3] = red<A, OP>(r[4 * j + 3], d);
         }
     } else {
 #pragma unroll
@@ -386,13 +468,17 @@ template <typename T, int OP, int VEC, typename A, typename P>
 __device__ __forceinline__ void vstore(P &o, const A (&r)[VEC],
                                        float alpha)
 {
-    if constexpr (is_fp8_v<T> && std::is_same<A, float>::value) {
+    if constexpr (is_fp8_v<T> && std::is_same<A, float>::value &&
+                  VEC % 4 == 0) {
+        PackW<VEC / 4> pw;
 #pragma unroll
-        for (int k = 0; k < VEC; k += 2) {
-            cvt2_store<T>(apply_alpha<A>(r[k], alpha),
-                          apply_alpha<A>(r[k + 1], alpha), o.v[k],
-                          o.v[k + 1]);
+        for (int j = 0; j < VEC / 4; j++) {
+            pw.w[j] = cvt4_store<T>(apply_alpha<A>(r[4 * j], alpha),
+                                    apply_alpha<A>(r[4 * j + 1], alpha),
+                                    apply_alpha<A>(r[4 * j + 2], alpha),
+                                    apply_alpha<A>(r[4 * j + 3], alpha));
         }
+        o = __builtin_bit_cast(P, pw);
     } else {
 #pragma unroll
         for (int k = 0; k < VEC; k++) {
