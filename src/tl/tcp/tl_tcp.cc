@@ -2949,24 +2949,57 @@ class TcpReduceScatterRingTask final : public TcpTask {
     {
         begin();
         const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
-        dt_   = a_.dst.info.datatype;
+        const bool is_v = a_.coll_type == UCC_COLL_TYPE_REDUCE_SCATTERV;
+        dt_   = is_v ? a_.dst.info_v.datatype : a_.dst.info.datatype;
         op_   = a_.op == UCC_OP_AVG ? UCC_OP_SUM : a_.op;
         dtsz_ = ucc_dt_size(dt_);
-        if (inplace) {
-            total_ = a_.dst.info.count;
-            per_   = total_ / n_;
+        cnt_.resize(n_);
+        dsp_.resize(n_);
+        uint64_t maxb = 0;
+        if (is_v) {
+            /* v-variant (reference reduce_scatterv ring role): block r
+             * is counts[r] elements, laid out contiguously in counts
+             * order in the source vector */
+            size_t off = 0;
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = ((a_.flags & UCC_COLL_ARGS_FLAG_COUNT_64BIT)
+                               ? ((const uint64_t *)
+                                      a_.dst.info_v.counts)[r]
+                               : ((const uint32_t *)
+                                      a_.dst.info_v.counts)[r]) *
+                          dtsz_;
+                dsp_[r] = off;
+                off += cnt_[r];
+                maxb = cnt_[r] > maxb ? cnt_[r] : maxb;
+            }
+            total_ = off;
         } else {
-            per_   = a_.dst.info.count;
-            total_ = per_ * n_;
+            uint64_t per =
+                (inplace ? a_.dst.info.count / n_ : a_.dst.info.count) *
+                dtsz_;
+            total_ = per * n_;
+            if (per == 0 ||
+                (inplace && per * n_ != a_.dst.info.count * dtsz_)) {
+                return UCC_ERR_NOT_SUPPORTED; /* ragged: linear */
+            }
+            for (uint32_t r = 0; r < n_; r++) {
+                cnt_[r] = per;
+                dsp_[r] = (size_t)r * per;
+            }
+            maxb = per;
         }
-        if (per_ == 0 || per_ * n_ != total_) {
-            return UCC_ERR_NOT_SUPPORTED; /* ragged: linear handles */
+        if (total_ == 0) {
+            return UCC_ERR_NOT_SUPPORTED;
         }
-        work_.resize(total_ * dtsz_);
-        memcpy(work_.data(),
-               inplace ? a_.dst.info.buffer : a_.src.info.buffer,
-               total_ * dtsz_);
-        tmp_.resize(per_ * dtsz_);
+        work_.resize(total_);
+        const void *src;
+        if (is_v) {
+            src = inplace ? a_.dst.info_v.buffer : a_.src.info.buffer;
+        } else {
+            src = inplace ? a_.dst.info.buffer : a_.src.info.buffer;
+        }
+        memcpy(work_.data(), src, total_);
+        tmp_.resize(maxb);
         round_ = 0;
         phase_ = 0;
         status = UCC_INPROGRESS;
@@ -2975,7 +3008,7 @@ class TcpReduceScatterRingTask final : public TcpTask {
     ucc_status_t progress() override { return progress_(); }
 
   private:
-    uint8_t *blk(uint32_t b) { return work_.data() + (size_t)b * per_ * dtsz_; }
+    uint8_t *blk(uint32_t b) { return work_.data() + dsp_[b]; }
 
     ucc_status_t progress_()
     {
@@ -2984,9 +3017,10 @@ class TcpReduceScatterRingTask final : public TcpTask {
         while (round_ < (int)n_ - 1) {
             if (phase_ == 0) {
                 uint32_t sb = (me_ + 2 * n_ - 1 - (uint32_t)round_) % n_;
-                send_to(right, (uint32_t)round_, blk(sb), per_ * dtsz_);
+                uint32_t rb = (me_ + 2 * n_ - 2 - (uint32_t)round_) % n_;
+                send_to(right, (uint32_t)round_, blk(sb), cnt_[sb]);
                 recv_from(left, (uint32_t)round_, tmp_.data(),
-                          per_ * dtsz_);
+                          cnt_[rb]);
                 phase_ = 1;
             }
             if (!ops_done()) {
@@ -2994,21 +3028,25 @@ class TcpReduceScatterRingTask final : public TcpTask {
             }
             clear_ops();
             uint32_t rb = (me_ + 2 * n_ - 2 - (uint32_t)round_) % n_;
-            const void *srcs[2] = {blk(rb), tmp_.data()};
-            ec_cpu::reduce(blk(rb), srcs, 2, per_, dt_, op_);
+            if (cnt_[rb]) {
+                const void *srcs[2] = {blk(rb), tmp_.data()};
+                ec_cpu::reduce(blk(rb), srcs, 2, cnt_[rb] / dtsz_, dt_,
+                               op_);
+            }
             phase_ = 0;
             round_++;
         }
-        uint8_t *dst = (uint8_t *)a_.dst.info.buffer;
+        const bool is_v = a_.coll_type == UCC_COLL_TYPE_REDUCE_SCATTERV;
+        uint8_t *dst = (uint8_t *)(is_v ? a_.dst.info_v.buffer
+                                        : a_.dst.info.buffer);
         if (a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE) {
-            dst = (uint8_t *)a_.dst.info.buffer + (size_t)me_ * per_ *
-                                                      dtsz_;
+            dst += dsp_[me_];
         }
-        memcpy(dst, blk(me_), per_ * dtsz_);
-        if (a_.op == UCC_OP_AVG) {
+        memcpy(dst, blk(me_), cnt_[me_]);
+        if (a_.op == UCC_OP_AVG && cnt_[me_]) {
             const void *srcs[1] = {dst};
-            ec_cpu::reduce(dst, srcs, 1, per_, dt_, UCC_OP_SUM,
-                           1.0 / (double)n_);
+            ec_cpu::reduce(dst, srcs, 1, cnt_[me_] / dtsz_, dt_,
+                           UCC_OP_SUM, 1.0 / (double)n_);
         }
         return UCC_OK;
     }
@@ -3016,7 +3054,8 @@ class TcpReduceScatterRingTask final : public TcpTask {
     ucc_datatype_t     dt_ = UCC_DT_FLOAT32;
     ucc_reduction_op_t op_ = UCC_OP_SUM;
     size_t   dtsz_ = 4;
-    uint64_t total_ = 0, per_ = 0;
+    uint64_t total_ = 0;
+    std::vector<size_t>  cnt_, dsp_;
     std::vector<uint8_t> work_, tmp_;
 };
 
@@ -4188,6 +4227,19 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
             return UCC_OK;
         };
         map.add(UCC_COLL_TYPE_REDUCE_SCATTER, UCC_MEMORY_TYPE_HOST, r);
+        /* v-variant through the same ring engine (per-block v sizes,
+         * reference reduce_scatterv ring role) */
+        ScoreRange rv = r;
+        rv.init       = [self](const ucc_coll_args_t &args, Team *t2,
+                         Task **task) -> ucc_status_t {
+            if (!ucc_dt_is_predefined(args.dst.info_v.datatype)) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpReduceScatterRingTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_REDUCE_SCATTERV, UCC_MEMORY_TYPE_HOST,
+                rv);
         /* recursive halving below the ring band: log2 rounds beat the
          * ring's n-1 while the message sizes still fit latency-bound
          * traffic (reference reduce_scatter knomial role) */
