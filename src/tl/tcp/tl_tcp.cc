@@ -2989,7 +2989,7 @@ class TcpReduceScatterRingTask final : public TcpTask {
             maxb = per;
         }
         if (total_ == 0) {
-            return UCC_ERR_NOT_SUPPORTED;
+            return UCC_OK; /* zero-length vector: nothing to move */
         }
         work_.resize(total_);
         const void *src;
@@ -4220,8 +4220,10 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
         r.alg_name = "ring";
         r.init     = [self](const ucc_coll_args_t &args, Team *t2,
                         Task **task) -> ucc_status_t {
-            if (!ucc_dt_is_predefined(args.dst.info.datatype)) {
-                return UCC_ERR_NOT_SUPPORTED;
+            if (!ucc_dt_is_predefined(args.dst.info.datatype) ||
+                ((args.flags & UCC_COLL_ARGS_FLAG_IN_PLACE) &&
+                 args.dst.info.count % t2->size != 0)) {
+                return UCC_ERR_NOT_SUPPORTED; /* ragged: linear */
             }
             *task = new TcpReduceScatterRingTask(t2->ctx, self, args);
             return UCC_OK;

@@ -712,6 +712,55 @@ def test_allreduce_knomial_radix():
             f"radix {radix}"
 
 
+def test_reduce_scatterv_ring():
+    """Ring reduce_scatterv (reference reduce_scatterv ring role): the
+    ring engine with per-block v sizes — ragged counts incl. zero
+    blocks, AVG, odd/even n."""
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "from ucc_amd import core, dtypes\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "for n in (2, 3, 5, 8):\n"
+        "    job = LocalJob(n)\n"
+        "    rng = np.random.default_rng(n)\n"
+        "    cnts = [137 * (r + 1) for r in range(n)]\n"
+        "    cnts[n // 2] = 0\n"
+        "    tot = sum(cnts)\n"
+        "    dsp = np.cumsum([0] + cnts[:-1]).tolist()\n"
+        "    srcs = [(rng.random(tot) - 0.5).astype(np.float32)\n"
+        "            for _ in range(n)]\n"
+        "    dsts = [np.zeros(max(cnts[r], 1), np.float32)\n"
+        "            for r in range(n)]\n"
+        "    for op, scale in ((dtypes.OP_SUM, 1.0),\n"
+        "                      (dtypes.OP_AVG, 1.0 / n)):\n"
+        "        reqs = job.coll('reduce_scatterv', [\n"
+        "            dict(src=srcs[r].ctypes.data,\n"
+        "                 dst=dsts[r].ctypes.data, count=cnts[r],\n"
+        "                 dt=dtypes.FLOAT32, op=op,\n"
+        "                 dst_counts=cnts, dst_displs=dsp)\n"
+        "            for r in range(n)])\n"
+        "        job.run(reqs)\n"
+        "        exp = np.sum(srcs, axis=0) * scale\n"
+        "        for r in range(n):\n"
+        "            if cnts[r]:\n"
+        "                np.testing.assert_allclose(\n"
+        "                    dsts[r][:cnts[r]],\n"
+        "                    exp[dsp[r]:dsp[r] + cnts[r]],\n"
+        "                    rtol=1e-5, atol=1e-5)\n"
+        "print('RSV_RING_OK')\n"
+    ) % (REPO,)
+    env = dict(os.environ)
+    env["UCC_TL_SHM_ENABLE"] = "0"
+    env["UCC_TUNE"] = "reduce_scatterv:@ring:99"
+    env["UCC_TL_TCP_RS_RING_MIN"] = "0"
+    p = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=300)
+    sys.stdout.write(p.stdout[-400:])
+    sys.stderr.write(p.stderr[-2000:])
+    assert p.returncode == 0 and "RSV_RING_OK" in p.stdout
+
+
 def test_reduce_scatter_halving():
     """Recursive-halving reduce_scatter (reference tl/ucp rs knomial /
     Rabenseifner role): log2(m) halving rounds with a fold for
