@@ -146,6 +146,7 @@ asan: build/asan_generic_dt build/asan_obj_size build/asan_perftest
 	ASAN_OPTIONS=detect_leaks=1 UCC_TL_SHM_ENABLE=0 UCC_TUNE=reduce_scatter:@ring:99 UCC_TL_TCP_RS_RING_MIN=256 ./build/asan_perftest -c reduce_scatter -j 5 -b 1024 -e 65536 -n 2 -w 1
 	ASAN_OPTIONS=detect_leaks=1 UCC_TL_SHM_ENABLE=0 UCC_TUNE=reduce_scatter:@knomial:99 ./build/asan_perftest -c reduce_scatter -j 7 -b 256 -e 65536 -n 2 -w 1
 	ASAN_OPTIONS=detect_leaks=1 UCC_FAKE_SOCKET_SPLIT=3 UCC_TL_SHM_CHUNK_SIZE=4096 ./build/asan_perftest -c bcast -j 6 -b 1024 -e 65536 -n 2 -w 1
+	ASAN_OPTIONS=detect_leaks=1 UCC_FAKE_SOCKET_SPLIT=3 UCC_TL_SHM_CHUNK_SIZE=4096 ./build/asan_perftest -c allreduce -j 6 -b 1024 -e 65536 -n 2 -w 1
 
 # ---------------------------------------------------------------- TSAN
 # Host ThreadSanitizer build + the native MT test (lock-free progress

@@ -48,6 +48,33 @@ class ShmCollTask final : public Task {
     inline bool rank_ge(uint32_t r, uint64_t k);
     inline bool root_ge(uint64_t k) { return rank_ge((uint32_t)a_.root, k); }
 
+    void calc_socket(const Team *team, uint32_t root)
+    {
+        if (!Config::instance().get_bool("TL_SHM", "SOCKET_STAGING",
+                                         true)) {
+            return;
+        }
+        const auto &procs = team->procs;
+        int16_t     rsck  = procs[root].socket_id;
+        bool        known = rsck >= 0;
+        bool        multi = false;
+        sck_ldr_          = me_;
+        for (uint32_t r = 0; r < n_ && known; r++) {
+            if (procs[r].socket_id < 0) {
+                known = false;
+            }
+            if (procs[r].socket_id != rsck) {
+                multi = true;
+            }
+            if (procs[r].socket_id == procs[me_].socket_id &&
+                r < sck_ldr_) {
+                sck_ldr_ = r;
+            }
+        }
+        sck_      = known && multi;
+        root_sck_ = procs[me_].socket_id == rsck;
+    }
+
     ucc_status_t prog_allreduce();
     ucc_status_t prog_reduce_scatter(); /* also v */
     ucc_status_t prog_allgather();      /* also v */
@@ -227,6 +254,11 @@ ucc_status_t ShmCollTask::post()
             alpha_ = 1.0 / n_;
         }
         nrounds_ = (total_ + chunk - 1) / chunk;
+        /* phase-C result reads relay through one leader per socket
+         * (the result area is written scattered by every rank, so all
+         * sockets benefit symmetrically) */
+        calc_socket(team, me_);
+        sck_leader_ = sck_ && sck_ldr_ == me_;
         break;
     case UCC_COLL_TYPE_REDUCE:
         dt_    = a_.src.info.datatype;
@@ -255,29 +287,8 @@ ucc_status_t ShmCollTask::post()
          * root's result area into its own data area and its socket
          * peers read THAT — one cross-socket transfer per socket per
          * chunk instead of one per remote reader. */
-        if (Config::instance().get_bool("TL_SHM", "SOCKET_STAGING",
-                                        true)) {
-            const auto &procs = team->procs;
-            int16_t     rsck  = procs[a_.root].socket_id;
-            bool        known = rsck >= 0;
-            bool        multi = false;
-            sck_ldr_          = me_;
-            for (uint32_t r = 0; r < n_ && known; r++) {
-                if (procs[r].socket_id < 0) {
-                    known = false;
-                }
-                if (procs[r].socket_id != rsck) {
-                    multi = true;
-                }
-                if (procs[r].socket_id == procs[me_].socket_id &&
-                    r < sck_ldr_) {
-                    sck_ldr_ = r;
-                }
-            }
-            sck_        = known && multi;
-            root_sck_   = procs[me_].socket_id == rsck;
-            sck_leader_ = sck_ && !root_sck_ && sck_ldr_ == me_;
-        }
+        calc_socket(team, (uint32_t)a_.root);
+        sck_leader_ = sck_ && !root_sck_ && sck_ldr_ == me_;
         break;
     case UCC_COLL_TYPE_BARRIER:
     case UCC_COLL_TYPE_FANIN:
@@ -520,7 +531,10 @@ ucc_status_t ShmCollTask::prog_allreduce()
         size_t   len = total_ - off < chunk ? total_ - off : chunk;
         switch (phase_) {
         case 0: /* A */
-            if (j < 2 ? !all_ge(0) : !all_ge(3 * (j - 2) + 2)) {
+            /* sck_: leaders' data areas double as phase-C relays, so
+             * reuse waits for the full previous parity round (+3) */
+            if (j < 2 ? !all_ge(0)
+                      : !all_ge(3 * (j - 2) + (sck_ ? 3 : 2))) {
                 return UCC_INPROGRESS;
             }
             memcpy(seg.data(slot_, p, me_), sbuf_ + off, len);
@@ -549,10 +563,21 @@ ucc_status_t ShmCollTask::prog_allreduce()
             break;
         }
         case 2: /* C */
-            if (!all_ge(3 * j + 2)) {
-                return UCC_INPROGRESS;
+            if (sck_ && !sck_leader_) {
+                /* read my socket leader's local relay copy */
+                if (!rank_ge(sck_ldr_, 3 * j + 3)) {
+                    return UCC_INPROGRESS;
+                }
+                memcpy(dbuf_ + off, seg.data(slot_, p, sck_ldr_), len);
+            } else {
+                if (!all_ge(3 * j + 2)) {
+                    return UCC_INPROGRESS;
+                }
+                memcpy(dbuf_ + off, seg.result(slot_, p), len);
+                if (sck_leader_) { /* publish the relay for my socket */
+                    memcpy(seg.data(slot_, p, me_), dbuf_ + off, len);
+                }
             }
-            memcpy(dbuf_ + off, seg.result(slot_, p), len);
             publish(3 * j + 3);
             phase_ = 0;
             round_++;

@@ -523,6 +523,20 @@ def test_shm_socket_staged_bcast():
         "        job.run(reqs)\n"
         "        for b in bufs:\n"
         "            np.testing.assert_array_equal(b, exp)\n"
+        "# allreduce: phase-C result reads relay through socket leaders\n"
+        "for count in (63, 5000, 300_000):\n"
+        "    arrs = [(rng.random(count) - 0.5).astype(np.float32)\n"
+        "            for _ in range(n)]\n"
+        "    outs = [np.zeros(count, np.float32) for _ in range(n)]\n"
+        "    reqs = job.coll('allreduce', [\n"
+        "        dict(src=arrs[r].ctypes.data, dst=outs[r].ctypes.data,\n"
+        "             count=count, dt=dtypes.FLOAT32)\n"
+        "        for r in range(n)])\n"
+        "    job.run(reqs)\n"
+        "    exp = np.sum(arrs, axis=0)\n"
+        "    for o in outs:\n"
+        "        np.testing.assert_allclose(o, exp, rtol=1e-5,\n"
+        "                                   atol=1e-5)\n"
         "print('SCK_BCAST_OK')\n"
     ) % (REPO,)
     for staging in ("1", "0"):
