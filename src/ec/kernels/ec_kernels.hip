@@ -499,22 +499,13 @@ __global__ void k_reduce(ReduceArgs a)
     for (; i < nv; i += str) {
         P acc = ((const P *)a.srcs[0])[i];
         A  r[VEC];
-#pragma unroll
-        for (int k = 0; k < VEC; k++) {
-            r[k] = Cvt<T>::load(acc.v[k]);
-        }
+        vload<T, OP, VEC, A>(acc, r);
         for (int s = 1; s < n; s++) {
             P x = ((const P *)a.srcs[s])[i];
-#pragma unroll
-            for (int k = 0; k < VEC; k++) {
-                r[k] = red<A, OP>(r[k], Cvt<T>::load(x.v[k]));
-            }
+            vaccum<T, OP, VEC, A>(r, x);
         }
         P out;
-#pragma unroll
-        for (int k = 0; k < VEC; k++) {
-            out.v[k] = Cvt<T>::store(apply_alpha<A>(r[k], a.alpha));
-        }
+        vstore<T, OP, VEC, A>(out, r, a.alpha);
         ((P *)a.dst)[i] = out;
     }
     /* tail (scalar) */
@@ -726,22 +717,13 @@ __global__ void k_fused_allreduce(FusedArgs a)
         for (uint64_t i = tid; i < nv; i += str) {
             P acc = ((const P *)a.peer_scratch[0])[i];
             A r[VEC];
-#pragma unroll
-            for (int k = 0; k < VEC; k++) {
-                r[k] = Cvt<T>::load(acc.v[k]);
-            }
+            vload<T, OP, VEC, A>(acc, r);
             for (int s = 1; s < n; s++) {
                 P x = ((const P *)a.peer_scratch[s])[i];
-#pragma unroll
-                for (int k = 0; k < VEC; k++) {
-                    r[k] = red<A, OP>(r[k], Cvt<T>::load(x.v[k]));
-                }
+                vaccum<T, OP, VEC, A>(r, x);
             }
             P out;
-#pragma unroll
-            for (int k = 0; k < VEC; k++) {
-                out.v[k] = Cvt<T>::store(apply_alpha<A>(r[k], a.alpha));
-            }
+            vstore<T, OP, VEC, A>(out, r, a.alpha);
             ((P *)a.dst)[i] = out;
         }
         for (uint64_t t = nv * VEC + tid; t < a.count; t += str) {
@@ -878,23 +860,14 @@ __global__ void k_fused_allreduce_graph(GraphFusedArgs a)
         for (uint64_t i = tid; i < nv; i += str) {
             P acc = ((const P *)((const uint8_t *)a.peer_scratch[0] + par))[i];
             A r[VEC];
-#pragma unroll
-            for (int k = 0; k < VEC; k++) {
-                r[k] = Cvt<T>::load(acc.v[k]);
-            }
+            vload<T, OP, VEC, A>(acc, r);
             for (int s = 1; s < n; s++) {
                 P x = ((const P *)((const uint8_t *)a.peer_scratch[s] +
                                    par))[i];
-#pragma unroll
-                for (int k = 0; k < VEC; k++) {
-                    r[k] = red<A, OP>(r[k], Cvt<T>::load(x.v[k]));
-                }
+                vaccum<T, OP, VEC, A>(r, x);
             }
             P out;
-#pragma unroll
-            for (int k = 0; k < VEC; k++) {
-                out.v[k] = Cvt<T>::store(apply_alpha<A>(r[k], a.alpha));
-            }
+            vstore<T, OP, VEC, A>(out, r, a.alpha);
             ((P *)a.dst)[i] = out;
         }
         for (uint64_t t = nv * VEC + tid; t < a.count; t += str) {
