@@ -2102,6 +2102,82 @@ class TcpAllgatherNeighborTask final : public TcpTask {
     std::vector<uint32_t> g_;
 };
 
+/* ---- DBT allreduce (reference tl/ucp allreduce dbt role): composed
+ * as DBT reduce to rank 0 followed by DBT bcast of the result — both
+ * message halves stream through the two shifted in-order BSTs, so
+ * per-rank send load stays ~2x the message across the whole
+ * composition (the latency*bandwidth-balanced medium band). */
+class TcpAllreduceDbtTask final : public TcpTask {
+  public:
+    using TcpTask::TcpTask;
+    ~TcpAllreduceDbtTask() override { delete sub_; }
+
+    ucc_status_t post() override
+    {
+        begin(); /* consume one team seq uniformly (composite itself
+                    sends nothing) */
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        dtsz_ = ucc_dt_size(a_.dst.info.datatype);
+        scr_.resize(a_.dst.info.count * dtsz_);
+        ucc_coll_args_t ra{};
+        ra.mask              = UCC_COLL_ARGS_FIELD_FLAGS;
+        ra.flags             = a_.flags & UCC_COLL_ARGS_FLAG_TIMEOUT;
+        ra.coll_type         = UCC_COLL_TYPE_REDUCE;
+        ra.root              = 0;
+        ra.op                = a_.op;
+        ra.src.info          = a_.dst.info;
+        ra.src.info.buffer   = inplace ? a_.dst.info.buffer
+                                       : a_.src.info.buffer;
+        ra.dst.info          = a_.dst.info;
+        ra.dst.info.buffer   = scr_.data();
+        sub_                 = new TcpReduceDbtTask(ctx_, tt_, ra);
+        ucc_status_t st      = sub_->post();
+        if (st == UCC_ERR_NOT_SUPPORTED) {
+            return st; /* tiny vectors: recursive doubling handles */
+        }
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return step(st);
+    }
+    ucc_status_t progress() override { return step(sub_->progress()); }
+
+  private:
+    ucc_status_t step(ucc_status_t st)
+    {
+        while (true) {
+            if (st == UCC_INPROGRESS) {
+                return UCC_INPROGRESS;
+            }
+            if (st != UCC_OK) {
+                return st;
+            }
+            if (phase_ == 0) { /* reduce done: bcast the result */
+                delete sub_;
+                sub_ = nullptr;
+                if (me_ == 0) {
+                    memcpy(a_.dst.info.buffer, scr_.data(),
+                           scr_.size());
+                }
+                ucc_coll_args_t ba{};
+                ba.mask            = UCC_COLL_ARGS_FIELD_FLAGS;
+                ba.flags           = a_.flags & UCC_COLL_ARGS_FLAG_TIMEOUT;
+                ba.coll_type       = UCC_COLL_TYPE_BCAST;
+                ba.root            = 0;
+                ba.src.info        = a_.dst.info;
+                sub_               = new TcpBcastDbtTask(ctx_, tt_, ba);
+                phase_             = 1;
+                st                 = sub_->post();
+                continue;
+            }
+            return UCC_OK; /* bcast complete */
+        }
+    }
+
+    size_t               dtsz_ = 4;
+    std::vector<uint8_t> scr_;
+    TcpTask             *sub_ = nullptr;
+};
+
 /* ---- sparbit-role allgather (reference tl/ucp allgather sparbit,
  * re-derived): ceil(log2 n) rounds, DATA-ORDERED — every block lands
  * directly at its absolute dst position, so there is no work buffer
@@ -3980,6 +4056,27 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
             return UCC_OK;
         };
         map.add(UCC_COLL_TYPE_REDUCE, UCC_MEMORY_TYPE_HOST, dr);
+
+        /* DBT allreduce: reduce + bcast through the two shifted BSTs
+         * (selectable; the recursive-doubling/SRA defaults measured
+         * better on loopback — profiles/tcp_host_crossovers_r02.md) */
+        ScoreRange ar;
+        ar.start    = dbt_min;
+        ar.end      = dbt_max;
+        ar.score    = sc; /* tie with the default: tune to enable */
+        ar.tl_name  = "tcp";
+        ar.alg_name = "dbt";
+        ar.init     = [self](const ucc_coll_args_t &args, Team *t2,
+                         Task **task) -> ucc_status_t {
+            if ((args.mask & UCC_COLL_ARGS_FIELD_ACTIVE_SET) ||
+                !ucc_dt_is_predefined(args.dst.info.datatype) ||
+                args.dst.info.count < 2) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpAllreduceDbtTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_ALLREDUCE, UCC_MEMORY_TYPE_HOST, ar);
 
         /* k-nomial reduce below the DBT band: log_k(n) hops instead
          * of the linear task's n-1 root fan */

@@ -614,6 +614,57 @@ def test_gather_scatter_knomial():
             f"radix {radix}"
 
 
+def test_allreduce_dbt():
+    """DBT allreduce (reference allreduce dbt role): DBT reduce to 0
+    composed with DBT bcast; odd/even n, AVG, odd counts (uneven
+    message halves)."""
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "from ucc_amd import core, dtypes\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "for n in (2, 3, 6, 8):\n"
+        "    job = LocalJob(n)\n"
+        "    smap = core().score_map_str(job.teams[0])\n"
+        "    assert 'allreduce' in smap and '@tcp/dbt' in smap, smap\n"
+        "    rng = np.random.default_rng(n)\n"
+        "    for count in (3001, 40_001):\n"
+        "        arrs = [(rng.random(count) - 0.5).astype(np.float32)\n"
+        "                for _ in range(n)]\n"
+        "        outs = [np.zeros(count, np.float32) for _ in range(n)]\n"
+        "        reqs = job.coll('allreduce', [\n"
+        "            dict(src=arrs[r].ctypes.data,\n"
+        "                 dst=outs[r].ctypes.data, count=count,\n"
+        "                 dt=dtypes.FLOAT32) for r in range(n)])\n"
+        "        job.run(reqs)\n"
+        "        exp = np.sum(arrs, axis=0)\n"
+        "        for o in outs:\n"
+        "            np.testing.assert_allclose(o, exp, rtol=1e-5,\n"
+        "                                       atol=1e-5)\n"
+        "    arrs = [(rng.random(7001) - 0.5).astype(np.float32)\n"
+        "            for _ in range(n)]\n"
+        "    outs = [np.zeros(7001, np.float32) for _ in range(n)]\n"
+        "    reqs = job.coll('allreduce', [\n"
+        "        dict(src=arrs[r].ctypes.data, dst=outs[r].ctypes.data,\n"
+        "             count=7001, dt=dtypes.FLOAT32, op=dtypes.OP_AVG)\n"
+        "        for r in range(n)])\n"
+        "    job.run(reqs)\n"
+        "    exp = np.mean(arrs, axis=0)\n"
+        "    for o in outs:\n"
+        "        np.testing.assert_allclose(o, exp, rtol=1e-5,\n"
+        "                                   atol=1e-6)\n"
+        "print('AR_DBT_OK')\n"
+    ) % (REPO,)
+    env = dict(os.environ)
+    env["UCC_TL_SHM_ENABLE"] = "0"
+    env["UCC_TUNE"] = "allreduce:@dbt:99"
+    p = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=300)
+    sys.stdout.write(p.stdout[-400:])
+    sys.stderr.write(p.stderr[-2000:])
+    assert p.returncode == 0 and "AR_DBT_OK" in p.stdout
+
+
 def test_reduce_knomial_radix():
     """Radix-k knomial reduce (reference tl/ucp reduce knomial role):
     children's subtree sums climb the reversed bcast tree; every root,
