@@ -2003,6 +2003,95 @@ class TcpAllgatherBruckTask final : public TcpTask {
     std::vector<uint8_t> work_;
 };
 
+/* ---- radix-k knomial allgather (reference tl/ucp allgather knomial
+ * role, re-derived as radix-k dissemination at absolute offsets): the
+ * sparbit task generalized — at round r each rank owns the circular
+ * run [me, me+own) with own = k^r and exchanges with the k-1 peers at
+ * distances i*own, receiving their runs which extend the ownership to
+ * k*own contiguously; ceil(log_k n) rounds, data-ordered, any n. */
+class TcpAllgatherKnomialTask final : public TcpTask {
+  public:
+    TcpAllgatherKnomialTask(Context *ctx, TcpTlTeam *tt,
+                            const ucc_coll_args_t &args, uint32_t radix)
+        : TcpTask(ctx, tt, args), k_(radix < 2 ? 2 : radix)
+    {
+    }
+
+    ucc_status_t post() override
+    {
+        begin();
+        const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
+        dtsz_ = ucc_dt_size(a_.dst.info.datatype);
+        blk_  = a_.dst.info.count * dtsz_ / n_;
+        dst_  = (uint8_t *)a_.dst.info.buffer;
+        if (blk_ == 0 || blk_ * n_ != a_.dst.info.count * dtsz_) {
+            return UCC_ERR_NOT_SUPPORTED; /* ragged: ring handles */
+        }
+        if (!inplace) {
+            memcpy(dst_ + me_ * blk_, a_.src.info.buffer, blk_);
+        }
+        own_   = 1;
+        round_ = 0;
+        phase_ = 0;
+        status = UCC_INPROGRESS;
+        return progress_();
+    }
+    ucc_status_t progress() override { return progress_(); }
+
+  private:
+    void post_run(uint32_t peer, uint32_t tag, uint64_t first,
+                  uint64_t cnt, bool is_send)
+    {
+        uint64_t head = std::min(cnt, (uint64_t)n_ - first);
+        if (is_send) {
+            send_to(peer, tag, dst_ + first * blk_, head * blk_);
+            if (cnt > head) {
+                send_to(peer, tag + 1, dst_, (cnt - head) * blk_);
+            }
+        } else {
+            recv_from(peer, tag, dst_ + first * blk_, head * blk_);
+            if (cnt > head) {
+                recv_from(peer, tag + 1, dst_, (cnt - head) * blk_);
+            }
+        }
+    }
+
+    ucc_status_t progress_()
+    {
+        while (own_ < n_) {
+            if (phase_ == 0) {
+                for (uint32_t i = 1; i < k_; i++) {
+                    uint64_t d = (uint64_t)i * own_;
+                    if (d >= n_) {
+                        break;
+                    }
+                    uint64_t cnt = std::min((uint64_t)own_,
+                                            (uint64_t)n_ - d);
+                    uint32_t to   = (uint32_t)((me_ + n_ - d) % n_);
+                    uint32_t from = (uint32_t)((me_ + d) % n_);
+                    uint32_t tag  = (uint32_t)round_ * 16 + i * 2;
+                    post_run(to, tag, me_, cnt, true);
+                    post_run(from, tag, from, cnt, false);
+                }
+                phase_ = 1;
+            }
+            if (!ops_done()) {
+                return UCC_INPROGRESS;
+            }
+            clear_ops();
+            own_ = std::min<uint64_t>((uint64_t)own_ * k_, n_);
+            round_++;
+            phase_ = 0;
+        }
+        return UCC_OK;
+    }
+
+    uint32_t k_ = 2;
+    uint64_t own_ = 1;
+    size_t   dtsz_ = 4, blk_ = 0;
+    uint8_t *dst_ = nullptr;
+};
+
 /* ---- neighbor-exchange allgather (reference tl/ucp allgather
  * neighbor role, Chan et al., re-derived): EVEN n only. Round 0 pairs
  * (2i, 2i+1) swap their own blocks; every later round alternates
@@ -4126,6 +4215,34 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
                 return UCC_ERR_NOT_SUPPORTED;
             }
             *task = new TcpAllgatherBruckTask(t2->ctx, self, args);
+            return UCC_OK;
+        };
+        map.add(UCC_COLL_TYPE_ALLGATHER, UCC_MEMORY_TYPE_HOST, r);
+    }
+    {
+        /* radix-k knomial allgather: ceil(log_k n) rounds with k-1
+         * exchanges each (selectable; sparbit is the k=2 default) */
+        uint32_t agk = (uint32_t)Config::instance().get_int(
+            "TL_TCP", "KN_RADIX", 4);
+        ScoreRange r;
+        r.start    = 0;
+        r.end      = Config::instance().get_size("TL_TCP",
+                                                 "AG_BRUCK_MAX",
+                                                 64 * 1024);
+        r.score    = sc; /* tie with default: tune to enable */
+        r.tl_name  = "tcp";
+        r.alg_name = "knomial";
+        r.init     = [self, agk](const ucc_coll_args_t &args, Team *t2,
+                        Task **task) -> ucc_status_t {
+            if (!ucc_dt_is_predefined(args.dst.info.datatype) ||
+                (args.dst.info.count *
+                 ucc_dt_size(args.dst.info.datatype)) %
+                        t2->size !=
+                    0) {
+                return UCC_ERR_NOT_SUPPORTED;
+            }
+            *task = new TcpAllgatherKnomialTask(t2->ctx, self, args,
+                                                agk);
             return UCC_OK;
         };
         map.add(UCC_COLL_TYPE_ALLGATHER, UCC_MEMORY_TYPE_HOST, r);

@@ -343,6 +343,47 @@ def test_allgather_bruck():
     assert p.returncode == 0 and "AG_BRUCK_OK" in p.stdout
 
 
+def test_allgather_knomial_radix():
+    """Radix-k knomial allgather (reference tl/ucp allgather knomial
+    role): radix-k dissemination at absolute offsets, radices 2-5 over
+    odd/even/prime team sizes."""
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "import numpy as np\n"
+        "from ucc_amd import core, dtypes\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "for n in (3, 8, 9, 13):\n"
+        "    job = LocalJob(n)\n"
+        "    assert '@tcp/knomial' in core().score_map_str(\n"
+        "        job.teams[0])\n"
+        "    rng = np.random.default_rng(n)\n"
+        "    for per in (1, 77, 900):\n"
+        "        blks = [rng.standard_normal(per).astype(np.float32)\n"
+        "                for _ in range(n)]\n"
+        "        dsts = [np.zeros(per * n, np.float32)\n"
+        "                for _ in range(n)]\n"
+        "        reqs = job.coll('allgather', [\n"
+        "            dict(src=blks[r].ctypes.data,\n"
+        "                 dst=dsts[r].ctypes.data, count=per * n,\n"
+        "                 dt=dtypes.FLOAT32) for r in range(n)])\n"
+        "        job.run(reqs)\n"
+        "        exp = np.concatenate(blks)\n"
+        "        for d in dsts:\n"
+        "            np.testing.assert_array_equal(d, exp)\n"
+        "print('AG_KN_OK')\n"
+    ) % (REPO,)
+    for radix in ("2", "3", "4"):
+        env = dict(os.environ)
+        env["UCC_TL_SHM_ENABLE"] = "0"
+        env["UCC_TUNE"] = "allgather:@knomial:99"
+        env["UCC_TL_TCP_KN_RADIX"] = radix
+        p = subprocess.run([sys.executable, "-c", code], env=env,
+                           capture_output=True, text=True, timeout=300)
+        sys.stdout.write(p.stdout[-300:])
+        sys.stderr.write(p.stderr[-2000:])
+        assert p.returncode == 0 and "AG_KN_OK" in p.stdout, radix
+
+
 def test_allgather_sparbit():
     """Sparbit-role allgather (reference tl/ucp allgather sparbit):
     ceil(log2 n) rounds, data-ordered (blocks land at absolute dst
