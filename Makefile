@@ -141,6 +141,9 @@ asan: build/asan_generic_dt build/asan_obj_size build/asan_perftest
 	ASAN_OPTIONS=detect_leaks=1 UCC_TL_SHM_ENABLE=0 UCC_TUNE=allgather:@bruck:99 ./build/asan_perftest -c allgather -j 5 -b 64 -e 16384 -n 2 -w 1
 	ASAN_OPTIONS=detect_leaks=1 UCC_TL_SHM_ENABLE=0 UCC_TUNE=allgather:@neighbor:99 UCC_TL_TCP_AG_NEIGHBOR_MIN=0 ./build/asan_perftest -c allgather -j 6 -b 64 -e 16384 -n 2 -w 1
 	ASAN_OPTIONS=detect_leaks=1 UCC_TL_SHM_ENABLE=0 UCC_TUNE=allgather:@sparbit:99 ./build/asan_perftest -c allgather -j 7 -b 64 -e 16384 -n 2 -w 1
+	ASAN_OPTIONS=detect_leaks=1 UCC_TL_SHM_ENABLE=0 UCC_TUNE=allgather:@knomial:99 UCC_TL_TCP_KN_RADIX=3 ./build/asan_perftest -c allgather -j 7 -b 64 -e 16384 -n 2 -w 1
+	ASAN_OPTIONS=detect_leaks=1 UCC_TL_SHM_ENABLE=0 UCC_TUNE=allreduce:@dbt:99 ./build/asan_perftest -c allreduce -j 6 -b 8192 -e 131072 -n 2 -w 1
+	ASAN_OPTIONS=detect_leaks=1 UCC_TL_SHM_ENABLE=0 UCC_TUNE=reduce_scatterv:@ring:99 UCC_TL_TCP_RS_RING_MIN=0 ./build/asan_perftest -c allgatherv -j 5 -b 1024 -e 32768 -n 2 -w 1
 	ASAN_OPTIONS=detect_leaks=1 UCC_TL_SHM_ENABLE=0 UCC_TUNE=allreduce:@knomial:99 UCC_TL_TCP_KN_RADIX=3 ./build/asan_perftest -c allreduce -j 7 -b 8 -e 16384 -n 2 -w 1
 	ASAN_OPTIONS=detect_leaks=1 UCC_TL_SHM_ENABLE=0 UCC_TUNE=gather:@knomial:99,scatter:@knomial:99 UCC_TL_TCP_KN_RADIX=3 ./build/asan_perftest -c gather -j 6 -b 64 -e 8192 -n 2 -w 1
 	ASAN_OPTIONS=detect_leaks=1 UCC_TL_SHM_ENABLE=0 UCC_TUNE=reduce_scatter:@ring:99 UCC_TL_TCP_RS_RING_MIN=256 ./build/asan_perftest -c reduce_scatter -j 5 -b 1024 -e 65536 -n 2 -w 1

@@ -2205,6 +2205,8 @@ class TcpAllreduceDbtTask final : public TcpTask {
     {
         begin(); /* consume one team seq uniformly (composite itself
                     sends nothing) */
+        delete sub_; /* re-post of a persistent/reused request */
+        sub_ = nullptr;
         const bool inplace = a_.flags & UCC_COLL_ARGS_FLAG_IN_PLACE;
         dtsz_ = ucc_dt_size(a_.dst.info.datatype);
         scr_.resize(a_.dst.info.count * dtsz_);
