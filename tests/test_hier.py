@@ -493,3 +493,91 @@ def test_hier_device_allreduce_fake_nodes():
     sys.stdout.write(p.stdout[-2000:])
     sys.stderr.write(p.stderr[-3000:])
     assert p.returncode == 0 and "HIER_DEV_OK" in p.stdout
+
+
+SCALE_WORKER = r"""
+import sys
+import numpy as np
+sys.path.insert(0, %r)
+from ucc_amd import core, dtypes
+from ucc_amd.testing import LocalJob
+
+n = 16  # 4 pseudo-nodes x 4 ranks: 5 sub-team bootstraps per rank
+job = LocalJob(n)
+c = core()
+smap = c.score_map_str(job.teams[0])
+assert "@hier/rab" in smap and "@hier/split_rail" in smap, smap
+rng = np.random.default_rng(161)
+
+# allreduce through RAB (small) and split_rail (>=64k) bands
+for count in (5000, 40_000):
+    arrs = [(rng.random(count) - 0.5).astype(np.float32)
+            for _ in range(n)]
+    outs = [np.zeros(count, np.float32) for _ in range(n)]
+    reqs = job.coll("allreduce", [
+        dict(src=arrs[r].ctypes.data, dst=outs[r].ctypes.data,
+             count=count, dt=dtypes.FLOAT32) for r in range(n)])
+    job.run(reqs)
+    exp = np.sum(arrs, axis=0)
+    for o in outs:
+        np.testing.assert_allclose(o, exp, rtol=1e-5, atol=1e-4)
+
+# 2step bcast from a non-leader root on the last node
+root = n - 1
+bufs = [np.zeros(9001, np.float64) for _ in range(n)]
+bufs[root][:] = rng.random(9001)
+exp_b = bufs[root].copy()
+reqs = job.coll("bcast", [
+    dict(src=b.ctypes.data, dst=0, count=9001, dt=dtypes.FLOAT64,
+         root=root) for b in bufs])
+job.run(reqs)
+for b in bufs:
+    np.testing.assert_array_equal(b, exp_b)
+
+# 2step reduce to a mid-team non-leader root
+root = 6
+rsrcs = [(rng.random(7777) - 0.5).astype(np.float32) for _ in range(n)]
+rdst = np.zeros(7777, np.float32)
+reqs = job.coll("reduce", [
+    dict(src=rsrcs[r].ctypes.data,
+         dst=rdst.ctypes.data if r == root else 0, count=7777,
+         dt=dtypes.FLOAT32, root=root) for r in range(n)])
+job.run(reqs)
+np.testing.assert_allclose(rdst, np.sum(rsrcs, axis=0), rtol=1e-5,
+                           atol=1e-4)
+
+# node-packed allgatherv with ragged counts
+cnts = [97 * ((r %% 5) + 1) for r in range(n)]
+dsps = np.concatenate([[0], np.cumsum(cnts)[:-1]]).astype(np.uint64)
+total = int(sum(cnts))
+gs = [(rng.random(cnts[r])).astype(np.float32) for r in range(n)]
+gd = [np.zeros(total, np.float32) for _ in range(n)]
+reqs = job.coll("allgatherv", [
+    dict(src=gs[r].ctypes.data, dst=gd[r].ctypes.data, count=cnts[r],
+         dt=dtypes.FLOAT32, dst_counts=cnts,
+         dst_displs=dsps.tolist()) for r in range(n)])
+job.run(reqs)
+exp_g = np.concatenate(gs)
+for d in gd:
+    np.testing.assert_array_equal(d, exp_g)
+
+# barrier
+reqs = job.coll("barrier", [dict(src=0, dst=0, count=0,
+                                 dt=dtypes.FLOAT32) for _ in range(n)])
+job.run(reqs)
+print("HIER_SCALE_OK")
+""" % (REPO,)
+
+
+def test_hier_scale_16_ranks_4_nodes():
+    """Hier bootstrap + collectives at 16 ranks over 4 pseudo-nodes:
+    every rank drives node/leaders/rail sub-team creation through the
+    padded parent-OOB rounds (the largest in-process hier scale; the
+    bootstrap alignment bugs are O(ranks x rounds))."""
+    env = dict(os.environ)
+    env["UCC_FAKE_NODE_SPLIT"] = "4"
+    p = subprocess.run([sys.executable, "-c", SCALE_WORKER], env=env,
+                       capture_output=True, text=True, timeout=600)
+    sys.stdout.write(p.stdout[-2000:])
+    sys.stderr.write(p.stderr[-3000:])
+    assert p.returncode == 0 and "HIER_SCALE_OK" in p.stdout
