@@ -1136,8 +1136,27 @@ class StagedTask final : public Cdna4Task {
         }
         copy_pending_ = len > 0;
         if (len > 0) {
-            HIPWARN(hipMemcpyAsync(tt_->area(me_, slot_, p, 0), src, len,
-                                   hipMemcpyDeviceToDevice, copy_s()));
+            /* stage with a COPY KERNEL by default: kernel writes are
+             * L2-coherent with the peer processes' reader kernels on
+             * this device by construction, while hipMemcpyAsync may
+             * ride an SDMA engine (one unreproduced data-mismatch on
+             * this path is on record — NEXT_STEPS.md). SDMA staging
+             * stays available via UCC_TL_CDNA4_STAGE_SDMA=1. */
+            static const bool sdma = Config::instance().get_bool(
+                "TL_CDNA4", "STAGE_SDMA", false);
+            if (sdma) {
+                HIPWARN(hipMemcpyAsync(tt_->area(me_, slot_, p, 0), src,
+                                       len, hipMemcpyDeviceToDevice,
+                                       copy_s()));
+            } else {
+                ec_hip::GatherArgs ga{};
+                ga.dst_base = tt_->area(me_, slot_, p, 0);
+                ga.srcs[0]  = src;
+                ga.offs[0]  = 0;
+                ga.lens[0]  = len;
+                ga.n        = 1;
+                ec_hip::gather_copy(ga, copy_s());
+            }
         }
         HIPWARN(hipEventRecord(ev(0 + (int)p), copy_s()));
     }
@@ -2557,6 +2576,9 @@ class Cdna4Tl final : public Tl {
         cfg.declare("TL_CDNA4", "GATED_BLOCKS", "0",
                     "gated-pipeline kernel grid size (workgroups of 256; "
                     "0 = built-in default; must match on every rank)");
+        cfg.declare("TL_CDNA4", "STAGE_SDMA", "0",
+                    "stage fragments with hipMemcpyAsync (SDMA) "
+                    "instead of a copy kernel");
         cfg.declare("TL_CDNA4", "CE_ALLTOALL", "1",
                     "move zero-copy alltoall data on SDMA copy engines "
                     "(hipMemcpyAsync) instead of gather kernels");
