@@ -2069,7 +2069,8 @@ class TcpAllgatherKnomialTask final : public TcpTask {
                                             (uint64_t)n_ - d);
                     uint32_t to   = (uint32_t)((me_ + n_ - d) % n_);
                     uint32_t from = (uint32_t)((me_ + d) % n_);
-                    uint32_t tag  = (uint32_t)round_ * 16 + i * 2;
+                    uint32_t tag  =
+                        (uint32_t)round_ * 2 * k_ + i * 2;
                     post_run(to, tag, me_, cnt, true);
                     post_run(from, tag, from, cnt, false);
                 }
