@@ -176,19 +176,40 @@ ucc_status_t ScoreMap::apply_str(const std::string &str)
                 if (mt_filter >= 0 && mi != mt_filter) {
                     continue;
                 }
+                /* ranged updates SPLIT partially-overlapping entries:
+                 * the score changes only inside [r_start, r_end); the
+                 * outside pieces keep their old score (reference
+                 * ucc_coll_score_update_from_str semantics) */
+                std::vector<ScoreRange> keep;
                 for (auto &r : ranges[ci][mi]) {
                     if (!alg_filter.empty() && r.alg_name != alg_filter) {
                         continue;
                     }
-                    if (have_range) {
-                        if (r_start > r.start) {
-                            r.start = r_start;
-                        }
-                        if (r_end < r.end) {
-                            r.end = r_end;
-                        }
+                    if (!have_range) {
+                        r.score = score;
+                        continue;
                     }
+                    size_t lo = r_start > r.start ? r_start : r.start;
+                    size_t hi = r_end < r.end ? r_end : r.end;
+                    if (lo >= hi) {
+                        continue; /* disjoint: untouched */
+                    }
+                    if (lo > r.start) {
+                        ScoreRange pre = r;
+                        pre.end        = lo;
+                        keep.push_back(pre);
+                    }
+                    if (hi < r.end) {
+                        ScoreRange post = r;
+                        post.start      = hi;
+                        keep.push_back(post);
+                    }
+                    r.start = lo;
+                    r.end   = hi;
                     r.score = score;
+                }
+                for (auto &k : keep) {
+                    ranges[ci][mi].push_back(k);
                 }
                 std::stable_sort(ranges[ci][mi].begin(), ranges[ci][mi].end(),
                                  [](const ScoreRange &a, const ScoreRange &b) {

@@ -530,3 +530,45 @@ def test_team_churn():
         for o in outs:
             np.testing.assert_allclose(o, exp)
         del reqs, teams, oob  # destroy before the next cycle
+
+
+RANGE_TUNE_WORKER = r"""
+import sys
+import numpy as np
+sys.path.insert(0, %r)
+from ucc_amd import core
+from ucc_amd.testing import LocalJob
+
+job = LocalJob(2)
+smap = core().score_map_str(job.teams[0])
+lines = [l for l in smap.splitlines()
+         if l.startswith("allreduce:host") and "@shm" in l]
+# ranged update splits the entry: score 0 only below 4096, the
+# 4096-inf piece keeps the original score
+assert any(l.startswith("allreduce:host:0-4096:@shm") and
+           l.endswith(":0") for l in lines), lines
+assert any(l.startswith("allreduce:host:4096-inf:@shm") and
+           not l.endswith(":0") for l in lines), lines
+for count in (100, 100_000):
+    arrs = [np.ones(count, np.float32),
+            np.full(count, 2.0, np.float32)]
+    outs = job.allreduce_np(arrs)
+    for o in outs:
+        np.testing.assert_allclose(o, np.full(count, 3.0, np.float32))
+print("RANGE_TUNE_OK")
+""" % (REPO,)
+
+
+def test_tuning_string_range_split():
+    """Ranged tuning entries (coll:lo-hi:mem:@alg:score) change the
+    score only INSIDE the range — partially-overlapping score-map
+    entries split and the outside pieces keep their score (reference
+    ucc_coll_score_update_from_str semantics)."""
+    env = dict(os.environ)
+    env["UCC_TUNE"] = "allreduce:0-4096:host:@slotted:0"
+    p = subprocess.run([sys.executable, "-c", RANGE_TUNE_WORKER],
+                       env=env, capture_output=True, text=True,
+                       timeout=120)
+    sys.stdout.write(p.stdout[-2000:])
+    sys.stderr.write(p.stderr[-2000:])
+    assert p.returncode == 0 and "RANGE_TUNE_OK" in p.stdout
