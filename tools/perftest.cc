@@ -59,6 +59,7 @@ struct Opts {
     int         nprocs = 0; /* 0 = in-process */
     int         nranks = 2;
     bool        skew = false; /* alltoallv: MoE-like skewed matrix */
+    bool        root_shift = false; /* rooted colls: rotate root */
     bool        persistent = false, inplace = false, check = false;
     bool        triggered = false;
 };
@@ -306,7 +307,8 @@ static ucc_coll_args_t make_args(const Opts &o, Rank &r, size_t bytes,
                                  std::vector<uint64_t> &cnts,
                                  std::vector<uint64_t> &dsps,
                                  std::vector<uint64_t> &rcnts,
-                                 std::vector<uint64_t> &rdsps)
+                                 std::vector<uint64_t> &rdsps,
+                                 int root = 0)
 {
     ucc_coll_type_t    ct = coll_from_name(o.coll);
     ucc_datatype_t     dt = dt_from_name(o.dtype);
@@ -342,7 +344,7 @@ static ucc_coll_args_t make_args(const Opts &o, Rank &r, size_t bytes,
                   UCC_COLL_ARGS_FLAG_DISPLACEMENTS_64BIT;
     a.coll_type = ct;
     a.op        = op;
-    a.root      = 0;
+    a.root      = (uint64_t)root;
     a.src.info.buffer   = r.bufs.src;
     a.src.info.count    = count;
     a.src.info.datatype = dt;
@@ -805,7 +807,29 @@ static int run_inproc(const Opts &o)
             continue;
         }
         bool coll_fail = false;
+        int  rs_iter   = 0;
+        const bool rooted = ct == UCC_COLL_TYPE_BCAST ||
+                            ct == UCC_COLL_TYPE_GATHER ||
+                            ct == UCC_COLL_TYPE_SCATTER ||
+                            ct == UCC_COLL_TYPE_REDUCE;
         auto iter = [&]() {
+            if (o.root_shift && rooted) {
+                /* -R: rotate the root every iteration (reference
+                 * root-shift mode) — re-init so placement effects
+                 * average out; the timing includes the re-init */
+                int root = rs_iter++ % n;
+                for (int i = 0; i < n; i++) {
+                    ucc_collective_finalize(reqs[i]);
+                    ucc_coll_args_t a2 = make_args(
+                        o, ranks[i], bytes, i, n, cnts[i], dsps[i],
+                        rcnts[i], rdsps[i], root);
+                    if (ucc_collective_init(&a2, &reqs[i],
+                                            ranks[i].team) != UCC_OK) {
+                        coll_fail = true;
+                        return;
+                    }
+                }
+            }
             for (int i = 0; i < n; i++) {
                 if (ucc_collective_post(reqs[i]) != UCC_OK) {
                     coll_fail = true;
@@ -877,7 +901,7 @@ int main(int argc, char **argv)
     setvbuf(stdout, nullptr, _IOLBF, 0); /* line-buffer under pipes */
     Opts o;
     int  c;
-    while ((c = getopt(argc, argv, "c:b:e:n:w:m:d:o:p:j:FiCTMh")) != -1) {
+    while ((c = getopt(argc, argv, "c:b:e:n:w:m:d:o:p:j:FiCTMRh")) != -1) {
         switch (c) {
         case 'c': o.coll = optarg; break;
         case 'b': o.min_b = strtoull(optarg, nullptr, 0); break;
@@ -894,12 +918,14 @@ int main(int argc, char **argv)
         case 'C': o.check = true; break;
         case 'T': o.triggered = true; o.persistent = true; break;
         case 'M': o.skew = true; break;
+        case 'R': o.root_shift = true; break;
         case 'h':
         default:
             printf("ucc_perftest [-c coll] [-b min] [-e max] [-n iters] "
                    "[-w warmup] [-m host|cuda] [-d dtype] [-o op] "
                    "[-p nprocs(fork)] [-j inproc_ranks] [-F persistent] "
-                   "[-i inplace] [-T triggered] [-M skewed-alltoallv]\n");
+                   "[-i inplace] [-T triggered] [-M skewed-alltoallv] "
+                   "[-R root-shift]\n");
             return c == 'h' ? 0 : 1;
         }
     }
