@@ -372,6 +372,20 @@ def _worker_pp_ring(rank, world, port, q):
         w = torch.full((16,), float(rank))
         pp.broadcast_stage_weights([w], root=0)
         assert torch.all(w == 0.0)
+        # Communicator.alltoall / reduce wrappers
+        a2a_src = torch.arange(4 * world, dtype=torch.float32) + \
+            rank * 100
+        a2a_dst = torch.zeros(4 * world)
+        comm.alltoall(a2a_src, a2a_dst)
+        for s in range(world):
+            exp = torch.arange(4, dtype=torch.float32) + rank * 4 + \
+                s * 100
+            assert torch.equal(a2a_dst[s * 4:(s + 1) * 4], exp)
+        red = torch.full((32,), float(rank + 1))
+        comm.reduce(red, root=1)
+        if rank == 1:
+            exp = float(sum(q + 1 for q in range(world)))
+            assert torch.all(red == exp), red[0]
         # --- ring attention: full KV sweep visits every shard once
         ring = RingKV(comm)
         kv = torch.full((128,), float(rank))

@@ -108,6 +108,29 @@ class Communicator:
         self._wait(req)
         return tensor
 
+    def alltoall(self, src, dst):
+        dt = dtypes.from_torch(src.dtype)
+        req = self.c.coll_init(self.team, "alltoall",
+                               src=src.data_ptr(), dst=dst.data_ptr(),
+                               count=dst.numel(), dt=dt,
+                               mem_type=self._mem(dst))
+        self._wait(req)
+        return dst
+
+    def reduce(self, tensor, root=0, op="sum", out=None):
+        dt = dtypes.from_torch(tensor.dtype)
+        is_root = self.rank == root
+        req = self.c.coll_init(
+            self.team, "reduce", src=tensor.data_ptr(),
+            dst=(out.data_ptr() if (is_root and out is not None)
+                 else tensor.data_ptr() if is_root else 0),
+            count=tensor.numel(), dt=dt, op=_OPMAP[op], root=root,
+            mem_type=self._mem(tensor),
+            flags=(self.c.FLAG_IN_PLACE
+                   if (is_root and out is None) else 0))
+        self._wait(req)
+        return out if (is_root and out is not None) else tensor
+
     def barrier(self):
         req = self.c.coll_init(self.team, "barrier", src=0, dst=0, count=0,
                                dt=dtypes.INT8)
