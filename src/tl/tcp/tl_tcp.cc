@@ -3168,6 +3168,16 @@ class TcpReduceScatterRingTask final : public TcpTask {
         }
         memcpy(work_.data(), src, total_);
         tmp_.resize(maxb);
+        /* bidirectional mode (reference tl_ucp reduce_scatter ring
+         * bidirectional knob): each block splits in two halves that
+         * travel opposite ring directions, using both duplex
+         * directions of every link each round */
+        bidir_ = Config::instance().get_bool("TL_TCP", "RS_RING_BIDIR",
+                                             false) &&
+                 n_ > 2;
+        if (bidir_) {
+            tmpb_.resize(maxb);
+        }
         round_ = 0;
         phase_ = 0;
         status = UCC_INPROGRESS;
@@ -3177,29 +3187,61 @@ class TcpReduceScatterRingTask final : public TcpTask {
 
   private:
     uint8_t *blk(uint32_t b) { return work_.data() + dsp_[b]; }
+    size_t   ha(uint32_t b) const /* fwd half, element-aligned */
+    {
+        return cnt_[b] / 2 / dtsz_ * dtsz_;
+    }
 
     ucc_status_t progress_()
     {
         const uint32_t right = (me_ + 1) % n_;
         const uint32_t left  = (me_ + n_ - 1) % n_;
         while (round_ < (int)n_ - 1) {
+            uint32_t sf = (me_ + 2 * n_ - 1 - (uint32_t)round_) % n_;
+            uint32_t rf = (me_ + 2 * n_ - 2 - (uint32_t)round_) % n_;
+            uint32_t sb = (me_ + 1 + (uint32_t)round_) % n_;
+            uint32_t rb = (me_ + 2 + (uint32_t)round_) % n_;
             if (phase_ == 0) {
-                uint32_t sb = (me_ + 2 * n_ - 1 - (uint32_t)round_) % n_;
-                uint32_t rb = (me_ + 2 * n_ - 2 - (uint32_t)round_) % n_;
-                send_to(right, (uint32_t)round_, blk(sb), cnt_[sb]);
-                recv_from(left, (uint32_t)round_, tmp_.data(),
-                          cnt_[rb]);
+                if (!bidir_) {
+                    send_to(right, (uint32_t)round_, blk(sf), cnt_[sf]);
+                    recv_from(left, (uint32_t)round_, tmp_.data(),
+                              cnt_[rf]);
+                } else {
+                    /* forward ring carries each block's first half */
+                    send_to(right, (uint32_t)round_, blk(sf), ha(sf));
+                    recv_from(left, (uint32_t)round_, tmp_.data(),
+                              ha(rf));
+                    /* backward ring carries the second half */
+                    send_to(left, 0x100u + (uint32_t)round_,
+                            blk(sb) + ha(sb), cnt_[sb] - ha(sb));
+                    recv_from(right, 0x100u + (uint32_t)round_,
+                              tmpb_.data(), cnt_[rb] - ha(rb));
+                }
                 phase_ = 1;
             }
             if (!ops_done()) {
                 return UCC_INPROGRESS;
             }
             clear_ops();
-            uint32_t rb = (me_ + 2 * n_ - 2 - (uint32_t)round_) % n_;
-            if (cnt_[rb]) {
-                const void *srcs[2] = {blk(rb), tmp_.data()};
-                ec_cpu::reduce(blk(rb), srcs, 2, cnt_[rb] / dtsz_, dt_,
-                               op_);
+            if (!bidir_) {
+                if (cnt_[rf]) {
+                    const void *srcs[2] = {blk(rf), tmp_.data()};
+                    ec_cpu::reduce(blk(rf), srcs, 2, cnt_[rf] / dtsz_,
+                                   dt_, op_);
+                }
+            } else {
+                if (ha(rf)) {
+                    const void *srcs[2] = {blk(rf), tmp_.data()};
+                    ec_cpu::reduce(blk(rf), srcs, 2, ha(rf) / dtsz_,
+                                   dt_, op_);
+                }
+                if (cnt_[rb] - ha(rb)) {
+                    const void *srcs[2] = {blk(rb) + ha(rb),
+                                           tmpb_.data()};
+                    ec_cpu::reduce(blk(rb) + ha(rb), srcs, 2,
+                                   (cnt_[rb] - ha(rb)) / dtsz_, dt_,
+                                   op_);
+                }
             }
             phase_ = 0;
             round_++;
@@ -3223,8 +3265,9 @@ class TcpReduceScatterRingTask final : public TcpTask {
     ucc_reduction_op_t op_ = UCC_OP_SUM;
     size_t   dtsz_ = 4;
     uint64_t total_ = 0;
+    bool     bidir_ = false;
     std::vector<size_t>  cnt_, dsp_;
-    std::vector<uint8_t> work_, tmp_;
+    std::vector<uint8_t> work_, tmp_, tmpb_;
 };
 
 /* ---- recursive-halving reduce_scatter (reference tl/ucp
@@ -3918,6 +3961,9 @@ class TcpTl final : public Tl {
         cfg.declare("TL_TCP", "AG_NEIGHBOR_MIN", "256k",
                     "neighbor-exchange allgather lower bound bytes "
                     "(even team sizes only)");
+        cfg.declare("TL_TCP", "RS_RING_BIDIR", "0",
+                    "reduce_scatter ring: split halves over both ring "
+                    "directions (full-duplex links)");
         cfg.declare("TL_TCP", "SOCKBUF", "4m",
                     "SO_SNDBUF/SO_RCVBUF bytes (0 = kernel default)");
         cfg.declare("TL_TCP", "AG_LINEAR_MIN", "512k",

@@ -953,6 +953,7 @@ def test_reduce_scatter_ring():
     env = dict(os.environ)
     env["UCC_TL_SHM_ENABLE"] = "0"
     env["UCC_TL_TCP_RS_RING_MIN"] = "1024"
+    env["UCC_TL_TCP_RS_RING_BIDIR"] = "1"  # both ring directions
     env["UCC_TUNE"] = "reduce_scatter:@ring:99"
     p = subprocess.run([sys.executable, "-c", code], env=env,
                        capture_output=True, text=True, timeout=600)
