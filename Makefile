@@ -88,7 +88,10 @@ $(SHLIB): $(LIB_OBJS)
 clean:
 	rm -rf $(BUILD) ucc_amd/_core*.so
 
-.PHONY: all clean
+check: all
+	python3 -m pytest tests/ -x -q -m "not gpu"
+
+.PHONY: all clean check
 
 # auto-generated header dependencies (-MMD)
 -include $(LIB_OBJS:.o=.d) $(BIND_OBJ:.o=.d)

@@ -572,3 +572,39 @@ def test_tuning_string_range_split():
     sys.stdout.write(p.stdout[-2000:])
     sys.stderr.write(p.stderr[-2000:])
     assert p.returncode == 0 and "RANGE_TUNE_OK" in p.stdout
+
+
+def test_tuning_string_disjoint_and_multi():
+    """Disjoint ranged entries leave the map untouched; comma-separated
+    entries and ':inf' scores compose (reference DSL grammar)."""
+    code = (
+        "import sys; sys.path.insert(0, %r)\n"
+        "from ucc_amd import core\n"
+        "from ucc_amd.testing import LocalJob\n"
+        "job = LocalJob(2)\n"
+        "smap = core().score_map_str(job.teams[0])\n"
+        "lines = [l for l in smap.splitlines()\n"
+        "         if l.startswith('allreduce:host')]\n"
+        "# disjoint range (beyond any entry start) left shm at 40\n"
+        "assert any('@shm/slotted:40' in l for l in lines), lines\n"
+        "# second entry boosted bcast dbt to the inf ceiling\n"
+        "blines = [l for l in smap.splitlines()\n"
+        "          if l.startswith('bcast:host') and '@tcp/dbt' in l]\n"
+        "assert any(l.rstrip().endswith(':2147483647')\n"
+        "           for l in blines), blines\n"
+        "print('DSL_OK')\n"
+    ) % (REPO,)
+    env = dict(os.environ)
+    # first entry: range [1TB, inf) matches nothing that overlaps the
+    # shm entry below it? shm is 0-inf so it DOES overlap -> to make a
+    # truly disjoint case, clamp an entry that ends before 1TB: tcp
+    # bruck allgather ends at 64k.
+    env["UCC_TUNE"] = ("allgather:1099511627776-inf:@bruck:1,"
+                       "bcast:@dbt:inf")
+    code2 = code.replace("allreduce:host", "allgather:host").replace(
+        "@shm/slotted:40", "@tcp/bruck:")
+    p = subprocess.run([sys.executable, "-c", code2], env=env,
+                       capture_output=True, text=True, timeout=120)
+    sys.stdout.write(p.stdout[-1500:])
+    sys.stderr.write(p.stderr[-1500:])
+    assert p.returncode == 0 and "DSL_OK" in p.stdout
