@@ -563,6 +563,8 @@ PYBIND11_MODULE(_core, m)
     m.attr("INPROGRESS") = (int)UCC_INPROGRESS;
     m.attr("FLAG_IN_PLACE")   = (uint64_t)UCC_COLL_ARGS_FLAG_IN_PLACE;
     m.attr("FLAG_PERSISTENT") = (uint64_t)UCC_COLL_ARGS_FLAG_PERSISTENT;
+    m.attr("FLAG_MEM_MAPPED") =
+        (uint64_t)UCC_COLL_ARGS_FLAG_MEM_MAPPED_BUFFERS;
     m.def("dt_size", [](int dt) { return ucc_dt_size((ucc_datatype_t)dt); });
     m.def("version", []() { return std::string(ucc_get_version_string()); });
     m.def("hip_device_count",

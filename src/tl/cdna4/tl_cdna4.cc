@@ -1425,6 +1425,12 @@ class GatedCollTask final : public Cdna4Task {
             Config::instance().get_bool("TL_CDNA4", "ZCOPY", true)) {
             if (a_.flags & UCC_COLL_ARGS_FLAG_PERSISTENT) {
                 zc_ = true; /* exchange amortized over re-posts */
+            } else if (a_.flags & UCC_COLL_ARGS_FLAG_MEM_MAPPED_BUFFERS) {
+                /* caller promises src/dst sit inside ucc_mem_map'd
+                 * regions: imports come from the memh/IPC caches, so
+                 * zero-copy pays at any size (reference MEM_MAPPED
+                 * flag role, ucc.h onesided semantics) */
+                zc_ = true;
             } else {
                 /* one-shot: the 4 host-gated exchange rounds (team
                  * IPC-import cache makes the opens free after first

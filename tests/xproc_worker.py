@@ -152,7 +152,8 @@ def main():
     r2d = c.coll_init(team, "allreduce", src=src3.data_ptr(),
                       dst=dst3.data_ptr(), count=count3,
                       dt=dtypes.FLOAT32, mem_type=dtypes.MEM_CUDA,
-                      flags=c.FLAG_PERSISTENT, src_memh=sh, dst_memh=dh)
+                      flags=c.FLAG_PERSISTENT | c.FLAG_MEM_MAPPED,
+                      src_memh=sh, dst_memh=dh)
     for it in range(2):
         src3.copy_(full3[rank] * (it + 1))
         torch.cuda.synchronize()
