@@ -55,24 +55,30 @@ class ShmCollTask final : public Task {
             return;
         }
         const auto &procs = team->procs;
-        int16_t     rsck  = procs[root].socket_id;
-        bool        known = rsck >= 0;
-        bool        multi = false;
-        sck_ldr_          = me_;
+        /* group by CPU socket; if socket ids are unavailable on this
+         * platform, NUMA ids carry the same locality signal (both are
+         * reference ucc_sbgp kinds) */
+        auto gid = [&](uint32_t r) -> int16_t {
+            return procs[r].socket_id >= 0 ? procs[r].socket_id
+                                           : procs[r].numa_id;
+        };
+        int16_t rsck  = gid(root);
+        bool    known = rsck >= 0;
+        bool    multi = false;
+        sck_ldr_      = me_;
         for (uint32_t r = 0; r < n_ && known; r++) {
-            if (procs[r].socket_id < 0) {
+            if (gid(r) < 0) {
                 known = false;
             }
-            if (procs[r].socket_id != rsck) {
+            if (gid(r) != rsck) {
                 multi = true;
             }
-            if (procs[r].socket_id == procs[me_].socket_id &&
-                r < sck_ldr_) {
+            if (gid(r) == gid(me_) && r < sck_ldr_) {
                 sck_ldr_ = r;
             }
         }
         sck_      = known && multi;
-        root_sck_ = procs[me_].socket_id == rsck;
+        root_sck_ = gid(me_) == rsck;
     }
 
     ucc_status_t prog_allreduce();
