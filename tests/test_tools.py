@@ -40,7 +40,9 @@ def test_info_version_configs_scoremap():
 
 
 @pytest.mark.parametrize("coll", ["allreduce", "allgather", "bcast",
-                                  "reduce_scatter"])
+                                  "reduce_scatter", "reduce_scatterv",
+                                  "gatherv", "scatterv",
+                                  "allgatherv"])
 def test_perftest_validated(coll):
     """-C runs a golden-pattern validation iteration per size."""
     p = _run([PERF, "-c", coll, "-j", "3", "-b", "64", "-e", "16384",

@@ -105,7 +105,10 @@ static const struct {
     {"gather", UCC_COLL_TYPE_GATHER},
     {"reduce", UCC_COLL_TYPE_REDUCE},
     {"reduce_scatter", UCC_COLL_TYPE_REDUCE_SCATTER},
+    {"reduce_scatterv", UCC_COLL_TYPE_REDUCE_SCATTERV},
     {"scatter", UCC_COLL_TYPE_SCATTER},
+    {"gatherv", UCC_COLL_TYPE_GATHERV},
+    {"scatterv", UCC_COLL_TYPE_SCATTERV},
 };
 
 static ucc_coll_type_t coll_from_name(const std::string &s)
@@ -131,6 +134,7 @@ static double busbw_factor(ucc_coll_type_t ct, int n)
     case UCC_COLL_TYPE_ALLTOALL:
     case UCC_COLL_TYPE_ALLTOALLV:
     case UCC_COLL_TYPE_REDUCE_SCATTER:
+    case UCC_COLL_TYPE_REDUCE_SCATTERV:
         return (double)(n - 1) / n;
     default:
         return 1.0;
@@ -365,6 +369,48 @@ static ucc_coll_args_t make_args(const Opts &o, Rank &r, size_t bytes,
     case UCC_COLL_TYPE_REDUCE_SCATTER:
         a.dst.info.count = count / nranks;
         break;
+    case UCC_COLL_TYPE_REDUCE_SCATTERV: {
+        cnts.assign(nranks, count / nranks);
+        dsps.resize(nranks);
+        for (int i = 0; i < nranks; i++) {
+            dsps[i] = (uint64_t)i * (count / nranks);
+        }
+        a.dst.info_v.buffer        = r.bufs.dst;
+        a.dst.info_v.counts        = cnts.data();
+        a.dst.info_v.displacements = dsps.data();
+        a.dst.info_v.datatype      = dt;
+        a.dst.info_v.mem_type      = mt;
+        a.src.info.count           = count;
+        break;
+    }
+    case UCC_COLL_TYPE_GATHERV: {
+        cnts.assign(nranks, count / nranks);
+        dsps.resize(nranks);
+        for (int i = 0; i < nranks; i++) {
+            dsps[i] = (uint64_t)i * (count / nranks);
+        }
+        a.src.info.count           = count / nranks;
+        a.dst.info_v.buffer        = r.bufs.dst;
+        a.dst.info_v.counts        = cnts.data();
+        a.dst.info_v.displacements = dsps.data();
+        a.dst.info_v.datatype      = dt;
+        a.dst.info_v.mem_type      = mt;
+        break;
+    }
+    case UCC_COLL_TYPE_SCATTERV: {
+        cnts.assign(nranks, count / nranks);
+        dsps.resize(nranks);
+        for (int i = 0; i < nranks; i++) {
+            dsps[i] = (uint64_t)i * (count / nranks);
+        }
+        a.src.info_v.buffer        = r.bufs.src;
+        a.src.info_v.counts        = cnts.data();
+        a.src.info_v.displacements = dsps.data();
+        a.src.info_v.datatype      = dt;
+        a.src.info_v.mem_type      = mt;
+        a.dst.info.count           = count / nranks;
+        break;
+    }
     case UCC_COLL_TYPE_ALLGATHERV: {
         cnts.assign(nranks, count / nranks);
         dsps.resize(nranks);
