@@ -483,6 +483,68 @@ static ucc_coll_args_t make_args(const Opts &o, Rank &r, size_t bytes,
     return a;
 }
 
+/* golden fp32 patterns shared by the fork- and inproc-mode -C checks */
+static void vpattern(ucc_coll_type_t ct, int rank, size_t count,
+                     std::vector<float> &h)
+{
+    h.resize(count);
+    switch (ct) {
+    case UCC_COLL_TYPE_ALLREDUCE:
+        for (size_t i = 0; i < count; i++) {
+            h[i] = (float)(rank + 1) + 0.25f * (float)(i % 7);
+        }
+        break;
+    case UCC_COLL_TYPE_BCAST:
+        for (size_t i = 0; i < count; i++) {
+            h[i] = rank == 0 ? (float)(i % 97) : -1.0f;
+        }
+        break;
+    default: /* allgather / reduce_scatter family */
+        for (size_t i = 0; i < count; i++) {
+            h[i] = (float)(rank + 1) + 0.5f * (float)(i % 5);
+        }
+        break;
+    }
+}
+
+static size_t vcheck(ucc_coll_type_t ct, int rank, int nranks,
+                     size_t count, const float *out)
+{
+    size_t bad = 0;
+    if (ct == UCC_COLL_TYPE_ALLREDUCE) {
+        for (size_t i = 0; i < count; i++) {
+            float s = 0;
+            for (int k = 0; k < nranks; k++) {
+                s += (float)(k + 1) + 0.25f * (float)(i % 7);
+            }
+            bad += out[i] != s;
+        }
+    } else if (ct == UCC_COLL_TYPE_ALLGATHER) {
+        size_t per = count / nranks;
+        for (size_t i = 0; i < per * (size_t)nranks; i++) {
+            int    src = (int)(i / per);
+            size_t li  = i % per;
+            bad += out[i] !=
+                   (float)(src + 1) + 0.5f * (float)(li % 5);
+        }
+    } else if (ct == UCC_COLL_TYPE_REDUCE_SCATTER) {
+        size_t per = count / nranks;
+        for (size_t i = 0; i < per; i++) {
+            size_t gi = (size_t)rank * per + i;
+            float  e  = 0;
+            for (int k = 0; k < nranks; k++) {
+                e += (float)(k + 1) + 0.5f * (float)(gi % 5);
+            }
+            bad += out[i] != e;
+        }
+    } else if (ct == UCC_COLL_TYPE_BCAST) {
+        for (size_t i = 0; i < count; i++) {
+            bad += out[i] != (float)(i % 97);
+        }
+    }
+    return bad;
+}
+
 /* -C validation: run one iteration with a known fp32 pattern and check
  * the result (allreduce/allgather/bcast/reduce_scatter). Returns false
  * on mismatch. Works for host and device memory (staged via host). */
@@ -895,6 +957,47 @@ static int run_inproc(const Opts &o)
                 }
             }
         };
+        const bool checkable =
+            o.check && o.dtype == "float32" && !o.inplace &&
+            o.mem == "host" && bytes >= 4 &&
+            (ct == UCC_COLL_TYPE_ALLREDUCE ||
+             ct == UCC_COLL_TYPE_ALLGATHER ||
+             ct == UCC_COLL_TYPE_REDUCE_SCATTER ||
+             ct == UCC_COLL_TYPE_BCAST);
+        if (checkable) {
+            std::vector<float> h;
+            for (int i = 0; i < n; i++) {
+                vpattern(ct, i, bytes / 4, h);
+                memcpy(ranks[i].bufs.src, h.data(), bytes);
+                memset(ranks[i].bufs.dst, 0, bytes);
+            }
+            iter();
+            size_t bad = 0;
+            for (int i = 0; i < n && !coll_fail; i++) {
+                const float *out = (const float *)(
+                    ct == UCC_COLL_TYPE_BCAST ? ranks[i].bufs.src
+                                              : ranks[i].bufs.dst);
+                /* note: make_args rounded the count for block colls */
+                size_t cnt2 =
+                    ct == UCC_COLL_TYPE_REDUCE_SCATTER
+                        ? ((bytes / 4) - (bytes / 4) % n) / n
+                        : ct == UCC_COLL_TYPE_ALLGATHER
+                              ? (bytes / 4) - (bytes / 4) % n
+                              : bytes / 4;
+                bad += vcheck(ct, i, n, ct == UCC_COLL_TYPE_REDUCE_SCATTER
+                                            ? cnt2 * n
+                                            : cnt2,
+                              out);
+            }
+            if (bad) {
+                printf("%12zu        - VALIDATION FAILED (%zu bad)\n",
+                       bytes, bad);
+                for (int i = 0; i < n; i++) {
+                    ucc_collective_finalize(reqs[i]);
+                }
+                exit(2);
+            }
+        }
         for (int i = 0; i < o.warmup && !coll_fail; i++) {
             iter();
         }
