@@ -559,27 +559,12 @@ static bool validate_once(const Opts &o, Rank &r, ucc_coll_req_h req,
     if (count == 0) {
         return true;
     }
-    std::vector<float> h(count), out(count, 0.0f);
-    switch (ct) {
-    case UCC_COLL_TYPE_ALLREDUCE:
-        for (size_t i = 0; i < count; i++) {
-            h[i] = (float)(rank + 1) + 0.25f * (float)(i % 7);
-        }
-        break;
-    case UCC_COLL_TYPE_ALLGATHER:
-    case UCC_COLL_TYPE_REDUCE_SCATTER:
-        for (size_t i = 0; i < count; i++) {
-            h[i] = (float)(rank + 1) + 0.5f * (float)(i % 5);
-        }
-        break;
-    case UCC_COLL_TYPE_BCAST:
-        for (size_t i = 0; i < count; i++) {
-            h[i] = rank == 0 ? (float)(i % 97) : -1.0f;
-        }
-        break;
-    default:
+    if (ct != UCC_COLL_TYPE_ALLREDUCE && ct != UCC_COLL_TYPE_ALLGATHER &&
+        ct != UCC_COLL_TYPE_REDUCE_SCATTER && ct != UCC_COLL_TYPE_BCAST) {
         return true;
     }
+    std::vector<float> h, out(count, 0.0f);
+    vpattern(ct, rank, count, h);
 #ifdef UCC_AMD_HAS_HIP
     if (o.mem == "cuda") {
         HIPWARN(hipMemcpy(r.bufs.src, h.data(), bytes, hipMemcpyHostToDevice));
@@ -617,49 +602,7 @@ static bool validate_once(const Opts &o, Rank &r, ucc_coll_req_h req,
                ct == UCC_COLL_TYPE_BCAST ? r.bufs.src : r.bufs.dst,
                bytes);
     }
-    auto expect_allreduce = [&](size_t i) {
-        float s = 0;
-        for (int k = 0; k < nranks; k++) {
-            s += (float)(k + 1) + 0.25f * (float)(i % 7);
-        }
-        return s;
-    };
-    size_t bad = 0;
-    if (ct == UCC_COLL_TYPE_ALLREDUCE) {
-        for (size_t i = 0; i < count; i++) {
-            if (out[i] != expect_allreduce(i)) {
-                bad++;
-            }
-        }
-    } else if (ct == UCC_COLL_TYPE_ALLGATHER) {
-        size_t per = count / nranks;
-        for (size_t i = 0; i < per * nranks; i++) {
-            int    src = (int)(i / per);
-            size_t li  = i % per;
-            float  e = (float)(src + 1) + 0.5f * (float)(li % 5);
-            if (out[i] != e) {
-                bad++;
-            }
-        }
-    } else if (ct == UCC_COLL_TYPE_REDUCE_SCATTER) {
-        size_t per = count / nranks;
-        for (size_t i = 0; i < per; i++) {
-            size_t gi = (size_t)rank * per + i;
-            float  e  = 0;
-            for (int k = 0; k < nranks; k++) {
-                e += (float)(k + 1) + 0.5f * (float)(gi % 5);
-            }
-            if (out[i] != e) {
-                bad++;
-            }
-        }
-    } else if (ct == UCC_COLL_TYPE_BCAST) {
-        for (size_t i = 0; i < count; i++) {
-            if (out[i] != (float)(i % 97)) {
-                bad++;
-            }
-        }
-    }
+    size_t bad = vcheck(ct, rank, nranks, count, out.data());
     if (bad) {
         fprintf(stderr, "[rank %d] VALIDATION FAILED: %zu/%zu bad at %zu "
                         "bytes\n", rank, bad, count, bytes);
