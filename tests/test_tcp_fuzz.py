@@ -1,0 +1,125 @@
+"""Seeded randomized collective-sequence stress over the tcp mesh:
+random collective types, sizes spanning every score band (so bruck /
+sparbit / knomial / linear / ring / SRA / DBT / halving all interleave
+on the same per-team tag space), random roots — each op validated
+against numpy. Catches cross-algorithm tag/sequence interactions that
+single-collective tests cannot."""
+
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+FUZZ = r"""
+import sys
+import numpy as np
+sys.path.insert(0, %r)
+from ucc_amd import core, dtypes
+from ucc_amd.testing import LocalJob
+
+seed = int(sys.argv[1])
+n = 5
+job = LocalJob(n)
+rng = np.random.default_rng(seed)
+
+for opno in range(60):
+    coll = rng.choice(["allreduce", "allgather", "bcast", "reduce",
+                       "reduce_scatter", "alltoall", "barrier"])
+    # sizes span the latency/mid/bandwidth bands (elements)
+    per = int(rng.choice([1, 7, 333, 4096, 20000, 70000]))
+    root = int(rng.integers(0, n))
+    if coll == "barrier":
+        reqs = job.coll("barrier", [
+            dict(src=0, dst=0, count=0, dt=dtypes.INT8)
+            for _ in range(n)])
+        job.run(reqs)
+        continue
+    if coll == "allreduce":
+        arrs = [(rng.random(per) - 0.5).astype(np.float32)
+                for _ in range(n)]
+        outs = [np.zeros(per, np.float32) for _ in range(n)]
+        reqs = job.coll("allreduce", [
+            dict(src=arrs[r].ctypes.data, dst=outs[r].ctypes.data,
+                 count=per, dt=dtypes.FLOAT32) for r in range(n)])
+        job.run(reqs)
+        exp = np.sum(arrs, axis=0)
+        for o in outs:
+            np.testing.assert_allclose(o, exp, rtol=1e-5, atol=1e-5,
+                                       err_msg=f"op{opno}")
+    elif coll == "allgather":
+        blks = [rng.standard_normal(per).astype(np.float32)
+                for _ in range(n)]
+        dsts = [np.zeros(per * n, np.float32) for _ in range(n)]
+        reqs = job.coll("allgather", [
+            dict(src=blks[r].ctypes.data, dst=dsts[r].ctypes.data,
+                 count=per * n, dt=dtypes.FLOAT32) for r in range(n)])
+        job.run(reqs)
+        exp = np.concatenate(blks)
+        for d in dsts:
+            np.testing.assert_array_equal(d, exp, err_msg=f"op{opno}")
+    elif coll == "bcast":
+        bufs = [np.zeros(per, np.float64) for _ in range(n)]
+        bufs[root][:] = rng.random(per)
+        exp = bufs[root].copy()
+        reqs = job.coll("bcast", [
+            dict(src=b.ctypes.data, dst=0, count=per,
+                 dt=dtypes.FLOAT64, root=root) for b in bufs])
+        job.run(reqs)
+        for b in bufs:
+            np.testing.assert_array_equal(b, exp, err_msg=f"op{opno}")
+    elif coll == "reduce":
+        srcs = [(rng.random(per) - 0.5).astype(np.float32)
+                for _ in range(n)]
+        dst = np.zeros(per, np.float32)
+        reqs = job.coll("reduce", [
+            dict(src=srcs[r].ctypes.data,
+                 dst=dst.ctypes.data if r == root else 0, count=per,
+                 dt=dtypes.FLOAT32, root=root) for r in range(n)])
+        job.run(reqs)
+        np.testing.assert_allclose(dst, np.sum(srcs, axis=0),
+                                   rtol=1e-5, atol=1e-5,
+                                   err_msg=f"op{opno}")
+    elif coll == "reduce_scatter":
+        srcs = [(rng.random(per * n) - 0.5).astype(np.float32)
+                for _ in range(n)]
+        dsts = [np.zeros(per, np.float32) for _ in range(n)]
+        reqs = job.coll("reduce_scatter", [
+            dict(src=srcs[r].ctypes.data, dst=dsts[r].ctypes.data,
+                 count=per, dt=dtypes.FLOAT32) for r in range(n)])
+        job.run(reqs)
+        exp = np.sum(srcs, axis=0)
+        for r in range(n):
+            np.testing.assert_allclose(
+                dsts[r], exp[r * per:(r + 1) * per], rtol=1e-5,
+                atol=1e-5, err_msg=f"op{opno}")
+    elif coll == "alltoall":
+        srcs = [rng.standard_normal(per * n).astype(np.float32)
+                for _ in range(n)]
+        dsts = [np.zeros(per * n, np.float32) for _ in range(n)]
+        reqs = job.coll("alltoall", [
+            dict(src=srcs[r].ctypes.data, dst=dsts[r].ctypes.data,
+                 count=per * n, dt=dtypes.FLOAT32) for r in range(n)])
+        job.run(reqs)
+        for d in range(n):
+            for s in range(n):
+                np.testing.assert_array_equal(
+                    dsts[d][s * per:(s + 1) * per],
+                    srcs[s][d * per:(d + 1) * per],
+                    err_msg=f"op{opno}")
+print("FUZZ_OK")
+""" % (REPO,)
+
+
+@pytest.mark.parametrize("seed", [11, 42])
+def test_tcp_random_sequence(seed):
+    env = dict(os.environ)
+    env["UCC_TL_SHM_ENABLE"] = "0"
+    p = subprocess.run([sys.executable, "-c", FUZZ, str(seed)],
+                       env=env, capture_output=True, text=True,
+                       timeout=600)
+    sys.stdout.write(p.stdout[-1500:])
+    sys.stderr.write(p.stderr[-2500:])
+    assert p.returncode == 0 and "FUZZ_OK" in p.stdout
