@@ -123,3 +123,18 @@ def test_tcp_random_sequence(seed):
     sys.stdout.write(p.stdout[-1500:])
     sys.stderr.write(p.stderr[-2500:])
     assert p.returncode == 0 and "FUZZ_OK" in p.stdout
+
+
+def test_shm_socket_random_sequence():
+    """Same randomized sequence with the shm TL serving and the
+    socket-relay staging active (pseudo-sockets) — slot/parity/relay
+    interleavings across mixed collectives."""
+    env = dict(os.environ)
+    env["UCC_FAKE_SOCKET_SPLIT"] = "3"
+    env["UCC_TL_SHM_CHUNK_SIZE"] = "65536"
+    p = subprocess.run([sys.executable, "-c", FUZZ, "7"],
+                       env=env, capture_output=True, text=True,
+                       timeout=600)
+    sys.stdout.write(p.stdout[-1500:])
+    sys.stderr.write(p.stderr[-2500:])
+    assert p.returncode == 0 and "FUZZ_OK" in p.stdout
