@@ -138,3 +138,19 @@ def test_shm_socket_random_sequence():
     sys.stdout.write(p.stdout[-1500:])
     sys.stderr.write(p.stderr[-2500:])
     assert p.returncode == 0 and "FUZZ_OK" in p.stdout
+
+
+def test_hier_random_sequence():
+    """Randomized sequence over the hier composition (fake 2-node
+    split): RAB allreduce, 2step bcast/reduce and the hierarchical
+    barrier interleave with flat tcp colls on shared sub-team
+    sequence spaces."""
+    env = dict(os.environ)
+    env["UCC_FAKE_NODE_SPLIT"] = "2"
+    env["UCC_TL_SHM_ENABLE"] = "0"
+    p = subprocess.run([sys.executable, "-c", FUZZ, "23"],
+                       env=env, capture_output=True, text=True,
+                       timeout=600)
+    sys.stdout.write(p.stdout[-1500:])
+    sys.stderr.write(p.stderr[-2500:])
+    assert p.returncode == 0 and "FUZZ_OK" in p.stdout
