@@ -179,3 +179,4 @@ build/tsan_mt: $(BUILD)/tsan/tests/test_mt.o $(TSAN_OBJS)
 .PHONY: tsan
 tsan: build/tsan_mt
 	TSAN_OPTIONS="halt_on_error=1" ./build/tsan_mt
+	TSAN_OPTIONS="halt_on_error=1" UCC_TL_SHM_ENABLE=0 ./build/tsan_mt

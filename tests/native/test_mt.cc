@@ -137,37 +137,87 @@ int main()
         });
     }
 
+    auto run2 = [&](ucc_coll_args_t (&aa)[2], ucc_team_h *teams) {
+        ucc_coll_req_h rq[2];
+        for (int r = 0; r < n; r++) {
+            CHECK(ucc_collective_init(&aa[r], &rq[r], teams[r]) ==
+                  UCC_OK);
+            CHECK(ucc_collective_post(rq[r]) >= 0);
+        }
+        while (ucc_collective_test(rq[0]) == UCC_INPROGRESS ||
+               ucc_collective_test(rq[1]) == UCC_INPROGRESS) {
+            std::this_thread::yield();
+        }
+        CHECK(ucc_collective_test(rq[0]) == UCC_OK);
+        CHECK(ucc_collective_test(rq[1]) == UCC_OK);
+        ucc_collective_finalize(rq[0]);
+        ucc_collective_finalize(rq[1]);
+    };
+    /* rotate collective types and sizes so the MT run crosses score
+     * bands (knomial/ring/SRA/etc.) instead of hammering one path */
     auto drive = [&](ucc_team_h *teams, float base) {
         for (int it = 0; it < 40; it++) {
-            const size_t       cnt = 700;
+            const size_t cnt = (it % 3 == 2) ? 20000 : 700;
             std::vector<float> s0(cnt, base + it), s1(cnt, base - it);
             std::vector<float> d0(cnt), d1(cnt);
-            ucc_coll_req_h     rq[2];
             std::vector<float> *sb[2] = {&s0, &s1}, *db[2] = {&d0, &d1};
-            for (int r = 0; r < n; r++) {
-                ucc_coll_args_t a{};
-                a.coll_type         = UCC_COLL_TYPE_ALLREDUCE;
-                a.op                = UCC_OP_SUM;
-                a.src.info.buffer   = sb[r]->data();
-                a.src.info.count    = cnt;
-                a.src.info.datatype = UCC_DT_FLOAT32;
-                a.src.info.mem_type = UCC_MEMORY_TYPE_HOST;
-                a.dst.info          = a.src.info;
-                a.dst.info.buffer   = db[r]->data();
-                CHECK(ucc_collective_init(&a, &rq[r], teams[r]) == UCC_OK);
-                CHECK(ucc_collective_post(rq[r]) >= 0);
+            ucc_coll_args_t aa[2]{};
+            switch (it % 4) {
+            case 0:
+            case 2: { /* allreduce (small + SRA-band sizes) */
+                for (int r = 0; r < n; r++) {
+                    aa[r].coll_type         = UCC_COLL_TYPE_ALLREDUCE;
+                    aa[r].op                = UCC_OP_SUM;
+                    aa[r].src.info.buffer   = sb[r]->data();
+                    aa[r].src.info.count    = cnt;
+                    aa[r].src.info.datatype = UCC_DT_FLOAT32;
+                    aa[r].src.info.mem_type = UCC_MEMORY_TYPE_HOST;
+                    aa[r].dst.info          = aa[r].src.info;
+                    aa[r].dst.info.buffer   = db[r]->data();
+                }
+                run2(aa, teams);
+                for (size_t i = 0; i < cnt; i++) {
+                    CHECK(d0[i] == 2 * base && d1[i] == 2 * base);
+                }
+                break;
             }
-            while (ucc_collective_test(rq[0]) == UCC_INPROGRESS ||
-                   ucc_collective_test(rq[1]) == UCC_INPROGRESS) {
-                std::this_thread::yield();
+            case 1: { /* bcast from rank 0 */
+                for (int r = 0; r < n; r++) {
+                    aa[r].coll_type         = UCC_COLL_TYPE_BCAST;
+                    aa[r].root              = 0;
+                    aa[r].src.info.buffer   = sb[r]->data();
+                    aa[r].src.info.count    = cnt;
+                    aa[r].src.info.datatype = UCC_DT_FLOAT32;
+                    aa[r].src.info.mem_type = UCC_MEMORY_TYPE_HOST;
+                }
+                run2(aa, teams);
+                for (size_t i = 0; i < cnt; i++) {
+                    CHECK(s1[i] == base + it); /* got rank 0's value */
+                }
+                break;
             }
-            CHECK(ucc_collective_test(rq[0]) == UCC_OK);
-            CHECK(ucc_collective_test(rq[1]) == UCC_OK);
-            for (size_t i = 0; i < cnt; i++) {
-                CHECK(d0[i] == 2 * base && d1[i] == 2 * base);
+            case 3: { /* allgather (own half + peer half) */
+                size_t per = cnt / 2;
+                for (int r = 0; r < n; r++) {
+                    aa[r].coll_type         = UCC_COLL_TYPE_ALLGATHER;
+                    aa[r].src.info.buffer   = sb[r]->data();
+                    aa[r].src.info.count    = per;
+                    aa[r].src.info.datatype = UCC_DT_FLOAT32;
+                    aa[r].src.info.mem_type = UCC_MEMORY_TYPE_HOST;
+                    aa[r].dst.info          = aa[r].src.info;
+                    aa[r].dst.info.count    = per * 2;
+                    aa[r].dst.info.buffer   = db[r]->data();
+                }
+                run2(aa, teams);
+                for (size_t i = 0; i < per; i++) {
+                    CHECK(d0[i] == base + it &&
+                          d0[per + i] == base - it);
+                    CHECK(d1[i] == base + it &&
+                          d1[per + i] == base - it);
+                }
+                break;
             }
-            ucc_collective_finalize(rq[0]);
-            ucc_collective_finalize(rq[1]);
+            }
         }
     };
     std::thread a([&] { drive(t1, 100.f); });
