@@ -274,21 +274,38 @@ size_t coll_args_msgsize(const ucc_coll_args_t &args, uint32_t rank,
     case UCC_COLL_TYPE_REDUCE_SCATTER:
     case UCC_COLL_TYPE_ALLTOALL:
     case UCC_COLL_TYPE_ALLGATHER:
-    case UCC_COLL_TYPE_GATHER:
-    case UCC_COLL_TYPE_SCATTER:
         /* total bytes moved through dst */
-        if (args.coll_type == UCC_COLL_TYPE_SCATTER ||
-            (args.flags & UCC_COLL_ARGS_FLAG_IN_PLACE &&
-             args.coll_type != UCC_COLL_TYPE_ALLGATHER)) {
+        if (args.flags & UCC_COLL_ARGS_FLAG_IN_PLACE &&
+            args.coll_type != UCC_COLL_TYPE_ALLGATHER) {
             return args.src.info.count * ucc_dt_size(args.src.info.datatype);
         }
         return args.dst.info.count * ucc_dt_size(args.dst.info.datatype);
+    case UCC_COLL_TYPE_GATHER:
+        /* selection MUST be rank-invariant (all ranks must pick the
+         * same algorithm): root passes the gathered TOTAL, leaves
+         * their equal block — scale the leaf view by team size */
+        if (rank == (uint32_t)args.root) {
+            return args.dst.info.count *
+                   ucc_dt_size(args.dst.info.datatype);
+        }
+        return args.src.info.count *
+               ucc_dt_size(args.src.info.datatype) * size;
+    case UCC_COLL_TYPE_SCATTER:
+        if (rank == (uint32_t)args.root) {
+            return args.src.info.count *
+                   ucc_dt_size(args.src.info.datatype);
+        }
+        return args.dst.info.count *
+               ucc_dt_size(args.dst.info.datatype) * size;
     case UCC_COLL_TYPE_ALLGATHERV:
         return counts_total(args, args.dst.info_v.counts, size) *
                ucc_dt_size(args.dst.info_v.datatype);
     case UCC_COLL_TYPE_GATHERV:
         /* v-args are significant at the root only (UCC semantics):
-         * non-roots size by their own contiguous src */
+         * non-roots size by their own contiguous src. NOTE: this is
+         * rank-divergent for ragged tables; keep every gatherv
+         * algorithm registered over the FULL size range (currently
+         * true: single linear task per transport). */
         if (rank != (uint32_t)args.root) {
             return args.src.info.count *
                    ucc_dt_size(args.src.info.datatype);
@@ -296,8 +313,14 @@ size_t coll_args_msgsize(const ucc_coll_args_t &args, uint32_t rank,
         return counts_total(args, args.dst.info_v.counts, size) *
                ucc_dt_size(args.dst.info_v.datatype);
     case UCC_COLL_TYPE_ALLTOALLV:
-        return counts_total(args, args.src.info_v.counts, size) *
-               ucc_dt_size(args.src.info_v.datatype);
+        /* the per-rank count ROWS differ across ranks, so any local
+         * byte sum is rank-DIVERGENT — ranks in different score bands
+         * would run different algorithms against each other and
+         * deadlock (caught by the randomized-sequence stress under
+         * the hier composition). Selection therefore uses a constant;
+         * size-adaptive behavior lives INSIDE the algorithms (hybrid
+         * per-pair threshold, cdna4 exchanged global max/sum). */
+        return 0;
     case UCC_COLL_TYPE_SCATTERV:
         if (rank != (uint32_t)args.root) {
             return args.dst.info.count *

@@ -27,7 +27,9 @@ rng = np.random.default_rng(seed)
 
 for opno in range(60):
     coll = rng.choice(["allreduce", "allgather", "bcast", "reduce",
-                       "reduce_scatter", "alltoall", "barrier"])
+                       "reduce_scatter", "alltoall", "barrier",
+                       "gather", "scatter", "alltoallv",
+                       "reduce_scatterv"])
     # sizes span the latency/mid/bandwidth bands (elements)
     per = int(rng.choice([1, 7, 333, 4096, 20000, 70000]))
     root = int(rng.integers(0, n))
@@ -109,6 +111,78 @@ for opno in range(60):
                     dsts[d][s * per:(s + 1) * per],
                     srcs[s][d * per:(d + 1) * per],
                     err_msg=f"op{opno}")
+    elif coll == "gather":
+        srcs = [rng.standard_normal(per).astype(np.float32)
+                for _ in range(n)]
+        gdst = np.zeros(per * n, np.float32)
+        reqs = job.coll("gather", [
+            dict(src=srcs[r].ctypes.data,
+                 dst=gdst.ctypes.data if r == root else 0,
+                 count=per * n if r == root else per,
+                 dt=dtypes.FLOAT32, root=root) for r in range(n)])
+        job.run(reqs)
+        np.testing.assert_array_equal(gdst, np.concatenate(srcs),
+                                      err_msg=f"op{opno}")
+    elif coll == "scatter":
+        big = rng.standard_normal(per * n).astype(np.float32)
+        sdst = [np.zeros(per, np.float32) for _ in range(n)]
+        reqs = job.coll("scatter", [
+            dict(src=big.ctypes.data if r == root else 0,
+                 dst=sdst[r].ctypes.data,
+                 count=per * n if r == root else per,
+                 dt=dtypes.FLOAT32, root=root) for r in range(n)])
+        job.run(reqs)
+        for r in range(n):
+            np.testing.assert_array_equal(
+                sdst[r], big[r * per:(r + 1) * per],
+                err_msg=f"op{opno}")
+    elif coll == "alltoallv":
+        scnt = [[int(rng.integers(0, max(per // 2, 2)))
+                 for _ in range(n)] for _ in range(n)]
+        rcnt = [[scnt[src][dd] for src in range(n)]
+                for dd in range(n)]
+        sdsp = [np.cumsum([0] + row[:-1]).tolist() for row in scnt]
+        rdsp = [np.cumsum([0] + row[:-1]).tolist() for row in rcnt]
+        stot = [max(sum(row), 1) for row in scnt]
+        rtot = [max(sum(row), 1) for row in rcnt]
+        srcs = [rng.standard_normal(stot[r]).astype(np.float32)
+                for r in range(n)]
+        dsts = [np.zeros(rtot[r], np.float32) for r in range(n)]
+        reqs = job.coll("alltoallv", [
+            dict(src=srcs[r].ctypes.data, dst=dsts[r].ctypes.data,
+                 count=0, dt=dtypes.FLOAT32,
+                 src_counts=scnt[r], src_displs=sdsp[r],
+                 dst_counts=rcnt[r], dst_displs=rdsp[r])
+            for r in range(n)])
+        job.run(reqs)
+        for dd in range(n):
+            for src in range(n):
+                cq = rcnt[dd][src]
+                np.testing.assert_array_equal(
+                    dsts[dd][rdsp[dd][src]:rdsp[dd][src] + cq],
+                    srcs[src][sdsp[src][dd]:sdsp[src][dd] + cq],
+                    err_msg=f"op{opno}")
+    elif coll == "reduce_scatterv":
+        cnts = [int(rng.integers(0, per + 1)) for _ in range(n)]
+        tot = max(sum(cnts), 1)
+        dsp = np.cumsum([0] + cnts[:-1]).tolist()
+        srcs = [(rng.random(tot) - 0.5).astype(np.float32)
+                for _ in range(n)]
+        dsts = [np.zeros(max(cnts[r], 1), np.float32)
+                for r in range(n)]
+        reqs = job.coll("reduce_scatterv", [
+            dict(src=srcs[r].ctypes.data, dst=dsts[r].ctypes.data,
+                 count=cnts[r], dt=dtypes.FLOAT32,
+                 dst_counts=cnts, dst_displs=dsp)
+            for r in range(n)])
+        job.run(reqs)
+        exp = np.sum(srcs, axis=0)
+        for r in range(n):
+            if cnts[r]:
+                np.testing.assert_allclose(
+                    dsts[r][:cnts[r]],
+                    exp[dsp[r]:dsp[r] + cnts[r]], rtol=1e-5,
+                    atol=1e-5, err_msg=f"op{opno}")
 print("FUZZ_OK")
 """ % (REPO,)
 
