@@ -2357,9 +2357,12 @@ void add_scores(Team *team)
      * rab_dev) */
     team->score_map.add(UCC_COLL_TYPE_BCAST, UCC_MEMORY_TYPE_CUDA, b);
 
-    /* node-aggregated a2av pays off while per-pair messages are small:
-     * nleaders^2 sockets instead of nranks^2. Above the threshold the
-     * flat pairwise path wins (no triple copy of big payloads). */
+    /* node-aggregated a2av: nleaders^2 sockets instead of nranks^2.
+     * NOTE: alltoallv SELECTION sizing is rank-invariant (constant) —
+     * per-rank row sums diverge across ranks and mixed selections
+     * deadlock — so this range knob acts as an on/off switch
+     * (0 disables); the small-vs-large adaptivity lives in the flat
+     * hybrid's per-pair threshold when this path is off. */
     size_t a2av_thresh = Config::instance().get_size(
         "CL_HIER", "A2AV_NODE_THRESH", 256 * 1024);
     if (a2av_thresh > 0) {
