@@ -24,6 +24,17 @@ seed = int(sys.argv[1])
 n = 5
 job = LocalJob(n)
 rng = np.random.default_rng(seed)
+c = core()
+
+# a persistent allreduce re-posted throughout the sequence (slot and
+# tag-space reuse interleaved with every other algorithm)
+P = 900
+psrc = [np.zeros(P, np.float32) for _ in range(n)]
+pdst = [np.zeros(P, np.float32) for _ in range(n)]
+preq = job.coll("allreduce", [
+    dict(src=psrc[r].ctypes.data, dst=pdst[r].ctypes.data, count=P,
+         dt=dtypes.FLOAT32, flags=c.FLAG_PERSISTENT)
+    for r in range(n)])
 
 for opno in range(60):
     coll = rng.choice(["allreduce", "allgather", "bcast", "reduce",
@@ -39,25 +50,53 @@ for opno in range(60):
             for _ in range(n)])
         job.run(reqs)
         continue
+    if opno %% 9 == 4:  # interleave a persistent re-post
+        for r in range(n):
+            psrc[r][:] = rng.random(P).astype(np.float32) + opno
+        pexp = np.sum(psrc, axis=0)
+        job.run(preq)
+        for o in pdst:
+            np.testing.assert_allclose(o, pexp, rtol=1e-5, atol=1e-4,
+                                       err_msg=f"persistent@op{opno}")
     if coll == "allreduce":
+        inplace = bool(rng.integers(0, 2))
         arrs = [(rng.random(per) - 0.5).astype(np.float32)
                 for _ in range(n)]
-        outs = [np.zeros(per, np.float32) for _ in range(n)]
-        reqs = job.coll("allreduce", [
-            dict(src=arrs[r].ctypes.data, dst=outs[r].ctypes.data,
-                 count=per, dt=dtypes.FLOAT32) for r in range(n)])
-        job.run(reqs)
         exp = np.sum(arrs, axis=0)
+        if inplace:
+            reqs = job.coll("allreduce", [
+                dict(src=0, dst=arrs[r].ctypes.data, count=per,
+                     dt=dtypes.FLOAT32, flags=c.FLAG_IN_PLACE)
+                for r in range(n)])
+            job.run(reqs)
+            outs = arrs
+        else:
+            outs = [np.zeros(per, np.float32) for _ in range(n)]
+            reqs = job.coll("allreduce", [
+                dict(src=arrs[r].ctypes.data, dst=outs[r].ctypes.data,
+                     count=per, dt=dtypes.FLOAT32)
+                for r in range(n)])
+            job.run(reqs)
         for o in outs:
             np.testing.assert_allclose(o, exp, rtol=1e-5, atol=1e-5,
                                        err_msg=f"op{opno}")
     elif coll == "allgather":
+        inplace = bool(rng.integers(0, 2))
         blks = [rng.standard_normal(per).astype(np.float32)
                 for _ in range(n)]
         dsts = [np.zeros(per * n, np.float32) for _ in range(n)]
-        reqs = job.coll("allgather", [
-            dict(src=blks[r].ctypes.data, dst=dsts[r].ctypes.data,
-                 count=per * n, dt=dtypes.FLOAT32) for r in range(n)])
+        if inplace:
+            for r in range(n):
+                dsts[r][r * per:(r + 1) * per] = blks[r]
+            reqs = job.coll("allgather", [
+                dict(src=0, dst=dsts[r].ctypes.data, count=per * n,
+                     dt=dtypes.FLOAT32, flags=c.FLAG_IN_PLACE)
+                for r in range(n)])
+        else:
+            reqs = job.coll("allgather", [
+                dict(src=blks[r].ctypes.data, dst=dsts[r].ctypes.data,
+                     count=per * n, dt=dtypes.FLOAT32)
+                for r in range(n)])
         job.run(reqs)
         exp = np.concatenate(blks)
         for d in dsts:
