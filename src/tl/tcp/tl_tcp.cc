@@ -3955,7 +3955,7 @@ class TcpTl final : public Tl {
                     "sliding-window allreduce windows in flight");
         cfg.declare("TL_TCP", "AG_BRUCK_MAX", "64k",
                     "Bruck allgather upper bound bytes");
-        cfg.declare("TL_TCP", "AG_SPARBIT_MAX", "512k",
+        cfg.declare("TL_TCP", "AG_SPARBIT_MAX", "256k",
                     "sparbit (data-ordered log-round) allgather upper "
                     "bound bytes");
         cfg.declare("TL_TCP", "AG_NEIGHBOR_MIN", "256k",
@@ -3966,7 +3966,7 @@ class TcpTl final : public Tl {
                     "directions (full-duplex links)");
         cfg.declare("TL_TCP", "SOCKBUF", "4m",
                     "SO_SNDBUF/SO_RCVBUF bytes (0 = kernel default)");
-        cfg.declare("TL_TCP", "AG_LINEAR_MIN", "512k",
+        cfg.declare("TL_TCP", "AG_LINEAR_MIN", "256k",
                     "linear direct allgather lower bound bytes");
         cfg.declare("TL_TCP", "AG_LINEAR_MAX", "0",
                     "linear direct allgather upper bound bytes "
@@ -4300,11 +4300,12 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
         /* sparbit role: log-round AND data-ordered (no work buffer, no
          * rotation) — preferred over bruck in the small-block band */
         size_t smax = Config::instance().get_size(
-            "TL_TCP", "AG_SPARBIT_MAX", 512 * 1024);
+            "TL_TCP", "AG_SPARBIT_MAX", 256 * 1024);
         ScoreRange r;
         r.start    = 0;
         r.end      = smax;
-        r.score    = sc + 2; /* wins 64k-512k in the measured sweep */
+        r.score    = sc + 2; /* wins 64k-256k in the measured sweep
+                                (crossover re-checked after SOCKBUF) */
         r.tl_name  = "tcp";
         r.alg_name = "sparbit";
         r.init     = [self](const ucc_coll_args_t &args, Team *t2,
@@ -4329,7 +4330,7 @@ void TcpTlTeam::get_scores(Team *team, ScoreMap &map)
         /* linear direct: one hop per block, all peers in flight —
          * wins the band between the log-round and ring regimes */
         size_t lmin = Config::instance().get_size(
-            "TL_TCP", "AG_LINEAR_MIN", 512 * 1024);
+            "TL_TCP", "AG_LINEAR_MIN", 256 * 1024);
         size_t lmax = Config::instance().get_size(
             "TL_TCP", "AG_LINEAR_MAX", 0); /* 0 = unbounded */
         if (lmax == 0) {

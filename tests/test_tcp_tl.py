@@ -397,7 +397,7 @@ def test_allgather_sparbit():
         "for n in (2, 3, 5, 6, 8, 11):\n"
         "    job = LocalJob(n)\n"
         "    smap = core().score_map_str(job.teams[0])\n"
-        "    assert 'allgather:host:0-524288:@tcp/sparbit' in smap, smap\n"
+        "    assert 'allgather:host:0-262144:@tcp/sparbit' in smap, smap\n"
         "    rng = np.random.default_rng(n)\n"
         "    for per in (1, 77, 4000):\n"
         "        blks = [rng.standard_normal(per).astype(np.float32)\n"
