@@ -21,7 +21,7 @@ from ucc_amd import core, dtypes
 from ucc_amd.testing import LocalJob
 
 seed = int(sys.argv[1])
-n = 5
+n = [2, 3, 4, 6, 7, 8][seed %% 6]  # cover folds/even-n/tree shapes
 job = LocalJob(n)
 rng = np.random.default_rng(seed)
 c = core()
